@@ -112,7 +112,8 @@ def lib():
                                           C.c_void_p, C.c_int64]
         _lib.sno_decode.restype = C.c_int32
         _lib.sno_decode.argtypes = [C.c_int32, C.c_void_p, C.c_int64, C.c_int32,
-                                    C.c_void_p, C.POINTER(C.c_int32), C.POINTER(C.c_uint8)]
+                                    C.c_void_p, C.c_int64,
+                                    C.POINTER(C.c_int32), C.POINTER(C.c_uint8)]
     return _lib
 
 
@@ -232,10 +233,11 @@ def decode(dtype, blob, count):
     b = np.frombuffer(blob, dtype=np.uint8)
     valid = np.zeros(count, dtype=np.uint8)
     if dtype == T_STRING:
-        out = (C.c_uint8 * max(1, len(blob) * 2))()
+        cap = max(1, len(blob) * 2 + count * 256)
+        out = (C.c_uint8 * cap)()
         lens = np.zeros(count, dtype=np.int32)
         rc = L.sno_decode(dtype, b.ctypes.data_as(C.c_void_p), len(blob), count,
-                          out, lens.ctypes.data_as(C.POINTER(C.c_int32)),
+                          out, cap, lens.ctypes.data_as(C.POINTER(C.c_int32)),
                           valid.ctypes.data_as(C.POINTER(C.c_uint8)))
         assert rc == 0, rc
         vals, off = [], 0
@@ -249,7 +251,7 @@ def decode(dtype, blob, count):
         return vals, valid
     arr = np.zeros(count, dtype=_NP_OF_T[dtype])
     rc = L.sno_decode(dtype, b.ctypes.data_as(C.c_void_p), len(blob), count,
-                      arr.ctypes.data_as(C.c_void_p), None,
+                      arr.ctypes.data_as(C.c_void_p), arr.nbytes, None,
                       valid.ctypes.data_as(C.POINTER(C.c_uint8)))
     assert rc == 0, rc
     return arr, valid

@@ -1108,10 +1108,12 @@ SNO_EXPORT int64_t sno_encode(int32_t dtype, int32_t encoding, int32_t nullable,
       }
       wb_i32(&w, dn);
       for (int32_t j = 0; j < dn; j++) { wb_i32(&w, dlen[j]); wb_bytes(&w, v8 + doff[j], dlen[j]); }
+      /* index array holds only NON-NULL entries (NullableEncoder.writeIsNull
+       * sets the bit and writes nothing, ColumnEncoding.scala:1239-1253) */
       for (int32_t i = 0; i < count; i++) {
-        int32_t ix = idx[i] < 0 ? dn : idx[i];   /* null == numElements */
-        if (big) wb_i32(&w, ix);
-        else { int16_t x = (int16_t)ix; wb_bytes(&w, &x, 2); }
+        if (idx[i] < 0) continue;
+        if (big) wb_i32(&w, idx[i]);
+        else { int16_t x = (int16_t)idx[i]; wb_bytes(&w, &x, 2); }
       }
       free(doff); free(dlen); free(idx);
     } else if (dtype == SN_TYPE_INT32 || dtype == SN_TYPE_INT64) {
@@ -1133,9 +1135,9 @@ SNO_EXPORT int64_t sno_encode(int32_t dtype, int32_t encoding, int32_t nullable,
         else wb_bytes(&w, &dv[j], 8);
       }
       for (int32_t i = 0; i < count; i++) {
-        int32_t ix = idx[i] < 0 ? dn : idx[i];
-        if (big) wb_i32(&w, ix);
-        else { int16_t x = (int16_t)ix; wb_bytes(&w, &x, 2); }
+        if (idx[i] < 0) continue;   /* nulls write no index entry */
+        if (big) wb_i32(&w, idx[i]);
+        else { int16_t x = (int16_t)idx[i]; wb_bytes(&w, &x, 2); }
       }
       free(dv); free(idx);
     } else return SN_ERR_UNSUPPORTED;
@@ -1230,7 +1232,8 @@ SNO_EXPORT int64_t sno_encode_stats(int32_t ncols, const int32_t *dtypes,
  * Decode helper for round-trip tests
  * ======================================================================= */
 SNO_EXPORT int32_t sno_decode(int32_t dtype, const uint8_t *blob, int64_t len,
-    int32_t count, void *out_values, int32_t *out_str_lens, uint8_t *out_valid) {
+    int32_t count, void *out_values, int64_t out_cap, int32_t *out_str_lens,
+    uint8_t *out_valid) {
   sno_dec d;
   int rc = dec_init(&d, dtype, blob, len, 0, NULL, NULL);
   if (rc != SN_OK) return rc;
@@ -1246,6 +1249,7 @@ SNO_EXPORT int32_t sno_decode(int32_t dtype, const uint8_t *blob, int64_t len,
       rc = dec_read_string(&d, nnp, &s, &sl);
       if (rc != SN_OK) { dec_free(&d); return rc; }
       if (s == NULL) { if (out_valid) out_valid[ord] = 0; if (out_str_lens) out_str_lens[ord] = 0; continue; }
+      if (so + sl > out_cap) { dec_free(&d); return SN_ERR_NOMEM; }
       memcpy(ov + so, s, (size_t)sl); so += sl;
       if (out_str_lens) out_str_lens[ord] = sl;
     } else if (dtype == SN_TYPE_DOUBLE) {
