@@ -1,0 +1,358 @@
+/*
+ * builder.cpp — product-side batch builder + seeded TPC-H lineitem generator.
+ *
+ * This is the engine's equivalent of the reference's ingest rollover path
+ * (ColumnBatchCreator.createAndStoreBatch, ColumnBatchCreator.scala:46-131 +
+ * ColumnInsertExec encoding calls): raw row data -> reference-format column
+ * blobs (header ColumnEncoding.scala:37-53; uncompressed bodies
+ * Uncompressed.scala:32-160; string dictionaries DictionaryEncoding.scala:
+ * 85-160 with the int16 -> int32 BigDictionary promotion at Short.MaxValue
+ * :313,338) + a ColumnStatsSchema stats UnsafeRow (ColumnEncoding.scala:
+ * 1015-1036) -> sn_batch_put.
+ *
+ * Encoder choice mirrors ColumnEncoding.getColumnEncoder (:838-870):
+ * dictionary for strings, uncompressed otherwise.
+ *
+ * The lineitem generator is the dbgen-equivalent synthetic source mandated by
+ * BASELINE.md (no network: seeded, reproducible; distributions chosen to
+ * match TPC-H Q6 selectivity ~1.9% — SURVEY.md §8(d)).
+ */
+#include <cstring>
+#include <string>
+#include <thread>
+#include <vector>
+#include <map>
+#include <atomic>
+#include <cstdio>
+
+#include "../../include/snappy_engine.h"
+
+extern "C" int32_t sn_batch_put(sn_engine *, int32_t, int64_t, int32_t, int32_t,
+                                const sn_buf *, const sn_buf *, const sn_buf *,
+                                const sn_buf *);
+
+/* ---------------- little-endian writers ---------------- */
+static inline void put_i32(std::vector<uint8_t> &b, int32_t v) {
+  const uint8_t *p = (const uint8_t *)&v; b.insert(b.end(), p, p + 4);
+}
+static inline void put_i64(std::vector<uint8_t> &b, int64_t v) {
+  const uint8_t *p = (const uint8_t *)&v; b.insert(b.end(), p, p + 8);
+}
+static inline void put_f64(std::vector<uint8_t> &b, double v) {
+  const uint8_t *p = (const uint8_t *)&v; b.insert(b.end(), p, p + 8);
+}
+static inline void put_i16(std::vector<uint8_t> &b, int16_t v) {
+  const uint8_t *p = (const uint8_t *)&v; b.insert(b.end(), p, p + 2);
+}
+
+typedef struct { const void *data; const int32_t *str_lens; const uint8_t *valid; } sn_ingest_col;
+
+/* header: typeId + null bitset (ColumnEncoding.scala:37-53) */
+static void put_header(std::vector<uint8_t> &out, int32_t type_id,
+                       const uint8_t *valid, int32_t count) {
+  put_i32(out, type_id);
+  int32_t nnull = 0;
+  if (valid) for (int32_t i = 0; i < count; i++) nnull += !valid[i];
+  if (!nnull) { put_i32(out, 0); return; }
+  int32_t words = (count + 63) >> 6;
+  put_i32(out, words * 8);
+  size_t base = out.size();
+  out.resize(base + (size_t)words * 8, 0);
+  for (int32_t i = 0; i < count; i++)
+    if (!valid[i]) out[base + (i >> 6) * 8 + ((i & 63) >> 3)] |= (uint8_t)(1u << (i & 7));
+}
+
+/* encode one column (uncompressed numerics / dictionary strings) */
+static int encode_column(sn_type_t dtype, const sn_ingest_col &col,
+                         int32_t count, std::vector<uint8_t> &out) {
+  const uint8_t *v8 = (const uint8_t *)col.data;
+  const uint8_t *valid = col.valid;
+  if (dtype == SN_TYPE_STRING) {
+    /* two-pass dictionary build in first-occurrence order; promote to
+     * BigDictionary when entries reach Short.MaxValue */
+    std::map<std::string, int32_t> didx;
+    std::vector<std::string> dict;
+    std::vector<int32_t> idx(count, -1);
+    int64_t so = 0;
+    for (int32_t i = 0; i < count; i++) {
+      int32_t L = col.str_lens[i];
+      if (valid && !valid[i]) { so += L; continue; }
+      std::string s((const char *)v8 + so, (size_t)L);
+      so += L;
+      auto it = didx.find(s);
+      if (it == didx.end()) {
+        idx[i] = (int32_t)dict.size();
+        didx.emplace(std::move(s), idx[i]);
+        dict.push_back(std::string((const char *)v8 + so - L, (size_t)L));
+      } else idx[i] = it->second;
+    }
+    bool big = dict.size() >= 32767;
+    put_header(out, big ? SN_ENC_BIG_DICTIONARY : SN_ENC_DICTIONARY, valid, count);
+    put_i32(out, (int32_t)dict.size());
+    for (auto &s : dict) { put_i32(out, (int32_t)s.size());
+      out.insert(out.end(), s.begin(), s.end()); }
+    for (int32_t i = 0; i < count; i++) {
+      if (idx[i] < 0) continue;          /* writeIsNull writes no index */
+      if (big) put_i32(out, idx[i]); else put_i16(out, (int16_t)idx[i]);
+    }
+    return SN_OK;
+  }
+  put_header(out, SN_ENC_UNCOMPRESSED, valid, count);
+  int w;
+  switch (dtype) {
+    case SN_TYPE_DOUBLE: case SN_TYPE_INT64: w = 8; break;
+    case SN_TYPE_INT32: case SN_TYPE_FLOAT: w = 4; break;
+    case SN_TYPE_INT16: w = 2; break;
+    case SN_TYPE_INT8: case SN_TYPE_BOOL: w = 1; break;
+    default: return SN_ERR_UNSUPPORTED;
+  }
+  if (!valid) {
+    out.insert(out.end(), v8, v8 + (size_t)count * w);
+  } else {
+    for (int32_t i = 0; i < count; i++)
+      if (valid[i]) out.insert(out.end(), v8 + (size_t)i * w, v8 + (size_t)(i + 1) * w);
+  }
+  return SN_OK;
+}
+
+/* stats row writer (UnsafeRow: ColumnStatsSchema) */
+static void encode_stats(const std::vector<sn_type_t> &dtypes,
+                         int32_t batch_count_signed,
+                         const std::vector<double> &lo_d, const std::vector<double> &hi_d,
+                         const std::vector<int64_t> &lo_i, const std::vector<int64_t> &hi_i,
+                         const std::vector<int32_t> &ncount,
+                         const std::vector<uint8_t> &has_bounds,
+                         std::vector<uint8_t> &out) {
+  int nc = (int)dtypes.size();
+  int32_t num_fields = nc * 3 + 1;
+  int32_t nwords = (num_fields + 63) >> 6;
+  out.assign((size_t)nwords * 8 + (size_t)num_fields * 8, 0);
+  uint8_t *bits = out.data();
+  uint8_t *slots = out.data() + (size_t)nwords * 8;
+  auto set_bit = [&](int f) { bits[(f >> 6) * 8 + ((f & 63) >> 3)] |= (uint8_t)(1u << (f & 7)); };
+  memcpy(slots, &batch_count_signed, 4);
+  for (int c = 0; c < nc; c++) {
+    int f_lo = 1 + c * 3, f_hi = 2 + c * 3, f_nc = 3 + c * 3;
+    memcpy(slots + (size_t)f_nc * 8, &ncount[c], 4);
+    if (!has_bounds[c]) { set_bit(f_lo); set_bit(f_hi); continue; }
+    switch (dtypes[c]) {
+      case SN_TYPE_DOUBLE: case SN_TYPE_FLOAT:
+        memcpy(slots + (size_t)f_lo * 8, &lo_d[c], 8);
+        memcpy(slots + (size_t)f_hi * 8, &hi_d[c], 8); break;
+      case SN_TYPE_INT64:
+        memcpy(slots + (size_t)f_lo * 8, &lo_i[c], 8);
+        memcpy(slots + (size_t)f_hi * 8, &hi_i[c], 8); break;
+      default: {
+        int32_t lo = (int32_t)lo_i[c], hi = (int32_t)hi_i[c];
+        memcpy(slots + (size_t)f_lo * 8, &lo, 4);
+        memcpy(slots + (size_t)f_hi * 8, &hi, 4);
+      }
+    }
+  }
+}
+
+/* schema/shard accessors implemented in engine.cpp */
+struct TableInfoProbe { int32_t ncols; sn_type_t dtypes[64]; uint8_t nullable[64]; };
+extern "C" int32_t sn_table_schema(sn_engine *e, int32_t table, TableInfoProbe *out);
+extern "C" void sn_engine_shard(sn_engine *e, int32_t *rank, int32_t *count);
+
+/* ingest raw row data: split into batches, encode, put.
+ * Returns rows ingested (on this shard) or negative error. */
+extern "C" int64_t sn_ingest_columns(sn_engine *e, int32_t table, int64_t nrows,
+                                     const sn_ingest_col *cols, int32_t batch_rows,
+                                     int32_t first_bucket) {
+  TableInfoProbe ti;
+  if (sn_table_schema(e, table, &ti) != SN_OK) return SN_ERR_BADARG;
+  if (batch_rows <= 0) batch_rows = 200000;
+  int nc = ti.ncols;
+  int64_t put_rows = 0;
+  std::vector<int64_t> str_off(nc, 0);
+  for (int64_t s = 0, bi = 0; s < nrows; s += batch_rows, bi++) {
+    int32_t n = (int32_t)std::min<int64_t>(batch_rows, nrows - s);
+    std::vector<std::vector<uint8_t>> blobs(nc);
+    std::vector<sn_buf> bufs(nc);
+    std::vector<sn_type_t> dtypes(nc);
+    std::vector<double> lo_d(nc, 0), hi_d(nc, 0);
+    std::vector<int64_t> lo_i(nc, 0), hi_i(nc, 0);
+    std::vector<int32_t> ncount(nc, 0);
+    std::vector<uint8_t> hb(nc, 0);
+    for (int c = 0; c < nc; c++) {
+      dtypes[c] = ti.dtypes[c];
+      sn_ingest_col view = cols[c];
+      const uint8_t *base = (const uint8_t *)cols[c].data;
+      if (ti.dtypes[c] == SN_TYPE_STRING) {
+        view.data = base + str_off[c];
+        view.str_lens = cols[c].str_lens + s;
+        int64_t bytes = 0;
+        for (int32_t i = 0; i < n; i++) bytes += cols[c].str_lens[s + i];
+        str_off[c] += bytes;
+      } else {
+        int w = (ti.dtypes[c] == SN_TYPE_DOUBLE || ti.dtypes[c] == SN_TYPE_INT64) ? 8 :
+                (ti.dtypes[c] == SN_TYPE_INT16) ? 2 :
+                (ti.dtypes[c] == SN_TYPE_INT8 || ti.dtypes[c] == SN_TYPE_BOOL) ? 1 : 4;
+        view.data = base + (int64_t)s * w;
+      }
+      if (cols[c].valid) view.valid = cols[c].valid + s;
+      int rc = encode_column(ti.dtypes[c], view, n, blobs[c]);
+      if (rc != SN_OK) return rc;
+      /* stats: min/max over non-null values */
+      bool first = true;
+      for (int32_t i = 0; i < n; i++) {
+        if (view.valid && !view.valid[i]) { ncount[c]++; continue; }
+        switch (ti.dtypes[c]) {
+          case SN_TYPE_DOUBLE: {
+            double v = ((const double *)view.data)[i];
+            if (first || v < lo_d[c]) lo_d[c] = v;
+            if (first || v > hi_d[c]) hi_d[c] = v;
+            hb[c] = 1; first = false; break;
+          }
+          case SN_TYPE_INT32: {
+            int32_t v = ((const int32_t *)view.data)[i];
+            if (first || v < lo_i[c]) lo_i[c] = v;
+            if (first || v > hi_i[c]) hi_i[c] = v;
+            hb[c] = 1; first = false; break;
+          }
+          case SN_TYPE_INT64: {
+            int64_t v = ((const int64_t *)view.data)[i];
+            if (first || v < lo_i[c]) lo_i[c] = v;
+            if (first || v > hi_i[c]) hi_i[c] = v;
+            hb[c] = 1; first = false; break;
+          }
+          default: break;   /* strings/other: no bounds in stats (round 1) */
+        }
+      }
+      bufs[c].data = blobs[c].data();
+      bufs[c].len = (int64_t)blobs[c].size();
+    }
+    std::vector<uint8_t> stats;
+    encode_stats(dtypes, n, lo_d, hi_d, lo_i, hi_i, ncount, hb, stats);
+    sn_buf sbuf = { stats.data(), (int64_t)stats.size() };
+    int32_t rc = sn_batch_put(e, table, bi, first_bucket + (int32_t)bi, n,
+                              bufs.data(), &sbuf, nullptr, nullptr);
+    if (rc != SN_OK) return rc;
+    put_rows += n;
+  }
+  return put_rows;
+}
+
+/* ---------------- seeded lineitem generator ---------------- */
+static inline uint64_t splitmix64(uint64_t x) {
+  x += 0x9E3779B97f4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+static int days_from_civil(int y, int m, int d) {
+  y -= m <= 2;
+  int era = (y >= 0 ? y : y - 399) / 400;
+  unsigned yoe = (unsigned)(y - era * 400);
+  unsigned doy = (153u * (unsigned)(m + (m > 2 ? -3 : 9)) + 2) / 5 + (unsigned)d - 1;
+  unsigned doe = yoe * 365 + yoe / 4 - yoe / 100 + doy;
+  return era * 146097 + (int)doe - 719468;
+}
+
+extern "C" void sn_gen_lineitem_arrays(int64_t start_row, int32_t n, int64_t seed,
+                                       double *qty, double *ep, double *disc,
+                                       double *tax, uint8_t *rf, uint8_t *ls,
+                                       int32_t *ship) {
+  const int ship_lo = days_from_civil(1992, 1, 1);
+  const int ship_hi = days_from_civil(1998, 12, 1);
+  static const char RF[3] = { 'A', 'N', 'R' };
+  static const char LS[2] = { 'F', 'O' };
+  for (int32_t i = 0; i < n; i++) {
+    uint64_t r = (uint64_t)(start_row + i);
+    uint64_t a = splitmix64(r * 2 + 1 + (uint64_t)seed * 0x100000001ull);
+    uint64_t b = splitmix64(r * 2 + 2 + (uint64_t)seed * 0x100000001ull);
+    qty[i] = (double)(1 + (a % 50));
+    ep[i] = (double)(90000 + ((a >> 8) % 10410000)) / 100.0;
+    disc[i] = (double)((a >> 32) % 11) / 100.0;
+    tax[i] = (double)((a >> 40) % 9) / 100.0;
+    rf[i] = (uint8_t)RF[(b >> 3) % 3];
+    ls[i] = (uint8_t)LS[(b >> 5) % 2];
+    ship[i] = ship_lo + (int32_t)((b >> 16) % (uint64_t)(ship_hi - ship_lo));
+  }
+}
+
+/* generate + ingest `total_rows` of lineitem into `table` (7-column hot-path
+ * projection: qty, ep, disc, tax, returnflag, linestatus, shipdate).
+ * Batches are deterministic in (seed, batch index) so every shard layout
+ * produces identical data; a batch's bucket = its index, and sn_batch_put
+ * keeps only buckets owned by this shard.  Multithreaded generation+encode.
+ * Returns rows resident on this shard. */
+extern "C" int64_t sn_datagen_lineitem(sn_engine *e, int32_t table,
+                                       int64_t total_rows, int64_t seed,
+                                       int32_t batch_rows, int32_t nthreads) {
+  TableInfoProbe ti;
+  if (sn_table_schema(e, table, &ti) != SN_OK) return SN_ERR_BADARG;
+  if (ti.ncols != 7) return SN_ERR_BADARG;
+  if (batch_rows <= 0) batch_rows = 600000;   /* ~24 MB across 7 columns */
+  if (nthreads <= 0) nthreads = (int32_t)std::thread::hardware_concurrency();
+  int64_t nbatches = (total_rows + batch_rows - 1) / batch_rows;
+  std::atomic<int64_t> next(0), put_rows(0);
+  std::atomic<int32_t> err(SN_OK);
+  std::vector<sn_type_t> dtypes(ti.dtypes, ti.dtypes + 7);
+
+  /* which buckets this shard owns is decided inside sn_batch_put; we probe
+   * it cheaply here to skip generating foreign batches */
+  auto worker = [&]() {
+    std::vector<double> qty(batch_rows), ep(batch_rows), disc(batch_rows), tax(batch_rows);
+    std::vector<uint8_t> rf(batch_rows), ls(batch_rows);
+    std::vector<int32_t> ship(batch_rows);
+    std::vector<int32_t> len1(batch_rows, 1);
+    int32_t sh_rank, sh_count;
+    sn_engine_shard(e, &sh_rank, &sh_count);
+    while (true) {
+      int64_t bi = next.fetch_add(1);
+      if (bi >= nbatches || err.load() != SN_OK) break;
+      /* skip batches another shard owns (bucket = batch index) */
+      if (sh_count > 1 && (int32_t)(bi % sh_count) != sh_rank) continue;
+      int32_t n = (int32_t)std::min<int64_t>(batch_rows, total_rows - bi * batch_rows);
+      int64_t rows_before = bi * batch_rows;
+      sn_gen_lineitem_arrays(rows_before, n, seed, qty.data(), ep.data(),
+                             disc.data(), tax.data(), rf.data(), ls.data(),
+                             ship.data());
+      /* encode the 7 columns */
+      std::vector<std::vector<uint8_t>> blobs(7);
+      sn_ingest_col c;
+      c.valid = nullptr; c.str_lens = nullptr;
+      c.data = qty.data();  encode_column(SN_TYPE_DOUBLE, c, n, blobs[0]);
+      c.data = ep.data();   encode_column(SN_TYPE_DOUBLE, c, n, blobs[1]);
+      c.data = disc.data(); encode_column(SN_TYPE_DOUBLE, c, n, blobs[2]);
+      c.data = tax.data();  encode_column(SN_TYPE_DOUBLE, c, n, blobs[3]);
+      c.data = rf.data();   c.str_lens = len1.data();
+      encode_column(SN_TYPE_STRING, c, n, blobs[4]);
+      c.data = ls.data();
+      encode_column(SN_TYPE_STRING, c, n, blobs[5]);
+      c.str_lens = nullptr;
+      c.data = ship.data(); encode_column(SN_TYPE_INT32, c, n, blobs[6]);
+      /* stats */
+      std::vector<double> lo_d(7, 0), hi_d(7, 0);
+      std::vector<int64_t> lo_i(7, 0), hi_i(7, 0);
+      std::vector<int32_t> ncnt(7, 0);
+      std::vector<uint8_t> hb(7, 0);
+      auto mm_d = [&](int ci, const double *v) {
+        double lo = v[0], hi = v[0];
+        for (int32_t i = 1; i < n; i++) { if (v[i] < lo) lo = v[i]; if (v[i] > hi) hi = v[i]; }
+        lo_d[ci] = lo; hi_d[ci] = hi; hb[ci] = 1;
+      };
+      mm_d(0, qty.data()); mm_d(1, ep.data()); mm_d(2, disc.data()); mm_d(3, tax.data());
+      int32_t slo = ship[0], shi = ship[0];
+      for (int32_t i = 1; i < n; i++) { if (ship[i] < slo) slo = ship[i]; if (ship[i] > shi) shi = ship[i]; }
+      lo_i[6] = slo; hi_i[6] = shi; hb[6] = 1;
+      std::vector<uint8_t> stats;
+      encode_stats(dtypes, n, lo_d, hi_d, lo_i, hi_i, ncnt, hb, stats);
+      sn_buf bufs[7], sbuf = { stats.data(), (int64_t)stats.size() };
+      for (int i = 0; i < 7; i++) { bufs[i].data = blobs[i].data(); bufs[i].len = (int64_t)blobs[i].size(); }
+      int32_t rc = sn_batch_put(e, table, bi, (int32_t)bi, n, bufs, &sbuf,
+                                nullptr, nullptr);
+      if (rc != SN_OK) { err.store(rc); break; }
+      put_rows += n;
+    }
+  };
+  std::vector<std::thread> ths;
+  for (int i = 0; i < nthreads; i++) ths.emplace_back(worker);
+  for (auto &th : ths) th.join();
+  if (err.load() != SN_OK) return err.load();
+  return put_rows.load();
+}

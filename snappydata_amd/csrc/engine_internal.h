@@ -1,0 +1,98 @@
+/*
+ * engine_internal.h — device-visible descriptors shared between the host
+ * runtime (engine.cpp) and the HIP kernels (kernels.hip).
+ *
+ * The engine stores column batches in HBM in the reference's encoded byte
+ * format and decodes INSIDE the scan kernel.  The host
+ * pre-parses each blob's header once at sn_batch_put and keeps these
+ * descriptors so the kernel never re-parses headers:
+ *   - body pointer (fixed-width array / dictionary index array)
+ *   - null bitset words + a per-64-row prefix-sum of null counts, so
+ *     nonNullPosition = prefix[row/64] + popcount(word & mask) is O(1) per
+ *     row on the GPU (the reference decodes sequentially on the JVM —
+ *     NullableDecoder, ColumnEncoding.scala:1069-1135 — which has no
+ *     data-parallel analogue, hence the auxiliary prefix array)
+ *   - delete mask re-expressed as a row bitmap
+ *   - update deltas (ColumnDeltaDecoder/UpdatedColumnDecoder semantics)
+ *     merged on host into a per-column sorted patch list applied by the
+ *     kernel during the scan (deltas are small by construction:
+ *     ColumnMaxDeltaRows=10000, Literals.scala:138-146)
+ */
+#ifndef SN_ENGINE_INTERNAL_H
+#define SN_ENGINE_INTERNAL_H
+
+#include <stdint.h>
+
+#define SN_DEV_MAX_COLS 8      /* referenced columns per plan */
+
+/* column value kinds as seen by the kernel */
+enum {
+  SN_K_F64 = 0, SN_K_I32 = 1, SN_K_I64 = 2, SN_K_F32 = 3,
+  SN_K_DICT16 = 4,   /* int16 dictionary index (group cols) */
+  SN_K_DICT32 = 5,   /* int32 dictionary index (BigDictionary) */
+  SN_K_I16 = 6, SN_K_BOOLBIT = 7
+};
+
+typedef struct {
+  const void     *body;        /* fixed-width values / dict index array */
+  const uint64_t *nullw;       /* null bitset words (NULL if none) */
+  const uint32_t *nullpfx;     /* nulls in words [0, w) — host-built */
+  const int32_t  *dictmap;     /* group col: local dict idx -> premultiplied
+                                  global group contribution */
+  /* update-delta patches (host-merged, sorted by row) */
+  const uint64_t *patch_bm;    /* bitmap: row has patch */
+  const int32_t  *patch_pos;   /* sorted patched rows */
+  const double   *patch_val;   /* patched value (f64 or int bits in .i64) */
+  const uint64_t *patch_nullbm;/* bitmap over patch index: patch writes NULL */
+  int32_t patch_n;
+  int32_t kind;
+  int32_t has_nulls;
+  int32_t null_gid;            /* group col: premultiplied global id of NULL */
+} sn_dev_col;
+
+typedef struct {
+  int32_t num_rows;
+  int32_t clean;               /* 1: no nulls/deletes/patches on any referenced
+                                  col and all plain fixed-width -> fast path */
+  const uint64_t *del_bm;      /* delete bitmap (NULL if none) */
+  sn_dev_col cols[SN_DEV_MAX_COLS];
+} sn_dev_batch;
+
+typedef struct { int32_t batch; int32_t row_start; } sn_dev_tile;
+
+typedef struct {
+  int32_t cslot;               /* index into sn_dev_batch.cols */
+  int32_t is_double;
+  double  lo_d, hi_d;
+  int64_t lo_i, hi_i;
+  int32_t has_lo, has_hi, lo_strict, hi_strict;
+} sn_dev_pred;
+
+typedef struct { int32_t cslot; double add, mul; } sn_dev_factor;
+typedef struct { int32_t kind; int32_t nf; sn_dev_factor f[3]; } sn_dev_agg;
+
+typedef struct {
+  int32_t npreds, naggs, ngroup, nslots;
+  int32_t gcol[2];             /* cslot of group columns (ngroup entries) */
+  sn_dev_pred preds[8];
+  sn_dev_agg  aggs[12];
+} sn_dev_plan;
+
+#define SN_TILE_ROWS 4096      /* rows per workgroup tile (256 thr x 16) */
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* launches the scan+filter+aggregate over all tiles.
+ * out: device array [nslots][naggs+1] doubles (last = group row count),
+ * zeroed by caller.  Returns hipError_t as int. */
+int sn_launch_scan_agg(const sn_dev_plan *plan,
+                       const sn_dev_batch *dev_batches,
+                       const sn_dev_tile *dev_tiles, int32_t ntiles,
+                       double *dev_out, void *stream);
+
+#ifdef __cplusplus
+}
+#endif
+#endif

@@ -1,0 +1,289 @@
+"""Python host layer over the MI355X engine (libsnappy_engine.so).
+
+Mirrors the reference's operator/data surfaces a drop-in must honor:
+  - Engine.table_define  -> ColumnFormatRelation metadata
+  - Engine.batch_put     -> ExternalStore.storeColumnBatch
+                            (ExternalStore.scala:43-45) / the
+                            ColumnBatchIterator buffer protocol
+                            (ColumnBatchIterator.scala:96-163)
+  - Engine.query         -> ColumnTableScan -> Filter ->
+                            SnappyHashAggregateExec execution
+  - Query.partials/merge -> the partial->final exchange (ShuffleExchange
+                            between SnappyHashAggregateExec stages)
+
+The compute path is the HIP engine; this layer is plumbing.  There is NO CPU
+fallback: querying without a GPU raises EngineError(SN_ERR_NOGPU).
+"""
+import ctypes as C
+import os
+import subprocess
+
+import numpy as np
+
+from . import abi
+
+_DIR = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_DIR, "libsnappy_engine.so")
+
+
+class EngineError(RuntimeError):
+    def __init__(self, code, msg=""):
+        super().__init__(f"engine error {code}: {msg}")
+        self.code = code
+
+
+def build(force=False):
+    """Compile the engine .so in-tree (hipcc --offload-arch=gfx950)."""
+    srcdir = os.path.join(_DIR, "csrc")
+    srcs = [os.path.join(srcdir, f) for f in
+            ("engine.cpp", "builder.cpp", "kernels.hip", "engine_internal.h")]
+    if force or (not os.path.exists(_SO)) or \
+            os.path.getmtime(_SO) < max(os.path.getmtime(s) for s in srcs):
+        subprocess.run(["make", "-C", srcdir], check=True)
+    return _SO
+
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(_SO):
+            build()
+        L = C.CDLL(_SO)
+        L.sn_engine_create.restype = C.c_void_p
+        L.sn_engine_create.argtypes = [C.POINTER(abi.SnConfig)]
+        L.sn_engine_destroy.argtypes = [C.c_void_p]
+        L.sn_last_error.restype = C.c_char_p
+        L.sn_engine_arch.restype = C.c_char_p
+        L.sn_table_define.restype = C.c_int32
+        L.sn_table_define.argtypes = [C.c_void_p, C.c_char_p, C.c_int32,
+                                      C.POINTER(abi.SnColSchema)]
+        L.sn_batch_put.restype = C.c_int32
+        L.sn_batch_put.argtypes = [C.c_void_p, C.c_int32, C.c_int64, C.c_int32,
+                                   C.c_int32, C.POINTER(abi.SnBuf),
+                                   C.POINTER(abi.SnBuf), C.POINTER(abi.SnBuf),
+                                   C.POINTER(abi.SnBuf)]
+        L.sn_table_num_batches.restype = C.c_int64
+        L.sn_table_num_batches.argtypes = [C.c_void_p, C.c_int32]
+        L.sn_table_num_rows.restype = C.c_int64
+        L.sn_table_num_rows.argtypes = [C.c_void_p, C.c_int32]
+        L.sn_table_get_blob.restype = C.c_int64
+        L.sn_table_get_blob.argtypes = [C.c_void_p, C.c_int32, C.c_int32,
+                                        C.c_int32, C.c_void_p, C.c_int64]
+        L.sn_query_submit.restype = C.c_void_p
+        L.sn_query_submit.argtypes = [C.c_void_p, C.POINTER(abi.SnPlan)]
+        L.sn_query_wait.restype = C.c_int32
+        L.sn_query_wait.argtypes = [C.c_void_p]
+        L.sn_query_result.restype = C.c_int32
+        L.sn_query_result.argtypes = [C.c_void_p, C.POINTER(abi.SnResult)]
+        L.sn_query_destroy.argtypes = [C.c_void_p]
+        L.sn_query_partial_bytes.restype = C.c_int64
+        L.sn_query_partial_bytes.argtypes = [C.c_void_p]
+        L.sn_query_partials.restype = C.c_int32
+        L.sn_query_partials.argtypes = [C.c_void_p, C.c_void_p, C.c_int32]
+        L.sn_query_merge.restype = C.c_int32
+        L.sn_query_merge.argtypes = [C.c_void_p, C.c_void_p, C.c_int64, C.c_int32]
+        L.sn_ingest_columns.restype = C.c_int64
+        L.sn_ingest_columns.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
+                                        C.POINTER(abi.SnIngestCol), C.c_int32,
+                                        C.c_int32]
+        L.sn_datagen_lineitem.restype = C.c_int64
+        L.sn_datagen_lineitem.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
+                                          C.c_int64, C.c_int32, C.c_int32]
+        L.sn_gen_lineitem_arrays.argtypes = [
+            C.c_int64, C.c_int32, C.c_int64,
+            C.POINTER(C.c_double), C.POINTER(C.c_double), C.POINTER(C.c_double),
+            C.POINTER(C.c_double), C.POINTER(C.c_uint8), C.POINTER(C.c_uint8),
+            C.POINTER(C.c_int32)]
+        _lib = L
+    return _lib
+
+
+def last_error():
+    return (lib().sn_last_error() or b"").decode()
+
+
+def _check(rc, what=""):
+    if rc < 0:
+        raise EngineError(rc, f"{what}: {last_error()}")
+    return rc
+
+
+def gen_lineitem_arrays(start_row, n, seed):
+    """Generate the synthetic lineitem columns (same generator as the engine's
+    sn_datagen_lineitem — used by tests and bench's CPU-baseline sample)."""
+    qty = np.empty(n); ep = np.empty(n); disc = np.empty(n); tax = np.empty(n)
+    rf = np.empty(n, dtype=np.uint8); ls = np.empty(n, dtype=np.uint8)
+    ship = np.empty(n, dtype=np.int32)
+    lib().sn_gen_lineitem_arrays(
+        start_row, n, seed,
+        qty.ctypes.data_as(C.POINTER(C.c_double)),
+        ep.ctypes.data_as(C.POINTER(C.c_double)),
+        disc.ctypes.data_as(C.POINTER(C.c_double)),
+        tax.ctypes.data_as(C.POINTER(C.c_double)),
+        rf.ctypes.data_as(C.POINTER(C.c_uint8)),
+        ls.ctypes.data_as(C.POINTER(C.c_uint8)),
+        ship.ctypes.data_as(C.POINTER(C.c_int32)))
+    return dict(qty=qty, ep=ep, disc=disc, tax=tax,
+                rf=[bytes([b]) for b in rf], ls=[bytes([b]) for b in ls],
+                ship=ship)
+
+
+class Query:
+    def __init__(self, eng, handle, plan):
+        self._e = eng
+        self._h = handle
+        self.plan = plan
+
+    def wait(self):
+        _check(lib().sn_query_wait(self._h), "query_wait")
+        return self
+
+    def result(self):
+        res = abi.SnResult()
+        _check(lib().sn_query_result(self._h, C.byref(res)), "query_result")
+        return res
+
+    def rows(self):
+        return abi.result_rows(self.result())
+
+    def partial_bytes(self):
+        return _check(lib().sn_query_partial_bytes(self._h))
+
+    def partials_host(self):
+        n = self.partial_bytes()
+        buf = np.zeros(n, dtype=np.uint8)
+        _check(lib().sn_query_partials(self._h, buf.ctypes.data, 0), "partials")
+        return buf
+
+    def partials_into_device(self, dev_ptr):
+        _check(lib().sn_query_partials(self._h, C.c_void_p(dev_ptr), 1), "partials")
+
+    def merge_host(self, blocks, stride, n_blocks):
+        """blocks: numpy uint8 array of n_blocks partial blocks."""
+        _check(lib().sn_query_merge(self._h, blocks.ctypes.data, stride, n_blocks),
+               "merge")
+        return self
+
+    def close(self):
+        if self._h:
+            lib().sn_query_destroy(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class Engine:
+    def __init__(self, device=0, shard_rank=0, shard_count=1, n_buckets=0,
+                 column_batch_size=0):
+        cfg = abi.SnConfig()
+        cfg.device = device
+        cfg.shard_rank = shard_rank
+        cfg.shard_count = shard_count
+        cfg.n_buckets = n_buckets
+        cfg.column_batch_size = column_batch_size
+        self._h = lib().sn_engine_create(C.byref(cfg))
+        if not self._h:
+            raise EngineError(-1, last_error())
+
+    def table_define(self, name, schema):
+        """schema: list of (dtype, nullable)."""
+        arr = (abi.SnColSchema * len(schema))()
+        for i, (dt, nl) in enumerate(schema):
+            arr[i].dtype = dt
+            arr[i].nullable = 1 if nl else 0
+        return _check(lib().sn_table_define(self._h, name.encode(), len(schema), arr),
+                      "table_define")
+
+    def batch_put(self, table, uuid, bucket, num_rows, col_blobs,
+                  stats=None, delete_mask=None, deltas=None):
+        nc = len(col_blobs)
+        bufs = (abi.SnBuf * nc)()
+        keep = []
+        for i, blob in enumerate(col_blobs):
+            a = np.frombuffer(blob, dtype=np.uint8)
+            keep.append(a)
+            bufs[i].data = a.ctypes.data
+            bufs[i].len = len(blob)
+        sb = abi.SnBuf()
+        if stats:
+            a = np.frombuffer(stats, dtype=np.uint8); keep.append(a)
+            sb.data, sb.len = a.ctypes.data, len(stats)
+        db = abi.SnBuf()
+        if delete_mask:
+            a = np.frombuffer(delete_mask, dtype=np.uint8); keep.append(a)
+            db.data, db.len = a.ctypes.data, len(delete_mask)
+        dl = None
+        if deltas is not None:
+            dl = (abi.SnBuf * (nc * 2))()
+            for i, pair in enumerate(deltas):
+                for j, d in enumerate(pair):
+                    if d:
+                        a = np.frombuffer(d, dtype=np.uint8); keep.append(a)
+                        dl[i * 2 + j].data, dl[i * 2 + j].len = a.ctypes.data, len(d)
+        _check(lib().sn_batch_put(self._h, table, uuid, bucket, num_rows, bufs,
+                                  C.byref(sb) if stats else None,
+                                  C.byref(db) if delete_mask else None, dl),
+               "batch_put")
+
+    def num_batches(self, table):
+        return _check(lib().sn_table_num_batches(self._h, table))
+
+    def num_rows(self, table):
+        return _check(lib().sn_table_num_rows(self._h, table))
+
+    def get_blob(self, table, batch, col, cap=1 << 26):
+        buf = np.zeros(cap, dtype=np.uint8)
+        n = _check(lib().sn_table_get_blob(self._h, table, batch, col,
+                                           buf.ctypes.data, cap), "get_blob")
+        return bytes(buf[:n].tobytes())
+
+    def ingest_columns(self, table, cols, nrows, batch_rows=200000, first_bucket=0):
+        """cols: list of dicts {data: np.ndarray|bytes payload, lens: np.int32
+        array (strings), valid: np.uint8 array or None}."""
+        arr = (abi.SnIngestCol * len(cols))()
+        keep = []
+        for i, col in enumerate(cols):
+            data = col["data"]
+            if isinstance(data, (bytes, bytearray)):
+                data = np.frombuffer(data, dtype=np.uint8)
+            data = np.ascontiguousarray(data)
+            keep.append(data)
+            arr[i].data = data.ctypes.data
+            if col.get("lens") is not None:
+                lens = np.ascontiguousarray(col["lens"], dtype=np.int32)
+                keep.append(lens)
+                arr[i].str_lens = lens.ctypes.data_as(C.POINTER(C.c_int32))
+            if col.get("valid") is not None:
+                v = np.ascontiguousarray(col["valid"], dtype=np.uint8)
+                keep.append(v)
+                arr[i].valid = v.ctypes.data_as(C.POINTER(C.c_uint8))
+        return _check(lib().sn_ingest_columns(self._h, table, nrows, arr,
+                                              batch_rows, first_bucket), "ingest")
+
+    def datagen_lineitem(self, table, total_rows, seed=42, batch_rows=0, threads=0):
+        return _check(lib().sn_datagen_lineitem(self._h, table, total_rows, seed,
+                                                batch_rows, threads), "datagen")
+
+    def query(self, plan):
+        h = lib().sn_query_submit(self._h, C.byref(plan))
+        if not h:
+            raise EngineError(-1, last_error())
+        return Query(self, h, plan)
+
+    def close(self):
+        if self._h:
+            lib().sn_engine_destroy(self._h)
+            self._h = None
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
