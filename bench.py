@@ -1,0 +1,273 @@
+#!/usr/bin/env python3
+"""bench.py — the driver-facing benchmark of the MI355X columnar engine.
+
+Workload (BASELINE.json): the hot path scan -> filter -> aggregate on TPC-H
+lineitem.  A "step" is one full pass of the hot path over the resident
+synthetic table (all column batches, one kernel launch).  Default workload at
+N=1 is `tpch_q6_lineitem_sf10` (BASELINE configs[1] — the first
+single-GPU configuration; configs[0] is the reference's CPU-only case).
+Inputs are generated once (seeded, synthetic — no network) and resident in
+HBM before the timed region.
+
+Multi-GPU (--gpus N, launched by torch.distributed.run, one rank per GPU over
+RCCL): column batches shard by bucket across ranks (weak scaling — per-GPU
+rows fixed); the keyless Q6 partial state is exchanged with an RCCL all_reduce
+over xGMI, mirroring the reference's partial->final aggregation exchange.
+
+Output: ONE JSON line from rank 0 per the driver contract, including
+`roofline` (achieved HBM GB/s of the scan kernel from HIP events on its
+launch stream vs the 8 TB/s chip peak) and `cpu_baseline` (the CPU oracle —
+a faithful restatement of the reference's WholeStageCodegen loop — timed on
+the host cores over a bounded sample; the reference itself is JVM-only and
+cannot run here: BASELINE.md).
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+
+from snappydata_amd import abi, engine as se  # noqa: E402
+
+HBM_PEAK_GBPS = 8000.0   # MI355X spec peak (MI355X_MICROARCH.md)
+
+
+def days(y, m, d):
+    import datetime
+    return (datetime.date(y, m, d) - datetime.date(1970, 1, 1)).days
+
+
+LINEITEM_SCHEMA = [(abi.T_DOUBLE, False)] * 4 + [(abi.T_STRING, False)] * 2 + \
+    [(abi.T_INT32, False)]
+COL_QTY, COL_EP, COL_DISC, COL_TAX, COL_RF, COL_LS, COL_SHIP = range(7)
+
+
+def q6_plan(table):
+    return abi.make_plan(
+        table=table,
+        preds=[dict(col=COL_SHIP, lo=days(1994, 1, 1), hi=days(1995, 1, 1),
+                    hi_strict=True),
+               dict(col=COL_DISC, is_double=True, lo=0.05, hi=0.07),
+               dict(col=COL_QTY, is_double=True, hi=24.0, hi_strict=True)],
+        aggs=[("sum", [(COL_EP, 0.0, 1.0), (COL_DISC, 0.0, 1.0)])])
+
+
+def q1_plan(table):
+    cutoff = days(1997, 12, 31) - 90
+    return abi.make_plan(
+        table=table,
+        preds=[dict(col=COL_SHIP, hi=cutoff)],
+        group_cols=[COL_RF, COL_LS],
+        aggs=[("sum", [(COL_QTY, 0.0, 1.0)]),
+              ("sum", [(COL_EP, 0.0, 1.0)]),
+              ("sum", [(COL_EP, 0.0, 1.0), (COL_DISC, 1.0, -1.0)]),
+              ("sum", [(COL_EP, 0.0, 1.0), (COL_DISC, 1.0, -1.0),
+                       (COL_TAX, 1.0, 1.0)]),
+              ("avg", [(COL_QTY, 0.0, 1.0)]),
+              ("avg", [(COL_EP, 0.0, 1.0)]),
+              ("avg", [(COL_DISC, 0.0, 1.0)]),
+              ("count", [])])
+
+
+WORKLOADS = {
+    # name: (plan fn, rows per GPU, algorithmic bytes per row read by the scan)
+    "tpch_q6_lineitem_sf10": (q6_plan, 60_000_000, 28),   # ship4+qty8+ep8+disc8
+    "tpch_q1_lineitem_sf100": (q1_plan, 600_000_000, 40),  # +tax8+2x2dict, ship4
+    "tpch_q1_lineitem_sf10": (q1_plan, 60_000_000, 40),
+}
+
+
+def cpu_baseline_leg(workload, seed, target_seconds=10.0):
+    """Time the CPU oracle (reference-loop restatement, OpenMP over all host
+    cores) on a bounded sample of the same workload.  Returns the dict for
+    the JSON line."""
+    from oracle import pyoracle as po
+    from tests import tpch_util as tu
+
+    sample_rows = 4_000_000
+    d = se.gen_lineitem_arrays(0, sample_rows, seed)
+    t = po.OracleTable(tu.LINEITEM_DTYPES)
+    for num_rows, cols, stats in tu.encode_lineitem_batches(d, 600_000):
+        t.add_batch(num_rows, cols, stats=stats)
+    cores = os.cpu_count()
+    plan = tu.q6_plan() if "q6" in workload else tu.q1_plan()
+    # one calibration pass, then enough reps for ~target_seconds
+    t0 = time.perf_counter()
+    t.query(plan, nthreads=cores)
+    t1 = time.perf_counter()
+    reps = max(1, int(target_seconds / max(1e-3, t1 - t0)))
+    t0 = time.perf_counter()
+    for _ in range(reps):
+        t.query(plan, nthreads=cores)
+    dt = time.perf_counter() - t0
+    return {
+        "value": sample_rows * reps / dt,
+        "unit": "rows/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": f"{sample_rows} rows x {reps} passes ({dt:.1f}s) of the same "
+                  f"synthetic workload through the CPU oracle (OpenMP, "
+                  f"{cores} threads; no JVM available for the reference itself)",
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--workload", default="tpch_q6_lineitem_sf10",
+                    choices=sorted(WORKLOADS))
+    ap.add_argument("--rows", type=int, default=0,
+                    help="override rows per GPU")
+    ap.add_argument("--seed", type=int, default=42)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(world, args.gpus if world == 1 else world)
+
+    dist = None
+    torch = None
+    if world > 1:
+        import torch as _torch
+        import torch.distributed as _dist
+        torch = _torch
+        torch.cuda.set_device(local_rank)
+        _dist.init_process_group("nccl")
+        dist = _dist
+
+    plan_fn, default_rows, bytes_per_row = WORKLOADS[args.workload]
+    rows_per_gpu = args.rows or default_rows
+    total_rows = rows_per_gpu * n_gpus
+
+    eng = se.Engine(device=local_rank, shard_rank=rank, shard_count=world,
+                    n_buckets=max(128, world * 16))
+    t = eng.table_define("lineitem", LINEITEM_SCHEMA)
+    eng.datagen_lineitem(t, total_rows, seed=args.seed, batch_rows=600_000)
+    resident = eng.num_rows(t)
+    plan = plan_fn(t)
+    grouped = plan.ngroup > 0
+
+    exchange_buf = None
+
+    def step():
+        q = eng.query(plan)
+        q.wait()
+        km = q.kernel_ms()
+        if dist is not None:
+            if not grouped:
+                # RCCL all_reduce of the fixed-width keyless partial block
+                nonlocal exchange_buf
+                n = q.partial_bytes()
+                assert n % 8 == 0
+                if exchange_buf is None:
+                    exchange_buf = torch.zeros(n // 8, dtype=torch.float64,
+                                               device=f"cuda:{local_rank}")
+                q.partials_into_device(exchange_buf.data_ptr())
+                dist.all_reduce(exchange_buf)
+                merged = exchange_buf.cpu().numpy().view(np.uint8)
+                q.merge_host(np.ascontiguousarray(merged), n, 1)
+            else:
+                # RCCL all_gather of self-describing grouped partial blocks
+                n = q.partial_bytes()
+                local = torch.from_numpy(q.partials_host()).to(f"cuda:{local_rank}")
+                gathered = torch.zeros(world * n, dtype=torch.uint8,
+                                       device=f"cuda:{local_rank}")
+                dist.all_gather_into_tensor(gathered, local)
+                blocks = gathered.cpu().numpy()
+                q.merge_host(np.ascontiguousarray(blocks), n, world)
+        rows = q.rows()
+        q.close()
+        return rows, km
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+            torch.cuda.synchronize()
+
+    # warmup
+    result = None
+    for _ in range(max(1, args.warmup)):
+        result, _ = step()
+
+    barrier_sync()
+    kms = []
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        result, km = step()
+        if km > 0:
+            kms.append(km)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    if dist is not None:
+        e_t = torch.tensor([elapsed], dtype=torch.float64,
+                           device=f"cuda:{local_rank}")
+        dist.all_reduce(e_t, op=dist.ReduceOp.MAX)
+        elapsed = float(e_t.cpu())
+
+    ms_per_step = elapsed * 1000.0 / args.steps
+    value = total_rows * args.steps / elapsed
+
+    if rank == 0:
+        # roofline: dominant kernel = the single fused scan kernel.
+        # achieved = algorithmic bytes per launch / avg kernel time.
+        roofline = None
+        if kms:
+            avg_ms = float(np.mean(kms))
+            alg_bytes = resident * bytes_per_row     # this rank's launch
+            achieved = alg_bytes / (avg_ms / 1000.0) / 1e9
+            roofline = {
+                "bound": "hbm",
+                "achieved": round(achieved, 1),
+                "peak": HBM_PEAK_GBPS,
+                "unit": "GB/s",
+                "frac": round(achieved / HBM_PEAK_GBPS, 4),
+                "traffic": None,   # PMC FETCH/WRITE via rocprofv3 — profiles/
+                "kernel_ms": round(avg_ms, 4),
+            }
+        cpu_baseline = None
+        if world == 1 and not args.no_cpu_baseline:
+            cpu_baseline = cpu_baseline_leg(args.workload, args.seed)
+
+        line = {
+            "metric": "rows/sec scan+filter+agg (TPC-H lineitem)",
+            "value": round(value, 1),
+            "unit": "rows/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 4),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,     # no absolute published numbers
+            "dtype": "f64",
+            "data": "synthetic",
+            "config": {
+                "workload": args.workload,
+                "rows_per_gpu": rows_per_gpu,
+                "total_rows": total_rows,
+                "bytes_per_row": bytes_per_row,
+                "batch_rows": 600_000,
+                "parallelism": f"bucket-dp{n_gpus}",
+                "result_rows": len(result) if result else 0,
+            },
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+        }
+        print(json.dumps(line))
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

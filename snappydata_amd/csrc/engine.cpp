@@ -617,10 +617,16 @@ struct sn_query {
   std::vector<double> host_out;
   int64_t rows_scanned = 0;             /* host metric: rows in unskipped batches */
   int64_t batches_seen = 0, batches_skipped = 0;
+  hipEvent_t ev_start = nullptr, ev_stop = nullptr;  /* brackets the scan kernel */
+  float kernel_ms = -1.0f;
   bool done = false;
   bool merged = false;
   std::vector<GroupOut> final_groups;
   int status = SN_OK;
+  ~sn_query() {
+    if (ev_start) (void)hipEventDestroy(ev_start);
+    if (ev_stop) (void)hipEventDestroy(ev_stop);
+  }
 };
 
 /* grouped partial-block slot layout (see include/snappy_engine.h) */
@@ -859,15 +865,22 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     void *db_dev = e->arena.alloc(hbatches.size() * sizeof(sn_dev_batch));
     void *tl_dev = e->arena.alloc(htiles.size() * sizeof(sn_dev_tile));
     if (!db_dev || !tl_dev) { fail(SN_ERR_NOMEM, "desc alloc"); return nullptr; }
-    if (hipMemcpyAsync(db_dev, hbatches.data(), hbatches.size() * sizeof(sn_dev_batch),
-                       hipMemcpyHostToDevice, e->stream) != hipSuccess ||
-        hipMemcpyAsync(tl_dev, htiles.data(), htiles.size() * sizeof(sn_dev_tile),
-                       hipMemcpyHostToDevice, e->stream) != hipSuccess) {
+    /* blocking copies: the host vectors are stack-local and die at return */
+    if (hipMemcpy(db_dev, hbatches.data(), hbatches.size() * sizeof(sn_dev_batch),
+                  hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemcpy(tl_dev, htiles.data(), htiles.size() * sizeof(sn_dev_tile),
+                  hipMemcpyHostToDevice) != hipSuccess) {
       fail(SN_ERR_GENERIC, "desc upload"); return nullptr;
     }
+    /* HIP events bracket the scan kernel on ITS stream for the roofline leg
+     * (torch.cuda.Event would only see torch's current stream) */
+    (void)hipEventCreate(&q->ev_start);
+    (void)hipEventCreate(&q->ev_stop);
+    if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
     int rc = sn_launch_scan_agg(&dp, (const sn_dev_batch *)db_dev,
                                 (const sn_dev_tile *)tl_dev, (int32_t)htiles.size(),
                                 q->dev_out, e->stream);
+    if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);
     if (rc != 0) {
       fail(SN_ERR_GENERIC, "kernel launch: %s", hipGetErrorString((hipError_t)rc));
       return nullptr;
@@ -884,9 +897,19 @@ extern "C" int32_t sn_query_wait(sn_query *q) {
     q->host_out.resize(out_n);
     HIP_OR_FAIL(hipMemcpy(q->host_out.data(), q->dev_out, out_n * 8,
                           hipMemcpyDeviceToHost));
+    if (q->ev_start && q->ev_stop)
+      (void)hipEventElapsedTime(&q->kernel_ms, q->ev_start, q->ev_stop);
     q->done = true;
   }
   return SN_OK;
+}
+
+/* scan-kernel duration in ms (HIP events on the launch stream); -1 if the
+ * query launched nothing */
+extern "C" double sn_query_kernel_ms(sn_query *q) {
+  if (!q) return -1.0;
+  (void)sn_query_wait(q);
+  return (double)q->kernel_ms;
 }
 
 /* local accumulators -> GroupOut list (pre-merge view) */

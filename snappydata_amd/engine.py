@@ -79,6 +79,8 @@ def lib():
         L.sn_query_result.restype = C.c_int32
         L.sn_query_result.argtypes = [C.c_void_p, C.POINTER(abi.SnResult)]
         L.sn_query_destroy.argtypes = [C.c_void_p]
+        L.sn_query_kernel_ms.restype = C.c_double
+        L.sn_query_kernel_ms.argtypes = [C.c_void_p]
         L.sn_query_partial_bytes.restype = C.c_int64
         L.sn_query_partial_bytes.argtypes = [C.c_void_p]
         L.sn_query_partials.restype = C.c_int32
@@ -148,6 +150,10 @@ class Query:
 
     def rows(self):
         return abi.result_rows(self.result())
+
+    def kernel_ms(self):
+        """Scan-kernel duration (HIP events on the launch stream)."""
+        return lib().sn_query_kernel_ms(self._h)
 
     def partial_bytes(self):
         return _check(lib().sn_query_partial_bytes(self._h))

@@ -1,0 +1,217 @@
+"""GPU parity tests: the HIP engine vs the CPU oracle on identical inputs,
+including the reference's TPC-H golden vectors replayed from the committed
+fixture (no /root/reference on GPU boxes).
+
+Bars (north star): bit-exact COUNT/integer results; <= 1e-6 relative for
+double SUM/AVG (GPU tree-reduction order differs from the JVM loop's).
+"""
+import os
+
+import numpy as np
+import pytest
+
+from oracle import pyoracle as po
+from snappydata_amd import abi, engine as se
+from tests import tpch_util as tu
+from tests.golden.make_fixtures import read_batches
+
+pytestmark = pytest.mark.gpu
+
+GOLD = os.path.join(os.path.dirname(os.path.abspath(__file__)), "golden")
+REL = 1e-6
+
+
+@pytest.fixture(scope="module")
+def eng():
+    e = se.Engine(device=0)
+    yield e
+    e.close()
+
+
+def assert_close(got, exp, rel=REL):
+    if exp is None or got is None:
+        assert got == exp
+        return
+    assert abs(got - exp) <= rel * max(1.0, abs(exp)), (got, exp)
+
+
+def compare_results(grows, orows, count_aggs=()):
+    assert len(grows) == len(orows), (grows, orows)
+    for (gk, gv), (ok, ov) in zip(grows, orows):
+        assert gk == ok
+        for a, (g, o) in enumerate(zip(gv, ov)):
+            if a in count_aggs:
+                assert g == o, f"agg {a} count mismatch: {g} != {o}"
+            else:
+                assert_close(g, o)
+
+
+def test_config1_sum_where(eng):
+    """BASELINE config 1: SELECT SUM(d) WHERE i > k, 10M rows."""
+    n = 10_000_000
+    rng = np.random.default_rng(42)
+    i32 = rng.integers(0, 10**6, n).astype(np.int32)
+    f64 = rng.random(n)
+    k = int(np.median(i32))
+    t = eng.table_define("c1", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": i32}, {"data": f64}], n, batch_rows=600_000)
+    plan = abi.make_plan(table=t, preds=[dict(col=0, lo=k, lo_strict=True)],
+                         aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    rows = eng.query(plan).rows()
+    m = i32 > k
+    assert rows[0][1][1] == float(m.sum())            # COUNT bit-exact
+    assert_close(rows[0][1][0], float(f64[m].sum()))
+
+
+@pytest.fixture(scope="module")
+def li_fixture_table(eng):
+    batches = read_batches(os.path.join(GOLD, "lineitem_batches.bin"))
+    t = eng.table_define("li_fix", [(abi.T_DOUBLE, False)] * 4 +
+                         [(abi.T_STRING, False)] * 2 + [(abi.T_INT32, False)])
+    for bi, (num_rows, cols, stats) in enumerate(batches):
+        eng.batch_put(t, bi, bi, num_rows, cols, stats=stats)
+    return t
+
+
+def q6_plan_engine(table):
+    return abi.make_plan(
+        table=table,
+        preds=[dict(col=tu.COL_SHIP, lo=tu.days(1994, 1, 1),
+                    hi=tu.days(1995, 1, 1), hi_strict=True),
+               dict(col=tu.COL_DISC, is_double=True, lo=0.05, hi=0.07),
+               dict(col=tu.COL_QTY, is_double=True, hi=24.0, hi_strict=True)],
+        aggs=[("sum", [(tu.COL_EP, 0.0, 1.0), (tu.COL_DISC, 0.0, 1.0)])])
+
+
+def q1_plan_engine(table):
+    cutoff = tu.days(1997, 12, 31) - 90
+    return abi.make_plan(
+        table=table,
+        preds=[dict(col=tu.COL_SHIP, hi=cutoff)],
+        group_cols=[tu.COL_RF, tu.COL_LS],
+        aggs=[("sum", [(tu.COL_QTY, 0.0, 1.0)]),
+              ("sum", [(tu.COL_EP, 0.0, 1.0)]),
+              ("sum", [(tu.COL_EP, 0.0, 1.0), (tu.COL_DISC, 1.0, -1.0)]),
+              ("sum", [(tu.COL_EP, 0.0, 1.0), (tu.COL_DISC, 1.0, -1.0),
+                       (tu.COL_TAX, 1.0, 1.0)]),
+              ("avg", [(tu.COL_QTY, 0.0, 1.0)]),
+              ("avg", [(tu.COL_EP, 0.0, 1.0)]),
+              ("avg", [(tu.COL_DISC, 0.0, 1.0)]),
+              ("count", [])])
+
+
+def test_q6_golden_on_gpu(eng, li_fixture_table):
+    rows = eng.query(q6_plan_engine(li_fixture_table)).rows()
+    assert len(rows) == 1
+    assert tu.fmt(rows[0][1][0]) == tu.GOLDEN_Q6[0]
+
+
+def test_q1_golden_on_gpu(eng, li_fixture_table):
+    rows = eng.query(q1_plan_engine(li_fixture_table)).rows()
+    assert tu.q1_result_lines(rows) == tu.GOLDEN_Q1
+
+
+def test_q6_stats_skip_on_gpu(eng, li_fixture_table):
+    q = eng.query(q6_plan_engine(li_fixture_table))
+    res = q.result()
+    assert res.batches_seen == 8
+    # fixture batches are unclustered; skip count may be 0 — just consistency
+    assert 0 <= res.batches_skipped < res.batches_seen
+
+
+def test_q1_synthetic_vs_oracle(eng):
+    """Engine and oracle on the same synthetic lineitem data (1M rows)."""
+    n = 1_000_000
+    t = eng.table_define("li_syn", [(abi.T_DOUBLE, False)] * 4 +
+                         [(abi.T_STRING, False)] * 2 + [(abi.T_INT32, False)])
+    eng.datagen_lineitem(t, n, seed=42, batch_rows=200_000)
+    assert eng.num_rows(t) == n
+    grows = eng.query(q1_plan_engine(t)).rows()
+
+    d = se.gen_lineitem_arrays(0, n, seed=42)
+    ot = po.OracleTable(tu.LINEITEM_DTYPES)
+    for num_rows, cols, stats in tu.encode_lineitem_batches(d, 200_000):
+        ot.add_batch(num_rows, cols, stats=stats)
+    orows = po.result_rows(ot.query(tu.q1_plan()))
+    compare_results(grows, orows, count_aggs={7})
+
+
+def test_nulls_on_gpu_keyless(eng):
+    """nullable inputs through the general path (null bitset + prefix)."""
+    n = 100_000
+    rng = np.random.default_rng(5)
+    vals = rng.random(n)
+    valid = (rng.random(n) >= 0.3).astype(np.uint8)
+    ivals = rng.integers(0, 1000, n).astype(np.int32)
+    t = eng.table_define("tn", [(abi.T_DOUBLE, True), (abi.T_INT32, False)])
+    eng.ingest_columns(t, [{"data": vals, "valid": valid}, {"data": ivals}],
+                       n, batch_rows=30_000)
+    plan = abi.make_plan(table=t, preds=[dict(col=1, hi=500, hi_strict=True)],
+                         aggs=[("sum", [(0, 0.0, 1.0)]),
+                               ("avg", [(0, 0.0, 1.0)]), ("count", [])])
+    rows = eng.query(plan).rows()
+    m = ivals < 500
+    mv = m & valid.astype(bool)
+    assert rows[0][1][2] == float(m.sum())
+    assert_close(rows[0][1][0], float(vals[mv].sum()))
+    assert_close(rows[0][1][1], float(vals[mv].sum()) / mv.sum())
+
+
+def test_deletes_and_deltas_on_gpu(eng):
+    """delete mask + 2-deep update deltas, engine vs oracle bit-for-bit on
+    integer COUNT and 1e-6 on sums (reference formats via oracle encoders)."""
+    n = 50_000
+    rng = np.random.default_rng(6)
+    f64 = rng.random(n)
+    i32 = rng.integers(0, 100, n).astype(np.int32)
+    cols = [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64),
+            po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, i32)]
+    del_pos = np.unique(rng.integers(0, n, 500)).astype(np.int32)
+    dmask = po.encode_delete(del_pos, n)
+    upd_pos = np.unique(rng.integers(0, n, 800)).astype(np.int32)
+    upd_vals = rng.random(len(upd_pos)) * 10
+    d1 = po.encode_delta(po.T_DOUBLE, po.ENC_UNCOMPRESSED, upd_pos, n, upd_vals)
+    upd2_pos = np.unique(rng.integers(0, n, 400)).astype(np.int32)
+    upd2_vals = rng.random(len(upd2_pos)) * 100
+    d2 = po.encode_delta(po.T_DOUBLE, po.ENC_UNCOMPRESSED, upd2_pos, n, upd2_vals)
+    deltas = [(d1, d2), (None, None)]
+
+    t = eng.table_define("tdd", [(abi.T_DOUBLE, False), (abi.T_INT32, False)])
+    eng.batch_put(t, 0, 0, -n, cols, delete_mask=dmask, deltas=deltas)
+    plan = abi.make_plan(table=t, preds=[dict(col=1, hi=50, hi_strict=True)],
+                         aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(plan).rows()
+
+    ot = po.OracleTable([po.T_DOUBLE, po.T_INT32])
+    ot.add_batch(-n, cols, delete_mask=dmask, deltas=deltas)
+    orows = po.result_rows(ot.query(po.make_plan(
+        preds=[dict(col=1, hi=50, hi_strict=True)],
+        aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])))
+    assert grows[0][1][1] == orows[0][1][1]
+    assert_close(grows[0][1][0], orows[0][1][0])
+
+
+def test_partials_roundtrip_single_gpu(eng, li_fixture_table):
+    """partials export + merge(1 block) == direct result (the exchange path)."""
+    q1 = eng.query(q6_plan_engine(li_fixture_table))
+    direct = q1.rows()
+    q2 = eng.query(q6_plan_engine(li_fixture_table))
+    block = q2.partials_host()
+    q2.merge_host(block, len(block), 1)
+    assert q2.rows() == direct
+
+    q3 = eng.query(q1_plan_engine(li_fixture_table))
+    direct = q3.rows()
+    q4 = eng.query(q1_plan_engine(li_fixture_table))
+    block = q4.partials_host()
+    q4.merge_host(block, len(block), 1)
+    assert q4.rows() == direct
+
+
+def test_empty_result_on_gpu(eng, li_fixture_table):
+    """predicate selecting nothing: SUM -> NULL, COUNT -> 0 (Spark semantics)."""
+    plan = abi.make_plan(table=li_fixture_table,
+                         preds=[dict(col=tu.COL_SHIP, lo=10**6)],
+                         aggs=[("sum", [(tu.COL_EP, 0.0, 1.0)]), ("count", [])])
+    rows = eng.query(plan).rows()
+    assert rows == [((), [None, 0.0])]
