@@ -608,7 +608,8 @@ struct sn_query {
   int cslot_of_col[64];                 /* table col -> device col slot */
   std::vector<int32_t> used_cols;       /* cslot -> table col */
   int nslots = 0;
-  int g1cap = 0, g2cap = 0;             /* per-group-col slot counts (incl null) */
+  int g1cap = 0, g2cap = 0;             /* per-group-col slot counts */
+  int gnull1 = -1, gnull2 = -1;         /* null slot index per group col (-1: none) */
   bool grouped_nonnull_ok = true;
   /* device buffers */
   double *dev_out = nullptr;
@@ -710,15 +711,22 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       if (use_col(c) < 0) { fail(SN_ERR_BADARG, "too many plan columns"); return nullptr; }
     }
 
-  /* group slot space: global dict sizes + a null slot per group col */
+  /* group slot space: global dict sizes, plus a null slot only when the
+   * schema allows null keys (non-nullable key columns waste no slots —
+   * keeps Q1's 3x2 keys in the register-friendly 8x8 kernel) */
   if (plan->ngroup >= 1) {
     int c0 = plan->group_cols[0];
-    q->g1cap = (int)t->gdict[c0].size() + 1;
+    q->g1cap = (int)t->gdict[c0].size() + (t->schema[c0].nullable ? 1 : 0);
     q->g2cap = 1;
     if (plan->ngroup == 2) {
       int c1 = plan->group_cols[1];
-      q->g2cap = (int)t->gdict[c1].size() + 1;
+      q->g2cap = (int)t->gdict[c1].size() + (t->schema[c1].nullable ? 1 : 0);
     }
+    if (t->schema[c0].nullable) q->gnull1 = q->g1cap - 1;
+    if (plan->ngroup == 2 && t->schema[plan->group_cols[1]].nullable)
+      q->gnull2 = q->g2cap - 1;
+    if (q->g1cap == 0) q->g1cap = 1;
+    if (q->g2cap == 0) q->g2cap = 1;
     q->nslots = q->g1cap * q->g2cap;
     if (q->nslots > SN_MAX_GROUP_SLOTS ||
         (q->nslots > 16)) {
@@ -737,6 +745,11 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   memset(&dp, 0, sizeof(dp));
   dp.npreds = plan->npreds; dp.naggs = plan->naggs;
   dp.ngroup = plan->ngroup; dp.nslots = q->nslots;
+  dp.nused = (int32_t)q->used_cols.size();
+  dp.i64_mask = 0;
+  for (size_t ui = 0; ui < q->used_cols.size(); ui++)
+    if (t->schema[q->used_cols[ui]].dtype == SN_TYPE_INT64)
+      dp.i64_mask |= 1u << ui;
   for (int i = 0; i < plan->ngroup; i++) dp.gcol[i] = q->cslot_of_col[plan->group_cols[i]];
   for (int i = 0; i < plan->npreds; i++) {
     const sn_pred &s = plan->preds[i];
@@ -813,8 +826,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
           std::vector<int32_t> map(m.local2global.size() + 1);
           for (size_t i = 0; i < m.local2global.size(); i++)
             map[i] = m.local2global[i] * mul;
-          /* index == numElements denotes null (DictionaryEncoding.scala:90) */
-          int null_gid_local = is_g2 ? (q->g2cap - 1) : (q->g1cap - 1) * q->g2cap;
+          /* index == numElements denotes null (DictionaryEncoding.scala:90);
+           * only nullable key columns reserve a null slot */
+          int gn = is_g2 ? q->gnull2 : q->gnull1;
+          int null_gid_local = gn >= 0 ? (is_g2 ? gn : gn * q->g2cap) : 0;
           map[m.local2global.size()] = null_gid_local;
           dc.null_gid = null_gid_local;
           dc.dictmap = (const int32_t *)up(e, map.data(), map.size() * 4);
@@ -935,12 +950,12 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
     if (p.ngroup >= 1) {
       int c0 = p.group_cols[0];
       int g1 = s / q->g2cap;
-      if (g1 == q->g1cap - 1) g.key_null[0] = true;
+      if (g1 == q->gnull1) g.key_null[0] = true;
       else g.keys[0] = t->gdict[c0][g1];
       if (p.ngroup == 2) {
         int c1 = p.group_cols[1];
         int g2 = s % q->g2cap;
-        if (g2 == q->g2cap - 1) g.key_null[1] = true;
+        if (g2 == q->gnull2) g.key_null[1] = true;
         else g.keys[1] = t->gdict[c1][g2];
       }
     }

@@ -73,6 +73,9 @@ typedef struct { int32_t kind; int32_t nf; sn_dev_factor f[3]; } sn_dev_agg;
 
 typedef struct {
   int32_t npreds, naggs, ngroup, nslots;
+  int32_t nused;               /* referenced columns (cslots 0..nused-1) */
+  uint32_t i64_mask;           /* bit c: cslot c is INT64 (raw-bitcast in LDS;
+                                  predicates compare exactly as int64) */
   int32_t gcol[2];             /* cslot of group columns (ngroup entries) */
   sn_dev_pred preds[8];
   sn_dev_agg  aggs[12];

@@ -5,28 +5,36 @@
  * loop (WholeStageCodegen of ColumnTableScan -> Filter ->
  * SnappyHashAggregateExec; ColumnTableScan.scala:636-815,
  * SnappyHashAggregateExec.scala:337-500).  The path is HBM-bandwidth-bound
- * (no dense contraction — MFMA unused by design; see DESIGN.md roofline):
- * the kernels stream encoded column bodies from HBM with coalesced loads,
- * keep per-group accumulators in registers (the reference's
- * DictionaryOptimizedMapAccessor idea: a direct accumulator slot per
- * dictionary combination), and fold block partials into a tiny global
- * accumulator with one atomic per value per wave.
+ * (no dense contraction — MFMA unused by design; DESIGN.md roofline).
  *
- * One launch covers EVERY column batch of the table: tiles of SN_TILE_ROWS
- * rows are mapped to workgroups through a host-built tile array, so launch
- * overhead is O(1) per query instead of O(batches).
+ * Structure (third iteration, driven by measurement):
+ *  v1 passed per-lane value arrays to helpers -> 112 B/lane HBM-backed
+ *     scratch -> 9.6% of HBM peak.
+ *  v2 used scalar registers + per-row kind switches -> 1863 basic blocks,
+ *     2782 SGPR lane spills (the compiler cannot hoist 8 uniform switches
+ *     out of an unrolled row loop) -> 15% of peak.
+ *  v3 (this file): per-tile COLUMNAR CONVERSION into LDS, then a branch-free
+ *     row phase.
+ *     - conversion pass: for each referenced column, one tight coalesced
+ *       loop (kind dispatched ONCE per column per chunk) decodes the encoded
+ *       body into a canonical LDS image: f64 value (int32/dict index
+ *       widened; int64 raw-bitcast to stay exact), plus validity/deleted
+ *       bitmap words built with one __ballot per 64 rows.
+ *     - row phase: predicates and aggregate factors read the LDS image by
+ *       RUNTIME column slot (LDS indexing is free, unlike register arrays),
+ *       so there are no per-row branches at all; per-group accumulators stay
+ *       in registers (dictionary-direct slots, the reference's
+ *       DictionaryOptimizedMapAccessor idea), then one wave-reduced atomic
+ *       per value.
  *
- * Register discipline: per-row column values are SCALAR variables (v0..v7)
- * selected by wave-uniform compile-time-unrolled chains — passing per-lane
- * arrays to helpers decays them to memory and allocates HBM-backed scratch
- * (measured: 112 B/lane scratch = 9.6% of HBM peak; scalars = none).
- * Batch column descriptors are hoisted into registers once per tile.
+ * One launch covers EVERY column batch of the table via a host-built tile
+ * array (launch overhead O(1) per query, not O(batches)).
  */
 #include <hip/hip_runtime.h>
 #include "engine_internal.h"
 
 #define WG 256
-#define VPT (SN_TILE_ROWS / WG)
+#define CHUNK 1024              /* rows converted per LDS round */
 
 __device__ __forceinline__ int bm_get(const uint64_t *bm, int row) {
   return (int)((bm[row >> 6] >> (row & 63)) & 1ull);
@@ -50,8 +58,7 @@ __device__ __forceinline__ int patch_find(const int32_t *pos, int n, int row) {
   return -1;
 }
 
-/* general-path read of one column value at `row` (nulls/patches/any kind).
- * Returns 0 if NULL. */
+/* general-path read (nulls/patches/any kind); returns 0 if NULL */
 __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
                                             double *vd, long long *vi, int *gid) {
   if (c.patch_n && bm_get(c.patch_bm, row)) {
@@ -86,38 +93,12 @@ __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
   return 1;
 }
 
-/* scalar-select chains over wave-uniform index (never pass per-lane arrays
- * to functions: that allocates scratch) */
-#define SELD(i) ((i) == 0 ? v0 : (i) == 1 ? v1 : (i) == 2 ? v2 : (i) == 3 ? v3 : \
-                 (i) == 4 ? v4 : (i) == 5 ? v5 : (i) == 6 ? v6 : v7)
-#define SELI(i) ((i) == 0 ? w0 : (i) == 1 ? w1 : (i) == 2 ? w2 : (i) == 3 ? w3 : \
-                 (i) == 4 ? w4 : (i) == 5 ? w5 : (i) == 6 ? w6 : w7)
-#define SELG(i) ((i) == 0 ? g0 : (i) == 1 ? g1 : (i) == 2 ? g2 : (i) == 3 ? g3 : \
-                 (i) == 4 ? g4 : (i) == 5 ? g5 : (i) == 6 ? g6 : g7)
-#define SELN(i) ((i) == 0 ? n0 : (i) == 1 ? n1 : (i) == 2 ? n2 : (i) == 3 ? n3 : \
-                 (i) == 4 ? n4 : (i) == 5 ? n5 : (i) == 6 ? n6 : n7)
-
-/* clean-path load of column slot `c` into scalars (kind/body in registers) */
-#define LOAD_CLEAN(c, vv, ww, gg)                                              \
-  if (kind##c != -1) {                                                         \
-    switch (kind##c) {                                                         \
-      case SN_K_F64: vv = ((const double *)body##c)[row]; ww = (long long)vv; break; \
-      case SN_K_I32: ww = ((const int32_t *)body##c)[row]; vv = (double)ww; break;   \
-      case SN_K_I64: ww = ((const long long *)body##c)[row]; vv = (double)ww; break; \
-      case SN_K_F32: vv = ((const float *)body##c)[row]; ww = (long long)vv; break;  \
-      case SN_K_I16: ww = ((const int16_t *)body##c)[row]; vv = (double)ww; break;   \
-      case SN_K_DICT16:                                                        \
-        gg = dmap##c[(int)(uint16_t)((const int16_t *)body##c)[row]]; break;   \
-      case SN_K_DICT32: gg = dmap##c[((const int32_t *)body##c)[row]]; break;  \
-      default: break;                                                          \
-    }                                                                          \
-  }
-
-#define LOAD_GENERAL(c, vv, ww, gg, nn)                                        \
-  if (kind##c != -1) {                                                         \
-    gg = b.cols[c].null_gid;                                                   \
-    nn = !read_general(b.cols[c], row, &vv, &ww, &gg);                         \
-  }
+__device__ __forceinline__ double i64_as_f64(long long x) {
+  return __longlong_as_double(x);
+}
+__device__ __forceinline__ long long f64_as_i64(double x) {
+  return __double_as_longlong(x);
+}
 
 template <int NSLOTS, int NAGGS>
 __launch_bounds__(WG, 2)
@@ -126,7 +107,16 @@ __global__ void k_scan_agg(sn_dev_plan plan,
                            const sn_dev_tile *__restrict__ tiles, int ntiles,
                            double *__restrict__ out /* [NSLOTS][2*NAGGS+1] */) {
   const int tid = threadIdx.x;
-  double sums[NSLOTS][NAGGS];           /* compile-time indexed only */
+  const int nused = plan.nused;
+
+  /* dynamic LDS: [nused][CHUNK] f64 values, then [nused][CHUNK/64] validity
+   * words, then [CHUNK/64] deleted words */
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double *sval = (double *)smem;
+  uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
+  uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+
+  double sums[NSLOTS][NAGGS];            /* compile-time indexed only */
   double cnts[NSLOTS == 1 ? NAGGS : 1];
   double rc[NSLOTS];
 #pragma unroll
@@ -141,101 +131,168 @@ __global__ void k_scan_agg(sn_dev_plan plan,
   }
 
   const int npreds = plan.npreds, naggs = plan.naggs, ngroup = plan.ngroup;
-  const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
 
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
-    const int row_end = min(tile.row_start + SN_TILE_ROWS, b.num_rows);
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
     const int clean = b.clean;
 
-    /* hoist per-column kind/body/dictmap into registers for the tile */
-#define HOIST(c)                                                               \
-    const void *body##c = b.cols[c].body;                                      \
-    const int kind##c = body##c ? b.cols[c].kind : -1;                         \
-    const int32_t *dmap##c = b.cols[c].dictmap;
-    HOIST(0) HOIST(1) HOIST(2) HOIST(3) HOIST(4) HOIST(5) HOIST(6) HOIST(7)
-#undef HOIST
+    for (int base = tile.row_start; base < tile_end; base += CHUNK) {
+      const int rows = min(CHUNK, tile_end - base);
 
-    for (int row = tile.row_start + tid; row < row_end; row += WG) {
-      double v0 = 0, v1 = 0, v2 = 0, v3 = 0, v4 = 0, v5 = 0, v6 = 0, v7 = 0;
-      long long w0 = 0, w1 = 0, w2 = 0, w3 = 0, w4 = 0, w5 = 0, w6 = 0, w7 = 0;
-      int g0 = 0, g1 = 0, g2 = 0, g3 = 0, g4 = 0, g5 = 0, g6 = 0, g7 = 0;
-      int n0 = 0, n1 = 0, n2 = 0, n3 = 0, n4 = 0, n5 = 0, n6 = 0, n7 = 0;
-      int alive = 1;
-
-      if (clean) {
-        LOAD_CLEAN(0, v0, w0, g0) LOAD_CLEAN(1, v1, w1, g1)
-        LOAD_CLEAN(2, v2, w2, g2) LOAD_CLEAN(3, v3, w3, g3)
-        LOAD_CLEAN(4, v4, w4, g4) LOAD_CLEAN(5, v5, w5, g5)
-        LOAD_CLEAN(6, v6, w6, g6) LOAD_CLEAN(7, v7, w7, g7)
-      } else {
-        if (b.del_bm && bm_get(b.del_bm, row)) continue;
-        LOAD_GENERAL(0, v0, w0, g0, n0) LOAD_GENERAL(1, v1, w1, g1, n1)
-        LOAD_GENERAL(2, v2, w2, g2, n2) LOAD_GENERAL(3, v3, w3, g3, n3)
-        LOAD_GENERAL(4, v4, w4, g4, n4) LOAD_GENERAL(5, v5, w5, g5, n5)
-        LOAD_GENERAL(6, v6, w6, g6, n6) LOAD_GENERAL(7, v7, w7, g7, n7)
-      }
-
-      /* predicates (NULL compares false) — compile-time unrolled over the
-       * max so plan fields read via uniform kernarg loads */
-#pragma unroll
-      for (int i = 0; i < 8; i++) {
-        if (i >= npreds) break;
-        const sn_dev_pred &p = plan.preds[i];
-        const int cs = p.cslot;
-        if (!clean && SELN(cs)) { alive = 0; continue; }
-        if (p.is_double) {
-          double x = SELD(cs);
-          if (p.has_lo && (p.lo_strict ? !(x > p.lo_d) : !(x >= p.lo_d))) alive = 0;
-          if (p.has_hi && (p.hi_strict ? !(x < p.hi_d) : !(x <= p.hi_d))) alive = 0;
+      /* ---- conversion pass: one tight loop per referenced column ---- */
+      for (int c = 0; c < nused; c++) {
+        const sn_dev_col col = b.cols[c];       /* scalar copy (uniform) */
+        double *dst = sval + (size_t)c * CHUNK;
+        if (clean) {
+          switch (col.kind) {
+            case SN_K_F64: {
+              const double *src = (const double *)col.body + base;
+              for (int r = tid; r < rows; r += WG) dst[r] = src[r];
+              break;
+            }
+            case SN_K_I32: {
+              const int32_t *src = (const int32_t *)col.body + base;
+              for (int r = tid; r < rows; r += WG) dst[r] = (double)src[r];
+              break;
+            }
+            case SN_K_I64: {
+              const long long *src = (const long long *)col.body + base;
+              for (int r = tid; r < rows; r += WG) dst[r] = i64_as_f64(src[r]);
+              break;
+            }
+            case SN_K_F32: {
+              const float *src = (const float *)col.body + base;
+              for (int r = tid; r < rows; r += WG) dst[r] = (double)src[r];
+              break;
+            }
+            case SN_K_I16: {
+              const int16_t *src = (const int16_t *)col.body + base;
+              for (int r = tid; r < rows; r += WG) dst[r] = (double)src[r];
+              break;
+            }
+            case SN_K_DICT16: {
+              const int16_t *src = (const int16_t *)col.body + base;
+              const int32_t *dm = col.dictmap;
+              for (int r = tid; r < rows; r += WG)
+                dst[r] = (double)dm[(int)(uint16_t)src[r]];
+              break;
+            }
+            case SN_K_DICT32: {
+              const int32_t *src = (const int32_t *)col.body + base;
+              const int32_t *dm = col.dictmap;
+              for (int r = tid; r < rows; r += WG) dst[r] = (double)dm[src[r]];
+              break;
+            }
+            case SN_K_BOOLBIT: {
+              const uint64_t *src = (const uint64_t *)col.body;
+              for (int r = tid; r < rows; r += WG)
+                dst[r] = (double)bm_get(src, base + r);
+              break;
+            }
+          }
         } else {
-          long long x = SELI(cs);
-          if (p.has_lo && (p.lo_strict ? !(x > p.lo_i) : !(x >= p.lo_i))) alive = 0;
-          if (p.has_hi && (p.hi_strict ? !(x < p.hi_i) : !(x <= p.hi_i))) alive = 0;
-        }
-      }
-      if (__popcll(__ballot(alive)) == 0) continue;
-
-      int slot = 0;
-      if (NSLOTS > 1) {
-        if (ngroup >= 1) slot = SELG(gc0);
-        if (ngroup >= 2) slot += SELG(gc1);
-      }
-
-      /* aggregates: value = product of (add + mul * col) factors */
-#pragma unroll
-      for (int a = 0; a < NAGGS; a++) {
-        if (a >= naggs) break;
-        const sn_dev_agg &ag = plan.aggs[a];
-        double aval = 1.0;
-        int anull = 0;
-        if (ag.kind != 1) {
-#pragma unroll
-          for (int j = 0; j < 3; j++) {
-            if (j >= ag.nf) break;
-            const int fc = ag.f[j].cslot;
-            if (!clean && SELN(fc)) anull = 1;
-            aval = aval * (ag.f[j].add + ag.f[j].mul * SELD(fc));
+          /* general: nulls/patches; validity words via one ballot per 64 rows */
+          uint64_t *vw = svalid + (size_t)c * (CHUNK / 64);
+          const int is_dict = col.kind == SN_K_DICT16 || col.kind == SN_K_DICT32;
+          const int is_i64 = col.kind == SN_K_I64;
+          for (int r = tid; r < CHUNK; r += WG) {
+            int row = base + r;
+            int ok = 0;
+            double vd = 0.0; long long vi = 0; int gid = col.null_gid;
+            if (row < num_rows)
+              ok = read_general(col, row, &vd, &vi, &gid);
+            double v = is_dict ? (double)gid : (is_i64 ? i64_as_f64(vi) : vd);
+            dst[r] = v;
+            uint64_t w = __ballot(ok);
+            if ((tid & 63) == 0) vw[r >> 6] = w;
           }
         }
-        const int m = alive && !anull;
+      }
+      /* deleted-row words (general only) */
+      if (!clean) {
+        const uint64_t *del = b.del_bm;
+        for (int r = tid; r < CHUNK; r += WG) {
+          int row = base + r;
+          int dead = (row >= num_rows) || (del && bm_get(del, row));
+          uint64_t w = __ballot(dead);
+          if ((tid & 63) == 0) sdead[r >> 6] = w;
+        }
+      }
+      __syncthreads();
+
+      /* ---- row phase: branch-free, LDS-indexed by runtime cslot ---- */
+      for (int r = tid; r < rows; r += WG) {
+        int alive = clean ? 1 : !((sdead[r >> 6] >> (r & 63)) & 1);
+
+#pragma unroll
+        for (int i = 0; i < 8; i++) {
+          if (i >= npreds) break;
+          const sn_dev_pred &p = plan.preds[i];
+          const int cs = p.cslot;
+          if (!clean)
+            alive &= (int)((svalid[(size_t)cs * (CHUNK / 64) + (r >> 6)]
+                            >> (r & 63)) & 1ull);
+          const double xv = sval[(size_t)cs * CHUNK + r];
+          if ((plan.i64_mask >> cs) & 1u) {
+            long long x = f64_as_i64(xv);
+            if (p.has_lo && (p.lo_strict ? !(x > p.lo_i) : !(x >= p.lo_i))) alive = 0;
+            if (p.has_hi && (p.hi_strict ? !(x < p.hi_i) : !(x <= p.hi_i))) alive = 0;
+          } else {
+            if (p.has_lo && (p.lo_strict ? !(xv > p.lo_d) : !(xv >= p.lo_d))) alive = 0;
+            if (p.has_hi && (p.hi_strict ? !(xv < p.hi_d) : !(xv <= p.hi_d))) alive = 0;
+          }
+        }
+        if (__popcll(__ballot(alive)) == 0) continue;
+
+        int slot = 0;
+        if (NSLOTS > 1) {
+          if (ngroup >= 1)
+            slot = (int)sval[(size_t)plan.gcol[0] * CHUNK + r];
+          if (ngroup >= 2)
+            slot += (int)sval[(size_t)plan.gcol[1] * CHUNK + r];
+        }
+
+#pragma unroll
+        for (int a = 0; a < NAGGS; a++) {
+          if (a >= naggs) break;
+          const sn_dev_agg &ag = plan.aggs[a];
+          double aval = 1.0;
+          int anull = 0;
+          if (ag.kind != 1) {
+#pragma unroll
+            for (int j = 0; j < 3; j++) {
+              if (j >= ag.nf) break;
+              const int fc = ag.f[j].cslot;
+              if (!clean)
+                anull |= !(int)((svalid[(size_t)fc * (CHUNK / 64) + (r >> 6)]
+                                 >> (r & 63)) & 1ull);
+              double x = sval[(size_t)fc * CHUNK + r];
+              if ((plan.i64_mask >> fc) & 1u) x = (double)f64_as_i64(x);
+              aval = aval * (ag.f[j].add + ag.f[j].mul * x);
+            }
+          }
+          const int m = alive && !anull;
+          if (NSLOTS == 1) {
+            sums[0][a] += m ? aval : 0.0;
+            cnts[a] += m ? 1.0 : 0.0;
+          } else {
+#pragma unroll
+            for (int s = 0; s < NSLOTS; s++)
+              sums[s][a] += (m && slot == s) ? aval : 0.0;
+          }
+        }
         if (NSLOTS == 1) {
-          sums[0][a] += m ? aval : 0.0;
-          cnts[a] += m ? 1.0 : 0.0;
+          rc[0] += alive ? 1.0 : 0.0;
         } else {
 #pragma unroll
           for (int s = 0; s < NSLOTS; s++)
-            sums[s][a] += (m && slot == s) ? aval : 0.0;
+            rc[s] += (alive && slot == s) ? 1.0 : 0.0;
         }
       }
-      if (NSLOTS == 1) {
-        rc[0] += alive ? 1.0 : 0.0;
-      } else {
-#pragma unroll
-        for (int s = 0; s < NSLOTS; s++)
-          rc[s] += (alive && slot == s) ? 1.0 : 0.0;
-      }
+      __syncthreads();   /* LDS reused by the next chunk */
     }
   }
 
@@ -274,9 +331,11 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
   hipStream_t s = (hipStream_t)stream;
   int grid = ntiles < 2048 ? (ntiles > 0 ? ntiles : 1) : 2048;
   const int ns = plan->nslots, na = plan->naggs;
+  size_t lds = (size_t)plan->nused * CHUNK * 8 +
+               (size_t)plan->nused * (CHUNK / 64) * 8 + (CHUNK / 64) * 8 + 16;
   hipError_t err;
 #define LAUNCH(S, A)                                                        \
-  hipLaunchKernelGGL((k_scan_agg<S, A>), dim3(grid), dim3(WG), 0, s,        \
+  hipLaunchKernelGGL((k_scan_agg<S, A>), dim3(grid), dim3(WG), lds, s,      \
                      *plan, dev_batches, dev_tiles, ntiles, dev_out);
   if (ns <= 1) {
     if (na <= 2) { LAUNCH(1, 2) }
