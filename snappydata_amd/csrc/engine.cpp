@@ -758,6 +758,13 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     d.cslot = q->cslot_of_col[s.col];
     d.is_double = (dt == SN_TYPE_DOUBLE || dt == SN_TYPE_FLOAT);
     d.lo_d = s.lo_d; d.hi_d = s.hi_d; d.lo_i = s.lo_i; d.hi_i = s.hi_i;
+    /* the kernel's LDS image is f64 for every non-INT64 column, so integer
+     * bounds must be widened (exact for int32/int16); INT64 stays raw and
+     * compares through lo_i/hi_i */
+    if (!d.is_double && dt != SN_TYPE_INT64) {
+      d.lo_d = (double)s.lo_i;
+      d.hi_d = (double)s.hi_i;
+    }
     d.has_lo = s.has_lo; d.has_hi = s.has_hi;
     d.lo_strict = s.lo_strict; d.hi_strict = s.hi_strict;
   }
