@@ -420,9 +420,22 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
         l2g.push_back(gid);
       }
     }
-    /* upload blob */
-    b.col_dev[c] = up(e, blob, (size_t)len);
-    if (!b.col_dev[c]) return fail(SN_ERR_NOMEM, "HBM upload failed");
+    /* upload blob, placed so the BODY is 16-byte aligned (the scan kernel
+     * issues 16 B/lane vector loads on the body; the 8-byte blob header
+     * would otherwise leave it 8-aligned) */
+    {
+      int64_t pad = (16 - (b.cols[c].body_off & 15)) & 15;
+      char *base = (char *)e->arena.alloc((size_t)len + 16);
+      if (!base) return fail(SN_ERR_NOMEM, "HBM upload failed");
+      char *dst = base + pad;
+      if (e->arena.device >= 0) {
+        if (hipMemcpy(dst, blob, (size_t)len, hipMemcpyHostToDevice) != hipSuccess)
+          return fail(SN_ERR_NOMEM, "HBM upload failed");
+      } else {
+        memcpy(dst, blob, (size_t)len);
+      }
+      b.col_dev[c] = dst;
+    }
     if (e->arena.device < 0) {
       b.host_blobs.emplace_back(blob, blob + len);
     }

@@ -1,40 +1,38 @@
 /*
- * kernels.hip — CDNA4 (gfx950) scan -> filter -> aggregate kernels.
+ * kernels.hip — CDNA4 (gfx950) scan -> filter -> aggregate kernels (v4).
  *
  * MI355X-native replacement for the reference's generated per-partition JVM
  * loop (WholeStageCodegen of ColumnTableScan -> Filter ->
  * SnappyHashAggregateExec; ColumnTableScan.scala:636-815,
- * SnappyHashAggregateExec.scala:337-500).  The path is HBM-bandwidth-bound
+ * SnappyHashAggregateExec.scala:337-500).  HBM-bandwidth-bound by nature
  * (no dense contraction — MFMA unused by design; DESIGN.md roofline).
  *
- * Structure (third iteration, driven by measurement):
- *  v1 passed per-lane value arrays to helpers -> 112 B/lane HBM-backed
- *     scratch -> 9.6% of HBM peak.
- *  v2 used scalar registers + per-row kind switches -> 1863 basic blocks,
- *     2782 SGPR lane spills (the compiler cannot hoist 8 uniform switches
- *     out of an unrolled row loop) -> 15% of peak.
- *  v3 (this file): per-tile COLUMNAR CONVERSION into LDS, then a branch-free
- *     row phase.
- *     - conversion pass: for each referenced column, one tight coalesced
- *       loop (kind dispatched ONCE per column per chunk) decodes the encoded
- *       body into a canonical LDS image: f64 value (int32/dict index
- *       widened; int64 raw-bitcast to stay exact), plus validity/deleted
- *       bitmap words built with one __ballot per 64 rows.
- *     - row phase: predicates and aggregate factors read the LDS image by
- *       RUNTIME column slot (LDS indexing is free, unlike register arrays),
- *       so there are no per-row branches at all; per-group accumulators stay
- *       in registers (dictionary-direct slots, the reference's
- *       DictionaryOptimizedMapAccessor idea), then one wave-reduced atomic
- *       per value.
+ * Iteration history (measured on MI355X):
+ *  v1 per-lane value arrays -> 112 B/lane scratch -> 9.6% of HBM peak.
+ *  v2 scalar registers + per-row kind switches -> 1863 blocks, SGPR spill
+ *     storm -> 15%.
+ *  v3 columnar LDS conversion + branch-free row phase -> 23% (Q6); grouped
+ *     8x8 register accumulators spilled 372 B/lane -> 3.9% (Q1).
+ *  v4 (this file):
+ *     - conversion loops compile-time unrolled and vectorized (double2 /
+ *       int2 loads, ds_write_b128) on full chunks;
+ *     - separate grouped kernel that evaluates predicates once into an LDS
+ *       (slot, alive) image, then processes aggregates in groups of 4 with
+ *       per-lane sums[NSLOTS][4] registers, wave-reducing each chunk into an
+ *       LDS block accumulator — register pressure bounded for any slot
+ *       count; one global atomic per accumulator per block at the end.
  *
- * One launch covers EVERY column batch of the table via a host-built tile
- * array (launch overhead O(1) per query, not O(batches)).
+ * One launch covers EVERY column batch via a host-built tile array.
  */
 #include <hip/hip_runtime.h>
 #include "engine_internal.h"
 
 #define WG 256
-#define CHUNK 1024              /* rows converted per LDS round */
+#define CHUNK 1024
+#define AGRP 4                  /* aggregates per register pass (grouped) */
+
+typedef double double2_t __attribute__((ext_vector_type(2)));
+typedef int int2_t __attribute__((ext_vector_type(2)));
 
 __device__ __forceinline__ int bm_get(const uint64_t *bm, int row) {
   return (int)((bm[row >> 6] >> (row & 63)) & 1ull);
@@ -93,44 +91,216 @@ __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
   return 1;
 }
 
-__device__ __forceinline__ double i64_as_f64(long long x) {
-  return __longlong_as_double(x);
-}
-__device__ __forceinline__ long long f64_as_i64(double x) {
-  return __double_as_longlong(x);
+/* ---- conversion pass: decode one chunk of every referenced column into the
+ * canonical LDS image (f64 values; int64 raw-bitcast; dict -> premultiplied
+ * group id as f64).  Vectorized (16 B/lane) on full chunks. ---- */
+__device__ __forceinline__ void convert_chunk(
+    const sn_dev_batch &b, int nused, int base, int rows, int num_rows,
+    double *sval, uint64_t *svalid, uint64_t *sdead) {
+  const int tid = threadIdx.x;
+  const int clean = b.clean;
+  for (int c = 0; c < nused; c++) {
+    const sn_dev_col col = b.cols[c];
+    double *dst = sval + (size_t)c * CHUNK;
+    if (clean) {
+      switch (col.kind) {
+        case SN_K_F64: {
+          const double *src = (const double *)col.body + base;
+          if (rows == CHUNK) {
+#pragma unroll
+            for (int k = 0; k < CHUNK / (2 * WG); k++) {
+              int h = tid + k * WG;   /* pair index */
+              ((double2_t *)dst)[h] = ((const double2_t *)src)[h];
+            }
+          } else {
+#pragma unroll
+            for (int k = 0; k < CHUNK / WG; k++) {
+              int r = tid + k * WG;
+              if (r < rows) dst[r] = src[r];
+            }
+          }
+          break;
+        }
+        case SN_K_I32: {
+          const int32_t *src = (const int32_t *)col.body + base;
+          if (rows == CHUNK) {
+#pragma unroll
+            for (int k = 0; k < CHUNK / (2 * WG); k++) {
+              int h = tid + k * WG;
+              int2_t x = ((const int2_t *)src)[h];
+              double2_t y; y.x = (double)x.x; y.y = (double)x.y;
+              ((double2_t *)dst)[h] = y;
+            }
+          } else {
+#pragma unroll
+            for (int k = 0; k < CHUNK / WG; k++) {
+              int r = tid + k * WG;
+              if (r < rows) dst[r] = (double)src[r];
+            }
+          }
+          break;
+        }
+        case SN_K_I64: {
+          const long long *src = (const long long *)col.body + base;
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = __longlong_as_double(src[r]);
+          }
+          break;
+        }
+        case SN_K_F32: {
+          const float *src = (const float *)col.body + base;
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = (double)src[r];
+          }
+          break;
+        }
+        case SN_K_I16: {
+          const int16_t *src = (const int16_t *)col.body + base;
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = (double)src[r];
+          }
+          break;
+        }
+        case SN_K_DICT16: {
+          const int16_t *src = (const int16_t *)col.body + base;
+          const int32_t *dm = col.dictmap;
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = (double)dm[(int)(uint16_t)src[r]];
+          }
+          break;
+        }
+        case SN_K_DICT32: {
+          const int32_t *src = (const int32_t *)col.body + base;
+          const int32_t *dm = col.dictmap;
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = (double)dm[src[r]];
+          }
+          break;
+        }
+        case SN_K_BOOLBIT: {
+          const uint64_t *src = (const uint64_t *)col.body;
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = (double)bm_get(src, base + r);
+          }
+          break;
+        }
+      }
+    } else {
+      uint64_t *vw = svalid + (size_t)c * (CHUNK / 64);
+      const int is_dict = col.kind == SN_K_DICT16 || col.kind == SN_K_DICT32;
+      const int is_i64 = col.kind == SN_K_I64;
+#pragma unroll
+      for (int k = 0; k < CHUNK / WG; k++) {
+        int r = tid + k * WG;
+        int row = base + r;
+        int ok = 0;
+        double vd = 0.0; long long vi = 0; int gid = col.null_gid;
+        if (row < num_rows) ok = read_general(col, row, &vd, &vi, &gid);
+        dst[r] = is_dict ? (double)gid : (is_i64 ? __longlong_as_double(vi) : vd);
+        uint64_t w = __ballot(ok);
+        if ((tid & 63) == 0) vw[r >> 6] = w;
+      }
+    }
+  }
+  if (!clean) {
+    const uint64_t *del = b.del_bm;
+#pragma unroll
+    for (int k = 0; k < CHUNK / WG; k++) {
+      int r = tid + k * WG;
+      int row = base + r;
+      int dead = (row >= num_rows) || (del && bm_get(del, row));
+      uint64_t w = __ballot(dead);
+      if ((tid & 63) == 0) sdead[r >> 6] = w;
+    }
+  }
 }
 
-template <int NSLOTS, int NAGGS>
+/* predicate evaluation for row r of the chunk (LDS image) */
+__device__ __forceinline__ int eval_preds(const sn_dev_plan &plan, int clean,
+                                          const double *sval,
+                                          const uint64_t *svalid,
+                                          const uint64_t *sdead, int r) {
+  int alive = clean ? 1 : !((sdead[r >> 6] >> (r & 63)) & 1ull);
+#pragma unroll
+  for (int i = 0; i < 8; i++) {
+    if (i >= plan.npreds) break;
+    const sn_dev_pred &p = plan.preds[i];
+    const int cs = p.cslot;
+    if (!clean)
+      alive &= (int)((svalid[(size_t)cs * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
+    const double xv = sval[(size_t)cs * CHUNK + r];
+    if ((plan.i64_mask >> cs) & 1u) {
+      long long x = __double_as_longlong(xv);
+      if (p.has_lo && (p.lo_strict ? !(x > p.lo_i) : !(x >= p.lo_i))) alive = 0;
+      if (p.has_hi && (p.hi_strict ? !(x < p.hi_i) : !(x <= p.hi_i))) alive = 0;
+    } else {
+      if (p.has_lo && (p.lo_strict ? !(xv > p.lo_d) : !(xv >= p.lo_d))) alive = 0;
+      if (p.has_hi && (p.hi_strict ? !(xv < p.hi_d) : !(xv <= p.hi_d))) alive = 0;
+    }
+  }
+  return alive;
+}
+
+/* aggregate input value for agg a, row r (non-null factors assumed checked) */
+__device__ __forceinline__ double eval_agg(const sn_dev_plan &plan, int a,
+                                           const double *sval, int r,
+                                           int clean, const uint64_t *svalid,
+                                           int *anull) {
+  const sn_dev_agg &ag = plan.aggs[a];
+  double aval = 1.0;
+  *anull = 0;
+  if (ag.kind == 1) return 1.0;   /* COUNT(*) */
+#pragma unroll
+  for (int j = 0; j < 3; j++) {
+    if (j >= ag.nf) break;
+    const int fc = ag.f[j].cslot;
+    if (!clean)
+      *anull |= !(int)((svalid[(size_t)fc * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
+    double x = sval[(size_t)fc * CHUNK + r];
+    if ((plan.i64_mask >> fc) & 1u) x = (double)__double_as_longlong(x);
+    aval = aval * (ag.f[j].add + ag.f[j].mul * x);
+  }
+  return aval;
+}
+
+/* wave (64-lane) sum reduction */
+__device__ __forceinline__ double wave_sum(double x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    x += __shfl_down(x, off, 64);
+  return x;
+}
+
+/* ================= keyless kernel ================= */
+template <int NAGGS>
 __launch_bounds__(WG, 2)
-__global__ void k_scan_agg(sn_dev_plan plan,
-                           const sn_dev_batch *__restrict__ batches,
-                           const sn_dev_tile *__restrict__ tiles, int ntiles,
-                           double *__restrict__ out /* [NSLOTS][2*NAGGS+1] */) {
+__global__ void k_keyless(sn_dev_plan plan,
+                          const sn_dev_batch *__restrict__ batches,
+                          const sn_dev_tile *__restrict__ tiles, int ntiles,
+                          double *__restrict__ out /* [2*NAGGS+1] */) {
   const int tid = threadIdx.x;
   const int nused = plan.nused;
-
-  /* dynamic LDS: [nused][CHUNK] f64 values, then [nused][CHUNK/64] validity
-   * words, then [CHUNK/64] deleted words */
   extern __shared__ __attribute__((aligned(16))) char smem[];
   double *sval = (double *)smem;
   uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
   uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
 
-  double sums[NSLOTS][NAGGS];            /* compile-time indexed only */
-  double cnts[NSLOTS == 1 ? NAGGS : 1];
-  double rc[NSLOTS];
+  double sums[NAGGS], cnts[NAGGS], rcnt = 0.0;
 #pragma unroll
-  for (int s = 0; s < NSLOTS; s++) {
-    rc[s] = 0.0;
-#pragma unroll
-    for (int a = 0; a < NAGGS; a++) sums[s][a] = 0.0;
-  }
-  if (NSLOTS == 1) {
-#pragma unroll
-    for (int a = 0; a < NAGGS; a++) cnts[a] = 0.0;
-  }
-
-  const int npreds = plan.npreds, naggs = plan.naggs, ngroup = plan.ngroup;
+  for (int a = 0; a < NAGGS; a++) { sums[a] = 0.0; cnts[a] = 0.0; }
+  const int naggs = plan.naggs;
 
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
@@ -141,187 +311,181 @@ __global__ void k_scan_agg(sn_dev_plan plan,
 
     for (int base = tile.row_start; base < tile_end; base += CHUNK) {
       const int rows = min(CHUNK, tile_end - base);
-
-      /* ---- conversion pass: one tight loop per referenced column ---- */
-      for (int c = 0; c < nused; c++) {
-        const sn_dev_col col = b.cols[c];       /* scalar copy (uniform) */
-        double *dst = sval + (size_t)c * CHUNK;
-        if (clean) {
-          switch (col.kind) {
-            case SN_K_F64: {
-              const double *src = (const double *)col.body + base;
-              for (int r = tid; r < rows; r += WG) dst[r] = src[r];
-              break;
-            }
-            case SN_K_I32: {
-              const int32_t *src = (const int32_t *)col.body + base;
-              for (int r = tid; r < rows; r += WG) dst[r] = (double)src[r];
-              break;
-            }
-            case SN_K_I64: {
-              const long long *src = (const long long *)col.body + base;
-              for (int r = tid; r < rows; r += WG) dst[r] = i64_as_f64(src[r]);
-              break;
-            }
-            case SN_K_F32: {
-              const float *src = (const float *)col.body + base;
-              for (int r = tid; r < rows; r += WG) dst[r] = (double)src[r];
-              break;
-            }
-            case SN_K_I16: {
-              const int16_t *src = (const int16_t *)col.body + base;
-              for (int r = tid; r < rows; r += WG) dst[r] = (double)src[r];
-              break;
-            }
-            case SN_K_DICT16: {
-              const int16_t *src = (const int16_t *)col.body + base;
-              const int32_t *dm = col.dictmap;
-              for (int r = tid; r < rows; r += WG)
-                dst[r] = (double)dm[(int)(uint16_t)src[r]];
-              break;
-            }
-            case SN_K_DICT32: {
-              const int32_t *src = (const int32_t *)col.body + base;
-              const int32_t *dm = col.dictmap;
-              for (int r = tid; r < rows; r += WG) dst[r] = (double)dm[src[r]];
-              break;
-            }
-            case SN_K_BOOLBIT: {
-              const uint64_t *src = (const uint64_t *)col.body;
-              for (int r = tid; r < rows; r += WG)
-                dst[r] = (double)bm_get(src, base + r);
-              break;
-            }
-          }
-        } else {
-          /* general: nulls/patches; validity words via one ballot per 64 rows */
-          uint64_t *vw = svalid + (size_t)c * (CHUNK / 64);
-          const int is_dict = col.kind == SN_K_DICT16 || col.kind == SN_K_DICT32;
-          const int is_i64 = col.kind == SN_K_I64;
-          for (int r = tid; r < CHUNK; r += WG) {
-            int row = base + r;
-            int ok = 0;
-            double vd = 0.0; long long vi = 0; int gid = col.null_gid;
-            if (row < num_rows)
-              ok = read_general(col, row, &vd, &vi, &gid);
-            double v = is_dict ? (double)gid : (is_i64 ? i64_as_f64(vi) : vd);
-            dst[r] = v;
-            uint64_t w = __ballot(ok);
-            if ((tid & 63) == 0) vw[r >> 6] = w;
-          }
-        }
-      }
-      /* deleted-row words (general only) */
-      if (!clean) {
-        const uint64_t *del = b.del_bm;
-        for (int r = tid; r < CHUNK; r += WG) {
-          int row = base + r;
-          int dead = (row >= num_rows) || (del && bm_get(del, row));
-          uint64_t w = __ballot(dead);
-          if ((tid & 63) == 0) sdead[r >> 6] = w;
-        }
-      }
+      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
       __syncthreads();
 
-      /* ---- row phase: branch-free, LDS-indexed by runtime cslot ---- */
-      for (int r = tid; r < rows; r += WG) {
-        int alive = clean ? 1 : !((sdead[r >> 6] >> (r & 63)) & 1);
-
-#pragma unroll
-        for (int i = 0; i < 8; i++) {
-          if (i >= npreds) break;
-          const sn_dev_pred &p = plan.preds[i];
-          const int cs = p.cslot;
-          if (!clean)
-            alive &= (int)((svalid[(size_t)cs * (CHUNK / 64) + (r >> 6)]
-                            >> (r & 63)) & 1ull);
-          const double xv = sval[(size_t)cs * CHUNK + r];
-          if ((plan.i64_mask >> cs) & 1u) {
-            long long x = f64_as_i64(xv);
-            if (p.has_lo && (p.lo_strict ? !(x > p.lo_i) : !(x >= p.lo_i))) alive = 0;
-            if (p.has_hi && (p.hi_strict ? !(x < p.hi_i) : !(x <= p.hi_i))) alive = 0;
-          } else {
-            if (p.has_lo && (p.lo_strict ? !(xv > p.lo_d) : !(xv >= p.lo_d))) alive = 0;
-            if (p.has_hi && (p.hi_strict ? !(xv < p.hi_d) : !(xv <= p.hi_d))) alive = 0;
-          }
-        }
+#pragma unroll 2
+      for (int k = 0; k < CHUNK / WG; k++) {
+        int r = tid + k * WG;
+        int inr = r < rows;
+        int alive = inr ? eval_preds(plan, clean, sval, svalid, sdead, r) : 0;
         if (__popcll(__ballot(alive)) == 0) continue;
-
-        int slot = 0;
-        if (NSLOTS > 1) {
-          if (ngroup >= 1)
-            slot = (int)sval[(size_t)plan.gcol[0] * CHUNK + r];
-          if (ngroup >= 2)
-            slot += (int)sval[(size_t)plan.gcol[1] * CHUNK + r];
-        }
-
 #pragma unroll
         for (int a = 0; a < NAGGS; a++) {
           if (a >= naggs) break;
-          const sn_dev_agg &ag = plan.aggs[a];
-          double aval = 1.0;
-          int anull = 0;
-          if (ag.kind != 1) {
-#pragma unroll
-            for (int j = 0; j < 3; j++) {
-              if (j >= ag.nf) break;
-              const int fc = ag.f[j].cslot;
-              if (!clean)
-                anull |= !(int)((svalid[(size_t)fc * (CHUNK / 64) + (r >> 6)]
-                                 >> (r & 63)) & 1ull);
-              double x = sval[(size_t)fc * CHUNK + r];
-              if ((plan.i64_mask >> fc) & 1u) x = (double)f64_as_i64(x);
-              aval = aval * (ag.f[j].add + ag.f[j].mul * x);
-            }
-          }
+          int anull;
+          double aval = eval_agg(plan, a, sval, r, clean, svalid, &anull);
           const int m = alive && !anull;
-          if (NSLOTS == 1) {
-            sums[0][a] += m ? aval : 0.0;
-            cnts[a] += m ? 1.0 : 0.0;
-          } else {
-#pragma unroll
-            for (int s = 0; s < NSLOTS; s++)
-              sums[s][a] += (m && slot == s) ? aval : 0.0;
-          }
+          sums[a] += m ? aval : 0.0;
+          cnts[a] += m ? 1.0 : 0.0;
         }
-        if (NSLOTS == 1) {
-          rc[0] += alive ? 1.0 : 0.0;
-        } else {
+        rcnt += alive ? 1.0 : 0.0;
+      }
+      __syncthreads();
+    }
+  }
+
+#pragma unroll
+  for (int a = 0; a < NAGGS; a++) {
+    double x = wave_sum(sums[a]);
+    if ((tid & 63) == 0 && x != 0.0) atomicAdd(&out[a], x);
+    x = wave_sum(cnts[a]);
+    if ((tid & 63) == 0 && x != 0.0) atomicAdd(&out[NAGGS + a], x);
+  }
+  double x = wave_sum(rcnt);
+  if ((tid & 63) == 0 && x != 0.0) atomicAdd(&out[2 * NAGGS], x);
+}
+
+/* ================= grouped kernel =================
+ * Aggregates processed in groups of AGRP with per-lane sums[NSLOTS][AGRP]
+ * registers; after each chunk the registers wave-reduce into the LDS block
+ * accumulator (nslots x (naggs+1)), which flushes once per block to global
+ * atomics.  Requires non-nullable aggregate inputs (engine-validated);
+ * output layout [slot][2*NA+1]: sums, counts(=rowcount written host-side),
+ * rowcount. */
+template <int NSLOTS>
+__launch_bounds__(WG, 2)
+__global__ void k_grouped(sn_dev_plan plan,
+                          const sn_dev_batch *__restrict__ batches,
+                          const sn_dev_tile *__restrict__ tiles, int ntiles,
+                          double *__restrict__ out, int out_stride) {
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int nused = plan.nused;
+  const int naggs = plan.naggs, ngroup = plan.ngroup;
+  const int nagg_grps = (naggs + AGRP - 1) / AGRP;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double *sval = (double *)smem;
+  uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
+  uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+  uint64_t *salive = sdead + CHUNK / 64;
+  int16_t *sslot = (int16_t *)(salive + CHUNK / 64);
+  /* block accumulator: [NSLOTS][naggs+1], init once */
+  double *bacc = (double *)(sslot + CHUNK);
+
+  for (int i = tid; i < NSLOTS * (naggs + 1); i += WG) bacc[i] = 0.0;
+  __syncthreads();
+
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const sn_dev_tile tile = tiles[t];
+    const sn_dev_batch &b = batches[tile.batch];
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
+    const int clean = b.clean;
+
+    for (int base = tile.row_start; base < tile_end; base += CHUNK) {
+      const int rows = min(CHUNK, tile_end - base);
+      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      __syncthreads();
+
+      /* pass A: predicates + slot once per row */
+      double rcreg = 0.0;
+#pragma unroll 2
+      for (int k = 0; k < CHUNK / WG; k++) {
+        int r = tid + k * WG;
+        int inr = r < rows;
+        int alive = inr ? eval_preds(plan, clean, sval, svalid, sdead, r) : 0;
+        int slot = 0;
+        if (alive) {
+          if (ngroup >= 1) slot = (int)sval[(size_t)plan.gcol[0] * CHUNK + r];
+          if (ngroup >= 2) slot += (int)sval[(size_t)plan.gcol[1] * CHUNK + r];
+        }
+        sslot[r] = (int16_t)slot;
+        uint64_t w = __ballot(alive);
+        if ((tid & 63) == 0) salive[r >> 6] = w;
+      }
+      __syncthreads();
+
+      /* rowcount per slot (registers, one pass) */
+      {
+        double rc[NSLOTS];
+#pragma unroll
+        for (int s = 0; s < NSLOTS; s++) rc[s] = 0.0;
+#pragma unroll 2
+        for (int k = 0; k < CHUNK / WG; k++) {
+          int r = tid + k * WG;
+          int alive = (int)((salive[r >> 6] >> (r & 63)) & 1ull);
+          int slot = sslot[r];
 #pragma unroll
           for (int s = 0; s < NSLOTS; s++)
             rc[s] += (alive && slot == s) ? 1.0 : 0.0;
         }
+#pragma unroll
+        for (int s = 0; s < NSLOTS; s++) {
+          double x = wave_sum(rc[s]);
+          if ((tid & 63) == 0 && x != 0.0)
+            atomicAdd(&bacc[s * (naggs + 1) + naggs], x);
+        }
+        (void)rcreg;
       }
-      __syncthreads();   /* LDS reused by the next chunk */
+
+      /* pass B: aggregates in groups of AGRP */
+      for (int g = 0; g < nagg_grps; g++) {
+        double sums[NSLOTS][AGRP];
+#pragma unroll
+        for (int s = 0; s < NSLOTS; s++)
+#pragma unroll
+          for (int j = 0; j < AGRP; j++) sums[s][j] = 0.0;
+
+#pragma unroll 2
+        for (int k = 0; k < CHUNK / WG; k++) {
+          int r = tid + k * WG;
+          int alive = (int)((salive[r >> 6] >> (r & 63)) & 1ull);
+          if (__popcll(__ballot(alive)) == 0) continue;
+          int slot = sslot[r];
+          double av[AGRP];
+#pragma unroll
+          for (int j = 0; j < AGRP; j++) {
+            int a = g * AGRP + j;
+            int anull;
+            av[j] = (a < naggs)
+                ? eval_agg(plan, a, sval, r, clean, svalid, &anull) : 0.0;
+            if (a < naggs && !clean && anull) av[j] = 0.0;  /* non-null enforced */
+          }
+#pragma unroll
+          for (int s = 0; s < NSLOTS; s++) {
+            const int m = alive && slot == s;
+#pragma unroll
+            for (int j = 0; j < AGRP; j++)
+              sums[s][j] += m ? av[j] : 0.0;
+          }
+        }
+        /* wave-reduce into the block accumulator */
+#pragma unroll
+        for (int s = 0; s < NSLOTS; s++)
+#pragma unroll
+          for (int j = 0; j < AGRP; j++) {
+            int a = g * AGRP + j;
+            if (a >= naggs) continue;
+            double x = wave_sum(sums[s][j]);
+            if ((tid & 63) == 0 && x != 0.0)
+              atomicAdd(&bacc[s * (naggs + 1) + a], x);
+          }
+      }
+      __syncthreads();
     }
   }
 
-  /* block reduce: wave shuffle reduce, then one global atomic per value per
-   * wave (Guideline 12) */
-  const int STRIDE = 2 * NAGGS + 1;
-#pragma unroll
-  for (int s = 0; s < NSLOTS; s++) {
-#pragma unroll
-    for (int a = 0; a < NAGGS + 1; a++) {
-      double x = (a < NAGGS) ? sums[s][a] : rc[s];
-#pragma unroll
-      for (int off = 32; off > 0; off >>= 1)
-        x += __shfl_down(x, off, 64);
-      if ((tid & 63) == 0 && x != 0.0)
-        atomicAdd(&out[s * STRIDE + (a < NAGGS ? a : 2 * NAGGS)], x);
-    }
+  /* flush block accumulator to global (one atomic per value per block) */
+  __syncthreads();
+  for (int i = tid; i < NSLOTS * (naggs + 1); i += WG) {
+    int s = i / (naggs + 1), a = i % (naggs + 1);
+    double x = bacc[i];
+    if (x != 0.0)
+      atomicAdd(&out[s * out_stride + (a < naggs ? a : out_stride - 1)], x);
   }
-  if (NSLOTS == 1) {
-#pragma unroll
-    for (int a = 0; a < NAGGS; a++) {
-      double x = cnts[a];
-#pragma unroll
-      for (int off = 32; off > 0; off >>= 1)
-        x += __shfl_down(x, off, 64);
-      if ((tid & 63) == 0 && x != 0.0)
-        atomicAdd(&out[NAGGS + a], x);
-    }
-  }
+  (void)wid;
 }
 
 extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
@@ -334,19 +498,34 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
   size_t lds = (size_t)plan->nused * CHUNK * 8 +
                (size_t)plan->nused * (CHUNK / 64) * 8 + (CHUNK / 64) * 8 + 16;
   hipError_t err;
-#define LAUNCH(S, A)                                                        \
-  hipLaunchKernelGGL((k_scan_agg<S, A>), dim3(grid), dim3(WG), lds, s,      \
-                     *plan, dev_batches, dev_tiles, ntiles, dev_out);
   if (ns <= 1) {
-    if (na <= 2) { LAUNCH(1, 2) }
-    else if (na <= 4) { LAUNCH(1, 4) }
-    else { LAUNCH(1, 12) }
-  } else if (ns <= 4 && na <= 4) { LAUNCH(4, 4) }
-  else if (ns <= 8 && na <= 8) { LAUNCH(8, 8) }
-  else if (ns <= 16 && na <= 4) { LAUNCH(16, 4) }
-  else if (ns <= 16 && na <= 8) { LAUNCH(16, 8) }
-  else return (int)hipErrorInvalidValue;
-#undef LAUNCH
+    if (na <= 2) {
+      hipLaunchKernelGGL((k_keyless<2>), dim3(grid), dim3(WG), lds, s,
+                         *plan, dev_batches, dev_tiles, ntiles, dev_out);
+    } else if (na <= 4) {
+      hipLaunchKernelGGL((k_keyless<4>), dim3(grid), dim3(WG), lds, s,
+                         *plan, dev_batches, dev_tiles, ntiles, dev_out);
+    } else {
+      hipLaunchKernelGGL((k_keyless<12>), dim3(grid), dim3(WG), lds, s,
+                         *plan, dev_batches, dev_tiles, ntiles, dev_out);
+    }
+  } else {
+    /* grouped: extra LDS for salive + slot array + block accumulator */
+    lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)ns * (na + 1) * 8 + 16;
+    int out_stride = 2 * (na <= 2 ? 2 : na <= 4 ? 4 : na <= 8 ? 8 : 12) + 1;
+    if (ns <= 4) {
+      hipLaunchKernelGGL((k_grouped<4>), dim3(grid), dim3(WG), lds, s,
+                         *plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
+    } else if (ns <= 8) {
+      hipLaunchKernelGGL((k_grouped<8>), dim3(grid), dim3(WG), lds, s,
+                         *plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
+    } else if (ns <= 16) {
+      hipLaunchKernelGGL((k_grouped<16>), dim3(grid), dim3(WG), lds, s,
+                         *plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
+    } else {
+      return (int)hipErrorInvalidValue;
+    }
+  }
   err = hipGetLastError();
   return (int)err;
 }
