@@ -81,7 +81,8 @@ typedef struct {
   sn_dev_agg  aggs[12];
 } sn_dev_plan;
 
-#define SN_TILE_ROWS 4096      /* rows per workgroup tile (256 thr x 16) */
+#define SN_TILE_ROWS 16384     /* rows per workgroup tile (16 LDS chunks;
+                                  long pipelines for the staged conversion) */
 
 #ifdef __cplusplus
 extern "C" {

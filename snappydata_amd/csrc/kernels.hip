@@ -227,6 +227,123 @@ __device__ __forceinline__ void convert_chunk(
   }
 }
 
+/* ---- staged (software-pipelined) conversion for clean full chunks ----
+ * T14 async-STAGE split: the NEXT chunk's global loads are issued into
+ * registers right after the barrier that frees LDS, so they stay in flight
+ * underneath the current chunk's row phase (the single biggest lever here:
+ * SQ_WAIT_ANY was 75% of wave cycles with the unstaged conversion).
+ * Raw loads are width-based (8/4/2 B); kind conversion happens at the LDS
+ * write.  I64 shares the w8 path (its LDS image is the raw bits). */
+/* one shared register buffer for every width class (64 VGPRs total): raw
+ * bits packed into double2 lanes; row mapping is pair-based for every width
+ * (pair h = tid + p*WG covers rows 2h, 2h+1) */
+struct Stage {
+  double2_t buf[SN_DEV_MAX_COLS][CHUNK / (2 * WG)];
+};
+__device__ __forceinline__ double pack_u64(unsigned lo, unsigned hi) {
+  return __longlong_as_double(((unsigned long long)hi << 32) | lo);
+}
+
+__device__ __forceinline__ int col_width_class(int kind) {
+  switch (kind) {
+    case SN_K_F64: case SN_K_I64: return 8;
+    case SN_K_I32: case SN_K_F32: case SN_K_DICT32: return 4;
+    case SN_K_I16: case SN_K_DICT16: return 2;
+    default: return 0;   /* BOOLBIT: not stageable */
+  }
+}
+
+__device__ __forceinline__ void stage_load(const sn_dev_batch &b, int nused,
+                                           int base, Stage &st) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int c = 0; c < SN_DEV_MAX_COLS; c++) {
+    if (c >= nused) break;
+    const sn_dev_col &col = b.cols[c];
+    const int w = col_width_class(col.kind);
+    if (w == 8) {
+      const double2_t *s2 = (const double2_t *)((const char *)col.body + (size_t)base * 8);
+#pragma unroll
+      for (int p = 0; p < CHUNK / (2 * WG); p++) st.buf[c][p] = s2[tid + p * WG];
+    } else if (w == 4) {
+      /* two int2 pair-loads -> raw bits in buf[c][0].x / .y */
+      const int2_t *s2 = (const int2_t *)((const char *)col.body + (size_t)base * 4);
+      int2_t a0 = s2[tid], a1 = s2[tid + WG];
+      st.buf[c][0].x = *(double *)&a0;
+      st.buf[c][0].y = *(double *)&a1;
+    } else {
+      /* two short2 (4 B) pair-loads -> packed into buf[c][0].x */
+      const unsigned *s2 = (const unsigned *)((const char *)col.body + (size_t)base * 2);
+      unsigned a0 = s2[tid], a1 = s2[tid + WG];
+      st.buf[c][0].x = pack_u64(a0, a1);
+    }
+  }
+}
+
+__device__ __forceinline__ void stage_write(const sn_dev_batch &b, int nused,
+                                            Stage &st, double *sval) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int c = 0; c < SN_DEV_MAX_COLS; c++) {
+    if (c >= nused) break;
+    const sn_dev_col &col = b.cols[c];
+    double *dst = sval + (size_t)c * CHUNK;
+    switch (col.kind) {
+      case SN_K_F64: case SN_K_I64:
+#pragma unroll
+        for (int p = 0; p < CHUNK / (2 * WG); p++)
+          ((double2_t *)dst)[tid + p * WG] = st.buf[c][p];
+        break;
+      case SN_K_I32: case SN_K_F32: case SN_K_DICT32: {
+        const int32_t *dm = col.dictmap;
+#pragma unroll
+        for (int p = 0; p < CHUNK / (2 * WG); p++) {
+          double raw = p == 0 ? st.buf[c][0].x : st.buf[c][0].y;
+          int2_t a = *(int2_t *)&raw;
+          double2_t y;
+          if (col.kind == SN_K_I32) { y.x = (double)a.x; y.y = (double)a.y; }
+          else if (col.kind == SN_K_F32) {
+            float2 f = *(float2 *)&raw;
+            y.x = (double)f.x; y.y = (double)f.y;
+          } else { y.x = (double)dm[a.x]; y.y = (double)dm[a.y]; }
+          ((double2_t *)dst)[tid + p * WG] = y;
+        }
+        break;
+      }
+      case SN_K_I16: case SN_K_DICT16: {
+        const int32_t *dm = col.dictmap;
+        unsigned long long raw = __double_as_longlong(st.buf[c][0].x);
+#pragma unroll
+        for (int p = 0; p < CHUNK / (2 * WG); p++) {
+          unsigned half = (unsigned)(raw >> (32 * p));
+          int16_t e0 = (int16_t)(half & 0xffff), e1 = (int16_t)(half >> 16);
+          double2_t y;
+          if (col.kind == SN_K_I16) { y.x = (double)e0; y.y = (double)e1; }
+          else {
+            y.x = (double)dm[(int)(uint16_t)e0];
+            y.y = (double)dm[(int)(uint16_t)e1];
+          }
+          ((double2_t *)dst)[tid + p * WG] = y;
+        }
+        break;
+      }
+      default: break;
+    }
+  }
+}
+
+/* can this batch use the staged pipeline? (clean + every col stageable) */
+__device__ __forceinline__ int batch_stageable(const sn_dev_batch &b, int nused) {
+  if (!b.clean) return 0;
+  int ok = 1;
+#pragma unroll
+  for (int c = 0; c < SN_DEV_MAX_COLS; c++) {
+    if (c >= nused) break;
+    ok &= col_width_class(b.cols[c].kind) != 0;
+  }
+  return ok;
+}
+
 /* predicate evaluation for row r of the chunk (LDS image) */
 __device__ __forceinline__ int eval_preds(const sn_dev_plan &plan, int clean,
                                           const double *sval,
@@ -302,17 +419,29 @@ __global__ void k_keyless(sn_dev_plan plan,
   for (int a = 0; a < NAGGS; a++) { sums[a] = 0.0; cnts[a] = 0.0; }
   const int naggs = plan.naggs;
 
+  Stage st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
     const int num_rows = b.num_rows;
     const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
     const int clean = b.clean;
+    const int pipe = batch_stageable(b, nused);
 
+    int staged = 0;
+    if (pipe && tile.row_start + CHUNK <= tile_end) {
+      stage_load(b, nused, tile.row_start, st);
+      staged = 1;
+    }
     for (int base = tile.row_start; base < tile_end; base += CHUNK) {
       const int rows = min(CHUNK, tile_end - base);
-      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      if (staged) stage_write(b, nused, st, sval);
+      else convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
       __syncthreads();
+      /* prefetch the NEXT full chunk under this chunk's row phase */
+      const int nbase = base + CHUNK;
+      const int next_staged = pipe && nbase + CHUNK <= tile_end;
+      if (next_staged) stage_load(b, nused, nbase, st);
 
 #pragma unroll 2
       for (int k = 0; k < CHUNK / WG; k++) {
@@ -332,6 +461,7 @@ __global__ void k_keyless(sn_dev_plan plan,
         rcnt += alive ? 1.0 : 0.0;
       }
       __syncthreads();
+      staged = next_staged;
     }
   }
 
@@ -377,17 +507,28 @@ __global__ void k_grouped(sn_dev_plan plan,
   for (int i = tid; i < NSLOTS * (naggs + 1); i += WG) bacc[i] = 0.0;
   __syncthreads();
 
+  Stage st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
     const int num_rows = b.num_rows;
     const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
     const int clean = b.clean;
+    const int pipe = batch_stageable(b, nused);
 
+    int staged = 0;
+    if (pipe && tile.row_start + CHUNK <= tile_end) {
+      stage_load(b, nused, tile.row_start, st);
+      staged = 1;
+    }
     for (int base = tile.row_start; base < tile_end; base += CHUNK) {
       const int rows = min(CHUNK, tile_end - base);
-      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      if (staged) stage_write(b, nused, st, sval);
+      else convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
       __syncthreads();
+      const int nbase = base + CHUNK;
+      const int next_staged = pipe && nbase + CHUNK <= tile_end;
+      if (next_staged) stage_load(b, nused, nbase, st);
 
       /* pass A: predicates + slot once per row */
       double rcreg = 0.0;
@@ -474,6 +615,7 @@ __global__ void k_grouped(sn_dev_plan plan,
           }
       }
       __syncthreads();
+      staged = next_staged;
     }
   }
 
