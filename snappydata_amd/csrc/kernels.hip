@@ -34,18 +34,30 @@
 typedef double double2_t __attribute__((ext_vector_type(2)));
 typedef int int2_t __attribute__((ext_vector_type(2)));
 
+/* Global-address-space pointer cast.  Pointers reaching the kernel through
+ * descriptor structs are generic, so plain derefs lower to flat_load_* —
+ * and flat loads tick BOTH vmcnt and lgkmcnt, so every LDS wait drains all
+ * in-flight global loads (measured: the whole prefetch pipeline collapsed).
+ * Casting to address_space(1) lowers them to global_load_* (vmcnt only). */
+#define GAS __attribute__((address_space(1)))
+template <typename T>
+__device__ __forceinline__ const GAS T *as_global(const T *p) {
+  return (const GAS T *)(unsigned long long)(uintptr_t)p;
+}
+
 __device__ __forceinline__ int bm_get(const uint64_t *bm, int row) {
-  return (int)((bm[row >> 6] >> (row & 63)) & 1ull);
+  return (int)((as_global(bm)[row >> 6] >> (row & 63)) & 1ull);
 }
 
 __device__ __forceinline__ int nonnull_pos(const uint64_t *nullw,
                                            const uint32_t *pfx, int row) {
-  uint64_t w = nullw[row >> 6];
+  uint64_t w = as_global(nullw)[row >> 6];
   uint64_t mask = (1ull << (row & 63)) - 1ull;
-  return row - (int)(pfx[row >> 6] + __popcll(w & mask));
+  return row - (int)(as_global(pfx)[row >> 6] + __popcll(w & mask));
 }
 
-__device__ __forceinline__ int patch_find(const int32_t *pos, int n, int row) {
+__device__ __forceinline__ int patch_find(const int32_t *pos_, int n, int row) {
+  const GAS int32_t *pos = as_global(pos_);
   int lo = 0, hi = n - 1;
   while (lo <= hi) {
     int mid = (lo + hi) >> 1;
@@ -63,7 +75,7 @@ __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
     int pi = patch_find(c.patch_pos, c.patch_n, row);
     if (pi >= 0) {
       if (c.patch_nullbm && bm_get(c.patch_nullbm, pi)) return 0;
-      double pv = c.patch_val[pi];
+      double pv = as_global(c.patch_val)[pi];
       if (c.kind == SN_K_DICT16 || c.kind == SN_K_DICT32) {
         *gid = (int)__double2ll_rn(pv);
         return 1;
@@ -79,13 +91,13 @@ __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
     nnp = nonnull_pos(c.nullw, c.nullpfx, row);
   }
   switch (c.kind) {
-    case SN_K_F64: *vd = ((const double *)c.body)[nnp]; *vi = (long long)*vd; break;
-    case SN_K_I32: *vi = ((const int32_t *)c.body)[nnp]; *vd = (double)*vi; break;
-    case SN_K_I64: *vi = ((const long long *)c.body)[nnp]; *vd = (double)*vi; break;
-    case SN_K_F32: *vd = ((const float *)c.body)[nnp]; *vi = (long long)*vd; break;
-    case SN_K_I16: *vi = ((const int16_t *)c.body)[nnp]; *vd = (double)*vi; break;
-    case SN_K_DICT16: *gid = c.dictmap[(int)(uint16_t)((const int16_t *)c.body)[nnp]]; break;
-    case SN_K_DICT32: *gid = c.dictmap[((const int32_t *)c.body)[nnp]]; break;
+    case SN_K_F64: *vd = as_global((const double *)c.body)[nnp]; *vi = (long long)*vd; break;
+    case SN_K_I32: *vi = as_global((const int32_t *)c.body)[nnp]; *vd = (double)*vi; break;
+    case SN_K_I64: *vi = as_global((const long long *)c.body)[nnp]; *vd = (double)*vi; break;
+    case SN_K_F32: *vd = as_global((const float *)c.body)[nnp]; *vi = (long long)*vd; break;
+    case SN_K_I16: *vi = as_global((const int16_t *)c.body)[nnp]; *vd = (double)*vi; break;
+    case SN_K_DICT16: *gid = as_global(c.dictmap)[(int)(uint16_t)as_global((const int16_t *)c.body)[nnp]]; break;
+    case SN_K_DICT32: *gid = as_global(c.dictmap)[as_global((const int32_t *)c.body)[nnp]]; break;
     case SN_K_BOOLBIT: *vi = bm_get((const uint64_t *)c.body, nnp); *vd = (double)*vi; break;
   }
   return 1;
@@ -105,12 +117,12 @@ __device__ __forceinline__ void convert_chunk(
     if (clean) {
       switch (col.kind) {
         case SN_K_F64: {
-          const double *src = (const double *)col.body + base;
+          const GAS double *src = as_global((const double *)col.body) + base;
           if (rows == CHUNK) {
 #pragma unroll
             for (int k = 0; k < CHUNK / (2 * WG); k++) {
               int h = tid + k * WG;   /* pair index */
-              ((double2_t *)dst)[h] = ((const double2_t *)src)[h];
+              ((double2_t *)dst)[h] = ((const GAS double2_t *)src)[h];
             }
           } else {
 #pragma unroll
@@ -122,12 +134,12 @@ __device__ __forceinline__ void convert_chunk(
           break;
         }
         case SN_K_I32: {
-          const int32_t *src = (const int32_t *)col.body + base;
+          const GAS int32_t *src = as_global((const int32_t *)col.body) + base;
           if (rows == CHUNK) {
 #pragma unroll
             for (int k = 0; k < CHUNK / (2 * WG); k++) {
               int h = tid + k * WG;
-              int2_t x = ((const int2_t *)src)[h];
+              int2_t x = ((const GAS int2_t *)src)[h];
               double2_t y; y.x = (double)x.x; y.y = (double)x.y;
               ((double2_t *)dst)[h] = y;
             }
@@ -141,7 +153,7 @@ __device__ __forceinline__ void convert_chunk(
           break;
         }
         case SN_K_I64: {
-          const long long *src = (const long long *)col.body + base;
+          const GAS long long *src = as_global((const long long *)col.body) + base;
 #pragma unroll
           for (int k = 0; k < CHUNK / WG; k++) {
             int r = tid + k * WG;
@@ -150,7 +162,7 @@ __device__ __forceinline__ void convert_chunk(
           break;
         }
         case SN_K_F32: {
-          const float *src = (const float *)col.body + base;
+          const GAS float *src = as_global((const float *)col.body) + base;
 #pragma unroll
           for (int k = 0; k < CHUNK / WG; k++) {
             int r = tid + k * WG;
@@ -159,7 +171,7 @@ __device__ __forceinline__ void convert_chunk(
           break;
         }
         case SN_K_I16: {
-          const int16_t *src = (const int16_t *)col.body + base;
+          const GAS int16_t *src = as_global((const int16_t *)col.body) + base;
 #pragma unroll
           for (int k = 0; k < CHUNK / WG; k++) {
             int r = tid + k * WG;
@@ -168,8 +180,8 @@ __device__ __forceinline__ void convert_chunk(
           break;
         }
         case SN_K_DICT16: {
-          const int16_t *src = (const int16_t *)col.body + base;
-          const int32_t *dm = col.dictmap;
+          const GAS int16_t *src = as_global((const int16_t *)col.body) + base;
+          const GAS int32_t *dm = as_global(col.dictmap);
 #pragma unroll
           for (int k = 0; k < CHUNK / WG; k++) {
             int r = tid + k * WG;
@@ -178,8 +190,8 @@ __device__ __forceinline__ void convert_chunk(
           break;
         }
         case SN_K_DICT32: {
-          const int32_t *src = (const int32_t *)col.body + base;
-          const int32_t *dm = col.dictmap;
+          const GAS int32_t *src = as_global((const int32_t *)col.body) + base;
+          const GAS int32_t *dm = as_global(col.dictmap);
 #pragma unroll
           for (int k = 0; k < CHUNK / WG; k++) {
             int r = tid + k * WG;
@@ -188,7 +200,7 @@ __device__ __forceinline__ void convert_chunk(
           break;
         }
         case SN_K_BOOLBIT: {
-          const uint64_t *src = (const uint64_t *)col.body;
+          const uint64_t *src = (const uint64_t *)col.body;  /* via bm_get */
 #pragma unroll
           for (int k = 0; k < CHUNK / WG; k++) {
             int r = tid + k * WG;
@@ -262,18 +274,21 @@ __device__ __forceinline__ void stage_load(const sn_dev_batch &b, int nused,
     const sn_dev_col &col = b.cols[c];
     const int w = col_width_class(col.kind);
     if (w == 8) {
-      const double2_t *s2 = (const double2_t *)((const char *)col.body + (size_t)base * 8);
+      const GAS double2_t *s2 = (const GAS double2_t *)
+          as_global((const char *)col.body + (size_t)base * 8);
 #pragma unroll
       for (int p = 0; p < CHUNK / (2 * WG); p++) st.buf[c][p] = s2[tid + p * WG];
     } else if (w == 4) {
       /* two int2 pair-loads -> raw bits in buf[c][0].x / .y */
-      const int2_t *s2 = (const int2_t *)((const char *)col.body + (size_t)base * 4);
+      const GAS int2_t *s2 = (const GAS int2_t *)
+          as_global((const char *)col.body + (size_t)base * 4);
       int2_t a0 = s2[tid], a1 = s2[tid + WG];
       st.buf[c][0].x = *(double *)&a0;
       st.buf[c][0].y = *(double *)&a1;
     } else {
       /* two short2 (4 B) pair-loads -> packed into buf[c][0].x */
-      const unsigned *s2 = (const unsigned *)((const char *)col.body + (size_t)base * 2);
+      const GAS unsigned *s2 = (const GAS unsigned *)
+          as_global((const char *)col.body + (size_t)base * 2);
       unsigned a0 = s2[tid], a1 = s2[tid + WG];
       st.buf[c][0].x = pack_u64(a0, a1);
     }
@@ -295,7 +310,7 @@ __device__ __forceinline__ void stage_write(const sn_dev_batch &b, int nused,
           ((double2_t *)dst)[tid + p * WG] = st.buf[c][p];
         break;
       case SN_K_I32: case SN_K_F32: case SN_K_DICT32: {
-        const int32_t *dm = col.dictmap;
+        const GAS int32_t *dm = as_global(col.dictmap);
 #pragma unroll
         for (int p = 0; p < CHUNK / (2 * WG); p++) {
           double raw = p == 0 ? st.buf[c][0].x : st.buf[c][0].y;
@@ -311,7 +326,7 @@ __device__ __forceinline__ void stage_write(const sn_dev_batch &b, int nused,
         break;
       }
       case SN_K_I16: case SN_K_DICT16: {
-        const int32_t *dm = col.dictmap;
+        const GAS int32_t *dm = as_global(col.dictmap);
         unsigned long long raw = __double_as_longlong(st.buf[c][0].x);
 #pragma unroll
         for (int p = 0; p < CHUNK / (2 * WG); p++) {
