@@ -34,6 +34,7 @@
 
 #include <algorithm>
 #include <atomic>
+#include <cmath>
 #include <cstdarg>
 #include <cstdio>
 #include <cstring>
@@ -753,10 +754,12 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   q->na_t = template_naggs(plan->naggs);
   q->out_stride = 2 * (size_t)q->na_t + 1;
 
-  /* build device plan */
+  /* build device plan — canonical branchless forms:
+   * predicates as closed intervals (strictness folded via nextafter / +-1,
+   * missing bounds as +-inf), aggregates as three neutral-padded factors */
   sn_dev_plan dp;
   memset(&dp, 0, sizeof(dp));
-  dp.npreds = plan->npreds; dp.naggs = plan->naggs;
+  dp.naggs = plan->naggs;
   dp.ngroup = plan->ngroup; dp.nslots = q->nslots;
   dp.nused = (int32_t)q->used_cols.size();
   dp.i64_mask = 0;
@@ -766,28 +769,51 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   for (int i = 0; i < plan->ngroup; i++) dp.gcol[i] = q->cslot_of_col[plan->group_cols[i]];
   for (int i = 0; i < plan->npreds; i++) {
     const sn_pred &s = plan->preds[i];
-    sn_dev_pred &d = dp.preds[i];
     sn_type_t dt = t->schema[s.col].dtype;
-    d.cslot = q->cslot_of_col[s.col];
-    d.is_double = (dt == SN_TYPE_DOUBLE || dt == SN_TYPE_FLOAT);
-    d.lo_d = s.lo_d; d.hi_d = s.hi_d; d.lo_i = s.lo_i; d.hi_i = s.hi_i;
-    /* the kernel's LDS image is f64 for every non-INT64 column, so integer
-     * bounds must be widened (exact for int32/int16); INT64 stays raw and
-     * compares through lo_i/hi_i */
-    if (!d.is_double && dt != SN_TYPE_INT64) {
-      d.lo_d = (double)s.lo_i;
-      d.hi_d = (double)s.hi_i;
+    int cslot = q->cslot_of_col[s.col];
+    if (dt == SN_TYPE_INT64) {
+      if (dp.npreds_i >= 4) return (fail(SN_ERR_BADARG, "too many int64 preds"), nullptr);
+      sn_dev_pred_i &d = dp.preds_i[dp.npreds_i++];
+      d.cslot = cslot;
+      d.lo = s.has_lo ? (s.lo_strict && s.lo_i < INT64_MAX ? s.lo_i + 1 : s.lo_i)
+                      : INT64_MIN;
+      d.hi = s.has_hi ? (s.hi_strict && s.hi_i > INT64_MIN ? s.hi_i - 1 : s.hi_i)
+                      : INT64_MAX;
+    } else {
+      bool is_d = (dt == SN_TYPE_DOUBLE || dt == SN_TYPE_FLOAT);
+      sn_dev_pred_d &d = dp.preds_d[dp.npreds_d++];
+      d.cslot = cslot;
+      double lo = s.has_lo ? (is_d ? s.lo_d : (double)s.lo_i) : -INFINITY;
+      double hi = s.has_hi ? (is_d ? s.hi_d : (double)s.hi_i) : INFINITY;
+      if (s.has_lo && s.lo_strict) lo = std::nextafter(lo, INFINITY);
+      if (s.has_hi && s.hi_strict) hi = std::nextafter(hi, -INFINITY);
+      d.lo = lo; d.hi = hi;
     }
-    d.has_lo = s.has_lo; d.has_hi = s.has_hi;
-    d.lo_strict = s.lo_strict; d.hi_strict = s.hi_strict;
   }
   for (int a = 0; a < plan->naggs; a++) {
-    dp.aggs[a].kind = plan->aggs[a].kind == SN_AGG_COUNT_STAR ? 1 : 0;
-    dp.aggs[a].nf = plan->aggs[a].nfactors;
-    for (int f = 0; f < plan->aggs[a].nfactors; f++) {
-      dp.aggs[a].f[f].cslot = q->cslot_of_col[plan->aggs[a].factors[f].col];
-      dp.aggs[a].f[f].add = plan->aggs[a].factors[f].add;
-      dp.aggs[a].f[f].mul = plan->aggs[a].factors[f].mul;
+    sn_dev_agg &da = dp.aggs[a];
+    da.a0 = da.a1 = da.a2 = 1.0;
+    da.m0 = da.m1 = da.m2 = 0.0;
+    da.c0 = da.c1 = da.c2 = 0;
+    da.nf = 0;
+    if (plan->aggs[a].kind == SN_AGG_COUNT_STAR) continue;
+    da.nf = plan->aggs[a].nfactors;
+    const sn_agg &sa = plan->aggs[a];
+    if (sa.nfactors >= 1) {
+      da.c0 = q->cslot_of_col[sa.factors[0].col];
+      da.a0 = sa.factors[0].add; da.m0 = sa.factors[0].mul;
+      if (dp.i64_mask & (1u << da.c0)) {
+        fail(SN_ERR_UNSUPPORTED, "int64 aggregate factors not in round-1 GPU path");
+        return nullptr;
+      }
+    }
+    if (sa.nfactors >= 2) {
+      da.c1 = q->cslot_of_col[sa.factors[1].col];
+      da.a1 = sa.factors[1].add; da.m1 = sa.factors[1].mul;
+    }
+    if (sa.nfactors >= 3) {
+      da.c2 = q->cslot_of_col[sa.factors[2].col];
+      da.a2 = sa.factors[2].add; da.m2 = sa.factors[2].mul;
     }
   }
 
@@ -909,10 +935,16 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     }
     /* HIP events bracket the scan kernel on ITS stream for the roofline leg
      * (torch.cuda.Event would only see torch's current stream) */
+    void *dp_dev = e->arena.alloc(sizeof(dp));
+    if (!dp_dev ||
+        hipMemcpy(dp_dev, &dp, sizeof(dp), hipMemcpyHostToDevice) != hipSuccess) {
+      fail(SN_ERR_NOMEM, "plan upload"); return nullptr;
+    }
     (void)hipEventCreate(&q->ev_start);
     (void)hipEventCreate(&q->ev_stop);
     if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
-    int rc = sn_launch_scan_agg(&dp, (const sn_dev_batch *)db_dev,
+    int rc = sn_launch_scan_agg(&dp, (const sn_dev_plan *)dp_dev,
+                                (const sn_dev_batch *)db_dev,
                                 (const sn_dev_tile *)tl_dev, (int32_t)htiles.size(),
                                 q->dev_out, e->stream);
     if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);

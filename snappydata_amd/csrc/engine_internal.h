@@ -60,25 +60,32 @@ typedef struct {
 
 typedef struct { int32_t batch; int32_t row_start; } sn_dev_tile;
 
-typedef struct {
-  int32_t cslot;               /* index into sn_dev_batch.cols */
-  int32_t is_double;
-  double  lo_d, hi_d;
-  int64_t lo_i, hi_i;
-  int32_t has_lo, has_hi, lo_strict, hi_strict;
-} sn_dev_pred;
+/* Canonical (branchless) predicate: alive &= (lo <= x && x <= hi).
+ * Strictness and missing bounds are folded by the host (nextafter for
+ * strict double bounds, +-1 for strict integer bounds, +-inf when absent),
+ * so the kernel does exactly two compares per predicate. */
+typedef struct { double lo, hi; int32_t cslot; int32_t _p; } sn_dev_pred_d;
+typedef struct { int64_t lo, hi; int32_t cslot; int32_t _p; } sn_dev_pred_i;
 
-typedef struct { int32_t cslot; double add, mul; } sn_dev_factor;
-typedef struct { int32_t kind; int32_t nf; sn_dev_factor f[3]; } sn_dev_agg;
+/* Canonical aggregate input: value = (a0+m0*x0)*(a1+m1*x1)*(a2+m2*x2) with
+ * unused factors neutralized to (1 + 0*x_c0) — three fmas, no branches.
+ * COUNT(*) is all-neutral (value 1).  nf/c-slots retained for the general
+ * path's per-factor null checks. */
+typedef struct {
+  double a0, m0, a1, m1, a2, m2;
+  int32_t c0, c1, c2;
+  int32_t nf;                  /* real factors (0 for COUNT(*)) */
+} sn_dev_agg;
 
 typedef struct {
-  int32_t npreds, naggs, ngroup, nslots;
-  int32_t nused;               /* referenced columns (cslots 0..nused-1) */
-  uint32_t i64_mask;           /* bit c: cslot c is INT64 (raw-bitcast in LDS;
-                                  predicates compare exactly as int64) */
-  int32_t gcol[2];             /* cslot of group columns (ngroup entries) */
-  sn_dev_pred preds[8];
-  sn_dev_agg  aggs[12];
+  int32_t npreds_d, npreds_i, naggs, ngroup;
+  int32_t nslots, nused;
+  uint32_t i64_mask;           /* bit c: cslot c is INT64 (raw bits in LDS) */
+  int32_t gcol[2];
+  int32_t _pad;
+  sn_dev_pred_d preds_d[8];
+  sn_dev_pred_i preds_i[4];
+  sn_dev_agg aggs[12];
 } sn_dev_plan;
 
 #define SN_TILE_ROWS 16384     /* rows per workgroup tile (16 LDS chunks;
@@ -92,6 +99,7 @@ extern "C" {
  * out: device array [nslots][naggs+1] doubles (last = group row count),
  * zeroed by caller.  Returns hipError_t as int. */
 int sn_launch_scan_agg(const sn_dev_plan *plan,
+                       const sn_dev_plan *dev_plan,  /* device copy (LDS mirror source) */
                        const sn_dev_batch *dev_batches,
                        const sn_dev_tile *dev_tiles, int32_t ntiles,
                        double *dev_out, void *stream);

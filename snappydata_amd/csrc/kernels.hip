@@ -359,52 +359,56 @@ __device__ __forceinline__ int batch_stageable(const sn_dev_batch &b, int nused)
   return ok;
 }
 
-/* predicate evaluation for row r of the chunk (LDS image) */
-__device__ __forceinline__ int eval_preds(const sn_dev_plan &plan, int clean,
-                                          const double *sval,
+/* predicate evaluation for row r (branchless; plan mirrored in LDS) */
+__device__ __forceinline__ int eval_preds(const sn_dev_plan *P, int npd, int npi,
+                                          int clean, const double *sval,
                                           const uint64_t *svalid,
                                           const uint64_t *sdead, int r) {
   int alive = clean ? 1 : !((sdead[r >> 6] >> (r & 63)) & 1ull);
 #pragma unroll
   for (int i = 0; i < 8; i++) {
-    if (i >= plan.npreds) break;
-    const sn_dev_pred &p = plan.preds[i];
-    const int cs = p.cslot;
+    if (i >= npd) break;
+    const int cs = P->preds_d[i].cslot;
+    const double x = sval[(size_t)cs * CHUNK + r];
+    alive &= (x >= P->preds_d[i].lo) & (x <= P->preds_d[i].hi);
     if (!clean)
       alive &= (int)((svalid[(size_t)cs * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
-    const double xv = sval[(size_t)cs * CHUNK + r];
-    if ((plan.i64_mask >> cs) & 1u) {
-      long long x = __double_as_longlong(xv);
-      if (p.has_lo && (p.lo_strict ? !(x > p.lo_i) : !(x >= p.lo_i))) alive = 0;
-      if (p.has_hi && (p.hi_strict ? !(x < p.hi_i) : !(x <= p.hi_i))) alive = 0;
-    } else {
-      if (p.has_lo && (p.lo_strict ? !(xv > p.lo_d) : !(xv >= p.lo_d))) alive = 0;
-      if (p.has_hi && (p.hi_strict ? !(xv < p.hi_d) : !(xv <= p.hi_d))) alive = 0;
-    }
+  }
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    if (i >= npi) break;
+    const int cs = P->preds_i[i].cslot;
+    const long long x = __double_as_longlong(sval[(size_t)cs * CHUNK + r]);
+    alive &= (x >= P->preds_i[i].lo) & (x <= P->preds_i[i].hi);
+    if (!clean)
+      alive &= (int)((svalid[(size_t)cs * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
   }
   return alive;
 }
 
-/* aggregate input value for agg a, row r (non-null factors assumed checked) */
-__device__ __forceinline__ double eval_agg(const sn_dev_plan &plan, int a,
-                                           const double *sval, int r,
-                                           int clean, const uint64_t *svalid,
-                                           int *anull) {
-  const sn_dev_agg &ag = plan.aggs[a];
-  double aval = 1.0;
+/* aggregate input value: three neutral-padded fmas, no branches */
+__device__ __forceinline__ double eval_agg_clean(const sn_dev_plan *P, int a,
+                                                 const double *sval, int r) {
+  const sn_dev_agg &A = P->aggs[a];
+  const double x0 = sval[(size_t)A.c0 * CHUNK + r];
+  const double x1 = sval[(size_t)A.c1 * CHUNK + r];
+  const double x2 = sval[(size_t)A.c2 * CHUNK + r];
+  return (A.a0 + A.m0 * x0) * (A.a1 + A.m1 * x1) * (A.a2 + A.m2 * x2);
+}
+
+__device__ __forceinline__ double eval_agg_general(const sn_dev_plan *P, int a,
+                                                   const double *sval, int r,
+                                                   const uint64_t *svalid,
+                                                   int *anull) {
+  const sn_dev_agg &A = P->aggs[a];
   *anull = 0;
-  if (ag.kind == 1) return 1.0;   /* COUNT(*) */
-#pragma unroll
-  for (int j = 0; j < 3; j++) {
-    if (j >= ag.nf) break;
-    const int fc = ag.f[j].cslot;
-    if (!clean)
-      *anull |= !(int)((svalid[(size_t)fc * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
-    double x = sval[(size_t)fc * CHUNK + r];
-    if ((plan.i64_mask >> fc) & 1u) x = (double)__double_as_longlong(x);
-    aval = aval * (ag.f[j].add + ag.f[j].mul * x);
-  }
-  return aval;
+  if (A.nf >= 1)
+    *anull |= !(int)((svalid[(size_t)A.c0 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
+  if (A.nf >= 2)
+    *anull |= !(int)((svalid[(size_t)A.c1 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
+  if (A.nf >= 3)
+    *anull |= !(int)((svalid[(size_t)A.c2 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
+  return eval_agg_clean(P, a, sval, r);
 }
 
 /* wave (64-lane) sum reduction */
@@ -419,6 +423,7 @@ __device__ __forceinline__ double wave_sum(double x) {
 template <int NAGGS>
 __launch_bounds__(WG, 2)
 __global__ void k_keyless(sn_dev_plan plan,
+                          const sn_dev_plan *__restrict__ plan_g,
                           const sn_dev_batch *__restrict__ batches,
                           const sn_dev_tile *__restrict__ tiles, int ntiles,
                           double *__restrict__ out /* [2*NAGGS+1] */) {
@@ -428,11 +433,20 @@ __global__ void k_keyless(sn_dev_plan plan,
   double *sval = (double *)smem;
   uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
   uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+  /* plan mirrored into LDS: row-phase field reads become broadcast ds_reads
+   * instead of SGPR-spilled kernarg loads */
+  sn_dev_plan *P = (sn_dev_plan *)(sdead + CHUNK / 64 + 2);
+  {
+    const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
+    unsigned *dst = (unsigned *)P;
+    for (unsigned i = tid; i < sizeof(sn_dev_plan) / 4; i += WG) dst[i] = src[i];
+  }
 
   double sums[NAGGS], cnts[NAGGS], rcnt = 0.0;
 #pragma unroll
   for (int a = 0; a < NAGGS; a++) { sums[a] = 0.0; cnts[a] = 0.0; }
   const int naggs = plan.naggs;
+  const int npd = plan.npreds_d, npi = plan.npreds_i;
 
   Stage st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
@@ -462,13 +476,14 @@ __global__ void k_keyless(sn_dev_plan plan,
       for (int k = 0; k < CHUNK / WG; k++) {
         int r = tid + k * WG;
         int inr = r < rows;
-        int alive = inr ? eval_preds(plan, clean, sval, svalid, sdead, r) : 0;
+        int alive = inr ? eval_preds(P, npd, npi, clean, sval, svalid, sdead, r) : 0;
         if (__popcll(__ballot(alive)) == 0) continue;
 #pragma unroll
         for (int a = 0; a < NAGGS; a++) {
           if (a >= naggs) break;
-          int anull;
-          double aval = eval_agg(plan, a, sval, r, clean, svalid, &anull);
+          int anull = 0;
+          double aval = clean ? eval_agg_clean(P, a, sval, r)
+                              : eval_agg_general(P, a, sval, r, svalid, &anull);
           const int m = alive && !anull;
           sums[a] += m ? aval : 0.0;
           cnts[a] += m ? 1.0 : 0.0;
@@ -501,6 +516,7 @@ __global__ void k_keyless(sn_dev_plan plan,
 template <int NSLOTS>
 __launch_bounds__(WG, 2)
 __global__ void k_grouped(sn_dev_plan plan,
+                          const sn_dev_plan *__restrict__ plan_g,
                           const sn_dev_batch *__restrict__ batches,
                           const sn_dev_tile *__restrict__ tiles, int ntiles,
                           double *__restrict__ out, int out_stride) {
@@ -508,6 +524,7 @@ __global__ void k_grouped(sn_dev_plan plan,
   const int wid = tid >> 6;
   const int nused = plan.nused;
   const int naggs = plan.naggs, ngroup = plan.ngroup;
+  const int npd = plan.npreds_d, npi = plan.npreds_i;
   const int nagg_grps = (naggs + AGRP - 1) / AGRP;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -518,6 +535,12 @@ __global__ void k_grouped(sn_dev_plan plan,
   int16_t *sslot = (int16_t *)(salive + CHUNK / 64);
   /* block accumulator: [NSLOTS][naggs+1], init once */
   double *bacc = (double *)(sslot + CHUNK);
+  sn_dev_plan *P = (sn_dev_plan *)(bacc + NSLOTS * (naggs + 1) + 2);
+  {
+    const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
+    unsigned *dst = (unsigned *)P;
+    for (unsigned i = tid; i < sizeof(sn_dev_plan) / 4; i += WG) dst[i] = src[i];
+  }
 
   for (int i = tid; i < NSLOTS * (naggs + 1); i += WG) bacc[i] = 0.0;
   __syncthreads();
@@ -551,7 +574,7 @@ __global__ void k_grouped(sn_dev_plan plan,
       for (int k = 0; k < CHUNK / WG; k++) {
         int r = tid + k * WG;
         int inr = r < rows;
-        int alive = inr ? eval_preds(plan, clean, sval, svalid, sdead, r) : 0;
+        int alive = inr ? eval_preds(P, npd, npi, clean, sval, svalid, sdead, r) : 0;
         int slot = 0;
         if (alive) {
           if (ngroup >= 1) slot = (int)sval[(size_t)plan.gcol[0] * CHUNK + r];
@@ -604,10 +627,7 @@ __global__ void k_grouped(sn_dev_plan plan,
 #pragma unroll
           for (int j = 0; j < AGRP; j++) {
             int a = g * AGRP + j;
-            int anull;
-            av[j] = (a < naggs)
-                ? eval_agg(plan, a, sval, r, clean, svalid, &anull) : 0.0;
-            if (a < naggs && !clean && anull) av[j] = 0.0;  /* non-null enforced */
+            av[j] = (a < naggs) ? eval_agg_clean(P, a, sval, r) : 0.0;
           }
 #pragma unroll
           for (int s = 0; s < NSLOTS; s++) {
@@ -646,6 +666,7 @@ __global__ void k_grouped(sn_dev_plan plan,
 }
 
 extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
+                                  const sn_dev_plan *dev_plan,
                                   const sn_dev_batch *dev_batches,
                                   const sn_dev_tile *dev_tiles, int32_t ntiles,
                                   double *dev_out, void *stream) {
@@ -653,18 +674,19 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
   int grid = ntiles < 2048 ? (ntiles > 0 ? ntiles : 1) : 2048;
   const int ns = plan->nslots, na = plan->naggs;
   size_t lds = (size_t)plan->nused * CHUNK * 8 +
-               (size_t)plan->nused * (CHUNK / 64) * 8 + (CHUNK / 64) * 8 + 16;
+               (size_t)plan->nused * (CHUNK / 64) * 8 + (CHUNK / 64) * 8 +
+               sizeof(sn_dev_plan) + 64;
   hipError_t err;
   if (ns <= 1) {
     if (na <= 2) {
       hipLaunchKernelGGL((k_keyless<2>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_batches, dev_tiles, ntiles, dev_out);
+                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out);
     } else if (na <= 4) {
       hipLaunchKernelGGL((k_keyless<4>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_batches, dev_tiles, ntiles, dev_out);
+                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out);
     } else {
       hipLaunchKernelGGL((k_keyless<12>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_batches, dev_tiles, ntiles, dev_out);
+                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out);
     }
   } else {
     /* grouped: extra LDS for salive + slot array + block accumulator */
@@ -672,13 +694,13 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
     int out_stride = 2 * (na <= 2 ? 2 : na <= 4 ? 4 : na <= 8 ? 8 : 12) + 1;
     if (ns <= 4) {
       hipLaunchKernelGGL((k_grouped<4>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
+                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
     } else if (ns <= 8) {
       hipLaunchKernelGGL((k_grouped<8>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
+                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
     } else if (ns <= 16) {
       hipLaunchKernelGGL((k_grouped<16>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
+                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
     } else {
       return (int)hipErrorInvalidValue;
     }

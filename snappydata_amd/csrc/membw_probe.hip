@@ -1,0 +1,151 @@
+/* membw_probe — isolate why the scan kernel plateaus at ~1.9 TB/s.
+ * Variants over the same 1.68 GB working set (4 f64-ish streams):
+ *   A: pure grid-stride streaming read (double2/lane), no LDS, no barriers
+ *   B: A + 33 KB dummy LDS per block (same blocks/CU as the scan kernel)
+ *   C: chunked LDS round trip: load 1024-row chunk -> ds_write -> barrier ->
+ *      ds_read+sum -> barrier (the scan kernel's skeleton)
+ *   D: C + register-staged prefetch of the next chunk (the T14 split)
+ * Build: hipcc --offload-arch=gfx950 -O3 membw_probe.hip -o membw_probe
+ */
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <vector>
+
+#define WG 256
+#define CHUNK 1024
+#define GAS __attribute__((address_space(1)))
+
+typedef double double2_t __attribute__((ext_vector_type(2)));
+
+__global__ __launch_bounds__(WG, 2) void k_streamA(
+    const double *__restrict__ a, const double *__restrict__ b,
+    const double *__restrict__ c, const double *__restrict__ d,
+    long long n, double *out) {
+  const GAS double2_t *A = (const GAS double2_t *)(uintptr_t)a;
+  const GAS double2_t *B = (const GAS double2_t *)(uintptr_t)b;
+  const GAS double2_t *C = (const GAS double2_t *)(uintptr_t)c;
+  const GAS double2_t *D = (const GAS double2_t *)(uintptr_t)d;
+  long long i = blockIdx.x * (long long)WG + threadIdx.x;
+  long long stride = gridDim.x * (long long)WG;
+  double2_t s = {0, 0};
+  for (; i < n / 2; i += stride) {
+    double2_t x = A[i], y = B[i], z = C[i], w = D[i];
+    s.x += x.x + y.x + z.x + w.x;
+    s.y += x.y + y.y + z.y + w.y;
+  }
+  double v = s.x + s.y;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  if ((threadIdx.x & 63) == 0 && v != 0.0) atomicAdd(out, v);
+}
+
+template <int LDSKB, int STAGED>
+__global__ __launch_bounds__(WG, 2) void k_chunked(
+    const double *__restrict__ a, const double *__restrict__ b,
+    const double *__restrict__ c, const double *__restrict__ d,
+    long long n, double *out) {
+  __shared__ __attribute__((aligned(16))) double lds[LDSKB * 128]; /* KB -> doubles */
+  const GAS double2_t *src[4] = {
+    (const GAS double2_t *)(uintptr_t)a, (const GAS double2_t *)(uintptr_t)b,
+    (const GAS double2_t *)(uintptr_t)c, (const GAS double2_t *)(uintptr_t)d };
+  const int tid = threadIdx.x;
+  long long nchunks = n / CHUNK;
+  double acc = 0.0;
+  double2_t st[4][2];
+  long long t0 = blockIdx.x;
+  if (STAGED && t0 < nchunks) {
+#pragma unroll
+    for (int cc = 0; cc < 4; cc++)
+#pragma unroll
+      for (int p = 0; p < 2; p++)
+        st[cc][p] = src[cc][t0 * (CHUNK / 2) + tid + p * WG];
+  }
+  for (long long t = t0; t < nchunks; t += gridDim.x) {
+    if (STAGED) {
+#pragma unroll
+      for (int cc = 0; cc < 4; cc++)
+#pragma unroll
+        for (int p = 0; p < 2; p++)
+          ((double2_t *)(lds + cc * CHUNK))[tid + p * WG] = st[cc][p];
+    } else {
+#pragma unroll
+      for (int cc = 0; cc < 4; cc++)
+#pragma unroll
+        for (int p = 0; p < 2; p++)
+          ((double2_t *)(lds + cc * CHUNK))[tid + p * WG] =
+              src[cc][t * (CHUNK / 2) + tid + p * WG];
+    }
+    __syncthreads();
+    long long tn = t + gridDim.x;
+    if (STAGED && tn < nchunks) {
+#pragma unroll
+      for (int cc = 0; cc < 4; cc++)
+#pragma unroll
+        for (int p = 0; p < 2; p++)
+          st[cc][p] = src[cc][tn * (CHUNK / 2) + tid + p * WG];
+    }
+    /* row phase: 4 passes, read every stream from LDS, cheap filter+sum */
+#pragma unroll 2
+    for (int k = 0; k < CHUNK / WG; k++) {
+      int r = tid + k * WG;
+      double ship = lds[3 * CHUNK + r];
+      int alive = ship > 0.1 && ship < 0.9;
+      double v = lds[0 * CHUNK + r] * lds[2 * CHUNK + r];
+      acc += (alive && lds[1 * CHUNK + r] < 2.0) ? v : 0.0;
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
+  if ((threadIdx.x & 63) == 0 && acc != 0.0) atomicAdd(out, acc);
+}
+
+static double bench(void (*launch)(int, const double *, const double *,
+                                   const double *, const double *, long long,
+                                   double *, hipStream_t),
+                    int grid, const double *a, const double *b, const double *c,
+                    const double *d, long long n, double *out) {
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0); hipEventCreate(&e1);
+  launch(grid, a, b, c, d, n, out, 0);   /* warmup */
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  for (int i = 0; i < 5; i++) launch(grid, a, b, c, d, n, out, 0);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  hipEventDestroy(e0); hipEventDestroy(e1);
+  return ms / 5.0;
+}
+
+#define LAUNCHER(name, kexpr)                                                 \
+  static void name(int grid, const double *a, const double *b,                \
+                   const double *c, const double *d, long long n,             \
+                   double *out, hipStream_t s) {                              \
+    hipLaunchKernelGGL(kexpr, dim3(grid), dim3(WG), 0, s, a, b, c, d, n, out); \
+  }
+LAUNCHER(la, k_streamA)
+LAUNCHER(lb, (k_chunked<33, 0>))
+LAUNCHER(lc, (k_chunked<33, 1>))
+
+int main(int argc, char **argv) {
+  long long n = 60LL * 1000 * 1000;   /* rows */
+  n = (n / CHUNK) * CHUNK;
+  double *a, *b, *c, *d, *out;
+  hipMalloc(&a, n * 8); hipMalloc(&b, n * 8);
+  hipMalloc(&c, n * 8); hipMalloc(&d, n * 8);
+  hipMalloc(&out, 8);
+  hipMemset(a, 0x11, n * 8); hipMemset(b, 0x12, n * 8);
+  hipMemset(c, 0x13, n * 8); hipMemset(d, 0x14, n * 8);
+  double bytes = 4.0 * n * 8;
+  for (int grid : {1024, 2048, 4096}) {
+    double msA = bench(la, grid, a, b, c, d, n, out);
+    double msB = bench(lb, grid, a, b, c, d, n, out);
+    double msC = bench(lc, grid, a, b, c, d, n, out);
+    printf("grid=%d  A(stream)=%.3fms %.0fGB/s  B(chunkLDS)=%.3fms %.0fGB/s  "
+           "C(staged)=%.3fms %.0fGB/s\n", grid,
+           msA, bytes / msA / 1e6, msB, bytes / msB / 1e6, msC, bytes / msC / 1e6);
+  }
+  return 0;
+}
