@@ -249,8 +249,9 @@ __device__ __forceinline__ void convert_chunk(
 /* one shared register buffer for every width class (64 VGPRs total): raw
  * bits packed into double2 lanes; row mapping is pair-based for every width
  * (pair h = tid + p*WG covers rows 2h, 2h+1) */
+template <int NC>
 struct Stage {
-  double2_t buf[SN_DEV_MAX_COLS][CHUNK / (2 * WG)];
+  double2_t buf[NC][CHUNK / (2 * WG)];
 };
 __device__ __forceinline__ double pack_u64(unsigned lo, unsigned hi) {
   return __longlong_as_double(((unsigned long long)hi << 32) | lo);
@@ -265,11 +266,12 @@ __device__ __forceinline__ int col_width_class(int kind) {
   }
 }
 
+template <int NC>
 __device__ __forceinline__ void stage_load(const sn_dev_batch &b, int nused,
-                                           int base, Stage &st) {
+                                           int base, Stage<NC> &st) {
   const int tid = threadIdx.x;
 #pragma unroll
-  for (int c = 0; c < SN_DEV_MAX_COLS; c++) {
+  for (int c = 0; c < NC; c++) {
     if (c >= nused) break;
     const sn_dev_col &col = b.cols[c];
     const int w = col_width_class(col.kind);
@@ -295,11 +297,12 @@ __device__ __forceinline__ void stage_load(const sn_dev_batch &b, int nused,
   }
 }
 
+template <int NC>
 __device__ __forceinline__ void stage_write(const sn_dev_batch &b, int nused,
-                                            Stage &st, double *sval) {
+                                            Stage<NC> &st, double *sval) {
   const int tid = threadIdx.x;
 #pragma unroll
-  for (int c = 0; c < SN_DEV_MAX_COLS; c++) {
+  for (int c = 0; c < NC; c++) {
     if (c >= nused) break;
     const sn_dev_col &col = b.cols[c];
     double *dst = sval + (size_t)c * CHUNK;
@@ -420,8 +423,8 @@ __device__ __forceinline__ double wave_sum(double x) {
 }
 
 /* ================= keyless kernel ================= */
-template <int NAGGS>
-__launch_bounds__(WG, 2)
+template <int NAGGS, int NC>
+__launch_bounds__(WG, NAGGS <= 4 ? 4 : 2)
 __global__ void k_keyless(sn_dev_plan plan,
                           const sn_dev_plan *__restrict__ plan_g,
                           const sn_dev_batch *__restrict__ batches,
@@ -448,7 +451,7 @@ __global__ void k_keyless(sn_dev_plan plan,
   const int naggs = plan.naggs;
   const int npd = plan.npreds_d, npi = plan.npreds_i;
 
-  Stage st;
+  Stage<NC> st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
@@ -513,7 +516,7 @@ __global__ void k_keyless(sn_dev_plan plan,
  * atomics.  Requires non-nullable aggregate inputs (engine-validated);
  * output layout [slot][2*NA+1]: sums, counts(=rowcount written host-side),
  * rowcount. */
-template <int NSLOTS>
+template <int NSLOTS, int NC>
 __launch_bounds__(WG, 2)
 __global__ void k_grouped(sn_dev_plan plan,
                           const sn_dev_plan *__restrict__ plan_g,
@@ -545,7 +548,7 @@ __global__ void k_grouped(sn_dev_plan plan,
   for (int i = tid; i < NSLOTS * (naggs + 1); i += WG) bacc[i] = 0.0;
   __syncthreads();
 
-  Stage st;
+  Stage<NC> st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
@@ -677,34 +680,26 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
                (size_t)plan->nused * (CHUNK / 64) * 8 + (CHUNK / 64) * 8 +
                sizeof(sn_dev_plan) + 64;
   hipError_t err;
+  const bool nc4 = plan->nused <= 4;
+#define KL(A, NCv) hipLaunchKernelGGL((k_keyless<A, NCv>), dim3(grid), dim3(WG), lds, s, \
+        *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out)
+#define KG(S, NCv) hipLaunchKernelGGL((k_grouped<S, NCv>), dim3(grid), dim3(WG), lds, s, \
+        *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride)
   if (ns <= 1) {
-    if (na <= 2) {
-      hipLaunchKernelGGL((k_keyless<2>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out);
-    } else if (na <= 4) {
-      hipLaunchKernelGGL((k_keyless<4>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out);
-    } else {
-      hipLaunchKernelGGL((k_keyless<12>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out);
-    }
+    if (na <= 2) { if (nc4) KL(2, 4); else KL(2, 8); }
+    else if (na <= 4) { if (nc4) KL(4, 4); else KL(4, 8); }
+    else { if (nc4) KL(12, 4); else KL(12, 8); }
   } else {
     /* grouped: extra LDS for salive + slot array + block accumulator */
     lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)ns * (na + 1) * 8 + 16;
     int out_stride = 2 * (na <= 2 ? 2 : na <= 4 ? 4 : na <= 8 ? 8 : 12) + 1;
-    if (ns <= 4) {
-      hipLaunchKernelGGL((k_grouped<4>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
-    } else if (ns <= 8) {
-      hipLaunchKernelGGL((k_grouped<8>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
-    } else if (ns <= 16) {
-      hipLaunchKernelGGL((k_grouped<16>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride);
-    } else {
-      return (int)hipErrorInvalidValue;
-    }
+    if (ns <= 4) { if (nc4) KG(4, 4); else KG(4, 8); }
+    else if (ns <= 8) { if (nc4) KG(8, 4); else KG(8, 8); }
+    else if (ns <= 16) { if (nc4) KG(16, 4); else KG(16, 8); }
+    else return (int)hipErrorInvalidValue;
   }
+#undef KL
+#undef KG
   err = hipGetLastError();
   return (int)err;
 }
