@@ -422,6 +422,63 @@ __device__ __forceinline__ double wave_sum(double x) {
   return x;
 }
 
+/* initialize the alive bitmap words (wave-owned: the wave that processes
+ * rows [w*64, w*64+64) is the only writer of word w in every later pass) */
+__device__ __forceinline__ void alive_init(uint64_t *salive,
+                                           const uint64_t *sdead, int rows,
+                                           int clean) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int k = 0; k < CHUNK / WG; k++) {
+    const int r = tid + k * WG;
+    const int w = r >> 6;
+    if ((tid & 63) == 0) {
+      uint64_t aw = clean ? ~0ull : ~sdead[w];
+      const int rbase = w << 6;
+      if (rbase + 64 > rows)
+        aw &= rows > rbase ? ((1ull << (rows - rbase)) - 1ull) : 0ull;
+      salive[w] = aw;
+    }
+  }
+}
+
+/* one sweep per predicate, params hoisted; wave-owned word updates */
+__device__ __forceinline__ void pred_sweeps(const sn_dev_plan *P, int npd,
+                                            int npi, int clean,
+                                            const double *sval,
+                                            const uint64_t *svalid,
+                                            uint64_t *salive) {
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < 8; i++) {
+    if (i >= npd) break;
+    const int cs = P->preds_d[i].cslot;
+    const double lo = P->preds_d[i].lo, hi = P->preds_d[i].hi;
+#pragma unroll
+    for (int k = 0; k < CHUNK / WG; k++) {
+      const int r = tid + k * WG;
+      const double x = sval[(size_t)cs * CHUNK + r];
+      uint64_t w = __ballot(x >= lo && x <= hi);
+      if (!clean) w &= svalid[(size_t)cs * (CHUNK / 64) + (r >> 6)];
+      if ((tid & 63) == 0) salive[r >> 6] &= w;
+    }
+  }
+#pragma unroll
+  for (int i = 0; i < 4; i++) {
+    if (i >= npi) break;
+    const int cs = P->preds_i[i].cslot;
+    const long long lo = P->preds_i[i].lo, hi = P->preds_i[i].hi;
+#pragma unroll
+    for (int k = 0; k < CHUNK / WG; k++) {
+      const int r = tid + k * WG;
+      const long long x = __double_as_longlong(sval[(size_t)cs * CHUNK + r]);
+      uint64_t w = __ballot(x >= lo && x <= hi);
+      if (!clean) w &= svalid[(size_t)cs * (CHUNK / 64) + (r >> 6)];
+      if ((tid & 63) == 0) salive[r >> 6] &= w;
+    }
+  }
+}
+
 /* ================= keyless kernel ================= */
 template <int NAGGS, int NC>
 __launch_bounds__(WG, NAGGS <= 4 ? 4 : 2)
@@ -436,9 +493,10 @@ __global__ void k_keyless(sn_dev_plan plan,
   double *sval = (double *)smem;
   uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
   uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+  uint64_t *salive = sdead + CHUNK / 64;
   /* plan mirrored into LDS: row-phase field reads become broadcast ds_reads
    * instead of SGPR-spilled kernarg loads */
-  sn_dev_plan *P = (sn_dev_plan *)(sdead + CHUNK / 64 + 2);
+  sn_dev_plan *P = (sn_dev_plan *)(salive + CHUNK / 64 + 2);
   {
     const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
     unsigned *dst = (unsigned *)P;
@@ -475,23 +533,37 @@ __global__ void k_keyless(sn_dev_plan plan,
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
       if (next_staged) stage_load(b, nused, nbase, st);
 
-#pragma unroll 2
-      for (int k = 0; k < CHUNK / WG; k++) {
-        int r = tid + k * WG;
-        int inr = r < rows;
-        int alive = inr ? eval_preds(P, npd, npi, clean, sval, svalid, sdead, r) : 0;
-        if (__popcll(__ballot(alive)) == 0) continue;
+      /* ---- row phase as sweep passes: each wave owns its 64-row words of
+       * the alive bitmap, so pred/agg passes need no barriers; plan params
+       * hoisted per pass (row-invariant LDS reads once, not per row) ---- */
+      alive_init(salive, sdead, rows, clean);
+      pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
 #pragma unroll
-        for (int a = 0; a < NAGGS; a++) {
-          if (a >= naggs) break;
-          int anull = 0;
-          double aval = clean ? eval_agg_clean(P, a, sval, r)
-                              : eval_agg_general(P, a, sval, r, svalid, &anull);
-          const int m = alive && !anull;
-          sums[a] += m ? aval : 0.0;
+      for (int a = 0; a < NAGGS; a++) {
+        if (a >= naggs) break;
+        const sn_dev_agg A = P->aggs[a];
+#pragma unroll 2
+        for (int k = 0; k < CHUNK / WG; k++) {
+          const int r = tid + k * WG;
+          uint64_t w = salive[r >> 6];
+          if (!clean) {
+            if (A.nf >= 1) w &= svalid[(size_t)A.c0 * (CHUNK / 64) + (r >> 6)];
+            if (A.nf >= 2) w &= svalid[(size_t)A.c1 * (CHUNK / 64) + (r >> 6)];
+            if (A.nf >= 3) w &= svalid[(size_t)A.c2 * (CHUNK / 64) + (r >> 6)];
+          }
+          if (w == 0) continue;
+          const int m = (int)((w >> (tid & 63)) & 1ull);
+          const double val = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
+                             (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
+                             (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+          sums[a] += m ? val : 0.0;
           cnts[a] += m ? 1.0 : 0.0;
         }
-        rcnt += alive ? 1.0 : 0.0;
+      }
+#pragma unroll 2
+      for (int k = 0; k < CHUNK / WG; k++) {
+        const int r = tid + k * WG;
+        rcnt += (double)((salive[r >> 6] >> (tid & 63)) & 1ull);
       }
       __syncthreads();
       staged = next_staged;
@@ -528,7 +600,6 @@ __global__ void k_grouped(sn_dev_plan plan,
   const int nused = plan.nused;
   const int naggs = plan.naggs, ngroup = plan.ngroup;
   const int npd = plan.npreds_d, npi = plan.npreds_i;
-  const int nagg_grps = (naggs + AGRP - 1) / AGRP;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   double *sval = (double *)smem;
@@ -571,25 +642,23 @@ __global__ void k_grouped(sn_dev_plan plan,
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
       if (next_staged) stage_load(b, nused, nbase, st);
 
-      /* pass A: predicates + slot once per row */
-      double rcreg = 0.0;
-#pragma unroll 2
-      for (int k = 0; k < CHUNK / WG; k++) {
-        int r = tid + k * WG;
-        int inr = r < rows;
-        int alive = inr ? eval_preds(P, npd, npi, clean, sval, svalid, sdead, r) : 0;
-        int slot = 0;
-        if (alive) {
-          if (ngroup >= 1) slot = (int)sval[(size_t)plan.gcol[0] * CHUNK + r];
-          if (ngroup >= 2) slot += (int)sval[(size_t)plan.gcol[1] * CHUNK + r];
+      /* pass A: alive bitmap via sweeps + slot per row (wave-owned words,
+       * no barrier needed before this block's own later passes) */
+      alive_init(salive, sdead, rows, clean);
+      pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
+      {
+        const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
+#pragma unroll
+        for (int k = 0; k < CHUNK / WG; k++) {
+          const int r = tid + k * WG;
+          int slot = 0;
+          if (ngroup >= 1) slot = (int)sval[(size_t)gc0 * CHUNK + r];
+          if (ngroup >= 2) slot += (int)sval[(size_t)gc1 * CHUNK + r];
+          sslot[r] = (int16_t)slot;
         }
-        sslot[r] = (int16_t)slot;
-        uint64_t w = __ballot(alive);
-        if ((tid & 63) == 0) salive[r >> 6] = w;
       }
-      __syncthreads();
 
-      /* rowcount per slot (registers, one pass) */
+      /* rowcount per slot */
       {
         double rc[NSLOTS];
 #pragma unroll
@@ -609,48 +678,35 @@ __global__ void k_grouped(sn_dev_plan plan,
           if ((tid & 63) == 0 && x != 0.0)
             atomicAdd(&bacc[s * (naggs + 1) + naggs], x);
         }
-        (void)rcreg;
       }
 
-      /* pass B: aggregates in groups of AGRP */
-      for (int g = 0; g < nagg_grps; g++) {
-        double sums[NSLOTS][AGRP];
+      /* pass B: one sweep per aggregate, params hoisted, slot-predicated
+       * register accumulators (NSLOTS live at a time) */
+      for (int a = 0; a < naggs; a++) {
+        const sn_dev_agg A = P->aggs[a];
+        double sums[NSLOTS];
 #pragma unroll
-        for (int s = 0; s < NSLOTS; s++)
-#pragma unroll
-          for (int j = 0; j < AGRP; j++) sums[s][j] = 0.0;
-
+        for (int s = 0; s < NSLOTS; s++) sums[s] = 0.0;
 #pragma unroll 2
         for (int k = 0; k < CHUNK / WG; k++) {
-          int r = tid + k * WG;
-          int alive = (int)((salive[r >> 6] >> (r & 63)) & 1ull);
-          if (__popcll(__ballot(alive)) == 0) continue;
-          int slot = sslot[r];
-          double av[AGRP];
+          const int r = tid + k * WG;
+          const uint64_t w = salive[r >> 6];
+          if (w == 0) continue;
+          const int m = (int)((w >> (tid & 63)) & 1ull);
+          const int slot = sslot[r];
+          const double val = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
+                             (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
+                             (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
 #pragma unroll
-          for (int j = 0; j < AGRP; j++) {
-            int a = g * AGRP + j;
-            av[j] = (a < naggs) ? eval_agg_clean(P, a, sval, r) : 0.0;
-          }
-#pragma unroll
-          for (int s = 0; s < NSLOTS; s++) {
-            const int m = alive && slot == s;
-#pragma unroll
-            for (int j = 0; j < AGRP; j++)
-              sums[s][j] += m ? av[j] : 0.0;
-          }
+          for (int s = 0; s < NSLOTS; s++)
+            sums[s] += (m && slot == s) ? val : 0.0;
         }
-        /* wave-reduce into the block accumulator */
 #pragma unroll
-        for (int s = 0; s < NSLOTS; s++)
-#pragma unroll
-          for (int j = 0; j < AGRP; j++) {
-            int a = g * AGRP + j;
-            if (a >= naggs) continue;
-            double x = wave_sum(sums[s][j]);
-            if ((tid & 63) == 0 && x != 0.0)
-              atomicAdd(&bacc[s * (naggs + 1) + a], x);
-          }
+        for (int s = 0; s < NSLOTS; s++) {
+          double x = wave_sum(sums[s]);
+          if ((tid & 63) == 0 && x != 0.0)
+            atomicAdd(&bacc[s * (naggs + 1) + a], x);
+        }
       }
       __syncthreads();
       staged = next_staged;
@@ -677,7 +733,8 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
   int grid = ntiles < 2048 ? (ntiles > 0 ? ntiles : 1) : 2048;
   const int ns = plan->nslots, na = plan->naggs;
   size_t lds = (size_t)plan->nused * CHUNK * 8 +
-               (size_t)plan->nused * (CHUNK / 64) * 8 + (CHUNK / 64) * 8 +
+               (size_t)plan->nused * (CHUNK / 64) * 8 +
+               2 * (CHUNK / 64) * 8 +            /* sdead + salive */
                sizeof(sn_dev_plan) + 64;
   hipError_t err;
   const bool nc4 = plan->nused <= 4;
