@@ -246,6 +246,27 @@ __device__ __forceinline__ void convert_chunk(
  * SQ_WAIT_ANY was 75% of wave cycles with the unstaged conversion).
  * Raw loads are width-based (8/4/2 B); kind conversion happens at the LDS
  * write.  I64 shares the w8 path (its LDS image is the raw bits). */
+/* per-tile register copy of the hot column-descriptor fields: reading
+ * b.cols[c].body/kind per CHUNK emits serialized global_load_dword +
+ * vmcnt(0) chains (measured: a fixed ~0.35 ms per query = the whole Q6
+ * budget).  Copy once per tile into compile-time-indexed registers. */
+struct ColRegs {
+  const void *body;
+  const int32_t *dictmap;
+  int kind;
+};
+
+template <int NC>
+__device__ __forceinline__ void hoist_cols(const sn_dev_batch &b, int nused,
+                                           ColRegs (&cr)[NC]) {
+#pragma unroll
+  for (int c = 0; c < NC; c++) {
+    cr[c].body = c < nused ? b.cols[c].body : nullptr;
+    cr[c].dictmap = c < nused ? b.cols[c].dictmap : nullptr;
+    cr[c].kind = c < nused ? b.cols[c].kind : -1;
+  }
+}
+
 /* one shared register buffer for every width class (64 VGPRs total): raw
  * bits packed into double2 lanes; row mapping is pair-based for every width
  * (pair h = tid + p*WG covers rows 2h, 2h+1) */
@@ -267,13 +288,13 @@ __device__ __forceinline__ int col_width_class(int kind) {
 }
 
 template <int NC>
-__device__ __forceinline__ void stage_load(const sn_dev_batch &b, int nused,
+__device__ __forceinline__ void stage_load(const ColRegs (&cr)[NC], int nused,
                                            int base, Stage<NC> &st) {
   const int tid = threadIdx.x;
 #pragma unroll
   for (int c = 0; c < NC; c++) {
     if (c >= nused) break;
-    const sn_dev_col &col = b.cols[c];
+    const ColRegs &col = cr[c];
     const int w = col_width_class(col.kind);
     if (w == 8) {
       const GAS double2_t *s2 = (const GAS double2_t *)
@@ -298,13 +319,13 @@ __device__ __forceinline__ void stage_load(const sn_dev_batch &b, int nused,
 }
 
 template <int NC>
-__device__ __forceinline__ void stage_write(const sn_dev_batch &b, int nused,
+__device__ __forceinline__ void stage_write(const ColRegs (&cr)[NC], int nused,
                                             Stage<NC> &st, double *sval) {
   const int tid = threadIdx.x;
 #pragma unroll
   for (int c = 0; c < NC; c++) {
     if (c >= nused) break;
-    const sn_dev_col &col = b.cols[c];
+    const ColRegs &col = cr[c];
     double *dst = sval + (size_t)c * CHUNK;
     switch (col.kind) {
       case SN_K_F64: case SN_K_I64:
@@ -351,13 +372,15 @@ __device__ __forceinline__ void stage_write(const sn_dev_batch &b, int nused,
 }
 
 /* can this batch use the staged pipeline? (clean + every col stageable) */
-__device__ __forceinline__ int batch_stageable(const sn_dev_batch &b, int nused) {
-  if (!b.clean) return 0;
+template <int NC>
+__device__ __forceinline__ int batch_stageable(int clean, int nused,
+                                               const ColRegs (&cr)[NC]) {
+  if (!clean) return 0;
   int ok = 1;
 #pragma unroll
-  for (int c = 0; c < SN_DEV_MAX_COLS; c++) {
+  for (int c = 0; c < NC; c++) {
     if (c >= nused) break;
-    ok &= col_width_class(b.cols[c].kind) != 0;
+    ok &= col_width_class(cr[c].kind) != 0;
   }
   return ok;
 }
@@ -516,22 +539,24 @@ __global__ void k_keyless(sn_dev_plan plan,
     const int num_rows = b.num_rows;
     const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
     const int clean = b.clean;
-    const int pipe = batch_stageable(b, nused);
+    ColRegs cr[NC];
+    hoist_cols(b, nused, cr);
+    const int pipe = batch_stageable(clean, nused, cr);
 
     int staged = 0;
     if (pipe && tile.row_start + CHUNK <= tile_end) {
-      stage_load(b, nused, tile.row_start, st);
+      stage_load(cr, nused, tile.row_start, st);
       staged = 1;
     }
     for (int base = tile.row_start; base < tile_end; base += CHUNK) {
       const int rows = min(CHUNK, tile_end - base);
-      if (staged) stage_write(b, nused, st, sval);
+      if (staged) stage_write(cr, nused, st, sval);
       else convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
       __syncthreads();
       /* prefetch the NEXT full chunk under this chunk's row phase */
       const int nbase = base + CHUNK;
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
-      if (next_staged) stage_load(b, nused, nbase, st);
+      if (next_staged) stage_load(cr, nused, nbase, st);
 
       /* ---- row phase as sweep passes: each wave owns its 64-row words of
        * the alive bitmap, so pred/agg passes need no barriers; plan params
@@ -626,21 +651,23 @@ __global__ void k_grouped(sn_dev_plan plan,
     const int num_rows = b.num_rows;
     const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
     const int clean = b.clean;
-    const int pipe = batch_stageable(b, nused);
+    ColRegs cr[NC];
+    hoist_cols(b, nused, cr);
+    const int pipe = batch_stageable(clean, nused, cr);
 
     int staged = 0;
     if (pipe && tile.row_start + CHUNK <= tile_end) {
-      stage_load(b, nused, tile.row_start, st);
+      stage_load(cr, nused, tile.row_start, st);
       staged = 1;
     }
     for (int base = tile.row_start; base < tile_end; base += CHUNK) {
       const int rows = min(CHUNK, tile_end - base);
-      if (staged) stage_write(b, nused, st, sval);
+      if (staged) stage_write(cr, nused, st, sval);
       else convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
       __syncthreads();
       const int nbase = base + CHUNK;
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
-      if (next_staged) stage_load(b, nused, nbase, st);
+      if (next_staged) stage_load(cr, nused, nbase, st);
 
       /* pass A: alive bitmap via sweeps + slot per row (wave-owned words,
        * no barrier needed before this block's own later passes) */

@@ -39,7 +39,7 @@ __global__ __launch_bounds__(WG, 2) void k_streamA(
   if ((threadIdx.x & 63) == 0 && v != 0.0) atomicAdd(out, v);
 }
 
-template <int LDSKB, int STAGED>
+template <int LDSKB, int STAGED, int TILE_CHUNKS = 1>
 __global__ __launch_bounds__(WG, 2) void k_chunked(
     const double *__restrict__ a, const double *__restrict__ b,
     const double *__restrict__ c, const double *__restrict__ d,
@@ -52,15 +52,19 @@ __global__ __launch_bounds__(WG, 2) void k_chunked(
   long long nchunks = n / CHUNK;
   double acc = 0.0;
   double2_t st[4][2];
-  long long t0 = blockIdx.x;
-  if (STAGED && t0 < nchunks) {
+  /* TILE_CHUNKS>1: contiguous multi-chunk tiles per block (the engine's
+   * mapping); 1: fine-grained grid-stride (chip-wide sequential sweep) */
+  long long ntiles = nchunks / TILE_CHUNKS;
+  for (long long tt = blockIdx.x; tt < ntiles; tt += gridDim.x) {
+  long long t0 = tt * TILE_CHUNKS;
+  if (STAGED) {
 #pragma unroll
     for (int cc = 0; cc < 4; cc++)
 #pragma unroll
       for (int p = 0; p < 2; p++)
         st[cc][p] = src[cc][t0 * (CHUNK / 2) + tid + p * WG];
   }
-  for (long long t = t0; t < nchunks; t += gridDim.x) {
+  for (long long t = t0; t < t0 + TILE_CHUNKS; t += 1) {
     if (STAGED) {
 #pragma unroll
       for (int cc = 0; cc < 4; cc++)
@@ -76,7 +80,7 @@ __global__ __launch_bounds__(WG, 2) void k_chunked(
               src[cc][t * (CHUNK / 2) + tid + p * WG];
     }
     __syncthreads();
-    long long tn = t + gridDim.x;
+    long long tn = t + 1;
     if (STAGED && tn < nchunks) {
 #pragma unroll
       for (int cc = 0; cc < 4; cc++)
@@ -94,6 +98,7 @@ __global__ __launch_bounds__(WG, 2) void k_chunked(
       acc += (alive && lds[1 * CHUNK + r] < 2.0) ? v : 0.0;
     }
     __syncthreads();
+  }
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
@@ -128,6 +133,8 @@ static double bench(void (*launch)(int, const double *, const double *,
 LAUNCHER(la, k_streamA)
 LAUNCHER(lb, (k_chunked<33, 0>))
 LAUNCHER(lc, (k_chunked<33, 1>))
+LAUNCHER(ld, (k_chunked<33, 0, 16>))
+LAUNCHER(le, (k_chunked<33, 1, 16>))
 
 int main(int argc, char **argv) {
   long long n = 60LL * 1000 * 1000;   /* rows */
@@ -143,9 +150,12 @@ int main(int argc, char **argv) {
     double msA = bench(la, grid, a, b, c, d, n, out);
     double msB = bench(lb, grid, a, b, c, d, n, out);
     double msC = bench(lc, grid, a, b, c, d, n, out);
-    printf("grid=%d  A(stream)=%.3fms %.0fGB/s  B(chunkLDS)=%.3fms %.0fGB/s  "
-           "C(staged)=%.3fms %.0fGB/s\n", grid,
-           msA, bytes / msA / 1e6, msB, bytes / msB / 1e6, msC, bytes / msC / 1e6);
+    double msD = bench(ld, grid, a, b, c, d, n, out);
+    double msE = bench(le, grid, a, b, c, d, n, out);
+    printf("grid=%d  A(stream)=%.0fGB/s  B(chunk-stride)=%.0fGB/s  C(staged-stride)=%.0fGB/s  "
+           "D(16chunk-tile)=%.0fGB/s  E(16tile-staged)=%.0fGB/s\n", grid,
+           bytes / msA / 1e6, bytes / msB / 1e6, bytes / msC / 1e6,
+           bytes / msD / 1e6, bytes / msE / 1e6);
   }
   return 0;
 }
