@@ -170,6 +170,10 @@ struct sn_engine {
   std::vector<std::unique_ptr<Table>> tables;
   hipStream_t stream = nullptr;
   bool has_gpu = false;
+  /* reusable per-engine scratch for block-partial reduction rows
+   * (queries on one engine serialize on its stream) */
+  double *scratch = nullptr;
+  size_t scratch_sz = 0;
   std::mutex mu;
 };
 
@@ -940,13 +944,23 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         hipMemcpy(dp_dev, &dp, sizeof(dp), hipMemcpyHostToDevice) != hipSuccess) {
       fail(SN_ERR_NOMEM, "plan upload"); return nullptr;
     }
+    /* block-partial scratch rows */
+    int grid = (int)htiles.size() < SN_GRID_CAP ? (int)htiles.size() : SN_GRID_CAP;
+    size_t nv = dp.nslots <= 1 ? (size_t)(2 * q->na_t + 1)
+                               : (size_t)dp.nslots * (dp.naggs + 1);
+    size_t need = (size_t)grid * nv * 8;
+    if (e->scratch_sz < need) {
+      e->scratch = (double *)e->arena.alloc(need);
+      e->scratch_sz = need;
+      if (!e->scratch) { fail(SN_ERR_NOMEM, "scratch alloc"); return nullptr; }
+    }
     (void)hipEventCreate(&q->ev_start);
     (void)hipEventCreate(&q->ev_stop);
     if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
     int rc = sn_launch_scan_agg(&dp, (const sn_dev_plan *)dp_dev,
                                 (const sn_dev_batch *)db_dev,
                                 (const sn_dev_tile *)tl_dev, (int32_t)htiles.size(),
-                                q->dev_out, e->stream);
+                                q->dev_out, e->scratch, e->stream);
     if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);
     if (rc != 0) {
       fail(SN_ERR_GENERIC, "kernel launch: %s", hipGetErrorString((hipError_t)rc));

@@ -88,6 +88,7 @@ typedef struct {
   sn_dev_agg aggs[12];
 } sn_dev_plan;
 
+#define SN_GRID_CAP 2048
 #define SN_TILE_ROWS 16384     /* rows per workgroup tile (16 LDS chunks;
                                   long pipelines for the staged conversion) */
 
@@ -102,7 +103,9 @@ int sn_launch_scan_agg(const sn_dev_plan *plan,
                        const sn_dev_plan *dev_plan,  /* device copy (LDS mirror source) */
                        const sn_dev_batch *dev_batches,
                        const sn_dev_tile *dev_tiles, int32_t ntiles,
-                       double *dev_out, void *stream);
+                       double *dev_out,
+                       double *dev_scratch,  /* >= min(ntiles,SN_GRID_CAP) x nv */
+                       void *stream);
 
 #ifdef __cplusplus
 }

@@ -595,15 +595,27 @@ __global__ void k_keyless(sn_dev_plan plan,
     }
   }
 
+  /* block partials to scratch (plain stores) — per-wave atomicAdd to the
+   * same few global addresses serialized at ~45ns each (8K+ per address =
+   * a fixed ~0.35 ms per query); a tiny reduce kernel sums the partials */
+  double *bacc = (double *)(P + 1);
+  {
+    const int NV = 2 * NAGGS + 1;
+    for (int i = tid; i < NV; i += WG) bacc[i] = 0.0;
+    __syncthreads();
 #pragma unroll
-  for (int a = 0; a < NAGGS; a++) {
-    double x = wave_sum(sums[a]);
-    if ((tid & 63) == 0 && x != 0.0) atomicAdd(&out[a], x);
-    x = wave_sum(cnts[a]);
-    if ((tid & 63) == 0 && x != 0.0) atomicAdd(&out[NAGGS + a], x);
+    for (int a = 0; a < NAGGS; a++) {
+      double x = wave_sum(sums[a]);
+      if ((tid & 63) == 0 && x != 0.0) atomicAdd(bacc + a, x);
+      x = wave_sum(cnts[a]);
+      if ((tid & 63) == 0 && x != 0.0) atomicAdd(bacc + NAGGS + a, x);
+    }
+    double x = wave_sum(rcnt);
+    if ((tid & 63) == 0 && x != 0.0) atomicAdd(bacc + 2 * NAGGS, x);
+    __syncthreads();
+    for (int i = tid; i < NV; i += WG)
+      out[(size_t)blockIdx.x * NV + i] = bacc[i];
   }
-  double x = wave_sum(rcnt);
-  if ((tid & 63) == 0 && x != 0.0) atomicAdd(&out[2 * NAGGS], x);
 }
 
 /* ================= grouped kernel =================
@@ -740,43 +752,76 @@ __global__ void k_grouped(sn_dev_plan plan,
     }
   }
 
-  /* flush block accumulator to global (one atomic per value per block) */
+  /* flush block accumulator to a per-block scratch row (plain stores);
+   * the reduce kernel folds rows into the final [slot][stride] layout */
   __syncthreads();
-  for (int i = tid; i < NSLOTS * (naggs + 1); i += WG) {
-    int s = i / (naggs + 1), a = i % (naggs + 1);
-    double x = bacc[i];
-    if (x != 0.0)
-      atomicAdd(&out[s * out_stride + (a < naggs ? a : out_stride - 1)], x);
+  const int NV = NSLOTS * (naggs + 1);
+  for (int i = tid; i < NV; i += WG)
+    out[(size_t)blockIdx.x * NV + i] = bacc[i];
+  (void)wid; (void)out_stride;
+}
+
+/* fold per-block partial rows into the final output.
+ * keyless: final[i] = sum_b scratch[b][i]  (NV = 2*NA_t+1, identical layout)
+ * grouped: scratch rows are [slot][naggs+1]; final is [slot][out_stride]
+ * with rowcount at out_stride-1. */
+__global__ void k_reduce(const double *__restrict__ scratch, int nblocks,
+                         int nv, double *__restrict__ out, int naggs1,
+                         int out_stride) {
+  /* one block per output value; 256 threads stride the partial rows */
+  __shared__ double red[4];
+  const GAS double *src = (const GAS double *)(uintptr_t)scratch;
+  const int i = blockIdx.x;
+  if (i >= nv) return;
+  double s = 0.0;
+  for (int b = threadIdx.x; b < nblocks; b += blockDim.x)
+    s += src[(size_t)b * nv + i];
+  s = wave_sum(s);
+  if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = s;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double t = red[0] + red[1] + red[2] + red[3];
+    if (naggs1 == 0) {
+      out[i] = t;
+    } else {
+      int slot = i / naggs1, a = i % naggs1;
+      out[(size_t)slot * out_stride + (a < naggs1 - 1 ? a : out_stride - 1)] = t;
+    }
   }
-  (void)wid;
 }
 
 extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
                                   const sn_dev_plan *dev_plan,
                                   const sn_dev_batch *dev_batches,
                                   const sn_dev_tile *dev_tiles, int32_t ntiles,
-                                  double *dev_out, void *stream) {
+                                  double *dev_out, double *dev_scratch,
+                                  void *stream) {
   hipStream_t s = (hipStream_t)stream;
-  int grid = ntiles < 2048 ? (ntiles > 0 ? ntiles : 1) : 2048;
+  int grid = ntiles < SN_GRID_CAP ? (ntiles > 0 ? ntiles : 1) : SN_GRID_CAP;
   const int ns = plan->nslots, na = plan->naggs;
+  const int na_t = na <= 2 ? 2 : na <= 4 ? 4 : na <= 8 ? 8 : 12;
   size_t lds = (size_t)plan->nused * CHUNK * 8 +
                (size_t)plan->nused * (CHUNK / 64) * 8 +
                2 * (CHUNK / 64) * 8 +            /* sdead + salive */
-               sizeof(sn_dev_plan) + 64;
+               sizeof(sn_dev_plan) + 512;        /* plan mirror + keyless bacc */
   hipError_t err;
   const bool nc4 = plan->nused <= 4;
+  int nv, naggs1, out_stride;
+  if (ns <= 1) {
+    nv = 2 * na_t + 1; naggs1 = 0; out_stride = nv;
+  } else {
+    nv = ns * (na + 1); naggs1 = na + 1; out_stride = 2 * na_t + 1;
+  }
 #define KL(A, NCv) hipLaunchKernelGGL((k_keyless<A, NCv>), dim3(grid), dim3(WG), lds, s, \
-        *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out)
+        *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_scratch)
 #define KG(S, NCv) hipLaunchKernelGGL((k_grouped<S, NCv>), dim3(grid), dim3(WG), lds, s, \
-        *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_out, out_stride)
+        *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_scratch, out_stride)
   if (ns <= 1) {
     if (na <= 2) { if (nc4) KL(2, 4); else KL(2, 8); }
     else if (na <= 4) { if (nc4) KL(4, 4); else KL(4, 8); }
     else { if (nc4) KL(12, 4); else KL(12, 8); }
   } else {
-    /* grouped: extra LDS for salive + slot array + block accumulator */
     lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)ns * (na + 1) * 8 + 16;
-    int out_stride = 2 * (na <= 2 ? 2 : na <= 4 ? 4 : na <= 8 ? 8 : 12) + 1;
     if (ns <= 4) { if (nc4) KG(4, 4); else KG(4, 8); }
     else if (ns <= 8) { if (nc4) KG(8, 4); else KG(8, 8); }
     else if (ns <= 16) { if (nc4) KG(16, 4); else KG(16, 8); }
@@ -784,6 +829,8 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
   }
 #undef KL
 #undef KG
+  hipLaunchKernelGGL(k_reduce, dim3(nv), dim3(WG), 0, s,
+                     dev_scratch, grid, nv, dev_out, naggs1, out_stride);
   err = hipGetLastError();
   return (int)err;
 }
