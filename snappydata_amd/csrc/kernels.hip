@@ -753,9 +753,11 @@ __global__ void k_grouped(sn_dev_plan plan,
   }
 
   /* flush block accumulator to a per-block scratch row (plain stores);
-   * the reduce kernel folds rows into the final [slot][stride] layout */
+   * the reduce kernel folds rows into the final [slot][stride] layout.
+   * Row width uses the RUNTIME slot count (the engine sizes scratch by it;
+   * the template NSLOTS is only the register-capacity ceiling). */
   __syncthreads();
-  const int NV = NSLOTS * (naggs + 1);
+  const int NV = plan.nslots * (naggs + 1);
   for (int i = tid; i < NV; i += WG)
     out[(size_t)blockIdx.x * NV + i] = bacc[i];
   (void)wid; (void)out_stride;
@@ -821,7 +823,10 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
     else if (na <= 4) { if (nc4) KL(4, 4); else KL(4, 8); }
     else { if (nc4) KL(12, 4); else KL(12, 8); }
   } else {
-    lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)ns * (na + 1) * 8 + 16;
+    /* bacc + plan-mirror offsets inside the kernel use the TEMPLATE slot
+     * count — size the dynamic LDS with it, not the runtime ns */
+    const int ns_t = ns <= 4 ? 4 : ns <= 8 ? 8 : 16;
+    lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)ns_t * (na + 1) * 8 + 16;
     if (ns <= 4) { if (nc4) KG(4, 4); else KG(4, 8); }
     else if (ns <= 8) { if (nc4) KG(8, 4); else KG(8, 8); }
     else if (ns <= 16) { if (nc4) KG(16, 4); else KG(16, 8); }
