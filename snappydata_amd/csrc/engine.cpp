@@ -629,6 +629,10 @@ struct sn_query {
   int g1cap = 0, g2cap = 0;             /* per-group-col slot counts */
   int gnull1 = -1, gnull2 = -1;         /* null slot index per group col (-1: none) */
   bool grouped_nonnull_ok = true;
+  /* grouped-mode device aggregate dedup: logical agg -> device sweep index
+   * (-1 = COUNT(*), derived from the per-slot rowcount) */
+  int agg_map[SN_MAX_AGGS];
+  int dev_naggs = 0;
   /* device buffers */
   double *dev_out = nullptr;
   size_t out_stride = 0;                /* 2*NA_t+1 of the launched template */
@@ -755,8 +759,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   } else {
     q->nslots = 1;
   }
-  q->na_t = template_naggs(plan->naggs);
-  q->out_stride = 2 * (size_t)q->na_t + 1;
+  /* (na_t set after device-aggregate dedup below) */
 
   /* build device plan — canonical branchless forms:
    * predicates as closed intervals (strictness folded via nextafter / +-1,
@@ -794,32 +797,56 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       d.lo = lo; d.hi = hi;
     }
   }
-  for (int a = 0; a < plan->naggs; a++) {
-    sn_dev_agg &da = dp.aggs[a];
-    da.a0 = da.a1 = da.a2 = 1.0;
-    da.m0 = da.m1 = da.m2 = 0.0;
-    da.c0 = da.c1 = da.c2 = 0;
-    da.nf = 0;
-    if (plan->aggs[a].kind == SN_AGG_COUNT_STAR) continue;
-    da.nf = plan->aggs[a].nfactors;
-    const sn_agg &sa = plan->aggs[a];
-    if (sa.nfactors >= 1) {
-      da.c0 = q->cslot_of_col[sa.factors[0].col];
-      da.a0 = sa.factors[0].add; da.m0 = sa.factors[0].mul;
-      if (dp.i64_mask & (1u << da.c0)) {
-        fail(SN_ERR_UNSUPPORTED, "int64 aggregate factors not in round-1 GPU path");
-        return nullptr;
+  {
+    /* canonicalize each aggregate; in grouped mode dedupe identical
+     * expressions and fold COUNT(*) into the per-slot rowcount, so e.g.
+     * Q1's 8 logical aggregates run 5 device sweeps (sum/avg pairs share) */
+    const bool grouped = plan->ngroup > 0;
+    int ndev = 0;
+    for (int a = 0; a < plan->naggs; a++) {
+      sn_dev_agg da;
+      da.a0 = da.a1 = da.a2 = 1.0;
+      da.m0 = da.m1 = da.m2 = 0.0;
+      da.c0 = da.c1 = da.c2 = 0;
+      da.nf = 0;
+      const sn_agg &sa = plan->aggs[a];
+      if (sa.kind != SN_AGG_COUNT_STAR) {
+        da.nf = sa.nfactors;
+        if (sa.nfactors >= 1) {
+          da.c0 = q->cslot_of_col[sa.factors[0].col];
+          da.a0 = sa.factors[0].add; da.m0 = sa.factors[0].mul;
+          if (dp.i64_mask & (1u << da.c0)) {
+            fail(SN_ERR_UNSUPPORTED, "int64 aggregate factors not in round-1 GPU path");
+            return nullptr;
+          }
+        }
+        if (sa.nfactors >= 2) {
+          da.c1 = q->cslot_of_col[sa.factors[1].col];
+          da.a1 = sa.factors[1].add; da.m1 = sa.factors[1].mul;
+        }
+        if (sa.nfactors >= 3) {
+          da.c2 = q->cslot_of_col[sa.factors[2].col];
+          da.a2 = sa.factors[2].add; da.m2 = sa.factors[2].mul;
+        }
       }
+      if (grouped && sa.kind == SN_AGG_COUNT_STAR) {
+        q->agg_map[a] = -1;
+        continue;
+      }
+      int idx = -1;
+      if (grouped) {
+        for (int j = 0; j < ndev; j++)
+          if (memcmp(&dp.aggs[j], &da, sizeof(da)) == 0) { idx = j; break; }
+      }
+      if (idx < 0) { idx = ndev; dp.aggs[ndev++] = da; }
+      q->agg_map[a] = idx;
     }
-    if (sa.nfactors >= 2) {
-      da.c1 = q->cslot_of_col[sa.factors[1].col];
-      da.a1 = sa.factors[1].add; da.m1 = sa.factors[1].mul;
-    }
-    if (sa.nfactors >= 3) {
-      da.c2 = q->cslot_of_col[sa.factors[2].col];
-      da.a2 = sa.factors[2].add; da.m2 = sa.factors[2].mul;
-    }
+    q->dev_naggs = grouped ? ndev : plan->naggs;
+    dp.naggs = q->dev_naggs;
   }
+
+  q->na_t = template_naggs(q->dev_naggs > 0 ? q->dev_naggs : 1);
+  q->out_stride = 2 * (size_t)q->na_t + 1;
 
   /* batch descriptors + tile map (stats skip applied here) */
   std::vector<sn_dev_batch> hbatches;
@@ -947,7 +974,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     /* block-partial scratch rows */
     int grid = (int)htiles.size() < SN_GRID_CAP ? (int)htiles.size() : SN_GRID_CAP;
     size_t nv = dp.nslots <= 1 ? (size_t)(2 * q->na_t + 1)
-                               : (size_t)dp.nslots * (dp.naggs + 1);
+                               : (size_t)dp.nslots * (q->dev_naggs + 1);
     size_t need = (size_t)grid * nv * 8;
     if (e->scratch_sz < need) {
       e->scratch = (double *)e->arena.alloc(need);
@@ -1004,13 +1031,15 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
     GroupOut g;
     g.rowcount = rowcount;
     for (int a = 0; a < p.naggs; a++) {
-      g.sums[a] = row[a];
       if (p.ngroup == 0) {
+        g.sums[a] = row[a];
         g.counts[a] = row[q->na_t + a];
       } else {
-        /* grouped kernel: non-null inputs enforced -> count = rowcount,
-         * COUNT(*) sums 1.0 per row */
-        g.counts[a] = p.aggs[a].kind == SN_AGG_COUNT_STAR ? g.sums[a] : rowcount;
+        /* grouped: deduped device sweeps; COUNT(*) = rowcount; non-null
+         * inputs enforced -> per-agg count = rowcount */
+        int di = q->agg_map[a];
+        g.sums[a] = di < 0 ? rowcount : row[di];
+        g.counts[a] = rowcount;
       }
     }
     if (p.ngroup >= 1) {

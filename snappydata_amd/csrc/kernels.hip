@@ -719,32 +719,45 @@ __global__ void k_grouped(sn_dev_plan plan,
         }
       }
 
-      /* pass B: one sweep per aggregate, params hoisted, slot-predicated
-       * register accumulators (NSLOTS live at a time) */
-      for (int a = 0; a < naggs; a++) {
-        const sn_dev_agg A = P->aggs[a];
-        double sums[NSLOTS];
+      /* pass B: aggregate PAIRS per sweep, params hoisted, slot-predicated
+       * register accumulators (2 x NSLOTS live at a time) */
+      for (int a0 = 0; a0 < naggs; a0 += 2) {
+        const sn_dev_agg A = P->aggs[a0];
+        const int has2 = a0 + 1 < naggs;
+        const sn_dev_agg B2 = P->aggs[has2 ? a0 + 1 : a0];
+        double sa[NSLOTS], sb[NSLOTS];
 #pragma unroll
-        for (int s = 0; s < NSLOTS; s++) sums[s] = 0.0;
-#pragma unroll 2
+        for (int s = 0; s < NSLOTS; s++) { sa[s] = 0.0; sb[s] = 0.0; }
+#pragma unroll 4
         for (int k = 0; k < CHUNK / WG; k++) {
           const int r = tid + k * WG;
           const uint64_t w = salive[r >> 6];
           if (w == 0) continue;
           const int m = (int)((w >> (tid & 63)) & 1ull);
           const int slot = sslot[r];
-          const double val = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
-                             (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
-                             (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+          const double va = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
+                            (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
+                            (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+          const double vb = (B2.a0 + B2.m0 * sval[(size_t)B2.c0 * CHUNK + r]) *
+                            (B2.a1 + B2.m1 * sval[(size_t)B2.c1 * CHUNK + r]) *
+                            (B2.a2 + B2.m2 * sval[(size_t)B2.c2 * CHUNK + r]);
 #pragma unroll
-          for (int s = 0; s < NSLOTS; s++)
-            sums[s] += (m && slot == s) ? val : 0.0;
+          for (int s = 0; s < NSLOTS; s++) {
+            const int ms = m && slot == s;
+            sa[s] += ms ? va : 0.0;
+            sb[s] += ms ? vb : 0.0;
+          }
         }
 #pragma unroll
         for (int s = 0; s < NSLOTS; s++) {
-          double x = wave_sum(sums[s]);
+          double x = wave_sum(sa[s]);
           if ((tid & 63) == 0 && x != 0.0)
-            atomicAdd(&bacc[s * (naggs + 1) + a], x);
+            atomicAdd(&bacc[s * (naggs + 1) + a0], x);
+          if (has2) {
+            x = wave_sum(sb[s]);
+            if ((tid & 63) == 0 && x != 0.0)
+              atomicAdd(&bacc[s * (naggs + 1) + a0 + 1], x);
+          }
         }
       }
       __syncthreads();

@@ -34,13 +34,27 @@ shapes = {
     "7col_sum_nopred":      dict(preds=[], aggs=[("sum", [(Q,0.,1.)]), ("sum", [(EP,0.,1.)]),
                                                   ("sum", [(DI,0.,1.)]), ("sum", [(TX,0.,1.)]),
                                                   ("sum", [(SH,0.,1.)])]),
+    "grp_1agg":             dict(preds=[], group_cols=[RF, LS], aggs=[("sum", [(Q,0.,1.)])]),
+    "grp_2agg":             dict(preds=[], group_cols=[RF, LS], aggs=[("sum", [(Q,0.,1.)]), ("count", [])]),
+    "grp_4agg":             dict(preds=[], group_cols=[RF, LS],
+                                 aggs=[("sum", [(Q,0.,1.)]), ("sum", [(EP,0.,1.)]),
+                                       ("sum", [(DI,0.,1.)]), ("count", [])]),
+    "grp_q1":               dict(preds=[dict(col=SH, hi=days(1997,10,2))], group_cols=[RF, LS],
+                                 aggs=[("sum", [(Q,0.,1.)]), ("sum", [(EP,0.,1.)]),
+                                       ("sum", [(EP,0.,1.), (DI,1.,-1.)]),
+                                       ("sum", [(EP,0.,1.), (DI,1.,-1.), (TX,1.,1.)]),
+                                       ("avg", [(Q,0.,1.)]), ("avg", [(EP,0.,1.)]),
+                                       ("avg", [(DI,0.,1.)]), ("count", [])]),
+    "grp_1agg_1grpcol":     dict(preds=[], group_cols=[RF], aggs=[("sum", [(Q,0.,1.)])]),
 }
 
 BYTES = {"1col_sum_nopred": 8, "2col_sum_1pred": 12, "4col_sum_nopred": 28,
-         "q6_cols_sum_3pred": 28, "q6_1pred": 28, "7col_sum_nopred": 36}
+         "q6_cols_sum_3pred": 28, "q6_1pred": 28, "7col_sum_nopred": 36,
+         "grp_1agg": 12, "grp_2agg": 12, "grp_4agg": 28, "grp_q1": 40,
+         "grp_1agg_1grpcol": 10}
 
 for name, sh in shapes.items():
-    plan = abi.make_plan(table=t, **sh)
+    print("running", name, flush=True); plan = abi.make_plan(table=t, **sh)
     for _ in range(3):
         q = eng.query(plan); q.wait(); km = q.kernel_ms(); q.close()
     kms = []
@@ -48,4 +62,4 @@ for name, sh in shapes.items():
         q = eng.query(plan); q.wait(); kms.append(q.kernel_ms()); q.close()
     km = float(np.median(kms))
     gbs = N * BYTES[name] / (km/1e3) / 1e9
-    print(f"{name:22s} kernel_ms={km:7.3f}  alg={BYTES[name]:2d}B/row  {gbs:7.0f} GB/s")
+    print(f"{name:22s} kernel_ms={km:7.3f}  alg={BYTES[name]:2d}B/row  {gbs:7.0f} GB/s", flush=True)
