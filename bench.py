@@ -79,7 +79,38 @@ WORKLOADS = {
     "tpch_q6_lineitem_sf10": (q6_plan, 60_000_000, 28),   # ship4+qty8+ep8+disc8
     "tpch_q1_lineitem_sf100": (q1_plan, 600_000_000, 40),  # +tax8+2x2dict, ship4
     "tpch_q1_lineitem_sf10": (q1_plan, 60_000_000, 40),
+    # BASELINE config 5: Q6 scan with ~2% of rows update-patched and ~1%
+    # deleted (ColumnDeltaDecoder/delete-mask merge during the scan)
+    "tpch_q6_mut_sf10": (q6_plan, 60_000_000, 28),
 }
+
+
+def build_mutable_lineitem(eng, t, total_rows, seed, batch_rows=600_000):
+    """Config-5 table: product-encoded batches carrying update deltas on
+    l_discount (~2% of rows) and delete masks (~1%)."""
+    rng = np.random.default_rng(seed + 1)
+    bi = 0
+    for start in range(0, total_rows, batch_rows):
+        n = min(batch_rows, total_rows - start)
+        d = se.gen_lineitem_arrays(start, n, seed)
+        cols = [se.encode_column(abi.T_DOUBLE, d["qty"]),
+                se.encode_column(abi.T_DOUBLE, d["ep"]),
+                se.encode_column(abi.T_DOUBLE, d["disc"]),
+                se.encode_column(abi.T_DOUBLE, d["tax"]),
+                se.encode_column(abi.T_STRING, b"".join(d["rf"]),
+                                 lens=np.ones(n, dtype=np.int32)),
+                se.encode_column(abi.T_STRING, b"".join(d["ls"]),
+                                 lens=np.ones(n, dtype=np.int32)),
+                se.encode_column(abi.T_INT32, d["ship"])]
+        upd = np.unique(rng.integers(0, n, max(1, n // 50))).astype(np.int32)
+        d1 = se.encode_update_delta(abi.T_DOUBLE, upd, n,
+                                    rng.integers(0, 11, len(upd)) / 100.0)
+        deltas = [(None, None), (None, None), (d1, None), (None, None),
+                  (None, None), (None, None), (None, None)]
+        dels = np.unique(rng.integers(0, n, max(1, n // 100))).astype(np.int32)
+        dmask = se.encode_delete_mask(dels, n)
+        eng.batch_put(t, bi, bi, -n, cols, delete_mask=dmask, deltas=deltas)
+        bi += 1
 
 
 def cpu_baseline_leg(workload, seed, target_seconds=10.0):
@@ -151,7 +182,10 @@ def main():
     eng = se.Engine(device=local_rank, shard_rank=rank, shard_count=world,
                     n_buckets=max(128, world * 16))
     t = eng.table_define("lineitem", LINEITEM_SCHEMA)
-    eng.datagen_lineitem(t, total_rows, seed=args.seed, batch_rows=600_000)
+    if args.workload == "tpch_q6_mut_sf10":
+        build_mutable_lineitem(eng, t, total_rows, args.seed)
+    else:
+        eng.datagen_lineitem(t, total_rows, seed=args.seed, batch_rows=600_000)
     resident = eng.num_rows(t)
     plan = plan_fn(t)
     grouped = plan.ngroup > 0

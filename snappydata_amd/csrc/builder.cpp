@@ -356,3 +356,68 @@ extern "C" int64_t sn_datagen_lineitem(sn_engine *e, int32_t table,
   if (err.load() != SN_OK) return err.load();
   return put_rows.load();
 }
+
+/* ---------------- product-side mutation encoders ----------------
+ * Mirrors the reference's write-side delta/delete encoders:
+ *   delete mask [int32 0][int32 numBaseRows][int32 numPositions][sorted
+ *   int32 positions] — ColumnDeleteEncoder.createFinalBuffer
+ *   (ColumnDeleteEncoder.scala:101-126);
+ *   update delta: standard blob header (null bitset over DELTA entries) +
+ *   [numBaseRows][numPositions][positions][pad8][uncompressed body] —
+ *   ColumnDeltaEncoder header doc (ColumnDeltaEncoder.scala:40-80).       */
+extern "C" int64_t sn_encode_delete_mask(const int32_t *pos, int32_t n,
+                                         int32_t num_base_rows, uint8_t *out,
+                                         int64_t cap) {
+  int64_t need = 12 + (int64_t)n * 4;
+  if (cap < need) return SN_ERR_NOMEM;
+  int32_t zero = 0;
+  memcpy(out, &zero, 4);
+  memcpy(out + 4, &num_base_rows, 4);
+  memcpy(out + 8, &n, 4);
+  memcpy(out + 12, pos, (size_t)n * 4);
+  return need;
+}
+
+extern "C" int64_t sn_encode_update_delta(int32_t dtype, const int32_t *pos,
+                                          int32_t n, int32_t num_base_rows,
+                                          const void *values,
+                                          const int32_t *str_lens,
+                                          const uint8_t *valid, uint8_t *out,
+                                          int64_t cap) {
+  sn_ingest_col col;
+  col.data = values; col.str_lens = str_lens; col.valid = valid;
+  std::vector<uint8_t> body;
+  int rc = encode_column((sn_type_t)dtype, col, n, body);
+  if (rc != SN_OK) return rc;
+  /* splice positions between the null header and the encoding body */
+  int32_t null_bytes;
+  memcpy(&null_bytes, body.data() + 4, 4);
+  int64_t hdr = 8 + null_bytes;
+  int64_t off = hdr + 8 + (int64_t)n * 4;
+  int64_t pad = ((off + 7) & ~7ll) - off;
+  int64_t need = (int64_t)body.size() + 8 + (int64_t)n * 4 + pad;
+  if (cap < need) return SN_ERR_NOMEM;
+  memcpy(out, body.data(), (size_t)hdr);
+  memcpy(out + hdr, &num_base_rows, 4);
+  memcpy(out + hdr + 4, &n, 4);
+  memcpy(out + hdr + 8, pos, (size_t)n * 4);
+  memset(out + off, 0, (size_t)pad);
+  memcpy(out + off + pad, body.data() + hdr, body.size() - (size_t)hdr);
+  return need;
+}
+
+/* expose the column encoder itself (ColumnEncoder mirror) so external hosts
+ * can build reference-format blobs without the engine's batching policy */
+extern "C" int64_t sn_encode_column(int32_t dtype, const void *data,
+                                    const int32_t *str_lens,
+                                    const uint8_t *valid, int32_t count,
+                                    uint8_t *out, int64_t cap) {
+  sn_ingest_col col;
+  col.data = data; col.str_lens = str_lens; col.valid = valid;
+  std::vector<uint8_t> body;
+  int rc = encode_column((sn_type_t)dtype, col, count, body);
+  if (rc != SN_OK) return rc;
+  if ((int64_t)body.size() > cap) return SN_ERR_NOMEM;
+  memcpy(out, body.data(), body.size());
+  return (int64_t)body.size();
+}

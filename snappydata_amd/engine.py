@@ -94,6 +94,19 @@ def lib():
         L.sn_datagen_lineitem.restype = C.c_int64
         L.sn_datagen_lineitem.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
                                           C.c_int64, C.c_int32, C.c_int32]
+        L.sn_encode_column.restype = C.c_int64
+        L.sn_encode_column.argtypes = [C.c_int32, C.c_void_p, C.POINTER(C.c_int32),
+                                       C.POINTER(C.c_uint8), C.c_int32,
+                                       C.c_void_p, C.c_int64]
+        L.sn_encode_delete_mask.restype = C.c_int64
+        L.sn_encode_delete_mask.argtypes = [C.POINTER(C.c_int32), C.c_int32,
+                                            C.c_int32, C.c_void_p, C.c_int64]
+        L.sn_encode_update_delta.restype = C.c_int64
+        L.sn_encode_update_delta.argtypes = [C.c_int32, C.POINTER(C.c_int32),
+                                             C.c_int32, C.c_int32, C.c_void_p,
+                                             C.POINTER(C.c_int32),
+                                             C.POINTER(C.c_uint8), C.c_void_p,
+                                             C.c_int64]
         L.sn_gen_lineitem_arrays.argtypes = [
             C.c_int64, C.c_int32, C.c_int64,
             C.POINTER(C.c_double), C.POINTER(C.c_double), C.POINTER(C.c_double),
@@ -111,6 +124,46 @@ def _check(rc, what=""):
     if rc < 0:
         raise EngineError(rc, f"{what}: {last_error()}")
     return rc
+
+
+def encode_column(dtype, data, lens=None, valid=None):
+    """Product column encoder (reference blob format)."""
+    data = np.ascontiguousarray(data) if not isinstance(data, (bytes, bytearray)) \
+        else np.frombuffer(data, dtype=np.uint8)
+    count = len(lens) if lens is not None else len(data)
+    cap = 64 + int(data.nbytes) * 2 + count * 8
+    out = np.zeros(cap, dtype=np.uint8)
+    lp = np.ascontiguousarray(lens, dtype=np.int32).ctypes.data_as(
+        C.POINTER(C.c_int32)) if lens is not None else None
+    vp = np.ascontiguousarray(valid, dtype=np.uint8).ctypes.data_as(
+        C.POINTER(C.c_uint8)) if valid is not None else None
+    n = _check(lib().sn_encode_column(dtype, data.ctypes.data, lp, vp, count,
+                                      out.ctypes.data, cap), "encode_column")
+    return out[:n].tobytes()
+
+
+def encode_delete_mask(positions, num_base_rows):
+    pos = np.ascontiguousarray(positions, dtype=np.int32)
+    cap = 16 + 4 * len(pos)
+    out = np.zeros(cap, dtype=np.uint8)
+    n = _check(lib().sn_encode_delete_mask(
+        pos.ctypes.data_as(C.POINTER(C.c_int32)), len(pos), num_base_rows,
+        out.ctypes.data, cap), "encode_delete_mask")
+    return out[:n].tobytes()
+
+
+def encode_update_delta(dtype, positions, num_base_rows, values, valid=None):
+    pos = np.ascontiguousarray(positions, dtype=np.int32)
+    vals = np.ascontiguousarray(values)
+    cap = 128 + len(pos) * 40 + int(vals.nbytes) * 2
+    out = np.zeros(cap, dtype=np.uint8)
+    vp = np.ascontiguousarray(valid, dtype=np.uint8).ctypes.data_as(
+        C.POINTER(C.c_uint8)) if valid is not None else None
+    n = _check(lib().sn_encode_update_delta(
+        dtype, pos.ctypes.data_as(C.POINTER(C.c_int32)), len(pos),
+        num_base_rows, vals.ctypes.data, None, vp, out.ctypes.data, cap),
+        "encode_update_delta")
+    return out[:n].tobytes()
 
 
 def gen_lineitem_arrays(start_row, n, seed):

@@ -216,3 +216,70 @@ def test_empty_result_on_gpu(eng, li_fixture_table):
                          aggs=[("sum", [(tu.COL_EP, 0.0, 1.0)]), ("count", [])])
     rows = eng.query(plan).rows()
     assert rows == [((), [None, 0.0])]
+
+
+def test_sharded_two_engines_merge(eng):
+    """The multi-GPU path on one device: two shard engines (rank 0/1 of 2)
+    each hold disjoint buckets; gathered partial blocks merge to the same
+    result as an unsharded engine — the exact semantics bench.py runs over
+    RCCL at N>1 (partial->final exchange, SnappyHashAggregateExec partial
+    modes)."""
+    n = 2_000_000
+    shards = []
+    for rank in range(2):
+        e = se.Engine(device=0, shard_rank=rank, shard_count=2)
+        t = e.table_define("li", [(abi.T_DOUBLE, False)] * 4 +
+                           [(abi.T_STRING, False)] * 2 + [(abi.T_INT32, False)])
+        e.datagen_lineitem(t, n, seed=7, batch_rows=100_000)
+        shards.append((e, t))
+    full = eng.table_define("li_full", [(abi.T_DOUBLE, False)] * 4 +
+                            [(abi.T_STRING, False)] * 2 + [(abi.T_INT32, False)])
+    eng.datagen_lineitem(full, n, seed=7, batch_rows=100_000)
+    assert shards[0][0].num_rows(0) + shards[1][0].num_rows(0) == n
+
+    for plan_fn, count_aggs in ((q6_plan_engine, set()), (q1_plan_engine, {7})):
+        qs = [e.query(plan_fn(t)) for e, t in shards]
+        blocks = [q.partials_host() for q in qs]
+        stride = len(blocks[0])
+        assert len(blocks[1]) == stride  # equal-size blocks (all_gather contract)
+        gathered = np.concatenate(blocks)
+        qs[0].merge_host(gathered, stride, 2)
+        merged = qs[0].rows()
+        direct = eng.query(plan_fn(full)).rows()
+        compare_results(merged, direct, count_aggs=count_aggs)
+        for q in qs:
+            q.close()
+    for e, _ in shards:
+        e.close()
+
+
+def test_q6_with_deltas_workload(eng):
+    """BASELINE config 5 shape: Q6 scan with ~2% of rows covered by update
+    deltas and ~1% deleted, vs the oracle on identical inputs."""
+    n = 500_000
+    rng = np.random.default_rng(11)
+    d = se.gen_lineitem_arrays(0, n, seed=11)
+    cols = [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, d["qty"]),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, d["ep"]),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, d["disc"]),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, d["tax"]),
+            po.encode(po.T_STRING, po.ENC_DICT, d["rf"]),
+            po.encode(po.T_STRING, po.ENC_DICT, d["ls"]),
+            po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, d["ship"])]
+    upd = np.unique(rng.integers(0, n, 10_000)).astype(np.int32)
+    upd_vals = rng.random(len(upd))  # new discounts
+    d1 = po.encode_delta(po.T_DOUBLE, po.ENC_UNCOMPRESSED, upd, n, upd_vals)
+    deltas = [(None, None), (None, None), (d1, None), (None, None),
+              (None, None), (None, None), (None, None)]
+    dels = np.unique(rng.integers(0, n, 5_000)).astype(np.int32)
+    dmask = po.encode_delete(dels, n)
+
+    t = eng.table_define("li_d5", [(abi.T_DOUBLE, False)] * 4 +
+                         [(abi.T_STRING, False)] * 2 + [(abi.T_INT32, False)])
+    eng.batch_put(t, 0, 0, -n, cols, delete_mask=dmask, deltas=deltas)
+    grows = eng.query(q6_plan_engine(t)).rows()
+
+    ot = po.OracleTable(tu.LINEITEM_DTYPES)
+    ot.add_batch(-n, cols, delete_mask=dmask, deltas=deltas)
+    orows = po.result_rows(ot.query(tu.q6_plan()))
+    assert_close(grows[0][1][0], orows[0][1][0])
