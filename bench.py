@@ -113,6 +113,14 @@ def build_mutable_lineitem(eng, t, total_rows, seed, batch_rows=600_000):
         bi += 1
 
 
+def _pmc_traffic(workload, resident_rows):
+    try:
+        cal = json.load(open(os.path.join(REPO, "profiles", "traffic.json")))
+        return round(cal[workload]["bytes_per_row"] * resident_rows)
+    except Exception:
+        return None
+
+
 def cpu_baseline_leg(workload, seed, target_seconds=10.0):
     """Time the CPU oracle (reference-loop restatement, OpenMP over all host
     cores) on a bounded sample of the same workload.  Returns the dict for
@@ -265,7 +273,10 @@ def main():
                 "peak": HBM_PEAK_GBPS,
                 "unit": "GB/s",
                 "frac": round(achieved / HBM_PEAK_GBPS, 4),
-                "traffic": None,   # PMC FETCH/WRITE via rocprofv3 — profiles/
+                # PMC-measured HBM bytes/launch (rocprofv3 FETCH_SIZE x2 +
+                # WRITE_SIZE, collected per profiles/README.md) scaled to
+                # this run's resident rows; null when unmeasured
+                "traffic": _pmc_traffic(args.workload, resident),
                 "kernel_ms": round(avg_ms, 4),
             }
         cpu_baseline = None
