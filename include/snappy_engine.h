@@ -168,6 +168,10 @@ typedef struct {
   sn_factor factors[SN_MAX_FACTORS];
 } sn_agg;
 
+#define SN_JOIN_NONE  -1
+#define SN_JOIN_SEMI   0   /* fact row survives iff key present in dim */
+#define SN_JOIN_GROUP  1   /* additionally GROUP BY the dim attribute */
+
 typedef struct {
   int32_t table;        /* handle from sn_table_define */
   int32_t npreds;
@@ -177,6 +181,15 @@ typedef struct {
   int32_t naggs;
   int32_t _pad;
   sn_agg  aggs[SN_MAX_AGGS];
+  /* broadcast-dimension join (HashJoinExec semantics, HashJoinExec.scala:
+   * 285-520: per-task ObjectHashSet build from the replicated row-store
+   * dimension, cached across tasks via HashedObjectCache :449-470; here the
+   * device hash table is built once per dimension and cached on the engine,
+   * resident in HBM — the broadcast).  join_dim = SN_JOIN_NONE disables. */
+  int32_t join_dim;       /* handle from sn_dim_define */
+  int32_t join_fact_col;  /* int32/int64 fact key column */
+  int32_t join_mode;      /* SN_JOIN_SEMI | SN_JOIN_GROUP */
+  int32_t _pad4;
 } sn_plan;
 
 /* ---- results ----
@@ -227,6 +240,18 @@ int32_t sn_batch_put(sn_engine *e, int32_t table,
 /* number of resident batches / rows for a table on this shard */
 int64_t sn_table_num_batches(sn_engine *e, int32_t table);
 int64_t sn_table_num_rows(sn_engine *e, int32_t table);
+
+/* ---- dimension tables (row-store stand-in for the broadcast join) ----
+ * The reference keeps dimension tables in the GemFire row store and builds
+ * the probe map per task from region get()s; here the host holds the rows
+ * and the engine broadcasts a key -> attribute hash table into HBM. */
+int32_t sn_dim_define(sn_engine *e, const char *name);
+/* keys must be unique (the dimension's primary key).  attrs: optional
+ * per-key attribute strings (payload + lens) for SN_JOIN_GROUP; NULL for
+ * semi-join-only dimensions. */
+int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
+                   const int64_t *keys, const char *attr_payload,
+                   const int32_t *attr_lens);
 
 /* ---- query plane ---- */
 sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan);

@@ -55,7 +55,9 @@ class SnPlan(C.Structure):
                 ("ngroup", C.c_int32),
                 ("group_cols", C.c_int32 * SN_MAX_GROUPS),
                 ("naggs", C.c_int32), ("_pad", C.c_int32),
-                ("aggs", SnAgg * SN_MAX_AGGS)]
+                ("aggs", SnAgg * SN_MAX_AGGS),
+                ("join_dim", C.c_int32), ("join_fact_col", C.c_int32),
+                ("join_mode", C.c_int32), ("_pad4", C.c_int32)]
 
 
 class SnResult(C.Structure):
@@ -90,6 +92,10 @@ def lib():
         _lib.sno_table_add_batch.restype = C.c_int32
         _lib.sno_table_add_batch.argtypes = [C.c_void_p, C.c_int32, C.POINTER(SnBuf),
                                              C.POINTER(SnBuf), C.POINTER(SnBuf), C.POINTER(SnBuf)]
+        _lib.sno_table_set_dim.restype = C.c_int32
+        _lib.sno_table_set_dim.argtypes = [C.c_void_p, C.POINTER(C.c_int64),
+                                           C.c_int64, C.c_char_p,
+                                           C.POINTER(C.c_int32)]
         _lib.sno_query.restype = C.c_int32
         _lib.sno_query.argtypes = [C.c_void_p, C.POINTER(SnPlan), C.POINTER(SnResult), C.c_int32]
         _lib.sno_encode.restype = C.c_int64
@@ -304,6 +310,19 @@ class OracleTable:
                                        dl)
         assert rc == 0, f"add_batch failed: {rc}"
 
+    def set_dim(self, keys, attrs=None):
+        """Register the broadcast dimension: keys int64 array; attrs optional
+        list of bytes (per key)."""
+        keys = np.ascontiguousarray(keys, dtype=np.int64)
+        payload, lens = None, None
+        if attrs is not None:
+            lens = np.array([len(a) for a in attrs], dtype=np.int32)
+            payload = b"".join(attrs)
+        rc = lib().sno_table_set_dim(
+            self._h, keys.ctypes.data_as(C.POINTER(C.c_int64)), len(keys),
+            payload, lens.ctypes.data_as(C.POINTER(C.c_int32)) if lens is not None else None)
+        assert rc == 0, rc
+
     def query(self, plan, nthreads=1):
         res = SnResult()
         rc = lib().sno_query(self._h, C.byref(plan), C.byref(res), nthreads)
@@ -318,7 +337,7 @@ class OracleTable:
             pass
 
 
-def make_plan(preds=(), group_cols=(), aggs=()):
+def make_plan(preds=(), group_cols=(), aggs=(), join=None):
     """preds: list of dicts {col, lo, hi, lo_strict, hi_strict, is_double}
     aggs: list of ('sum'|'avg'|'count', [(col, add, mul), ...])"""
     p = SnPlan()
@@ -342,6 +361,11 @@ def make_plan(preds=(), group_cols=(), aggs=()):
     p.ngroup = len(group_cols)
     for i, c in enumerate(group_cols):
         p.group_cols[i] = c
+    p.join_dim = -1
+    if join is not None:
+        p.join_dim = join["dim"]
+        p.join_fact_col = join["fact_col"]
+        p.join_mode = 1 if join.get("group") else 0
     p.naggs = len(aggs)
     for i, (kind, factors) in enumerate(aggs):
         ag = p.aggs[i]

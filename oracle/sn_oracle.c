@@ -485,6 +485,14 @@ typedef struct sno_table {
   uint8_t *nullable;
   int32_t nbatches, cap;
   sno_batch *batches;
+  /* broadcast dimension (HashJoinExec stand-in): open-address key table +
+   * per-key attribute strings (independent restatement for parity checks) */
+  int64_t *dim_hk;         /* capacity dim_cap, sentinel INT64_MIN */
+  int32_t *dim_hidx;       /* key ordinal per slot */
+  int64_t  dim_cap;
+  char    *dim_attr_pay;
+  int32_t *dim_attr_off;   /* nkeys+1 offsets into dim_attr_pay */
+  int64_t  dim_n;
 } sno_table;
 
 SNO_EXPORT sno_table *sno_table_create(int32_t ncols, const int32_t *dtypes,
@@ -545,6 +553,59 @@ SNO_EXPORT int32_t sno_table_add_batch(sno_table *t, int32_t num_rows,
   return SN_OK;
 }
 
+static uint64_t sno_mix64(uint64_t x) {
+  x += 0x9E3779B97f4A7C15ULL;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+  return x ^ (x >> 31);
+}
+
+SNO_EXPORT int32_t sno_table_set_dim(sno_table *t, const int64_t *keys,
+                                     int64_t n, const char *attr_payload,
+                                     const int32_t *attr_lens) {
+  if (!t || !keys || n <= 0) return SN_ERR_BADARG;
+  int64_t cap = 2;
+  while (cap < 2 * n + 1) cap <<= 1;
+  t->dim_hk = (int64_t *)malloc((size_t)cap * 8);
+  t->dim_hidx = (int32_t *)malloc((size_t)cap * 4);
+  for (int64_t i = 0; i < cap; i++) t->dim_hk[i] = INT64_MIN;
+  t->dim_cap = cap;
+  t->dim_n = n;
+  int64_t total = 0;
+  if (attr_payload && attr_lens)
+    for (int64_t i = 0; i < n; i++) total += attr_lens[i];
+  t->dim_attr_pay = (char *)malloc((size_t)(total ? total : 1));
+  t->dim_attr_off = (int32_t *)malloc((size_t)(n + 1) * 4);
+  int64_t off = 0;
+  for (int64_t i = 0; i < n; i++) {
+    t->dim_attr_off[i] = (int32_t)off;
+    if (attr_payload && attr_lens) {
+      memcpy(t->dim_attr_pay + off, attr_payload + off, (size_t)attr_lens[i]);
+      off += attr_lens[i];
+    }
+    uint64_t h = sno_mix64((uint64_t)keys[i]) & (uint64_t)(cap - 1);
+    while (t->dim_hk[h] != INT64_MIN) {
+      if (t->dim_hk[h] == keys[i]) return SN_ERR_BADARG;
+      h = (h + 1) & (uint64_t)(cap - 1);
+    }
+    t->dim_hk[h] = keys[i];
+    t->dim_hidx[h] = (int32_t)i;
+  }
+  t->dim_attr_off[n] = (int32_t)off;
+  return SN_OK;
+}
+
+static int dim_lookup(const sno_table *t, int64_t key) {
+  if (!t->dim_hk) return -1;
+  uint64_t h = sno_mix64((uint64_t)key) & (uint64_t)(t->dim_cap - 1);
+  while (1) {
+    int64_t k0 = t->dim_hk[h];
+    if (k0 == key) return t->dim_hidx[h];
+    if (k0 == INT64_MIN) return -1;
+    h = (h + 1) & (uint64_t)(t->dim_cap - 1);
+  }
+}
+
 SNO_EXPORT void sno_table_destroy(sno_table *t) {
   if (!t) return;
   for (int32_t bi = 0; bi < t->nbatches; bi++) {
@@ -557,7 +618,9 @@ SNO_EXPORT void sno_table_destroy(sno_table *t) {
       free(b->deltas); free(b->delta_lens);
     }
   }
-  free(t->batches); free(t->dtypes); free(t->nullable); free(t);
+  free(t->batches); free(t->dtypes); free(t->nullable);
+  free(t->dim_hk); free(t->dim_hidx); free(t->dim_attr_pay); free(t->dim_attr_off);
+  free(t);
 }
 
 /* =======================================================================
@@ -744,6 +807,9 @@ static int eval_batch(const sno_table *t, const sno_batch *b, const sn_plan *p,
   int rc = SN_OK;
   int nc = t->ncols;
   col_state *cs = (col_state *)calloc((size_t)nc, sizeof(col_state));
+  const int has_join = p->join_dim != SN_JOIN_NONE && t->dim_hk != NULL;
+  const int join_group = has_join && p->join_mode == SN_JOIN_GROUP;
+  if (has_join) cs[p->join_fact_col].needed = 1;
   for (int i = 0; i < p->npreds; i++) cs[p->preds[i].col].needed = 1;
   for (int i = 0; i < p->ngroup; i++) { cs[p->group_cols[i]].needed = 1; cs[p->group_cols[i]].is_group = 1; }
   for (int a = 0; a < p->naggs; a++)
@@ -833,6 +899,15 @@ static int eval_batch(const sno_table *t, const sno_batch *b, const sn_plan *p,
         if (pr->has_hi && (pr->hi_strict ? !(v < pr->hi_i) : !(v <= pr->hi_i))) row_ok = 0;
       }
     }
+    /* broadcast-dimension probe (inner join on unique key) */
+    int dim_ord = -1;
+    if (row_ok && has_join) {
+      if (val_null[p->join_fact_col]) row_ok = 0;
+      else {
+        dim_ord = dim_lookup(t, val_i[p->join_fact_col]);
+        if (dim_ord < 0) row_ok = 0;
+      }
+    }
     if (!row_ok) continue;
     (*rows_passed)++;
 
@@ -840,6 +915,12 @@ static int eval_batch(const sno_table *t, const sno_batch *b, const sn_plan *p,
     char keys[SN_MAX_GROUPS][SN_KEY_MAX];
     uint8_t knull[SN_MAX_GROUPS];
     memset(knull, 0, sizeof(knull));
+    if (join_group) {
+      int32_t a0 = t->dim_attr_off[dim_ord], a1 = t->dim_attr_off[dim_ord + 1];
+      int32_t L = a1 - a0 < SN_KEY_MAX - 1 ? a1 - a0 : SN_KEY_MAX - 1;
+      memcpy(keys[0], t->dim_attr_pay + a0, (size_t)L);
+      keys[0][L] = 0;
+    }
     for (int i = 0; i < p->ngroup; i++) {
       int c = p->group_cols[i];
       if (val_null[c]) { knull[i] = 1; keys[i][0] = 0; }
@@ -893,9 +974,12 @@ SNO_EXPORT int32_t sno_query(sno_table *t, const sn_plan *p, sn_result *out,
     return SN_ERR_BADARG;
   memset(out, 0, sizeof(*out));
   int rc = SN_OK;
+  const int join_group = p->join_dim != SN_JOIN_NONE &&
+                         p->join_mode == SN_JOIN_GROUP && t->dim_hk != NULL;
+  const int eff_ngroup = join_group ? 1 : p->ngroup;
 
   sno_gtab g;
-  gtab_init(&g, p->ngroup, p->naggs);
+  gtab_init(&g, eff_ngroup, p->naggs);
   int64_t rows_scanned = 0, rows_passed = 0, seen = 0, skipped = 0;
 
   if (nthreads <= 1) {
@@ -923,7 +1007,7 @@ SNO_EXPORT int32_t sno_query(sno_table *t, const sn_plan *p, sn_result *out,
     int64_t *tpass = (int64_t *)calloc((size_t)nthreads, 8);
     int64_t *tseen = (int64_t *)calloc((size_t)nthreads, 8);
     int64_t *tskip = (int64_t *)calloc((size_t)nthreads, 8);
-    for (int i = 0; i < nthreads; i++) gtab_init(&tg[i], p->ngroup, p->naggs);
+    for (int i = 0; i < nthreads; i++) gtab_init(&tg[i], eff_ngroup, p->naggs);
 #pragma omp parallel for schedule(dynamic) num_threads(nthreads)
     for (int bi = 0; bi < nb; bi++) {
       int tid = omp_get_thread_num();
@@ -965,7 +1049,7 @@ SNO_EXPORT int32_t sno_query(sno_table *t, const sn_plan *p, sn_result *out,
   if (rc != SN_OK) { gtab_free(&g); return rc; }
 
   /* keyless aggregate with zero rows still yields one row (Spark semantics) */
-  if (p->ngroup == 0 && g.n == 0) {
+  if (eff_ngroup == 0 && g.n == 0) {
     char keys[SN_MAX_GROUPS][SN_KEY_MAX]; uint8_t knull[SN_MAX_GROUPS];
     memset(keys, 0, sizeof(keys)); memset(knull, 0, sizeof(knull));
     gtab_get(&g, (const char (*)[SN_KEY_MAX])keys, knull);
@@ -973,12 +1057,12 @@ SNO_EXPORT int32_t sno_query(sno_table *t, const sn_plan *p, sn_result *out,
 
   qsort(g.groups, (size_t)g.n, sizeof(sno_group), group_cmp);
 
-  out->nrows = g.n; out->ngroup = p->ngroup; out->naggs = p->naggs;
+  out->nrows = g.n; out->ngroup = eff_ngroup; out->naggs = p->naggs;
   out->rows_scanned = rows_scanned; out->rows_passed = rows_passed;
   out->batches_seen = seen; out->batches_skipped = skipped;
   for (int32_t i = 0; i < g.n; i++) {
     sno_group *grp = &g.groups[i];
-    for (int k = 0; k < p->ngroup; k++) {
+    for (int k = 0; k < eff_ngroup; k++) {
       strncpy(out->keys[i][k], grp->keys[k], SN_KEY_MAX);
       out->key_is_null[i][k] = grp->key_null[k];
     }

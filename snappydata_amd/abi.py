@@ -62,7 +62,9 @@ class SnPlan(C.Structure):
                 ("ngroup", C.c_int32),
                 ("group_cols", C.c_int32 * SN_MAX_GROUPS),
                 ("naggs", C.c_int32), ("_pad", C.c_int32),
-                ("aggs", SnAgg * SN_MAX_AGGS)]
+                ("aggs", SnAgg * SN_MAX_AGGS),
+                ("join_dim", C.c_int32), ("join_fact_col", C.c_int32),
+                ("join_mode", C.c_int32), ("_pad4", C.c_int32)]
 
 
 class SnResult(C.Structure):
@@ -81,7 +83,7 @@ class SnIngestCol(C.Structure):
                 ("valid", C.POINTER(C.c_uint8))]
 
 
-def make_plan(table=0, preds=(), group_cols=(), aggs=()):
+def make_plan(table=0, preds=(), group_cols=(), aggs=(), join=None):
     """Same plan-construction convention as the reference's thin planner
     (SnappyStrategies shapes): preds = [{col, lo, hi, lo_strict, hi_strict,
     is_double}], aggs = [(kind, [(col, add, mul), ...])]."""
@@ -106,6 +108,11 @@ def make_plan(table=0, preds=(), group_cols=(), aggs=()):
     p.ngroup = len(group_cols)
     for i, c in enumerate(group_cols):
         p.group_cols[i] = c
+    p.join_dim = -1
+    if join is not None:
+        p.join_dim = join["dim"]
+        p.join_fact_col = join["fact_col"]
+        p.join_mode = 1 if join.get("group") else 0
     p.naggs = len(aggs)
     for i, (kind, factors) in enumerate(aggs):
         ag = p.aggs[i]

@@ -34,6 +34,7 @@
 
 #include <algorithm>
 #include <atomic>
+#include <climits>
 #include <cmath>
 #include <cstdarg>
 #include <cstdio>
@@ -164,10 +165,25 @@ struct Table {
   std::mutex mu;
 };
 
+/* broadcast dimension (row-store stand-in): host rows + cached device
+ * open-address table (HashedObjectCache analogue, HashJoinExec.scala:449-470) */
+struct Dim {
+  std::string name;
+  std::vector<int64_t> keys;
+  std::vector<std::string> attrs;          /* per-key attr (may be empty) */
+  std::vector<std::string> attr_dict;      /* distinct attrs (gid order) */
+  std::vector<int32_t> attr_gid;           /* per-key gid */
+  /* device table (built lazily, cached) */
+  const int64_t *dev_keys = nullptr;
+  const int32_t *dev_payload = nullptr;
+  int32_t cap_log2 = 0;
+};
+
 struct sn_engine {
   sn_config cfg;
   Arena arena;
   std::vector<std::unique_ptr<Table>> tables;
+  std::vector<std::unique_ptr<Dim>> dims;
   hipStream_t stream = nullptr;
   bool has_gpu = false;
   /* reusable per-engine scratch for block-partial reduction rows
@@ -218,6 +234,77 @@ extern "C" int32_t sn_table_define(sn_engine *e, const char *name,
   t->gdict_idx.resize(ncols);
   e->tables.push_back(std::move(t));
   return (int32_t)e->tables.size() - 1;
+}
+
+extern "C" int32_t sn_dim_define(sn_engine *e, const char *name) {
+  if (!e) return SN_ERR_BADARG;
+  std::lock_guard<std::mutex> g(e->mu);
+  auto d = std::make_unique<Dim>();
+  d->name = name ? name : "";
+  e->dims.push_back(std::move(d));
+  return (int32_t)e->dims.size() - 1;
+}
+
+extern "C" int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
+                              const int64_t *keys, const char *attr_payload,
+                              const int32_t *attr_lens) {
+  if (!e || dim < 0 || dim >= (int32_t)e->dims.size() || !keys || nkeys <= 0)
+    return fail(SN_ERR_BADARG, "bad dim args");
+  Dim *d = e->dims[dim].get();
+  int64_t off = 0;
+  for (int64_t i = 0; i < nkeys; i++) {
+    d->keys.push_back(keys[i]);
+    if (attr_payload && attr_lens) {
+      std::string a(attr_payload + off, (size_t)attr_lens[i]);
+      off += attr_lens[i];
+      int32_t gid = -1;
+      for (size_t j = 0; j < d->attr_dict.size(); j++)
+        if (d->attr_dict[j] == a) { gid = (int32_t)j; break; }
+      if (gid < 0) { gid = (int32_t)d->attr_dict.size(); d->attr_dict.push_back(a); }
+      d->attrs.push_back(std::move(a));
+      d->attr_gid.push_back(gid);
+    } else {
+      d->attrs.emplace_back();
+      d->attr_gid.push_back(0);
+    }
+  }
+  /* invalidate the cached device table */
+  d->dev_keys = nullptr; d->dev_payload = nullptr;
+  return SN_OK;
+}
+
+static void *up(sn_engine *e, const void *host, size_t n);
+
+static inline uint64_t mix64h(uint64_t x) {
+  x += 0x9E3779B97f4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+/* build (or reuse) the device open-address table: the broadcast into HBM */
+static int dim_device_table(sn_engine *e, Dim *d) {
+  if (d->dev_keys) return SN_OK;
+  int32_t lg = 1;
+  while ((1u << lg) < 2 * d->keys.size() + 1) lg++;
+  size_t cap = 1ull << lg;
+  std::vector<int64_t> hk(cap, INT64_MIN);
+  std::vector<int32_t> hp(cap, -1);
+  for (size_t i = 0; i < d->keys.size(); i++) {
+    int64_t k = d->keys[i];
+    uint32_t h = (uint32_t)mix64h((uint64_t)k) & (cap - 1);
+    while (hk[h] != INT64_MIN) {
+      if (hk[h] == k) return fail(SN_ERR_BADARG, "duplicate dimension key");
+      h = (h + 1) & (cap - 1);
+    }
+    hk[h] = k;
+    hp[h] = d->attr_gid[i];
+  }
+  d->dev_keys = (const int64_t *)up(e, hk.data(), cap * 8);
+  d->dev_payload = (const int32_t *)up(e, hp.data(), cap * 4);
+  d->cap_log2 = lg;
+  if (!d->dev_keys || !d->dev_payload) return fail(SN_ERR_NOMEM, "dim upload");
+  return SN_OK;
 }
 
 static Table *get_table(sn_engine *e, int32_t t) {
@@ -633,6 +720,8 @@ struct sn_query {
    * (-1 = COUNT(*), derived from the per-slot rowcount) */
   int agg_map[SN_MAX_AGGS];
   int dev_naggs = 0;
+  Dim *join_dim = nullptr;          /* group-by-dim-attr key source */
+  bool join_group = false;
   /* device buffers */
   double *dev_out = nullptr;
   size_t out_stride = 0;                /* 2*NA_t+1 of the launched template */
@@ -733,6 +822,31 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       if (use_col(c) < 0) { fail(SN_ERR_BADARG, "too many plan columns"); return nullptr; }
     }
 
+  /* broadcast-dimension join */
+  Dim *jd = nullptr;
+  if (plan->join_dim != SN_JOIN_NONE) {
+    if (plan->join_dim < 0 || plan->join_dim >= (int32_t)e->dims.size()) {
+      fail(SN_ERR_BADARG, "unknown dimension %d", plan->join_dim); return nullptr;
+    }
+    jd = e->dims[plan->join_dim].get();
+    if (jd->keys.empty()) { fail(SN_ERR_BADARG, "empty dimension"); return nullptr; }
+    sn_type_t kt = t->schema[plan->join_fact_col].dtype;
+    if (kt != SN_TYPE_INT32 && kt != SN_TYPE_INT64) {
+      fail(SN_ERR_UNSUPPORTED, "join key must be int32/int64"); return nullptr;
+    }
+    if (plan->join_mode == SN_JOIN_GROUP && plan->ngroup > 0) {
+      fail(SN_ERR_UNSUPPORTED, "dim-attr grouping combined with fact group "
+           "columns is not in the round-1 path");
+      return nullptr;
+    }
+    if (use_col(plan->join_fact_col) < 0) {
+      fail(SN_ERR_BADARG, "too many plan columns"); return nullptr;
+    }
+    if (dim_device_table(e, jd) != SN_OK) return nullptr;
+    q->join_dim = jd;
+    q->join_group = plan->join_mode == SN_JOIN_GROUP;
+  }
+
   /* group slot space: global dict sizes, plus a null slot only when the
    * schema allows null keys (non-nullable key columns waste no slots —
    * keeps Q1's 3x2 keys in the register-friendly 8x8 kernel) */
@@ -756,6 +870,13 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
            q->nslots);
       return nullptr;
     }
+  } else if (q->join_group) {
+    q->nslots = (int)q->join_dim->attr_dict.size();
+    if (q->nslots > 16) {
+      fail(SN_ERR_UNSUPPORTED, "dim attr cardinality %d > round-1 slot limit 16",
+           q->nslots);
+      return nullptr;
+    }
   } else {
     q->nslots = 1;
   }
@@ -774,6 +895,13 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (t->schema[q->used_cols[ui]].dtype == SN_TYPE_INT64)
       dp.i64_mask |= 1u << ui;
   for (int i = 0; i < plan->ngroup; i++) dp.gcol[i] = q->cslot_of_col[plan->group_cols[i]];
+  if (jd) {
+    dp.jkeys = jd->dev_keys;
+    dp.jpayload = jd->dev_payload;
+    dp.jcap_log2 = jd->cap_log2;
+    dp.jcslot = q->cslot_of_col[plan->join_fact_col];
+    dp.jmode = plan->join_mode == SN_JOIN_GROUP ? 1 : 0;
+  }
   for (int i = 0; i < plan->npreds; i++) {
     const sn_pred &s = plan->preds[i];
     sn_type_t dt = t->schema[s.col].dtype;
@@ -801,7 +929,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     /* canonicalize each aggregate; in grouped mode dedupe identical
      * expressions and fold COUNT(*) into the per-slot rowcount, so e.g.
      * Q1's 8 logical aggregates run 5 device sweeps (sum/avg pairs share) */
-    const bool grouped = plan->ngroup > 0;
+    const bool grouped = plan->ngroup > 0 || q->join_group;
     int ndev = 0;
     for (int a = 0; a < plan->naggs; a++) {
       sn_dev_agg da;
@@ -1027,11 +1155,11 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
   for (int s = 0; s < q->nslots; s++) {
     const double *row = &q->host_out[(size_t)s * q->out_stride];
     double rowcount = row[2 * q->na_t];
-    if (p.ngroup > 0 && rowcount == 0.0) continue;
+    if ((p.ngroup > 0 || q->join_group) && rowcount == 0.0) continue;
     GroupOut g;
     g.rowcount = rowcount;
     for (int a = 0; a < p.naggs; a++) {
-      if (p.ngroup == 0) {
+      if (p.ngroup == 0 && !q->join_group) {
         g.sums[a] = row[a];
         g.counts[a] = row[q->na_t + a];
       } else {
@@ -1042,7 +1170,9 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
         g.counts[a] = rowcount;
       }
     }
-    if (p.ngroup >= 1) {
+    if (q->join_group) {
+      g.keys[0] = q->join_dim->attr_dict[s];
+    } else if (p.ngroup >= 1) {
       int c0 = p.group_cols[0];
       int g1 = s / q->g2cap;
       if (g1 == q->gnull1) g.key_null[0] = true;
@@ -1056,7 +1186,7 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
     }
     out->push_back(std::move(g));
   }
-  if (p.ngroup == 0 && out->empty()) out->push_back(GroupOut());
+  if (p.ngroup == 0 && !q->join_group && out->empty()) out->push_back(GroupOut());
 }
 
 static void finalize_groups(sn_query *q, std::vector<GroupOut> &groups) {
@@ -1082,14 +1212,15 @@ extern "C" int32_t sn_query_result(sn_query *q, sn_result *out) {
   }
   memset(out, 0, sizeof(*out));
   const sn_plan &p = q->plan;
-  out->ngroup = p.ngroup; out->naggs = p.naggs;
+  out->ngroup = q->join_group ? 1 : p.ngroup;
+  out->naggs = p.naggs;
   out->nrows = (int32_t)std::min((size_t)SN_MAX_GROUP_SLOTS, q->final_groups.size());
   out->rows_scanned = q->rows_scanned;
   out->batches_seen = q->batches_seen;
   out->batches_skipped = q->batches_skipped;
   for (int32_t i = 0; i < out->nrows; i++) {
     GroupOut &g = q->final_groups[i];
-    for (int k = 0; k < p.ngroup; k++) {
+    for (int k = 0; k < out->ngroup; k++) {
       strncpy(out->keys[i][k], g.keys[k].c_str(), SN_KEY_MAX - 1);
       out->key_is_null[i][k] = g.key_null[k] ? 1 : 0;
     }
@@ -1115,7 +1246,8 @@ extern "C" void sn_query_destroy(sn_query *q) { delete q; }
 /* ---- partial exchange ---- */
 extern "C" int64_t sn_query_partial_bytes(sn_query *q) {
   if (!q) return SN_ERR_BADARG;
-  if (q->plan.ngroup == 0) return (int64_t)(2 * q->plan.naggs + 1) * 8;
+  if (q->plan.ngroup == 0 && !q->join_group)
+    return (int64_t)(2 * q->plan.naggs + 1) * 8;
   return 8 + (int64_t)SN_MAX_GROUP_SLOTS * sizeof(PartialSlot);
 }
 
@@ -1125,7 +1257,7 @@ extern "C" int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_devi
   if (rc != SN_OK) return rc;
   const sn_plan &p = q->plan;
   std::vector<uint8_t> block((size_t)sn_query_partial_bytes(q), 0);
-  if (p.ngroup == 0) {
+  if (p.ngroup == 0 && !q->join_group) {
     double *o = (double *)block.data();
     const double *row = q->host_out.data();
     for (int a = 0; a < p.naggs; a++) { o[a] = row[a]; o[p.naggs + a] = row[q->na_t + a]; }
@@ -1162,7 +1294,7 @@ extern "C" int32_t sn_query_merge(sn_query *q, const void *blocks, int64_t strid
   if (!q || !blocks || n_blocks <= 0) return SN_ERR_BADARG;
   const sn_plan &p = q->plan;
   std::vector<GroupOut> merged;
-  if (p.ngroup == 0) {
+  if (p.ngroup == 0 && !q->join_group) {
     GroupOut g;
     for (int32_t bi = 0; bi < n_blocks; bi++) {
       const double *o = (const double *)((const uint8_t *)blocks + bi * stride);
@@ -1176,15 +1308,16 @@ extern "C" int32_t sn_query_merge(sn_query *q, const void *blocks, int64_t strid
       const uint8_t *bp = (const uint8_t *)blocks + bi * stride;
       int32_t n; memcpy(&n, bp, 4);
       const PartialSlot *slots = (const PartialSlot *)(bp + 8);
+      int eff_ngroup = q->join_group ? 1 : p.ngroup;
       for (int32_t i = 0; i < n; i++) {
         std::string key;
-        for (int k = 0; k < p.ngroup; k++) {
+        for (int k = 0; k < eff_ngroup; k++) {
           key += slots[i].key_null[k] ? std::string(1, '\x01')
                                       : std::string(slots[i].keys[k]);
           key += '\x00';
         }
         GroupOut &g = bykey[key];
-        for (int k = 0; k < p.ngroup; k++) {
+        for (int k = 0; k < eff_ngroup; k++) {
           g.keys[k] = slots[i].keys[k];
           g.key_null[k] = slots[i].key_null[k] != 0;
         }

@@ -82,6 +82,12 @@ typedef struct {
   int32_t nslots, nused;
   uint32_t i64_mask;           /* bit c: cslot c is INT64 (raw bits in LDS) */
   int32_t gcol[2];
+  /* broadcast-dimension probe (empty-slot sentinel key = INT64_MIN) */
+  const int64_t *jkeys;        /* open-address key array (NULL = no join) */
+  const int32_t *jpayload;     /* per-slot payload: dim attr gid (or 0) */
+  int32_t jcap_log2;           /* table capacity = 1 << jcap_log2 */
+  int32_t jcslot;              /* fact key column slot */
+  int32_t jmode;               /* 0 semi, 1 group-by-dim-attr */
   int32_t _pad;
   sn_dev_pred_d preds_d[8];
   sn_dev_pred_i preds_i[4];

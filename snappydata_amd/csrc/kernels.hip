@@ -502,6 +502,46 @@ __device__ __forceinline__ void pred_sweeps(const sn_dev_plan *P, int npd,
   }
 }
 
+/* broadcast-dimension probe sweep (HashJoinExec probe, device-side):
+ * open-address linear-probe table in HBM (L2/L3-resident by the 100 MB
+ * HashJoinSize design bound); filters alive and, for group-by-dim-attr
+ * joins, assigns the group slot from the dimension payload. */
+__device__ __forceinline__ unsigned long long mix64(unsigned long long x) {
+  x += 0x9E3779B97f4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
+
+__device__ __forceinline__ void probe_sweep(const sn_dev_plan *P,
+                                            const double *sval,
+                                            uint64_t *salive,
+                                            int16_t *sslot) {
+  const int tid = threadIdx.x;
+  const GAS long long *jk = (const GAS long long *)(uintptr_t)P->jkeys;
+  const GAS int32_t *jp = (const GAS int32_t *)(uintptr_t)P->jpayload;
+  const int cs = P->jcslot;
+  const unsigned mask = (1u << P->jcap_log2) - 1;
+  const int is_i64 = (P->i64_mask >> cs) & 1u;
+#pragma unroll 2
+  for (int k = 0; k < CHUNK / WG; k++) {
+    const int r = tid + k * WG;
+    const double xv = sval[(size_t)cs * CHUNK + r];
+    const long long key = is_i64 ? __double_as_longlong(xv) : (long long)xv;
+    unsigned h = (unsigned)mix64((unsigned long long)key) & mask;
+    int pay = -1;
+    while (true) {
+      long long k0 = jk[h];
+      if (k0 == key) { pay = jp[h]; break; }
+      if (k0 == LLONG_MIN) break;
+      h = (h + 1) & mask;
+    }
+    const uint64_t w = __ballot(pay >= 0);
+    if ((tid & 63) == 0) salive[r >> 6] &= w;
+    if (sslot) sslot[r] = (int16_t)(pay > 0 ? pay : 0);
+  }
+}
+
 /* ================= keyless kernel ================= */
 template <int NAGGS, int NC>
 __launch_bounds__(WG, NAGGS <= 4 ? 4 : 2)
@@ -563,6 +603,7 @@ __global__ void k_keyless(sn_dev_plan plan,
        * hoisted per pass (row-invariant LDS reads once, not per row) ---- */
       alive_init(salive, sdead, rows, clean);
       pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
+      if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
 #pragma unroll
       for (int a = 0; a < NAGGS; a++) {
         if (a >= naggs) break;
@@ -685,7 +726,12 @@ __global__ void k_grouped(sn_dev_plan plan,
        * no barrier needed before this block's own later passes) */
       alive_init(salive, sdead, rows, clean);
       pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
-      {
+      if (plan.jkeys) {
+        /* join: filter by probe; group-by-dim-attr takes its slot from the
+         * dimension payload */
+        probe_sweep(P, sval, salive, plan.jmode == 1 ? sslot : nullptr);
+      }
+      if (!(plan.jkeys && plan.jmode == 1)) {
         const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
 #pragma unroll
         for (int k = 0; k < CHUNK / WG; k++) {

@@ -94,6 +94,12 @@ def lib():
         L.sn_datagen_lineitem.restype = C.c_int64
         L.sn_datagen_lineitem.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
                                           C.c_int64, C.c_int32, C.c_int32]
+        L.sn_dim_define.restype = C.c_int32
+        L.sn_dim_define.argtypes = [C.c_void_p, C.c_char_p]
+        L.sn_dim_put.restype = C.c_int32
+        L.sn_dim_put.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
+                                 C.POINTER(C.c_int64), C.c_char_p,
+                                 C.POINTER(C.c_int32)]
         L.sn_encode_column.restype = C.c_int64
         L.sn_encode_column.argtypes = [C.c_int32, C.c_void_p, C.POINTER(C.c_int32),
                                        C.POINTER(C.c_uint8), C.c_int32,
@@ -329,6 +335,23 @@ class Engine:
     def datagen_lineitem(self, table, total_rows, seed=42, batch_rows=0, threads=0):
         return _check(lib().sn_datagen_lineitem(self._h, table, total_rows, seed,
                                                 batch_rows, threads), "datagen")
+
+    def dim_define(self, name):
+        return _check(lib().sn_dim_define(self._h, name.encode()), "dim_define")
+
+    def dim_put(self, dim, keys, attrs=None):
+        """Load dimension rows (the row-store stand-in): keys int64, attrs
+        optional list of bytes per key (for group-by-dim-attr joins)."""
+        keys = np.ascontiguousarray(keys, dtype=np.int64)
+        payload, lens = None, None
+        if attrs is not None:
+            lens = np.array([len(a) for a in attrs], dtype=np.int32)
+            payload = b"".join(attrs)
+        _check(lib().sn_dim_put(
+            self._h, dim, len(keys),
+            keys.ctypes.data_as(C.POINTER(C.c_int64)), payload,
+            lens.ctypes.data_as(C.POINTER(C.c_int32)) if lens is not None else None),
+            "dim_put")
 
     def query(self, plan):
         h = lib().sn_query_submit(self._h, C.byref(plan))
