@@ -82,7 +82,33 @@ WORKLOADS = {
     # BASELINE config 5: Q6 scan with ~2% of rows update-patched and ~1%
     # deleted (ColumnDeltaDecoder/delete-mask merge during the scan)
     "tpch_q6_mut_sf10": (q6_plan, 60_000_000, 28),
+    # BASELINE config 4: star schema — column fact (suppkey, extendedprice)
+    # joined to a broadcast row-store dimension, group by dim attribute
+    "star_join_sf10": (None, 60_000_000, 12),
 }
+
+
+def build_star_join(eng, t, total_rows, seed, batch_rows=2_000_000):
+    """Fact (suppkey int32, extendedprice f64) + dimension covering 40% of a
+    100K keyspace with 8 nation attributes (dimension << HashJoinSize)."""
+    rng = np.random.default_rng(seed)
+    keyspace = 100_000
+    for start in range(0, total_rows, batch_rows):
+        n = min(batch_rows, total_rows - start)
+        keys = rng.integers(0, keyspace, n).astype(np.int32)
+        ep = rng.random(n) * 1e5
+        eng.ingest_columns(t, [{"data": keys}, {"data": ep}], n,
+                           batch_rows=batch_rows,
+                           first_bucket=start // batch_rows)
+    dim = eng.dim_define("supplier")
+    dk = np.sort(rng.choice(keyspace, size=keyspace * 2 // 5,
+                            replace=False)).astype(np.int64)
+    attrs = [b"NATION_%d" % (int(k) % 8) for k in dk]
+    eng.dim_put(dim, dk, attrs)
+    plan = abi.make_plan(table=t,
+                         aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])],
+                         join=dict(dim=dim, fact_col=0, group=True))
+    return plan
 
 
 def build_mutable_lineitem(eng, t, total_rows, seed, batch_rows=600_000):
@@ -189,14 +215,18 @@ def main():
 
     eng = se.Engine(device=local_rank, shard_rank=rank, shard_count=world,
                     n_buckets=max(128, world * 16))
-    t = eng.table_define("lineitem", LINEITEM_SCHEMA)
-    if args.workload == "tpch_q6_mut_sf10":
-        build_mutable_lineitem(eng, t, total_rows, args.seed)
+    if args.workload == "star_join_sf10":
+        t = eng.table_define("fact", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+        plan = build_star_join(eng, t, total_rows, args.seed)
     else:
-        eng.datagen_lineitem(t, total_rows, seed=args.seed, batch_rows=600_000)
+        t = eng.table_define("lineitem", LINEITEM_SCHEMA)
+        if args.workload == "tpch_q6_mut_sf10":
+            build_mutable_lineitem(eng, t, total_rows, args.seed)
+        else:
+            eng.datagen_lineitem(t, total_rows, seed=args.seed, batch_rows=600_000)
+        plan = plan_fn(t)
     resident = eng.num_rows(t)
-    plan = plan_fn(t)
-    grouped = plan.ngroup > 0
+    grouped = plan.ngroup > 0 or plan.join_mode == 1 and plan.join_dim >= 0
 
     exchange_buf = None
 
@@ -280,7 +310,8 @@ def main():
                 "kernel_ms": round(avg_ms, 4),
             }
         cpu_baseline = None
-        if world == 1 and not args.no_cpu_baseline:
+        if world == 1 and not args.no_cpu_baseline and "lineitem" in args.workload \
+                or world == 1 and not args.no_cpu_baseline and "mut" in args.workload:
             cpu_baseline = cpu_baseline_leg(args.workload, args.seed)
 
         line = {
