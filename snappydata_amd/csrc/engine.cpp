@@ -154,10 +154,22 @@ struct Batch {
   std::vector<std::vector<uint8_t>> host_blobs;
 };
 
+/* cached device descriptor set for one plan shape (the reference caches
+ * generated plans per query signature, SnappySession plan cache) */
+struct DescCache {
+  uint64_t sig = 0;
+  size_t batch_count = 0;       /* invalidation: table grew */
+  const void *db_dev = nullptr;
+  const void *tl_dev = nullptr;
+  int32_t ntiles = 0;
+  int64_t rows = 0;
+};
+
 struct Table {
   std::string name;
   std::vector<sn_col_schema> schema;
   std::vector<Batch> batches;
+  std::vector<DescCache> desc_caches;
   /* global dictionary per column (string dict cols) */
   std::vector<std::vector<std::string>> gdict;
   std::vector<std::map<std::string, int32_t>> gdict_idx;
@@ -976,14 +988,38 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   q->na_t = template_naggs(q->dev_naggs > 0 ? q->dev_naggs : 1);
   q->out_stride = 2 * (size_t)q->na_t + 1;
 
-  /* batch descriptors + tile map (stats skip applied here) */
+  /* batch descriptors + tile map.  Expensive at many batches (dict-map
+   * premultiply + uploads), so cache per plan signature; a cached set is
+   * only reusable when stats-skip removes nothing for THIS plan (skipping
+   * is an optimization — correctness is unaffected either way). */
+  std::lock_guard<std::mutex> g(t->mu);
+  uint64_t sig = 1469598103934665603ull;
+  auto mix = [&](uint64_t v) { sig ^= v; sig *= 1099511628211ull; };
+  for (int32_t c : q->used_cols) mix((uint64_t)c + 1);
+  mix((uint64_t)plan->ngroup << 8);
+  for (int i = 0; i < plan->ngroup; i++) mix((uint64_t)plan->group_cols[i] + 17);
+  mix((uint64_t)(q->g1cap * 131 + q->g2cap));
+  mix((uint64_t)(plan->join_dim + 3) * 29 + (uint64_t)plan->join_mode);
+  mix((uint64_t)t->batches.size());
+  for (auto &c : t->gdict) mix(c.size() * 7919);
+
+  int64_t skip_count = 0;
+  for (auto &b : t->batches)
+    if (batch_skippable(b, plan, t)) skip_count++;
+  q->batches_seen = (int64_t)t->batches.size();
+  q->batches_skipped = skip_count;
+
+  DescCache *hit = nullptr;
+  if (skip_count == 0) {
+    for (auto &dc : t->desc_caches)
+      if (dc.sig == sig && dc.batch_count == t->batches.size()) { hit = &dc; break; }
+  }
+
   std::vector<sn_dev_batch> hbatches;
   std::vector<sn_dev_tile> htiles;
-  std::vector<const void *> tmp_keepalive;
-  std::lock_guard<std::mutex> g(t->mu);
   for (auto &b : t->batches) {
-    q->batches_seen++;
-    if (batch_skippable(b, plan, t)) { q->batches_skipped++; continue; }
+    if (hit) break;
+    if (skip_count && batch_skippable(b, plan, t)) continue;
     sn_dev_batch db;
     memset(&db, 0, sizeof(db));
     db.num_rows = b.num_rows;
@@ -1069,7 +1105,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     db.clean = clean ? 1 : 0;
     int32_t bi = (int32_t)hbatches.size();
     hbatches.push_back(db);
-    q->rows_scanned += b.num_rows;  /* refined below for deletes */
+    q->rows_scanned += b.num_rows;
     for (int32_t r = 0; r < b.num_rows; r += SN_TILE_ROWS)
       htiles.push_back({ bi, r });
   }
@@ -1081,17 +1117,35 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   if (hipMemsetAsync(q->dev_out, 0, out_n * 8, e->stream) != hipSuccess) {
     fail(SN_ERR_GENERIC, "memset out"); return nullptr;
   }
-  if (!htiles.empty()) {
-    void *db_dev = e->arena.alloc(hbatches.size() * sizeof(sn_dev_batch));
-    void *tl_dev = e->arena.alloc(htiles.size() * sizeof(sn_dev_tile));
-    if (!db_dev || !tl_dev) { fail(SN_ERR_NOMEM, "desc alloc"); return nullptr; }
+  const void *db_dev = nullptr, *tl_dev = nullptr;
+  int32_t ntiles = 0;
+  if (hit) {
+    db_dev = hit->db_dev; tl_dev = hit->tl_dev;
+    ntiles = hit->ntiles;
+    q->rows_scanned = hit->rows;
+  } else if (!htiles.empty()) {
+    void *dbp = e->arena.alloc(hbatches.size() * sizeof(sn_dev_batch));
+    void *tlp = e->arena.alloc(htiles.size() * sizeof(sn_dev_tile));
+    if (!dbp || !tlp) { fail(SN_ERR_NOMEM, "desc alloc"); return nullptr; }
     /* blocking copies: the host vectors are stack-local and die at return */
-    if (hipMemcpy(db_dev, hbatches.data(), hbatches.size() * sizeof(sn_dev_batch),
+    if (hipMemcpy(dbp, hbatches.data(), hbatches.size() * sizeof(sn_dev_batch),
                   hipMemcpyHostToDevice) != hipSuccess ||
-        hipMemcpy(tl_dev, htiles.data(), htiles.size() * sizeof(sn_dev_tile),
+        hipMemcpy(tlp, htiles.data(), htiles.size() * sizeof(sn_dev_tile),
                   hipMemcpyHostToDevice) != hipSuccess) {
       fail(SN_ERR_GENERIC, "desc upload"); return nullptr;
     }
+    db_dev = dbp; tl_dev = tlp;
+    ntiles = (int32_t)htiles.size();
+    if (skip_count == 0) {
+      if (t->desc_caches.size() > 16) t->desc_caches.clear();
+      DescCache dc;
+      dc.sig = sig; dc.batch_count = t->batches.size();
+      dc.db_dev = db_dev; dc.tl_dev = tl_dev; dc.ntiles = ntiles;
+      dc.rows = q->rows_scanned;
+      t->desc_caches.push_back(dc);
+    }
+  }
+  if (ntiles > 0) {
     /* HIP events bracket the scan kernel on ITS stream for the roofline leg
      * (torch.cuda.Event would only see torch's current stream) */
     void *dp_dev = e->arena.alloc(sizeof(dp));
@@ -1100,7 +1154,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       fail(SN_ERR_NOMEM, "plan upload"); return nullptr;
     }
     /* block-partial scratch rows */
-    int grid = (int)htiles.size() < SN_GRID_CAP ? (int)htiles.size() : SN_GRID_CAP;
+    int grid = ntiles < SN_GRID_CAP ? ntiles : SN_GRID_CAP;
     size_t nv = dp.nslots <= 1 ? (size_t)(2 * q->na_t + 1)
                                : (size_t)dp.nslots * (q->dev_naggs + 1);
     size_t need = (size_t)grid * nv * 8;
@@ -1114,7 +1168,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
     int rc = sn_launch_scan_agg(&dp, (const sn_dev_plan *)dp_dev,
                                 (const sn_dev_batch *)db_dev,
-                                (const sn_dev_tile *)tl_dev, (int32_t)htiles.size(),
+                                (const sn_dev_tile *)tl_dev, ntiles,
                                 q->dev_out, e->scratch, e->stream);
     if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);
     if (rc != 0) {
