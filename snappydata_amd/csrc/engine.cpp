@@ -951,6 +951,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       da.nf = 0;
       const sn_agg &sa = plan->aggs[a];
       if (sa.kind != SN_AGG_COUNT_STAR) {
+        /* neutral factors get c = c0 below so their LDS reads CSE away */
         da.nf = sa.nfactors;
         if (sa.nfactors >= 1) {
           da.c0 = q->cslot_of_col[sa.factors[0].col];
@@ -968,6 +969,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
           da.c2 = q->cslot_of_col[sa.factors[2].col];
           da.a2 = sa.factors[2].add; da.m2 = sa.factors[2].mul;
         }
+        if (sa.nfactors < 2) da.c1 = da.c0;
+        if (sa.nfactors < 3) da.c2 = sa.nfactors >= 2 ? da.c1 : da.c0;
       }
       if (grouped && sa.kind == SN_AGG_COUNT_STAR) {
         q->agg_map[a] = -1;

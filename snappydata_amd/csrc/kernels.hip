@@ -822,6 +822,119 @@ __global__ void k_grouped(sn_dev_plan plan,
   (void)wid; (void)out_stride;
 }
 
+/* ---- register-accumulator grouped kernel (small shapes: NSLOTS x NA
+ * accumulators live across the WHOLE kernel) ----
+ * The sweep kernel's per-chunk wave reductions made it LDS-latency-bound
+ * (measured: 61% of wave cycles parked, LDS array 40% busy, HBM 12%).
+ * Here each lane accumulates slot-predicated sums in registers through all
+ * tiles and reduces ONCE at kernel end, one LDS read per referenced value
+ * per row.  Dict-group plans with nslots <= NSLOTS and deduped aggs <= NA;
+ * join-group and larger shapes use k_grouped. */
+template <int NSLOTS, int NA, int NC>
+__launch_bounds__(WG, 2)
+__global__ void k_grouped_reg(sn_dev_plan plan,
+                              const sn_dev_plan *__restrict__ plan_g,
+                              const sn_dev_batch *__restrict__ batches,
+                              const sn_dev_tile *__restrict__ tiles, int ntiles,
+                              double *__restrict__ out, int out_stride) {
+  const int tid = threadIdx.x;
+  const int nused = plan.nused;
+  const int naggs = plan.naggs, ngroup = plan.ngroup;
+  const int npd = plan.npreds_d, npi = plan.npreds_i;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double *sval = (double *)smem;
+  uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
+  uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+  uint64_t *salive = sdead + CHUNK / 64;
+  sn_dev_plan *P = (sn_dev_plan *)(salive + CHUNK / 64 + 2);
+  double *bacc = (double *)(P + 1);
+  {
+    const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
+    unsigned *dst = (unsigned *)P;
+    for (unsigned i = tid; i < sizeof(sn_dev_plan) / 4; i += WG) dst[i] = src[i];
+  }
+  __syncthreads();
+
+  const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
+
+  double sums[NSLOTS][NA];
+  double rc[NSLOTS];
+#pragma unroll
+  for (int s = 0; s < NSLOTS; s++) {
+    rc[s] = 0.0;
+#pragma unroll
+    for (int a = 0; a < NA; a++) sums[s][a] = 0.0;
+  }
+
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const sn_dev_tile tile = tiles[t];
+    const sn_dev_batch &b = batches[tile.batch];
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
+    const int clean = b.clean;
+
+    for (int base = tile.row_start; base < tile_end; base += CHUNK) {
+      const int rows = min(CHUNK, tile_end - base);
+      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      __syncthreads();
+
+      alive_init(salive, sdead, rows, clean);
+      pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
+      if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
+
+#pragma unroll 2
+      for (int k = 0; k < CHUNK / WG; k++) {
+        const int r = tid + k * WG;
+        const uint64_t w = salive[r >> 6];
+        if (w == 0) continue;
+        const int m = (int)((w >> (tid & 63)) & 1ull);
+        int slot = 0;
+        if (ngroup >= 1) slot = (int)sval[(size_t)gc0 * CHUNK + r];
+        if (ngroup >= 2) slot += (int)sval[(size_t)gc1 * CHUNK + r];
+        double va[NA];
+#pragma unroll
+        for (int a = 0; a < NA; a++) {
+          if (a >= naggs) { va[a] = 0.0; continue; }
+          const sn_dev_agg &A = P->aggs[a];
+          va[a] = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
+                  (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
+                  (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+        }
+#pragma unroll
+        for (int s = 0; s < NSLOTS; s++) {
+          const int ms = m && slot == s;
+          rc[s] += ms ? 1.0 : 0.0;
+#pragma unroll
+          for (int a = 0; a < NA; a++)
+            sums[s][a] += ms ? va[a] : 0.0;
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  /* final block reduction into a scratch row (like the keyless kernel) */
+  const int NV = plan.nslots * (naggs + 1);
+  for (int i = tid; i < NV; i += WG) bacc[i] = 0.0;
+  __syncthreads();
+#pragma unroll
+  for (int s = 0; s < NSLOTS; s++) {
+    if (s >= plan.nslots) break;
+#pragma unroll
+    for (int a = 0; a < NA + 1; a++) {
+      if (a > naggs) break;
+      double x = wave_sum(a < naggs && a < NA ? sums[s][a] : rc[s]);
+      if ((tid & 63) == 0 && x != 0.0)
+        atomicAdd(&bacc[s * (naggs + 1) + (a < naggs ? a : naggs)], x);
+    }
+  }
+  __syncthreads();
+  for (int i = tid; i < NV; i += WG)
+    out[(size_t)blockIdx.x * NV + i] = bacc[i];
+  (void)out_stride;
+}
+
 /* fold per-block partial rows into the final output.
  * keyless: final[i] = sum_b scratch[b][i]  (NV = 2*NA_t+1, identical layout)
  * grouped: scratch rows are [slot][naggs+1]; final is [slot][out_stride]
@@ -881,6 +994,14 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
     if (na <= 2) { if (nc4) KL(2, 4); else KL(2, 8); }
     else if (na <= 4) { if (nc4) KL(4, 4); else KL(4, 8); }
     else { if (nc4) KL(12, 4); else KL(12, 8); }
+  } else if (plan->jmode != 1 && ns <= 8 && na <= 6) {
+    /* register-accumulator grouped kernel (Q1's shape) */
+    lds += (CHUNK / 64) * 8 + (size_t)8 * (na + 1) * 8 + 64;
+#define KGR(S, A, NCv) hipLaunchKernelGGL((k_grouped_reg<S, A, NCv>), dim3(grid), \
+        dim3(WG), lds, s, *plan, dev_plan, dev_batches, dev_tiles, ntiles, \
+        dev_scratch, out_stride)
+    if (nc4) KGR(8, 6, 4); else KGR(8, 6, 8);
+#undef KGR
   } else {
     /* bacc + plan-mirror offsets inside the kernel use the TEMPLATE slot
      * count — size the dynamic LDS with it, not the runtime ns */
