@@ -867,17 +867,30 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
     for (int a = 0; a < NA; a++) sums[s][a] = 0.0;
   }
 
+  Stage<NC> st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
     const int num_rows = b.num_rows;
     const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
     const int clean = b.clean;
+    ColRegs cr[NC];
+    hoist_cols(b, nused, cr);
+    const int pipe = batch_stageable(clean, nused, cr);
 
+    int staged = 0;
+    if (pipe && tile.row_start + CHUNK <= tile_end) {
+      stage_load(cr, nused, tile.row_start, st);
+      staged = 1;
+    }
     for (int base = tile.row_start; base < tile_end; base += CHUNK) {
       const int rows = min(CHUNK, tile_end - base);
-      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      if (staged) stage_write(cr, nused, st, sval);
+      else convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
       __syncthreads();
+      const int nbase = base + CHUNK;
+      const int next_staged = pipe && nbase + CHUNK <= tile_end;
+      if (next_staged) stage_load(cr, nused, nbase, st);
 
       alive_init(salive, sdead, rows, clean);
       pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
@@ -911,6 +924,7 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
         }
       }
       __syncthreads();
+      staged = next_staged;
     }
   }
 
