@@ -143,6 +143,10 @@ struct Batch {
   };
   std::vector<PatchDev> patch_dev;
   std::vector<Patch> patch_host;      /* kept for dict premultiply at query */
+  /* RLE aux (device): cumulative run ends + values widened to f64 */
+  std::vector<const int32_t *> rle_ends_dev;
+  std::vector<const double *> rle_vals_dev;
+  std::vector<int32_t> rle_n;
   std::vector<ColMeta> cols;
   /* stats (parsed) */
   bool stats_valid = false;
@@ -502,6 +506,9 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
   b.nullpfx_dev.resize(nc, nullptr);
   b.patch_dev.resize(nc);
   b.patch_host.resize(nc);
+  b.rle_ends_dev.resize(nc, nullptr);
+  b.rle_vals_dev.resize(nc, nullptr);
+  b.rle_n.resize(nc, 0);
 
   std::lock_guard<std::mutex> g(t->mu);
   for (int c = 0; c < nc; c++) {
@@ -543,6 +550,36 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
     if (e->arena.device < 0) {
       b.host_blobs.emplace_back(blob, blob + len);
     }
+    /* RLE aux: extract (cumulative end, value-as-f64) per run on host
+     * (the blob stores [value][int32 run-length] entries; the decoder
+     * accumulates — RunLengthEncoding.scala:99-172) */
+    if (b.cols[c].type_id == SN_ENC_RUNLENGTH) {
+      sn_type_t dt = t->schema[c].dtype;
+      int w = dt == SN_TYPE_INT16 ? 2 : dt == SN_TYPE_INT32 ? 4 :
+              dt == SN_TYPE_INT64 ? 8 : 0;
+      if (w == 0)
+        return fail(SN_ERR_UNSUPPORTED, "RLE col %d dtype %d on GPU path", c, dt);
+      std::vector<int32_t> ends;
+      std::vector<double> vals;
+      int64_t cur = b.cols[c].body_off;
+      int32_t acc = 0;
+      while (cur + w + 4 <= len) {
+        int64_t v = w == 2 ? rd_i16(blob + cur) : w == 4 ? rd_i32(blob + cur)
+                                                         : rd_i64(blob + cur);
+        acc += rd_i32(blob + cur + w);
+        ends.push_back(acc);
+        if (dt == SN_TYPE_INT64) {
+          double d; memcpy(&d, &v, 8); vals.push_back(d);  /* raw bits */
+        } else vals.push_back((double)v);
+        cur += w + 4;
+      }
+      if (!ends.empty()) {
+        b.rle_ends_dev[c] = (const int32_t *)up(e, ends.data(), ends.size() * 4);
+        b.rle_vals_dev[c] = (const double *)up(e, vals.data(), vals.size() * 8);
+        b.rle_n[c] = (int32_t)ends.size();
+      }
+    }
+
     /* null prefix aux */
     if (b.cols[c].num_null_words) {
       const uint8_t *nw = blob + b.cols[c].null_off;
@@ -1049,13 +1086,25 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
             case SN_TYPE_INT64: dc.kind = SN_K_I64; break;
             case SN_TYPE_FLOAT: dc.kind = SN_K_F32; break;
             case SN_TYPE_INT16: dc.kind = SN_K_I16; break;
-            case SN_TYPE_BOOL: dc.kind = SN_K_I16; /* byte body */
-              fail(SN_ERR_UNSUPPORTED, "bool col on GPU path not yet supported");
-              return nullptr;
+            case SN_TYPE_BOOL: dc.kind = SN_K_U8; break;
+            case SN_TYPE_INT8: dc.kind = SN_K_S8; break;
             default:
               fail(SN_ERR_UNSUPPORTED, "uncompressed %d on GPU path", (int)dt);
               return nullptr;
           }
+          break;
+        case SN_ENC_RUNLENGTH:
+          if (b.rle_n[c] <= 0 && b.num_rows > 0) {
+            fail(SN_ERR_UNSUPPORTED, "RLE col %d has no run aux", c);
+            return nullptr;
+          }
+          dc.kind = SN_K_RLE;
+          dc.rle_ends = b.rle_ends_dev[c];
+          dc.rle_vals = b.rle_vals_dev[c];
+          dc.rle_n = b.rle_n[c];
+          break;
+        case SN_ENC_BOOLEAN_BITSET:
+          dc.kind = SN_K_BOOLBIT;
           break;
         case SN_ENC_DICTIONARY:
         case SN_ENC_BIG_DICTIONARY: {

@@ -30,7 +30,10 @@ enum {
   SN_K_F64 = 0, SN_K_I32 = 1, SN_K_I64 = 2, SN_K_F32 = 3,
   SN_K_DICT16 = 4,   /* int16 dictionary index (group cols) */
   SN_K_DICT32 = 5,   /* int32 dictionary index (BigDictionary) */
-  SN_K_I16 = 6, SN_K_BOOLBIT = 7
+  SN_K_I16 = 6, SN_K_BOOLBIT = 7,
+  SN_K_U8 = 8,       /* uncompressed boolean byte body */
+  SN_K_S8 = 9,       /* uncompressed int8 */
+  SN_K_RLE = 10      /* run-length: host-built run-ends + values aux */
 };
 
 typedef struct {
@@ -44,6 +47,13 @@ typedef struct {
   const int32_t  *patch_pos;   /* sorted patched rows */
   const double   *patch_val;   /* patched value (f64 or int bits in .i64) */
   const uint64_t *patch_nullbm;/* bitmap over patch index: patch writes NULL */
+  /* RLE aux (host-extracted): cumulative run END positions (exclusive) and
+   * run values widened to f64 (i64 raw-bitcast); kernel binary-searches the
+   * run for each nonNullPosition (RunLengthEncoding.scala decodes
+   * sequentially — no parallel analogue) */
+  const int32_t *rle_ends;
+  const double  *rle_vals;
+  int32_t rle_n;
   int32_t patch_n;
   int32_t kind;
   int32_t has_nulls;

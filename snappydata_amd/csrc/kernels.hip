@@ -68,6 +68,18 @@ __device__ __forceinline__ int patch_find(const int32_t *pos_, int n, int row) {
   return -1;
 }
 
+/* RLE: value of nonNullPosition nnp = vals[first run whose end > nnp] */
+__device__ __forceinline__ double rle_value(const sn_dev_col &c, int nnp) {
+  const GAS int32_t *ends = (const GAS int32_t *)(uintptr_t)c.rle_ends;
+  const GAS double *vals = (const GAS double *)(uintptr_t)c.rle_vals;
+  int lo = 0, hi = c.rle_n - 1;
+  while (lo < hi) {
+    int mid = (lo + hi) >> 1;
+    if (ends[mid] > nnp) hi = mid; else lo = mid + 1;
+  }
+  return vals[lo];
+}
+
 /* general-path read (nulls/patches/any kind); returns 0 if NULL */
 __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
                                             double *vd, long long *vi, int *gid) {
@@ -99,6 +111,15 @@ __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
     case SN_K_DICT16: *gid = as_global(c.dictmap)[(int)(uint16_t)as_global((const int16_t *)c.body)[nnp]]; break;
     case SN_K_DICT32: *gid = as_global(c.dictmap)[as_global((const int32_t *)c.body)[nnp]]; break;
     case SN_K_BOOLBIT: *vi = bm_get((const uint64_t *)c.body, nnp); *vd = (double)*vi; break;
+    case SN_K_U8: *vi = as_global((const uint8_t *)c.body)[nnp] == 1; *vd = (double)*vi; break;
+    case SN_K_S8: *vi = as_global((const int8_t *)c.body)[nnp]; *vd = (double)*vi; break;
+    case SN_K_RLE: {
+      *vd = rle_value(c, nnp);
+      const int is_i64 = (c.kind == SN_K_RLE);  /* vals carry raw i64 bits for i64 */
+      (void)is_i64;
+      *vi = __double2ll_rn(*vd);
+      break;
+    }
   }
   return 1;
 }
@@ -205,6 +226,32 @@ __device__ __forceinline__ void convert_chunk(
           for (int k = 0; k < CHUNK / WG; k++) {
             int r = tid + k * WG;
             if (r < rows) dst[r] = (double)bm_get(src, base + r);
+          }
+          break;
+        }
+        case SN_K_U8: {
+          const GAS uint8_t *src = as_global((const uint8_t *)col.body) + base;
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = (double)(src[r] == 1);
+          }
+          break;
+        }
+        case SN_K_S8: {
+          const GAS int8_t *src = as_global((const int8_t *)col.body) + base;
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = (double)src[r];
+          }
+          break;
+        }
+        case SN_K_RLE: {
+#pragma unroll
+          for (int k = 0; k < CHUNK / WG; k++) {
+            int r = tid + k * WG;
+            if (r < rows) dst[r] = rle_value(col, base + r);
           }
           break;
         }

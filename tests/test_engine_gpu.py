@@ -283,3 +283,49 @@ def test_q6_with_deltas_workload(eng):
     ot.add_batch(-n, cols, delete_mask=dmask, deltas=deltas)
     orows = po.result_rows(ot.query(tu.q6_plan()))
     assert_close(grows[0][1][0], orows[0][1][0])
+
+
+def test_rle_and_bool_columns_on_gpu(eng):
+    """RunLength-encoded int columns (host-built run aux + device binary
+    search) and bool columns (byte body + boolean bitset) vs the oracle."""
+    n = 200_000
+    rng = np.random.default_rng(13)
+    rle_vals = np.repeat(rng.integers(0, 50, 4000),
+                         rng.integers(20, 80, 4000))[:n].astype(np.int32)
+    assert len(rle_vals) == n
+    rle64 = (rle_vals.astype(np.int64) * 3) << 33   # exercise raw-i64 path
+    bools = (rng.random(n) < 0.3).astype(np.uint8)
+    meas = rng.random(n)
+    cols = [po.encode(po.T_INT32, po.ENC_RLE, rle_vals),
+            po.encode(po.T_INT64, po.ENC_RLE, rle64),
+            po.encode(po.T_BOOL, po.ENC_BOOLBITSET, bools),
+            po.encode(po.T_BOOL, po.ENC_UNCOMPRESSED, bools),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, meas)]
+    schema = [(abi.T_INT32, False), (abi.T_INT64, False), (abi.T_BOOL, False),
+              (abi.T_BOOL, False), (abi.T_DOUBLE, False)]
+    t = eng.table_define("trle", schema)
+    for st in range(0, n, 70_000):
+        e_ = min(n, st + 70_000)
+        sub = [po.encode(po.T_INT32, po.ENC_RLE, rle_vals[st:e_]),
+               po.encode(po.T_INT64, po.ENC_RLE, rle64[st:e_]),
+               po.encode(po.T_BOOL, po.ENC_BOOLBITSET, bools[st:e_]),
+               po.encode(po.T_BOOL, po.ENC_UNCOMPRESSED, bools[st:e_]),
+               po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, meas[st:e_])]
+        eng.batch_put(t, st, st // 70_000, e_ - st, sub)
+    k = int(np.median(rle_vals))
+    i64cut = int(np.median(rle64))
+    plan_kw = dict(
+        preds=[dict(col=0, hi=k),
+               dict(col=1, lo=i64cut, lo_strict=True),
+               dict(col=2, lo=1), dict(col=3, lo=1)],   # bools == true
+        aggs=[("sum", [(4, 0.0, 1.0)]), ("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_INT32, po.T_INT64, po.T_BOOL, po.T_BOOL, po.T_DOUBLE])
+    ot.add_batch(n, cols)
+    orows = po.result_rows(ot.query(po.make_plan(**plan_kw)))
+    assert grows[0][1][2] == orows[0][1][2]           # COUNT bit-exact
+    assert grows[0][1][1] == orows[0][1][1]           # integer SUM bit-exact
+    assert_close(grows[0][1][0], orows[0][1][0])
+    # sanity vs numpy too
+    m = (rle_vals <= k) & (rle64 > i64cut) & (bools == 1)
+    assert grows[0][1][2] == float(m.sum())
