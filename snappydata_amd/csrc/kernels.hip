@@ -512,6 +512,37 @@ __device__ __forceinline__ void alive_init(uint64_t *salive,
   }
 }
 
+/* fused sweep for <= 3 double predicates (the common fast path): one
+ * traversal, params hoisted to ~20 registers, one alive-word RMW */
+__device__ __forceinline__ void pred_sweep_fused(const sn_dev_plan *P, int npd,
+                                                 int clean, const double *sval,
+                                                 const uint64_t *svalid,
+                                                 uint64_t *salive) {
+  const int tid = threadIdx.x;
+  double lo0 = -1e308, hi0 = 1e308, lo1 = lo0, hi1 = hi0, lo2 = lo0, hi2 = hi0;
+  int c0 = 0, c1 = 0, c2 = 0;
+  if (npd >= 1) { lo0 = P->preds_d[0].lo; hi0 = P->preds_d[0].hi; c0 = P->preds_d[0].cslot; }
+  if (npd >= 2) { lo1 = P->preds_d[1].lo; hi1 = P->preds_d[1].hi; c1 = P->preds_d[1].cslot; }
+  if (npd >= 3) { lo2 = P->preds_d[2].lo; hi2 = P->preds_d[2].hi; c2 = P->preds_d[2].cslot; }
+#pragma unroll 2
+  for (int k = 0; k < CHUNK / WG; k++) {
+    const int r = tid + k * WG;
+    const double x0 = sval[(size_t)c0 * CHUNK + r];
+    const double x1 = sval[(size_t)c1 * CHUNK + r];
+    const double x2 = sval[(size_t)c2 * CHUNK + r];
+    int ok = (x0 >= lo0 && x0 <= hi0) &&
+             (npd < 2 || (x1 >= lo1 && x1 <= hi1)) &&
+             (npd < 3 || (x2 >= lo2 && x2 <= hi2));
+    uint64_t w = __ballot(ok);
+    if (!clean) {
+      if (npd >= 1) w &= svalid[(size_t)c0 * (CHUNK / 64) + (r >> 6)];
+      if (npd >= 2) w &= svalid[(size_t)c1 * (CHUNK / 64) + (r >> 6)];
+      if (npd >= 3) w &= svalid[(size_t)c2 * (CHUNK / 64) + (r >> 6)];
+    }
+    if ((tid & 63) == 0) salive[r >> 6] &= w;
+  }
+}
+
 /* one sweep per predicate, params hoisted; wave-owned word updates */
 __device__ __forceinline__ void pred_sweeps(const sn_dev_plan *P, int npd,
                                             int npi, int clean,
@@ -519,6 +550,10 @@ __device__ __forceinline__ void pred_sweeps(const sn_dev_plan *P, int npd,
                                             const uint64_t *svalid,
                                             uint64_t *salive) {
   const int tid = threadIdx.x;
+  if (npd <= 3 && npi == 0) {
+    if (npd > 0) pred_sweep_fused(P, npd, clean, sval, svalid, salive);
+    return;
+  }
 #pragma unroll
   for (int i = 0; i < 8; i++) {
     if (i >= npd) break;
