@@ -680,6 +680,45 @@ __global__ void k_keyless(sn_dev_plan plan,
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
       if (next_staged) stage_load(cr, nused, nbase, st);
 
+      /* fully-fused single pass (clean chunks, <=3 double preds, no join):
+       * one traversal computes predicates, aggregates and counts — the
+       * minimal LDS-read structure (matches the membw probe's shape) */
+      if (clean && npd <= 3 && npi == 0 && !plan.jkeys) {
+        double lo0 = -1e308, hi0 = 1e308, lo1 = lo0, hi1 = hi0, lo2 = lo0, hi2 = hi0;
+        int c0 = 0, c1 = 0, c2 = 0;
+        if (npd >= 1) { lo0 = P->preds_d[0].lo; hi0 = P->preds_d[0].hi; c0 = P->preds_d[0].cslot; }
+        if (npd >= 2) { lo1 = P->preds_d[1].lo; hi1 = P->preds_d[1].hi; c1 = P->preds_d[1].cslot; }
+        if (npd >= 3) { lo2 = P->preds_d[2].lo; hi2 = P->preds_d[2].hi; c2 = P->preds_d[2].cslot; }
+#pragma unroll 2
+        for (int k = 0; k < CHUNK / WG; k++) {
+          const int r = tid + k * WG;
+          const int inr = r < rows;
+          const double x0 = sval[(size_t)c0 * CHUNK + r];
+          const double x1 = sval[(size_t)c1 * CHUNK + r];
+          const double x2 = sval[(size_t)c2 * CHUNK + r];
+          const int ok = inr &&
+              (npd < 1 || (x0 >= lo0 && x0 <= hi0)) &&
+              (npd < 2 || (x1 >= lo1 && x1 <= hi1)) &&
+              (npd < 3 || (x2 >= lo2 && x2 <= hi2));
+          if (__popcll(__ballot(ok)) == 0) continue;
+#pragma unroll
+          for (int a = 0; a < NAGGS; a++) {
+            if (a >= naggs) break;
+            const sn_dev_agg &A = P->aggs[a];
+            const double va =
+                (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
+                (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
+                (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+            sums[a] += ok ? va : 0.0;
+            cnts[a] += ok ? 1.0 : 0.0;
+          }
+          rcnt += ok ? 1.0 : 0.0;
+        }
+        __syncthreads();
+        staged = next_staged;
+        continue;
+      }
+
       /* ---- row phase as sweep passes: each wave owns its 64-row words of
        * the alive bitmap, so pred/agg passes need no barriers; plan params
        * hoisted per pass (row-invariant LDS reads once, not per row) ---- */
