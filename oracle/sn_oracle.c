@@ -68,6 +68,7 @@
 #include <string.h>
 #include <stdio.h>
 #include <math.h>
+#include <dlfcn.h>
 
 #include "../include/snappy_engine.h"
 
@@ -515,6 +516,37 @@ static uint8_t *dup_buf(const void *p, int64_t n) {
   return b;
 }
 
+/* LZ4 wrapper (CompressionUtils.scala:53-61): [-codecId][ulen][payload] */
+typedef int (*sno_lz4_fn)(const char *, char *, int, int);
+static sno_lz4_fn sno_get_lz4(void) {
+  static sno_lz4_fn fn = NULL;
+  static int tried = 0;
+  if (!tried) {
+    tried = 1;
+    void *h = dlopen("liblz4.so.1", RTLD_NOW);
+    if (!h) h = dlopen("liblz4.so", RTLD_NOW);
+    if (h) fn = (sno_lz4_fn)dlsym(h, "LZ4_decompress_safe");
+  }
+  return fn;
+}
+
+/* returns malloc'd decompressed buffer (caller frees) or NULL if plain */
+static uint8_t *sno_maybe_decompress(const uint8_t *blob, int64_t len,
+                                     int64_t *out_len, int *err) {
+  *err = 0;
+  if (len < 8 || rd_i32(blob) >= 0) return NULL;
+  int32_t codec = -rd_i32(blob);
+  int32_t ulen = rd_i32(blob + 4);
+  if (codec != 1 || ulen <= 0) { *err = SN_ERR_UNSUPPORTED; return NULL; }
+  sno_lz4_fn fn = sno_get_lz4();
+  if (!fn) { *err = SN_ERR_UNSUPPORTED; return NULL; }
+  uint8_t *buf = (uint8_t *)malloc((size_t)ulen);
+  int n = fn((const char *)blob + 8, (char *)buf, (int)(len - 8), ulen);
+  if (n != ulen) { free(buf); *err = SN_ERR_BADFORMAT; return NULL; }
+  *out_len = ulen;
+  return buf;
+}
+
 SNO_EXPORT int32_t sno_table_add_batch(sno_table *t, int32_t num_rows,
     const sn_buf *cols, const sn_buf *stats, const sn_buf *delete_mask,
     const sn_buf *deltas /* [ncols*2] or NULL */) {
@@ -532,8 +564,16 @@ SNO_EXPORT int32_t sno_table_add_batch(sno_table *t, int32_t num_rows,
   b->col_lens = (int64_t *)calloc((size_t)t->ncols, sizeof(int64_t));
   for (int i = 0; i < t->ncols; i++) {
     if (!cols[i].data || cols[i].len < 8) { return SN_ERR_BADARG; }
-    b->cols[i] = dup_buf(cols[i].data, cols[i].len);
-    b->col_lens[i] = cols[i].len;
+    int err = 0;
+    int64_t dlen = 0;
+    uint8_t *dec = sno_maybe_decompress((const uint8_t *)cols[i].data,
+                                        cols[i].len, &dlen, &err);
+    if (err) return err;
+    if (dec) { b->cols[i] = dec; b->col_lens[i] = dlen; }
+    else {
+      b->cols[i] = dup_buf(cols[i].data, cols[i].len);
+      b->col_lens[i] = cols[i].len;
+    }
   }
   if (stats && stats->data) { b->stats = dup_buf(stats->data, stats->len); b->stats_len = stats->len; }
   if (delete_mask && delete_mask->data) {

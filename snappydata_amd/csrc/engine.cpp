@@ -31,6 +31,7 @@
  * device fails with SN_ERR_NOGPU.
  */
 #include <hip/hip_runtime.h>
+#include <dlfcn.h>
 
 #include <algorithm>
 #include <atomic>
@@ -328,6 +329,47 @@ static Table *get_table(sn_engine *e, int32_t t) {
   return e->tables[t].get();
 }
 
+/* ---- LZ4 compression wrapper (CompressionUtils.scala:53-61,132-160):
+ * [int32 -codecId][int32 uncompressedLen][payload]; codec 1 = LZ4 (default),
+ * 2 = Snappy.  The engine decompresses on put, mirroring
+ * ColumnFormatValue.getValueRetain(DECOMPRESS)
+ * (ColumnFormatEntry.scala:267-330).  liblz4 loaded lazily via dlopen
+ * (runtime-only .so in this image). ---- */
+typedef int (*lz4_decomp_fn)(const char *, char *, int, int);
+static lz4_decomp_fn get_lz4(void) {
+  static lz4_decomp_fn fn = nullptr;
+  static bool tried = false;
+  if (!tried) {
+    tried = true;
+    void *h = dlopen("liblz4.so.1", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) h = dlopen("liblz4.so", RTLD_NOW | RTLD_GLOBAL);
+    if (h) fn = (lz4_decomp_fn)dlsym(h, "LZ4_decompress_safe");
+  }
+  return fn;
+}
+
+/* if blob is codec-wrapped, decompress into *out and return 1; 0 if plain;
+ * negative = error */
+static int maybe_decompress(const uint8_t *blob, int64_t len,
+                            std::vector<uint8_t> *out) {
+  if (len < 8) return 0;
+  int32_t tag = rd_i32(blob);
+  if (tag >= 0) return 0;                  /* plain typeId */
+  int32_t codec = -tag;
+  int32_t ulen = rd_i32(blob + 4);
+  if (ulen <= 0) return fail(SN_ERR_BADFORMAT, "bad compressed length");
+  if (codec == 1) {
+    lz4_decomp_fn fn = get_lz4();
+    if (!fn) return fail(SN_ERR_UNSUPPORTED, "liblz4 unavailable");
+    out->resize((size_t)ulen);
+    int n = fn((const char *)blob + 8, (char *)out->data(),
+               (int)(len - 8), ulen);
+    if (n != ulen) return fail(SN_ERR_BADFORMAT, "LZ4 decompress failed (%d)", n);
+    return 1;
+  }
+  return fail(SN_ERR_UNSUPPORTED, "codec %d (Snappy) not supported", codec);
+}
+
 /* ---- blob header parse (PRODUCT-side restatement of
  *      ColumnEncoding.scala:37-53,764-832; DictionaryEncoding.scala:85-160) */
 static int parse_blob(const uint8_t *blob, int64_t len, sn_type_t dtype,
@@ -511,9 +553,13 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
   b.rle_n.resize(nc, 0);
 
   std::lock_guard<std::mutex> g(t->mu);
+  std::vector<uint8_t> decomp;
   for (int c = 0; c < nc; c++) {
     const uint8_t *blob = (const uint8_t *)columns[c].data;
     int64_t len = columns[c].len;
+    int dec = maybe_decompress(blob, len, &decomp);
+    if (dec < 0) return dec;
+    if (dec == 1) { blob = decomp.data(); len = (int64_t)decomp.size(); }
     int rc = parse_blob(blob, len, t->schema[c].dtype, &b.cols[c]);
     if (rc != SN_OK) return fail(rc, "column %d blob parse failed", c);
     /* intern dictionary into table-global dict (string dict cols) */
