@@ -35,12 +35,6 @@ extern "C" int32_t sn_batch_put(sn_engine *, int32_t, int64_t, int32_t, int32_t,
 static inline void put_i32(std::vector<uint8_t> &b, int32_t v) {
   const uint8_t *p = (const uint8_t *)&v; b.insert(b.end(), p, p + 4);
 }
-static inline void put_i64(std::vector<uint8_t> &b, int64_t v) {
-  const uint8_t *p = (const uint8_t *)&v; b.insert(b.end(), p, p + 8);
-}
-static inline void put_f64(std::vector<uint8_t> &b, double v) {
-  const uint8_t *p = (const uint8_t *)&v; b.insert(b.end(), p, p + 8);
-}
 static inline void put_i16(std::vector<uint8_t> &b, int16_t v) {
   const uint8_t *p = (const uint8_t *)&v; b.insert(b.end(), p, p + 2);
 }
