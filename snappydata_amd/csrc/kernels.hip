@@ -1017,7 +1017,7 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
       pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
       if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
 
-#pragma unroll 2
+#pragma unroll 4
       for (int k = 0; k < CHUNK / WG; k++) {
         const int r = tid + k * WG;
         const uint64_t w = salive[r >> 6];
@@ -1106,7 +1106,15 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
                                   double *dev_out, double *dev_scratch,
                                   void *stream) {
   hipStream_t s = (hipStream_t)stream;
-  int grid = ntiles < SN_GRID_CAP ? (ntiles > 0 ? ntiles : 1) : SN_GRID_CAP;
+  /* balance: every block gets the same tile count (a ragged grid-stride
+   * leaves half the blocks with 2x work and CUs idle in the tail) */
+  int grid;
+  if (ntiles <= 0) grid = 1;
+  else if (ntiles <= SN_GRID_CAP) grid = ntiles;
+  else {
+    int rounds = (ntiles + SN_GRID_CAP - 1) / SN_GRID_CAP;
+    grid = (ntiles + rounds - 1) / rounds;
+  }
   const int ns = plan->nslots, na = plan->naggs;
   const int na_t = na <= 2 ? 2 : na <= 4 ? 4 : na <= 8 ? 8 : 12;
   size_t lds = (size_t)plan->nused * CHUNK * 8 +
