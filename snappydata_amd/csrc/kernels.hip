@@ -334,9 +334,10 @@ __device__ __forceinline__ int col_width_class(int kind) {
   }
 }
 
-template <int NC>
+template <int NC, int NB>
 __device__ __forceinline__ void stage_load(const ColRegs (&cr)[NC], int nused,
-                                           int base, Stage<NC> &st) {
+                                           int base, Stage<NB> &st) {
+  static_assert(NB >= NC || NB == 1, "stage buffer too small");
   const int tid = threadIdx.x;
 #pragma unroll
   for (int c = 0; c < NC; c++) {
@@ -347,27 +348,28 @@ __device__ __forceinline__ void stage_load(const ColRegs (&cr)[NC], int nused,
       const GAS double2_t *s2 = (const GAS double2_t *)
           as_global((const char *)col.body + (size_t)base * 8);
 #pragma unroll
-      for (int p = 0; p < CHUNK / (2 * WG); p++) st.buf[c][p] = s2[tid + p * WG];
+      for (int p = 0; p < CHUNK / (2 * WG); p++) st.buf[NB == 1 ? 0 : c][p] = s2[tid + p * WG];
     } else if (w == 4) {
       /* two int2 pair-loads -> raw bits in buf[c][0].x / .y */
       const GAS int2_t *s2 = (const GAS int2_t *)
           as_global((const char *)col.body + (size_t)base * 4);
       int2_t a0 = s2[tid], a1 = s2[tid + WG];
-      st.buf[c][0].x = *(double *)&a0;
-      st.buf[c][0].y = *(double *)&a1;
+      st.buf[NB == 1 ? 0 : c][0].x = *(double *)&a0;
+      st.buf[NB == 1 ? 0 : c][0].y = *(double *)&a1;
     } else {
       /* two short2 (4 B) pair-loads -> packed into buf[c][0].x */
       const GAS unsigned *s2 = (const GAS unsigned *)
           as_global((const char *)col.body + (size_t)base * 2);
       unsigned a0 = s2[tid], a1 = s2[tid + WG];
-      st.buf[c][0].x = pack_u64(a0, a1);
+      st.buf[NB == 1 ? 0 : c][0].x = pack_u64(a0, a1);
     }
   }
 }
 
-template <int NC>
+template <int NC, int NB>
 __device__ __forceinline__ void stage_write(const ColRegs (&cr)[NC], int nused,
-                                            Stage<NC> &st, double *sval) {
+                                            Stage<NB> &st, double *sval) {
+  static_assert(NB >= NC || NB == 1, "stage buffer too small");
   const int tid = threadIdx.x;
 #pragma unroll
   for (int c = 0; c < NC; c++) {
@@ -378,13 +380,13 @@ __device__ __forceinline__ void stage_write(const ColRegs (&cr)[NC], int nused,
       case SN_K_F64: case SN_K_I64:
 #pragma unroll
         for (int p = 0; p < CHUNK / (2 * WG); p++)
-          ((double2_t *)dst)[tid + p * WG] = st.buf[c][p];
+          ((double2_t *)dst)[tid + p * WG] = st.buf[NB == 1 ? 0 : c][p];
         break;
       case SN_K_I32: case SN_K_F32: case SN_K_DICT32: {
         const GAS int32_t *dm = as_global(col.dictmap);
 #pragma unroll
         for (int p = 0; p < CHUNK / (2 * WG); p++) {
-          double raw = p == 0 ? st.buf[c][0].x : st.buf[c][0].y;
+          double raw = p == 0 ? st.buf[NB == 1 ? 0 : c][0].x : st.buf[NB == 1 ? 0 : c][0].y;
           int2_t a = *(int2_t *)&raw;
           double2_t y;
           if (col.kind == SN_K_I32) { y.x = (double)a.x; y.y = (double)a.y; }
@@ -398,7 +400,7 @@ __device__ __forceinline__ void stage_write(const ColRegs (&cr)[NC], int nused,
       }
       case SN_K_I16: case SN_K_DICT16: {
         const GAS int32_t *dm = as_global(col.dictmap);
-        unsigned long long raw = __double_as_longlong(st.buf[c][0].x);
+        unsigned long long raw = __double_as_longlong(st.buf[NB == 1 ? 0 : c][0].x);
 #pragma unroll
         for (int p = 0; p < CHUNK / (2 * WG); p++) {
           unsigned half = (unsigned)(raw >> (32 * p));
@@ -653,8 +655,9 @@ __global__ void k_keyless(sn_dev_plan plan,
   for (int a = 0; a < NAGGS; a++) { sums[a] = 0.0; cnts[a] = 0.0; }
   const int naggs = plan.naggs;
   const int npd = plan.npreds_d, npi = plan.npreds_i;
+  constexpr int STAGED = 1;
 
-  Stage<NC> st;
+  Stage<STAGED ? NC : 1> st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
@@ -663,7 +666,7 @@ __global__ void k_keyless(sn_dev_plan plan,
     const int clean = b.clean;
     ColRegs cr[NC];
     hoist_cols(b, nused, cr);
-    const int pipe = batch_stageable(clean, nused, cr);
+    const int pipe = STAGED && batch_stageable(clean, nused, cr);
 
     int staged = 0;
     if (pipe && tile.row_start + CHUNK <= tile_end) {
@@ -818,7 +821,8 @@ __global__ void k_grouped(sn_dev_plan plan,
   for (int i = tid; i < NSLOTS * (naggs + 1); i += WG) bacc[i] = 0.0;
   __syncthreads();
 
-  Stage<NC> st;
+  constexpr int STAGED = 1;
+  Stage<STAGED ? NC : 1> st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
@@ -827,7 +831,7 @@ __global__ void k_grouped(sn_dev_plan plan,
     const int clean = b.clean;
     ColRegs cr[NC];
     hoist_cols(b, nused, cr);
-    const int pipe = batch_stageable(clean, nused, cr);
+    const int pipe = STAGED && batch_stageable(clean, nused, cr);
 
     int staged = 0;
     if (pipe && tile.row_start + CHUNK <= tile_end) {
@@ -951,8 +955,8 @@ __global__ void k_grouped(sn_dev_plan plan,
  * tiles and reduces ONCE at kernel end, one LDS read per referenced value
  * per row.  Dict-group plans with nslots <= NSLOTS and deduped aggs <= NA;
  * join-group and larger shapes use k_grouped. */
-template <int NSLOTS, int NA, int NC>
-__launch_bounds__(WG, 2)
+template <int NSLOTS, int NA, int NC, int STAGED = 1>
+__launch_bounds__(WG, STAGED ? 2 : 3)
 __global__ void k_grouped_reg(sn_dev_plan plan,
                               const sn_dev_plan *__restrict__ plan_g,
                               const sn_dev_batch *__restrict__ batches,
@@ -988,7 +992,7 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
     for (int a = 0; a < NA; a++) sums[s][a] = 0.0;
   }
 
-  Stage<NC> st;
+  Stage<STAGED ? NC : 1> st;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
     const sn_dev_batch &b = batches[tile.batch];
@@ -997,21 +1001,21 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
     const int clean = b.clean;
     ColRegs cr[NC];
     hoist_cols(b, nused, cr);
-    const int pipe = batch_stageable(clean, nused, cr);
+    const int pipe = STAGED && batch_stageable(clean, nused, cr);
 
     int staged = 0;
-    if (pipe && tile.row_start + CHUNK <= tile_end) {
-      stage_load(cr, nused, tile.row_start, st);
+    if (STAGED && pipe && tile.row_start + CHUNK <= tile_end) {
+      if constexpr (STAGED) stage_load(cr, nused, tile.row_start, st);
       staged = 1;
     }
     for (int base = tile.row_start; base < tile_end; base += CHUNK) {
       const int rows = min(CHUNK, tile_end - base);
-      if (staged) stage_write(cr, nused, st, sval);
+      if (STAGED && staged) { if constexpr (STAGED) stage_write(cr, nused, st, sval); }
       else convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
       __syncthreads();
       const int nbase = base + CHUNK;
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
-      if (next_staged) stage_load(cr, nused, nbase, st);
+      if (STAGED && next_staged) { if constexpr (STAGED) stage_load(cr, nused, nbase, st); }
 
       alive_init(salive, sdead, rows, clean);
       pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
@@ -1143,7 +1147,12 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
 #define KGR(S, A, NCv) hipLaunchKernelGGL((k_grouped_reg<S, A, NCv>), dim3(grid), \
         dim3(WG), lds, s, *plan, dev_plan, dev_batches, dev_tiles, ntiles, \
         dev_scratch, out_stride)
-    if (nc4) KGR(8, 6, 4); else KGR(8, 6, 8);
+    if (ns <= 6) {
+      /* unstaged variant runs at 3 waves/SIMD (occupancy > staging here) */
+      hipLaunchKernelGGL((k_grouped_reg<6, 6, 8, 0>), dim3(grid), dim3(WG), lds, s,
+                         *plan, dev_plan, dev_batches, dev_tiles, ntiles,
+                         dev_scratch, out_stride);
+    } else if (nc4) KGR(8, 6, 4); else KGR(8, 6, 8);
 #undef KGR
   } else {
     /* bacc + plan-mirror offsets inside the kernel use the TEMPLATE slot
