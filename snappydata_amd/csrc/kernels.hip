@@ -1147,12 +1147,9 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
 #define KGR(S, A, NCv) hipLaunchKernelGGL((k_grouped_reg<S, A, NCv>), dim3(grid), \
         dim3(WG), lds, s, *plan, dev_plan, dev_batches, dev_tiles, ntiles, \
         dev_scratch, out_stride)
-    if (ns <= 6) {
-      /* unstaged variant runs at 3 waves/SIMD (occupancy > staging here) */
-      hipLaunchKernelGGL((k_grouped_reg<6, 6, 8, 0>), dim3(grid), dim3(WG), lds, s,
-                         *plan, dev_plan, dev_batches, dev_tiles, ntiles,
-                         dev_scratch, out_stride);
-    } else if (nc4) KGR(8, 6, 4); else KGR(8, 6, 8);
+    /* (measured: an unstaged 3-waves/SIMD variant loses 20% — staging
+     * beats occupancy for the grouped shape as well) */
+    if (nc4) KGR(8, 6, 4); else KGR(8, 6, 8);
 #undef KGR
   } else {
     /* bacc + plan-mirror offsets inside the kernel use the TEMPLATE slot
