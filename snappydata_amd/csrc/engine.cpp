@@ -959,10 +959,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (q->g1cap == 0) q->g1cap = 1;
     if (q->g2cap == 0) q->g2cap = 1;
     q->nslots = q->g1cap * q->g2cap;
-    if (q->nslots > SN_MAX_GROUP_SLOTS ||
-        (q->nslots > 16)) {
-      fail(SN_ERR_UNSUPPORTED, "group cardinality %d > round-1 GPU slot limit 16",
-           q->nslots);
+    if (q->nslots > SN_MAX_GROUP_SLOTS) {
+      fail(SN_ERR_UNSUPPORTED, "group cardinality %d > %d (open-address "
+           "hash table for arbitrary cardinality is round-2)",
+           q->nslots, SN_MAX_GROUP_SLOTS);
       return nullptr;
     }
   } else if (q->join_group) {
@@ -1251,8 +1251,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         hipMemcpy(dp_dev, &dp, sizeof(dp), hipMemcpyHostToDevice) != hipSuccess) {
       fail(SN_ERR_NOMEM, "plan upload"); return nullptr;
     }
-    /* block-partial scratch rows */
+    /* block-partial scratch rows (the >16-slot path caps its grid) */
     int grid = ntiles < SN_GRID_CAP ? ntiles : SN_GRID_CAP;
+    if (dp.nslots > 16 && grid > SN_GRID_BIGSLOT) grid = SN_GRID_BIGSLOT;
     size_t nv = dp.nslots <= 1 ? (size_t)(2 * q->na_t + 1)
                                : (size_t)dp.nslots * (q->dev_naggs + 1);
     size_t need = (size_t)grid * nv * 8;

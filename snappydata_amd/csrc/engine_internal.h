@@ -105,6 +105,7 @@ typedef struct {
 } sn_dev_plan;
 
 #define SN_GRID_CAP 2048
+#define SN_GRID_BIGSLOT 256   /* grid cap for the >16-slot LDS hash-agg path */
 #define SN_TILE_ROWS 16384     /* rows per workgroup tile (16 LDS chunks;
                                   long pipelines for the staged conversion) */
 

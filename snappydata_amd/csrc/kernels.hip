@@ -947,6 +947,90 @@ __global__ void k_grouped(sn_dev_plan plan,
   (void)wid; (void)out_stride;
 }
 
+/* ---- LDS-accumulator grouped kernel (large slot counts: 17..1024) ----
+ * The ByteBufferHashMap/SHAMap analogue for dictionary group keys: the
+ * premultiplied global dictionary id IS the table slot (the reference's
+ * DictionaryOptimizedMapAccessor direct-array idea scaled up), and the
+ * per-block accumulator lives in LDS with f64 atomics — collisions are
+ * per-wave lanes hitting one slot, rare at high cardinality.  One LDS
+ * accumulator per block accumulates across all its tiles and flushes to a
+ * per-block scratch row at the end (grid is capped so scratch stays small).
+ * Open-address hashing of arbitrary (non-dictionary) keys is round-2. */
+__global__ __launch_bounds__(WG, 1)
+void k_grouped_lds(sn_dev_plan plan,
+                   const sn_dev_plan *__restrict__ plan_g,
+                   const sn_dev_batch *__restrict__ batches,
+                   const sn_dev_tile *__restrict__ tiles, int ntiles,
+                   double *__restrict__ out, int out_stride) {
+  const int tid = threadIdx.x;
+  const int nused = plan.nused;
+  const int naggs = plan.naggs, ngroup = plan.ngroup;
+  const int npd = plan.npreds_d, npi = plan.npreds_i;
+  const int nslots = plan.nslots;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double *sval = (double *)smem;
+  uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
+  uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+  uint64_t *salive = sdead + CHUNK / 64;
+  sn_dev_plan *P = (sn_dev_plan *)(salive + CHUNK / 64 + 2);
+  double *bacc = (double *)(P + 1);   /* [nslots][naggs+1] */
+  {
+    const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
+    unsigned *dst = (unsigned *)P;
+    for (unsigned i = tid; i < sizeof(sn_dev_plan) / 4; i += WG) dst[i] = src[i];
+  }
+  const int NV = nslots * (naggs + 1);
+  for (int i = tid; i < NV; i += WG) bacc[i] = 0.0;
+  __syncthreads();
+
+  const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const sn_dev_tile tile = tiles[t];
+    const sn_dev_batch &b = batches[tile.batch];
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
+    const int clean = b.clean;
+
+    for (int base = tile.row_start; base < tile_end; base += CHUNK) {
+      const int rows = min(CHUNK, tile_end - base);
+      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      __syncthreads();
+      alive_init(salive, sdead, rows, clean);
+      pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
+      if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
+
+#pragma unroll 2
+      for (int k = 0; k < CHUNK / WG; k++) {
+        const int r = tid + k * WG;
+        const uint64_t w = salive[r >> 6];
+        if (w == 0) continue;
+        const int m = (int)((w >> (tid & 63)) & 1ull);
+        if (!m) continue;
+        int slot = 0;
+        if (ngroup >= 1) slot = (int)sval[(size_t)gc0 * CHUNK + r];
+        if (ngroup >= 2) slot += (int)sval[(size_t)gc1 * CHUNK + r];
+        double *row_acc = bacc + (size_t)slot * (naggs + 1);
+        for (int a = 0; a < naggs; a++) {
+          const sn_dev_agg &A = P->aggs[a];
+          const double va =
+              (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
+              (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
+              (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+          atomicAdd(&row_acc[a], va);
+        }
+        atomicAdd(&row_acc[naggs], 1.0);
+      }
+      __syncthreads();
+    }
+  }
+
+  __syncthreads();
+  for (int i = tid; i < NV; i += WG)
+    out[(size_t)blockIdx.x * NV + i] = bacc[i];
+  (void)out_stride;
+}
+
 /* ---- register-accumulator grouped kernel (small shapes: NSLOTS x NA
  * accumulators live across the WHOLE kernel) ----
  * The sweep kernel's per-chunk wave reductions made it LDS-latency-bound
@@ -1141,6 +1225,15 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
     if (na <= 2) { if (nc4) KL(2, 4); else KL(2, 8); }
     else if (na <= 4) { if (nc4) KL(4, 4); else KL(4, 8); }
     else { if (nc4) KL(12, 4); else KL(12, 8); }
+  } else if (ns > 16) {
+    /* large-cardinality LDS hash-aggregate path: LDS accumulator bounds the
+     * grid so scratch rows stay small */
+    if (grid > SN_GRID_BIGSLOT) grid = SN_GRID_BIGSLOT;
+    lds += (CHUNK / 64) * 8 + (size_t)ns * (na + 1) * 8 + 64;
+    if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
+    hipLaunchKernelGGL(k_grouped_lds, dim3(grid), dim3(WG), lds, s,
+                       *plan, dev_plan, dev_batches, dev_tiles, ntiles,
+                       dev_scratch, out_stride);
   } else if (plan->jmode != 1 && ns <= 8 && na <= 6) {
     /* register-accumulator grouped kernel (Q1's shape) */
     lds += (CHUNK / 64) * 8 + (size_t)8 * (na + 1) * 8 + 64;

@@ -329,3 +329,39 @@ def test_rle_and_bool_columns_on_gpu(eng):
     # sanity vs numpy too
     m = (rle_vals <= k) & (rle64 > i64cut) & (bools == 1)
     assert grows[0][1][2] == float(m.sum())
+
+
+def test_large_group_cardinality_on_gpu(eng):
+    """>16 group slots (the LDS hash-aggregate path, SHAMap analogue for
+    dictionary keys): 40x25 = 1000 groups vs the oracle."""
+    n = 1_000_000
+    rng = np.random.default_rng(31)
+    k1 = [b"K%02d" % v for v in rng.integers(0, 40, n)]
+    k2 = [b"J%02d" % v for v in rng.integers(0, 25, n)]
+    meas = rng.random(n)
+    mask_col = rng.integers(0, 100, n).astype(np.int32)
+
+    def blobs(sl):
+        return [po.encode(po.T_STRING, po.ENC_DICT, k1[sl]),
+                po.encode(po.T_STRING, po.ENC_DICT, k2[sl]),
+                po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, meas[sl]),
+                po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, mask_col[sl])]
+
+    t = eng.table_define("tbig", [(abi.T_STRING, False), (abi.T_STRING, False),
+                                  (abi.T_DOUBLE, False), (abi.T_INT32, False)])
+    ot = po.OracleTable([po.T_STRING, po.T_STRING, po.T_DOUBLE, po.T_INT32])
+    for st in range(0, n, 250_000):
+        e_ = min(n, st + 250_000)
+        b = blobs(slice(st, e_))
+        eng.batch_put(t, st, st // 250_000, e_ - st, b)
+        ot.add_batch(e_ - st, b)
+    kw = dict(preds=[dict(col=3, hi=80, hi_strict=True)], group_cols=[0, 1],
+              aggs=[("sum", [(2, 0.0, 1.0)]), ("avg", [(2, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **kw)).rows()
+    orows = po.result_rows(ot.query(po.make_plan(**kw)))
+    assert len(grows) == len(orows) == 1000
+    for (gk, gv), (ok_, ov) in zip(grows, orows):
+        assert gk == ok_
+        assert gv[2] == ov[2]
+        for a in (0, 1):
+            assert abs(gv[a] - ov[a]) <= 1e-6 * max(1.0, abs(ov[a]))
