@@ -1057,7 +1057,7 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
   uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
   uint64_t *salive = sdead + CHUNK / 64;
   int16_t *sslot = (int16_t *)(salive + CHUNK / 64);
-  sn_dev_plan *P = (sn_dev_plan *)(sslot + CHUNK + 2);
+  sn_dev_plan *P = (sn_dev_plan *)(((uintptr_t)(sslot + CHUNK) + 15) & ~(uintptr_t)15);
   double *bacc = (double *)(P + 1);
   {
     const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
