@@ -1102,51 +1102,6 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
       if (STAGED && next_staged) { if constexpr (STAGED) stage_load(cr, nused, nbase, st); }
 
-      /* fused single pass (probe-F structure): predicates inline, no alive
-       * bitmap, one traversal computing slot + all aggregate values */
-      if (clean && npd <= 3 && npi == 0 && !plan.jkeys) {
-        double lo0 = -1e308, hi0 = 1e308, lo1 = lo0, hi1 = hi0, lo2 = lo0, hi2 = hi0;
-        int c0 = 0, c1 = 0, c2 = 0;
-        if (npd >= 1) { lo0 = P->preds_d[0].lo; hi0 = P->preds_d[0].hi; c0 = P->preds_d[0].cslot; }
-        if (npd >= 2) { lo1 = P->preds_d[1].lo; hi1 = P->preds_d[1].hi; c1 = P->preds_d[1].cslot; }
-        if (npd >= 3) { lo2 = P->preds_d[2].lo; hi2 = P->preds_d[2].hi; c2 = P->preds_d[2].cslot; }
-#pragma unroll 4
-        for (int k = 0; k < CHUNK / WG; k++) {
-          const int r = tid + k * WG;
-          const int inr = r < rows;
-          const double x0 = sval[(size_t)c0 * CHUNK + r];
-          const double x1 = sval[(size_t)c1 * CHUNK + r];
-          const double x2 = sval[(size_t)c2 * CHUNK + r];
-          const int m = inr &&
-              (npd < 1 || (x0 >= lo0 && x0 <= hi0)) &&
-              (npd < 2 || (x1 >= lo1 && x1 <= hi1)) &&
-              (npd < 3 || (x2 >= lo2 && x2 <= hi2));
-          int slot = 0;
-          if (ngroup >= 1) slot = (int)sval[(size_t)gc0 * CHUNK + r];
-          if (ngroup >= 2) slot += (int)sval[(size_t)gc1 * CHUNK + r];
-          double va[NA];
-#pragma unroll
-          for (int a = 0; a < NA; a++) {
-            if (a >= naggs) { va[a] = 0.0; continue; }
-            const sn_dev_agg &A = P->aggs[a];
-            va[a] = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
-                    (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
-                    (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
-          }
-#pragma unroll
-          for (int s = 0; s < NSLOTS; s++) {
-            const int ms = m && slot == s;
-            rc[s] += ms ? 1.0 : 0.0;
-#pragma unroll
-            for (int a = 0; a < NA; a++)
-              sums[s][a] += ms ? va[a] : 0.0;
-          }
-        }
-        __syncthreads();
-        staged = next_staged;
-        continue;
-      }
-
       alive_init(salive, sdead, rows, clean);
       pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
       if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
