@@ -1056,8 +1056,7 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
   uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
   uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
   uint64_t *salive = sdead + CHUNK / 64;
-  int16_t *sslot = (int16_t *)(salive + CHUNK / 64);
-  sn_dev_plan *P = (sn_dev_plan *)(((uintptr_t)(sslot + CHUNK) + 15) & ~(uintptr_t)15);
+  sn_dev_plan *P = (sn_dev_plan *)(salive + CHUNK / 64 + 2);
   double *bacc = (double *)(P + 1);
   {
     const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
@@ -1237,7 +1236,7 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
                        dev_scratch, out_stride);
   } else if (plan->jmode != 1 && ns <= 8 && na <= 6) {
     /* register-accumulator grouped kernel (Q1's shape) */
-    lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)8 * (na + 1) * 8 + 64;
+    lds += (CHUNK / 64) * 8 + (size_t)8 * (na + 1) * 8 + 64;
 #define KGR(S, A, NCv) hipLaunchKernelGGL((k_grouped_reg<S, A, NCv>), dim3(grid), \
         dim3(WG), lds, s, *plan, dev_plan, dev_batches, dev_tiles, ntiles, \
         dev_scratch, out_stride)
