@@ -1101,16 +1101,29 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
       if (STAGED && next_staged) { if constexpr (STAGED) stage_load(cr, nused, nbase, st); }
 
-      alive_init(salive, sdead, rows, clean);
-      pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
-      if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
+      const int inline_preds = clean && npi == 0 && !plan.jkeys;
+      if (!inline_preds) {
+        alive_init(salive, sdead, rows, clean);
+        pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
+        if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
+      }
 
 #pragma unroll 2
       for (int k = 0; k < CHUNK / WG; k++) {
         const int r = tid + k * WG;
-        const uint64_t w = salive[r >> 6];
-        if (w == 0) continue;
-        const int m = (int)((w >> (tid & 63)) & 1ull);
+        int m;
+        if (inline_preds) {
+          m = r < rows;
+          for (int i = 0; i < npd; i++) {
+            const double x = sval[(size_t)P->preds_d[i].cslot * CHUNK + r];
+            m &= (x >= P->preds_d[i].lo) & (x <= P->preds_d[i].hi);
+          }
+          if (__popcll(__ballot(m)) == 0) continue;
+        } else {
+          const uint64_t w = salive[r >> 6];
+          if (w == 0) continue;
+          m = (int)((w >> (tid & 63)) & 1ull);
+        }
         int slot = 0;
         if (ngroup >= 1) slot = (int)sval[(size_t)gc0 * CHUNK + r];
         if (ngroup >= 2) slot += (int)sval[(size_t)gc1 * CHUNK + r];
