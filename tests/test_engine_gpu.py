@@ -365,3 +365,64 @@ def test_large_group_cardinality_on_gpu(eng):
         assert gv[2] == ov[2]
         for a in (0, 1):
             assert abs(gv[a] - ov[a]) <= 1e-6 * max(1.0, abs(ov[a]))
+
+
+def test_single_group_col_and_avg_only(eng):
+    """ngroup=1 grouped path + AVG-only aggregates vs oracle."""
+    n = 300_000
+    rng = np.random.default_rng(41)
+    keys = [b"G%d" % v for v in rng.integers(0, 5, n)]
+    vals = rng.random(n) * 7
+    blobs = [po.encode(po.T_STRING, po.ENC_DICT, keys),
+             po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, vals)]
+    t = eng.table_define("tg1", [(abi.T_STRING, False), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 0, 0, n, blobs)
+    ot = po.OracleTable([po.T_STRING, po.T_DOUBLE])
+    ot.add_batch(n, blobs)
+    kw = dict(group_cols=[0], aggs=[("avg", [(1, 0.0, 1.0)])])
+    grows = eng.query(abi.make_plan(table=t, **kw)).rows()
+    orows = po.result_rows(ot.query(po.make_plan(**kw)))
+    assert len(grows) == len(orows) == 5
+    for (gk, gv), (ok_, ov) in zip(grows, orows):
+        assert gk == ok_
+        assert abs(gv[0] - ov[0]) <= 1e-6 * max(1.0, abs(ov[0]))
+
+
+def test_sharded_merge_large_groups(eng):
+    """sharded partial merge at high cardinality (the path an 8-GPU Q1
+    overflow run exercises): 2 shard engines x 600 groups."""
+    n = 400_000
+    rng = np.random.default_rng(43)
+    keys = [b"K%03d" % v for v in rng.integers(0, 600, n)]
+    vals = rng.random(n)
+    shards = []
+    for rank in range(2):
+        e = se.Engine(device=0, shard_rank=rank, shard_count=2)
+        t = e.table_define("tb", [(abi.T_STRING, False), (abi.T_DOUBLE, False)])
+        for bi, st in enumerate(range(0, n, 50_000)):
+            e_ = min(n, st + 50_000)
+            e.batch_put(t, bi, bi, e_ - st,
+                        [po.encode(po.T_STRING, po.ENC_DICT, keys[st:e_]),
+                         po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, vals[st:e_])])
+        shards.append((e, t))
+    kw = dict(group_cols=[0], aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    qs = [e.query(abi.make_plan(table=t, **kw)) for e, t in shards]
+    blocks = [q.partials_host() for q in qs]
+    qs[0].merge_host(np.concatenate(blocks), len(blocks[0]), 2)
+    merged = qs[0].rows()
+
+    full = eng.table_define("tbf", [(abi.T_STRING, False), (abi.T_DOUBLE, False)])
+    for bi, st in enumerate(range(0, n, 50_000)):
+        e_ = min(n, st + 50_000)
+        eng.batch_put(full, bi, bi, e_ - st,
+                      [po.encode(po.T_STRING, po.ENC_DICT, keys[st:e_]),
+                       po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, vals[st:e_])])
+    direct = eng.query(abi.make_plan(table=full, **kw)).rows()
+    assert len(merged) == len(direct) == 600
+    for (mk, mv), (dk, dv) in zip(merged, direct):
+        assert mk == dk and mv[1] == dv[1]
+        assert abs(mv[0] - dv[0]) <= 1e-6 * max(1.0, abs(dv[0]))
+    for q in qs:
+        q.close()
+    for e, _ in shards:
+        e.close()
