@@ -692,7 +692,7 @@ __global__ void k_keyless(sn_dev_plan plan,
         if (npd >= 1) { lo0 = P->preds_d[0].lo; hi0 = P->preds_d[0].hi; c0 = P->preds_d[0].cslot; }
         if (npd >= 2) { lo1 = P->preds_d[1].lo; hi1 = P->preds_d[1].hi; c1 = P->preds_d[1].cslot; }
         if (npd >= 3) { lo2 = P->preds_d[2].lo; hi2 = P->preds_d[2].hi; c2 = P->preds_d[2].cslot; }
-#pragma unroll 4
+#pragma unroll 2
         for (int k = 0; k < CHUNK / WG; k++) {
           const int r = tid + k * WG;
           const int inr = r < rows;
