@@ -78,6 +78,7 @@ static inline int16_t rd_i16(const uint8_t *p) { int16_t v; memcpy(&v, p, 2); re
 struct Arena {
   int device = -1;               /* -1: host-only shadow mode */
   std::vector<void *> slabs;
+  std::vector<void *> host_allocs;   /* host-only mode: freed on destroy */
   size_t slab_sz = 512ull << 20;
   size_t off = 0;
   std::mutex mu;
@@ -85,7 +86,11 @@ struct Arena {
   void *alloc(size_t n) {
     std::lock_guard<std::mutex> g(mu);
     n = (n + 255) & ~size_t(255);
-    if (device < 0) { void *p = malloc(n); return p; }
+    if (device < 0) {
+      void *p = malloc(n);
+      if (p) host_allocs.push_back(p);
+      return p;
+    }
     if (slabs.empty() || off + n > slab_sz) {
       size_t sz = std::max(slab_sz, n);
       void *p = nullptr;
@@ -100,6 +105,7 @@ struct Arena {
   }
   ~Arena() {
     for (void *p : slabs) (void)hipFree(p);
+    for (void *p : host_allocs) free(p);
   }
 };
 
