@@ -260,6 +260,11 @@ int32_t   sn_query_wait(sn_query *q);
 /* fills final (or this shard's partial, before any merge) result */
 int32_t   sn_query_result(sn_query *q, sn_result *out);
 void      sn_query_destroy(sn_query *q);
+/* scan-kernel duration in ms (HIP events on the launch stream); -1 if the
+ * query launched nothing */
+double    sn_query_kernel_ms(sn_query *q);
+/* 1 when the query ran a query-compiled (hipRTC) kernel, 0 interpreted */
+int32_t   sn_query_used_jit(sn_query *q);
 
 /* ---- multi-GPU partial exchange (caller runs the RCCL collective) ----
  * The caller (one process per GPU, torch.distributed over RCCL/xGMI)

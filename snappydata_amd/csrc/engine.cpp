@@ -174,6 +174,11 @@ struct DescCache {
   const void *tl_dev = nullptr;
   int32_t ntiles = 0;
   int64_t rows = 0;
+  /* JIT eligibility of this descriptor set: every batch clean with these
+   * uniform stageable kinds (needed again on cache hits, when the
+   * per-batch loop that derives them is skipped) */
+  int jit_ok = 0;
+  int jit_kinds[SN_DEV_MAX_COLS] = {0};
 };
 
 struct Table {
@@ -213,6 +218,8 @@ struct sn_engine {
    * (queries on one engine serialize on its stream) */
   double *scratch = nullptr;
   size_t scratch_sz = 0;
+  /* per-engine cache of query-compiled kernels (jit.cpp) */
+  void *jit = nullptr;
   std::mutex mu;
 };
 
@@ -236,12 +243,15 @@ extern "C" sn_engine *sn_engine_create(const sn_config *cfg) {
   }
   if (!e->has_gpu) {
     e->arena.device = -1;   /* host-only: ingest/encode testable, queries fail */
+  } else {
+    e->jit = sn_jit_cache_create();
   }
   return e;
 }
 
 extern "C" void sn_engine_destroy(sn_engine *e) {
   if (!e) return;
+  if (e->jit) sn_jit_cache_destroy(e->jit);
   if (e->stream) (void)hipStreamDestroy(e->stream);
   delete e;
 }
@@ -832,6 +842,7 @@ struct sn_query {
   int64_t batches_seen = 0, batches_skipped = 0;
   hipEvent_t ev_start = nullptr, ev_stop = nullptr;  /* brackets the scan kernel */
   float kernel_ms = -1.0f;
+  bool used_jit = false;                /* query-compiled kernel ran */
   bool done = false;
   bool merged = false;
   std::vector<GroupOut> final_groups;
@@ -1109,6 +1120,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
 
   std::vector<sn_dev_batch> hbatches;
   std::vector<sn_dev_tile> htiles;
+  /* JIT eligibility: every unskipped batch clean, kinds uniform + stageable
+   * (derived here; stored in the DescCache for reuse on hits) */
+  bool jit_ok_b = true, jit_first = true;
+  int jit_kinds[SN_DEV_MAX_COLS] = {0};
   for (auto &b : t->batches) {
     if (hit) break;
     if (skip_count && batch_skippable(b, plan, t)) continue;
@@ -1207,6 +1222,17 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       }
     }
     db.clean = clean ? 1 : 0;
+    if (!clean) jit_ok_b = false;
+    for (size_t ui = 0; ui < q->used_cols.size() && jit_ok_b; ui++) {
+      int k = db.cols[ui].kind;
+      bool stageable = k == SN_K_F64 || k == SN_K_I64 || k == SN_K_I32 ||
+                       k == SN_K_F32 || k == SN_K_I16 || k == SN_K_DICT16 ||
+                       k == SN_K_DICT32;
+      if (!stageable) jit_ok_b = false;
+      else if (jit_first) jit_kinds[ui] = k;
+      else if (jit_kinds[ui] != k) jit_ok_b = false;
+    }
+    jit_first = false;
     int32_t bi = (int32_t)hbatches.size();
     hbatches.push_back(db);
     q->rows_scanned += b.num_rows;
@@ -1246,6 +1272,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       dc.sig = sig; dc.batch_count = t->batches.size();
       dc.db_dev = db_dev; dc.tl_dev = tl_dev; dc.ntiles = ntiles;
       dc.rows = q->rows_scanned;
+      dc.jit_ok = (jit_ok_b && !jit_first) ? 1 : 0;
+      memcpy(dc.jit_kinds, jit_kinds, sizeof(jit_kinds));
       t->desc_caches.push_back(dc);
     }
   }
@@ -1270,11 +1298,41 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     }
     (void)hipEventCreate(&q->ev_start);
     (void)hipEventCreate(&q->ev_stop);
+    /* query-compiled kernel (jit.cpp): plan constants baked as literals —
+     * the WholeStageCodegen analogue; measured 3x on Q1's shape over the
+     * interpreted runtime-plan kernel.  Any miss falls back, still on GPU. */
+    void *jfn = nullptr;
+    if (e->jit && plan->join_dim < 0 &&
+        (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first)) &&
+        (dp.nslots <= 1 ? dp.naggs <= 12
+                        : (dp.nslots <= 8 && dp.naggs <= 6))) {
+      const int *jk = hit ? hit->jit_kinds : jit_kinds;
+      jfn = sn_jit_get(e->jit, &dp, jk, dp.nslots, q->na_t);
+    }
     if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
-    int rc = sn_launch_scan_agg(&dp, (const sn_dev_plan *)dp_dev,
-                                (const sn_dev_batch *)db_dev,
-                                (const sn_dev_tile *)tl_dev, ntiles,
-                                q->dev_out, e->scratch, e->stream);
+    int rc = -1;
+    if (jfn) {
+      int jgrid;
+      if (ntiles <= SN_GRID_CAP) jgrid = ntiles;
+      else {
+        int rounds = (ntiles + SN_GRID_CAP - 1) / SN_GRID_CAP;
+        jgrid = (ntiles + rounds - 1) / rounds;
+      }
+      int naggs1 = dp.nslots <= 1 ? 0 : dp.naggs + 1;
+      rc = sn_jit_launch(jfn, jgrid, (const sn_dev_batch *)db_dev,
+                         (const sn_dev_tile *)tl_dev, ntiles, e->scratch,
+                         e->stream);
+      if (rc == 0)
+        rc = sn_launch_reduce(e->scratch, jgrid, (int)nv, q->dev_out,
+                              naggs1, (int)q->out_stride, e->stream);
+      q->used_jit = rc == 0;
+      if (rc != 0) jfn = nullptr;   /* interpreted kernels take over */
+    }
+    if (!jfn)
+      rc = sn_launch_scan_agg(&dp, (const sn_dev_plan *)dp_dev,
+                              (const sn_dev_batch *)db_dev,
+                              (const sn_dev_tile *)tl_dev, ntiles,
+                              q->dev_out, e->scratch, e->stream);
     if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);
     if (rc != 0) {
       fail(SN_ERR_GENERIC, "kernel launch: %s", hipGetErrorString((hipError_t)rc));
@@ -1305,6 +1363,11 @@ extern "C" double sn_query_kernel_ms(sn_query *q) {
   if (!q) return -1.0;
   (void)sn_query_wait(q);
   return (double)q->kernel_ms;
+}
+
+/* 1 when the query ran a query-compiled (hipRTC) kernel, 0 interpreted */
+extern "C" int32_t sn_query_used_jit(sn_query *q) {
+  return q && q->used_jit ? 1 : 0;
 }
 
 /* local accumulators -> GroupOut list (pre-merge view) */

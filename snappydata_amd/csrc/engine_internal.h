@@ -124,6 +124,23 @@ int sn_launch_scan_agg(const sn_dev_plan *plan,
                        double *dev_scratch,  /* >= min(ntiles,SN_GRID_CAP) x nv */
                        void *stream);
 
+/* launches only the partial-fold (k_reduce): scratch[nblocks][nv] -> out.
+ * Used by the JIT path, whose scan kernel writes the same scratch rows. */
+int sn_launch_reduce(const double *dev_scratch, int nblocks, int nv,
+                     double *dev_out, int naggs1, int out_stride,
+                     void *stream);
+
+/* query-compiled scan kernels (jit.cpp, hipRTC).  sn_jit_get returns an
+ * opaque hipFunction_t for the plan (compiling + caching on first use) or
+ * NULL -> caller falls back to the interpreted kernels. */
+void *sn_jit_cache_create(void);
+void sn_jit_cache_destroy(void *cache);
+void *sn_jit_get(void *cache, const sn_dev_plan *p, const int *kinds,
+                 int nslots, int na_t);
+int sn_jit_launch(void *fn, int grid, const sn_dev_batch *batches,
+                  const sn_dev_tile *tiles, int ntiles, double *scratch,
+                  void *stream);
+
 #ifdef __cplusplus
 }
 #endif

@@ -1,0 +1,399 @@
+/*
+ * jit.cpp — query-compiled scan kernels via hipRTC.
+ *
+ * The MI355X-native analogue of the reference's WholeStageCodegen
+ * (SnappySession.scala:2388-2460 compiles one Java class per query stage;
+ * here one gfx950 kernel per plan signature).  Motivation is measured, not
+ * aesthetic: a probe kernel with Q1's exact shape and the plan constants
+ * compile-time reaches 5.7 TB/s, while the best runtime-plan kernel stops at
+ * 1.76 TB/s — literal predicates/aggregates let the compiler
+ * CSE shared factors, keep every parameter in registers and emit zero
+ * uniform branches in the row loop.
+ *
+ * Scope: plans whose unskipped batches are all CLEAN (no nulls, deletes or
+ * patches) with uniform stageable column kinds, <= 4 double predicates, no
+ * join, dictionary-slot or keyless grouping.  Everything else falls back to
+ * the interpreted kernels (still GPU — never CPU).
+ *
+ * Compiled modules cache per engine by source hash (the reference caches
+ * generated classes the same way; literal re-tokenization is round-2).
+ */
+#include <hip/hip_runtime.h>
+#include <hip/hiprtc.h>
+
+#include <cstdarg>
+#include <cstdio>
+#include <cstring>
+#include <map>
+#include <mutex>
+#include <string>
+#include <vector>
+
+#include "engine_internal.h"
+
+struct JitFn {
+  hipModule_t mod = nullptr;
+  hipFunction_t fn = nullptr;
+};
+
+struct JitCache {
+  std::map<uint64_t, JitFn> fns;
+  std::mutex mu;
+};
+
+extern "C" void *sn_jit_cache_create(void) { return new JitCache(); }
+extern "C" void sn_jit_cache_destroy(void *c) {
+  auto *jc = (JitCache *)c;
+  if (!jc) return;
+  for (auto &kv : jc->fns)
+    if (kv.second.mod) (void)hipModuleUnload(kv.second.mod);
+  delete jc;
+}
+
+static uint64_t fnv1a(const std::string &s) {
+  uint64_t h = 1469598103934665603ull;
+  for (char c : s) { h ^= (uint8_t)c; h *= 1099511628211ull; }
+  return h;
+}
+
+static void emitf(std::string &o, const char *fmt, ...) {
+  char buf[1024];
+  va_list ap; va_start(ap, fmt);
+  vsnprintf(buf, sizeof(buf), fmt, ap);
+  va_end(ap);
+  o += buf;
+}
+
+/* double literal that round-trips exactly */
+static std::string dlit(double v) {
+  char b[64];
+  snprintf(b, sizeof(b), "__longlong_as_double(%lldll)",
+           (long long)*(long long *)&v);
+  return b;
+}
+
+/* Generate the specialized kernel source.
+ * kinds[c]: SN_K_* per used column slot (uniform across batches).
+ * Layout contract identical to the interpreted kernels:
+ *   keyless: scratch row [2*na_t+1] = [sums][counts][rowcount]
+ *   grouped: scratch row [nslots*(naggs+1)], rowcount at +naggs.         */
+static std::string gen_source(const sn_dev_plan *p, const int *kinds,
+                              int nslots, int na_t) {
+  const int NC = p->nused;
+  const int NA = p->naggs;
+  const int grouped = nslots > 1;
+  std::string o;
+  o += R"(
+typedef double double2_t __attribute__((ext_vector_type(2)));
+typedef int int2_t __attribute__((ext_vector_type(2)));
+typedef float float2_t __attribute__((ext_vector_type(2)));
+#define GAS __attribute__((address_space(1)))
+typedef unsigned long long u64;
+typedef long long i64;
+/* layout mirror of engine_internal.h (pointers + i32 fields only) */
+struct sn_dev_col {
+  const void *body; const u64 *nullw; const unsigned *nullpfx;
+  const int *dictmap; const u64 *patch_bm; const int *patch_pos;
+  const double *patch_val; const u64 *patch_nullbm;
+  const int *rle_ends; const double *rle_vals; int rle_n;
+  int patch_n; int kind; int has_nulls; int null_gid;
+};
+struct sn_dev_batch { int num_rows; int clean; const u64 *del_bm; sn_dev_col cols[)";
+  emitf(o, "%d", SN_DEV_MAX_COLS);
+  o += R"(]; };
+struct sn_dev_tile { int batch; int row_start; };
+__device__ __forceinline__ double wsum(double x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_down(x, off, 64);
+  return x;
+}
+)";
+  emitf(o, "#define CHUNK %d\n#define WG %d\n#define TILE %d\n",
+        1024, 256, SN_TILE_ROWS);
+  emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, 2)\n"
+           "void jit_scan(const sn_dev_batch *__restrict__ batches,\n"
+           "              const sn_dev_tile *__restrict__ tiles, int ntiles,\n"
+           "              double *__restrict__ out) {\n");
+  emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
+  emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
+        grouped ? nslots * (NA + 1) : 2 * na_t + 1);
+  o += "  const int tid = threadIdx.x;\n";
+
+  /* accumulators */
+  if (grouped) {
+    emitf(o, "  double sums[%d][%d]; double rc[%d];\n", nslots, NA, nslots);
+    emitf(o, "#pragma unroll\n  for (int s = 0; s < %d; s++) { rc[s] = 0;\n"
+             "#pragma unroll\n    for (int a = 0; a < %d; a++) sums[s][a] = 0; }\n",
+          nslots, NA);
+  } else {
+    emitf(o, "  double sums[%d], cnts[%d], rcnt = 0.0;\n", NA, NA);
+    emitf(o, "#pragma unroll\n  for (int a = 0; a < %d; a++) { sums[a] = 0; cnts[a] = 0; }\n", NA);
+  }
+
+  /* staged register buffers: per column, by width class */
+  for (int c = 0; c < NC; c++) {
+    int k = kinds[c];
+    if (k == SN_K_F64 || k == SN_K_I64)
+      emitf(o, "  double2_t st%d_0, st%d_1;\n", c, c);
+    else if (k == SN_K_I32 || k == SN_K_F32 || k == SN_K_DICT32)
+      emitf(o, "  int2_t st%d_0, st%d_1;\n", c, c);
+    else /* I16/DICT16 */
+      emitf(o, "  unsigned st%d_0, st%d_1;\n", c, c);
+  }
+
+  o += R"(
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const sn_dev_tile tile = tiles[t];
+    const sn_dev_batch &b = batches[tile.batch];
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + TILE, num_rows);
+)";
+  /* hoist body/dict pointers */
+  for (int c = 0; c < NC; c++) {
+    emitf(o, "    const GAS char *body%d = (const GAS char *)(unsigned long long)b.cols[%d].body;\n", c, c);
+    if (kinds[c] == SN_K_DICT16 || kinds[c] == SN_K_DICT32)
+      emitf(o, "    const GAS int *dm%d = (const GAS int *)(unsigned long long)b.cols[%d].dictmap;\n", c, c);
+  }
+
+  /* stage_load macro body as a lambda-ish emitted twice (prologue + in-loop) */
+  auto emit_load = [&](const char *base_expr, const char *ind) {
+    for (int c = 0; c < NC; c++) {
+      int k = kinds[c];
+      if (k == SN_K_F64 || k == SN_K_I64) {
+        emitf(o, "%sst%d_0 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid];\n", ind, c, c, base_expr);
+        emitf(o, "%sst%d_1 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid + WG];\n", ind, c, c, base_expr);
+      } else if (k == SN_K_I32 || k == SN_K_F32 || k == SN_K_DICT32) {
+        emitf(o, "%sst%d_0 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid];\n", ind, c, c, base_expr);
+        emitf(o, "%sst%d_1 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid + WG];\n", ind, c, c, base_expr);
+      } else {
+        emitf(o, "%sst%d_0 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid];\n", ind, c, c, base_expr);
+        emitf(o, "%sst%d_1 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid + WG];\n", ind, c, c, base_expr);
+      }
+    }
+  };
+  auto emit_write = [&](const char *ind) {
+    for (int c = 0; c < NC; c++) {
+      int k = kinds[c];
+      if (k == SN_K_F64 || k == SN_K_I64) {
+        emitf(o, "%s((double2_t *)sval[%d])[tid] = st%d_0;\n", ind, c, c);
+        emitf(o, "%s((double2_t *)sval[%d])[tid + WG] = st%d_1;\n", ind, c, c);
+      } else if (k == SN_K_I32) {
+        emitf(o, "%s{ double2_t y0; y0.x = (double)st%d_0.x; y0.y = (double)st%d_0.y;\n"
+                 "%s  ((double2_t *)sval[%d])[tid] = y0;\n"
+                 "%s  double2_t y1; y1.x = (double)st%d_1.x; y1.y = (double)st%d_1.y;\n"
+                 "%s  ((double2_t *)sval[%d])[tid + WG] = y1; }\n",
+              ind, c, c, ind, c, ind, c, c, ind, c);
+      } else if (k == SN_K_F32) {
+        emitf(o, "%s{ float2_t f0 = *(float2_t *)&st%d_0, f1 = *(float2_t *)&st%d_1;\n"
+                 "%s  double2_t y0; y0.x = (double)f0.x; y0.y = (double)f0.y;\n"
+                 "%s  double2_t y1; y1.x = (double)f1.x; y1.y = (double)f1.y;\n"
+                 "%s  ((double2_t *)sval[%d])[tid] = y0;\n"
+                 "%s  ((double2_t *)sval[%d])[tid + WG] = y1; }\n",
+              ind, c, c, ind, ind, ind, c, ind, c);
+      } else if (k == SN_K_DICT32) {
+        emitf(o, "%s{ double2_t y0; y0.x = (double)dm%d[st%d_0.x]; y0.y = (double)dm%d[st%d_0.y];\n"
+                 "%s  double2_t y1; y1.x = (double)dm%d[st%d_1.x]; y1.y = (double)dm%d[st%d_1.y];\n"
+                 "%s  ((double2_t *)sval[%d])[tid] = y0;\n"
+                 "%s  ((double2_t *)sval[%d])[tid + WG] = y1; }\n",
+              ind, c, c, c, c, ind, c, c, c, c, ind, c, ind, c);
+      } else if (k == SN_K_DICT16) {
+        emitf(o, "%s{ double2_t y0; y0.x = (double)dm%d[st%d_0 & 0xffff]; y0.y = (double)dm%d[st%d_0 >> 16];\n"
+                 "%s  double2_t y1; y1.x = (double)dm%d[st%d_1 & 0xffff]; y1.y = (double)dm%d[st%d_1 >> 16];\n"
+                 "%s  ((double2_t *)sval[%d])[tid] = y0;\n"
+                 "%s  ((double2_t *)sval[%d])[tid + WG] = y1; }\n",
+              ind, c, c, c, c, ind, c, c, c, c, ind, c, ind, c);
+      } else { /* I16 signed */
+        emitf(o, "%s{ double2_t y0; y0.x = (double)(short)(st%d_0 & 0xffff); y0.y = (double)(short)(st%d_0 >> 16);\n"
+                 "%s  double2_t y1; y1.x = (double)(short)(st%d_1 & 0xffff); y1.y = (double)(short)(st%d_1 >> 16);\n"
+                 "%s  ((double2_t *)sval[%d])[tid] = y0;\n"
+                 "%s  ((double2_t *)sval[%d])[tid + WG] = y1; }\n",
+              ind, c, c, ind, c, c, ind, c, ind, c);
+      }
+    }
+  };
+
+  o += "    int staged = 0;\n"
+       "    if (tile.row_start + CHUNK <= tile_end) {\n";
+  emit_load("tile.row_start", "      ");
+  o += "      staged = 1;\n    }\n"
+       "    for (int base = tile.row_start; base < tile_end; base += CHUNK) {\n"
+       "      const int rows = min(CHUNK, tile_end - base);\n"
+       "      if (staged) {\n";
+  emit_write("        ");
+  o += "      } else {\n"
+       "        /* scalar tail conversion */\n";
+  for (int c = 0; c < NC; c++) {
+    int k = kinds[c];
+    emitf(o, "        for (int r = tid; r < rows; r += WG) sval[%d][r] = ", c);
+    switch (k) {
+      case SN_K_F64:
+        emitf(o, "((const GAS double *)(body%d))[base + r];\n", c); break;
+      case SN_K_I64:
+        emitf(o, "((const GAS double *)(body%d))[base + r];\n", c); break; /* raw bits */
+      case SN_K_I32:
+        emitf(o, "(double)((const GAS int *)(body%d))[base + r];\n", c); break;
+      case SN_K_F32:
+        emitf(o, "(double)((const GAS float *)(body%d))[base + r];\n", c); break;
+      case SN_K_DICT32:
+        emitf(o, "(double)dm%d[((const GAS int *)(body%d))[base + r]];\n", c, c); break;
+      case SN_K_DICT16:
+        emitf(o, "(double)dm%d[(int)((const GAS unsigned short *)(body%d))[base + r]];\n", c, c); break;
+      default: /* I16 */
+        emitf(o, "(double)((const GAS short *)(body%d))[base + r];\n", c); break;
+    }
+  }
+  o += "      }\n"
+       "      __syncthreads();\n"
+       "      const int nbase = base + CHUNK;\n"
+       "      const int next_staged = nbase + CHUNK <= tile_end;\n"
+       "      if (next_staged) {\n";
+  emit_load("nbase", "        ");
+  o += "      }\n";
+
+  /* fused row pass */
+  o += "#pragma unroll 2\n"
+       "      for (int k = 0; k < CHUNK / WG; k++) {\n"
+       "        const int r = tid + k * WG;\n"
+       "        int ok = r < rows;\n";
+  for (int i = 0; i < p->npreds_d; i++) {
+    emitf(o, "        { const double x = sval[%d][r];\n"
+             "          ok &= (x >= %s) & (x <= %s); }\n",
+          p->preds_d[i].cslot, dlit(p->preds_d[i].lo).c_str(),
+          dlit(p->preds_d[i].hi).c_str());
+  }
+  for (int i = 0; i < p->npreds_i; i++) {
+    emitf(o, "        { const i64 x = __double_as_longlong(sval[%d][r]);\n"
+             "          ok &= (x >= %lldll) & (x <= %lldll); }\n",
+          p->preds_i[i].cslot, (long long)p->preds_i[i].lo,
+          (long long)p->preds_i[i].hi);
+  }
+  o += "        if (__popcll(__ballot(ok)) == 0) continue;\n";
+  if (grouped) {
+    emitf(o, "        int slot = (int)sval[%d][r];\n", p->gcol[0]);
+    if (p->ngroup >= 2)
+      emitf(o, "        slot += (int)sval[%d][r];\n", p->gcol[1]);
+  }
+  for (int a = 0; a < NA; a++) {
+    const sn_dev_agg &A = p->aggs[a];
+    emitf(o, "        const double va%d = (%s + %s * sval[%d][r])", a,
+          dlit(A.a0).c_str(), dlit(A.m0).c_str(), A.c0);
+    emitf(o, " * (%s + %s * sval[%d][r])", dlit(A.a1).c_str(),
+          dlit(A.m1).c_str(), A.c1);
+    emitf(o, " * (%s + %s * sval[%d][r]);\n", dlit(A.a2).c_str(),
+          dlit(A.m2).c_str(), A.c2);
+  }
+  if (grouped) {
+    emitf(o, "#pragma unroll\n        for (int s = 0; s < %d; s++) {\n"
+             "          const int ms = ok && slot == s;\n"
+             "          rc[s] += ms ? 1.0 : 0.0;\n", nslots);
+    for (int a = 0; a < NA; a++)
+      emitf(o, "          sums[s][%d] += ms ? va%d : 0.0;\n", a, a);
+    o += "        }\n";
+  } else {
+    for (int a = 0; a < NA; a++) {
+      emitf(o, "        sums[%d] += ok ? va%d : 0.0;\n", a, a);
+      emitf(o, "        cnts[%d] += ok ? 1.0 : 0.0;\n", a);
+    }
+    o += "        rcnt += ok ? 1.0 : 0.0;\n";
+  }
+  o += "      }\n"
+       "      __syncthreads();\n"
+       "      staged = next_staged;\n"
+       "    }\n"
+       "  }\n";
+
+  /* block reduce into LDS bacc then scratch row */
+  int nv = grouped ? nslots * (NA + 1) : 2 * na_t + 1;
+  emitf(o, "  for (int i = tid; i < %d; i += WG) bacc[i] = 0.0;\n"
+           "  __syncthreads();\n", nv);
+  if (grouped) {
+    emitf(o, "#pragma unroll\n  for (int s = 0; s < %d; s++) {\n", nslots);
+    for (int a = 0; a < NA; a++)
+      emitf(o, "    { double x = wsum(sums[s][%d]);\n"
+               "      if ((tid & 63) == 0 && x != 0.0) atomicAdd(&bacc[s * %d + %d], x); }\n",
+            a, NA + 1, a);
+    emitf(o, "    { double x = wsum(rc[s]);\n"
+             "      if ((tid & 63) == 0 && x != 0.0) atomicAdd(&bacc[s * %d + %d], x); }\n",
+          NA + 1, NA);
+    o += "  }\n";
+  } else {
+    for (int a = 0; a < NA; a++) {
+      emitf(o, "  { double x = wsum(sums[%d]);\n"
+               "    if ((tid & 63) == 0 && x != 0.0) atomicAdd(&bacc[%d], x); }\n", a, a);
+      emitf(o, "  { double x = wsum(cnts[%d]);\n"
+               "    if ((tid & 63) == 0 && x != 0.0) atomicAdd(&bacc[%d], x); }\n",
+            a, na_t + a);
+    }
+    emitf(o, "  { double x = wsum(rcnt);\n"
+             "    if ((tid & 63) == 0 && x != 0.0) atomicAdd(&bacc[%d], x); }\n",
+          2 * na_t);
+  }
+  emitf(o, "  __syncthreads();\n"
+           "  for (int i = tid; i < %d; i += WG)\n"
+           "    out[(u64)blockIdx.x * %d + i] = bacc[i];\n"
+           "}\n", nv, nv);
+  return o;
+}
+
+/* compile (or fetch) the kernel for this plan; returns NULL on any failure
+ * (caller falls back to the interpreted kernels) */
+extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
+                            const int *kinds, int nslots, int na_t) {
+  auto *jc = (JitCache *)cache;
+  if (!jc) return nullptr;
+  std::string src = gen_source(p, kinds, nslots, na_t);
+  if (const char *dump = getenv("SN_JIT_DUMP")) {
+    if (FILE *f = fopen(dump, "a")) {
+      fputs(src.c_str(), f);
+      fputs("\n/* ---- */\n", f);
+      fclose(f);
+    }
+  }
+  uint64_t h = fnv1a(src);
+  std::lock_guard<std::mutex> g(jc->mu);
+  auto it = jc->fns.find(h);
+  if (it != jc->fns.end()) return (void *)it->second.fn;
+
+  hiprtcProgram prog;
+  if (hiprtcCreateProgram(&prog, src.c_str(), "sn_jit.cu", 0, nullptr,
+                          nullptr) != HIPRTC_SUCCESS)
+    return nullptr;
+  const char *opts[] = { "--offload-arch=gfx950", "-O3" };
+  hiprtcResult crc = hiprtcCompileProgram(prog, 2, opts);
+  if (crc != HIPRTC_SUCCESS) {
+    size_t lsz = 0;
+    hiprtcGetProgramLogSize(prog, &lsz);
+    std::vector<char> log(lsz + 1, 0);
+    if (lsz) hiprtcGetProgramLog(prog, log.data());
+    fprintf(stderr, "[sn_jit] compile failed:\n%s\n", log.data());
+    hiprtcDestroyProgram(&prog);
+    jc->fns[h] = JitFn();   /* negative-cache */
+    return nullptr;
+  }
+  size_t csz = 0;
+  hiprtcGetCodeSize(prog, &csz);
+  std::vector<char> code(csz);
+  hiprtcGetCode(prog, code.data());
+  hiprtcDestroyProgram(&prog);
+
+  JitFn jf;
+  if (hipModuleLoadData(&jf.mod, code.data()) != hipSuccess) return nullptr;
+  if (hipModuleGetFunction(&jf.fn, jf.mod, "jit_scan") != hipSuccess) {
+    (void)hipModuleUnload(jf.mod);
+    return nullptr;
+  }
+  jc->fns[h] = jf;
+  return (void *)jf.fn;
+}
+
+extern "C" int sn_jit_launch(void *fn, int grid,
+                             const sn_dev_batch *batches,
+                             const sn_dev_tile *tiles, int ntiles,
+                             double *scratch, void *stream) {
+  void *args[] = { (void *)&batches, (void *)&tiles, (void *)&ntiles,
+                   (void *)&scratch };
+  hipError_t e = hipModuleLaunchKernel((hipFunction_t)fn, grid, 1, 1,
+                                       256, 1, 1, 0, (hipStream_t)stream,
+                                       args, nullptr);
+  return (int)e;
+}

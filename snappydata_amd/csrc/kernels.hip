@@ -1200,6 +1200,14 @@ __global__ void k_reduce(const double *__restrict__ scratch, int nblocks,
   }
 }
 
+extern "C" int sn_launch_reduce(const double *dev_scratch, int nblocks,
+                                int nv, double *dev_out, int naggs1,
+                                int out_stride, void *stream) {
+  hipLaunchKernelGGL(k_reduce, dim3(nv), dim3(WG), 0, (hipStream_t)stream,
+                     dev_scratch, nblocks, nv, dev_out, naggs1, out_stride);
+  return (int)hipGetLastError();
+}
+
 extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
                                   const sn_dev_plan *dev_plan,
                                   const sn_dev_batch *dev_batches,

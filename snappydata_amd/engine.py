@@ -81,6 +81,8 @@ def lib():
         L.sn_query_destroy.argtypes = [C.c_void_p]
         L.sn_query_kernel_ms.restype = C.c_double
         L.sn_query_kernel_ms.argtypes = [C.c_void_p]
+        L.sn_query_used_jit.restype = C.c_int32
+        L.sn_query_used_jit.argtypes = [C.c_void_p]
         L.sn_query_partial_bytes.restype = C.c_int64
         L.sn_query_partial_bytes.argtypes = [C.c_void_p]
         L.sn_query_partials.restype = C.c_int32
@@ -213,6 +215,10 @@ class Query:
     def kernel_ms(self):
         """Scan-kernel duration (HIP events on the launch stream)."""
         return lib().sn_query_kernel_ms(self._h)
+
+    def used_jit(self):
+        """True when the query ran a query-compiled (hipRTC) kernel."""
+        return bool(lib().sn_query_used_jit(self._h))
 
     def partial_bytes(self):
         return _check(lib().sn_query_partial_bytes(self._h))
