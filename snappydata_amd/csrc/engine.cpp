@@ -1302,7 +1302,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
      * the WholeStageCodegen analogue; measured 3x on Q1's shape over the
      * interpreted runtime-plan kernel.  Any miss falls back, still on GPU. */
     void *jfn = nullptr;
-    if (e->jit && plan->join_dim < 0 &&
+    if (e->jit &&
         (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first)) &&
         (dp.nslots <= 1 ? dp.naggs <= 12
                         : (dp.nslots <= 8 && dp.naggs <= 6))) {
@@ -1321,7 +1321,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       int naggs1 = dp.nslots <= 1 ? 0 : dp.naggs + 1;
       rc = sn_jit_launch(jfn, jgrid, (const sn_dev_batch *)db_dev,
                          (const sn_dev_tile *)tl_dev, ntiles, e->scratch,
-                         e->stream);
+                         dp.jkeys, dp.jpayload, e->stream);
       if (rc == 0)
         rc = sn_launch_reduce(e->scratch, jgrid, (int)nv, q->dev_out,
                               naggs1, (int)q->out_stride, e->stream);

@@ -139,6 +139,7 @@ void *sn_jit_get(void *cache, const sn_dev_plan *p, const int *kinds,
                  int nslots, int na_t);
 int sn_jit_launch(void *fn, int grid, const sn_dev_batch *batches,
                   const sn_dev_tile *tiles, int ntiles, double *scratch,
+                  const int64_t *jkeys, const int32_t *jpayload,
                   void *stream);
 
 #ifdef __cplusplus

@@ -107,13 +107,24 @@ __device__ __forceinline__ double wsum(double x) {
   for (int off = 32; off > 0; off >>= 1) x += __shfl_down(x, off, 64);
   return x;
 }
+__device__ __forceinline__ u64 mix64(u64 x) {
+  x += 0x9E3779B97f4A7C15ull;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
+  return x ^ (x >> 31);
+}
 )";
   emitf(o, "#define CHUNK %d\n#define WG %d\n#define TILE %d\n",
         1024, 256, SN_TILE_ROWS);
   emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, 2)\n"
            "void jit_scan(const sn_dev_batch *__restrict__ batches,\n"
            "              const sn_dev_tile *__restrict__ tiles, int ntiles,\n"
-           "              double *__restrict__ out) {\n");
+           "              double *__restrict__ out,\n"
+           "              const i64 *__restrict__ jkeys_p,\n"
+           "              const int *__restrict__ jpayload_p) {\n"
+           "  const GAS i64 *jkeys = (const GAS i64 *)(u64)jkeys_p;\n"
+           "  const GAS int *jpayload = (const GAS int *)(u64)jpayload_p;\n"
+           "  (void)jkeys; (void)jpayload;\n");
   emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
   emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
         grouped ? nslots * (NA + 1) : 2 * na_t + 1);
@@ -268,10 +279,36 @@ __device__ __forceinline__ double wsum(double x) {
           (long long)p->preds_i[i].hi);
   }
   o += "        if (__popcll(__ballot(ok)) == 0) continue;\n";
+  if (p->jkeys) {
+    /* broadcast-dimension probe with literal table shape (mask, key slot,
+     * i64-ness); empty-slot sentinel = INT64_MIN, same as the interpreted
+     * probe_sweep */
+    int is_i64 = (p->i64_mask >> p->jcslot) & 1u;
+    emitf(o, "        int pay = -1;\n"
+             "        if (ok) {\n"
+             "          const double jx = sval[%d][r];\n"
+             "          const i64 key = %s;\n"
+             "          unsigned h = (unsigned)mix64((u64)key) & %uu;\n"
+             "          while (true) {\n"
+             "            const i64 k0 = jkeys[h];\n"
+             "            if (k0 == key) { pay = jpayload[h]; break; }\n"
+             "            if (k0 == (i64)0x8000000000000000ll) break;\n"
+             "            h = (h + 1) & %uu;\n"
+             "          }\n"
+             "          ok = pay >= 0;\n"
+             "        }\n",
+          p->jcslot,
+          is_i64 ? "__double_as_longlong(jx)" : "(i64)jx",
+          (1u << p->jcap_log2) - 1, (1u << p->jcap_log2) - 1);
+  }
   if (grouped) {
-    emitf(o, "        int slot = (int)sval[%d][r];\n", p->gcol[0]);
-    if (p->ngroup >= 2)
-      emitf(o, "        slot += (int)sval[%d][r];\n", p->gcol[1]);
+    if (p->jkeys && p->jmode == 1) {
+      o += "        const int slot = pay > 0 ? pay : 0;\n";
+    } else {
+      emitf(o, "        int slot = (int)sval[%d][r];\n", p->gcol[0]);
+      if (p->ngroup >= 2)
+        emitf(o, "        slot += (int)sval[%d][r];\n", p->gcol[1]);
+    }
   }
   for (int a = 0; a < NA; a++) {
     const sn_dev_agg &A = p->aggs[a];
@@ -389,9 +426,10 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
 extern "C" int sn_jit_launch(void *fn, int grid,
                              const sn_dev_batch *batches,
                              const sn_dev_tile *tiles, int ntiles,
-                             double *scratch, void *stream) {
+                             double *scratch, const int64_t *jkeys,
+                             const int32_t *jpayload, void *stream) {
   void *args[] = { (void *)&batches, (void *)&tiles, (void *)&ntiles,
-                   (void *)&scratch };
+                   (void *)&scratch, (void *)&jkeys, (void *)&jpayload };
   hipError_t e = hipModuleLaunchKernel((hipFunction_t)fn, grid, 1, 1,
                                        256, 1, 1, 0, (hipStream_t)stream,
                                        args, nullptr);

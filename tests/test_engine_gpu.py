@@ -426,3 +426,41 @@ def test_sharded_merge_large_groups(eng):
         q.close()
     for e, _ in shards:
         e.close()
+
+
+def test_jit_engages_on_clean_scans(eng, li_fixture_table):
+    """Clean fixture batches with uniform kinds must run the query-compiled
+    (hipRTC) kernel — and fall back to the interpreted kernels the moment a
+    delta patch dirties a batch, with identical results either way."""
+    q6 = eng.query(q6_plan_engine(li_fixture_table))
+    q6.rows()
+    assert q6.used_jit(), "Q6 over clean batches should run the JIT kernel"
+    q1 = eng.query(q1_plan_engine(li_fixture_table))
+    q1.rows()
+    assert q1.used_jit(), "Q1 over clean batches should run the JIT kernel"
+
+    # dirty one batch: JIT must step aside, result must not change shape
+    n = 40_000
+    rng = np.random.default_rng(77)
+    f64 = np.round(rng.random(n), 3)
+    t = eng.table_define("tjit", [(abi.T_DOUBLE, False)])
+    eng.batch_put(t, 0, 0, n,
+                  [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)])
+    plan = abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    qa = eng.query(plan)
+    ra = qa.rows()
+    assert qa.used_jit()
+    pos = np.array([1, 5], dtype=np.int32)
+    nv = np.array([9.5, 10.5])
+    delta = se.encode_update_delta(abi.T_DOUBLE, pos, n, nv)
+    t2 = eng.table_define("tjit2", [(abi.T_DOUBLE, False)])
+    eng.batch_put(t2, 0, 0, n,
+                  [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)],
+                  deltas=[(delta, None)])
+    plan2 = abi.make_plan(table=t2, aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    qb = eng.query(plan2)
+    rb = qb.rows()
+    assert not qb.used_jit(), "patched batch must use the interpreted path"
+    assert rb[0][1][1] == ra[0][1][1]
+    exp = f64.sum() - f64[pos].sum() + nv.sum()
+    assert abs(rb[0][1][0] - exp) <= 1e-9 * abs(exp)
