@@ -178,6 +178,7 @@ struct DescCache {
    * uniform stageable kinds (needed again on cache hits, when the
    * per-batch loop that derives them is skipped) */
   int jit_ok = 0;
+  int jit_del = 0;
   int jit_kinds[SN_DEV_MAX_COLS] = {0};
 };
 
@@ -716,6 +717,32 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
       pd.pos = (const int32_t *)up(e, P.pos.data(), P.pos.size() * 4);
       pd.val = (const double *)up(e, P.val.data(), P.val.size() * 8);
       pd.nullbm = any_null ? (const uint64_t *)up(e, nbm.data(), nbm.size() * 8) : nullptr;
+
+      /* materialize value-only patches straight into the device body when
+       * the base column is null-free fixed-width: the batch then scans
+       * clean (query-compiled kernels apply).  Scan results are identical
+       * by construction — read_general would hand back exactly these
+       * values; the host blob (sn_table_get_blob) keeps the base bytes. */
+      if (e->has_gpu && !any_null && b.cols[c].num_null_words == 0 &&
+          b.cols[c].type_id == SN_ENC_UNCOMPRESSED) {
+        int k = -1;
+        switch (t->schema[c].dtype) {
+          case SN_TYPE_DOUBLE: k = SN_K_F64; break;
+          case SN_TYPE_FLOAT:  k = SN_K_F32; break;
+          case SN_TYPE_INT32:  k = SN_K_I32; break;
+          case SN_TYPE_INT64:  k = SN_K_I64; break;
+          case SN_TYPE_INT16:  k = SN_K_I16; break;
+          default: break;
+        }
+        if (k >= 0) {
+          void *body = (uint8_t *)(uintptr_t)b.col_dev[c] + b.cols[c].body_off;
+          if (sn_launch_patch_apply(body, pd.pos, pd.val, pd.n, k,
+                                    e->stream) == 0) {
+            pd = Batch::PatchDev();
+            P = Patch();
+          }
+        }
+      }
     }
   }
 
@@ -1122,7 +1149,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   std::vector<sn_dev_tile> htiles;
   /* JIT eligibility: every unskipped batch clean, kinds uniform + stageable
    * (derived here; stored in the DescCache for reuse on hits) */
-  bool jit_ok_b = true, jit_first = true;
+  bool jit_ok_b = true, jit_first = true, jit_any_del = false;
   int jit_kinds[SN_DEV_MAX_COLS] = {0};
   for (auto &b : t->batches) {
     if (hit) break;
@@ -1222,13 +1249,16 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       }
     }
     db.clean = clean ? 1 : 0;
-    if (!clean) jit_ok_b = false;
+    /* deletes alone don't disqualify the JIT (the generated kernel reads
+     * del_bm); nulls or unmaterialized patches on a used column do */
+    if (db.del_bm) jit_any_del = true;
     for (size_t ui = 0; ui < q->used_cols.size() && jit_ok_b; ui++) {
-      int k = db.cols[ui].kind;
+      const sn_dev_col &jc = db.cols[ui];
+      int k = jc.kind;
       bool stageable = k == SN_K_F64 || k == SN_K_I64 || k == SN_K_I32 ||
                        k == SN_K_F32 || k == SN_K_I16 || k == SN_K_DICT16 ||
                        k == SN_K_DICT32;
-      if (!stageable) jit_ok_b = false;
+      if (!stageable || jc.has_nulls || jc.patch_n > 0) jit_ok_b = false;
       else if (jit_first) jit_kinds[ui] = k;
       else if (jit_kinds[ui] != k) jit_ok_b = false;
     }
@@ -1273,6 +1303,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       dc.db_dev = db_dev; dc.tl_dev = tl_dev; dc.ntiles = ntiles;
       dc.rows = q->rows_scanned;
       dc.jit_ok = (jit_ok_b && !jit_first) ? 1 : 0;
+      dc.jit_del = jit_any_del ? 1 : 0;
       memcpy(dc.jit_kinds, jit_kinds, sizeof(jit_kinds));
       t->desc_caches.push_back(dc);
     }
@@ -1307,7 +1338,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         (dp.nslots <= 1 ? dp.naggs <= 12
                         : (dp.nslots <= 8 && dp.naggs <= 6))) {
       const int *jk = hit ? hit->jit_kinds : jit_kinds;
-      jfn = sn_jit_get(e->jit, &dp, jk, dp.nslots, q->na_t);
+      int jdel = hit ? hit->jit_del : (jit_any_del ? 1 : 0);
+      jfn = sn_jit_get(e->jit, &dp, jk, dp.nslots, q->na_t, jdel);
     }
     if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
     int rc = -1;

@@ -124,6 +124,10 @@ int sn_launch_scan_agg(const sn_dev_plan *plan,
                        double *dev_scratch,  /* >= min(ntiles,SN_GRID_CAP) x nv */
                        void *stream);
 
+/* put-time patch materialization into a null-free fixed-width device body */
+int sn_launch_patch_apply(void *body, const int32_t *pos, const double *val,
+                          int n, int kind, void *stream);
+
 /* launches only the partial-fold (k_reduce): scratch[nblocks][nv] -> out.
  * Used by the JIT path, whose scan kernel writes the same scratch rows. */
 int sn_launch_reduce(const double *dev_scratch, int nblocks, int nv,
@@ -136,7 +140,7 @@ int sn_launch_reduce(const double *dev_scratch, int nblocks, int nv,
 void *sn_jit_cache_create(void);
 void sn_jit_cache_destroy(void *cache);
 void *sn_jit_get(void *cache, const sn_dev_plan *p, const int *kinds,
-                 int nslots, int na_t);
+                 int nslots, int na_t, int has_del);
 int sn_jit_launch(void *fn, int grid, const sn_dev_batch *batches,
                   const sn_dev_tile *tiles, int ntiles, double *scratch,
                   const int64_t *jkeys, const int32_t *jpayload,

@@ -78,7 +78,7 @@ static std::string dlit(double v) {
  *   keyless: scratch row [2*na_t+1] = [sums][counts][rowcount]
  *   grouped: scratch row [nslots*(naggs+1)], rowcount at +naggs.         */
 static std::string gen_source(const sn_dev_plan *p, const int *kinds,
-                              int nslots, int na_t) {
+                              int nslots, int na_t, int has_del) {
   const int NC = p->nused;
   const int NA = p->naggs;
   const int grouped = nslots > 1;
@@ -160,6 +160,8 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     const int tile_end = min(tile.row_start + TILE, num_rows);
 )";
   /* hoist body/dict pointers */
+  if (has_del)
+    o += "    const GAS u64 *del = (const GAS u64 *)(u64)b.del_bm;\n";
   for (int c = 0; c < NC; c++) {
     emitf(o, "    const GAS char *body%d = (const GAS char *)(unsigned long long)b.cols[%d].body;\n", c, c);
     if (kinds[c] == SN_K_DICT16 || kinds[c] == SN_K_DICT32)
@@ -266,6 +268,11 @@ __device__ __forceinline__ u64 mix64(u64 x) {
        "      for (int k = 0; k < CHUNK / WG; k++) {\n"
        "        const int r = tid + k * WG;\n"
        "        int ok = r < rows;\n";
+  if (has_del)
+    o += "        if (del) {\n"
+         "          const int gr = base + r;\n"
+         "          ok &= (int)(~(del[(u64)gr >> 6] >> (gr & 63)) & 1ull);\n"
+         "        }\n";
   for (int i = 0; i < p->npreds_d; i++) {
     emitf(o, "        { const double x = sval[%d][r];\n"
              "          ok &= (x >= %s) & (x <= %s); }\n",
@@ -375,10 +382,11 @@ __device__ __forceinline__ u64 mix64(u64 x) {
 /* compile (or fetch) the kernel for this plan; returns NULL on any failure
  * (caller falls back to the interpreted kernels) */
 extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
-                            const int *kinds, int nslots, int na_t) {
+                            const int *kinds, int nslots, int na_t,
+                            int has_del) {
   auto *jc = (JitCache *)cache;
   if (!jc) return nullptr;
-  std::string src = gen_source(p, kinds, nslots, na_t);
+  std::string src = gen_source(p, kinds, nslots, na_t, has_del);
   if (const char *dump = getenv("SN_JIT_DUMP")) {
     if (FILE *f = fopen(dump, "a")) {
       fputs(src.c_str(), f);

@@ -1200,6 +1200,36 @@ __global__ void k_reduce(const double *__restrict__ scratch, int nblocks,
   }
 }
 
+/* put-time patch materialization: write value-only update patches straight
+ * into the (null-free, fixed-width) device body so the batch scans clean.
+ * Values are plain doubles (decode_delta widening); integer bodies take the
+ * round-nearest bits, matching read_general's __double2ll_rn. */
+__global__ void k_patch_apply(void *__restrict__ body,
+                              const int32_t *__restrict__ pos,
+                              const double *__restrict__ val, int n,
+                              int kind) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const int p = as_global(pos)[i];
+  const double v = as_global(val)[i];
+  switch (kind) {
+    case SN_K_F64: ((GAS double *)(uintptr_t)body)[p] = v; break;
+    case SN_K_F32: ((GAS float *)(uintptr_t)body)[p] = (float)v; break;
+    case SN_K_I32: ((GAS int32_t *)(uintptr_t)body)[p] = (int32_t)__double2ll_rn(v); break;
+    case SN_K_I64: ((GAS long long *)(uintptr_t)body)[p] = __double2ll_rn(v); break;
+    case SN_K_I16: ((GAS int16_t *)(uintptr_t)body)[p] = (int16_t)__double2ll_rn(v); break;
+  }
+}
+
+extern "C" int sn_launch_patch_apply(void *body, const int32_t *pos,
+                                     const double *val, int n, int kind,
+                                     void *stream) {
+  if (n <= 0) return 0;
+  hipLaunchKernelGGL(k_patch_apply, dim3((n + 255) / 256), dim3(256), 0,
+                     (hipStream_t)stream, body, pos, val, n, kind);
+  return (int)hipGetLastError();
+}
+
 extern "C" int sn_launch_reduce(const double *dev_scratch, int nblocks,
                                 int nv, double *dev_out, int naggs1,
                                 int out_stride, void *stream) {

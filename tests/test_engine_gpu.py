@@ -439,7 +439,8 @@ def test_jit_engages_on_clean_scans(eng, li_fixture_table):
     q1.rows()
     assert q1.used_jit(), "Q1 over clean batches should run the JIT kernel"
 
-    # dirty one batch: JIT must step aside, result must not change shape
+    # value-only patches materialize into the device body at put time, so
+    # the batch still runs the JIT kernel with the patched values
     n = 40_000
     rng = np.random.default_rng(77)
     f64 = np.round(rng.random(n), 3)
@@ -460,7 +461,39 @@ def test_jit_engages_on_clean_scans(eng, li_fixture_table):
     plan2 = abi.make_plan(table=t2, aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
     qb = eng.query(plan2)
     rb = qb.rows()
-    assert not qb.used_jit(), "patched batch must use the interpreted path"
+    assert qb.used_jit(), "value-only patches materialize; JIT should engage"
     assert rb[0][1][1] == ra[0][1][1]
     exp = f64.sum() - f64[pos].sum() + nv.sum()
     assert abs(rb[0][1][0] - exp) <= 1e-9 * abs(exp)
+
+    # a patch that writes NULL cannot materialize -> interpreted path
+    nvn = np.array([3.25, 0.0])
+    deltan = se.encode_update_delta(abi.T_DOUBLE, pos, n, nvn,
+                                    valid=np.array([1, 0], dtype=np.uint8))
+    t3 = eng.table_define("tjit3", [(abi.T_DOUBLE, True)])
+    eng.batch_put(t3, 0, 0, n,
+                  [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)],
+                  deltas=[(deltan, None)])
+    plan3 = abi.make_plan(table=t3, aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    qc = eng.query(plan3)
+    rcc = qc.rows()
+    assert not qc.used_jit(), "null-writing patch must use the interpreted path"
+    assert rcc[0][1][1] == float(n)      # COUNT(*) keeps the row
+    exp3 = f64.sum() - f64[1] - f64[5] + 3.25
+    assert abs(rcc[0][1][0] - exp3) <= 1e-9 * abs(exp3)
+
+    # deletes alone stay on the JIT path (del_bm honored by generated code)
+    dels = np.arange(0, n, 97, dtype=np.int32)
+    dmask = po.encode_delete(dels, n)
+    t4 = eng.table_define("tjit4", [(abi.T_DOUBLE, False)])
+    eng.batch_put(t4, 0, 0, n,
+                  [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)],
+                  delete_mask=dmask)
+    plan4 = abi.make_plan(table=t4, aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    qd = eng.query(plan4)
+    rd = qd.rows()
+    assert qd.used_jit(), "delete-only batches should stay on the JIT path"
+    keep = np.ones(n, dtype=bool); keep[dels] = False
+    assert rd[0][1][1] == float(keep.sum())
+    exp4 = f64[keep].sum()
+    assert abs(rd[0][1][0] - exp4) <= 1e-9 * abs(exp4)
