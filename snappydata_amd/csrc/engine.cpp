@@ -206,6 +206,9 @@ struct Dim {
   const int64_t *dev_keys = nullptr;
   const int32_t *dev_payload = nullptr;
   int32_t cap_log2 = 0;
+  /* dense payload LUT over [lut_min, lut_max] when the span is small */
+  const int32_t *dev_lut = nullptr;
+  int64_t lut_min = 0, lut_max = -1;
 };
 
 struct sn_engine {
@@ -338,6 +341,21 @@ static int dim_device_table(sn_engine *e, Dim *d) {
   d->dev_payload = (const int32_t *)up(e, hp.data(), cap * 4);
   d->cap_log2 = lg;
   if (!d->dev_keys || !d->dev_payload) return fail(SN_ERR_NOMEM, "dim upload");
+  /* dense LUT when the key span is small (16M entries = 64 MB worst case;
+   * typical dimensions are far smaller and sit in L2): replaces the
+   * dependent open-address chain with one load per probed row */
+  if (!d->keys.empty()) {
+    int64_t mn = d->keys[0], mx = d->keys[0];
+    for (int64_t k : d->keys) { mn = k < mn ? k : mn; mx = k > mx ? k : mx; }
+    int64_t span = mx - mn + 1;
+    if (span > 0 && span <= (1ll << 24)) {
+      std::vector<int32_t> lut((size_t)span, -1);
+      for (size_t i = 0; i < d->keys.size(); i++)
+        lut[(size_t)(d->keys[i] - mn)] = d->attr_gid[i];
+      d->dev_lut = (const int32_t *)up(e, lut.data(), lut.size() * 4);
+      d->lut_min = mn; d->lut_max = mx;
+    }
+  }
   return SN_OK;
 }
 
@@ -1040,6 +1058,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     dp.jcap_log2 = jd->cap_log2;
     dp.jcslot = q->cslot_of_col[plan->join_fact_col];
     dp.jmode = plan->join_mode == SN_JOIN_GROUP ? 1 : 0;
+    dp.jlut = jd->dev_lut;
+    dp.jlut_min = jd->lut_min;
+    dp.jlut_max = jd->lut_max;
   }
   for (int i = 0; i < plan->npreds; i++) {
     const sn_pred &s = plan->preds[i];
@@ -1353,7 +1374,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       int naggs1 = dp.nslots <= 1 ? 0 : dp.naggs + 1;
       rc = sn_jit_launch(jfn, jgrid, (const sn_dev_batch *)db_dev,
                          (const sn_dev_tile *)tl_dev, ntiles, e->scratch,
-                         dp.jkeys, dp.jpayload, e->stream);
+                         dp.jkeys, dp.jpayload, dp.jlut, e->stream);
       if (rc == 0)
         rc = sn_launch_reduce(e->scratch, jgrid, (int)nv, q->dev_out,
                               naggs1, (int)q->out_stride, e->stream);

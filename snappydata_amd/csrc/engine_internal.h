@@ -99,6 +99,11 @@ typedef struct {
   int32_t jcslot;              /* fact key column slot */
   int32_t jmode;               /* 0 semi, 1 group-by-dim-attr */
   int32_t _pad;
+  /* dense probe LUT (built when the dim key span fits 16M entries):
+   * pay = key in [jlut_min, jlut_max] ? jlut[key - jlut_min] : -1.
+   * One dependent load per row instead of the open-address chain. */
+  const int32_t *jlut;
+  int64_t jlut_min, jlut_max;
   sn_dev_pred_d preds_d[8];
   sn_dev_pred_i preds_i[4];
   sn_dev_agg aggs[12];
@@ -144,7 +149,7 @@ void *sn_jit_get(void *cache, const sn_dev_plan *p, const int *kinds,
 int sn_jit_launch(void *fn, int grid, const sn_dev_batch *batches,
                   const sn_dev_tile *tiles, int ntiles, double *scratch,
                   const int64_t *jkeys, const int32_t *jpayload,
-                  void *stream);
+                  const int32_t *jlut, void *stream);
 
 #ifdef __cplusplus
 }

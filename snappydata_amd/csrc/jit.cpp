@@ -121,10 +121,12 @@ __device__ __forceinline__ u64 mix64(u64 x) {
            "              const sn_dev_tile *__restrict__ tiles, int ntiles,\n"
            "              double *__restrict__ out,\n"
            "              const i64 *__restrict__ jkeys_p,\n"
-           "              const int *__restrict__ jpayload_p) {\n"
+           "              const int *__restrict__ jpayload_p,\n"
+           "              const int *__restrict__ jlut_p) {\n"
            "  const GAS i64 *jkeys = (const GAS i64 *)(u64)jkeys_p;\n"
            "  const GAS int *jpayload = (const GAS int *)(u64)jpayload_p;\n"
-           "  (void)jkeys; (void)jpayload;\n");
+           "  const GAS int *jlut = (const GAS int *)(u64)jlut_p;\n"
+           "  (void)jkeys; (void)jpayload; (void)jlut;\n");
   emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
   emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
         grouped ? nslots * (NA + 1) : 2 * na_t + 1);
@@ -291,22 +293,38 @@ __device__ __forceinline__ u64 mix64(u64 x) {
      * i64-ness); empty-slot sentinel = INT64_MIN, same as the interpreted
      * probe_sweep */
     int is_i64 = (p->i64_mask >> p->jcslot) & 1u;
-    emitf(o, "        int pay = -1;\n"
-             "        if (ok) {\n"
-             "          const double jx = sval[%d][r];\n"
-             "          const i64 key = %s;\n"
-             "          unsigned h = (unsigned)mix64((u64)key) & %uu;\n"
-             "          while (true) {\n"
-             "            const i64 k0 = jkeys[h];\n"
-             "            if (k0 == key) { pay = jpayload[h]; break; }\n"
-             "            if (k0 == (i64)0x8000000000000000ll) break;\n"
-             "            h = (h + 1) & %uu;\n"
-             "          }\n"
-             "          ok = pay >= 0;\n"
-             "        }\n",
-          p->jcslot,
-          is_i64 ? "__double_as_longlong(jx)" : "(i64)jx",
-          (1u << p->jcap_log2) - 1, (1u << p->jcap_log2) - 1);
+    if (p->jlut) {
+      /* dense key span: direct payload lookup, bounds as literals */
+      emitf(o, "        int pay = -1;\n"
+               "        if (ok) {\n"
+               "          const double jx = sval[%d][r];\n"
+               "          const i64 key = %s;\n"
+               "          if (key >= %lldll && key <= %lldll)\n"
+               "            pay = jlut[key - %lldll];\n"
+               "          ok = pay >= 0;\n"
+               "        }\n",
+            p->jcslot,
+            is_i64 ? "__double_as_longlong(jx)" : "(i64)jx",
+            (long long)p->jlut_min, (long long)p->jlut_max,
+            (long long)p->jlut_min);
+    } else {
+      emitf(o, "        int pay = -1;\n"
+               "        if (ok) {\n"
+               "          const double jx = sval[%d][r];\n"
+               "          const i64 key = %s;\n"
+               "          unsigned h = (unsigned)mix64((u64)key) & %uu;\n"
+               "          while (true) {\n"
+               "            const i64 k0 = jkeys[h];\n"
+               "            if (k0 == key) { pay = jpayload[h]; break; }\n"
+               "            if (k0 == (i64)0x8000000000000000ll) break;\n"
+               "            h = (h + 1) & %uu;\n"
+               "          }\n"
+               "          ok = pay >= 0;\n"
+               "        }\n",
+            p->jcslot,
+            is_i64 ? "__double_as_longlong(jx)" : "(i64)jx",
+            (1u << p->jcap_log2) - 1, (1u << p->jcap_log2) - 1);
+    }
   }
   if (grouped) {
     if (p->jkeys && p->jmode == 1) {
@@ -435,9 +453,11 @@ extern "C" int sn_jit_launch(void *fn, int grid,
                              const sn_dev_batch *batches,
                              const sn_dev_tile *tiles, int ntiles,
                              double *scratch, const int64_t *jkeys,
-                             const int32_t *jpayload, void *stream) {
+                             const int32_t *jpayload, const int32_t *jlut,
+                             void *stream) {
   void *args[] = { (void *)&batches, (void *)&tiles, (void *)&ntiles,
-                   (void *)&scratch, (void *)&jkeys, (void *)&jpayload };
+                   (void *)&scratch, (void *)&jkeys, (void *)&jpayload,
+                   (void *)&jlut };
   hipError_t e = hipModuleLaunchKernel((hipFunction_t)fn, grid, 1, 1,
                                        256, 1, 1, 0, (hipStream_t)stream,
                                        args, nullptr);

@@ -607,18 +607,25 @@ __device__ __forceinline__ void probe_sweep(const sn_dev_plan *P,
   const int cs = P->jcslot;
   const unsigned mask = (1u << P->jcap_log2) - 1;
   const int is_i64 = (P->i64_mask >> cs) & 1u;
+  const GAS int32_t *lut = (const GAS int32_t *)(uintptr_t)P->jlut;
+  const long long lmin = P->jlut_min, lmax = P->jlut_max;
 #pragma unroll 2
   for (int k = 0; k < CHUNK / WG; k++) {
     const int r = tid + k * WG;
     const double xv = sval[(size_t)cs * CHUNK + r];
     const long long key = is_i64 ? __double_as_longlong(xv) : (long long)xv;
-    unsigned h = (unsigned)mix64((unsigned long long)key) & mask;
     int pay = -1;
-    while (true) {
-      long long k0 = jk[h];
-      if (k0 == key) { pay = jp[h]; break; }
-      if (k0 == LLONG_MIN) break;
-      h = (h + 1) & mask;
+    if (lut) {
+      /* dense span: one dependent load, no hash chain */
+      if (key >= lmin && key <= lmax) pay = lut[key - lmin];
+    } else {
+      unsigned h = (unsigned)mix64((unsigned long long)key) & mask;
+      while (true) {
+        long long k0 = jk[h];
+        if (k0 == key) { pay = jp[h]; break; }
+        if (k0 == LLONG_MIN) break;
+        h = (h + 1) & mask;
+      }
     }
     const uint64_t w = __ballot(pay >= 0);
     if ((tid & 63) == 0) salive[r >> 6] &= w;
