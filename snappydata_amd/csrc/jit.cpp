@@ -265,6 +265,29 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   emit_load("nbase", "        ");
   o += "      }\n";
 
+  /* LUT probe as its own pass: the 4 per-lane loads issue back-to-back
+   * (4-deep MLP) instead of serializing behind each row's predicate check;
+   * out-of-span keys load a clamped (valid) address and are masked after */
+  if (p->jkeys && p->jlut) {
+    int is_i64 = (p->i64_mask >> p->jcslot) & 1u;
+    emitf(o, "      int payv[CHUNK / WG];\n"
+             "#pragma unroll\n"
+             "      for (int k = 0; k < CHUNK / WG; k++) {\n"
+             "        const int r = tid + k * WG;\n"
+             "        const double jx = sval[%d][r];\n"
+             "        const i64 key = %s;\n"
+             "        const int inr = (key >= %lldll) & (key <= %lldll);\n"
+             "        i64 ck = key < %lldll ? %lldll : (key > %lldll ? %lldll : key);\n"
+             "        const int pv = jlut[ck - %lldll];\n"
+             "        payv[k] = inr ? pv : -1;\n"
+             "      }\n",
+          p->jcslot, is_i64 ? "__double_as_longlong(jx)" : "(i64)jx",
+          (long long)p->jlut_min, (long long)p->jlut_max,
+          (long long)p->jlut_min, (long long)p->jlut_min,
+          (long long)p->jlut_max, (long long)p->jlut_max,
+          (long long)p->jlut_min);
+  }
+
   /* fused row pass */
   o += "#pragma unroll 2\n"
        "      for (int k = 0; k < CHUNK / WG; k++) {\n"
@@ -294,19 +317,9 @@ __device__ __forceinline__ u64 mix64(u64 x) {
      * probe_sweep */
     int is_i64 = (p->i64_mask >> p->jcslot) & 1u;
     if (p->jlut) {
-      /* dense key span: direct payload lookup, bounds as literals */
-      emitf(o, "        int pay = -1;\n"
-               "        if (ok) {\n"
-               "          const double jx = sval[%d][r];\n"
-               "          const i64 key = %s;\n"
-               "          if (key >= %lldll && key <= %lldll)\n"
-               "            pay = jlut[key - %lldll];\n"
-               "          ok = pay >= 0;\n"
-               "        }\n",
-            p->jcslot,
-            is_i64 ? "__double_as_longlong(jx)" : "(i64)jx",
-            (long long)p->jlut_min, (long long)p->jlut_max,
-            (long long)p->jlut_min);
+      (void)is_i64;   /* folded into the prefetch pass */
+      o += "        const int pay = payv[k];\n"
+           "        ok &= pay >= 0;\n";
     } else {
       emitf(o, "        int pay = -1;\n"
                "        if (ok) {\n"
