@@ -192,6 +192,8 @@ def main():
                     help="override rows per GPU")
     ap.add_argument("--seed", type=int, default=42)
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--a2a", action="store_true",
+                    help="key-sharded all-to-all for grouped exchange (N>1)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -247,6 +249,18 @@ def main():
                 dist.all_reduce(exchange_buf)
                 merged = exchange_buf.cpu().numpy().view(np.uint8)
                 q.merge_host(np.ascontiguousarray(merged), n, 1)
+            elif args.a2a:
+                # key-sharded RCCL all-to-all: rank r keeps
+                # only groups hashing to shard r; each rank merges the
+                # world blocks it received and holds its final key subset
+                # (the reference's hash-partitioned partial->final shuffle)
+                n = q.partial_bytes()
+                send = torch.from_numpy(
+                    q.partials_sharded(world).reshape(-1)).to(f"cuda:{local_rank}")
+                recv = torch.empty_like(send)
+                dist.all_to_all_single(recv, send)
+                blocks = recv.cpu().numpy()
+                q.merge_host(np.ascontiguousarray(blocks), n, world)
             else:
                 # RCCL all_gather of self-describing grouped partial blocks
                 n = q.partial_bytes()

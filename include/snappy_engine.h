@@ -284,6 +284,12 @@ int32_t   sn_query_used_jit(sn_query *q);
 int64_t sn_query_partial_bytes(sn_query *q);
 /* export this shard's partial block; dst_is_device: 1 = HIP device memory */
 int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_device);
+/* key-sharded split for the grouped all-to-all: write
+ * `world` same-format blocks into dst (world * partial_bytes), block d
+ * holding only the groups whose key hashes to shard d.  After the
+ * all-to-all, each rank merges the world blocks it received (all of which
+ * carry only its keys) with sn_query_merge. */
+int32_t sn_query_partials_sharded(sn_query *q, int32_t world, void *dst);
 /* merge n_blocks partial blocks (stride bytes apart, host memory) into the
  * final result; n_blocks=1 imports an already-all-reduced keyless block */
 int32_t sn_query_merge(sn_query *q, const void *blocks, int64_t stride,

@@ -1564,6 +1564,58 @@ extern "C" int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_devi
   return SN_OK;
 }
 
+/* Split this shard's grouped partials into `world` same-format blocks by
+ * key-hash shard — the key-sharded all-to-all of SURVEY §8(e): rank r keeps
+ * only keys hashing to shard r; every other key's partial travels to its
+ * owner (the reference's partial->final ShuffleExchange with hash
+ * partitioning, SnappyStrategies.scala:560-601).  dst must hold
+ * world * sn_query_partial_bytes(q) bytes; block d goes to rank d.  The
+ * shard hash is FNV over the key bytes — layout-internal, any deterministic
+ * function gives identical results. */
+extern "C" int32_t sn_query_partials_sharded(sn_query *q, int32_t world,
+                                             void *dst) {
+  if (!q || !dst || world <= 0 || world > 1024) return SN_ERR_BADARG;
+  const sn_plan &p = q->plan;
+  if (p.ngroup == 0 && !q->join_group) return SN_ERR_UNSUPPORTED;
+  int rc = sn_query_wait(q);
+  if (rc != SN_OK) return rc;
+  int64_t bb = sn_query_partial_bytes(q);
+  memset(dst, 0, (size_t)bb * world);
+  std::vector<GroupOut> groups;
+  local_groups(q, &groups);
+  int eff_ngroup = q->join_group ? 1 : p.ngroup;
+  std::vector<int32_t> counts(world, 0);
+  for (auto &g : groups) {
+    uint64_t h = 1469598103934665603ull;
+    for (int k = 0; k < eff_ngroup; k++) {
+      if (g.key_null[k]) { h ^= 1; h *= 1099511628211ull; }
+      else
+        for (char c : g.keys[k]) { h ^= (uint8_t)c; h *= 1099511628211ull; }
+      h ^= 0xff; h *= 1099511628211ull;   /* key separator */
+    }
+    int d = (int)(h % (uint64_t)world);
+    uint8_t *bp = (uint8_t *)dst + (int64_t)d * bb;
+    int32_t &n = counts[d];
+    if (n >= SN_MAX_GROUP_SLOTS) return SN_ERR_NOMEM;
+    PartialSlot *slots = (PartialSlot *)(bp + 8);
+    for (int k = 0; k < SN_MAX_GROUPS; k++) {
+      strncpy(slots[n].keys[k], g.keys[k].c_str(), SN_KEY_MAX - 1);
+      slots[n].key_null[k] = g.key_null[k] ? 1 : 0;
+    }
+    memcpy(slots[n].sums, g.sums, sizeof(g.sums));
+    memcpy(slots[n].counts, g.counts, sizeof(g.counts));
+    slots[n].rowcount = g.rowcount;
+    n++;
+  }
+  int32_t cap = SN_MAX_GROUP_SLOTS;
+  for (int d = 0; d < world; d++) {
+    uint8_t *bp = (uint8_t *)dst + (int64_t)d * bb;
+    memcpy(bp, &counts[d], 4);
+    memcpy(bp + 4, &cap, 4);
+  }
+  return SN_OK;
+}
+
 extern "C" int32_t sn_query_merge(sn_query *q, const void *blocks, int64_t stride,
                                   int32_t n_blocks) {
   if (!q || !blocks || n_blocks <= 0) return SN_ERR_BADARG;

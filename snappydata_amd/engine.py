@@ -89,6 +89,8 @@ def lib():
         L.sn_query_partials.argtypes = [C.c_void_p, C.c_void_p, C.c_int32]
         L.sn_query_merge.restype = C.c_int32
         L.sn_query_merge.argtypes = [C.c_void_p, C.c_void_p, C.c_int64, C.c_int32]
+        L.sn_query_partials_sharded.restype = C.c_int32
+        L.sn_query_partials_sharded.argtypes = [C.c_void_p, C.c_int32, C.c_void_p]
         L.sn_ingest_columns.restype = C.c_int64
         L.sn_ingest_columns.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
                                         C.POINTER(abi.SnIngestCol), C.c_int32,
@@ -228,6 +230,15 @@ class Query:
         buf = np.zeros(n, dtype=np.uint8)
         _check(lib().sn_query_partials(self._h, buf.ctypes.data, 0), "partials")
         return buf
+
+    def partials_sharded(self, world):
+        """Key-sharded split for the grouped all-to-all:
+        returns (world, partial_bytes) uint8; row d travels to rank d."""
+        bb = self.partial_bytes()
+        buf = np.zeros(world * bb, dtype=np.uint8)
+        _check(lib().sn_query_partials_sharded(self._h, world, buf.ctypes.data),
+               "partials_sharded")
+        return buf.reshape(world, bb)
 
     def partials_into_device(self, dev_ptr):
         _check(lib().sn_query_partials(self._h, C.c_void_p(dev_ptr), 1), "partials")

@@ -60,3 +60,45 @@ def test_gloo_partial_exchange():
         p.join(timeout=60)
     for rank, status in results:
         assert status == "ok", (rank, status)
+
+
+def _a2a_worker(rank, world, port, q):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        dist.init_process_group("gloo", rank=rank, world_size=world)
+        # key-sharded exchange transport: each rank holds `world` equal-size
+        # destination blocks; the all-to-all delivers column `rank`.  gloo
+        # has no all_to_all, so the CPU test uses the all_gather + slice
+        # emulation bench.py's nccl path replaces with all_to_all_single.
+        bb = 32
+        send = np.zeros((world, bb), dtype=np.uint8)
+        for d in range(world):
+            send[d, 0] = 100 + 10 * rank + d   # marker: src, dst
+        gathered = [torch.zeros(world * bb, dtype=torch.uint8)
+                    for _ in range(world)]
+        dist.all_gather(gathered, torch.from_numpy(send.reshape(-1)))
+        recv = np.concatenate(
+            [gathered[src].numpy().reshape(world, bb)[rank]
+             for src in range(world)])
+        # rank r must hold blocks (src=0,dst=r), (src=1,dst=r), ...
+        for src in range(world):
+            assert recv[src * bb] == 100 + 10 * src + rank
+        dist.destroy_process_group()
+        q.put((rank, "ok"))
+    except Exception as ex:  # pragma: no cover
+        q.put((rank, f"fail: {ex!r}"))
+
+
+def test_gloo_key_sharded_a2a_transport():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29000 + (os.getpid() + 17) % 1000
+    procs = [ctx.Process(target=_a2a_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=120) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, status in results:
+        assert status == "ok", (rank, status)
