@@ -358,11 +358,15 @@ __device__ __forceinline__ u64 mix64(u64 x) {
           dlit(A.m2).c_str(), A.c2);
   }
   if (grouped) {
+    /* m is 0.0 or 1.0; fma(m, va, sum) is bit-identical to the select+add
+     * form (m=0 -> sum exactly, m=1 -> one rounding like the add) but one
+     * VALU op per aggregate instead of two */
     emitf(o, "#pragma unroll\n        for (int s = 0; s < %d; s++) {\n"
-             "          const int ms = ok && slot == s;\n"
-             "          rc[s] += ms ? 1.0 : 0.0;\n", nslots);
+             "          const double m = (ok && slot == s) ? 1.0 : 0.0;\n"
+             "          rc[s] += m;\n", nslots);
     for (int a = 0; a < NA; a++)
-      emitf(o, "          sums[s][%d] += ms ? va%d : 0.0;\n", a, a);
+      emitf(o, "          sums[s][%d] = __builtin_fma(m, va%d, sums[s][%d]);\n",
+            a, a, a);
     o += "        }\n";
   } else {
     for (int a = 0; a < NA; a++) {
