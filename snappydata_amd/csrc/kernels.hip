@@ -921,9 +921,11 @@ __global__ void k_grouped(sn_dev_plan plan,
                             (B2.a2 + B2.m2 * sval[(size_t)B2.c2 * CHUNK + r]);
 #pragma unroll
           for (int s = 0; s < NSLOTS; s++) {
-            const int ms = m && slot == s;
-            sa[s] += ms ? va : 0.0;
-            sb[s] += ms ? vb : 0.0;
+            /* fma(md, x, acc) is bit-identical to the select+add form and
+             * halves the VALU ops per slot */
+            const double md = (m && slot == s) ? 1.0 : 0.0;
+            sa[s] = __builtin_fma(md, va, sa[s]);
+            sb[s] = __builtin_fma(md, vb, sb[s]);
           }
         }
 #pragma unroll
@@ -1145,11 +1147,11 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
         }
 #pragma unroll
         for (int s = 0; s < NSLOTS; s++) {
-          const int ms = m && slot == s;
-          rc[s] += ms ? 1.0 : 0.0;
+          const double md = (m && slot == s) ? 1.0 : 0.0;
+          rc[s] += md;
 #pragma unroll
           for (int a = 0; a < NA; a++)
-            sums[s][a] += ms ? va[a] : 0.0;
+            sums[s][a] = __builtin_fma(md, va[a], sums[s][a]);
         }
       }
       __syncthreads();
