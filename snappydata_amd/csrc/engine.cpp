@@ -1354,10 +1354,16 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
      * the WholeStageCodegen analogue; measured 3x on Q1's shape over the
      * interpreted runtime-plan kernel.  Any miss falls back, still on GPU. */
     void *jfn = nullptr;
-    if (e->jit &&
-        (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first)) &&
-        (dp.nslots <= 1 ? dp.naggs <= 12
-                        : (dp.nslots <= 8 && dp.naggs <= 6))) {
+    bool jit_shape_ok;
+    if (dp.nslots <= 1) jit_shape_ok = dp.naggs <= 12;
+    else if (dp.nslots <= 8) jit_shape_ok = dp.naggs <= 6;
+    else
+      /* LDS-accumulator mode: static shared = LDS image + gacc must fit */
+      jit_shape_ok = dp.nslots <= 1024 &&
+                     (size_t)dp.nused * 8192 +
+                     (size_t)dp.nslots * (dp.naggs + 1) * 8 + 1024 <= 160 * 1024;
+    if (e->jit && jit_shape_ok &&
+        (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first))) {
       const int *jk = hit ? hit->jit_kinds : jit_kinds;
       int jdel = hit ? hit->jit_del : (jit_any_del ? 1 : 0);
       jfn = sn_jit_get(e->jit, &dp, jk, dp.nslots, q->na_t, jdel);
@@ -1371,6 +1377,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         int rounds = (ntiles + SN_GRID_CAP - 1) / SN_GRID_CAP;
         jgrid = (ntiles + rounds - 1) / rounds;
       }
+      /* >16-slot scratch rows are wide; mirror the interpreted grid cap
+       * (scratch was sized with the same bound above) */
+      if (dp.nslots > 16 && jgrid > SN_GRID_BIGSLOT) jgrid = SN_GRID_BIGSLOT;
       int naggs1 = dp.nslots <= 1 ? 0 : dp.naggs + 1;
       rc = sn_jit_launch(jfn, jgrid, (const sn_dev_batch *)db_dev,
                          (const sn_dev_tile *)tl_dev, ntiles, e->scratch,

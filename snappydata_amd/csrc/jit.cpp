@@ -116,7 +116,11 @@ __device__ __forceinline__ u64 mix64(u64 x) {
 )";
   emitf(o, "#define CHUNK %d\n#define WG %d\n#define TILE %d\n",
         1024, 256, SN_TILE_ROWS);
-  emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, 2)\n"
+  /* >8 slots: block-level LDS accumulators (f64 LDS atomics) instead of
+   * per-lane registers — the high-cardinality SHAMap analogue.  Occupancy 1
+   * (the LDS image + accumulator array leave no room for a second group). */
+  const int lds_mode = grouped && nslots > 8;
+  emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, %d)\n"
            "void jit_scan(const sn_dev_batch *__restrict__ batches,\n"
            "              const sn_dev_tile *__restrict__ tiles, int ntiles,\n"
            "              double *__restrict__ out,\n"
@@ -126,14 +130,22 @@ __device__ __forceinline__ u64 mix64(u64 x) {
            "  const GAS i64 *jkeys = (const GAS i64 *)(u64)jkeys_p;\n"
            "  const GAS int *jpayload = (const GAS int *)(u64)jpayload_p;\n"
            "  const GAS int *jlut = (const GAS int *)(u64)jlut_p;\n"
-           "  (void)jkeys; (void)jpayload; (void)jlut;\n");
+           "  (void)jkeys; (void)jpayload; (void)jlut;\n",
+        lds_mode ? 1 : 2);
   emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
-  emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
-        grouped ? nslots * (NA + 1) : 2 * na_t + 1);
+  if (!lds_mode)
+    emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
+          grouped ? nslots * (NA + 1) : 2 * na_t + 1);
   o += "  const int tid = threadIdx.x;\n";
 
   /* accumulators */
-  if (grouped) {
+  if (lds_mode) {
+    /* block-level LDS accumulator, zeroed once, flushed once at the end */
+    emitf(o, "  __shared__ __attribute__((aligned(16))) double gacc[%d];\n"
+             "  for (int i = tid; i < %d; i += WG) gacc[i] = 0.0;\n"
+             "  __syncthreads();\n",
+          nslots * (NA + 1), nslots * (NA + 1));
+  } else if (grouped) {
     emitf(o, "  double sums[%d][%d]; double rc[%d];\n", nslots, NA, nslots);
     emitf(o, "#pragma unroll\n  for (int s = 0; s < %d; s++) { rc[s] = 0;\n"
              "#pragma unroll\n    for (int a = 0; a < %d; a++) sums[s][a] = 0; }\n",
@@ -357,7 +369,14 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emitf(o, " * (%s + %s * sval[%d][r]);\n", dlit(A.a2).c_str(),
           dlit(A.m2).c_str(), A.c2);
   }
-  if (grouped) {
+  if (lds_mode) {
+    emitf(o, "        if (ok) {\n"
+             "          double *row = &gacc[slot * %d];\n"
+             "          atomicAdd(&row[%d], 1.0);\n", NA + 1, NA);
+    for (int a = 0; a < NA; a++)
+      emitf(o, "          atomicAdd(&row[%d], va%d);\n", a, a);
+    o += "        }\n";
+  } else if (grouped) {
     /* m is 0.0 or 1.0; fma(m, va, sum) is bit-identical to the select+add
      * form (m=0 -> sum exactly, m=1 -> one rounding like the add) but one
      * VALU op per aggregate instead of two */
@@ -383,6 +402,14 @@ __device__ __forceinline__ u64 mix64(u64 x) {
 
   /* block reduce into LDS bacc then scratch row */
   int nv = grouped ? nslots * (NA + 1) : 2 * na_t + 1;
+  if (lds_mode) {
+    /* gacc IS the block accumulator — flush it straight to the scratch row */
+    emitf(o, "  __syncthreads();\n"
+             "  for (int i = tid; i < %d; i += WG)\n"
+             "    out[(u64)blockIdx.x * %d + i] = gacc[i];\n"
+             "}\n", nv, nv);
+    return o;
+  }
   emitf(o, "  for (int i = tid; i < %d; i += WG) bacc[i] = 0.0;\n"
            "  __syncthreads();\n", nv);
   if (grouped) {
