@@ -77,6 +77,7 @@ def q1_plan(table):
 WORKLOADS = {
     # name: (plan fn, rows per GPU, algorithmic bytes per row read by the scan)
     "tpch_q6_lineitem_sf10": (q6_plan, 60_000_000, 28),   # ship4+qty8+ep8+disc8
+    "tpch_q6_lineitem_sf100": (q6_plan, 600_000_000, 28),
     "tpch_q1_lineitem_sf100": (q1_plan, 600_000_000, 40),  # +tax8+2x2dict, ship4
     "tpch_q1_lineitem_sf10": (q1_plan, 60_000_000, 40),
     # BASELINE config 5: Q6 scan with ~2% of rows update-patched and ~1%
@@ -192,8 +193,9 @@ def main():
                     help="override rows per GPU")
     ap.add_argument("--seed", type=int, default=42)
     ap.add_argument("--no-cpu-baseline", action="store_true")
-    ap.add_argument("--a2a", action="store_true",
-                    help="key-sharded all-to-all for grouped exchange (N>1)")
+    ap.add_argument("--no-a2a", action="store_true",
+                    help="use all_gather instead of the key-sharded "
+                         "all-to-all for the grouped exchange (N>1)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -249,8 +251,9 @@ def main():
                 dist.all_reduce(exchange_buf)
                 merged = exchange_buf.cpu().numpy().view(np.uint8)
                 q.merge_host(np.ascontiguousarray(merged), n, 1)
-            elif args.a2a:
-                # key-sharded RCCL all-to-all: rank r keeps
+            elif not args.no_a2a:
+                # key-sharded RCCL all-to-all (the default
+                # grouped exchange — BASELINE's 8-GPU Q1 shape): rank r keeps
                 # only groups hashing to shard r; each rank merges the
                 # world blocks it received and holds its final key subset
                 # (the reference's hash-partitioned partial->final shuffle)
