@@ -554,3 +554,84 @@ def test_key_sharded_a2a_split_merge(eng):
     for e, _ in shards:
         e.close()
     full_e.close()
+
+
+def test_edge_all_rows_deleted(eng):
+    """A batch whose every row is deleted: scans to an empty/NULL result
+    (SUM of nothing is NULL, COUNT(*) 0 — Spark semantics)."""
+    n = 10_000
+    f64 = np.arange(n, dtype=np.float64)
+    dmask = po.encode_delete(np.arange(n, dtype=np.int32), n)
+    t = eng.table_define("tdel_all", [(abi.T_DOUBLE, False)])
+    eng.batch_put(t, 0, 0, n, [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)],
+                  delete_mask=dmask)
+    plan = abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    rows = eng.query(plan).rows()
+    assert rows[0][1][1] == 0.0
+    assert rows[0][1][0] is None
+
+
+def test_edge_delta_covers_every_row(eng):
+    """An update delta patching EVERY row (materialized at put): result must
+    reflect only the new values."""
+    n = 20_000
+    rng = np.random.default_rng(13)
+    f64 = rng.random(n)
+    nv = rng.random(n) * 7
+    delta = se.encode_update_delta(abi.T_DOUBLE, np.arange(n, dtype=np.int32),
+                                   n, nv)
+    t = eng.table_define("tdelta_all", [(abi.T_DOUBLE, False)])
+    eng.batch_put(t, 0, 0, n, [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)],
+                  deltas=[(delta, None)])
+    q = eng.query(abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)]),
+                                               ("count", [])]))
+    rows = q.rows()
+    assert q.used_jit()
+    assert rows[0][1][1] == float(n)
+    assert abs(rows[0][1][0] - nv.sum()) <= 1e-9 * abs(nv.sum())
+
+
+def test_edge_delete_wins_over_delta(eng):
+    """A row both updated and deleted: the delete wins (the reference checks
+    the delete bitmap before the delta merge) — engine vs oracle."""
+    n = 5_000
+    rng = np.random.default_rng(14)
+    f64 = rng.random(n)
+    pos = np.array([10, 20, 30], dtype=np.int32)
+    nv = np.array([100.0, 200.0, 300.0])
+    delta = po.encode_delta(po.T_DOUBLE, po.ENC_UNCOMPRESSED, pos, n, nv)
+    dmask = po.encode_delete(np.array([20], dtype=np.int32), n)
+    cols = [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)]
+    t = eng.table_define("tdd2", [(abi.T_DOUBLE, False)])
+    eng.batch_put(t, 0, 0, -n, cols, delete_mask=dmask,
+                  deltas=[(delta, None)])
+    grows = eng.query(abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)]),
+                                                   ("count", [])])).rows()
+    ot = po.OracleTable([po.T_DOUBLE])
+    ot.add_batch(-n, cols, delete_mask=dmask, deltas=[(delta, None)])
+    orows = po.result_rows(ot.query(po.make_plan(
+        aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])))
+    assert grows[0][1][1] == orows[0][1][1] == float(n - 1)
+    assert abs(grows[0][1][0] - orows[0][1][0]) <= 1e-9 * abs(orows[0][1][0])
+
+
+def test_edge_mixed_clean_and_null_patched_batches(eng):
+    """A table mixing clean batches and a null-writing-patched batch must
+    fall back to the interpreted kernels as a whole and stay correct."""
+    n = 30_000
+    rng = np.random.default_rng(15)
+    a, b = rng.random(n), rng.random(n)
+    t = eng.table_define("tmix", [(abi.T_DOUBLE, True)])
+    eng.batch_put(t, 0, 0, n, [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, a)])
+    deltan = se.encode_update_delta(abi.T_DOUBLE, np.array([3], dtype=np.int32),
+                                    n, np.array([0.0]),
+                                    valid=np.array([0], dtype=np.uint8))
+    eng.batch_put(t, 1, 1, n, [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, b)],
+                  deltas=[(deltan, None)])
+    q = eng.query(abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)]),
+                                               ("count", [])]))
+    rows = q.rows()
+    assert not q.used_jit()
+    exp = a.sum() + b.sum() - b[3]
+    assert rows[0][1][1] == float(2 * n)
+    assert abs(rows[0][1][0] - exp) <= 1e-9 * abs(exp)
