@@ -530,6 +530,64 @@ static sno_lz4_fn sno_get_lz4(void) {
   return fn;
 }
 
+/* Raw Snappy block decode (google/snappy format_description.txt — the raw
+ * block format snappy-java emits; reference codec 2,
+ * CompressionCodecId.scala:31).  varint32 uncompressed length, then
+ * literal/copy elements; copies may overlap (byte-by-byte semantics). */
+static int sno_snappy_decompress(const uint8_t *src, int64_t slen,
+                                 uint8_t *dst, uint32_t ulen) {
+  int64_t ip = 0;
+  uint32_t declared = 0;
+  int shift = 0;
+  while (ip < slen) {
+    uint8_t b = src[ip++];
+    declared |= (uint32_t)(b & 0x7fu) << shift;
+    if (!(b & 0x80u)) break;
+    shift += 7;
+    if (shift > 28) return -1;
+  }
+  if (declared != ulen) return -1;
+  uint32_t op = 0;
+  while (ip < slen) {
+    uint8_t tag = src[ip++];
+    if ((tag & 3) == 0) {
+      uint32_t n = (uint32_t)(tag >> 2) + 1;
+      if (n > 60) {
+        int nb = (int)n - 60;
+        if (ip + nb > slen) return -1;
+        n = 0;
+        for (int i = 0; i < nb; i++) n |= (uint32_t)src[ip + i] << (8 * i);
+        n += 1;
+        ip += nb;
+      }
+      if (ip + n > slen || (uint64_t)op + n > ulen) return -1;
+      memcpy(dst + op, src + ip, n);
+      ip += n; op += n;
+    } else {
+      uint32_t n, off;
+      if ((tag & 3) == 1) {
+        if (ip >= slen) return -1;
+        n = ((uint32_t)(tag >> 2) & 7u) + 4;
+        off = ((uint32_t)(tag >> 5) << 8) | src[ip++];
+      } else if ((tag & 3) == 2) {
+        if (ip + 2 > slen) return -1;
+        n = (uint32_t)(tag >> 2) + 1;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
+        ip += 2;
+      } else {
+        if (ip + 4 > slen) return -1;
+        n = (uint32_t)(tag >> 2) + 1;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8) |
+              ((uint32_t)src[ip + 2] << 16) | ((uint32_t)src[ip + 3] << 24);
+        ip += 4;
+      }
+      if (off == 0 || off > op || (uint64_t)op + n > ulen) return -1;
+      for (uint32_t i = 0; i < n; i++) { dst[op] = dst[op - off]; op++; }
+    }
+  }
+  return op == ulen ? 0 : -1;
+}
+
 /* returns malloc'd decompressed buffer (caller frees) or NULL if plain */
 static uint8_t *sno_maybe_decompress(const uint8_t *blob, int64_t len,
                                      int64_t *out_len, int *err) {
@@ -537,12 +595,21 @@ static uint8_t *sno_maybe_decompress(const uint8_t *blob, int64_t len,
   if (len < 8 || rd_i32(blob) >= 0) return NULL;
   int32_t codec = -rd_i32(blob);
   int32_t ulen = rd_i32(blob + 4);
-  if (codec != 1 || ulen <= 0) { *err = SN_ERR_UNSUPPORTED; return NULL; }
-  sno_lz4_fn fn = sno_get_lz4();
-  if (!fn) { *err = SN_ERR_UNSUPPORTED; return NULL; }
+  if ((codec != 1 && codec != 2) || ulen <= 0) {
+    *err = SN_ERR_UNSUPPORTED;
+    return NULL;
+  }
   uint8_t *buf = (uint8_t *)malloc((size_t)ulen);
-  int n = fn((const char *)blob + 8, (char *)buf, (int)(len - 8), ulen);
-  if (n != ulen) { free(buf); *err = SN_ERR_BADFORMAT; return NULL; }
+  if (codec == 1) {
+    sno_lz4_fn fn = sno_get_lz4();
+    if (!fn) { free(buf); *err = SN_ERR_UNSUPPORTED; return NULL; }
+    int n = fn((const char *)blob + 8, (char *)buf, (int)(len - 8), ulen);
+    if (n != ulen) { free(buf); *err = SN_ERR_BADFORMAT; return NULL; }
+  } else {
+    if (sno_snappy_decompress(blob + 8, len - 8, buf, (uint32_t)ulen) != 0) {
+      free(buf); *err = SN_ERR_BADFORMAT; return NULL;
+    }
+  }
   *out_len = ulen;
   return buf;
 }

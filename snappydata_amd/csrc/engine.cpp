@@ -385,6 +385,67 @@ static lz4_decomp_fn get_lz4(void) {
 
 /* if blob is codec-wrapped, decompress into *out and return 1; 0 if plain;
  * negative = error */
+/* Raw Snappy block decompress (format_description.txt of google/snappy —
+ * the raw block format snappy-java's Snappy.compress emits, which the
+ * reference wraps as codec 2, CompressionCodecId.scala:31).  No libsnappy
+ * ships in this image, so the decoder is implemented here: varint32
+ * uncompressed length, then literal (tag&3==0) and copy (1/2/4-byte
+ * offset) elements; copies may overlap and run byte-by-byte. */
+static int snappy_decompress(const uint8_t *src, int64_t slen,
+                             std::vector<uint8_t> *out) {
+  int64_t ip = 0;
+  uint32_t ulen = 0;
+  int shift = 0;
+  while (ip < slen) {
+    uint8_t b = src[ip++];
+    ulen |= (uint32_t)(b & 0x7f) << shift;
+    if (!(b & 0x80)) break;
+    shift += 7;
+    if (shift > 28) return -1;
+  }
+  out->resize(ulen);
+  uint8_t *dst = out->data();
+  uint32_t op = 0;
+  while (ip < slen) {
+    uint8_t tag = src[ip++];
+    if ((tag & 3) == 0) {                        /* literal */
+      uint32_t n = (uint32_t)(tag >> 2) + 1;
+      if (n > 60) {
+        int nb = (int)n - 60;
+        if (ip + nb > slen) return -1;
+        n = 0;
+        for (int i = 0; i < nb; i++) n |= (uint32_t)src[ip + i] << (8 * i);
+        n += 1;
+        ip += nb;
+      }
+      if (ip + n > slen || (uint64_t)op + n > ulen) return -1;
+      memcpy(dst + op, src + ip, n);
+      ip += n; op += n;
+    } else {                                     /* copy */
+      uint32_t n, off;
+      if ((tag & 3) == 1) {
+        if (ip >= slen) return -1;
+        n = ((uint32_t)(tag >> 2) & 7) + 4;
+        off = ((uint32_t)(tag >> 5) << 8) | src[ip++];
+      } else if ((tag & 3) == 2) {
+        if (ip + 2 > slen) return -1;
+        n = (uint32_t)(tag >> 2) + 1;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8);
+        ip += 2;
+      } else {
+        if (ip + 4 > slen) return -1;
+        n = (uint32_t)(tag >> 2) + 1;
+        off = (uint32_t)src[ip] | ((uint32_t)src[ip + 1] << 8) |
+              ((uint32_t)src[ip + 2] << 16) | ((uint32_t)src[ip + 3] << 24);
+        ip += 4;
+      }
+      if (off == 0 || off > op || (uint64_t)op + n > ulen) return -1;
+      for (uint32_t i = 0; i < n; i++) { dst[op] = dst[op - off]; op++; }
+    }
+  }
+  return op == ulen ? 0 : -1;
+}
+
 static int maybe_decompress(const uint8_t *blob, int64_t len,
                             std::vector<uint8_t> *out) {
   if (len < 8) return 0;
@@ -402,7 +463,13 @@ static int maybe_decompress(const uint8_t *blob, int64_t len,
     if (n != ulen) return fail(SN_ERR_BADFORMAT, "LZ4 decompress failed (%d)", n);
     return 1;
   }
-  return fail(SN_ERR_UNSUPPORTED, "codec %d (Snappy) not supported", codec);
+  if (codec == 2) {
+    if (snappy_decompress(blob + 8, len - 8, out) != 0 ||
+        (int64_t)out->size() != ulen)
+      return fail(SN_ERR_BADFORMAT, "Snappy decompress failed");
+    return 1;
+  }
+  return fail(SN_ERR_UNSUPPORTED, "codec %d not supported", codec);
 }
 
 /* ---- blob header parse (PRODUCT-side restatement of
