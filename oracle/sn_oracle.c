@@ -1031,7 +1031,11 @@ static int eval_batch(const sno_table *t, const sno_batch *b, const sn_plan *p,
     for (int i = 0; i < p->ngroup; i++) {
       int c = p->group_cols[i];
       if (val_null[c]) { knull[i] = 1; keys[i][0] = 0; }
-      else {
+      else if (t->dtypes[c] != SN_TYPE_STRING) {
+        /* integer group key: decimal text (SHAMapAccessor serializes the
+         * raw fixed-width key; text keeps the partial-block format shared) */
+        snprintf(keys[i], SN_KEY_MAX, "%lld", (long long)val_i[c]);
+      } else {
         int32_t L = val_slen[c] < SN_KEY_MAX - 1 ? val_slen[c] : SN_KEY_MAX - 1;
         memcpy(keys[i], val_s[c], (size_t)L); keys[i][L] = 0;
       }

@@ -150,6 +150,9 @@ struct Batch {
   };
   std::vector<PatchDev> patch_dev;
   std::vector<Patch> patch_host;      /* kept for dict premultiply at query */
+  std::vector<uint8_t> had_patches;   /* survives materialization: int group
+                                         keys must reject patched columns
+                                         (stats no longer bound the values) */
   /* RLE aux (device): cumulative run ends + values widened to f64 */
   std::vector<const int32_t *> rle_ends_dev;
   std::vector<const double *> rle_vals_dev;
@@ -650,6 +653,7 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
   b.nullpfx_dev.resize(nc, nullptr);
   b.patch_dev.resize(nc);
   b.patch_host.resize(nc);
+  b.had_patches.assign(nc, 0);
   b.rle_ends_dev.resize(nc, nullptr);
   b.rle_vals_dev.resize(nc, nullptr);
   b.rle_n.resize(nc, 0);
@@ -789,6 +793,7 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
         P.isnull.push_back(kv.second.second);
       }
       if (P.pos.empty()) continue;
+      b.had_patches[c] = 1;
       /* device structures */
       std::vector<uint64_t> bm(((size_t)b.num_rows + 63) / 64, 0);
       for (int32_t p : P.pos) if (p >= 0 && p < b.num_rows) bm[p >> 6] |= 1ull << (p & 63);
@@ -937,6 +942,8 @@ struct sn_query {
   std::vector<int32_t> used_cols;       /* cslot -> table col */
   int nslots = 0;
   int g1cap = 0, g2cap = 0;             /* per-group-col slot counts */
+  bool gint[2] = { false, false };      /* integer group key (stats-ranged) */
+  int64_t gmin[2] = { 0, 0 };           /* integer key minimum (slot base) */
   int gnull1 = -1, gnull2 = -1;         /* null slot index per group col (-1: none) */
   bool grouped_nonnull_ok = true;
   /* grouped-mode device aggregate dedup: logical agg -> device sweep index
@@ -1027,8 +1034,15 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (use_col(plan->preds[i].col) < 0) { fail(SN_ERR_BADARG, "bad pred col"); return nullptr; }
   for (int i = 0; i < plan->ngroup; i++) {
     int c = plan->group_cols[i];
-    if (t->schema[c].dtype != SN_TYPE_STRING) {
-      fail(SN_ERR_UNSUPPORTED, "group-by supports dictionary string columns (round-1 path)");
+    sn_type_t gdt = t->schema[c].dtype;
+    bool int_key = gdt == SN_TYPE_INT32 || gdt == SN_TYPE_INT16;
+    if (gdt != SN_TYPE_STRING && !int_key) {
+      fail(SN_ERR_UNSUPPORTED,
+           "group-by supports dictionary string and int32/int16 key columns");
+      return nullptr;
+    }
+    if (int_key && t->schema[c].nullable) {
+      fail(SN_ERR_UNSUPPORTED, "nullable integer group keys not in round-1");
       return nullptr;
     }
     if (use_col(c) < 0) { fail(SN_ERR_BADARG, "bad group col"); return nullptr; }
@@ -1074,16 +1088,65 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   /* group slot space: global dict sizes, plus a null slot only when the
    * schema allows null keys (non-nullable key columns waste no slots —
    * keeps Q1's 3x2 keys in the register-friendly 8x8 kernel) */
+  /* integer group keys: the slot space is the stats-derived value range
+     (the DictionaryOptimizedMapAccessor direct-slot idea applied to ints).
+     Requires every batch to carry valid bounds for the key column and no
+     update patches on it (patches may move values outside the bounds). */
+  auto int_key_span = [&](int c, int64_t *mn_out) -> int64_t {
+    int64_t mn = 0, mx = -1;
+    bool first = true;
+    for (auto &b : t->batches) {
+      if (!b.stats_valid || b.bounds_null[c])
+        return fail(SN_ERR_UNSUPPORTED,
+                    "integer group key col %d needs stats bounds on every batch", c), -1;
+      if (b.had_patches.size() > (size_t)c && b.had_patches[c])
+        return fail(SN_ERR_UNSUPPORTED,
+                    "integer group key col %d has update patches", c), -1;
+      if (first) { mn = b.lo_i[c]; mx = b.hi_i[c]; first = false; }
+      else {
+        mn = b.lo_i[c] < mn ? b.lo_i[c] : mn;
+        mx = b.hi_i[c] > mx ? b.hi_i[c] : mx;
+      }
+    }
+    if (first) { *mn_out = 0; return 1; }   /* empty table */
+    *mn_out = mn;
+    return mx - mn + 1;
+  };
   if (plan->ngroup >= 1) {
     int c0 = plan->group_cols[0];
-    q->g1cap = (int)t->gdict[c0].size() + (t->schema[c0].nullable ? 1 : 0);
+    if (t->schema[c0].dtype != SN_TYPE_STRING) {
+      int64_t span = int_key_span(c0, &q->gmin[0]);
+      if (span < 0) return nullptr;
+      if (span > SN_MAX_GROUP_SLOTS) {
+        fail(SN_ERR_UNSUPPORTED, "integer key span %lld > %d", (long long)span,
+             SN_MAX_GROUP_SLOTS);
+        return nullptr;
+      }
+      q->gint[0] = true;
+      q->g1cap = (int)span;
+    } else {
+      q->g1cap = (int)t->gdict[c0].size() + (t->schema[c0].nullable ? 1 : 0);
+    }
     q->g2cap = 1;
     if (plan->ngroup == 2) {
       int c1 = plan->group_cols[1];
-      q->g2cap = (int)t->gdict[c1].size() + (t->schema[c1].nullable ? 1 : 0);
+      if (t->schema[c1].dtype != SN_TYPE_STRING) {
+        int64_t span = int_key_span(c1, &q->gmin[1]);
+        if (span < 0) return nullptr;
+        if (span > SN_MAX_GROUP_SLOTS) {
+          fail(SN_ERR_UNSUPPORTED, "integer key span %lld > %d",
+               (long long)span, SN_MAX_GROUP_SLOTS);
+          return nullptr;
+        }
+        q->gint[1] = true;
+        q->g2cap = (int)span;
+      } else {
+        q->g2cap = (int)t->gdict[c1].size() + (t->schema[c1].nullable ? 1 : 0);
+      }
     }
-    if (t->schema[c0].nullable) q->gnull1 = q->g1cap - 1;
-    if (plan->ngroup == 2 && t->schema[plan->group_cols[1]].nullable)
+    if (!q->gint[0] && t->schema[c0].nullable) q->gnull1 = q->g1cap - 1;
+    if (plan->ngroup == 2 && !q->gint[1] &&
+        t->schema[plan->group_cols[1]].nullable)
       q->gnull2 = q->g2cap - 1;
     if (q->g1cap == 0) q->g1cap = 1;
     if (q->g2cap == 0) q->g2cap = 1;
@@ -1119,6 +1182,12 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (t->schema[q->used_cols[ui]].dtype == SN_TYPE_INT64)
       dp.i64_mask |= 1u << ui;
   for (int i = 0; i < plan->ngroup; i++) dp.gcol[i] = q->cslot_of_col[plan->group_cols[i]];
+  dp.gmul0 = 1;
+  if (plan->ngroup >= 1 && q->gint[0]) {
+    dp.gbase[0] = q->gmin[0];
+    dp.gmul0 = q->g2cap;       /* dict col0 premultiplies via its dictmap */
+  }
+  if (plan->ngroup == 2 && q->gint[1]) dp.gbase[1] = q->gmin[1];
   if (jd) {
     dp.jkeys = jd->dev_keys;
     dp.jpayload = jd->dev_payload;
@@ -1526,12 +1595,14 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
     } else if (p.ngroup >= 1) {
       int c0 = p.group_cols[0];
       int g1 = s / q->g2cap;
-      if (g1 == q->gnull1) g.key_null[0] = true;
+      if (q->gint[0]) g.keys[0] = std::to_string(q->gmin[0] + g1);
+      else if (g1 == q->gnull1) g.key_null[0] = true;
       else g.keys[0] = t->gdict[c0][g1];
       if (p.ngroup == 2) {
         int c1 = p.group_cols[1];
         int g2 = s % q->g2cap;
-        if (g2 == q->gnull2) g.key_null[1] = true;
+        if (q->gint[1]) g.keys[1] = std::to_string(q->gmin[1] + g2);
+        else if (g2 == q->gnull2) g.key_null[1] = true;
         else g.keys[1] = t->gdict[c1][g2];
       }
     }

@@ -104,6 +104,12 @@ typedef struct {
    * One dependent load per row instead of the open-address chain. */
   const int32_t *jlut;
   int64_t jlut_min, jlut_max;
+  /* group slot formula: slot = ((v0 - gbase[0]) * gmul0) + (v1 - gbase[1]).
+   * Dictionary key columns arrive premultiplied via dictmap (base 0, and
+   * gmul0 folded into the map), integer key columns use their stats-derived
+   * minimum as base and the second key's span as gmul0. */
+  int64_t gbase[2];
+  int32_t gmul0, _pad2;
   sn_dev_pred_d preds_d[8];
   sn_dev_pred_i preds_i[4];
   sn_dev_agg aggs[12];

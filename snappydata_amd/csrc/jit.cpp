@@ -355,9 +355,14 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     if (p->jkeys && p->jmode == 1) {
       o += "        const int slot = pay > 0 ? pay : 0;\n";
     } else {
-      emitf(o, "        int slot = (int)sval[%d][r];\n", p->gcol[0]);
+      /* slot = (v0 - base0)*mul0 + (v1 - base1), all literals; dictionary
+       * keys arrive premultiplied (base 0, mul 1) so this folds to the
+       * plain cast for them */
+      emitf(o, "        int slot = (int)(((i64)sval[%d][r] - %lldll) * %d);\n",
+            p->gcol[0], (long long)p->gbase[0], p->gmul0 ? p->gmul0 : 1);
       if (p->ngroup >= 2)
-        emitf(o, "        slot += (int)sval[%d][r];\n", p->gcol[1]);
+        emitf(o, "        slot += (int)((i64)sval[%d][r] - %lldll);\n",
+              p->gcol[1], (long long)p->gbase[1]);
     }
   }
   for (int a = 0; a < NA; a++) {

@@ -869,8 +869,12 @@ __global__ void k_grouped(sn_dev_plan plan,
         for (int k = 0; k < CHUNK / WG; k++) {
           const int r = tid + k * WG;
           int slot = 0;
-          if (ngroup >= 1) slot = (int)sval[(size_t)gc0 * CHUNK + r];
-          if (ngroup >= 2) slot += (int)sval[(size_t)gc1 * CHUNK + r];
+          if (ngroup >= 1)
+            slot = (int)(((long long)sval[(size_t)gc0 * CHUNK + r] -
+                          P->gbase[0]) * P->gmul0);
+          if (ngroup >= 2)
+            slot += (int)((long long)sval[(size_t)gc1 * CHUNK + r] -
+                          P->gbase[1]);
           sslot[r] = (int16_t)slot;
         }
       }
@@ -1017,8 +1021,12 @@ void k_grouped_lds(sn_dev_plan plan,
         const int m = (int)((w >> (tid & 63)) & 1ull);
         if (!m) continue;
         int slot = 0;
-        if (ngroup >= 1) slot = (int)sval[(size_t)gc0 * CHUNK + r];
-        if (ngroup >= 2) slot += (int)sval[(size_t)gc1 * CHUNK + r];
+        if (ngroup >= 1)
+          slot = (int)(((long long)sval[(size_t)gc0 * CHUNK + r] -
+                        P->gbase[0]) * P->gmul0);
+        if (ngroup >= 2)
+          slot += (int)((long long)sval[(size_t)gc1 * CHUNK + r] -
+                        P->gbase[1]);
         double *row_acc = bacc + (size_t)slot * (naggs + 1);
         for (int a = 0; a < naggs; a++) {
           const sn_dev_agg &A = P->aggs[a];
@@ -1134,8 +1142,12 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
           m = (int)((w >> (tid & 63)) & 1ull);
         }
         int slot = 0;
-        if (ngroup >= 1) slot = (int)sval[(size_t)gc0 * CHUNK + r];
-        if (ngroup >= 2) slot += (int)sval[(size_t)gc1 * CHUNK + r];
+        if (ngroup >= 1)
+          slot = (int)(((long long)sval[(size_t)gc0 * CHUNK + r] -
+                        P->gbase[0]) * P->gmul0);
+        if (ngroup >= 2)
+          slot += (int)((long long)sval[(size_t)gc1 * CHUNK + r] -
+                        P->gbase[1]);
         double va[NA];
 #pragma unroll
         for (int a = 0; a < NA; a++) {
