@@ -1113,6 +1113,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     return mx - mn + 1;
   };
   if (plan->ngroup >= 1) {
+    /* short lock: reads batch stats + global dictionaries, both of which a
+     * concurrent sn_batch_put may be growing (config-5 ingest+scan) */
+    std::lock_guard<std::mutex> gslot(t->mu);
     int c0 = plan->group_cols[0];
     if (t->schema[c0].dtype != SN_TYPE_STRING) {
       int64_t span = int_key_span(c0, &q->gmin[0]);
@@ -1572,6 +1575,10 @@ extern "C" int32_t sn_query_used_jit(sn_query *q) {
 static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
   const sn_plan &p = q->plan;
   Table *t = q->t;
+  /* concurrent ingest may be interning new dictionary entries; group ids
+   * recorded at submit stay valid (dictionaries only grow) but the vector
+   * needs the lock for stability */
+  std::lock_guard<std::mutex> g(t->mu);
   for (int s = 0; s < q->nslots; s++) {
     const double *row = &q->host_out[(size_t)s * q->out_stride];
     double rowcount = row[2 * q->na_t];

@@ -635,3 +635,58 @@ def test_edge_mixed_clean_and_null_patched_batches(eng):
     exp = a.sum() + b.sum() - b[3]
     assert rows[0][1][1] == float(2 * n)
     assert abs(rows[0][1][0] - exp) <= 1e-9 * abs(exp)
+
+
+def test_concurrent_ingest_and_scan(eng):
+    """BASELINE config 5's 'concurrent' part: one thread keeps ingesting
+    batches while another queries; every result must be a consistent
+    prefix of the batches (rowcount never decreases, sums match the
+    prefix expectation) and the final query sees everything."""
+    import threading
+    n_batches, rows = 40, 20_000
+    rng = np.random.default_rng(17)
+    data = [np.round(rng.random(rows), 3) for _ in range(n_batches)]
+    t = eng.table_define("tconc", [(abi.T_DOUBLE, False)])
+    plan = abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    stop = threading.Event()
+    errs = []
+
+    def ingest():
+        try:
+            for bi in range(n_batches):
+                eng.batch_put(t, bi, bi, rows,
+                              [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED,
+                                         data[bi])])
+        except Exception as ex:  # pragma: no cover
+            errs.append(ex)
+        finally:
+            stop.set()
+
+    prefix_sums = np.cumsum([d.sum() for d in data])
+    seen = []
+
+    def scan():
+        try:
+            while not stop.is_set() or not seen:
+                res = eng.query(plan).rows()
+                cnt = res[0][1][1]
+                if cnt == 0.0:
+                    continue
+                k = int(cnt // rows)
+                assert cnt == k * float(rows), "partial batch visible"
+                assert abs(res[0][1][0] - prefix_sums[k - 1]) <= \
+                    1e-9 * prefix_sums[k - 1]
+                if seen and k < seen[-1]:
+                    raise AssertionError("rowcount went backwards")
+                seen.append(k)
+        except Exception as ex:  # pragma: no cover
+            errs.append(ex)
+
+    ti = threading.Thread(target=ingest)
+    ts = threading.Thread(target=scan)
+    ti.start(); ts.start()
+    ti.join(timeout=120); ts.join(timeout=120)
+    assert not errs, errs
+    final = eng.query(plan).rows()
+    assert final[0][1][1] == float(n_batches * rows)
+    assert abs(final[0][1][0] - prefix_sums[-1]) <= 1e-9 * prefix_sums[-1]
