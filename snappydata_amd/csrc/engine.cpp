@@ -1162,9 +1162,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     }
   } else if (q->join_group) {
     q->nslots = (int)q->join_dim->attr_dict.size();
-    if (q->nslots > 16) {
-      fail(SN_ERR_UNSUPPORTED, "dim attr cardinality %d > round-1 slot limit 16",
-           q->nslots);
+    if (q->nslots > SN_MAX_GROUP_SLOTS) {
+      fail(SN_ERR_UNSUPPORTED, "dim attr cardinality %d > %d",
+           q->nslots, SN_MAX_GROUP_SLOTS);
       return nullptr;
     }
   } else {

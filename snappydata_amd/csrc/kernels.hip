@@ -986,7 +986,8 @@ void k_grouped_lds(sn_dev_plan plan,
   uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
   uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
   uint64_t *salive = sdead + CHUNK / 64;
-  sn_dev_plan *P = (sn_dev_plan *)(salive + CHUNK / 64 + 2);
+  int16_t *sslot = (int16_t *)(salive + CHUNK / 64 + 2);   /* join-group */
+  sn_dev_plan *P = (sn_dev_plan *)((char *)sslot + CHUNK * 2);
   double *bacc = (double *)(P + 1);   /* [nslots][naggs+1] */
   {
     const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
@@ -1011,7 +1012,8 @@ void k_grouped_lds(sn_dev_plan plan,
       __syncthreads();
       alive_init(salive, sdead, rows, clean);
       pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
-      if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
+      const int jslot = plan.jkeys && plan.jmode == 1;
+      if (plan.jkeys) probe_sweep(P, sval, salive, jslot ? sslot : nullptr);
 
 #pragma unroll 2
       for (int k = 0; k < CHUNK / WG; k++) {
@@ -1021,12 +1023,15 @@ void k_grouped_lds(sn_dev_plan plan,
         const int m = (int)((w >> (tid & 63)) & 1ull);
         if (!m) continue;
         int slot = 0;
-        if (ngroup >= 1)
-          slot = (int)(((long long)sval[(size_t)gc0 * CHUNK + r] -
-                        P->gbase[0]) * P->gmul0);
-        if (ngroup >= 2)
-          slot += (int)((long long)sval[(size_t)gc1 * CHUNK + r] -
-                        P->gbase[1]);
+        if (jslot) slot = sslot[r];
+        else {
+          if (ngroup >= 1)
+            slot = (int)(((long long)sval[(size_t)gc0 * CHUNK + r] -
+                          P->gbase[0]) * P->gmul0);
+          if (ngroup >= 2)
+            slot += (int)((long long)sval[(size_t)gc1 * CHUNK + r] -
+                          P->gbase[1]);
+        }
         double *row_acc = bacc + (size_t)slot * (naggs + 1);
         for (int a = 0; a < naggs; a++) {
           const sn_dev_agg &A = P->aggs[a];
@@ -1301,7 +1306,7 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
     /* large-cardinality LDS hash-aggregate path: LDS accumulator bounds the
      * grid so scratch rows stay small */
     if (grid > SN_GRID_BIGSLOT) grid = SN_GRID_BIGSLOT;
-    lds += (CHUNK / 64) * 8 + (size_t)ns * (na + 1) * 8 + 64;
+    lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)ns * (na + 1) * 8 + 64;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     hipLaunchKernelGGL(k_grouped_lds, dim3(grid), dim3(WG), lds, s,
                        *plan, dev_plan, dev_batches, dev_tiles, ntiles,

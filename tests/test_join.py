@@ -129,3 +129,33 @@ class TestJoinGpu:
         for (gk, gv), (ok_, ov) in zip(direct, orows):
             assert gk == ok_ and gv[1] == ov[1]
             assert abs(gv[0] - ov[0]) <= 1e-6 * max(1.0, abs(ov[0]))
+
+
+@pytest.mark.gpu
+def test_group_join_25_nations(setup_factory=None):
+    """TPC-H-realistic attr cardinality: 25 nations (> the 16-slot sweep
+    kernel; routes through the LDS-accumulator paths)."""
+    eng = se.Engine(device=0)
+    keys, measure = make_fact(1_000_000, 20_000)
+    dk = np.arange(20_000, dtype=np.int64)
+    attrs = [b"NATION_%02d" % (int(k) % 25) for k in dk]
+    t = eng.table_define("fact25", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": keys}, {"data": measure}],
+                       len(keys), batch_rows=200_000)
+    dim = eng.dim_define("nation25")
+    eng.dim_put(dim, dk, attrs)
+    plan = abi.make_plan(table=t, aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])],
+                         join=dict(dim=dim, fact_col=0, group=True))
+    rows = eng.query(plan).rows()
+    attr_of = {int(k): a.decode() for k, a in zip(dk, attrs)}
+    exp = {}
+    for k, v in zip(keys.tolist(), measure):
+        a = attr_of[k]
+        s, c = exp.get(a, (0.0, 0))
+        exp[a] = (s + v, c + 1)
+    assert len(rows) == 25 == len(exp)
+    for (gk,), vals in rows:
+        s, c = exp[gk]
+        assert vals[1] == float(c)
+        assert abs(vals[0] - s) <= 1e-6 * max(1.0, abs(s))
+    eng.close()
