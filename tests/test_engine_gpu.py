@@ -690,3 +690,23 @@ def test_concurrent_ingest_and_scan(eng):
     final = eng.query(plan).rows()
     assert final[0][1][1] == float(n_batches * rows)
     assert abs(final[0][1][0] - prefix_sums[-1]) <= 1e-9 * prefix_sums[-1]
+
+
+def test_partials_into_device_matches_host(eng):
+    """The N>1 keyless exchange writes partials straight into a device
+    buffer (bench all_reduces a torch CUDA tensor) — must equal the host
+    export bit-for-bit."""
+    torch = pytest.importorskip("torch")
+    n = 100_000
+    rng = np.random.default_rng(19)
+    f64 = rng.random(n)
+    t = eng.table_define("tpd", [(abi.T_DOUBLE, False)])
+    eng.batch_put(t, 0, 0, n, [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)])
+    q = eng.query(abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)]),
+                                               ("avg", [(0, 0.0, 1.0)]),
+                                               ("count", [])]))
+    host = q.partials_host()
+    buf = torch.zeros(len(host) // 8, dtype=torch.float64, device="cuda:0")
+    q.partials_into_device(buf.data_ptr())
+    dev = buf.cpu().numpy().view(np.uint8)
+    assert bytes(dev) == bytes(host)
