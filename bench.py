@@ -86,7 +86,21 @@ WORKLOADS = {
     # BASELINE config 4: star schema — column fact (suppkey, extendedprice)
     # joined to a broadcast row-store dimension, group by dim attribute
     "star_join_sf10": (None, 60_000_000, 12),
+    # BASELINE config 1 (the reference's CPU-runnable case): 10M rows
+    # (int32, f64), SELECT SUM(d), COUNT(*) WHERE i > median
+    "config1_sum_where": (None, 10_000_000, 12),
 }
+
+
+def build_config1(eng, t, total_rows, seed, batch_rows=600_000):
+    rng = np.random.default_rng(seed)
+    i32 = rng.integers(0, 10**6, total_rows).astype(np.int32)
+    d = rng.random(total_rows)
+    eng.ingest_columns(t, [{"data": i32}, {"data": d}], total_rows,
+                       batch_rows=batch_rows)
+    k = int(np.median(i32))
+    return abi.make_plan(table=t, preds=[dict(col=0, lo=k, lo_strict=True)],
+                         aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
 
 
 def build_star_join(eng, t, total_rows, seed, batch_rows=2_000_000):
@@ -222,6 +236,9 @@ def main():
     if args.workload == "star_join_sf10":
         t = eng.table_define("fact", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
         plan = build_star_join(eng, t, total_rows, args.seed)
+    elif args.workload == "config1_sum_where":
+        t = eng.table_define("c1", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+        plan = build_config1(eng, t, total_rows, args.seed)
     else:
         t = eng.table_define("lineitem", LINEITEM_SCHEMA)
         if args.workload == "tpch_q6_mut_sf10":
