@@ -170,12 +170,26 @@ def cpu_baseline_leg(workload, seed, target_seconds=10.0):
     from tests import tpch_util as tu
 
     sample_rows = 4_000_000
-    d = se.gen_lineitem_arrays(0, sample_rows, seed)
-    t = po.OracleTable(tu.LINEITEM_DTYPES)
-    for num_rows, cols, stats in tu.encode_lineitem_batches(d, 600_000):
-        t.add_batch(num_rows, cols, stats=stats)
     cores = os.cpu_count()
-    plan = tu.q6_plan() if "q6" in workload else tu.q1_plan()
+    if workload == "config1_sum_where":
+        rng = np.random.default_rng(seed)
+        i32 = rng.integers(0, 10**6, sample_rows).astype(np.int32)
+        dv = rng.random(sample_rows)
+        t = po.OracleTable([po.T_INT32, po.T_DOUBLE])
+        for st in range(0, sample_rows, 600_000):
+            en = min(sample_rows, st + 600_000)
+            t.add_batch(en - st,
+                        [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, i32[st:en]),
+                         po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, dv[st:en])])
+        plan = po.make_plan(preds=[dict(col=0, lo=int(np.median(i32)),
+                                        lo_strict=True)],
+                            aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    else:
+        d = se.gen_lineitem_arrays(0, sample_rows, seed)
+        t = po.OracleTable(tu.LINEITEM_DTYPES)
+        for num_rows, cols, stats in tu.encode_lineitem_batches(d, 600_000):
+            t.add_batch(num_rows, cols, stats=stats)
+        plan = tu.q6_plan() if "q6" in workload else tu.q1_plan()
     # one calibration pass, then enough reps for ~target_seconds
     t0 = time.perf_counter()
     t.query(plan, nthreads=cores)
@@ -344,8 +358,9 @@ def main():
                 "kernel_ms": round(avg_ms, 4),
             }
         cpu_baseline = None
-        if world == 1 and not args.no_cpu_baseline and "lineitem" in args.workload \
-                or world == 1 and not args.no_cpu_baseline and "mut" in args.workload:
+        if world == 1 and not args.no_cpu_baseline and (
+                "lineitem" in args.workload or "mut" in args.workload
+                or args.workload == "config1_sum_where"):
             cpu_baseline = cpu_baseline_leg(args.workload, args.seed)
 
         line = {
