@@ -132,8 +132,14 @@ __device__ __forceinline__ u64 mix64(u64 x) {
            "  const GAS int *jlut = (const GAS int *)(u64)jlut_p;\n"
            "  (void)jkeys; (void)jpayload; (void)jlut;\n",
         lds_mode ? 1 : 2);
+  /* wbin mode: few aggregates over many slots makes the per-slot
+   * select-accumulate VALU-bound (star join: 8 slots x 2 updates vs 2 LDS
+   * atomics per row) — per-wave LDS bins shift the work to the LDS pipe,
+   * which the row phase barely uses. */
+  const int wbin_mode = grouped && !lds_mode && NA <= 2 &&
+                        nslots * (NA + 1) >= 12;
   emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
-  if (!lds_mode)
+  if (!lds_mode && !wbin_mode)
     emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
           grouped ? nslots * (NA + 1) : 2 * na_t + 1);
   o += "  const int tid = threadIdx.x;\n";
@@ -143,6 +149,12 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     /* block-level LDS accumulator, zeroed once, flushed once at the end */
     emitf(o, "  __shared__ __attribute__((aligned(16))) double gacc[%d];\n"
              "  for (int i = tid; i < %d; i += WG) gacc[i] = 0.0;\n"
+             "  __syncthreads();\n",
+          nslots * (NA + 1), nslots * (NA + 1));
+  } else if (wbin_mode) {
+    emitf(o, "  __shared__ __attribute__((aligned(16))) double wbin[4][%d];\n"
+             "  for (int i = tid; i < 4 * %d; i += WG)\n"
+             "    ((double *)wbin)[i] = 0.0;\n"
              "  __syncthreads();\n",
           nslots * (NA + 1), nslots * (NA + 1));
   } else if (grouped) {
@@ -381,6 +393,13 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     for (int a = 0; a < NA; a++)
       emitf(o, "          atomicAdd(&row[%d], va%d);\n", a, a);
     o += "        }\n";
+  } else if (wbin_mode) {
+    emitf(o, "        if (ok) {\n"
+             "          double *row = &wbin[tid >> 6][slot * %d];\n"
+             "          atomicAdd(&row[%d], 1.0);\n", NA + 1, NA);
+    for (int a = 0; a < NA; a++)
+      emitf(o, "          atomicAdd(&row[%d], va%d);\n", a, a);
+    o += "        }\n";
   } else if (grouped) {
     /* m is 0.0 or 1.0; fma(m, va, sum) is bit-identical to the select+add
      * form (m=0 -> sum exactly, m=1 -> one rounding like the add) but one
@@ -412,6 +431,14 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emitf(o, "  __syncthreads();\n"
              "  for (int i = tid; i < %d; i += WG)\n"
              "    out[(u64)blockIdx.x * %d + i] = gacc[i];\n"
+             "}\n", nv, nv);
+    return o;
+  }
+  if (wbin_mode) {
+    emitf(o, "  __syncthreads();\n"
+             "  for (int i = tid; i < %d; i += WG)\n"
+             "    out[(u64)blockIdx.x * %d + i] =\n"
+             "        wbin[0][i] + wbin[1][i] + wbin[2][i] + wbin[3][i];\n"
              "}\n", nv, nv);
     return o;
   }
