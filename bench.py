@@ -169,15 +169,16 @@ def cpu_baseline_leg(workload, seed, target_seconds=10.0):
     from oracle import pyoracle as po
     from tests import tpch_util as tu
 
-    sample_rows = 4_000_000
+    sample_rows = 24_000_000
+    batch_rows = 300_000        # >= one batch per thread, or cores idle
     cores = os.cpu_count()
     if workload == "config1_sum_where":
         rng = np.random.default_rng(seed)
         i32 = rng.integers(0, 10**6, sample_rows).astype(np.int32)
         dv = rng.random(sample_rows)
         t = po.OracleTable([po.T_INT32, po.T_DOUBLE])
-        for st in range(0, sample_rows, 600_000):
-            en = min(sample_rows, st + 600_000)
+        for st in range(0, sample_rows, batch_rows):
+            en = min(sample_rows, st + batch_rows)
             t.add_batch(en - st,
                         [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, i32[st:en]),
                          po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, dv[st:en])])
@@ -187,26 +188,28 @@ def cpu_baseline_leg(workload, seed, target_seconds=10.0):
     else:
         d = se.gen_lineitem_arrays(0, sample_rows, seed)
         t = po.OracleTable(tu.LINEITEM_DTYPES)
-        for num_rows, cols, stats in tu.encode_lineitem_batches(d, 600_000):
+        for num_rows, cols, stats in tu.encode_lineitem_batches(d, batch_rows):
             t.add_batch(num_rows, cols, stats=stats)
         plan = tu.q6_plan() if "q6" in workload else tu.q1_plan()
+    # threads beyond the batch count only add fork/merge overhead
+    nthreads = min(cores, sample_rows // batch_rows)
     # one calibration pass, then enough reps for ~target_seconds
     t0 = time.perf_counter()
-    t.query(plan, nthreads=cores)
+    t.query(plan, nthreads=nthreads)
     t1 = time.perf_counter()
     reps = max(1, int(target_seconds / max(1e-3, t1 - t0)))
     t0 = time.perf_counter()
     for _ in range(reps):
-        t.query(plan, nthreads=cores)
+        t.query(plan, nthreads=nthreads)
     dt = time.perf_counter() - t0
     return {
         "value": sample_rows * reps / dt,
         "unit": "rows/s",
-        "cores": cores,
+        "cores": nthreads,
         "kind": "port",
         "sample": f"{sample_rows} rows x {reps} passes ({dt:.1f}s) of the same "
                   f"synthetic workload through the CPU oracle (OpenMP, "
-                  f"{cores} threads; no JVM available for the reference itself)",
+                  f"{nthreads} of {cores} host threads; no JVM available for the reference itself)",
     }
 
 
