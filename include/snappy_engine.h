@@ -134,7 +134,12 @@ typedef struct sn_query  sn_query;
 #define SN_KEY_MAX      48    /* max bytes of one group-key string */
 
 /* conjunct range predicate on a scanned column:
- * lo OP value OP hi, with each bound optional.                          */
+ * lo OP value OP hi, with each bound optional.
+ * Dictionary string columns take equality only (str_eq/str_len): the
+ * engine resolves the literal to its global dictionary id and compares
+ * ids — the reference's dictionary filter pushdown
+ * (ColumnTableScan + DictionaryOptimizedMapAccessor consume the
+ * dictionary INDEX instead of the decoded string).                      */
 typedef struct {
   int32_t col;          /* scan-table column ordinal */
   int32_t _pad;
@@ -143,6 +148,10 @@ typedef struct {
   uint8_t has_lo, has_hi;
   uint8_t lo_strict, hi_strict;  /* 1: strict inequality */
   uint8_t _pad2[4];
+  const char *str_eq;   /* string equality literal (STRING cols only);
+                           valid for the duration of the submit call */
+  int32_t str_len;
+  int32_t _pad3;
 } sn_pred;
 
 /* one multiplicative factor of an aggregate input: (add + mul * col) */

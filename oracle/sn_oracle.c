@@ -995,6 +995,14 @@ static int eval_batch(const sno_table *t, const sno_batch *b, const sn_plan *p,
       const sn_pred *pr = &p->preds[i];
       int c = pr->col;
       if (val_null[c]) { row_ok = 0; break; }
+      if (pr->str_eq && pr->str_len > 0) {
+        /* dictionary string equality (engine pushes this down to a dict-id
+         * compare; the oracle compares the decoded bytes) */
+        if (t->dtypes[c] != SN_TYPE_STRING || val_slen[c] != pr->str_len ||
+            memcmp(val_s[c], pr->str_eq, (size_t)pr->str_len) != 0)
+          row_ok = 0;
+        continue;
+      }
       int is_d = (t->dtypes[c] == SN_TYPE_DOUBLE || t->dtypes[c] == SN_TYPE_FLOAT);
       if (is_d) {
         double v = val_d[c];

@@ -43,7 +43,9 @@ class SnPred(C.Structure):
                 ("lo_i", C.c_int64), ("hi_i", C.c_int64),
                 ("has_lo", C.c_uint8), ("has_hi", C.c_uint8),
                 ("lo_strict", C.c_uint8), ("hi_strict", C.c_uint8),
-                ("_pad2", C.c_uint8 * 4)]
+                ("_pad2", C.c_uint8 * 4),
+                ("str_eq", C.c_char_p), ("str_len", C.c_int32),
+                ("_pad3", C.c_int32)]
 
 
 class SnFactor(C.Structure):
@@ -105,6 +107,10 @@ def make_plan(table=0, preds=(), group_cols=(), aggs=(), join=None):
                 sp.hi_i = int(pr["hi"]); sp.has_hi = 1
         sp.lo_strict = 1 if pr.get("lo_strict") else 0
         sp.hi_strict = 1 if pr.get("hi_strict") else 0
+        if "eq" in pr:
+            lit = pr["eq"] if isinstance(pr["eq"], bytes) else pr["eq"].encode()
+            sp.str_eq = lit          # ctypes keeps the bytes alive via _objects
+            sp.str_len = len(lit)
     p.ngroup = len(group_cols)
     for i, c in enumerate(group_cols):
         p.group_cols[i] = c

@@ -1051,8 +1051,16 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     }
     return q->cslot_of_col[c];
   };
-  for (int i = 0; i < plan->npreds; i++)
-    if (use_col(plan->preds[i].col) < 0) { fail(SN_ERR_BADARG, "bad pred col"); return nullptr; }
+  for (int i = 0; i < plan->npreds; i++) {
+    const sn_pred &pr = plan->preds[i];
+    if (use_col(pr.col) < 0) { fail(SN_ERR_BADARG, "bad pred col"); return nullptr; }
+    if (t->schema[pr.col].dtype == SN_TYPE_STRING &&
+        (!pr.str_eq || pr.str_len <= 0)) {
+      fail(SN_ERR_UNSUPPORTED,
+           "string predicate supports dictionary equality only (str_eq)");
+      return nullptr;
+    }
+  }
   for (int i = 0; i < plan->ngroup; i++) {
     int c = plan->group_cols[i];
     sn_type_t gdt = t->schema[c].dtype;
@@ -1226,6 +1234,24 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     const sn_pred &s = plan->preds[i];
     sn_type_t dt = t->schema[s.col].dtype;
     int cslot = q->cslot_of_col[s.col];
+    if (dt == SN_TYPE_STRING) {
+      /* dictionary pushdown: literal -> global dict id, compared against
+       * the (possibly premultiplied) id the conversion pass writes */
+      bool is_g2 = plan->ngroup == 2 && plan->group_cols[1] == s.col;
+      int64_t mul = is_g2 ? 1 : q->g2cap;
+      int64_t gid = -1;
+      {
+        std::lock_guard<std::mutex> gl(t->mu);
+        auto it = t->gdict_idx[s.col].find(
+            std::string(s.str_eq, (size_t)s.str_len));
+        if (it != t->gdict_idx[s.col].end()) gid = it->second;
+      }
+      sn_dev_pred_d &d = dp.preds_d[dp.npreds_d++];
+      d.cslot = cslot;
+      if (gid < 0) { d.lo = 1.0; d.hi = 0.0; }   /* literal absent: no rows */
+      else { d.lo = d.hi = (double)(gid * mul); }
+      continue;
+    }
     if (dt == SN_TYPE_INT64) {
       if (dp.npreds_i >= 4) return (fail(SN_ERR_BADARG, "too many int64 preds"), nullptr);
       sn_dev_pred_i &d = dp.preds_i[dp.npreds_i++];
