@@ -1238,7 +1238,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       /* dictionary pushdown: literal -> global dict id, compared against
        * the (possibly premultiplied) id the conversion pass writes */
       bool is_g2 = plan->ngroup == 2 && plan->group_cols[1] == s.col;
-      int64_t mul = is_g2 ? 1 : q->g2cap;
+      int64_t mul = is_g2 ? 1 : (q->g2cap > 0 ? q->g2cap : 1);
       int64_t gid = -1;
       {
         std::lock_guard<std::mutex> gl(t->mu);
@@ -1416,7 +1416,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
           dc.kind = m.type_id == SN_ENC_DICTIONARY ? SN_K_DICT16 : SN_K_DICT32;
           /* build the local->premultiplied-global map for this query */
           bool is_g2 = plan->ngroup == 2 && plan->group_cols[1] == c;
-          int mul = is_g2 ? 1 : q->g2cap;
+          int mul = is_g2 ? 1 : (q->g2cap > 0 ? q->g2cap : 1);
           std::vector<int32_t> map(m.local2global.size() + 1);
           for (size_t i = 0; i < m.local2global.size(); i++)
             map[i] = m.local2global[i] * mul;
@@ -1445,7 +1445,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         if (dt == SN_TYPE_STRING) {
           /* premultiply global ids for this query */
           bool is_g2 = plan->ngroup == 2 && plan->group_cols[1] == c;
-          int mul = is_g2 ? 1 : q->g2cap;
+          int mul = is_g2 ? 1 : (q->g2cap > 0 ? q->g2cap : 1);
           std::vector<double> pv(b.patch_host[c].val.size());
           for (size_t i = 0; i < pv.size(); i++)
             pv[i] = b.patch_host[c].val[i] * mul;
