@@ -120,6 +120,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
    * per-lane registers — the high-cardinality SHAMap analogue.  Occupancy 1
    * (the LDS image + accumulator array leave no room for a second group). */
   const int lds_mode = grouped && nslots > 8;
+  const int wbin_pre = grouped && !lds_mode && NA <= 2 && nslots * (NA + 1) >= 12;
   emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, %d)\n"
            "void jit_scan(const sn_dev_batch *__restrict__ batches,\n"
            "              const sn_dev_tile *__restrict__ tiles, int ntiles,\n"
@@ -131,13 +132,12 @@ __device__ __forceinline__ u64 mix64(u64 x) {
            "  const GAS int *jpayload = (const GAS int *)(u64)jpayload_p;\n"
            "  const GAS int *jlut = (const GAS int *)(u64)jlut_p;\n"
            "  (void)jkeys; (void)jpayload; (void)jlut;\n",
-        lds_mode ? 1 : 2);
+        lds_mode ? 1 : (wbin_pre ? 4 : 2));
   /* wbin mode: few aggregates over many slots makes the per-slot
    * select-accumulate VALU-bound (star join: 8 slots x 2 updates vs 2 LDS
    * atomics per row) — per-wave LDS bins shift the work to the LDS pipe,
    * which the row phase barely uses. */
-  const int wbin_mode = grouped && !lds_mode && NA <= 2 &&
-                        nslots * (NA + 1) >= 12;
+  const int wbin_mode = wbin_pre;
   emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
   if (!lds_mode && !wbin_mode)
     emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
