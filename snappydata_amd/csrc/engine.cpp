@@ -41,6 +41,7 @@
 #include <cstdio>
 #include <cstring>
 #include <map>
+#include <set>
 #include <memory>
 #include <mutex>
 #include <string>
@@ -231,6 +232,9 @@ struct sn_engine {
    * small HIP-event pool (create/destroy costs ~µs per query otherwise) */
   std::map<uint64_t, void *> dp_cache;
   std::vector<hipEvent_t> ev_pool;
+  std::set<sn_query *> live_q;   /* outstanding queries: detached (e=NULL)
+                                    at engine destroy so a later
+                                    sn_query_destroy stays safe */
   std::mutex aux_mu;
   std::mutex mu;
 
@@ -277,6 +281,11 @@ extern "C" sn_engine *sn_engine_create(const sn_config *cfg) {
 
 extern "C" void sn_engine_destroy(sn_engine *e) {
   if (!e) return;
+  {
+    std::lock_guard<std::mutex> g(e->aux_mu);
+    for (sn_query *q : e->live_q) q->e = nullptr;
+    e->live_q.clear();
+  }
   for (hipEvent_t ev : e->ev_pool) (void)hipEventDestroy(ev);
   if (e->jit) sn_jit_cache_destroy(e->jit);
   if (e->stream) (void)hipStreamDestroy(e->stream);
@@ -1600,6 +1609,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       return nullptr;
     }
   }
+  {
+    std::lock_guard<std::mutex> ga(e->aux_mu);
+    e->live_q.insert(q.get());
+  }
   return q.release();
 }
 
@@ -1730,7 +1743,13 @@ extern "C" int32_t sn_query_result(sn_query *q, sn_result *out) {
   return SN_OK;
 }
 
-extern "C" void sn_query_destroy(sn_query *q) { delete q; }
+extern "C" void sn_query_destroy(sn_query *q) {
+  if (q && q->e) {
+    std::lock_guard<std::mutex> g(q->e->aux_mu);
+    q->e->live_q.erase(q);
+  }
+  delete q;
+}
 
 /* ---- partial exchange ---- */
 extern "C" int64_t sn_query_partial_bytes(sn_query *q) {
