@@ -279,13 +279,11 @@ extern "C" sn_engine *sn_engine_create(const sn_config *cfg) {
   return e;
 }
 
+static void sn_detach_queries(sn_engine *e);   /* defined below sn_query */
+
 extern "C" void sn_engine_destroy(sn_engine *e) {
   if (!e) return;
-  {
-    std::lock_guard<std::mutex> g(e->aux_mu);
-    for (sn_query *q : e->live_q) q->e = nullptr;
-    e->live_q.clear();
-  }
+  sn_detach_queries(e);
   for (hipEvent_t ev : e->ev_pool) (void)hipEventDestroy(ev);
   if (e->jit) sn_jit_cache_destroy(e->jit);
   if (e->stream) (void)hipStreamDestroy(e->stream);
@@ -1741,6 +1739,12 @@ extern "C" int32_t sn_query_result(sn_query *q, sn_result *out) {
     out->rows_passed += (int64_t)g.rowcount;
   }
   return SN_OK;
+}
+
+static void sn_detach_queries(sn_engine *e) {
+  std::lock_guard<std::mutex> g(e->aux_mu);
+  for (sn_query *q : e->live_q) q->e = nullptr;
+  e->live_q.clear();
 }
 
 extern "C" void sn_query_destroy(sn_query *q) {
