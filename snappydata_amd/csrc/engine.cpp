@@ -335,8 +335,9 @@ extern "C" int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
       d->attr_gid.push_back(0);
     }
   }
-  /* invalidate the cached device table */
+  /* invalidate the cached device table (and the dense LUT with it) */
   d->dev_keys = nullptr; d->dev_payload = nullptr;
+  d->dev_lut = nullptr; d->lut_min = 0; d->lut_max = -1;
   return SN_OK;
 }
 
