@@ -159,3 +159,30 @@ def test_group_join_25_nations(setup_factory=None):
         assert vals[1] == float(c)
         assert abs(vals[0] - s) <= 1e-6 * max(1.0, abs(s))
     eng.close()
+
+
+@pytest.mark.gpu
+def test_incremental_dim_put_invalidates_tables():
+    """a second dim_put must invalidate both the cached hash table AND the
+    dense LUT (stale-LUT regression test)."""
+    eng = se.Engine(device=0)
+    keys, measure = make_fact(400_000, 5_000)
+    t = eng.table_define("factinc", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": keys}, {"data": measure}],
+                       len(keys), batch_rows=100_000)
+    dim = eng.dim_define("dinc")
+    dk1 = np.arange(0, 2_000, dtype=np.int64)
+    eng.dim_put(dim, dk1, [b"A" for _ in dk1])
+    plan = abi.make_plan(table=t, aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])],
+                         join=dict(dim=dim, fact_col=0))
+    r1 = eng.query(plan).rows()
+    m1 = keys < 2_000
+    assert r1[0][1][1] == float(m1.sum())
+    # extend the dimension: results must now include the new keys
+    dk2 = np.arange(2_000, 3_500, dtype=np.int64)
+    eng.dim_put(dim, dk2, [b"B" for _ in dk2])
+    r2 = eng.query(plan).rows()
+    m2 = keys < 3_500
+    assert r2[0][1][1] == float(m2.sum())
+    assert abs(r2[0][1][0] - measure[m2].sum()) <= 1e-6 * abs(measure[m2].sum())
+    eng.close()
