@@ -172,7 +172,26 @@ def cpu_baseline_leg(workload, seed, target_seconds=10.0):
     sample_rows = 24_000_000
     batch_rows = 300_000        # >= one batch per thread, or cores idle
     cores = os.cpu_count()
-    if workload == "config1_sum_where":
+    if workload == "star_join_sf10":
+        # generic oracle loop (the join keeps the per-row path): smaller
+        # sample so the calibration pass stays bounded
+        sample_rows = 2_400_000
+        rng = np.random.default_rng(seed)
+        keyspace = 100_000
+        keys = rng.integers(0, keyspace, sample_rows).astype(np.int32)
+        ep = rng.random(sample_rows) * 1e5
+        t = po.OracleTable([po.T_INT32, po.T_DOUBLE])
+        for st in range(0, sample_rows, batch_rows):
+            en = min(sample_rows, st + batch_rows)
+            t.add_batch(en - st,
+                        [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, keys[st:en]),
+                         po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, ep[st:en])])
+        dk = np.sort(rng.choice(keyspace, size=keyspace * 2 // 5,
+                                replace=False)).astype(np.int64)
+        t.set_dim(dk, [b"NATION_%d" % (int(k) % 8) for k in dk])
+        plan = po.make_plan(aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])],
+                            join=dict(dim=0, fact_col=0, group=True))
+    elif workload == "config1_sum_where":
         rng = np.random.default_rng(seed)
         i32 = rng.integers(0, 10**6, sample_rows).astype(np.int32)
         dv = rng.random(sample_rows)
@@ -209,7 +228,11 @@ def cpu_baseline_leg(workload, seed, target_seconds=10.0):
         "kind": "port",
         "sample": f"{sample_rows} rows x {reps} passes ({dt:.1f}s) of the same "
                   f"synthetic workload through the CPU oracle (OpenMP, "
-                  f"{nthreads} of {cores} host threads; no JVM available for the reference itself)",
+                  f"{nthreads} of {cores} host threads"
+                  + (", generic per-row loop — the oracle has no "
+                     "codegen-shaped fast path for joins"
+                     if workload == "star_join_sf10" else "")
+                  + "; no JVM available for the reference itself)",
     }
 
 
@@ -363,7 +386,7 @@ def main():
         cpu_baseline = None
         if world == 1 and not args.no_cpu_baseline and (
                 "lineitem" in args.workload or "mut" in args.workload
-                or args.workload == "config1_sum_where"):
+                or args.workload in ("config1_sum_where", "star_join_sf10")):
             cpu_baseline = cpu_baseline_leg(args.workload, args.seed)
 
         line = {
