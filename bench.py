@@ -291,10 +291,13 @@ def main():
 
     exchange_buf = None
 
+    jit_used = {"v": False}
+
     def step():
         q = eng.query(plan)
         q.wait()
         km = q.kernel_ms()
+        jit_used["v"] = q.used_jit()
         if dist is not None:
             if not grouped:
                 # RCCL all_reduce of the fixed-width keyless partial block
@@ -410,6 +413,7 @@ def main():
                 "batch_rows": 600_000,
                 "parallelism": f"bucket-dp{n_gpus}",
                 "result_rows": len(result) if result else 0,
+                "jit": jit_used["v"],   # query-compiled (hipRTC) kernel ran
             },
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
