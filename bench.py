@@ -103,9 +103,13 @@ def build_config1(eng, t, total_rows, seed, batch_rows=600_000):
                          aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
 
 
-def build_star_join(eng, t, total_rows, seed, batch_rows=2_000_000):
+def build_star_join(eng, t, total_rows, seed, batch_rows=2_000_000,
+                    dist=None, rank=0, device="cpu"):
     """Fact (suppkey int32, extendedprice f64) + dimension covering 40% of a
-    100K keyspace with 8 nation attributes (dimension << HashJoinSize)."""
+    100K keyspace with 8 nation attributes (dimension << HashJoinSize).
+    At N>1 rank 0 generates the dimension and BROADCASTS it over RCCL into
+    every rank's HBM — the reference's replicated row-store region handed
+    to each executor once (host region get() -> broadcast)."""
     rng = np.random.default_rng(seed)
     keyspace = 100_000
     for start in range(0, total_rows, batch_rows):
@@ -116,8 +120,20 @@ def build_star_join(eng, t, total_rows, seed, batch_rows=2_000_000):
                            batch_rows=batch_rows,
                            first_bucket=start // batch_rows)
     dim = eng.dim_define("supplier")
-    dk = np.sort(rng.choice(keyspace, size=keyspace * 2 // 5,
-                            replace=False)).astype(np.int64)
+    ndim = keyspace * 2 // 5
+    if dist is None:
+        dk = np.sort(rng.choice(keyspace, size=ndim,
+                                replace=False)).astype(np.int64)
+    else:
+        import torch
+        if rank == 0:
+            dk_np = np.sort(rng.choice(keyspace, size=ndim,
+                                       replace=False)).astype(np.int64)
+            dk_t = torch.from_numpy(dk_np).to(device)
+        else:
+            dk_t = torch.empty(ndim, dtype=torch.int64, device=device)
+        dist.broadcast(dk_t, src=0)
+        dk = dk_t.cpu().numpy()
     attrs = [b"NATION_%d" % (int(k) % 8) for k in dk]
     eng.dim_put(dim, dk, attrs)
     plan = abi.make_plan(table=t,
@@ -275,7 +291,9 @@ def main():
                     n_buckets=max(128, world * 16))
     if args.workload == "star_join_sf10":
         t = eng.table_define("fact", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
-        plan = build_star_join(eng, t, total_rows, args.seed)
+        plan = build_star_join(eng, t, total_rows, args.seed, dist=dist,
+                               rank=rank,
+                               device=f"cuda:{local_rank}" if dist else "cpu")
     elif args.workload == "config1_sum_where":
         t = eng.table_define("c1", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
         plan = build_config1(eng, t, total_rows, args.seed)
