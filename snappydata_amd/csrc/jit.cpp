@@ -164,16 +164,8 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emitf(o, "#pragma unroll\n  for (int a = 0; a < %d; a++) { sums[a] = 0; cnts[a] = 0; }\n", NA);
   }
 
-  const int two_ahead = p->jkeys && p->jlut && kinds[p->jcslot] == SN_K_I32;
-  if (p->jkeys && p->jlut) {
-    if (two_ahead)
-      o += "  __shared__ int spay2[2][CHUNK];\n"
-           "  int scur = 0;\n"
-           "#define spay spay2[scur]\n"
-           "  int2_t kq_0, kq_1;\n";
-    else
-      o += "  __shared__ int spay[CHUNK];\n";
-  }
+  if (p->jkeys && p->jlut)
+    o += "  __shared__ int spay[CHUNK];\n";
 
   /* staged register buffers: per column, by width class */
   for (int c = 0; c < NC; c++) {
@@ -283,17 +275,10 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   o += "    int staged = 0;\n"
        "    if (tile.row_start + CHUNK <= tile_end) {\n";
   emit_load("tile.row_start", "      ");
-  if (two_ahead)
-    emitf(o, "      if (tile.row_start + 2 * CHUNK <= tile_end) {\n"
-             "        kq_0 = ((const GAS int2_t *)(body%d + (u64)(tile.row_start + CHUNK) * 4))[tid];\n"
-             "        kq_1 = ((const GAS int2_t *)(body%d + (u64)(tile.row_start + CHUNK) * 4))[tid + WG];\n"
-             "      }\n", p->jcslot, p->jcslot);
   o += "      staged = 1;\n    }\n"
        "    for (int base = tile.row_start; base < tile_end; base += CHUNK) {\n"
        "      const int rows = min(CHUNK, tile_end - base);\n"
        "      if (staged) {\n";
-  if (use_spay && two_ahead)
-    o += "        if (base == tile.row_start) {\n";   /* prime chunk 0 */
   if (use_spay) {
     const int jc = p->jcslot;
     const int jk = kinds[jc];
@@ -316,7 +301,6 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emit_lut_probe(b1, "spay[2 * tid + 1]", "        ");
     emit_lut_probe(b2, "spay[2 * (tid + WG)]", "        ");
     emit_lut_probe(b3, "spay[2 * (tid + WG) + 1]", "        ");
-    if (two_ahead) o += "        }\n";
   }
   emit_write("        ");
   o += "      } else {\n"
@@ -348,23 +332,6 @@ __device__ __forceinline__ u64 mix64(u64 x) {
        "      if (next_staged) {\n";
   emit_load("nbase", "        ");
   o += "      }\n";
-  if (two_ahead) {
-    /* gather the NEXT chunk's payloads from the dedicated key registers
-     * (loaded a full iteration ago, so no vmcnt stall) into the other spay
-     * buffer — the gathers then have this chunk's whole row pass to land */
-    o += "      int gathered = 0;\n"
-         "      if (staged && next_staged) {\n";
-    emit_lut_probe("(i64)kq_0.x", "spay2[scur ^ 1][2 * tid]", "        ");
-    emit_lut_probe("(i64)kq_0.y", "spay2[scur ^ 1][2 * tid + 1]", "        ");
-    emit_lut_probe("(i64)kq_1.x", "spay2[scur ^ 1][2 * (tid + WG)]", "        ");
-    emit_lut_probe("(i64)kq_1.y", "spay2[scur ^ 1][2 * (tid + WG) + 1]", "        ");
-    o += "        gathered = 1;\n"
-         "      }\n";
-    emitf(o, "      if (staged && base + 3 * CHUNK <= tile_end) {\n"
-             "        kq_0 = ((const GAS int2_t *)(body%d + (u64)(base + 2 * CHUNK) * 4))[tid];\n"
-             "        kq_1 = ((const GAS int2_t *)(body%d + (u64)(base + 2 * CHUNK) * 4))[tid + WG];\n"
-             "      }\n", p->jcslot, p->jcslot);
-  }
 
   /* tail / unstaged chunks: fill spay from the LDS image (lane-local
    * mapping, so no extra barrier before the row pass reads it) */
@@ -490,9 +457,8 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   }
   o += "      }\n"
        "      __syncthreads();\n"
-       "      staged = next_staged;\n";
-  if (two_ahead) o += "      if (gathered) scur ^= 1;\n";
-  o += "    }\n"
+       "      staged = next_staged;\n"
+       "    }\n"
        "  }\n";
 
   /* block reduce into LDS bacc then scratch row */
