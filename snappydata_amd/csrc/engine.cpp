@@ -1590,7 +1590,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       int naggs1 = dp.nslots <= 1 ? 0 : dp.naggs + 1;
       rc = sn_jit_launch(jfn, jgrid, (const sn_dev_batch *)db_dev,
                          (const sn_dev_tile *)tl_dev, ntiles, e->scratch,
-                         dp.jkeys, dp.jpayload, dp.jlut, e->stream);
+                         dp.jkeys, dp.jpayload, dp.jlut,
+                         (const sn_dev_plan *)dp_dev, e->stream);
       if (rc == 0)
         rc = sn_launch_reduce(e->scratch, jgrid, (int)nv, q->dev_out,
                               naggs1, (int)q->out_stride, e->stream);
@@ -1641,6 +1642,12 @@ extern "C" double sn_query_kernel_ms(sn_query *q) {
 /* 1 when the query ran a query-compiled (hipRTC) kernel, 0 interpreted */
 extern "C" int32_t sn_query_used_jit(sn_query *q) {
   return q && q->used_jit ? 1 : 0;
+}
+
+/* number of compiled kernels in this engine's JIT cache (tokenized plan
+ * shapes — different literal values share one entry) */
+extern "C" int32_t sn_engine_jit_count(sn_engine *e) {
+  return e ? sn_jit_cache_count(e->jit) : 0;
 }
 
 /* local accumulators -> GroupOut list (pre-merge view) */

@@ -150,12 +150,14 @@ int sn_launch_reduce(const double *dev_scratch, int nblocks, int nv,
  * NULL -> caller falls back to the interpreted kernels. */
 void *sn_jit_cache_create(void);
 void sn_jit_cache_destroy(void *cache);
+int sn_jit_cache_count(void *cache);
 void *sn_jit_get(void *cache, const sn_dev_plan *p, const int *kinds,
                  int nslots, int na_t, int has_del);
 int sn_jit_launch(void *fn, int grid, const sn_dev_batch *batches,
                   const sn_dev_tile *tiles, int ntiles, double *scratch,
                   const int64_t *jkeys, const int32_t *jpayload,
-                  const int32_t *jlut, void *stream);
+                  const int32_t *jlut, const sn_dev_plan *plan_dev,
+                  void *stream);
 
 #ifdef __cplusplus
 }

@@ -58,14 +58,6 @@ static void emitf(std::string &o, const char *fmt, ...) {
   o += buf;
 }
 
-/* double literal that round-trips exactly */
-static std::string dlit(double v) {
-  char b[64];
-  snprintf(b, sizeof(b), "__longlong_as_double(%lldll)",
-           (long long)*(long long *)&v);
-  return b;
-}
-
 /* Generate the specialized kernel source.
  * kinds[c]: SN_K_* per used column slot (uniform across batches).
  * Layout contract identical to the interpreted kernels:
@@ -96,6 +88,20 @@ struct sn_dev_batch { int num_rows; int clean; const u64 *del_bm; sn_dev_col col
   emitf(o, "%d", SN_DEV_MAX_COLS);
   o += R"(]; };
 struct sn_dev_tile { int batch; int row_start; };
+struct sn_dev_pred_d { double lo, hi; int cslot, _p; };
+struct sn_dev_pred_i { i64 lo, hi; int cslot, _p; };
+struct sn_dev_agg { double a0, m0, a1, m1, a2, m2; int c0, c1, c2, nf; };
+struct sn_dev_plan {
+  int npreds_d, npreds_i, naggs, ngroup, nslots, nused;
+  unsigned i64_mask; int gcol[2];
+  const i64 *jkeys; const int *jpayload;
+  int jcap_log2, jcslot, jmode, _pad;
+  const int *jlut; i64 jlut_min, jlut_max;
+  i64 gbase[2]; int gmul0, _pad2;
+  sn_dev_pred_d preds_d[8];
+  sn_dev_pred_i preds_i[4];
+  sn_dev_agg aggs[12];
+};
 __device__ __forceinline__ double wsum(double x) {
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) x += __shfl_down(x, off, 64);
@@ -361,15 +367,13 @@ __device__ __forceinline__ u64 mix64(u64 x) {
          "        }\n";
   for (int i = 0; i < p->npreds_d; i++) {
     emitf(o, "        { const double x = sval[%d][r];\n"
-             "          ok &= (x >= %s) & (x <= %s); }\n",
-          p->preds_d[i].cslot, dlit(p->preds_d[i].lo).c_str(),
-          dlit(p->preds_d[i].hi).c_str());
+             "          ok &= (x >= pd%d_lo) & (x <= pd%d_hi); }\n",
+          p->preds_d[i].cslot, i, i);
   }
   for (int i = 0; i < p->npreds_i; i++) {
     emitf(o, "        { const i64 x = __double_as_longlong(sval[%d][r]);\n"
-             "          ok &= (x >= %lldll) & (x <= %lldll); }\n",
-          p->preds_i[i].cslot, (long long)p->preds_i[i].lo,
-          (long long)p->preds_i[i].hi);
+             "          ok &= (x >= pi%d_lo) & (x <= pi%d_hi); }\n",
+          p->preds_i[i].cslot, i, i);
   }
   o += "        if (__popcll(__ballot(ok)) == 0) continue;\n";
   if (p->jkeys) {
@@ -416,12 +420,14 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   }
   for (int a = 0; a < NA; a++) {
     const sn_dev_agg &A = p->aggs[a];
-    emitf(o, "        const double va%d = (%s + %s * sval[%d][r])", a,
-          dlit(A.a0).c_str(), dlit(A.m0).c_str(), A.c0);
-    emitf(o, " * (%s + %s * sval[%d][r])", dlit(A.a1).c_str(),
-          dlit(A.m1).c_str(), A.c1);
-    emitf(o, " * (%s + %s * sval[%d][r]);\n", dlit(A.a2).c_str(),
-          dlit(A.m2).c_str(), A.c2);
+    if (A.nf < 1) { emitf(o, "        const double va%d = 1.0;\n", a); continue; }
+    emitf(o, "        const double va%d = __builtin_fma(ag%d_m0, sval[%d][r], ag%d_a0)",
+          a, a, A.c0, a);
+    if (A.nf >= 2)
+      emitf(o, " * __builtin_fma(ag%d_m1, sval[%d][r], ag%d_a1)", a, A.c1, a);
+    if (A.nf >= 3)
+      emitf(o, " * __builtin_fma(ag%d_m2, sval[%d][r], ag%d_a2)", a, A.c2, a);
+    o += ";\n";
   }
   if (lds_mode) {
     emitf(o, "        if (ok) {\n"
@@ -520,14 +526,23 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
                             int has_del) {
   auto *jc = (JitCache *)cache;
   if (!jc) return nullptr;
-  /* hash the INPUTS, not the generated source: cache hits (every steady-
-   * state query) must not pay the ~tens of µs of source generation */
+  /* hash the plan SHAPE only — predicate bounds and aggregate
+   * coefficients are tokenized (read from the device plan at run time),
+   * so every literal value of the same shape reuses one compiled kernel
+   * (the reference's tokenized plan cache) */
+  sn_dev_plan shape = *p;
+  for (int i = 0; i < 8; i++) { shape.preds_d[i].lo = shape.preds_d[i].hi = 0.0; }
+  for (int i = 0; i < 4; i++) { shape.preds_i[i].lo = shape.preds_i[i].hi = 0; }
+  for (int a = 0; a < 12; a++) {
+    shape.aggs[a].a0 = shape.aggs[a].m0 = shape.aggs[a].a1 = 0.0;
+    shape.aggs[a].m1 = shape.aggs[a].a2 = shape.aggs[a].m2 = 0.0;
+  }
   uint64_t h = 1469598103934665603ull;
   auto mix = [&](const void *d, size_t n) {
     const uint8_t *b = (const uint8_t *)d;
     for (size_t i = 0; i < n; i++) { h ^= b[i]; h *= 1099511628211ull; }
   };
-  mix(p, sizeof(*p));
+  mix(&shape, sizeof(shape));
   mix(kinds, sizeof(int) * SN_DEV_MAX_COLS);
   mix(&nslots, 4); mix(&na_t, 4); mix(&has_del, 4);
   {
@@ -579,15 +594,22 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
   return (void *)jf.fn;
 }
 
+extern "C" int sn_jit_cache_count(void *cache) {
+  auto *jc = (JitCache *)cache;
+  if (!jc) return 0;
+  std::lock_guard<std::mutex> g(jc->mu);
+  return (int)jc->fns.size();
+}
+
 extern "C" int sn_jit_launch(void *fn, int grid,
                              const sn_dev_batch *batches,
                              const sn_dev_tile *tiles, int ntiles,
                              double *scratch, const int64_t *jkeys,
                              const int32_t *jpayload, const int32_t *jlut,
-                             void *stream) {
+                             const sn_dev_plan *plan_dev, void *stream) {
   void *args[] = { (void *)&batches, (void *)&tiles, (void *)&ntiles,
                    (void *)&scratch, (void *)&jkeys, (void *)&jpayload,
-                   (void *)&jlut };
+                   (void *)&jlut, (void *)&plan_dev };
   hipError_t e = hipModuleLaunchKernel((hipFunction_t)fn, grid, 1, 1,
                                        256, 1, 1, 0, (hipStream_t)stream,
                                        args, nullptr);

@@ -83,6 +83,8 @@ def lib():
         L.sn_query_kernel_ms.argtypes = [C.c_void_p]
         L.sn_query_used_jit.restype = C.c_int32
         L.sn_query_used_jit.argtypes = [C.c_void_p]
+        L.sn_engine_jit_count.restype = C.c_int32
+        L.sn_engine_jit_count.argtypes = [C.c_void_p]
         L.sn_query_partial_bytes.restype = C.c_int64
         L.sn_query_partial_bytes.argtypes = [C.c_void_p]
         L.sn_query_partials.restype = C.c_int32
@@ -352,6 +354,11 @@ class Engine:
     def datagen_lineitem(self, table, total_rows, seed=42, batch_rows=0, threads=0):
         return _check(lib().sn_datagen_lineitem(self._h, table, total_rows, seed,
                                                 batch_rows, threads), "datagen")
+
+    def jit_count(self):
+        """Compiled kernels in the tokenized plan cache (shapes, not
+        literal values)."""
+        return lib().sn_engine_jit_count(self._h)
 
     def dim_define(self, name):
         return _check(lib().sn_dim_define(self._h, name.encode()), "dim_define")

@@ -715,3 +715,28 @@ def test_partials_into_device_matches_host(eng):
                          C.c_size_t(len(host)), 2) == 0   # DeviceToHost
     hip.hipFree(ptr)
     assert bytes(back) == bytes(host)
+
+
+def test_tokenized_plan_cache_reuse(eng):
+    """The reference tokenizes literals so different predicate values share
+    one generated class (TokenizationTest); here different bounds and
+    aggregate coefficients must reuse ONE compiled kernel per plan shape."""
+    n = 100_000
+    rng = np.random.default_rng(23)
+    i32 = rng.integers(0, 1000, n).astype(np.int32)
+    f64 = rng.random(n)
+    t = eng.table_define("ttok", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": i32}, {"data": f64}], n, batch_rows=50_000)
+    before = eng.jit_count()
+    for k, mul in ((100, 1.0), (500, 2.5), (900, -1.0)):
+        q = eng.query(abi.make_plan(
+            table=t, preds=[dict(col=0, hi=k, hi_strict=True)],
+            aggs=[("sum", [(1, 0.0, mul)]), ("count", [])]))
+        rows = q.rows()
+        assert q.used_jit()
+        m = i32 < k
+        assert rows[0][1][1] == float(m.sum())
+        exp = mul * f64[m].sum()
+        assert abs(rows[0][1][0] - exp) <= 1e-9 * max(1.0, abs(exp))
+    after = eng.jit_count()
+    assert after == before + 1, (before, after)
