@@ -127,12 +127,36 @@ __device__ __forceinline__ u64 mix64(u64 x) {
            "              double *__restrict__ out,\n"
            "              const i64 *__restrict__ jkeys_p,\n"
            "              const int *__restrict__ jpayload_p,\n"
-           "              const int *__restrict__ jlut_p) {\n"
+           "              const int *__restrict__ jlut_p,\n"
+           "              const sn_dev_plan *__restrict__ plan_p) {\n"
            "  const GAS i64 *jkeys = (const GAS i64 *)(u64)jkeys_p;\n"
            "  const GAS int *jpayload = (const GAS int *)(u64)jpayload_p;\n"
            "  const GAS int *jlut = (const GAS int *)(u64)jlut_p;\n"
-           "  (void)jkeys; (void)jpayload; (void)jlut;\n",
+           "  const GAS sn_dev_plan *P = (const GAS sn_dev_plan *)(u64)plan_p;\n"
+           "  (void)jkeys; (void)jpayload; (void)jlut; (void)P;\n",
         lds_mode ? 1 : (wbin_pre ? 4 : 2));
+  /* tokenized plan values (the reference's ParamLiteral tokenization,
+   * TokenizationTest / SnappySession plan cache): predicate bounds and
+   * aggregate coefficients load once per wave from the cached device plan
+   * (uniform scalar loads hoisted to kernel entry), so one compiled
+   * kernel serves every literal value of the same plan SHAPE.  Structure
+   * (counts, columns, neutral factors, group/join/LUT geometry) stays
+   * compile-time. */
+  for (int i = 0; i < p->npreds_d; i++)
+    emitf(o, "  const double pd%d_lo = P->preds_d[%d].lo, pd%d_hi = P->preds_d[%d].hi;\n",
+          i, i, i, i);
+  for (int i = 0; i < p->npreds_i; i++)
+    emitf(o, "  const i64 pi%d_lo = P->preds_i[%d].lo, pi%d_hi = P->preds_i[%d].hi;\n",
+          i, i, i, i);
+  for (int a = 0; a < NA; a++) {
+    const sn_dev_agg &A = p->aggs[a];
+    if (A.nf >= 1)
+      emitf(o, "  const double ag%d_a0 = P->aggs[%d].a0, ag%d_m0 = P->aggs[%d].m0;\n", a, a, a, a);
+    if (A.nf >= 2)
+      emitf(o, "  const double ag%d_a1 = P->aggs[%d].a1, ag%d_m1 = P->aggs[%d].m1;\n", a, a, a, a);
+    if (A.nf >= 3)
+      emitf(o, "  const double ag%d_a2 = P->aggs[%d].a2, ag%d_m2 = P->aggs[%d].m2;\n", a, a, a, a);
+  }
   /* wbin mode: few aggregates over many slots makes the per-slot
    * select-accumulate VALU-bound (star join: 8 slots x 2 updates vs 2 LDS
    * atomics per row) — per-wave LDS bins shift the work to the LDS pipe,
