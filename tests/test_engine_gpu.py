@@ -739,4 +739,6 @@ def test_tokenized_plan_cache_reuse(eng):
         exp = mul * f64[m].sum()
         assert abs(rows[0][1][0] - exp) <= 1e-9 * max(1.0, abs(exp))
     after = eng.jit_count()
-    assert after == before + 1, (before, after)
+    # three literal variants add AT MOST one shape entry (zero when an
+    # earlier test already compiled this shape)
+    assert after - before <= 1, (before, after)
