@@ -148,13 +148,16 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   for (int i = 0; i < p->npreds_i; i++)
     emitf(o, "  const i64 pi%d_lo = P->preds_i[%d].lo, pi%d_hi = P->preds_i[%d].hi;\n",
           i, i, i, i);
+  /* trivial factors (a=0, m=1 — plain column reads, the common case) fold
+   * back to the raw product; their triviality is part of the shape hash */
+  auto ftriv = [](double a, double m) { return a == 0.0 && m == 1.0; };
   for (int a = 0; a < NA; a++) {
     const sn_dev_agg &A = p->aggs[a];
-    if (A.nf >= 1)
+    if (A.nf >= 1 && !ftriv(A.a0, A.m0))
       emitf(o, "  const double ag%d_a0 = P->aggs[%d].a0, ag%d_m0 = P->aggs[%d].m0;\n", a, a, a, a);
-    if (A.nf >= 2)
+    if (A.nf >= 2 && !ftriv(A.a1, A.m1))
       emitf(o, "  const double ag%d_a1 = P->aggs[%d].a1, ag%d_m1 = P->aggs[%d].m1;\n", a, a, a, a);
-    if (A.nf >= 3)
+    if (A.nf >= 3 && !ftriv(A.a2, A.m2))
       emitf(o, "  const double ag%d_a2 = P->aggs[%d].a2, ag%d_m2 = P->aggs[%d].m2;\n", a, a, a, a);
   }
   /* wbin mode: few aggregates over many slots makes the per-slot
@@ -445,12 +448,16 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   for (int a = 0; a < NA; a++) {
     const sn_dev_agg &A = p->aggs[a];
     if (A.nf < 1) { emitf(o, "        const double va%d = 1.0;\n", a); continue; }
-    emitf(o, "        const double va%d = __builtin_fma(ag%d_m0, sval[%d][r], ag%d_a0)",
-          a, a, A.c0, a);
-    if (A.nf >= 2)
-      emitf(o, " * __builtin_fma(ag%d_m1, sval[%d][r], ag%d_a1)", a, A.c1, a);
-    if (A.nf >= 3)
-      emitf(o, " * __builtin_fma(ag%d_m2, sval[%d][r], ag%d_a2)", a, A.c2, a);
+    char t0[80], t1[80], t2[80];
+    if (ftriv(A.a0, A.m0)) snprintf(t0, 80, "sval[%d][r]", A.c0);
+    else snprintf(t0, 80, "__builtin_fma(ag%d_m0, sval[%d][r], ag%d_a0)", a, A.c0, a);
+    if (ftriv(A.a1, A.m1)) snprintf(t1, 80, "sval[%d][r]", A.c1);
+    else snprintf(t1, 80, "__builtin_fma(ag%d_m1, sval[%d][r], ag%d_a1)", a, A.c1, a);
+    if (ftriv(A.a2, A.m2)) snprintf(t2, 80, "sval[%d][r]", A.c2);
+    else snprintf(t2, 80, "__builtin_fma(ag%d_m2, sval[%d][r], ag%d_a2)", a, A.c2, a);
+    emitf(o, "        const double va%d = %s", a, t0);
+    if (A.nf >= 2) emitf(o, " * %s", t1);
+    if (A.nf >= 3) emitf(o, " * %s", t2);
     o += ";\n";
   }
   if (lds_mode) {
@@ -555,11 +562,17 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
    * so every literal value of the same shape reuses one compiled kernel
    * (the reference's tokenized plan cache) */
   sn_dev_plan shape = *p;
+  auto trivm = [](double a, double m) { return (a == 0.0 && m == 1.0) ? 1.0 : 0.0; };
   for (int i = 0; i < 8; i++) { shape.preds_d[i].lo = shape.preds_d[i].hi = 0.0; }
   for (int i = 0; i < 4; i++) { shape.preds_i[i].lo = shape.preds_i[i].hi = 0; }
   for (int a = 0; a < 12; a++) {
-    shape.aggs[a].a0 = shape.aggs[a].m0 = shape.aggs[a].a1 = 0.0;
-    shape.aggs[a].m1 = shape.aggs[a].a2 = shape.aggs[a].m2 = 0.0;
+    /* keep only the per-factor triviality bit (it changes the source) */
+    double t0 = trivm(shape.aggs[a].a0, shape.aggs[a].m0);
+    double t1 = trivm(shape.aggs[a].a1, shape.aggs[a].m1);
+    double t2 = trivm(shape.aggs[a].a2, shape.aggs[a].m2);
+    shape.aggs[a].a0 = t0; shape.aggs[a].m0 = 0.0;
+    shape.aggs[a].a1 = t1; shape.aggs[a].m1 = 0.0;
+    shape.aggs[a].a2 = t2; shape.aggs[a].m2 = 0.0;
   }
   uint64_t h = 1469598103934665603ull;
   auto mix = [&](const void *d, size_t n) {
