@@ -15,8 +15,10 @@
  * join, dictionary-slot or keyless grouping.  Everything else falls back to
  * the interpreted kernels (still GPU — never CPU).
  *
- * Compiled modules cache per engine by source hash (the reference caches
- * generated classes the same way; literal re-tokenization is round-2).
+ * Compiled modules cache per engine by plan SHAPE: predicate bounds and
+ * aggregate coefficients are tokenized (read from the cached device plan
+ * at run time), so different literal values of the same shape reuse one
+ * compiled kernel — the reference's ParamLiteral tokenization.
  */
 #include <hip/hip_runtime.h>
 #include <hip/hiprtc.h>
