@@ -178,20 +178,20 @@ def _pmc_traffic(workload, resident_rows):
         return None
 
 
-def cpu_baseline_leg(workload, seed, target_seconds=10.0):
+def cpu_baseline_leg(workload, seed, target_seconds=10.0, sample_rows=None):
     """Time the CPU oracle (reference-loop restatement, OpenMP over all host
     cores) on a bounded sample of the same workload.  Returns the dict for
     the JSON line."""
     from oracle import pyoracle as po
     from tests import tpch_util as tu
 
-    sample_rows = 24_000_000
+    sample_rows = sample_rows or 24_000_000
     batch_rows = 300_000        # >= one batch per thread, or cores idle
     cores = os.cpu_count()
     if workload == "star_join_sf10":
         # generic oracle loop (the join keeps the per-row path): smaller
         # sample so the calibration pass stays bounded
-        sample_rows = 2_400_000
+        sample_rows = min(sample_rows, 2_400_000)
         rng = np.random.default_rng(seed)
         keyspace = 100_000
         keys = rng.integers(0, keyspace, sample_rows).astype(np.int32)
