@@ -797,6 +797,10 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
     for (int c = 0; c < nc; c++) {
       const sn_buf &d1 = deltas[c * 2], &d2 = deltas[c * 2 + 1];
       if (!d1.data && !d2.data) continue;
+      /* deltas may arrive with a positive row count too (the reference
+       * signals them via the stats row's negative batchCount); either way
+       * the batch must never stats-skip — base bounds don't cover patches */
+      b.has_deltas = true;
       std::vector<int32_t> p1, p2;
       std::vector<double> v1, v2;
       std::vector<uint8_t> n1, n2;

@@ -742,3 +742,31 @@ def test_tokenized_plan_cache_reuse(eng):
     # three literal variants add AT MOST one shape entry (zero when an
     # earlier test already compiled this shape)
     assert after - before <= 1, (before, after)
+
+
+def test_stats_skip_never_drops_patched_batches(eng):
+    """A batch whose update delta moves values INTO the predicate range must
+    not be stats-skipped off the base bounds (has_deltas regression: deltas
+    with a positive row count)."""
+    n = 50_000
+    rng = np.random.default_rng(29)
+    f64 = rng.random(n)          # all < 1.0: base stats say hi < 1.0
+    i32 = rng.integers(0, 100, n).astype(np.int32)
+    pos = np.array([7, 9], dtype=np.int32)
+    nv = np.array([5.0, 6.0])    # patched values OUTSIDE the base bounds
+    delta = se.encode_update_delta(abi.T_DOUBLE, pos, n, nv)
+    t = eng.table_define("tsk", [(abi.T_DOUBLE, False), (abi.T_INT32, False)])
+    stats = po.encode_stats([po.T_DOUBLE, po.T_INT32], n,
+                            [float(f64.min()), int(i32.min())],
+                            [float(f64.max()), int(i32.max())])
+    eng.batch_put(t, 0, 0, n,
+                  [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64),
+                   po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, i32)],
+                  stats=stats, deltas=[(delta, None), (None, None)])
+    # predicate selects ONLY the patched values (> 2.0); base stats would skip
+    q = eng.query(abi.make_plan(table=t,
+                                preds=[dict(col=0, is_double=True, lo=2.0)],
+                                aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])]))
+    rows = q.rows()
+    assert rows[0][1][1] == 2.0
+    assert abs(rows[0][1][0] - 11.0) <= 1e-9 * 11.0
