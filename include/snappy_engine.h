@@ -262,6 +262,15 @@ int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
                    const int64_t *keys, const char *attr_payload,
                    const int32_t *attr_lens);
 
+/* attach the CURRENT (cumulative) mutation state to an existing batch —
+ * the UPDATE/DELETE seam: delete_mask and the per-column delta pairs
+ * replace any previous state for that batch; stats (if given) replace the
+ * stats row, else stats-based skipping is disabled for the batch.
+ * (ColumnDelta.scala:300-301; UpdatedColumnDecoder.scala:69-115) */
+int32_t sn_batch_mutate(sn_engine *e, int32_t table, int64_t uuid,
+                        int32_t bucket_id, const sn_buf *delete_mask,
+                        const sn_buf *deltas, const sn_buf *stats);
+
 /* ---- query plane ---- */
 sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan);
 /* blocks until device work completes; returns status */

@@ -660,6 +660,169 @@ static void *up(sn_engine *e, const void *host, size_t n) {
   return d;
 }
 
+/* ---- batch mutation processing (shared by sn_batch_put and
+ * sn_batch_mutate — the reference writes delete masks and delta blobs to
+ * region entries of an EXISTING batch after UPDATE/DELETE statements;
+ * the ColumnBatchIterator hands the current state per scan) ---- */
+
+static int32_t apply_delete_mask(sn_engine *e, Batch &b,
+                                 const sn_buf *delete_mask) {
+  /* delete mask -> bitmap (ColumnDeleteDecoder.scala:24-55 semantics);
+   * the mask is cumulative, so it REPLACES any previous one */
+  b.has_deletes = false;
+  b.del_bm_dev = nullptr;
+  if (delete_mask && delete_mask->data && delete_mask->len >= 12) {
+    const uint8_t *dm = (const uint8_t *)delete_mask->data;
+    int32_t n = rd_i32(dm + 8);
+    if (n < 0 || 12 + (int64_t)n * 4 > delete_mask->len)
+      return fail(SN_ERR_BADFORMAT, "bad delete mask");
+    if (n > 0) {
+      b.has_deletes = true;
+      std::vector<uint64_t> bm(((size_t)b.num_rows + 63) / 64, 0);
+      for (int32_t i = 0; i < n; i++) {
+        int32_t p = rd_i32(dm + 12 + (int64_t)i * 4);
+        if (p >= 0 && p < b.num_rows) bm[p >> 6] |= 1ull << (p & 63);
+      }
+      b.del_bm_dev = (const uint64_t *)up(e, bm.data(), bm.size() * 8);
+    }
+  }
+
+  return SN_OK;
+}
+
+static int32_t apply_deltas(sn_engine *e, Table *t, Batch &b,
+                            const sn_buf *deltas) {
+  /* update deltas -> host-merged patches (delta1 overrides delta2,
+   * UpdatedColumnDecoder.scala:69-115).  Delta blobs are cumulative for
+   * their batch (the encoder merges upward), so the decoded set REPLACES
+   * any previous patch state; values already materialized into the body
+   * are simply rewritten with the same bytes. */
+  const int nc = (int)t->schema.size();
+  if (deltas) {
+    for (int c = 0; c < nc; c++) {
+      const sn_buf &d1 = deltas[c * 2], &d2 = deltas[c * 2 + 1];
+      if (!d1.data && !d2.data) continue;
+      /* deltas may arrive with a positive row count too (the reference
+       * signals them via the stats row's negative batchCount); either way
+       * the batch must never stats-skip — base bounds don't cover patches */
+      b.has_deltas = true;
+      b.patch_host[c] = Patch();          /* cumulative: replace prior state */
+      b.patch_dev[c] = Batch::PatchDev();
+      std::vector<int32_t> p1, p2;
+      std::vector<double> v1, v2;
+      std::vector<uint8_t> n1, n2;
+      if (d2.data) {
+        int rc = decode_delta((const uint8_t *)d2.data, d2.len, t->schema[c].dtype,
+                              t, c, &p2, &v2, &n2);
+        if (rc != SN_OK) return fail(rc, "delta2 decode col %d", c);
+      }
+      if (d1.data) {
+        int rc = decode_delta((const uint8_t *)d1.data, d1.len, t->schema[c].dtype,
+                              t, c, &p1, &v1, &n1);
+        if (rc != SN_OK) return fail(rc, "delta1 decode col %d", c);
+      }
+      /* merge: start from delta2, override with delta1 */
+      std::map<int32_t, std::pair<double, uint8_t>> merged;
+      for (size_t i = 0; i < p2.size(); i++) merged[p2[i]] = { v2[i], n2[i] };
+      for (size_t i = 0; i < p1.size(); i++) merged[p1[i]] = { v1[i], n1[i] };
+      Patch &P = b.patch_host[c];
+      for (auto &kv : merged) {
+        P.pos.push_back(kv.first);
+        P.val.push_back(kv.second.first);
+        P.isnull.push_back(kv.second.second);
+      }
+      if (P.pos.empty()) continue;
+      b.had_patches[c] = 1;
+      /* device structures */
+      std::vector<uint64_t> bm(((size_t)b.num_rows + 63) / 64, 0);
+      for (int32_t p : P.pos) if (p >= 0 && p < b.num_rows) bm[p >> 6] |= 1ull << (p & 63);
+      std::vector<uint64_t> nbm((P.pos.size() + 63) / 64, 0);
+      bool any_null = false;
+      for (size_t i = 0; i < P.isnull.size(); i++)
+        if (P.isnull[i]) { nbm[i >> 6] |= 1ull << (i & 63); any_null = true; }
+      auto &pd = b.patch_dev[c];
+      pd.n = (int32_t)P.pos.size();
+      pd.bm = (const uint64_t *)up(e, bm.data(), bm.size() * 8);
+      pd.pos = (const int32_t *)up(e, P.pos.data(), P.pos.size() * 4);
+      pd.val = (const double *)up(e, P.val.data(), P.val.size() * 8);
+      pd.nullbm = any_null ? (const uint64_t *)up(e, nbm.data(), nbm.size() * 8) : nullptr;
+
+      /* materialize value-only patches straight into the device body when
+       * the base column is null-free fixed-width: the batch then scans
+       * clean (query-compiled kernels apply).  Scan results are identical
+       * by construction — read_general would hand back exactly these
+       * values; the host blob (sn_table_get_blob) keeps the base bytes. */
+      if (e->has_gpu && !any_null && b.cols[c].num_null_words == 0 &&
+          b.cols[c].type_id == SN_ENC_UNCOMPRESSED) {
+        int k = -1;
+        switch (t->schema[c].dtype) {
+          case SN_TYPE_DOUBLE: k = SN_K_F64; break;
+          case SN_TYPE_FLOAT:  k = SN_K_F32; break;
+          case SN_TYPE_INT32:  k = SN_K_I32; break;
+          case SN_TYPE_INT64:  k = SN_K_I64; break;
+          case SN_TYPE_INT16:  k = SN_K_I16; break;
+          default: break;
+        }
+        if (k >= 0) {
+          void *body = (uint8_t *)(uintptr_t)b.col_dev[c] + b.cols[c].body_off;
+          if (sn_launch_patch_apply(body, pd.pos, pd.val, pd.n, k,
+                                    e->stream) == 0) {
+            pd = Batch::PatchDev();
+            P = Patch();
+          }
+        }
+      }
+    }
+  }
+
+  return SN_OK;
+}
+
+static void apply_stats(Table *t, Batch &b, const sn_buf *stats) {
+  /* stats row parse (UnsafeRow: [null words][3*ncols+1 x 8B slots],
+   * ColumnStatsSchema, ColumnEncoding.scala:1015-1036) */
+  const int nc = (int)t->schema.size();
+  if (stats && stats->data) {
+    const uint8_t *sp = (const uint8_t *)stats->data;
+    int32_t num_fields = nc * 3 + 1;
+    int32_t nwords = (num_fields + 63) >> 6;
+    if (stats->len >= (int64_t)nwords * 8 + (int64_t)num_fields * 8) {
+      const uint8_t *bits = sp;
+      const uint8_t *slots = sp + (int64_t)nwords * 8;
+      auto bit = [&](int f) {
+        return (rd_i64(bits + ((f >> 6) << 3)) >> (f & 63)) & 1;
+      };
+      b.lo_d.resize(nc); b.hi_d.resize(nc);
+      b.lo_i.resize(nc); b.hi_i.resize(nc);
+      b.null_count.resize(nc); b.bounds_null.resize(nc);
+      for (int c = 0; c < nc; c++) {
+        int f_lo = 1 + c * 3, f_hi = 2 + c * 3, f_nc = 3 + c * 3;
+        b.bounds_null[c] = bit(f_lo) || bit(f_hi);
+        b.null_count[c] = bit(f_nc) ? 0 : rd_i32(slots + (int64_t)f_nc * 8);
+        if (b.bounds_null[c]) continue;
+        switch (t->schema[c].dtype) {
+          case SN_TYPE_DOUBLE:
+            b.lo_d[c] = rd_f64(slots + (int64_t)f_lo * 8);
+            b.hi_d[c] = rd_f64(slots + (int64_t)f_hi * 8); break;
+          case SN_TYPE_FLOAT:
+            b.lo_d[c] = rd_f32(slots + (int64_t)f_lo * 8);
+            b.hi_d[c] = rd_f32(slots + (int64_t)f_hi * 8); break;
+          case SN_TYPE_INT64:
+            b.lo_i[c] = rd_i64(slots + (int64_t)f_lo * 8);
+            b.hi_i[c] = rd_i64(slots + (int64_t)f_hi * 8); break;
+          case SN_TYPE_STRING:
+            b.bounds_null[c] = 1; break;
+          default:
+            b.lo_i[c] = rd_i32(slots + (int64_t)f_lo * 8);
+            b.hi_i[c] = rd_i32(slots + (int64_t)f_hi * 8);
+        }
+      }
+      b.stats_valid = true;
+    }
+  }
+
+}
+
 extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
                                 int64_t uuid, int32_t bucket_id, int32_t num_rows,
                                 const sn_buf *columns, const sn_buf *stats,
@@ -774,143 +937,53 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
     }
   }
 
-  /* delete mask -> bitmap (ColumnDeleteDecoder.scala:24-55 semantics) */
-  if (delete_mask && delete_mask->data && delete_mask->len >= 12) {
-    const uint8_t *dm = (const uint8_t *)delete_mask->data;
-    int32_t n = rd_i32(dm + 8);
-    if (n < 0 || 12 + (int64_t)n * 4 > delete_mask->len)
-      return fail(SN_ERR_BADFORMAT, "bad delete mask");
-    if (n > 0) {
-      b.has_deletes = true;
-      std::vector<uint64_t> bm(((size_t)b.num_rows + 63) / 64, 0);
-      for (int32_t i = 0; i < n; i++) {
-        int32_t p = rd_i32(dm + 12 + (int64_t)i * 4);
-        if (p >= 0 && p < b.num_rows) bm[p >> 6] |= 1ull << (p & 63);
-      }
-      b.del_bm_dev = (const uint64_t *)up(e, bm.data(), bm.size() * 8);
-    }
-  }
+  { int32_t rc_ = apply_delete_mask(e, b, delete_mask); if (rc_ != SN_OK) return rc_; }
 
-  /* update deltas -> host-merged patches (delta1 overrides delta2,
-   * UpdatedColumnDecoder.scala:69-115) */
-  if (deltas) {
-    for (int c = 0; c < nc; c++) {
-      const sn_buf &d1 = deltas[c * 2], &d2 = deltas[c * 2 + 1];
-      if (!d1.data && !d2.data) continue;
-      /* deltas may arrive with a positive row count too (the reference
-       * signals them via the stats row's negative batchCount); either way
-       * the batch must never stats-skip — base bounds don't cover patches */
-      b.has_deltas = true;
-      std::vector<int32_t> p1, p2;
-      std::vector<double> v1, v2;
-      std::vector<uint8_t> n1, n2;
-      if (d2.data) {
-        int rc = decode_delta((const uint8_t *)d2.data, d2.len, t->schema[c].dtype,
-                              t, c, &p2, &v2, &n2);
-        if (rc != SN_OK) return fail(rc, "delta2 decode col %d", c);
-      }
-      if (d1.data) {
-        int rc = decode_delta((const uint8_t *)d1.data, d1.len, t->schema[c].dtype,
-                              t, c, &p1, &v1, &n1);
-        if (rc != SN_OK) return fail(rc, "delta1 decode col %d", c);
-      }
-      /* merge: start from delta2, override with delta1 */
-      std::map<int32_t, std::pair<double, uint8_t>> merged;
-      for (size_t i = 0; i < p2.size(); i++) merged[p2[i]] = { v2[i], n2[i] };
-      for (size_t i = 0; i < p1.size(); i++) merged[p1[i]] = { v1[i], n1[i] };
-      Patch &P = b.patch_host[c];
-      for (auto &kv : merged) {
-        P.pos.push_back(kv.first);
-        P.val.push_back(kv.second.first);
-        P.isnull.push_back(kv.second.second);
-      }
-      if (P.pos.empty()) continue;
-      b.had_patches[c] = 1;
-      /* device structures */
-      std::vector<uint64_t> bm(((size_t)b.num_rows + 63) / 64, 0);
-      for (int32_t p : P.pos) if (p >= 0 && p < b.num_rows) bm[p >> 6] |= 1ull << (p & 63);
-      std::vector<uint64_t> nbm((P.pos.size() + 63) / 64, 0);
-      bool any_null = false;
-      for (size_t i = 0; i < P.isnull.size(); i++)
-        if (P.isnull[i]) { nbm[i >> 6] |= 1ull << (i & 63); any_null = true; }
-      auto &pd = b.patch_dev[c];
-      pd.n = (int32_t)P.pos.size();
-      pd.bm = (const uint64_t *)up(e, bm.data(), bm.size() * 8);
-      pd.pos = (const int32_t *)up(e, P.pos.data(), P.pos.size() * 4);
-      pd.val = (const double *)up(e, P.val.data(), P.val.size() * 8);
-      pd.nullbm = any_null ? (const uint64_t *)up(e, nbm.data(), nbm.size() * 8) : nullptr;
+  { int32_t rc_ = apply_deltas(e, t, b, deltas); if (rc_ != SN_OK) return rc_; }
 
-      /* materialize value-only patches straight into the device body when
-       * the base column is null-free fixed-width: the batch then scans
-       * clean (query-compiled kernels apply).  Scan results are identical
-       * by construction — read_general would hand back exactly these
-       * values; the host blob (sn_table_get_blob) keeps the base bytes. */
-      if (e->has_gpu && !any_null && b.cols[c].num_null_words == 0 &&
-          b.cols[c].type_id == SN_ENC_UNCOMPRESSED) {
-        int k = -1;
-        switch (t->schema[c].dtype) {
-          case SN_TYPE_DOUBLE: k = SN_K_F64; break;
-          case SN_TYPE_FLOAT:  k = SN_K_F32; break;
-          case SN_TYPE_INT32:  k = SN_K_I32; break;
-          case SN_TYPE_INT64:  k = SN_K_I64; break;
-          case SN_TYPE_INT16:  k = SN_K_I16; break;
-          default: break;
-        }
-        if (k >= 0) {
-          void *body = (uint8_t *)(uintptr_t)b.col_dev[c] + b.cols[c].body_off;
-          if (sn_launch_patch_apply(body, pd.pos, pd.val, pd.n, k,
-                                    e->stream) == 0) {
-            pd = Batch::PatchDev();
-            P = Patch();
-          }
-        }
-      }
-    }
-  }
-
-  /* stats row parse (UnsafeRow: [null words][3*ncols+1 x 8B slots],
-   * ColumnStatsSchema, ColumnEncoding.scala:1015-1036) */
-  if (stats && stats->data) {
-    const uint8_t *sp = (const uint8_t *)stats->data;
-    int32_t num_fields = nc * 3 + 1;
-    int32_t nwords = (num_fields + 63) >> 6;
-    if (stats->len >= (int64_t)nwords * 8 + (int64_t)num_fields * 8) {
-      const uint8_t *bits = sp;
-      const uint8_t *slots = sp + (int64_t)nwords * 8;
-      auto bit = [&](int f) {
-        return (rd_i64(bits + ((f >> 6) << 3)) >> (f & 63)) & 1;
-      };
-      b.lo_d.resize(nc); b.hi_d.resize(nc);
-      b.lo_i.resize(nc); b.hi_i.resize(nc);
-      b.null_count.resize(nc); b.bounds_null.resize(nc);
-      for (int c = 0; c < nc; c++) {
-        int f_lo = 1 + c * 3, f_hi = 2 + c * 3, f_nc = 3 + c * 3;
-        b.bounds_null[c] = bit(f_lo) || bit(f_hi);
-        b.null_count[c] = bit(f_nc) ? 0 : rd_i32(slots + (int64_t)f_nc * 8);
-        if (b.bounds_null[c]) continue;
-        switch (t->schema[c].dtype) {
-          case SN_TYPE_DOUBLE:
-            b.lo_d[c] = rd_f64(slots + (int64_t)f_lo * 8);
-            b.hi_d[c] = rd_f64(slots + (int64_t)f_hi * 8); break;
-          case SN_TYPE_FLOAT:
-            b.lo_d[c] = rd_f32(slots + (int64_t)f_lo * 8);
-            b.hi_d[c] = rd_f32(slots + (int64_t)f_hi * 8); break;
-          case SN_TYPE_INT64:
-            b.lo_i[c] = rd_i64(slots + (int64_t)f_lo * 8);
-            b.hi_i[c] = rd_i64(slots + (int64_t)f_hi * 8); break;
-          case SN_TYPE_STRING:
-            b.bounds_null[c] = 1; break;
-          default:
-            b.lo_i[c] = rd_i32(slots + (int64_t)f_lo * 8);
-            b.hi_i[c] = rd_i32(slots + (int64_t)f_hi * 8);
-        }
-      }
-      b.stats_valid = true;
-    }
-  }
+  apply_stats(t, b, stats);
 
   t->total_rows += b.num_rows;
   t->batches.push_back(std::move(b));
+  return SN_OK;
+}
+
+/* Attach the CURRENT mutation state to an existing batch — the seam the
+ * reference exercises when UPDATE/DELETE statements write delta blobs and
+ * delete masks to the region entries of a batch that already exists
+ * (ColumnDelta.scala:300-301 key addressing; ColumnBatchIterator then hands
+ * the latest buffers to every scan).  delete_mask and the delta pairs are
+ * CUMULATIVE (the reference merges delta chains upward), so they replace
+ * any previous state; stats, when given, replace the stats row (the
+ * reference rewrites it with a negative batchCount). */
+extern "C" int32_t sn_batch_mutate(sn_engine *e, int32_t table, int64_t uuid,
+                                   int32_t bucket_id,
+                                   const sn_buf *delete_mask,
+                                   const sn_buf *deltas,
+                                   const sn_buf *stats) {
+  Table *t = get_table(e, table);
+  if (!t) return fail(SN_ERR_BADARG, "bad table");
+  if (e->cfg.shard_count > 1 &&
+      (bucket_id % e->cfg.shard_count) != e->cfg.shard_rank)
+    return SN_OK;
+  std::lock_guard<std::mutex> g(t->mu);
+  Batch *b = nullptr;
+  for (auto &bb : t->batches)
+    if (bb.uuid == uuid && bb.bucket == bucket_id) { b = &bb; break; }
+  if (!b) return fail(SN_ERR_BADARG, "no batch uuid=%lld bucket=%d",
+                      (long long)uuid, bucket_id);
+  if (delete_mask) {
+    int32_t rc = apply_delete_mask(e, *b, delete_mask);
+    if (rc != SN_OK) return rc;
+  }
+  if (deltas) {
+    int32_t rc = apply_deltas(e, t, *b, deltas);
+    if (rc != SN_OK) return rc;
+  }
+  if (stats) apply_stats(t, b[0], stats);
+  else b->stats_valid = false;   /* old bounds no longer trustworthy */
+  /* cached descriptor sets reference the old patch/delete device data */
+  t->desc_caches.clear();
   return SN_OK;
 }
 

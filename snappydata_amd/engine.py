@@ -93,6 +93,10 @@ def lib():
         L.sn_query_merge.argtypes = [C.c_void_p, C.c_void_p, C.c_int64, C.c_int32]
         L.sn_query_partials_sharded.restype = C.c_int32
         L.sn_query_partials_sharded.argtypes = [C.c_void_p, C.c_int32, C.c_void_p]
+        L.sn_batch_mutate.restype = C.c_int32
+        L.sn_batch_mutate.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
+                                      C.c_int32, C.POINTER(abi.SnBuf),
+                                      C.POINTER(abi.SnBuf), C.POINTER(abi.SnBuf)]
         L.sn_ingest_columns.restype = C.c_int64
         L.sn_ingest_columns.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
                                         C.POINTER(abi.SnIngestCol), C.c_int32,
@@ -359,6 +363,36 @@ class Engine:
         """Compiled kernels in the tokenized plan cache (shapes, not
         literal values)."""
         return lib().sn_engine_jit_count(self._h)
+
+    def batch_mutate(self, table, uuid, bucket, delete_mask=None,
+                     deltas=None, stats=None):
+        """Attach the CURRENT cumulative mutation state to an existing batch
+        (the UPDATE/DELETE seam): delete_mask/deltas replace prior state."""
+        keep = []
+
+        def buf(blob):
+            b = abi.SnBuf()
+            if blob:
+                a = np.frombuffer(blob, dtype=np.uint8)
+                keep.append(a)
+                b.data = a.ctypes.data
+                b.len = len(blob)
+            return b
+
+        dm = buf(delete_mask) if delete_mask else None
+        st = buf(stats) if stats else None
+        darr = None
+        if deltas is not None:
+            nc = len(deltas)
+            darr = (abi.SnBuf * (2 * nc))()
+            for i, pair in enumerate(deltas):
+                d1, d2 = pair if isinstance(pair, tuple) else (pair, None)
+                darr[2 * i] = buf(d1)
+                darr[2 * i + 1] = buf(d2)
+        _check(lib().sn_batch_mutate(
+            self._h, table, uuid, bucket,
+            C.byref(dm) if dm else None, darr,
+            C.byref(st) if st else None), "batch_mutate")
 
     def dim_define(self, name):
         return _check(lib().sn_dim_define(self._h, name.encode()), "dim_define")

@@ -770,3 +770,40 @@ def test_stats_skip_never_drops_patched_batches(eng):
     rows = q.rows()
     assert rows[0][1][1] == 2.0
     assert abs(rows[0][1][0] - 11.0) <= 1e-9 * 11.0
+
+
+def test_batch_mutate_after_put(eng):
+    """UPDATE/DELETE after the batch exists (sn_batch_mutate): cumulative
+    delta + delete state replaces the old, descriptor caches invalidate,
+    and repeated identical plans see each new state."""
+    n = 40_000
+    rng = np.random.default_rng(31)
+    f64 = np.round(rng.random(n), 3)
+    t = eng.table_define("tmut", [(abi.T_DOUBLE, False)])
+    eng.batch_put(t, 77, 0, n, [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)])
+    plan = abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    r0 = eng.query(plan).rows()
+    assert abs(r0[0][1][0] - f64.sum()) <= 1e-9 * f64.sum()
+
+    # UPDATE rows 3 and 5
+    pos = np.array([3, 5], dtype=np.int32)
+    nv = np.array([10.0, 20.0])
+    d1 = se.encode_update_delta(abi.T_DOUBLE, pos, n, nv)
+    eng.batch_mutate(t, 77, 0, deltas=[(d1, None)])
+    r1 = eng.query(plan).rows()
+    exp1 = f64.sum() - f64[pos].sum() + nv.sum()
+    assert r1[0][1][1] == float(n)
+    assert abs(r1[0][1][0] - exp1) <= 1e-9 * exp1
+
+    # later, a LARGER cumulative delta (includes the first updates) + DELETE
+    pos2 = np.array([3, 5, 9], dtype=np.int32)
+    nv2 = np.array([10.0, 20.0, 30.0])
+    d2 = se.encode_update_delta(abi.T_DOUBLE, pos2, n, nv2)
+    dmask = po.encode_delete(np.array([5], dtype=np.int32), n)
+    eng.batch_mutate(t, 77, 0, deltas=[(d2, None)], delete_mask=dmask)
+    r2 = eng.query(plan).rows()
+    ref = f64.copy(); ref[pos2] = nv2
+    keep = np.ones(n, bool); keep[5] = False
+    assert r2[0][1][1] == float(n - 1)
+    exp2 = ref[keep].sum()
+    assert abs(r2[0][1][0] - exp2) <= 1e-9 * exp2
