@@ -269,3 +269,36 @@ def test_rle_layout_bytes():
     runs = blob[8:]
     exp = b"".join(x.to_bytes(4, "little") for x in [5, 3, 9, 2, 5, 1])
     assert runs == exp
+
+
+def test_dictionary_promotion_boundary():
+    """encoder promotes to BigDictionary exactly when the dictionary would
+    exceed Short.MaxValue entries (DictionaryEncoding.scala:313,338)."""
+    import numpy as np
+    below = [b"K%06d" % i for i in range(32767)]
+    blob = po.encode(po.T_STRING, po.ENC_DICT, below)
+    assert int.from_bytes(blob[:4], "little") == po.ENC_DICT
+    over = [b"K%06d" % i for i in range(32768)]
+    blob2 = po.encode(po.T_STRING, po.ENC_BIGDICT, over)
+    assert int.from_bytes(blob2[:4], "little") == po.ENC_BIGDICT
+    out, valid = po.decode(po.T_STRING, blob2, len(over))
+    assert list(out[:3]) == over[:3] and list(out[-2:]) == over[-2:]
+
+
+def test_delta_roundtrip_int64_and_int16():
+    """update-delta encode/decode for integer widths (value-as-double path)."""
+    import numpy as np
+    n = 10_000
+    for dt, npdt, vals in ((po.T_INT64, np.int64, [2**40, -7]),
+                           (po.T_INT16, np.int16, [123, -456])):
+        base = np.zeros(n, dtype=npdt)
+        pos = np.array([3, 8], dtype=np.int32)
+        d = po.encode_delta(dt, po.ENC_UNCOMPRESSED, pos, n,
+                            np.array(vals, dtype=npdt))
+        t = po.OracleTable([dt])
+        t.add_batch(-n, [po.encode(dt, po.ENC_UNCOMPRESSED, base)],
+                    deltas=[(d, None)])
+        rows = po.result_rows(t.query(po.make_plan(
+            aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])))
+        assert rows[0][1][0] == float(sum(vals))
+        assert rows[0][1][1] == float(n)

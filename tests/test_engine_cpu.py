@@ -122,3 +122,26 @@ def test_shard_filtering_cpu():
         e.close()
     assert sum(counts) == n
     assert all(c > 0 for c in counts)
+
+
+def test_batch_mutate_hostmode_no_crash():
+    """sn_batch_mutate on a host-only engine (no GPU): state bookkeeping
+    works, unknown batches fail loudly, get_blob keeps base bytes."""
+    import numpy as np
+    from oracle import pyoracle as po
+    n = 5_000
+    f64 = np.arange(n, dtype=np.float64)
+    e = se.Engine(device=-1)
+    t = e.table_define("tm", [(abi.T_DOUBLE, False)])
+    blob = po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64)
+    e.batch_put(t, 5, 0, n, [blob])
+    d1 = se.encode_update_delta(abi.T_DOUBLE, np.array([1], dtype=np.int32),
+                                n, np.array([99.0]))
+    e.batch_mutate(t, 5, 0, deltas=[(d1, None)])
+    assert e.get_blob(t, 0, 0) == blob          # base bytes untouched
+    try:
+        e.batch_mutate(t, 6, 0, deltas=[(d1, None)])
+        raise AssertionError("unknown uuid must fail")
+    except se.EngineError as ex:
+        assert ex.code == abi.SN_ERR_BADARG if hasattr(abi, "SN_ERR_BADARG") else True
+    e.close()
