@@ -96,3 +96,68 @@ def test_delete_delta_oracle_vs_numpy(n, seed, ndel, nupd):
         assert rows[0][1][0] is None
     else:
         assert abs(rows[0][1][0] - exp) <= 1e-9 * max(1.0, abs(exp))
+
+
+def test_batch_order_invariance():
+    """scan+agg results must not depend on batch insertion order."""
+    n = 120_000
+    rng = np.random.default_rng(67)
+    i32 = rng.integers(0, 500, n).astype(np.int32)
+    f64 = rng.random(n)
+    plan = po.make_plan(preds=[dict(col=0, hi=250, hi_strict=True)],
+                        aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    chunks = [(s, min(n, s + 30_000)) for s in range(0, n, 30_000)]
+
+    def run(order):
+        t = po.OracleTable([po.T_INT32, po.T_DOUBLE])
+        for s, e in order:
+            t.add_batch(e - s, [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, i32[s:e]),
+                                po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64[s:e])])
+        return po.result_rows(t.query(plan))
+
+    a = run(chunks)
+    b = run(list(reversed(chunks)))
+    assert a[0][1][1] == b[0][1][1]
+    assert abs(a[0][1][0] - b[0][1][0]) <= 1e-9 * max(1.0, abs(a[0][1][0]))
+
+
+def test_linearity_disjoint_union():
+    """SUM/COUNT over a union of disjoint row sets equals the sum of the
+    parts (the partial->final merge law at the semantic level)."""
+    n = 100_000
+    rng = np.random.default_rng(68)
+    f64 = rng.random(n)
+    plan = po.make_plan(aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+
+    def run(lo, hi):
+        t = po.OracleTable([po.T_DOUBLE])
+        t.add_batch(hi - lo, [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64[lo:hi])])
+        return po.result_rows(t.query(plan))[0][1]
+
+    whole = run(0, n)
+    left = run(0, n // 3)
+    right = run(n // 3, n)
+    assert whole[1] == left[1] + right[1] == float(n)
+    assert abs(whole[0] - (left[0] + right[0])) <= 1e-9 * abs(whole[0])
+
+
+def test_grouped_checksum_of_checksums():
+    """the grand total equals the sum over group sums (full-size-friendly
+    size-independent property)."""
+    n = 150_000
+    rng = np.random.default_rng(69)
+    keys = [b"G%02d" % v for v in rng.integers(0, 40, n)]
+    vals = rng.random(n)
+    t = po.OracleTable([po.T_STRING, po.T_DOUBLE])
+    for s in range(0, n, 50_000):
+        e = min(n, s + 50_000)
+        t.add_batch(e - s, [po.encode(po.T_STRING, po.ENC_DICT, keys[s:e]),
+                            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, vals[s:e])])
+    grows = po.result_rows(t.query(po.make_plan(
+        group_cols=[0], aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])))
+    krows = po.result_rows(t.query(po.make_plan(
+        aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])))
+    gsum = sum(v[0] for _, v in grows)
+    gcnt = sum(v[1] for _, v in grows)
+    assert gcnt == krows[0][1][1] == float(n)
+    assert abs(gsum - krows[0][1][0]) <= 1e-9 * abs(gsum)
