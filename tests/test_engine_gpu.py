@@ -807,3 +807,25 @@ def test_batch_mutate_after_put(eng):
     assert r2[0][1][1] == float(n - 1)
     exp2 = ref[keep].sum()
     assert abs(r2[0][1][0] - exp2) <= 1e-9 * exp2
+
+
+def test_fullsize_checksum_of_checksums(eng):
+    """size-independent property at scale (SURVEY §8 contract): the grouped
+    sums roll up to the keyless totals on a 10M-row table."""
+    n = 10_000_000
+    t = eng.table_define("li_cs", [(abi.T_DOUBLE, False)] * 4 +
+                         [(abi.T_STRING, False)] * 2 + [(abi.T_INT32, False)])
+    eng.datagen_lineitem(t, n, seed=7, batch_rows=600_000)
+    g = eng.query(abi.make_plan(table=t, group_cols=[tu.COL_RF, tu.COL_LS],
+                                aggs=[("sum", [(tu.COL_EP, 0.0, 1.0)]),
+                                      ("count", [])]))
+    grows = g.rows()
+    k = eng.query(abi.make_plan(table=t,
+                                aggs=[("sum", [(tu.COL_EP, 0.0, 1.0)]),
+                                      ("count", [])]))
+    krows = k.rows()
+    assert g.used_jit() and k.used_jit()
+    gsum = sum(v[0] for _, v in grows)
+    gcnt = sum(v[1] for _, v in grows)
+    assert gcnt == krows[0][1][1] == float(n)
+    assert abs(gsum - krows[0][1][0]) <= 1e-9 * abs(gsum)
