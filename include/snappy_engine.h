@@ -281,6 +281,11 @@ void      sn_query_destroy(sn_query *q);
 /* scan-kernel duration in ms (HIP events on the launch stream); -1 if the
  * query launched nothing */
 double    sn_query_kernel_ms(sn_query *q);
+/* result paging for group counts beyond SN_MAX_GROUP_SLOTS (dense group
+ * spaces up to 2^20 slots run through a global-atomic accumulate path);
+ * sn_query_result returns the first page */
+int32_t   sn_query_result_page(sn_query *q, int64_t offset, sn_result *out);
+int64_t   sn_query_num_groups(sn_query *q);
 /* 1 when the query ran a query-compiled (hipRTC) kernel, 0 interpreted */
 int32_t   sn_query_used_jit(sn_query *q);
 

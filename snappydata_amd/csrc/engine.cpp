@@ -1234,9 +1234,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (t->schema[c0].dtype != SN_TYPE_STRING) {
       int64_t span = int_key_span(c0, &q->gmin[0]);
       if (span < 0) return nullptr;
-      if (span > SN_MAX_GROUP_SLOTS) {
+      if (span > SN_BIG_GROUP_CAP) {
         fail(SN_ERR_UNSUPPORTED, "integer key span %lld > %d", (long long)span,
-             SN_MAX_GROUP_SLOTS);
+             SN_BIG_GROUP_CAP);
         return nullptr;
       }
       q->gint[0] = true;
@@ -1250,9 +1250,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       if (t->schema[c1].dtype != SN_TYPE_STRING) {
         int64_t span = int_key_span(c1, &q->gmin[1]);
         if (span < 0) return nullptr;
-        if (span > SN_MAX_GROUP_SLOTS) {
+        if (span > SN_BIG_GROUP_CAP) {
           fail(SN_ERR_UNSUPPORTED, "integer key span %lld > %d",
-               (long long)span, SN_MAX_GROUP_SLOTS);
+               (long long)span, SN_BIG_GROUP_CAP);
           return nullptr;
         }
         q->gint[1] = true;
@@ -1268,10 +1268,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (q->g1cap == 0) q->g1cap = 1;
     if (q->g2cap == 0) q->g2cap = 1;
     q->nslots = q->g1cap * q->g2cap;
-    if (q->nslots > SN_MAX_GROUP_SLOTS) {
-      fail(SN_ERR_UNSUPPORTED, "group cardinality %d > %d (open-address "
-           "hash table for arbitrary cardinality is round-2)",
-           q->nslots, SN_MAX_GROUP_SLOTS);
+    if (q->nslots > SN_BIG_GROUP_CAP) {
+      fail(SN_ERR_UNSUPPORTED, "group cardinality %d > %d",
+           q->nslots, SN_BIG_GROUP_CAP);
       return nullptr;
     }
   } else if (q->join_group) {
@@ -1626,11 +1625,16 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (dp.nslots > 16 && grid > SN_GRID_BIGSLOT) grid = SN_GRID_BIGSLOT;
     size_t nv = dp.nslots <= 1 ? (size_t)(2 * q->na_t + 1)
                                : (size_t)dp.nslots * (q->dev_naggs + 1);
-    size_t need = (size_t)grid * nv * 8;
+    const bool big_groups = dp.nslots > SN_RESULT_PAGE;
+    size_t need = (big_groups ? 1 : (size_t)grid) * nv * 8;
     if (e->scratch_sz < need) {
       e->scratch = (double *)e->arena.alloc(need);
       e->scratch_sz = need;
       if (!e->scratch) { fail(SN_ERR_NOMEM, "scratch alloc"); return nullptr; }
+    }
+    if (big_groups &&
+        hipMemsetAsync(e->scratch, 0, nv * 8, e->stream) != hipSuccess) {
+      fail(SN_ERR_GENERIC, "accumulator zero"); return nullptr;
     }
     q->ev_start = e->ev_acquire();
     q->ev_stop = e->ev_acquire();
@@ -1786,8 +1790,8 @@ static void finalize_groups(sn_query *q, std::vector<GroupOut> &groups) {
   q->final_groups = groups;
 }
 
-extern "C" int32_t sn_query_result(sn_query *q, sn_result *out) {
-  if (!q || !out) return SN_ERR_BADARG;
+static int32_t result_fill_page(sn_query *q, int64_t offset, sn_result *out) {
+  if (!q || !out || offset < 0) return SN_ERR_BADARG;
   int rc = sn_query_wait(q);
   if (rc != SN_OK) return rc;
   if (q->final_groups.empty() && !q->merged) {
@@ -1799,12 +1803,14 @@ extern "C" int32_t sn_query_result(sn_query *q, sn_result *out) {
   const sn_plan &p = q->plan;
   out->ngroup = q->join_group ? 1 : p.ngroup;
   out->naggs = p.naggs;
-  out->nrows = (int32_t)std::min((size_t)SN_MAX_GROUP_SLOTS, q->final_groups.size());
+  int64_t total = (int64_t)q->final_groups.size();
+  int64_t start = offset < total ? offset : total;
+  out->nrows = (int32_t)std::min((int64_t)SN_MAX_GROUP_SLOTS, total - start);
   out->rows_scanned = q->rows_scanned;
   out->batches_seen = q->batches_seen;
   out->batches_skipped = q->batches_skipped;
   for (int32_t i = 0; i < out->nrows; i++) {
-    GroupOut &g = q->final_groups[i];
+    GroupOut &g = q->final_groups[(size_t)(start + i)];
     for (int k = 0; k < out->ngroup; k++) {
       strncpy(out->keys[i][k], g.keys[k].c_str(), SN_KEY_MAX - 1);
       out->key_is_null[i][k] = g.key_null[k] ? 1 : 0;
@@ -1824,6 +1830,30 @@ extern "C" int32_t sn_query_result(sn_query *q, sn_result *out) {
     out->rows_passed += (int64_t)g.rowcount;
   }
   return SN_OK;
+}
+
+extern "C" int32_t sn_query_result(sn_query *q, sn_result *out) {
+  return result_fill_page(q, 0, out);
+}
+
+/* result paging for group counts beyond SN_MAX_GROUP_SLOTS (the
+ * global-atomic grouped path): fills up to one page from `offset`. */
+extern "C" int32_t sn_query_result_page(sn_query *q, int64_t offset,
+                                        sn_result *out) {
+  return result_fill_page(q, offset, out);
+}
+
+/* total group rows of the finalized result */
+extern "C" int64_t sn_query_num_groups(sn_query *q) {
+  if (!q) return SN_ERR_BADARG;
+  int rc = sn_query_wait(q);
+  if (rc != SN_OK) return rc;
+  if (q->final_groups.empty() && !q->merged) {
+    std::vector<GroupOut> groups;
+    local_groups(q, &groups);
+    finalize_groups(q, groups);
+  }
+  return (int64_t)q->final_groups.size();
 }
 
 static void sn_detach_queries(sn_engine *e) {
@@ -1862,6 +1892,11 @@ extern "C" int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_devi
   } else {
     std::vector<GroupOut> groups;
     local_groups(q, &groups);
+    if (groups.size() > (size_t)SN_MAX_GROUP_SLOTS)
+      return fail(SN_ERR_OVERFLOW,
+                  "%zu groups exceed the partial-block capacity %d (the "
+                  "key-sharded all-to-all overflow exchange is round-2)",
+                  groups.size(), SN_MAX_GROUP_SLOTS);
     int32_t n = (int32_t)std::min((size_t)SN_MAX_GROUP_SLOTS, groups.size());
     memcpy(block.data(), &n, 4);
     int32_t cap = SN_MAX_GROUP_SLOTS;

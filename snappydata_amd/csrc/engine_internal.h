@@ -116,6 +116,9 @@ typedef struct {
 } sn_dev_plan;
 
 #define SN_GRID_CAP 2048
+#define SN_RESULT_PAGE 1024      /* == SN_MAX_GROUP_SLOTS (result page) */
+#define SN_BIG_GROUP_CAP (1 << 20) /* dense-slot cap of the global-atomic
+                                      grouped path (>SN_RESULT_PAGE) */
 #define SN_GRID_BIGSLOT 256   /* grid cap for the >16-slot LDS hash-agg path */
 #define SN_TILE_ROWS 16384     /* rows per workgroup tile (16 LDS chunks;
                                   long pipelines for the staged conversion) */

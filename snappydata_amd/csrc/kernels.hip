@@ -1197,6 +1197,86 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
   (void)out_stride;
 }
 
+/* ---- unbounded-cardinality grouped scan (the SHAMap-overflow analogue):
+ * accumulates straight into a GLOBAL [nslots][naggs+1] array with f64
+ * atomics (zeroed by the host).  High cardinality means low per-address
+ * contention, which is exactly when global atomics are cheap; the slot
+ * space is still dense (dictionary ids / stats-ranged integers). ---- */
+__global__ __launch_bounds__(WG, 1)
+void k_grouped_global(sn_dev_plan plan,
+                      const sn_dev_plan *__restrict__ plan_g,
+                      const sn_dev_batch *__restrict__ batches,
+                      const sn_dev_tile *__restrict__ tiles, int ntiles,
+                      double *__restrict__ gacc) {
+  const int tid = threadIdx.x;
+  const int nused = plan.nused;
+  const int naggs = plan.naggs, ngroup = plan.ngroup;
+  const int npd = plan.npreds_d, npi = plan.npreds_i;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double *sval = (double *)smem;
+  uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
+  uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+  uint64_t *salive = sdead + CHUNK / 64;
+  int16_t *sslot = (int16_t *)(salive + CHUNK / 64 + 2);
+  sn_dev_plan *P = (sn_dev_plan *)((char *)sslot + CHUNK * 2);
+  {
+    const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
+    unsigned *dst = (unsigned *)P;
+    for (unsigned i = tid; i < sizeof(sn_dev_plan) / 4; i += WG) dst[i] = src[i];
+  }
+  __syncthreads();
+
+  GAS double *acc = (GAS double *)(uintptr_t)gacc;
+  const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const sn_dev_tile tile = tiles[t];
+    const sn_dev_batch &b = batches[tile.batch];
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
+    const int clean = b.clean;
+
+    for (int base = tile.row_start; base < tile_end; base += CHUNK) {
+      const int rows = min(CHUNK, tile_end - base);
+      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      __syncthreads();
+      alive_init(salive, sdead, rows, clean);
+      pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
+      const int jslot = plan.jkeys && plan.jmode == 1;
+      if (plan.jkeys) probe_sweep(P, sval, salive, jslot ? sslot : nullptr);
+
+#pragma unroll 2
+      for (int k = 0; k < CHUNK / WG; k++) {
+        const int r = tid + k * WG;
+        const uint64_t w = salive[r >> 6];
+        if (w == 0) continue;
+        const int m = (int)((w >> (tid & 63)) & 1ull);
+        if (!m) continue;
+        long long slot = 0;
+        if (jslot) slot = sslot[r];
+        else {
+          if (ngroup >= 1)
+            slot = ((long long)sval[(size_t)gc0 * CHUNK + r] -
+                    P->gbase[0]) * P->gmul0;
+          if (ngroup >= 2)
+            slot += (long long)sval[(size_t)gc1 * CHUNK + r] - P->gbase[1];
+        }
+        GAS double *row_acc = acc + (size_t)slot * (naggs + 1);
+        for (int a = 0; a < naggs; a++) {
+          const sn_dev_agg &A = P->aggs[a];
+          const double va =
+              (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
+              (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
+              (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+          (void)atomicAdd((double *)&row_acc[a], va);
+        }
+        (void)atomicAdd((double *)&row_acc[naggs], 1.0);
+      }
+      __syncthreads();
+    }
+  }
+}
+
 /* fold per-block partial rows into the final output.
  * keyless: final[i] = sum_b scratch[b][i]  (NV = 2*NA_t+1, identical layout)
  * grouped: scratch rows are [slot][naggs+1]; final is [slot][out_stride]
@@ -1302,6 +1382,15 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
     if (na <= 2) { if (nc4) KL(2, 4); else KL(2, 8); }
     else if (na <= 4) { if (nc4) KL(4, 4); else KL(4, 8); }
     else { if (nc4) KL(12, 4); else KL(12, 8); }
+  } else if (ns > SN_RESULT_PAGE) {
+    /* unbounded cardinality: global f64 atomics into the (host-zeroed)
+     * scratch accumulator — one "block" of partials for k_reduce */
+    lds += (CHUNK / 64) * 8 + CHUNK * 2 + sizeof(sn_dev_plan) + 64;
+    if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
+    hipLaunchKernelGGL(k_grouped_global, dim3(grid), dim3(WG), lds, s,
+                       *plan, dev_plan, dev_batches, dev_tiles, ntiles,
+                       dev_scratch);
+    grid = 1;   /* k_reduce folds ONE partial row set */
   } else if (ns > 16) {
     /* large-cardinality LDS hash-aggregate path: LDS accumulator bounds the
      * grid so scratch rows stay small */

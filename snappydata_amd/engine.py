@@ -83,6 +83,11 @@ def lib():
         L.sn_query_kernel_ms.argtypes = [C.c_void_p]
         L.sn_query_used_jit.restype = C.c_int32
         L.sn_query_used_jit.argtypes = [C.c_void_p]
+        L.sn_query_result_page.restype = C.c_int32
+        L.sn_query_result_page.argtypes = [C.c_void_p, C.c_int64,
+                                           C.POINTER(abi.SnResult)]
+        L.sn_query_num_groups.restype = C.c_int64
+        L.sn_query_num_groups.argtypes = [C.c_void_p]
         L.sn_engine_jit_count.restype = C.c_int32
         L.sn_engine_jit_count.argtypes = [C.c_void_p]
         L.sn_query_partial_bytes.restype = C.c_int64
@@ -218,7 +223,20 @@ class Query:
         return res
 
     def rows(self):
-        return abi.result_rows(self.result())
+        total = _check(lib().sn_query_num_groups(self._h), "num_groups")
+        if total <= abi.SN_MAX_GROUP_SLOTS:
+            return abi.result_rows(self.result())
+        out = []
+        off = 0
+        while off < total:
+            res = abi.SnResult()
+            _check(lib().sn_query_result_page(self._h, off, C.byref(res)),
+                   "result_page")
+            if res.nrows == 0:
+                break
+            out.extend(abi.result_rows(res))
+            off += res.nrows
+        return out
 
     def kernel_ms(self):
         """Scan-kernel duration (HIP events on the launch stream)."""

@@ -117,3 +117,30 @@ def test_engine_int_group_high_cardinality_and_partials():
     q.merge_host(block, len(block), 1)
     assert q.rows() == direct
     eng.close()
+
+
+@pytest.mark.gpu
+def test_big_group_cardinality_20k(rng_seed=75):
+    """>1024 dense group slots (global-atomic accumulate path) with paged
+    results: 20K distinct int keys vs a numpy model; partial export must
+    refuse loudly (the overflow exchange is round-2)."""
+    n = 2_000_000
+    rng = np.random.default_rng(rng_seed)
+    keys = rng.integers(0, 20_000, n).astype(np.int32)
+    vals = rng.random(n)
+    eng = se.Engine(device=0)
+    t = eng.table_define("tbig", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": keys}, {"data": vals}], n,
+                       batch_rows=250_000)
+    q = eng.query(abi.make_plan(table=t, group_cols=[0],
+                                aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])]))
+    rows = q.rows()
+    exp = expected(keys, vals)
+    assert len(rows) == len(exp)
+    check(rows, exp)
+    try:
+        q.partials_host()
+        raise AssertionError("partial export must refuse > 1024 groups")
+    except se.EngineError:
+        pass
+    eng.close()
