@@ -1645,11 +1645,13 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     bool jit_shape_ok;
     if (dp.nslots <= 1) jit_shape_ok = dp.naggs <= 12;
     else if (dp.nslots <= 8) jit_shape_ok = dp.naggs <= 6;
-    else
+    else if (dp.nslots <= 1024)
       /* LDS-accumulator mode: static shared = LDS image + gacc must fit */
-      jit_shape_ok = dp.nslots <= 1024 &&
-                     (size_t)dp.nused * 8192 +
+      jit_shape_ok = (size_t)dp.nused * 8192 +
                      (size_t)dp.nslots * (dp.naggs + 1) * 8 + 1024 <= 160 * 1024;
+    else
+      /* global-atomic mode: only the LDS image constrains */
+      jit_shape_ok = dp.nslots <= SN_BIG_GROUP_CAP;
     if (e->jit && jit_shape_ok &&
         (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first))) {
       const int *jk = hit ? hit->jit_kinds : jit_kinds;
@@ -1674,8 +1676,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
                          dp.jkeys, dp.jpayload, dp.jlut,
                          (const sn_dev_plan *)dp_dev, e->stream);
       if (rc == 0)
-        rc = sn_launch_reduce(e->scratch, jgrid, (int)nv, q->dev_out,
-                              naggs1, (int)q->out_stride, e->stream);
+        rc = sn_launch_reduce(e->scratch, big_groups ? 1 : jgrid, (int)nv,
+                              q->dev_out, naggs1, (int)q->out_stride,
+                              e->stream);
       q->used_jit = rc == 0;
       if (rc != 0) jfn = nullptr;   /* interpreted kernels take over */
     }

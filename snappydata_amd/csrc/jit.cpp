@@ -121,8 +121,10 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   /* >8 slots: block-level LDS accumulators (f64 LDS atomics) instead of
    * per-lane registers — the high-cardinality SHAMap analogue.  Occupancy 1
    * (the LDS image + accumulator array leave no room for a second group). */
-  const int lds_mode = grouped && nslots > 8;
-  const int wbin_pre = grouped && !lds_mode && NA <= 2 && nslots * (NA + 1) >= 12;
+  const int glob_mode = grouped && nslots > 1024;   /* HBM accumulator */
+  const int lds_mode = grouped && nslots > 8 && !glob_mode;
+  const int wbin_pre = grouped && !lds_mode && !glob_mode && NA <= 2 &&
+                       nslots * (NA + 1) >= 12;
   emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, %d)\n"
            "void jit_scan(const sn_dev_batch *__restrict__ batches,\n"
            "              const sn_dev_tile *__restrict__ tiles, int ntiles,\n"
@@ -136,7 +138,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
            "  const GAS int *jlut = (const GAS int *)(u64)jlut_p;\n"
            "  const GAS sn_dev_plan *P = (const GAS sn_dev_plan *)(u64)plan_p;\n"
            "  (void)jkeys; (void)jpayload; (void)jlut; (void)P;\n",
-        lds_mode ? 1 : (wbin_pre ? 4 : 2));
+        (lds_mode || glob_mode) ? 1 : (wbin_pre ? 4 : 2));
   /* tokenized plan values (the reference's ParamLiteral tokenization,
    * TokenizationTest / SnappySession plan cache): predicate bounds and
    * aggregate coefficients load once per wave from the cached device plan
@@ -168,13 +170,17 @@ __device__ __forceinline__ u64 mix64(u64 x) {
    * which the row phase barely uses. */
   const int wbin_mode = wbin_pre;
   emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
-  if (!lds_mode && !wbin_mode)
+  if (!lds_mode && !wbin_mode && !glob_mode)
     emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
           grouped ? nslots * (NA + 1) : 2 * na_t + 1);
   o += "  const int tid = threadIdx.x;\n";
 
   /* accumulators */
-  if (lds_mode) {
+  if (glob_mode) {
+    /* HBM accumulator IS the out/scratch pointer (host-zeroed); f64
+     * atomics are cheap at high cardinality (low per-address contention) */
+    o += "  GAS double *gacc = (GAS double *)(u64)out;\n";
+  } else if (lds_mode) {
     /* block-level LDS accumulator, zeroed once, flushed once at the end */
     emitf(o, "  __shared__ __attribute__((aligned(16))) double gacc[%d];\n"
              "  for (int i = tid; i < %d; i += WG) gacc[i] = 0.0;\n"
@@ -462,12 +468,13 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     if (A.nf >= 3) emitf(o, " * %s", t2);
     o += ";\n";
   }
-  if (lds_mode) {
+  if (lds_mode || glob_mode) {
     emitf(o, "        if (ok) {\n"
-             "          double *row = &gacc[slot * %d];\n"
-             "          atomicAdd(&row[%d], 1.0);\n", NA + 1, NA);
+             "          %sdouble *row = &gacc[(u64)slot * %d];\n"
+             "          atomicAdd((double *)&row[%d], 1.0);\n",
+          glob_mode ? "GAS " : "", NA + 1, NA);
     for (int a = 0; a < NA; a++)
-      emitf(o, "          atomicAdd(&row[%d], va%d);\n", a, a);
+      emitf(o, "          atomicAdd((double *)&row[%d], va%d);\n", a, a);
     o += "        }\n";
   } else if (wbin_mode) {
     emitf(o, "        if (ok) {\n"
@@ -502,6 +509,11 @@ __device__ __forceinline__ u64 mix64(u64 x) {
 
   /* block reduce into LDS bacc then scratch row */
   int nv = grouped ? nslots * (NA + 1) : 2 * na_t + 1;
+  if (glob_mode) {
+    /* nothing to flush — the HBM accumulator holds the single partial set */
+    o += "}\n";
+    return o;
+  }
   if (lds_mode) {
     /* gacc IS the block accumulator — flush it straight to the scratch row */
     emitf(o, "  __syncthreads();\n"
