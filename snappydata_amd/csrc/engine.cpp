@@ -1626,14 +1626,14 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     size_t nv = dp.nslots <= 1 ? (size_t)(2 * q->na_t + 1)
                                : (size_t)dp.nslots * (q->dev_naggs + 1);
     const bool big_groups = dp.nslots > SN_RESULT_PAGE;
-    size_t need = (big_groups ? 1 : (size_t)grid) * nv * 8;
+    size_t need = (big_groups ? 8 : (size_t)grid) * nv * 8;
     if (e->scratch_sz < need) {
       e->scratch = (double *)e->arena.alloc(need);
       e->scratch_sz = need;
       if (!e->scratch) { fail(SN_ERR_NOMEM, "scratch alloc"); return nullptr; }
     }
     if (big_groups &&
-        hipMemsetAsync(e->scratch, 0, nv * 8, e->stream) != hipSuccess) {
+        hipMemsetAsync(e->scratch, 0, 8 * nv * 8, e->stream) != hipSuccess) {
       fail(SN_ERR_GENERIC, "accumulator zero"); return nullptr;
     }
     q->ev_start = e->ev_acquire();
@@ -1676,7 +1676,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
                          dp.jkeys, dp.jpayload, dp.jlut,
                          (const sn_dev_plan *)dp_dev, e->stream);
       if (rc == 0)
-        rc = sn_launch_reduce(e->scratch, big_groups ? 1 : jgrid, (int)nv,
+        rc = sn_launch_reduce(e->scratch, big_groups ? 8 : jgrid, (int)nv,
                               q->dev_out, naggs1, (int)q->out_stride,
                               e->stream);
       q->used_jit = rc == 0;

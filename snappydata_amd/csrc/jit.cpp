@@ -177,9 +177,12 @@ __device__ __forceinline__ u64 mix64(u64 x) {
 
   /* accumulators */
   if (glob_mode) {
-    /* HBM accumulator IS the out/scratch pointer (host-zeroed); f64
-     * atomics are cheap at high cardinality (low per-address contention) */
-    o += "  GAS double *gacc = (GAS double *)(u64)out;\n";
+    /* HBM accumulator IS the out/scratch pointer (host-zeroed), privatized
+     * 8 ways by XCD (blockIdx & 7 matches the dispatch round-robin) so
+     * same-slot atomics from different XCDs never contend and stay in the
+     * local L2; k_reduce folds the 8 copies */
+    emitf(o, "  GAS double *gacc = (GAS double *)(u64)out"
+             " + (u64)(blockIdx.x & 7) * %d;\n", nslots * (NA + 1));
   } else if (lds_mode) {
     /* block-level LDS accumulator, zeroed once, flushed once at the end */
     emitf(o, "  __shared__ __attribute__((aligned(16))) double gacc[%d];\n"

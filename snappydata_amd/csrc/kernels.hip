@@ -1227,7 +1227,9 @@ void k_grouped_global(sn_dev_plan plan,
   }
   __syncthreads();
 
-  GAS double *acc = (GAS double *)(uintptr_t)gacc;
+  /* 8-way XCD privatization (see the JIT twin): fold happens in k_reduce */
+  GAS double *acc = (GAS double *)(uintptr_t)gacc +
+                    (size_t)(blockIdx.x & 7) * (size_t)plan.nslots * (naggs + 1);
   const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
@@ -1390,7 +1392,7 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
     hipLaunchKernelGGL(k_grouped_global, dim3(grid), dim3(WG), lds, s,
                        *plan, dev_plan, dev_batches, dev_tiles, ntiles,
                        dev_scratch);
-    grid = 1;   /* k_reduce folds ONE partial row set */
+    grid = 8;   /* k_reduce folds the 8 XCD-private copies */
   } else if (ns > 16) {
     /* large-cardinality LDS hash-aggregate path: LDS accumulator bounds the
      * grid so scratch rows stay small */
