@@ -671,6 +671,8 @@ static int32_t apply_delete_mask(sn_engine *e, Batch &b,
    * the mask is cumulative, so it REPLACES any previous one */
   b.has_deletes = false;
   b.del_bm_dev = nullptr;
+  if (delete_mask && delete_mask->data && delete_mask->len < 12)
+    return fail(SN_ERR_BADFORMAT, "delete mask shorter than its header");
   if (delete_mask && delete_mask->data && delete_mask->len >= 12) {
     const uint8_t *dm = (const uint8_t *)delete_mask->data;
     int32_t n = rd_i32(dm + 8);

@@ -302,3 +302,69 @@ def test_delta_roundtrip_int64_and_int16():
             aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])))
         assert rows[0][1][0] == float(sum(vals))
         assert rows[0][1][1] == float(n)
+
+
+def test_malformed_blobs_rejected():
+    """truncated/corrupt blobs must fail loudly, not crash (host put path
+    and oracle decoder)."""
+    import numpy as np
+    from snappydata_amd import abi, engine as se
+    good = po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, np.arange(100.0))
+    bad_cases = [
+        good[:6],                                     # shorter than header
+        b"\xff\xff\xff\x7f" + good[4:],               # absurd typeId
+        good[:4] + (7).to_bytes(4, "little") + good[8:],   # nullBytes % 8 != 0
+        good[:4] + (1 << 20).to_bytes(4, "little"),   # null bytes beyond blob
+    ]
+    e = se.Engine(device=-1)
+    t = e.table_define("tbad", [(abi.T_DOUBLE, False)])
+    for i, blob in enumerate(bad_cases):
+        try:
+            e.batch_put(t, i, 0, 100, [blob])
+            raise AssertionError(f"case {i} must be rejected")
+        except se.EngineError:
+            pass
+    # a put that failed must not have appended a batch
+    assert e.num_rows(t) == 0
+    e.close()
+
+    # the oracle validates lazily (decode at query time): either the put or
+    # the first query must fail — never crash or return wrong numbers
+    plan = po.make_plan(aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    for i, blob in enumerate(bad_cases):
+        ot = po.OracleTable([po.T_DOUBLE])
+        try:
+            ot.add_batch(100, [blob])
+        except Exception:
+            continue
+        try:
+            ot.query(plan)
+            raise AssertionError(f"oracle case {i} must fail at query")
+        except AssertionError as ex:
+            if "must fail" in str(ex):
+                raise
+        except Exception:
+            pass
+
+
+def test_truncated_delete_mask_and_delta_rejected():
+    import numpy as np
+    from snappydata_amd import abi, engine as se
+    n = 1000
+    blob = po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, np.arange(float(n)))
+    dmask = po.encode_delete(np.array([1, 2], dtype=np.int32), n)
+    e = se.Engine(device=-1)
+    t = e.table_define("tbad2", [(abi.T_DOUBLE, False)])
+    try:
+        e.batch_put(t, 0, 0, n, [blob], delete_mask=dmask[:10])
+        raise AssertionError("truncated delete mask must be rejected")
+    except se.EngineError:
+        pass
+    d1 = po.encode_delta(po.T_DOUBLE, po.ENC_UNCOMPRESSED,
+                         np.array([5], dtype=np.int32), n, np.array([1.0]))
+    try:
+        e.batch_put(t, 1, 0, -n, [blob], deltas=[(d1[:8], None)])
+        raise AssertionError("truncated delta must be rejected")
+    except se.EngineError:
+        pass
+    e.close()
