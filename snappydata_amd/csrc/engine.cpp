@@ -875,6 +875,29 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
         } else gid = it->second;
         l2g.push_back(gid);
       }
+      /* validate the index array: a corrupt index would walk the kernel
+       * off the end of the local->global map (index == dict size is the
+       * NULL sentinel, DictionaryEncoding.scala:90) */
+      const ColMeta &dm = b.cols[c];
+      const int32_t dn = (int32_t)dm.dict.size();
+      const uint8_t *bodyp = blob + dm.body_off;
+      int w = dm.type_id == SN_ENC_DICTIONARY ? 2 : 4;
+      /* exactly one index per NON-NULL row; padding beyond is not data */
+      int64_t nnull = 0;
+      for (int32_t wi = 0; wi < dm.num_null_words; wi++)
+        nnull += __builtin_popcountll(
+            (unsigned long long)rd_i64(blob + dm.null_off + (int64_t)wi * 8));
+      int64_t nidx = (int64_t)b.num_rows - nnull;
+      if (nidx * w > len - dm.body_off)
+        return fail(SN_ERR_BADFORMAT, "dictionary index array truncated col %d", c);
+      for (int64_t i = 0; i < nidx; i++) {
+        int32_t ix = w == 2 ? (int32_t)(uint16_t)rd_i16(bodyp + i * 2)
+                            : rd_i32(bodyp + i * 4);
+        if (ix < 0 || ix > dn)
+          return fail(SN_ERR_BADFORMAT,
+                      "dictionary index %d out of range [0,%d] at row %lld col %d",
+                      ix, dn, (long long)i, c);
+      }
     }
     /* upload blob, placed so the BODY is 16-byte aligned (the scan kernel
      * issues 16 B/lane vector loads on the body; the 8-byte blob header

@@ -368,3 +368,25 @@ def test_truncated_delete_mask_and_delta_rejected():
     except se.EngineError:
         pass
     e.close()
+
+
+def test_corrupt_dictionary_index_rejected():
+    """an index past the dictionary (beyond the NULL sentinel) must be
+    rejected at put — the scan kernel would otherwise walk off the map."""
+    import numpy as np
+    from snappydata_amd import abi, engine as se
+    keys = [b"A", b"B", b"A", b"C"] * 100
+    blob = bytearray(po.encode(po.T_STRING, po.ENC_DICT, keys))
+    # index array is the tail: poke one int16 index to 999 (> dict size 3)
+    blob[-2:] = (999).to_bytes(2, "little")
+    e = se.Engine(device=-1)
+    t = e.table_define("tdx", [(abi.T_STRING, False)])
+    try:
+        e.batch_put(t, 0, 0, len(keys), [bytes(blob)])
+        raise AssertionError("corrupt dictionary index must be rejected")
+    except se.EngineError:
+        pass
+    # the untouched blob still loads
+    e.batch_put(t, 1, 0, len(keys), [po.encode(po.T_STRING, po.ENC_DICT, keys)])
+    assert e.num_rows(t) == len(keys)
+    e.close()
