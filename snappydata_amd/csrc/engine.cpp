@@ -194,6 +194,10 @@ struct Table {
   /* global dictionary per column (string dict cols) */
   std::vector<std::vector<std::string>> gdict;
   std::vector<std::map<std::string, int32_t>> gdict_idx;
+  /* longest interned entry per column: group keys beyond SN_KEY_MAX-1
+   * bytes cannot travel in results/partial blocks — queries grouping on
+   * such a column fail loudly instead of silently merging truncated keys */
+  std::vector<int32_t> gdict_maxlen;
   int64_t total_rows = 0;
   std::mutex mu;
 };
@@ -213,6 +217,11 @@ struct Dim {
   /* dense payload LUT over [lut_min, lut_max] when the span is small */
   const int32_t *dev_lut = nullptr;
   int64_t lut_min = 0, lut_max = -1;
+  int32_t attr_maxlen = 0;
+  /* guards keys/attrs/device-table pointers: sn_dim_put may run while a
+   * submit builds/reads the device table (the reference's replicated
+   * region puts are similarly concurrent with task-side get()s) */
+  std::mutex mu;
 };
 
 struct sn_engine {
@@ -299,6 +308,7 @@ extern "C" int32_t sn_table_define(sn_engine *e, const char *name,
   t->schema.assign(schema, schema + ncols);
   t->gdict.resize(ncols);
   t->gdict_idx.resize(ncols);
+  t->gdict_maxlen.assign(ncols, 0);
   e->tables.push_back(std::move(t));
   return (int32_t)e->tables.size() - 1;
 }
@@ -318,6 +328,7 @@ extern "C" int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
   if (!e || dim < 0 || dim >= (int32_t)e->dims.size() || !keys || nkeys <= 0)
     return fail(SN_ERR_BADARG, "bad dim args");
   Dim *d = e->dims[dim].get();
+  std::lock_guard<std::mutex> gd(d->mu);
   int64_t off = 0;
   for (int64_t i = 0; i < nkeys; i++) {
     d->keys.push_back(keys[i]);
@@ -328,6 +339,7 @@ extern "C" int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
       for (size_t j = 0; j < d->attr_dict.size(); j++)
         if (d->attr_dict[j] == a) { gid = (int32_t)j; break; }
       if (gid < 0) { gid = (int32_t)d->attr_dict.size(); d->attr_dict.push_back(a); }
+      if ((int32_t)a.size() > d->attr_maxlen) d->attr_maxlen = (int32_t)a.size();
       d->attrs.push_back(std::move(a));
       d->attr_gid.push_back(gid);
     } else {
@@ -352,6 +364,7 @@ static inline uint64_t mix64h(uint64_t x) {
 
 /* build (or reuse) the device open-address table: the broadcast into HBM */
 static int dim_device_table(sn_engine *e, Dim *d) {
+  std::lock_guard<std::mutex> gd(d->mu);
   if (d->dev_keys) return SN_OK;
   int32_t lg = 1;
   while ((1u << lg) < 2 * d->keys.size() + 1) lg++;
@@ -602,7 +615,13 @@ static int decode_delta(const uint8_t *blob, int64_t len, sn_type_t dtype,
         case SN_TYPE_DOUBLE: v = rd_f64(body + (int64_t)nnp * 8); break;
         case SN_TYPE_FLOAT:  v = rd_f32(body + (int64_t)nnp * 4); break;
         case SN_TYPE_INT32:  v = (double)rd_i32(body + (int64_t)nnp * 4); break;
-        case SN_TYPE_INT64:  v = (double)rd_i64(body + (int64_t)nnp * 8); break;
+        case SN_TYPE_INT64: {
+          /* carry RAW BITS (exact beyond 2^53, like the base-column path);
+           * every consumer of int64 patch values bitcasts back */
+          int64_t iv = rd_i64(body + (int64_t)nnp * 8);
+          memcpy(&v, &iv, 8);
+          break;
+        }
         case SN_TYPE_INT16:  v = (double)rd_i16(body + (int64_t)nnp * 2); break;
         case SN_TYPE_STRING: {
           int32_t sz = rd_i32(body + var_cur);
@@ -616,6 +635,8 @@ static int decode_delta(const uint8_t *blob, int64_t len, sn_type_t dtype,
             gid = (int32_t)tab->gdict[col].size();
             tab->gdict[col].push_back(s);
             gi.emplace(s, gid);
+            if ((int32_t)s.size() > tab->gdict_maxlen[col])
+              tab->gdict_maxlen[col] = (int32_t)s.size();
           } else gid = it->second;
           v = (double)gid;
           break;
@@ -636,8 +657,13 @@ static int decode_delta(const uint8_t *blob, int64_t len, sn_type_t dtype,
           gid = (int32_t)tab->gdict[col].size();
           tab->gdict[col].push_back(s);
           gi.emplace(s, gid);
+          if ((int32_t)s.size() > tab->gdict_maxlen[col])
+            tab->gdict_maxlen[col] = (int32_t)s.size();
         } else gid = it->second;
         v = (double)gid;
+      } else if (dtype == SN_TYPE_INT64) {
+        int64_t iv = dict_i[idx];
+        memcpy(&v, &iv, 8);              /* raw bits, same as above */
       } else {
         v = (double)dict_i[idx];
       }
@@ -872,6 +898,8 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
           gid = (int32_t)t->gdict[c].size();
           t->gdict[c].push_back(s);
           t->gdict_idx[c].emplace(s, gid);
+          if ((int32_t)s.size() > t->gdict_maxlen[c])
+            t->gdict_maxlen[c] = (int32_t)s.size();
         } else gid = it->second;
         l2g.push_back(gid);
       }
@@ -899,6 +927,66 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
                       ix, dn, (long long)i, c);
       }
     }
+    /* int-typed dictionary columns (DictionaryEncoding over int32/int64,
+     * DictionaryEncoding.scala:85-137): materialize the values into a plain
+     * fixed-width body at put — decompress-on-put, like the LZ4 wrapper —
+     * so the scan runs the plain I32/I64 path (JIT-eligible).  Index ==
+     * numElements is the null sentinel (DictionaryEncoding.scala:90): such
+     * rows are folded into the synthesized null bitset. */
+    std::vector<uint8_t> synth;
+    if ((b.cols[c].type_id == SN_ENC_DICTIONARY ||
+         b.cols[c].type_id == SN_ENC_BIG_DICTIONARY) &&
+        t->schema[c].dtype != SN_TYPE_STRING) {
+      const ColMeta dm = b.cols[c];
+      const sn_type_t dt = t->schema[c].dtype;
+      const int w = dm.type_id == SN_ENC_DICTIONARY ? 2 : 4;
+      const int vw = dt == SN_TYPE_INT64 ? 8 : 4;
+      const int32_t dn = dm.dict_n;
+      const int32_t rows = b.num_rows;
+      const int32_t nwords = (rows + 63) >> 6;
+      std::vector<uint64_t> nullw(nwords, 0);
+      if (dm.num_null_words) {
+        for (int32_t wi = 0; wi < dm.num_null_words && wi < nwords; wi++)
+          nullw[wi] = (uint64_t)rd_i64(blob + dm.null_off + (int64_t)wi * 8);
+      }
+      std::vector<int64_t> vals;
+      vals.reserve(rows);
+      const uint8_t *bodyp = blob + dm.body_off;
+      int64_t ii = 0;                     /* index cursor (one per non-null row) */
+      const int64_t navail = (len - dm.body_off) / w;
+      for (int32_t r = 0; r < rows; r++) {
+        if ((nullw[r >> 6] >> (r & 63)) & 1) continue;
+        if (ii >= navail)
+          return fail(SN_ERR_BADFORMAT, "dictionary index array truncated col %d", c);
+        int32_t ix = w == 2 ? (int32_t)(uint16_t)rd_i16(bodyp + ii * 2)
+                            : rd_i32(bodyp + ii * 4);
+        ii++;
+        if (ix < 0 || ix > dn)
+          return fail(SN_ERR_BADFORMAT,
+                      "dictionary index %d out of range [0,%d] col %d", ix, dn, c);
+        if (ix == dn) { nullw[r >> 6] |= 1ull << (r & 63); continue; }
+        vals.push_back(dm.dict_i[ix]);
+      }
+      bool any_null = false;
+      for (uint64_t wv : nullw) any_null |= wv != 0;
+      const int64_t nb = any_null ? (int64_t)nwords * 8 : 0;
+      synth.resize(8 + nb + (int64_t)vals.size() * vw);
+      int32_t tid0 = SN_ENC_UNCOMPRESSED, nb32 = (int32_t)nb;
+      memcpy(synth.data(), &tid0, 4);
+      memcpy(synth.data() + 4, &nb32, 4);
+      if (nb) memcpy(synth.data() + 8, nullw.data(), (size_t)nb);
+      uint8_t *vb = synth.data() + 8 + nb;
+      for (size_t i = 0; i < vals.size(); i++) {
+        if (vw == 8) memcpy(vb + i * 8, &vals[i], 8);
+        else { int32_t v32 = (int32_t)vals[i]; memcpy(vb + i * 4, &v32, 4); }
+      }
+      blob = synth.data();
+      len = (int64_t)synth.size();
+      b.cols[c] = ColMeta();
+      int rc2 = parse_blob(blob, len, dt, &b.cols[c]);
+      if (rc2 != SN_OK) return fail(rc2, "materialized dict col %d", c);
+    }
+
     /* upload blob, placed so the BODY is 16-byte aligned (the scan kernel
      * issues 16 B/lane vector loads on the body; the 8-byte blob header
      * would otherwise leave it 8-aligned) */
@@ -1219,10 +1307,24 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (use_col(plan->join_fact_col) < 0) {
       fail(SN_ERR_BADARG, "too many plan columns"); return nullptr;
     }
+    if (plan->join_mode == SN_JOIN_GROUP && jd->attr_maxlen > SN_KEY_MAX - 1) {
+      fail(SN_ERR_UNSUPPORTED,
+           "dimension attribute exceeds the %d-byte group-key limit",
+           SN_KEY_MAX - 1);
+      return nullptr;
+    }
     if (dim_device_table(e, jd) != SN_OK) return nullptr;
     q->join_dim = jd;
     q->join_group = plan->join_mode == SN_JOIN_GROUP;
   }
+
+  /* ONE t->mu critical section from group-slot derivation through the
+   * descriptor build below: a concurrent sn_batch_put (config-5
+   * ingest+scan) may grow the global dictionary and append batches; slot
+   * capacities, premultiplied dictmaps and the descriptor set must all see
+   * the same consistent table state or grouped kernels write accumulator
+   * rows out of bounds. */
+  std::lock_guard<std::mutex> g(t->mu);
 
   /* group slot space: global dict sizes, plus a null slot only when the
    * schema allows null keys (non-nullable key columns waste no slots —
@@ -1252,9 +1354,20 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     return mx - mn + 1;
   };
   if (plan->ngroup >= 1) {
-    /* short lock: reads batch stats + global dictionaries, both of which a
-     * concurrent sn_batch_put may be growing (config-5 ingest+scan) */
-    std::lock_guard<std::mutex> gslot(t->mu);
+    /* group keys travel as fixed SN_KEY_MAX-byte strings in results and
+     * partial blocks; distinct keys sharing a 47-byte prefix would silently
+     * merge — reject loudly instead (the reference has no key-length limit,
+     * so this is a declared engine restriction, not silent divergence) */
+    for (int i = 0; i < plan->ngroup; i++) {
+      int gc = plan->group_cols[i];
+      if (t->schema[gc].dtype == SN_TYPE_STRING &&
+          t->gdict_maxlen[gc] > SN_KEY_MAX - 1) {
+        fail(SN_ERR_UNSUPPORTED,
+             "group key col %d has a %d-byte dictionary entry > %d-byte key limit",
+             gc, t->gdict_maxlen[gc], SN_KEY_MAX - 1);
+        return nullptr;
+      }
+    }
     int c0 = plan->group_cols[0];
     if (t->schema[c0].dtype != SN_TYPE_STRING) {
       int64_t span = int_key_span(c0, &q->gmin[0]);
@@ -1350,7 +1463,6 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       int64_t mul = is_g2 ? 1 : (q->g2cap > 0 ? q->g2cap : 1);
       int64_t gid = -1;
       {
-        std::lock_guard<std::mutex> gl(t->mu);
         auto it = t->gdict_idx[s.col].find(
             std::string(s.str_eq, (size_t)s.str_len));
         if (it != t->gdict_idx[s.col].end()) gid = it->second;
@@ -1399,9 +1511,35 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         if (sa.nfactors >= 1) {
           da.c0 = q->cslot_of_col[sa.factors[0].col];
           da.a0 = sa.factors[0].add; da.m0 = sa.factors[0].mul;
-          if (dp.i64_mask & (1u << da.c0)) {
-            fail(SN_ERR_UNSUPPORTED, "int64 aggregate factors not in round-1 GPU path");
-            return nullptr;
+          /* INT64 factors: kernels convert the raw-bit i64 LDS image to
+           * double per i64_mask.  A pure SUM(bigint_col) must stay
+           * BIT-EXACT (north star): f64 accumulation of integers is exact
+           * while every partial sum < 2^53, which the stats bounds prove;
+           * when they cannot, fail loudly rather than round silently.
+           * (Affine/multi-factor expressions are double-typed in SQL —
+           * the 1e-6 double budget applies, no gate.) */
+          if ((dp.i64_mask & (1u << da.c0)) && sa.kind == SN_AGG_SUM &&
+              sa.nfactors == 1 && sa.factors[0].add == 0.0 &&
+              sa.factors[0].mul == 1.0) {
+            int gc = sa.factors[0].col;
+            double bound = 0.0;
+            bool provable = true;
+            for (auto &bb : t->batches) {
+              if (!bb.stats_valid || bb.bounds_null[gc] ||
+                  (bb.had_patches.size() > (size_t)gc && bb.had_patches[gc])) {
+                provable = false;
+                break;
+              }
+              double mx = std::max(std::fabs((double)bb.lo_i[gc]),
+                                   std::fabs((double)bb.hi_i[gc]));
+              bound += (double)bb.num_rows * mx;
+            }
+            if (!provable || bound >= 9007199254740992.0 /* 2^53 */) {
+              fail(SN_ERR_UNSUPPORTED,
+                   "SUM over int64 col %d cannot be proven f64-exact "
+                   "(needs stats bounds with rows*max|v| < 2^53)", gc);
+              return nullptr;
+            }
           }
         }
         if (sa.nfactors >= 2) {
@@ -1437,8 +1575,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   /* batch descriptors + tile map.  Expensive at many batches (dict-map
    * premultiply + uploads), so cache per plan signature; a cached set is
    * only reusable when stats-skip removes nothing for THIS plan (skipping
-   * is an optimization — correctness is unaffected either way). */
-  std::lock_guard<std::mutex> g(t->mu);
+   * is an optimization — correctness is unaffected either way).
+   * (t->mu is still held — same critical section as the slot derivation.) */
   uint64_t sig = 1469598103934665603ull;
   auto mix = [&](uint64_t v) { sig ^= v; sig *= 1099511628211ull; };
   for (int32_t c : q->used_cols) mix((uint64_t)c + 1);
@@ -1508,7 +1646,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
             fail(SN_ERR_UNSUPPORTED, "RLE col %d has no run aux", c);
             return nullptr;
           }
-          dc.kind = SN_K_RLE;
+          /* INT64 runs carry raw bits in rle_vals (put-time extraction) —
+           * a distinct kind so read_general bitcasts instead of rounding */
+          dc.kind = dt == SN_TYPE_INT64 ? SN_K_RLE_I64 : SN_K_RLE;
           dc.rle_ends = b.rle_ends_dev[c];
           dc.rle_vals = b.rle_vals_dev[c];
           dc.rle_n = b.rle_n[c];
@@ -1786,6 +1926,7 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
       }
     }
     if (q->join_group) {
+      std::lock_guard<std::mutex> gd(q->join_dim->mu);
       g.keys[0] = q->join_dim->attr_dict[s];
     } else if (p.ngroup >= 1) {
       int c0 = p.group_cols[0];

@@ -33,7 +33,14 @@ enum {
   SN_K_I16 = 6, SN_K_BOOLBIT = 7,
   SN_K_U8 = 8,       /* uncompressed boolean byte body */
   SN_K_S8 = 9,       /* uncompressed int8 */
-  SN_K_RLE = 10      /* run-length: host-built run-ends + values aux */
+  SN_K_RLE = 10,     /* run-length: host-built run-ends + values aux */
+  SN_K_RLE_I64 = 11  /* run-length over an INT64 column: rle_vals carry the
+                        RAW int64 bits memcpy'd into doubles (the LDS-image
+                        convention for i64), not numeric doubles.
+                        (Int-typed DICTIONARY columns have no kernel kind:
+                        the host materializes their values into a plain
+                        fixed-width body at put — decompress-on-put, like
+                        the LZ4 wrapper.) */
 };
 
 typedef struct {

@@ -456,16 +456,25 @@ __device__ __forceinline__ u64 mix64(u64 x) {
               p->gcol[1], (long long)p->gbase[1]);
     }
   }
+  /* factor read: i64 slots hold raw bits in the LDS image — emit the
+   * convert only for those slots (compile-time; i64_mask is in the shape) */
+  auto fread = [&](char *dst, int cs) {
+    if ((p->i64_mask >> cs) & 1u)
+      snprintf(dst, 72, "(double)__double_as_longlong(sval[%d][r])", cs);
+    else
+      snprintf(dst, 72, "sval[%d][r]", cs);
+  };
   for (int a = 0; a < NA; a++) {
     const sn_dev_agg &A = p->aggs[a];
     if (A.nf < 1) { emitf(o, "        const double va%d = 1.0;\n", a); continue; }
-    char t0[80], t1[80], t2[80];
-    if (ftriv(A.a0, A.m0)) snprintf(t0, 80, "sval[%d][r]", A.c0);
-    else snprintf(t0, 80, "__builtin_fma(ag%d_m0, sval[%d][r], ag%d_a0)", a, A.c0, a);
-    if (ftriv(A.a1, A.m1)) snprintf(t1, 80, "sval[%d][r]", A.c1);
-    else snprintf(t1, 80, "__builtin_fma(ag%d_m1, sval[%d][r], ag%d_a1)", a, A.c1, a);
-    if (ftriv(A.a2, A.m2)) snprintf(t2, 80, "sval[%d][r]", A.c2);
-    else snprintf(t2, 80, "__builtin_fma(ag%d_m2, sval[%d][r], ag%d_a2)", a, A.c2, a);
+    char t0[128], t1[128], t2[128], f0[72], f1[72], f2[72];
+    fread(f0, A.c0); fread(f1, A.c1); fread(f2, A.c2);
+    if (ftriv(A.a0, A.m0)) snprintf(t0, 128, "%s", f0);
+    else snprintf(t0, 128, "__builtin_fma(ag%d_m0, %s, ag%d_a0)", a, f0, a);
+    if (ftriv(A.a1, A.m1)) snprintf(t1, 128, "%s", f1);
+    else snprintf(t1, 128, "__builtin_fma(ag%d_m1, %s, ag%d_a1)", a, f1, a);
+    if (ftriv(A.a2, A.m2)) snprintf(t2, 128, "%s", f2);
+    else snprintf(t2, 128, "__builtin_fma(ag%d_m2, %s, ag%d_a2)", a, f2, a);
     emitf(o, "        const double va%d = %s", a, t0);
     if (A.nf >= 2) emitf(o, " * %s", t1);
     if (A.nf >= 3) emitf(o, " * %s", t2);

@@ -93,6 +93,12 @@ __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
         return 1;
       }
       if (c.kind == SN_K_F64 || c.kind == SN_K_F32) { *vd = pv; *vi = (long long)pv; }
+      else if (c.kind == SN_K_I64 || c.kind == SN_K_RLE_I64) {
+        /* INT64 patch values travel as RAW BITS (decode_delta keeps them
+         * exact beyond 2^53); vd keeps the raw-bit double so the LDS image
+         * stays in the i64 raw-bit convention */
+        *vi = __double_as_longlong(pv); *vd = pv;
+      }
       else { long long b = __double2ll_rn(pv); *vi = b; *vd = (double)b; }
       return 1;
     }
@@ -113,13 +119,14 @@ __device__ __forceinline__ int read_general(const sn_dev_col &c, int row,
     case SN_K_BOOLBIT: *vi = bm_get((const uint64_t *)c.body, nnp); *vd = (double)*vi; break;
     case SN_K_U8: *vi = as_global((const uint8_t *)c.body)[nnp] == 1; *vd = (double)*vi; break;
     case SN_K_S8: *vi = as_global((const int8_t *)c.body)[nnp]; *vd = (double)*vi; break;
-    case SN_K_RLE: {
+    case SN_K_RLE:
       *vd = rle_value(c, nnp);
-      const int is_i64 = (c.kind == SN_K_RLE);  /* vals carry raw i64 bits for i64 */
-      (void)is_i64;
-      *vi = __double2ll_rn(*vd);
+      *vi = __double2ll_rn(*vd);        /* numeric doubles (i16/i32 runs) */
       break;
-    }
+    case SN_K_RLE_I64:
+      *vd = rle_value(c, nnp);          /* raw i64 bits as double */
+      *vi = __double_as_longlong(*vd);  /* bitcast back to the exact value */
+      break;
   }
   return 1;
 }
@@ -247,7 +254,10 @@ __device__ __forceinline__ void convert_chunk(
           }
           break;
         }
-        case SN_K_RLE: {
+        case SN_K_RLE:
+        case SN_K_RLE_I64: {
+          /* RLE_I64 run values are raw i64 bits as doubles — exactly the
+           * i64 LDS-image convention, so the same copy is correct */
 #pragma unroll
           for (int k = 0; k < CHUNK / WG; k++) {
             int r = tid + k * WG;
@@ -259,7 +269,7 @@ __device__ __forceinline__ void convert_chunk(
     } else {
       uint64_t *vw = svalid + (size_t)c * (CHUNK / 64);
       const int is_dict = col.kind == SN_K_DICT16 || col.kind == SN_K_DICT32;
-      const int is_i64 = col.kind == SN_K_I64;
+      const int is_i64 = col.kind == SN_K_I64 || col.kind == SN_K_RLE_I64;
 #pragma unroll
       for (int k = 0; k < CHUNK / WG; k++) {
         int r = tid + k * WG;
@@ -461,29 +471,24 @@ __device__ __forceinline__ int eval_preds(const sn_dev_plan *P, int npd, int npi
   return alive;
 }
 
-/* aggregate input value: three neutral-padded fmas, no branches */
-__device__ __forceinline__ double eval_agg_clean(const sn_dev_plan *P, int a,
-                                                 const double *sval, int r) {
-  const sn_dev_agg &A = P->aggs[a];
-  const double x0 = sval[(size_t)A.c0 * CHUNK + r];
-  const double x1 = sval[(size_t)A.c1 * CHUNK + r];
-  const double x2 = sval[(size_t)A.c2 * CHUNK + r];
-  return (A.a0 + A.m0 * x0) * (A.a1 + A.m1 * x1) * (A.a2 + A.m2 * x2);
+/* aggregate-factor read: i64 column slots hold RAW BITS in the LDS image
+ * (so integer predicates stay exact); aggregate arithmetic is double, so
+ * convert on read.  The mask test is scalar-uniform — a scalar branch,
+ * free for plans without i64 factors. */
+__device__ __forceinline__ double sv_agg(const sn_dev_plan *P,
+                                         const double *sval, int cs, int r) {
+  const double x = sval[(size_t)cs * CHUNK + r];
+  const double xc = (double)__double_as_longlong(x);   /* select, not branch */
+  return ((P->i64_mask >> cs) & 1u) ? xc : x;
 }
 
-__device__ __forceinline__ double eval_agg_general(const sn_dev_plan *P, int a,
-                                                   const double *sval, int r,
-                                                   const uint64_t *svalid,
-                                                   int *anull) {
-  const sn_dev_agg &A = P->aggs[a];
-  *anull = 0;
-  if (A.nf >= 1)
-    *anull |= !(int)((svalid[(size_t)A.c0 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
-  if (A.nf >= 2)
-    *anull |= !(int)((svalid[(size_t)A.c1 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
-  if (A.nf >= 3)
-    *anull |= !(int)((svalid[(size_t)A.c2 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
-  return eval_agg_clean(P, a, sval, r);
+/* aggregate input value: three neutral-padded fmas */
+__device__ __forceinline__ double eval_agg(const sn_dev_plan *P,
+                                           const sn_dev_agg &A,
+                                           const double *sval, int r) {
+  return (A.a0 + A.m0 * sv_agg(P, sval, A.c0, r)) *
+         (A.a1 + A.m1 * sv_agg(P, sval, A.c1, r)) *
+         (A.a2 + A.m2 * sv_agg(P, sval, A.c2, r));
 }
 
 /* wave (64-lane) sum reduction */
@@ -714,11 +719,7 @@ __global__ void k_keyless(sn_dev_plan plan,
 #pragma unroll
           for (int a = 0; a < NAGGS; a++) {
             if (a >= naggs) break;
-            const sn_dev_agg &A = P->aggs[a];
-            const double va =
-                (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
-                (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
-                (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+            const double va = eval_agg(P, P->aggs[a], sval, r);
             sums[a] += ok ? va : 0.0;
             cnts[a] += ok ? 1.0 : 0.0;
           }
@@ -750,9 +751,7 @@ __global__ void k_keyless(sn_dev_plan plan,
           }
           if (w == 0) continue;
           const int m = (int)((w >> (tid & 63)) & 1ull);
-          const double val = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
-                             (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
-                             (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+          const double val = eval_agg(P, A, sval, r);
           sums[a] += m ? val : 0.0;
           cnts[a] += m ? 1.0 : 0.0;
         }
@@ -917,12 +916,8 @@ __global__ void k_grouped(sn_dev_plan plan,
           if (w == 0) continue;
           const int m = (int)((w >> (tid & 63)) & 1ull);
           const int slot = sslot[r];
-          const double va = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
-                            (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
-                            (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
-          const double vb = (B2.a0 + B2.m0 * sval[(size_t)B2.c0 * CHUNK + r]) *
-                            (B2.a1 + B2.m1 * sval[(size_t)B2.c1 * CHUNK + r]) *
-                            (B2.a2 + B2.m2 * sval[(size_t)B2.c2 * CHUNK + r]);
+          const double va = eval_agg(P, A, sval, r);
+          const double vb = eval_agg(P, B2, sval, r);
 #pragma unroll
           for (int s = 0; s < NSLOTS; s++) {
             /* fma(md, x, acc) is bit-identical to the select+add form and
@@ -1033,14 +1028,8 @@ void k_grouped_lds(sn_dev_plan plan,
                           P->gbase[1]);
         }
         double *row_acc = bacc + (size_t)slot * (naggs + 1);
-        for (int a = 0; a < naggs; a++) {
-          const sn_dev_agg &A = P->aggs[a];
-          const double va =
-              (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
-              (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
-              (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
-          atomicAdd(&row_acc[a], va);
-        }
+        for (int a = 0; a < naggs; a++)
+          atomicAdd(&row_acc[a], eval_agg(P, P->aggs[a], sval, r));
         atomicAdd(&row_acc[naggs], 1.0);
       }
       __syncthreads();
@@ -1157,10 +1146,7 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
 #pragma unroll
         for (int a = 0; a < NA; a++) {
           if (a >= naggs) { va[a] = 0.0; continue; }
-          const sn_dev_agg &A = P->aggs[a];
-          va[a] = (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
-                  (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
-                  (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
+          va[a] = eval_agg(P, P->aggs[a], sval, r);
         }
 #pragma unroll
         for (int s = 0; s < NSLOTS; s++) {
@@ -1264,14 +1250,8 @@ void k_grouped_global(sn_dev_plan plan,
             slot += (long long)sval[(size_t)gc1 * CHUNK + r] - P->gbase[1];
         }
         GAS double *row_acc = acc + (size_t)slot * (naggs + 1);
-        for (int a = 0; a < naggs; a++) {
-          const sn_dev_agg &A = P->aggs[a];
-          const double va =
-              (A.a0 + A.m0 * sval[(size_t)A.c0 * CHUNK + r]) *
-              (A.a1 + A.m1 * sval[(size_t)A.c1 * CHUNK + r]) *
-              (A.a2 + A.m2 * sval[(size_t)A.c2 * CHUNK + r]);
-          (void)atomicAdd((double *)&row_acc[a], va);
-        }
+        for (int a = 0; a < naggs; a++)
+          (void)atomicAdd((double *)&row_acc[a], eval_agg(P, P->aggs[a], sval, r));
         (void)atomicAdd((double *)&row_acc[naggs], 1.0);
       }
       __syncthreads();
@@ -1310,8 +1290,9 @@ __global__ void k_reduce(const double *__restrict__ scratch, int nblocks,
 
 /* put-time patch materialization: write value-only update patches straight
  * into the (null-free, fixed-width) device body so the batch scans clean.
- * Values are plain doubles (decode_delta widening); integer bodies take the
- * round-nearest bits, matching read_general's __double2ll_rn. */
+ * Values are plain doubles (decode_delta widening) except INT64, whose
+ * patch values travel as raw bits (exact beyond 2^53); the narrower integer
+ * bodies take round-nearest, matching read_general's __double2ll_rn. */
 __global__ void k_patch_apply(void *__restrict__ body,
                               const int32_t *__restrict__ pos,
                               const double *__restrict__ val, int n,
@@ -1324,7 +1305,8 @@ __global__ void k_patch_apply(void *__restrict__ body,
     case SN_K_F64: ((GAS double *)(uintptr_t)body)[p] = v; break;
     case SN_K_F32: ((GAS float *)(uintptr_t)body)[p] = (float)v; break;
     case SN_K_I32: ((GAS int32_t *)(uintptr_t)body)[p] = (int32_t)__double2ll_rn(v); break;
-    case SN_K_I64: ((GAS long long *)(uintptr_t)body)[p] = __double2ll_rn(v); break;
+    /* INT64 patch values are raw bits (exact beyond 2^53) */
+    case SN_K_I64: ((GAS long long *)(uintptr_t)body)[p] = __double_as_longlong(v); break;
     case SN_K_I16: ((GAS int16_t *)(uintptr_t)body)[p] = (int16_t)__double2ll_rn(v); break;
   }
 }
