@@ -1,0 +1,221 @@
+"""Round-2 surface widening, parity-tested against the oracle:
+- RLE int64 columns on the GENERAL path (delete mask forces read_general) —
+  the raw-bit run values must not be rounded (advisor regression)
+- int64 aggregate factors (SUM(bigint) bit-exact under the stats-proven
+  2^53 bound; int64 inside double-typed expressions within 1e-6)
+- int-typed dictionary columns (materialized to plain bodies at put)
+- group keys longer than SN_KEY_MAX-1 rejected loudly (never silently
+  merged by truncation)
+"""
+import numpy as np
+import pytest
+
+from oracle import pyoracle as po
+from snappydata_amd import abi, engine as se
+
+REL = 1e-6
+
+
+@pytest.fixture
+def eng():
+    e = se.Engine(device=0)
+    yield e
+    e.close()
+
+
+def assert_close(got, exp, rel=REL):
+    assert abs(got - exp) <= rel * max(1.0, abs(exp)), (got, exp)
+
+
+@pytest.mark.gpu
+def test_rle_int64_general_path_with_deletes(eng):
+    """RLE int64 + delete mask: the delete forces the general read path,
+    whose RLE case must bitcast the raw-bit run values (SN_K_RLE_I64), not
+    round the bit pattern (a value of 100 would read as 0)."""
+    n = 120_000
+    rng = np.random.default_rng(17)
+    runs = np.repeat(rng.integers(1, 200, 3000), rng.integers(10, 90, 3000))[:n]
+    i64 = (runs.astype(np.int64) * 5) + (1 << 58)   # beyond 2^53 exactness
+    w = rng.random(n)
+    dels = np.unique(rng.integers(0, n, n // 20)).astype(np.int32)
+    dmask = se.encode_delete_mask(dels, n)
+    cols = [po.encode(po.T_INT64, po.ENC_RLE, i64),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
+    t = eng.table_define("trle64", [(abi.T_INT64, False), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 1, 0, n, cols, delete_mask=dmask)
+    cut = int(np.median(i64))
+    plan_kw = dict(preds=[dict(col=0, lo=cut, lo_strict=True)],
+                   aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_INT64, po.T_DOUBLE])
+    ot.add_batch(n, cols, delete_mask=dmask)
+    orows = po.result_rows(ot.query(po.make_plan(**plan_kw)))
+    assert grows[0][1][1] == orows[0][1][1]            # COUNT bit-exact
+    assert_close(grows[0][1][0], orows[0][1][0])
+    # numpy cross-check
+    alive = np.ones(n, dtype=bool)
+    alive[dels] = False
+    m = alive & (i64 > cut)
+    assert grows[0][1][1] == float(m.sum()) and m.sum() > 0
+
+
+@pytest.mark.gpu
+def test_int64_sum_bit_exact(eng):
+    """SUM over an int64 column: bit-exact (stats prove every partial sum
+    stays below 2^53); also int64 as a factor of a double expression."""
+    n = 300_000
+    rng = np.random.default_rng(23)
+    v = rng.integers(-10**9, 10**9, n).astype(np.int64)
+    d = rng.random(n)
+    t = eng.table_define("ti64agg", [(abi.T_INT64, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": v}, {"data": d}], n, batch_rows=70_000)
+    plan_kw = dict(preds=[dict(col=0, lo=0)],
+                   aggs=[("sum", [(0, 0.0, 1.0)]),                 # SUM(bigint)
+                         ("sum", [(0, 0.0, 1.0), (1, 0.0, 1.0)]),  # i64 * double
+                         ("avg", [(0, 0.0, 1.0)]),
+                         ("count", [])])
+    q = eng.query(abi.make_plan(table=t, **plan_kw))
+    grows = q.rows()
+    assert q.used_jit()        # i64 factors stay on the query-compiled path
+    ot = po.OracleTable([po.T_INT64, po.T_DOUBLE])
+    ot.add_batch(n, [po.encode(po.T_INT64, po.ENC_UNCOMPRESSED, v),
+                     po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, d)])
+    orows = po.result_rows(ot.query(po.make_plan(**plan_kw)))
+    m = v >= 0
+    assert grows[0][1][0] == orows[0][1][0] == float(v[m].sum())
+    assert_close(grows[0][1][1], orows[0][1][1])
+    assert_close(grows[0][1][2], orows[0][1][2])
+    assert grows[0][1][3] == orows[0][1][3] == float(m.sum())
+
+
+@pytest.mark.gpu
+def test_int64_sum_rejects_unprovable_range(eng):
+    """SUM(bigint) whose stats bound reaches 2^53 must fail loudly, not
+    round silently (the reference's LongType sum is exact)."""
+    n = 10_000
+    v = np.full(n, 1 << 55, dtype=np.int64)
+    t = eng.table_define("ti64big", [(abi.T_INT64, False)])
+    eng.ingest_columns(t, [{"data": v}], n, batch_rows=n)
+    with pytest.raises(se.EngineError) as ei:
+        eng.query(abi.make_plan(table=t, aggs=[("sum", [(0, 0.0, 1.0)])]))
+    assert "f64-exact" in str(ei.value)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("dt,pdt,vals", [
+    ("i32", po.T_INT32, lambda rng, n: rng.integers(0, 40, n).astype(np.int32) * 7),
+    ("i64", po.T_INT64, lambda rng, n: (rng.integers(0, 40, n).astype(np.int64) << 40) + 3),
+])
+def test_int_dictionary_columns(eng, dt, pdt, vals):
+    """Int-typed dictionary columns (DictionaryEncoding over int32/int64):
+    materialized to plain bodies at put; predicates exact (raw bits for i64)."""
+    n = 150_000
+    rng = np.random.default_rng(29)
+    v = vals(rng, n)
+    w = rng.random(n)
+    abit = abi.T_INT32 if dt == "i32" else abi.T_INT64
+    cols = [po.encode(pdt, po.ENC_DICT, v),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
+    t = eng.table_define("tdict_" + dt, [(abit, False), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 1, 0, n, cols)
+    cut = int(np.median(v))
+    plan_kw = dict(preds=[dict(col=0, hi=cut)],
+                   aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([pdt, po.T_DOUBLE])
+    ot.add_batch(n, cols)
+    orows = po.result_rows(ot.query(po.make_plan(**plan_kw)))
+    assert grows[0][1][1] == orows[0][1][1]
+    assert_close(grows[0][1][0], orows[0][1][0])
+    m = v <= cut
+    assert grows[0][1][1] == float(m.sum()) and 0 < m.sum() < n
+
+
+@pytest.mark.gpu
+def test_int_dictionary_nullable_with_sentinel(eng):
+    """Nullable int dictionary: nulls from the bitset AND from the
+    index==numElements sentinel both fold into the materialized bitset."""
+    n = 50_000
+    rng = np.random.default_rng(31)
+    v = rng.integers(0, 25, n).astype(np.int32)
+    valid = (rng.random(n) > 0.1).astype(np.uint8)
+    w = rng.random(n)
+    cols = [po.encode(po.T_INT32, po.ENC_DICT, v, valid=valid),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
+    t = eng.table_define("tdictnull", [(abi.T_INT32, True), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 1, 0, n, cols)
+    plan_kw = dict(preds=[dict(col=0, lo=10)],
+                   aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_INT32, po.T_DOUBLE], nullable=[True, False])
+    ot.add_batch(n, cols)
+    orows = po.result_rows(ot.query(po.make_plan(**plan_kw)))
+    assert grows[0][1][1] == orows[0][1][1]
+    assert_close(grows[0][1][0], orows[0][1][0])
+    m = (valid == 1) & (v >= 10)
+    assert grows[0][1][1] == float(m.sum())
+
+
+@pytest.mark.gpu
+def test_long_group_key_rejected(eng):
+    """A >47-byte dictionary entry on a group column fails the query loudly
+    (SN_ERR_UNSUPPORTED) instead of silently merging truncated keys."""
+    n = 1000
+    longkey = b"X" * 60
+    keys = [longkey if i % 2 else b"short" for i in range(n)]
+    lens = np.array([len(k) for k in keys], dtype=np.int32)
+    payload = b"".join(keys)
+    w = np.ones(n)
+    t = eng.table_define("tlongkey", [(abi.T_STRING, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": payload, "lens": lens}, {"data": w}], n,
+                       batch_rows=n)
+    with pytest.raises(se.EngineError) as ei:
+        eng.query(abi.make_plan(table=t, group_cols=[0],
+                                aggs=[("sum", [(1, 0.0, 1.0)])]))
+    assert "key" in str(ei.value)
+    # non-grouping queries over the same table still run
+    rows = eng.query(abi.make_plan(table=t,
+                                   aggs=[("count", [])])).rows()
+    assert rows[0][1][0] == float(n)
+
+
+def test_oracle_int64_delta_exact_cpu():
+    """Oracle-side: int64 update deltas keep exactness beyond 2^53."""
+    n = 10_000
+    base = np.arange(n, dtype=np.int64) + (1 << 57)
+    pos = np.arange(0, n, 7, dtype=np.int32)
+    newv = base[pos] + 3
+    d1 = po.encode_delta(po.T_INT64, po.ENC_UNCOMPRESSED, pos, n, newv)
+    t = po.OracleTable([po.T_INT64])
+    t.add_batch(-n, [po.encode(po.T_INT64, po.ENC_UNCOMPRESSED, base)],
+                deltas=[(d1, None)])
+    cut = int(base[pos[5]] + 3)
+    rows = po.result_rows(t.query(po.make_plan(
+        preds=[dict(col=0, lo=cut, hi=cut)], aggs=[("count", [])])))
+    merged = base.copy()
+    merged[pos] = newv
+    assert rows[0][1][0] == float((merged == cut).sum())
+
+
+@pytest.mark.gpu
+def test_engine_int64_delta_exact(eng):
+    """Engine-side: int64 update-delta values travel as raw bits (patch
+    materialization writes them back exactly)."""
+    n = 20_000
+    base = np.arange(n, dtype=np.int64) + (1 << 57)
+    w = np.ones(n)
+    pos = np.arange(0, n, 11, dtype=np.int32)
+    newv = base[pos] + 5
+    d1 = se.encode_update_delta(abi.T_INT64, pos, n, newv)
+    cols = [po.encode(po.T_INT64, po.ENC_UNCOMPRESSED, base),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
+    t = eng.table_define("tdelta64", [(abi.T_INT64, False), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 1, 0, -n, cols,
+                  deltas=[(d1, None), (None, None)])
+    cut = int(base[pos[7]] + 5)
+    grows = eng.query(abi.make_plan(
+        table=t, preds=[dict(col=0, lo=cut, hi=cut)],
+        aggs=[("count", [])])).rows()
+    merged = base.copy()
+    merged[pos] = newv
+    assert grows[0][1][0] == float((merged == cut).sum()) == 1.0
