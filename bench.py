@@ -335,18 +335,33 @@ def main():
                 # grouped exchange — BASELINE's 8-GPU Q1 shape): rank r keeps
                 # only groups hashing to shard r; each rank merges the
                 # world blocks it received and holds its final key subset
-                # (the reference's hash-partitioned partial->final shuffle)
-                n = q.partial_bytes()
+                # (the reference's hash-partitioned partial->final shuffle).
+                # Block capacity is negotiated per step (allreduce-MAX of
+                # local group counts) so overflow group-bys — the north-star
+                # "hash table overflows one GPU" demo — exchange cleanly.
+                cap_t = torch.tensor([max(1024, q.num_groups())],
+                                     dtype=torch.int64,
+                                     device=f"cuda:{local_rank}")
+                dist.all_reduce(cap_t, op=dist.ReduceOp.MAX)
+                cap = int(cap_t.item())
+                n = q.partial_bytes(cap)
                 send = torch.from_numpy(
-                    q.partials_sharded(world).reshape(-1)).to(f"cuda:{local_rank}")
+                    q.partials_sharded(world, cap).reshape(-1)).to(
+                        f"cuda:{local_rank}")
                 recv = torch.empty_like(send)
                 dist.all_to_all_single(recv, send)
                 blocks = recv.cpu().numpy()
                 q.merge_host(np.ascontiguousarray(blocks), n, world)
             else:
                 # RCCL all_gather of self-describing grouped partial blocks
-                n = q.partial_bytes()
-                local = torch.from_numpy(q.partials_host()).to(f"cuda:{local_rank}")
+                cap_t = torch.tensor([max(1024, q.num_groups())],
+                                     dtype=torch.int64,
+                                     device=f"cuda:{local_rank}")
+                dist.all_reduce(cap_t, op=dist.ReduceOp.MAX)
+                cap = int(cap_t.item())
+                n = q.partial_bytes(cap)
+                local = torch.from_numpy(q.partials_host(cap)).to(
+                    f"cuda:{local_rank}")
                 gathered = torch.zeros(world * n, dtype=torch.uint8,
                                        device=f"cuda:{local_rank}")
                 dist.all_gather_into_tensor(gathered, local)

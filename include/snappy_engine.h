@@ -186,7 +186,12 @@ typedef struct {
   int32_t npreds;
   sn_pred preds[SN_MAX_PREDS];
   int32_t ngroup;       /* 0 = keyless aggregate */
-  int32_t group_cols[SN_MAX_GROUPS]; /* dictionary-encoded string columns */
+  int32_t group_cols[SN_MAX_GROUPS]; /* dictionary string columns, or integer
+                           (int16/int32/int64) key columns.  Integer keys
+                           with dense stats-derived spans use direct slots;
+                           sparse/unbounded integer keys run the
+                           open-address hash aggregate (ByteBufferHashMap /
+                           SHAMapAccessor analogue) */
   int32_t naggs;
   int32_t _pad;
   sn_agg  aggs[SN_MAX_AGGS];
@@ -307,6 +312,18 @@ int32_t   sn_query_used_jit(sn_query *q);
 int64_t sn_query_partial_bytes(sn_query *q);
 /* export this shard's partial block; dst_is_device: 1 = HIP device memory */
 int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_device);
+/* variable-capacity variants for group counts beyond SN_MAX_GROUP_SLOTS
+ * (big dense slot spaces and the sparse hash aggregate): the caller
+ * negotiates one capacity across ranks (e.g. an allreduce-MAX of
+ * sn_query_num_groups) so every block in the collective is the same size —
+ * mirroring how the reference's hash-partitioned partial->final exchange
+ * sizes its shuffle blocks dynamically (SnappyHashAggregateExec.scala:
+ * 161-167).  Block layout is unchanged; capacity rides in the header. */
+int64_t sn_query_partial_bytes2(sn_query *q, int32_t cap_slots);
+int32_t sn_query_partials2(sn_query *q, void *dst, int32_t dst_is_device,
+                           int32_t cap_slots);
+int32_t sn_query_partials_sharded2(sn_query *q, int32_t world, void *dst,
+                                   int32_t cap_slots);
 /* key-sharded split for the grouped all-to-all: write
  * `world` same-format blocks into dst (world * partial_bytes), block d
  * holding only the groups whose key hashes to shard d.  After the

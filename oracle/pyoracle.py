@@ -100,6 +100,12 @@ def lib():
                                            C.POINTER(C.c_int32)]
         _lib.sno_query.restype = C.c_int32
         _lib.sno_query.argtypes = [C.c_void_p, C.POINTER(SnPlan), C.POINTER(SnResult), C.c_int32]
+        _lib.sno_query_groups.restype = C.c_int64
+        _lib.sno_query_groups.argtypes = [C.c_void_p, C.POINTER(SnPlan),
+                                          C.c_int32, C.c_int64, C.c_char_p,
+                                          C.POINTER(C.c_uint8),
+                                          C.POINTER(C.c_double),
+                                          C.POINTER(C.c_uint8)]
         _lib.sno_encode.restype = C.c_int64
         _lib.sno_encode.argtypes = [C.c_int32, C.c_int32, C.c_int32, C.c_void_p,
                                     C.POINTER(C.c_int32), C.POINTER(C.c_uint8),
@@ -330,6 +336,34 @@ class OracleTable:
         rc = lib().sno_query(self._h, C.byref(plan), C.byref(res), nthreads)
         assert rc == 0, f"sno_query failed: {rc}"
         return res
+
+    def query_groups(self, plan, nthreads=1, cap=1 << 21):
+        """Flat result for group counts beyond the sn_result page: returns
+        the full sorted [(keys tuple, vals list), ...] like result_rows()."""
+        keys = np.zeros((cap, SN_MAX_GROUPS, SN_KEY_MAX), dtype=np.uint8)
+        knull = np.zeros((cap, SN_MAX_GROUPS), dtype=np.uint8)
+        vals = np.zeros((cap, SN_MAX_AGGS), dtype=np.float64)
+        vnull = np.zeros((cap, SN_MAX_AGGS), dtype=np.uint8)
+        n = lib().sno_query_groups(
+            self._h, C.byref(plan), nthreads, cap,
+            keys.ctypes.data_as(C.c_char_p),
+            knull.ctypes.data_as(C.POINTER(C.c_uint8)),
+            vals.ctypes.data_as(C.POINTER(C.c_double)),
+            vnull.ctypes.data_as(C.POINTER(C.c_uint8)))
+        assert n >= 0, f"sno_query_groups failed: {n}"
+        assert n <= cap, f"{n} groups exceed cap {cap}"
+        ngroup = plan.ngroup if plan.join_dim < 0 or plan.join_mode != 1 else 1
+        naggs = plan.naggs
+        out = []
+        for i in range(n):
+            ks = tuple(
+                None if knull[i][k] else
+                bytes(keys[i][k]).split(b"\0")[0].decode()
+                for k in range(ngroup))
+            vs = [None if vnull[i][a] else float(vals[i][a])
+                  for a in range(naggs)]
+            out.append((ks, vs))
+        return out
 
     def __del__(self):
         try:

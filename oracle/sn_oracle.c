@@ -837,6 +837,7 @@ typedef struct {
   sno_group *groups;
   int32_t n, cap;
   int32_t ngroup_cols, naggs;
+  int32_t max_groups;     /* hard cardinality bound */
   /* hash index: key -> group slot */
   int32_t *index;  /* size index_cap, -1 empty */
   int32_t index_cap;
@@ -852,9 +853,13 @@ static uint64_t key_hash(const char (*keys)[SN_KEY_MAX], const uint8_t *knull, i
   return h;
 }
 
+/* max_groups 0 -> default 1<<22: the oracle follows ByteBufferHashMap's
+ * grow-and-rehash (ByteBufferHashMap.scala:245-292) so cardinalities well
+ * beyond the engine's one-page result bound are checkable */
 static int gtab_init(sno_gtab *g, int ngroup_cols, int naggs) {
   memset(g, 0, sizeof(*g));
   g->ngroup_cols = ngroup_cols; g->naggs = naggs;
+  g->max_groups = 1 << 22;
   g->cap = 64;
   g->groups = (sno_group *)calloc((size_t)g->cap, sizeof(sno_group));
   g->index_cap = 4 * SN_MAX_GROUP_SLOTS;
@@ -864,14 +869,32 @@ static int gtab_init(sno_gtab *g, int ngroup_cols, int naggs) {
 }
 static void gtab_free(sno_gtab *g) { free(g->groups); free(g->index); }
 
+static void gtab_rehash(sno_gtab *g) {
+  int32_t ncap = g->index_cap * 2;
+  int32_t *ni = (int32_t *)malloc(sizeof(int32_t) * (size_t)ncap);
+  if (!ni) return;                 /* keep probing the old (denser) index */
+  memset(ni, 0xff, sizeof(int32_t) * (size_t)ncap);
+  for (int32_t gi = 0; gi < g->n; gi++) {
+    uint64_t h = key_hash((const char (*)[SN_KEY_MAX])g->groups[gi].keys,
+                          g->groups[gi].key_null, g->ngroup_cols);
+    int32_t s = (int32_t)(h & (uint64_t)(ncap - 1));
+    while (ni[s] >= 0) s = (s + 1) & (ncap - 1);
+    ni[s] = gi;
+  }
+  free(g->index);
+  g->index = ni;
+  g->index_cap = ncap;
+}
+
 static sno_group *gtab_get(sno_gtab *g, const char (*keys)[SN_KEY_MAX],
                            const uint8_t *knull) {
+  if (g->n * 2 >= g->index_cap) gtab_rehash(g);
   uint64_t h = key_hash(keys, knull, g->ngroup_cols);
   int32_t slot = (int32_t)(h & (uint64_t)(g->index_cap - 1));
   while (1) {
     int32_t gi = g->index[slot];
     if (gi < 0) {
-      if (g->n >= SN_MAX_GROUP_SLOTS) return NULL;
+      if (g->n >= g->max_groups) return NULL;
       if (g->n == g->cap) {
         g->cap *= 2;
         g->groups = (sno_group *)realloc(g->groups, sizeof(sno_group) * (size_t)g->cap);
@@ -1270,12 +1293,13 @@ static int group_cmp(const void *a, const void *b) {
   return 0;
 }
 
-SNO_EXPORT int32_t sno_query(sno_table *t, const sn_plan *p, sn_result *out,
-                             int32_t nthreads) {
-  if (!t || !p || !out) return SN_ERR_BADARG;
+/* shared evaluation: fills the group table (sorted) + metrics
+ * m = {rows_scanned, rows_passed, batches_seen, batches_skipped} */
+static int32_t sno_run(sno_table *t, const sn_plan *p, int32_t nthreads,
+                       sno_gtab *gout, int64_t m[4]) {
+  if (!t || !p) return SN_ERR_BADARG;
   if (p->ngroup > SN_MAX_GROUPS || p->naggs > SN_MAX_AGGS || p->npreds > SN_MAX_PREDS)
     return SN_ERR_BADARG;
-  memset(out, 0, sizeof(*out));
   int rc = SN_OK;
   const int join_group = p->join_dim != SN_JOIN_NONE &&
                          p->join_mode == SN_JOIN_GROUP && t->dim_hk != NULL;
@@ -1361,31 +1385,84 @@ SNO_EXPORT int32_t sno_query(sno_table *t, const sn_plan *p, sn_result *out,
   }
 
   qsort(g.groups, (size_t)g.n, sizeof(sno_group), group_cmp);
+  *gout = g;                     /* caller owns (gtab_free) */
+  m[0] = rows_scanned; m[1] = rows_passed; m[2] = seen; m[3] = skipped;
+  return SN_OK;
+}
 
+/* fill the agg value + null flag per Spark semantics */
+static void sno_fill_val(const sn_agg *ag, const sno_group *grp, int a,
+                         double *val, uint8_t *isnull) {
+  if (ag->kind == SN_AGG_COUNT_STAR) { *val = grp->sums[a]; *isnull = 0; }
+  else if (ag->kind == SN_AGG_AVG) {
+    if (grp->counts[a] > 0) { *val = grp->sums[a] / grp->counts[a]; *isnull = 0; }
+    else { *val = 0; *isnull = 1; }
+  } else {
+    if (grp->counts[a] > 0) { *val = grp->sums[a]; *isnull = 0; }
+    else { *val = 0; *isnull = 1; }
+  }
+}
+
+SNO_EXPORT int32_t sno_query(sno_table *t, const sn_plan *p, sn_result *out,
+                             int32_t nthreads) {
+  if (!out) return SN_ERR_BADARG;
+  memset(out, 0, sizeof(*out));
+  sno_gtab g;
+  int64_t m[4] = { 0, 0, 0, 0 };
+  int32_t rc = sno_run(t, p, nthreads, &g, m);
+  if (rc != SN_OK) return rc;
+  const int join_group = p->join_dim != SN_JOIN_NONE &&
+                         p->join_mode == SN_JOIN_GROUP && t->dim_hk != NULL;
+  const int eff_ngroup = join_group ? 1 : p->ngroup;
+  if (g.n > SN_MAX_GROUP_SLOTS) {     /* fixed result page: use sno_query_groups */
+    gtab_free(&g);
+    return SN_ERR_OVERFLOW;
+  }
   out->nrows = g.n; out->ngroup = eff_ngroup; out->naggs = p->naggs;
-  out->rows_scanned = rows_scanned; out->rows_passed = rows_passed;
-  out->batches_seen = seen; out->batches_skipped = skipped;
+  out->rows_scanned = m[0]; out->rows_passed = m[1];
+  out->batches_seen = m[2]; out->batches_skipped = m[3];
   for (int32_t i = 0; i < g.n; i++) {
     sno_group *grp = &g.groups[i];
     for (int k = 0; k < eff_ngroup; k++) {
       strncpy(out->keys[i][k], grp->keys[k], SN_KEY_MAX);
       out->key_is_null[i][k] = grp->key_null[k];
     }
-    for (int a = 0; a < p->naggs; a++) {
-      const sn_agg *ag = &p->aggs[a];
-      if (ag->kind == SN_AGG_COUNT_STAR) {
-        out->vals[i][a] = grp->sums[a]; out->val_is_null[i][a] = 0;
-      } else if (ag->kind == SN_AGG_AVG) {
-        if (grp->counts[a] > 0) { out->vals[i][a] = grp->sums[a] / grp->counts[a]; out->val_is_null[i][a] = 0; }
-        else { out->vals[i][a] = 0; out->val_is_null[i][a] = 1; }
-      } else { /* SUM */
-        if (grp->counts[a] > 0) { out->vals[i][a] = grp->sums[a]; out->val_is_null[i][a] = 0; }
-        else { out->vals[i][a] = 0; out->val_is_null[i][a] = 1; }
-      }
-    }
+    for (int a = 0; a < p->naggs; a++)
+      sno_fill_val(&p->aggs[a], grp, a, &out->vals[i][a], &out->val_is_null[i][a]);
   }
   gtab_free(&g);
   return SN_OK;
+}
+
+/* flat-array result export for group cardinalities beyond the sn_result
+ * page (the oracle leg of the engine's sparse hash-aggregate / big-group
+ * parity tests).  keys = [cap][SN_MAX_GROUPS][SN_KEY_MAX] char,
+ * knull = [cap][SN_MAX_GROUPS], vals/vnull = [cap][SN_MAX_AGGS].
+ * Returns the TOTAL group count (sorted; only min(total, cap) rows are
+ * written) or a negative error. */
+SNO_EXPORT int64_t sno_query_groups(sno_table *t, const sn_plan *p,
+                                    int32_t nthreads, int64_t cap,
+                                    char *keys, uint8_t *knull,
+                                    double *vals, uint8_t *vnull) {
+  if (!keys || !knull || !vals || !vnull || cap < 0) return SN_ERR_BADARG;
+  sno_gtab g;
+  int64_t m[4] = { 0, 0, 0, 0 };
+  int32_t rc = sno_run(t, p, nthreads, &g, m);
+  if (rc != SN_OK) return rc;
+  int64_t n = g.n < cap ? g.n : cap;
+  for (int64_t i = 0; i < n; i++) {
+    sno_group *grp = &g.groups[i];
+    for (int k = 0; k < SN_MAX_GROUPS; k++) {
+      memcpy(keys + (i * SN_MAX_GROUPS + k) * SN_KEY_MAX, grp->keys[k], SN_KEY_MAX);
+      knull[i * SN_MAX_GROUPS + k] = grp->key_null[k];
+    }
+    for (int a = 0; a < p->naggs; a++)
+      sno_fill_val(&p->aggs[a], grp, a, &vals[i * SN_MAX_AGGS + a],
+                   &vnull[i * SN_MAX_AGGS + a]);
+  }
+  int64_t total = g.n;
+  gtab_free(&g);
+  return total;
 }
 
 /* =======================================================================

@@ -235,6 +235,15 @@ struct sn_engine {
    * (queries on one engine serialize on its stream) */
   double *scratch = nullptr;
   size_t scratch_sz = 0;
+  /* sparse hash-aggregate workspace (reused: the sparse path reads its
+   * results back inside submit, so no two queries share it live) */
+  long long *hws_keys = nullptr;
+  long long *hws_okeys = nullptr;
+  double *hws_acc = nullptr;
+  double *hws_orows = nullptr;
+  int32_t *hws_flags = nullptr;   /* [0] overflow, [1] compact counter */
+  int hws_cap_log2 = 0;
+  size_t hws_acc_bytes = 0;
   /* per-engine cache of query-compiled kernels (jit.cpp) */
   void *jit = nullptr;
   /* steady-state submit caches: device plan copies by content hash, and a
@@ -1160,6 +1169,9 @@ struct sn_query {
   std::vector<int32_t> used_cols;       /* cslot -> table col */
   int nslots = 0;
   int g1cap = 0, g2cap = 0;             /* per-group-col slot counts */
+  bool sparse = false;                  /* open-address hash-aggregate mode */
+  std::vector<long long> sparse_keys;   /* compacted group keys */
+  std::vector<double> sparse_rows;      /* [n][dev_naggs+1] accumulator rows */
   bool gint[2] = { false, false };      /* integer group key (stats-ranged) */
   int64_t gmin[2] = { 0, 0 };           /* integer key minimum (slot base) */
   int gnull1 = -1, gnull2 = -1;         /* null slot index per group col (-1: none) */
@@ -1225,6 +1237,30 @@ static bool batch_skippable(const Batch &b, const sn_plan *p, const Table *t) {
   return false;
 }
 
+/* engine-cached device workspace for the open-address hash aggregate */
+static int ensure_sparse_ws(sn_engine *e, int cap_log2, int naggs1) {
+  size_t cap = 1ull << cap_log2;
+  if (!e->hws_keys || e->hws_cap_log2 < cap_log2) {
+    e->hws_keys = (long long *)e->arena.alloc(cap * 8);
+    e->hws_okeys = (long long *)e->arena.alloc((cap + 1) * 8);
+    e->hws_cap_log2 = cap_log2;
+    e->hws_acc_bytes = 0;
+    if (!e->hws_keys || !e->hws_okeys) return SN_ERR_NOMEM;
+  }
+  size_t accb = ((1ull << e->hws_cap_log2) + 1) * (size_t)naggs1 * 8;
+  if (e->hws_acc_bytes < accb) {
+    e->hws_acc = (double *)e->arena.alloc(accb);
+    e->hws_orows = (double *)e->arena.alloc(accb);
+    e->hws_acc_bytes = accb;
+    if (!e->hws_acc || !e->hws_orows) return SN_ERR_NOMEM;
+  }
+  if (!e->hws_flags) {
+    e->hws_flags = (int32_t *)e->arena.alloc(64);
+    if (!e->hws_flags) return SN_ERR_NOMEM;
+  }
+  return SN_OK;
+}
+
 extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   if (!e || !plan) { fail(SN_ERR_BADARG, "null engine/plan"); return nullptr; }
   Table *t = get_table(e, plan->table);
@@ -1262,14 +1298,15 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   for (int i = 0; i < plan->ngroup; i++) {
     int c = plan->group_cols[i];
     sn_type_t gdt = t->schema[c].dtype;
-    bool int_key = gdt == SN_TYPE_INT32 || gdt == SN_TYPE_INT16;
+    bool int_key = gdt == SN_TYPE_INT32 || gdt == SN_TYPE_INT16 ||
+                   gdt == SN_TYPE_INT64;
     if (gdt != SN_TYPE_STRING && !int_key) {
       fail(SN_ERR_UNSUPPORTED,
-           "group-by supports dictionary string and int32/int16 key columns");
+           "group-by supports dictionary string and int16/int32/int64 key columns");
       return nullptr;
     }
     if (int_key && t->schema[c].nullable) {
-      fail(SN_ERR_UNSUPPORTED, "nullable integer group keys not in round-1");
+      fail(SN_ERR_UNSUPPORTED, "nullable integer group keys not supported");
       return nullptr;
     }
     if (use_col(c) < 0) { fail(SN_ERR_BADARG, "bad group col"); return nullptr; }
@@ -1333,16 +1370,15 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
      (the DictionaryOptimizedMapAccessor direct-slot idea applied to ints).
      Requires every batch to carry valid bounds for the key column and no
      update patches on it (patches may move values outside the bounds). */
+  /* dense-slot span of an integer key column, or -1 when the dense path is
+   * unavailable (missing stats bounds, update patches, overflowing span) —
+   * the caller then falls back to the open-address hash aggregate */
   auto int_key_span = [&](int c, int64_t *mn_out) -> int64_t {
     int64_t mn = 0, mx = -1;
     bool first = true;
     for (auto &b : t->batches) {
-      if (!b.stats_valid || b.bounds_null[c])
-        return fail(SN_ERR_UNSUPPORTED,
-                    "integer group key col %d needs stats bounds on every batch", c), -1;
-      if (b.had_patches.size() > (size_t)c && b.had_patches[c])
-        return fail(SN_ERR_UNSUPPORTED,
-                    "integer group key col %d has update patches", c), -1;
+      if (!b.stats_valid || b.bounds_null[c]) return -1;
+      if (b.had_patches.size() > (size_t)c && b.had_patches[c]) return -1;
       if (first) { mn = b.lo_i[c]; mx = b.hi_i[c]; first = false; }
       else {
         mn = b.lo_i[c] < mn ? b.lo_i[c] : mn;
@@ -1368,48 +1404,67 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         return nullptr;
       }
     }
-    int c0 = plan->group_cols[0];
-    if (t->schema[c0].dtype != SN_TYPE_STRING) {
-      int64_t span = int_key_span(c0, &q->gmin[0]);
-      if (span < 0) return nullptr;
-      if (span > SN_BIG_GROUP_CAP) {
-        fail(SN_ERR_UNSUPPORTED, "integer key span %lld > %d", (long long)span,
-             SN_BIG_GROUP_CAP);
+    /* try the dense direct-slot space (dictionary ids / stats-ranged
+     * integers); integer keys whose span is unbounded, statless, patched or
+     * beyond SN_BIG_GROUP_CAP fall back to the open-address hash aggregate
+     * (the ByteBufferHashMap/SHAMap analogue, k_grouped_hash) */
+    bool col_sparse[2] = { false, false };
+    bool any_string = false;
+    int caps[2] = { 1, 1 };
+    for (int i = 0; i < plan->ngroup; i++) {
+      int c = plan->group_cols[i];
+      sn_type_t gdt = t->schema[c].dtype;
+      if (gdt == SN_TYPE_STRING) {
+        any_string = true;
+        caps[i] = (int)t->gdict[c].size() + (t->schema[c].nullable ? 1 : 0);
+        continue;
+      }
+      int64_t span = gdt == SN_TYPE_INT64 ? -1 : int_key_span(c, &q->gmin[i]);
+      if (span < 0 || span > SN_BIG_GROUP_CAP) { col_sparse[i] = true; continue; }
+      q->gint[i] = true;
+      caps[i] = (int)span;
+    }
+    int64_t dense_slots = (int64_t)std::max(caps[0], 1) *
+                          (int64_t)std::max(caps[1], 1);
+    q->sparse = col_sparse[0] || col_sparse[1] ||
+                (!any_string && dense_slots > SN_BIG_GROUP_CAP);
+    if (q->sparse) {
+      if (any_string) {
+        fail(SN_ERR_UNSUPPORTED,
+             "mixed string + sparse integer group keys not supported");
         return nullptr;
       }
-      q->gint[0] = true;
-      q->g1cap = (int)span;
-    } else {
-      q->g1cap = (int)t->gdict[c0].size() + (t->schema[c0].nullable ? 1 : 0);
-    }
-    q->g2cap = 1;
-    if (plan->ngroup == 2) {
-      int c1 = plan->group_cols[1];
-      if (t->schema[c1].dtype != SN_TYPE_STRING) {
-        int64_t span = int_key_span(c1, &q->gmin[1]);
-        if (span < 0) return nullptr;
-        if (span > SN_BIG_GROUP_CAP) {
-          fail(SN_ERR_UNSUPPORTED, "integer key span %lld > %d",
-               (long long)span, SN_BIG_GROUP_CAP);
-          return nullptr;
-        }
-        q->gint[1] = true;
-        q->g2cap = (int)span;
-      } else {
-        q->g2cap = (int)t->gdict[c1].size() + (t->schema[c1].nullable ? 1 : 0);
+      if (plan->ngroup == 2 &&
+          (t->schema[plan->group_cols[0]].dtype == SN_TYPE_INT64 ||
+           t->schema[plan->group_cols[1]].dtype == SN_TYPE_INT64)) {
+        fail(SN_ERR_UNSUPPORTED,
+             "two-column group keys with an int64 column not supported "
+             "(keys pack into one 64-bit word)");
+        return nullptr;
       }
-    }
-    if (!q->gint[0] && t->schema[c0].nullable) q->gnull1 = q->g1cap - 1;
-    if (plan->ngroup == 2 && !q->gint[1] &&
-        t->schema[plan->group_cols[1]].nullable)
-      q->gnull2 = q->g2cap - 1;
-    if (q->g1cap == 0) q->g1cap = 1;
-    if (q->g2cap == 0) q->g2cap = 1;
-    q->nslots = q->g1cap * q->g2cap;
-    if (q->nslots > SN_BIG_GROUP_CAP) {
-      fail(SN_ERR_UNSUPPORTED, "group cardinality %d > %d",
-           q->nslots, SN_BIG_GROUP_CAP);
-      return nullptr;
+      if (jd) {
+        fail(SN_ERR_UNSUPPORTED, "join combined with sparse group keys");
+        return nullptr;
+      }
+      q->gint[0] = q->gint[1] = false;
+      q->nslots = 0;
+      q->g1cap = q->g2cap = 1;
+    } else {
+      q->g1cap = caps[0];
+      q->g2cap = plan->ngroup == 2 ? caps[1] : 1;
+      if (!q->gint[0] && t->schema[plan->group_cols[0]].nullable)
+        q->gnull1 = q->g1cap - 1;
+      if (plan->ngroup == 2 && !q->gint[1] &&
+          t->schema[plan->group_cols[1]].nullable)
+        q->gnull2 = q->g2cap - 1;
+      if (q->g1cap == 0) q->g1cap = 1;
+      if (q->g2cap == 0) q->g2cap = 1;
+      q->nslots = q->g1cap * q->g2cap;
+      if (q->nslots > SN_BIG_GROUP_CAP) {
+        fail(SN_ERR_UNSUPPORTED, "group cardinality %d > %d",
+             q->nslots, SN_BIG_GROUP_CAP);
+        return nullptr;
+      }
     }
   } else if (q->join_group) {
     q->nslots = (int)q->join_dim->attr_dict.size();
@@ -1728,10 +1783,12 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
 
   /* upload + launch */
   size_t out_n = (size_t)q->nslots * q->out_stride;
-  q->dev_out = (double *)e->arena.alloc(out_n * 8);
-  if (!q->dev_out) { fail(SN_ERR_NOMEM, "out alloc"); return nullptr; }
-  if (hipMemsetAsync(q->dev_out, 0, out_n * 8, e->stream) != hipSuccess) {
-    fail(SN_ERR_GENERIC, "memset out"); return nullptr;
+  if (!q->sparse) {
+    q->dev_out = (double *)e->arena.alloc(out_n * 8);
+    if (!q->dev_out) { fail(SN_ERR_NOMEM, "out alloc"); return nullptr; }
+    if (hipMemsetAsync(q->dev_out, 0, out_n * 8, e->stream) != hipSuccess) {
+      fail(SN_ERR_GENERIC, "memset out"); return nullptr;
+    }
   }
   const void *db_dev = nullptr, *tl_dev = nullptr;
   int32_t ntiles = 0;
@@ -1763,6 +1820,93 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       memcpy(dc.jit_kinds, jit_kinds, sizeof(jit_kinds));
       t->desc_caches.push_back(dc);
     }
+  }
+  if (q->sparse && ntiles > 0) {
+    /* ---- open-address hash-aggregate launch (k_grouped_hash) ----
+     * The workspace (key table + accumulators) is engine-cached; results
+     * are compacted on device and read back HERE (inside submit, stream
+     * synchronized), so concurrent queries never share the live table. */
+    const int naggs1 = q->dev_naggs + 1;
+    int cap_log2 = 20;
+    while (cap_log2 > 12 && (1ll << (cap_log2 - 1)) >= 4 * q->rows_scanned)
+      cap_log2--;                      /* small tables: smaller table */
+    q->ev_start = e->ev_acquire();
+    q->ev_stop = e->ev_acquire();
+    bool done_h = false;
+    for (int attempt = 0; attempt < 4 && !done_h; attempt++) {
+      if (ensure_sparse_ws(e, cap_log2, naggs1) != SN_OK) {
+        fail(SN_ERR_NOMEM, "sparse hash workspace"); return nullptr;
+      }
+      const size_t cap = 1ull << cap_log2;
+      sn_dev_plan dps = dp;
+      dps.sparse = 1;
+      dps.hkeys = e->hws_keys;
+      dps.hacc = e->hws_acc;
+      dps.hflags = e->hws_flags;
+      dps.hcap_log2 = cap_log2;
+      void *dps_dev = e->arena.alloc(sizeof(dps));
+      if (!dps_dev ||
+          hipMemcpy(dps_dev, &dps, sizeof(dps), hipMemcpyHostToDevice) != hipSuccess) {
+        fail(SN_ERR_NOMEM, "sparse plan upload"); return nullptr;
+      }
+      if (hipMemsetAsync(e->hws_keys, 0xff, cap * 8, e->stream) != hipSuccess ||
+          hipMemsetAsync(e->hws_acc, 0, (cap + 1) * (size_t)naggs1 * 8,
+                         e->stream) != hipSuccess ||
+          hipMemsetAsync(e->hws_flags, 0, 8, e->stream) != hipSuccess) {
+        fail(SN_ERR_GENERIC, "sparse workspace zero"); return nullptr;
+      }
+      if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
+      int rc = sn_launch_hash_scan(&dps, (const sn_dev_plan *)dps_dev,
+                                   (const sn_dev_batch *)db_dev,
+                                   (const sn_dev_tile *)tl_dev, ntiles,
+                                   e->stream);
+      if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);
+      if (rc != 0) {
+        fail(SN_ERR_GENERIC, "hash-agg launch: %s",
+             hipGetErrorString((hipError_t)rc));
+        return nullptr;
+      }
+      if (hipStreamSynchronize(e->stream) != hipSuccess) {
+        fail(SN_ERR_GENERIC, "hash-agg sync"); return nullptr;
+      }
+      int32_t ovf = 0;
+      (void)hipMemcpy(&ovf, e->hws_flags, 4, hipMemcpyDeviceToHost);
+      if (ovf) {
+        if (cap_log2 >= 24) {
+          fail(SN_ERR_OVERFLOW,
+               "sparse group cardinality exceeds the 2^24 hash table");
+          return nullptr;
+        }
+        cap_log2 = std::min(cap_log2 + 2, 24);
+        continue;
+      }
+      if (hipMemsetAsync(e->hws_flags + 1, 0, 4, e->stream) != hipSuccess ||
+          sn_launch_hash_compact(e->hws_keys, e->hws_acc, (int)cap, naggs1,
+                                 e->hws_okeys, e->hws_orows,
+                                 (int *)(e->hws_flags + 1), e->stream) != 0 ||
+          hipStreamSynchronize(e->stream) != hipSuccess) {
+        fail(SN_ERR_GENERIC, "hash-agg compact"); return nullptr;
+      }
+      int32_t ngrp = 0;
+      (void)hipMemcpy(&ngrp, e->hws_flags + 1, 4, hipMemcpyDeviceToHost);
+      q->sparse_keys.resize((size_t)ngrp);
+      q->sparse_rows.resize((size_t)ngrp * naggs1);
+      if (ngrp > 0) {
+        if (hipMemcpy(q->sparse_keys.data(), e->hws_okeys, (size_t)ngrp * 8,
+                      hipMemcpyDeviceToHost) != hipSuccess ||
+            hipMemcpy(q->sparse_rows.data(), e->hws_orows,
+                      (size_t)ngrp * naggs1 * 8,
+                      hipMemcpyDeviceToHost) != hipSuccess) {
+          fail(SN_ERR_GENERIC, "hash-agg readback"); return nullptr;
+        }
+      }
+      done_h = true;
+    }
+    {
+      std::lock_guard<std::mutex> ga(e->aux_mu);
+      e->live_q.insert(q.get());
+    }
+    return q.release();
   }
   if (ntiles > 0) {
     /* HIP events bracket the scan kernel on ITS stream for the roofline leg
@@ -1871,8 +2015,9 @@ extern "C" int32_t sn_query_wait(sn_query *q) {
     HIP_OR_FAIL(hipStreamSynchronize(q->e->stream));
     size_t out_n = (size_t)q->nslots * q->out_stride;
     q->host_out.resize(out_n);
-    HIP_OR_FAIL(hipMemcpy(q->host_out.data(), q->dev_out, out_n * 8,
-                          hipMemcpyDeviceToHost));
+    if (out_n > 0 && q->dev_out)
+      HIP_OR_FAIL(hipMemcpy(q->host_out.data(), q->dev_out, out_n * 8,
+                            hipMemcpyDeviceToHost));
     if (q->ev_start && q->ev_stop)
       (void)hipEventElapsedTime(&q->kernel_ms, q->ev_start, q->ev_stop);
     q->done = true;
@@ -1903,6 +2048,34 @@ extern "C" int32_t sn_engine_jit_count(sn_engine *e) {
 static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
   const sn_plan &p = q->plan;
   Table *t = q->t;
+  if (q->sparse) {
+    /* hash-aggregate results: compacted (key, accumulator-row) pairs.
+     * Keys surface as decimal text like the dense integer-key path (the
+     * partial-block and result formats stay shared). */
+    const int naggs1 = q->dev_naggs + 1;
+    const bool packed = p.ngroup == 2;
+    for (size_t i = 0; i < q->sparse_keys.size(); i++) {
+      const double *row = &q->sparse_rows[i * naggs1];
+      double rowcount = row[q->dev_naggs];
+      if (rowcount == 0.0) continue;
+      GroupOut g;
+      g.rowcount = rowcount;
+      for (int a = 0; a < p.naggs; a++) {
+        int di = q->agg_map[a];
+        g.sums[a] = di < 0 ? rowcount : row[di];
+        g.counts[a] = rowcount;
+      }
+      long long key = q->sparse_keys[i];
+      if (packed) {
+        g.keys[0] = std::to_string((int32_t)((unsigned long long)key >> 32));
+        g.keys[1] = std::to_string((int32_t)(unsigned)key);
+      } else {
+        g.keys[0] = std::to_string(key);
+      }
+      out->push_back(std::move(g));
+    }
+    return;
+  }
   /* concurrent ingest may be interning new dictionary entries; group ids
    * recorded at submit stay valid (dictionaries only grow) but the vector
    * needs the lock for stability */
@@ -2039,20 +2212,35 @@ extern "C" void sn_query_destroy(sn_query *q) {
   delete q;
 }
 
-/* ---- partial exchange ---- */
-extern "C" int64_t sn_query_partial_bytes(sn_query *q) {
-  if (!q) return SN_ERR_BADARG;
+/* ---- partial exchange ----
+ * Grouped blocks carry a caller-chosen slot CAPACITY (the *2 entry points):
+ * group counts beyond SN_MAX_GROUP_SLOTS — the big dense-slot and sparse
+ * hash-aggregate paths — export by negotiating a capacity across ranks
+ * (e.g. allreduce-MAX of sn_query_num_groups) before the collective.  The
+ * legacy entry points keep the fixed SN_MAX_GROUP_SLOTS capacity. */
+static int64_t partial_bytes_cap(sn_query *q, int32_t cap) {
   if (q->plan.ngroup == 0 && !q->join_group)
     return (int64_t)(2 * q->plan.naggs + 1) * 8;
-  return 8 + (int64_t)SN_MAX_GROUP_SLOTS * sizeof(PartialSlot);
+  return 8 + (int64_t)cap * sizeof(PartialSlot);
 }
 
-extern "C" int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_device) {
-  if (!q || !dst) return SN_ERR_BADARG;
+extern "C" int64_t sn_query_partial_bytes(sn_query *q) {
+  if (!q) return SN_ERR_BADARG;
+  return partial_bytes_cap(q, SN_MAX_GROUP_SLOTS);
+}
+
+extern "C" int64_t sn_query_partial_bytes2(sn_query *q, int32_t cap_slots) {
+  if (!q || cap_slots <= 0) return SN_ERR_BADARG;
+  return partial_bytes_cap(q, cap_slots);
+}
+
+extern "C" int32_t sn_query_partials2(sn_query *q, void *dst,
+                                      int32_t dst_is_device, int32_t cap_slots) {
+  if (!q || !dst || cap_slots <= 0) return SN_ERR_BADARG;
   int rc = sn_query_wait(q);
   if (rc != SN_OK) return rc;
   const sn_plan &p = q->plan;
-  std::vector<uint8_t> block((size_t)sn_query_partial_bytes(q), 0);
+  std::vector<uint8_t> block((size_t)partial_bytes_cap(q, cap_slots), 0);
   if (p.ngroup == 0 && !q->join_group) {
     double *o = (double *)block.data();
     const double *row = q->host_out.data();
@@ -2061,15 +2249,14 @@ extern "C" int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_devi
   } else {
     std::vector<GroupOut> groups;
     local_groups(q, &groups);
-    if (groups.size() > (size_t)SN_MAX_GROUP_SLOTS)
+    if (groups.size() > (size_t)cap_slots)
       return fail(SN_ERR_OVERFLOW,
-                  "%zu groups exceed the partial-block capacity %d (the "
-                  "key-sharded all-to-all overflow exchange is round-2)",
-                  groups.size(), SN_MAX_GROUP_SLOTS);
-    int32_t n = (int32_t)std::min((size_t)SN_MAX_GROUP_SLOTS, groups.size());
+                  "%zu groups exceed the partial-block capacity %d "
+                  "(negotiate a larger capacity via sn_query_partial_bytes2)",
+                  groups.size(), cap_slots);
+    int32_t n = (int32_t)groups.size();
     memcpy(block.data(), &n, 4);
-    int32_t cap = SN_MAX_GROUP_SLOTS;
-    memcpy(block.data() + 4, &cap, 4);
+    memcpy(block.data() + 4, &cap_slots, 4);
     PartialSlot *slots = (PartialSlot *)(block.data() + 8);
     for (int32_t i = 0; i < n; i++) {
       GroupOut &g = groups[i];
@@ -2090,6 +2277,10 @@ extern "C" int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_devi
   return SN_OK;
 }
 
+extern "C" int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_device) {
+  return sn_query_partials2(q, dst, dst_is_device, SN_MAX_GROUP_SLOTS);
+}
+
 /* Split this shard's grouped partials into `world` same-format blocks by
  * key-hash shard — the key-sharded all-to-all of SURVEY §8(e): rank r keeps
  * only keys hashing to shard r; every other key's partial travels to its
@@ -2098,14 +2289,15 @@ extern "C" int32_t sn_query_partials(sn_query *q, void *dst, int32_t dst_is_devi
  * world * sn_query_partial_bytes(q) bytes; block d goes to rank d.  The
  * shard hash is FNV over the key bytes — layout-internal, any deterministic
  * function gives identical results. */
-extern "C" int32_t sn_query_partials_sharded(sn_query *q, int32_t world,
-                                             void *dst) {
-  if (!q || !dst || world <= 0 || world > 1024) return SN_ERR_BADARG;
+extern "C" int32_t sn_query_partials_sharded2(sn_query *q, int32_t world,
+                                              void *dst, int32_t cap_slots) {
+  if (!q || !dst || world <= 0 || world > 1024 || cap_slots <= 0)
+    return SN_ERR_BADARG;
   const sn_plan &p = q->plan;
   if (p.ngroup == 0 && !q->join_group) return SN_ERR_UNSUPPORTED;
   int rc = sn_query_wait(q);
   if (rc != SN_OK) return rc;
-  int64_t bb = sn_query_partial_bytes(q);
+  int64_t bb = partial_bytes_cap(q, cap_slots);
   memset(dst, 0, (size_t)bb * world);
   std::vector<GroupOut> groups;
   local_groups(q, &groups);
@@ -2122,7 +2314,9 @@ extern "C" int32_t sn_query_partials_sharded(sn_query *q, int32_t world,
     int d = (int)(h % (uint64_t)world);
     uint8_t *bp = (uint8_t *)dst + (int64_t)d * bb;
     int32_t &n = counts[d];
-    if (n >= SN_MAX_GROUP_SLOTS) return SN_ERR_NOMEM;
+    if (n >= cap_slots)
+      return fail(SN_ERR_OVERFLOW,
+                  "shard %d exceeds the partial-block capacity %d", d, cap_slots);
     PartialSlot *slots = (PartialSlot *)(bp + 8);
     for (int k = 0; k < SN_MAX_GROUPS; k++) {
       strncpy(slots[n].keys[k], g.keys[k].c_str(), SN_KEY_MAX - 1);
@@ -2133,13 +2327,17 @@ extern "C" int32_t sn_query_partials_sharded(sn_query *q, int32_t world,
     slots[n].rowcount = g.rowcount;
     n++;
   }
-  int32_t cap = SN_MAX_GROUP_SLOTS;
   for (int d = 0; d < world; d++) {
     uint8_t *bp = (uint8_t *)dst + (int64_t)d * bb;
     memcpy(bp, &counts[d], 4);
-    memcpy(bp + 4, &cap, 4);
+    memcpy(bp + 4, &cap_slots, 4);
   }
   return SN_OK;
+}
+
+extern "C" int32_t sn_query_partials_sharded(sn_query *q, int32_t world,
+                                             void *dst) {
+  return sn_query_partials_sharded2(q, world, dst, SN_MAX_GROUP_SLOTS);
 }
 
 extern "C" int32_t sn_query_merge(sn_query *q, const void *blocks, int64_t stride,

@@ -120,7 +120,26 @@ typedef struct {
   sn_dev_pred_d preds_d[8];
   sn_dev_pred_i preds_i[4];
   sn_dev_agg aggs[12];
+  /* sparse-key open-address hash aggregate (the ByteBufferHashMap /
+   * SHAMapAccessor analogue, ByteBufferHashMap.scala:140-183,
+   * SHAMapAccessor.scala:716-830): integer group keys WITHOUT dense-slot
+   * structure (int64 keys; int32/int16 spans beyond the dense cap or
+   * without stats).  hkeys = open-address key array of 1<<hcap_log2
+   * (sentinel SN_HASH_EMPTY = -1; a row whose key IS -1 lands in the
+   * reserved row at index 1<<hcap_log2); hacc = accumulators
+   * [(1<<hcap_log2)+1][naggs+1]; hflags[0] = overflow (table full — host
+   * grows and relaunches).  ngroup==2 packs two 32-bit keys into one i64. */
+  long long *hkeys;
+  double *hacc;
+  int32_t *hflags;
+  int32_t hcap_log2;
+  int32_t sparse;              /* 1: hash-aggregate mode */
+  int32_t _pad3[2];
 } sn_dev_plan;
+
+/* sparse hash-aggregate empty-slot sentinel: -1 so the host can memset the
+ * key array (a REAL key of -1 routes to the reserved row at index cap) */
+#define SN_HASH_EMPTY (-1ll)
 
 #define SN_GRID_CAP 2048
 #define SN_RESULT_PAGE 1024      /* == SN_MAX_GROUP_SLOTS (result page) */
@@ -148,6 +167,16 @@ int sn_launch_scan_agg(const sn_dev_plan *plan,
 /* put-time patch materialization into a null-free fixed-width device body */
 int sn_launch_patch_apply(void *body, const int32_t *pos, const double *val,
                           int n, int kind, void *stream);
+
+/* sparse-key open-address hash aggregate: scan into plan->hkeys/hacc, then
+ * compact the used rows into (key, accumulator-row) pairs for readback */
+int sn_launch_hash_scan(const sn_dev_plan *plan, const sn_dev_plan *dev_plan,
+                        const sn_dev_batch *dev_batches,
+                        const sn_dev_tile *dev_tiles, int32_t ntiles,
+                        void *stream);
+int sn_launch_hash_compact(const long long *hk, const double *hacc,
+                           int cap, int naggs1, long long *okeys,
+                           double *orows, int *counter, void *stream);
 
 /* launches only the partial-fold (k_reduce): scratch[nblocks][nv] -> out.
  * Used by the JIT path, whose scan kernel writes the same scratch rows. */

@@ -1259,6 +1259,173 @@ void k_grouped_global(sn_dev_plan plan,
   }
 }
 
+/* ---- sparse-key open-address hash aggregate ----
+ * The ByteBufferHashMap / SHAMapAccessor analogue for integer group keys
+ * WITHOUT dense-slot structure (ByteBufferHashMap.putBufferIfAbsent,
+ * ByteBufferHashMap.scala:140-183 — open addressing, hash+key probe,
+ * append-on-miss; SHAMapAccessor.generateMapGetOrInsert,
+ * SHAMapAccessor.scala:716-830).  Instead of the JVM's byte-buffer value
+ * records, the device keeps an open-address KEY array in HBM (sentinel
+ * SN_HASH_EMPTY, atomicCAS insert) whose probe position IS the accumulator row
+ * index — high key cardinality means low per-address contention, exactly
+ * when HBM f64 atomics are cheap (same argument as k_grouped_global). */
+__device__ __forceinline__ long long sparse_key(const sn_dev_plan *P,
+                                                const double *sval, int r) {
+  const int gc0 = P->gcol[0];
+  if (P->ngroup == 1) {
+    const double x = sval[(size_t)gc0 * CHUNK + r];
+    return ((P->i64_mask >> gc0) & 1u) ? __double_as_longlong(x)
+                                       : (long long)x;
+  }
+  /* two <=32-bit keys packed into one i64 (reversible; a pack that lands
+   * exactly on the sentinel routes to the reserved row like any other) */
+  const unsigned k0 = (unsigned)(int)sval[(size_t)gc0 * CHUNK + r];
+  const unsigned k1 = (unsigned)(int)sval[(size_t)P->gcol[1] * CHUNK + r];
+  return (long long)(((unsigned long long)k0 << 32) | k1);
+}
+
+__device__ __forceinline__ int hash_probe(long long key, long long *hk_,
+                                          int cap_log2, int32_t *ovf) {
+  GAS long long *hk = (GAS long long *)(uintptr_t)hk_;
+  if (key == SN_HASH_EMPTY) return 1 << cap_log2;   /* reserved row */
+  const unsigned mask = (1u << cap_log2) - 1;
+  unsigned h = (unsigned)mix64((unsigned long long)key) & mask;
+  for (unsigned it = 0; it <= mask; ++it) {
+    const long long k0 = hk[h];
+    if (k0 == key) return (int)h;
+    if (k0 == SN_HASH_EMPTY) {
+      const long long old = (long long)atomicCAS(
+          (unsigned long long *)&hk[h], (unsigned long long)SN_HASH_EMPTY,
+          (unsigned long long)key);
+      if (old == SN_HASH_EMPTY || old == key) return (int)h;
+      /* lost the insert race to a DIFFERENT key: step on */
+    }
+    h = (h + 1) & mask;
+  }
+  atomicOr((int *)(uintptr_t)ovf, 1);               /* table full */
+  return -1;
+}
+
+__global__ __launch_bounds__(WG, 1)
+void k_grouped_hash(sn_dev_plan plan,
+                    const sn_dev_plan *__restrict__ plan_g,
+                    const sn_dev_batch *__restrict__ batches,
+                    const sn_dev_tile *__restrict__ tiles, int ntiles) {
+  const int tid = threadIdx.x;
+  const int nused = plan.nused;
+  const int naggs = plan.naggs;
+  const int npd = plan.npreds_d, npi = plan.npreds_i;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double *sval = (double *)smem;
+  uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
+  uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+  uint64_t *salive = sdead + CHUNK / 64;
+  sn_dev_plan *P = (sn_dev_plan *)(salive + CHUNK / 64 + 2);
+  {
+    const GAS unsigned *src = (const GAS unsigned *)(uintptr_t)plan_g;
+    unsigned *dst = (unsigned *)P;
+    for (unsigned i = tid; i < sizeof(sn_dev_plan) / 4; i += WG) dst[i] = src[i];
+  }
+  __syncthreads();
+
+  GAS double *acc = (GAS double *)(uintptr_t)plan.hacc;
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const sn_dev_tile tile = tiles[t];
+    const sn_dev_batch &b = batches[tile.batch];
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
+    const int clean = b.clean;
+
+    for (int base = tile.row_start; base < tile_end; base += CHUNK) {
+      const int rows = min(CHUNK, tile_end - base);
+      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      __syncthreads();
+      alive_init(salive, sdead, rows, clean);
+      pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
+      if (plan.jkeys) probe_sweep(P, sval, salive, nullptr);
+
+#pragma unroll 2
+      for (int k = 0; k < CHUNK / WG; k++) {
+        const int r = tid + k * WG;
+        const uint64_t w = salive[r >> 6];
+        if (w == 0) continue;
+        const int m = (int)((w >> (tid & 63)) & 1ull);
+        if (!m) continue;
+        const long long key = sparse_key(P, sval, r);
+        const int slot = hash_probe(key, plan.hkeys, plan.hcap_log2,
+                                    plan.hflags);
+        if (slot < 0) continue;                     /* overflow: host retries */
+        GAS double *row_acc = acc + (size_t)slot * (naggs + 1);
+        for (int a = 0; a < naggs; a++)
+          (void)atomicAdd((double *)&row_acc[a], eval_agg(P, P->aggs[a], sval, r));
+        (void)atomicAdd((double *)&row_acc[naggs], 1.0);
+      }
+      __syncthreads();
+    }
+  }
+}
+
+/* compact the used hash-table rows into dense (key, accumulator-row) pairs
+ * so the host reads back only the live groups, not the whole table */
+__global__ void k_hash_compact(const long long *__restrict__ hk_,
+                               const double *__restrict__ hacc_,
+                               int cap, int naggs1,
+                               long long *__restrict__ okeys,
+                               double *__restrict__ orows,
+                               int *__restrict__ counter) {
+  const GAS long long *hk = (const GAS long long *)(uintptr_t)hk_;
+  const GAS double *hacc = (const GAS double *)(uintptr_t)hacc_;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i > cap) return;
+  long long key;
+  int used;
+  if (i == cap) {                                   /* reserved sentinel-key row */
+    key = SN_HASH_EMPTY;
+    used = hacc[(size_t)cap * naggs1 + (naggs1 - 1)] != 0.0;
+  } else {
+    key = hk[i];
+    used = key != SN_HASH_EMPTY;
+  }
+  if (!used) return;
+  const int o = atomicAdd((int *)(uintptr_t)counter, 1);
+  ((GAS long long *)(uintptr_t)okeys)[o] = key;
+  GAS double *dst = (GAS double *)(uintptr_t)orows + (size_t)o * naggs1;
+  for (int a = 0; a < naggs1; a++) dst[a] = hacc[(size_t)i * naggs1 + a];
+}
+
+extern "C" int sn_launch_hash_scan(const sn_dev_plan *plan,
+                                   const sn_dev_plan *dev_plan,
+                                   const sn_dev_batch *dev_batches,
+                                   const sn_dev_tile *dev_tiles, int32_t ntiles,
+                                   void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int grid;
+  if (ntiles <= 0) return 0;
+  if (ntiles <= SN_GRID_CAP) grid = ntiles;
+  else {
+    int rounds = (ntiles + SN_GRID_CAP - 1) / SN_GRID_CAP;
+    grid = (ntiles + rounds - 1) / rounds;
+  }
+  size_t lds = (size_t)plan->nused * CHUNK * 8 +
+               (size_t)plan->nused * (CHUNK / 64) * 8 +
+               2 * (CHUNK / 64) * 8 + sizeof(sn_dev_plan) + 64;
+  if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
+  hipLaunchKernelGGL(k_grouped_hash, dim3(grid), dim3(WG), lds, s,
+                     *plan, dev_plan, dev_batches, dev_tiles, ntiles);
+  return (int)hipGetLastError();
+}
+
+extern "C" int sn_launch_hash_compact(const long long *hk, const double *hacc,
+                                      int cap, int naggs1, long long *okeys,
+                                      double *orows, int *counter,
+                                      void *stream) {
+  hipLaunchKernelGGL(k_hash_compact, dim3((cap + 1 + 255) / 256), dim3(256), 0,
+                     (hipStream_t)stream, hk, hacc, cap, naggs1, okeys, orows,
+                     counter);
+  return (int)hipGetLastError();
+}
+
 /* fold per-block partial rows into the final output.
  * keyless: final[i] = sum_b scratch[b][i]  (NV = 2*NA_t+1, identical layout)
  * grouped: scratch rows are [slot][naggs+1]; final is [slot][out_stride]

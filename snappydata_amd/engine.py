@@ -98,6 +98,14 @@ def lib():
         L.sn_query_merge.argtypes = [C.c_void_p, C.c_void_p, C.c_int64, C.c_int32]
         L.sn_query_partials_sharded.restype = C.c_int32
         L.sn_query_partials_sharded.argtypes = [C.c_void_p, C.c_int32, C.c_void_p]
+        L.sn_query_partial_bytes2.restype = C.c_int64
+        L.sn_query_partial_bytes2.argtypes = [C.c_void_p, C.c_int32]
+        L.sn_query_partials2.restype = C.c_int32
+        L.sn_query_partials2.argtypes = [C.c_void_p, C.c_void_p, C.c_int32,
+                                         C.c_int32]
+        L.sn_query_partials_sharded2.restype = C.c_int32
+        L.sn_query_partials_sharded2.argtypes = [C.c_void_p, C.c_int32,
+                                                 C.c_void_p, C.c_int32]
         L.sn_batch_mutate.restype = C.c_int32
         L.sn_batch_mutate.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
                                       C.c_int32, C.POINTER(abi.SnBuf),
@@ -246,22 +254,41 @@ class Query:
         """True when the query ran a query-compiled (hipRTC) kernel."""
         return bool(lib().sn_query_used_jit(self._h))
 
-    def partial_bytes(self):
-        return _check(lib().sn_query_partial_bytes(self._h))
+    def num_groups(self):
+        """Total group rows of the local (pre-merge) result."""
+        return _check(lib().sn_query_num_groups(self._h), "num_groups")
 
-    def partials_host(self):
-        n = self.partial_bytes()
+    def partial_bytes(self, cap=None):
+        """Partial-block size; cap overrides the slot capacity for group
+        counts beyond SN_MAX_GROUP_SLOTS (negotiate one cap across ranks)."""
+        if cap is None:
+            return _check(lib().sn_query_partial_bytes(self._h))
+        return _check(lib().sn_query_partial_bytes2(self._h, cap))
+
+    def partials_host(self, cap=None):
+        n = self.partial_bytes(cap)
         buf = np.zeros(n, dtype=np.uint8)
-        _check(lib().sn_query_partials(self._h, buf.ctypes.data, 0), "partials")
+        if cap is None:
+            _check(lib().sn_query_partials(self._h, buf.ctypes.data, 0),
+                   "partials")
+        else:
+            _check(lib().sn_query_partials2(self._h, buf.ctypes.data, 0, cap),
+                   "partials")
         return buf
 
-    def partials_sharded(self, world):
+    def partials_sharded(self, world, cap=None):
         """Key-sharded split for the grouped all-to-all:
         returns (world, partial_bytes) uint8; row d travels to rank d."""
-        bb = self.partial_bytes()
+        bb = self.partial_bytes(cap)
         buf = np.zeros(world * bb, dtype=np.uint8)
-        _check(lib().sn_query_partials_sharded(self._h, world, buf.ctypes.data),
-               "partials_sharded")
+        if cap is None:
+            _check(lib().sn_query_partials_sharded(self._h, world,
+                                                   buf.ctypes.data),
+                   "partials_sharded")
+        else:
+            _check(lib().sn_query_partials_sharded2(self._h, world,
+                                                    buf.ctypes.data, cap),
+                   "partials_sharded")
         return buf.reshape(world, bb)
 
     def partials_into_device(self, dev_ptr):
