@@ -13,7 +13,8 @@ from snappydata_amd import abi
 def test_workload_table_shape():
     for name, (plan_fn, rows, bpr) in bench.WORKLOADS.items():
         assert rows > 0 and bpr > 0
-        assert name in ("star_join_sf10", "config1_sum_where") or plan_fn
+        assert name in ("star_join_sf10", "config1_sum_where",
+                        "sparse_group_sf10") or plan_fn
 
 
 def test_q6_q1_plan_construction():
