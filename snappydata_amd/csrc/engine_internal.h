@@ -134,7 +134,13 @@ typedef struct {
   int32_t *hflags;
   int32_t hcap_log2;
   int32_t sparse;              /* 1: hash-aggregate mode */
-  int32_t _pad3[2];
+  /* 1: per-agg counts — grouped accumulator rows widen to
+   * [sums naggs][counts naggs][rowcount] because some aggregate input
+   * column carries ACTUAL nulls (Spark Sum/Average skip null inputs,
+   * SnappyHashAggregateExec.scala:450-500); 0 keeps [sums][rowcount] with
+   * counts == rowcount (non-null inputs) */
+  int32_t pac;
+  int32_t _pad3;
 } sn_dev_plan;
 
 /* sparse hash-aggregate empty-slot sentinel: -1 so the host can memset the

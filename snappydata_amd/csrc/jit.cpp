@@ -104,7 +104,7 @@ struct sn_dev_plan {
   sn_dev_pred_i preds_i[4];
   sn_dev_agg aggs[12];
   i64 *hkeys; double *hacc; int *hflags;
-  int hcap_log2, sparse, _pad3[2];
+  int hcap_log2, sparse, pac, _pad3;
 };
 __device__ __forceinline__ double wsum(double x) {
 #pragma unroll

@@ -491,6 +491,21 @@ __device__ __forceinline__ double eval_agg(const sn_dev_plan *P,
          (A.a2 + A.m2 * sv_agg(P, sval, A.c2, r));
 }
 
+/* per-agg factor validity on general batches: 1 when every referenced
+ * factor column is non-null for row r (Spark Sum/Average skip null
+ * inputs; COUNT(*) has nf == 0 and is always valid) */
+__device__ __forceinline__ int agg_valid(const sn_dev_agg &A,
+                                         const uint64_t *svalid, int r) {
+  int ok = 1;
+  if (A.nf >= 1)
+    ok &= (int)((svalid[(size_t)A.c0 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
+  if (A.nf >= 2)
+    ok &= (int)((svalid[(size_t)A.c1 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
+  if (A.nf >= 3)
+    ok &= (int)((svalid[(size_t)A.c2 * (CHUNK / 64) + (r >> 6)] >> (r & 63)) & 1ull);
+  return ok;
+}
+
 /* wave (64-lane) sum reduction */
 __device__ __forceinline__ double wave_sum(double x) {
 #pragma unroll

@@ -207,7 +207,9 @@ def test_key_sharded_a2a_20k_groups_split_merge(eng):
         q0.merge_host(np.ascontiguousarray(blocks0), bb, 2)
         q1.merge_host(np.ascontiguousarray(blocks1), bb, 2)
         merged = sorted(q0.rows() + q1.rows())
-        assert merged == sorted(ref_rows)
+        # sharded partials sum in a different order than the unsharded
+        # reference: counts/keys exact, double sums within 1e-9 relative
+        assert_rows_match(merged, sorted(ref_rows), count_aggs={1})
     finally:
         e0.close()
         e1.close()

@@ -205,17 +205,18 @@ def test_engine_int64_delta_exact(eng):
     base = np.arange(n, dtype=np.int64) + (1 << 57)
     w = np.ones(n)
     pos = np.arange(0, n, 11, dtype=np.int32)
-    newv = base[pos] + 5
+    newv = base[pos] + n + 5          # collision-free with unpatched values
     d1 = se.encode_update_delta(abi.T_INT64, pos, n, newv)
     cols = [po.encode(po.T_INT64, po.ENC_UNCOMPRESSED, base),
             po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
     t = eng.table_define("tdelta64", [(abi.T_INT64, False), (abi.T_DOUBLE, False)])
     eng.batch_put(t, 1, 0, -n, cols,
                   deltas=[(d1, None), (None, None)])
-    cut = int(base[pos[7]] + 5)
+    cut = int(base[pos[7]] + n + 5)
     grows = eng.query(abi.make_plan(
         table=t, preds=[dict(col=0, lo=cut, hi=cut)],
         aggs=[("count", [])])).rows()
     merged = base.copy()
     merged[pos] = newv
-    assert grows[0][1][0] == float((merged == cut).sum()) == 1.0
+    assert (merged == cut).sum() == 1    # collision-free by construction
+    assert grows[0][1][0] == 1.0
