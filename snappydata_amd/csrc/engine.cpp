@@ -184,6 +184,7 @@ struct DescCache {
   int jit_ok = 0;
   int jit_del = 0;
   int jit_kinds[SN_DEV_MAX_COLS] = {0};
+  int pac = 0;        /* batches carry nulls on aggregate-input columns */
 };
 
 struct Table {
@@ -1170,8 +1171,9 @@ struct sn_query {
   int nslots = 0;
   int g1cap = 0, g2cap = 0;             /* per-group-col slot counts */
   bool sparse = false;                  /* open-address hash-aggregate mode */
+  bool pac = false;                     /* per-agg counts (nullable agg inputs) */
   std::vector<long long> sparse_keys;   /* compacted group keys */
-  std::vector<double> sparse_rows;      /* [n][dev_naggs+1] accumulator rows */
+  std::vector<double> sparse_rows;      /* [n][na1] accumulator rows */
   bool gint[2] = { false, false };      /* integer group key (stats-ranged) */
   int64_t gmin[2] = { 0, 0 };           /* integer key minimum (slot base) */
   int gnull1 = -1, gnull2 = -1;         /* null slot index per group col (-1: none) */
@@ -1316,11 +1318,6 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       int c = plan->aggs[a].factors[f].col;
       sn_type_t dt = t->schema[c].dtype;
       if (dt == SN_TYPE_STRING) { fail(SN_ERR_UNSUPPORTED, "string agg input"); return nullptr; }
-      if (plan->ngroup > 0 && t->schema[c].nullable) {
-        fail(SN_ERR_UNSUPPORTED,
-             "grouped aggregates over nullable inputs not in the round-1 GPU path");
-        return nullptr;
-      }
       if (use_col(c) < 0) { fail(SN_ERR_BADARG, "too many plan columns"); return nullptr; }
     }
 
@@ -1660,6 +1657,17 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
    * (derived here; stored in the DescCache for reuse on hits) */
   bool jit_ok_b = true, jit_first = true, jit_any_del = false;
   int jit_kinds[SN_DEV_MAX_COLS] = {0};
+  /* per-agg counts needed?  Only when a grouped plan's aggregate-input
+   * columns carry ACTUAL nulls (null-free nullable schemas keep the
+   * counts==rowcount fast layout, JIT included) */
+  uint32_t agg_cmask = 0;
+  for (int a = 0; a < q->dev_naggs; a++) {
+    const sn_dev_agg &A = dp.aggs[a];
+    if (A.nf >= 1) agg_cmask |= 1u << A.c0;
+    if (A.nf >= 2) agg_cmask |= 1u << A.c1;
+    if (A.nf >= 3) agg_cmask |= 1u << A.c2;
+  }
+  bool agg_nulls = false;
   for (auto &b : t->batches) {
     if (hit) break;
     if (skip_count && batch_skippable(b, plan, t)) continue;
@@ -1760,6 +1768,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       }
     }
     db.clean = clean ? 1 : 0;
+    for (size_t ui = 0; ui < q->used_cols.size(); ui++)
+      if (((agg_cmask >> ui) & 1) &&
+          (db.cols[ui].has_nulls || db.cols[ui].patch_nullbm))
+        agg_nulls = true;
     /* deletes alone don't disqualify the JIT (the generated kernel reads
      * del_bm); nulls or unmaterialized patches on a used column do */
     if (db.del_bm) jit_any_del = true;
@@ -1818,15 +1830,21 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       dc.jit_ok = (jit_ok_b && !jit_first) ? 1 : 0;
       dc.jit_del = jit_any_del ? 1 : 0;
       memcpy(dc.jit_kinds, jit_kinds, sizeof(jit_kinds));
+      dc.pac = agg_nulls ? 1 : 0;
       t->desc_caches.push_back(dc);
     }
+  }
+  {
+    const bool grouped_mode = plan->ngroup > 0 || q->join_group;
+    q->pac = grouped_mode && (hit ? hit->pac != 0 : agg_nulls);
+    dp.pac = q->pac ? 1 : 0;
   }
   if (q->sparse && ntiles > 0) {
     /* ---- open-address hash-aggregate launch (k_grouped_hash) ----
      * The workspace (key table + accumulators) is engine-cached; results
      * are compacted on device and read back HERE (inside submit, stream
      * synchronized), so concurrent queries never share the live table. */
-    const int naggs1 = q->dev_naggs + 1;
+    const int naggs1 = (q->pac ? 2 : 1) * q->dev_naggs + 1;
     int cap_log2 = 20;
     while (cap_log2 > 12 && (1ll << (cap_log2 - 1)) >= 4 * q->rows_scanned)
       cap_log2--;                      /* small tables: smaller table */
@@ -1932,9 +1950,12 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     /* block-partial scratch rows (the >16-slot path caps its grid) */
     int grid = ntiles < SN_GRID_CAP ? ntiles : SN_GRID_CAP;
     if (dp.nslots > 16 && grid > SN_GRID_BIGSLOT) grid = SN_GRID_BIGSLOT;
+    const int na1 = dp.pac ? 2 * q->dev_naggs + 1 : q->dev_naggs + 1;
     size_t nv = dp.nslots <= 1 ? (size_t)(2 * q->na_t + 1)
-                               : (size_t)dp.nslots * (q->dev_naggs + 1);
-    const bool big_groups = dp.nslots > SN_RESULT_PAGE;
+                               : (size_t)dp.nslots * na1;
+    const bool big_groups =
+        dp.nslots > SN_RESULT_PAGE ||
+        (dp.pac && sn_grouped_needs_global(dp.nused, dp.nslots, na1));
     size_t need = (big_groups ? 8 : (size_t)grid) * nv * 8;
     if (e->scratch_sz < need) {
       e->scratch = (double *)e->arena.alloc(need);
@@ -1979,7 +2000,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       /* >16-slot scratch rows are wide; mirror the interpreted grid cap
        * (scratch was sized with the same bound above) */
       if (dp.nslots > 16 && jgrid > SN_GRID_BIGSLOT) jgrid = SN_GRID_BIGSLOT;
-      int naggs1 = dp.nslots <= 1 ? 0 : dp.naggs + 1;
+      int naggs1 = dp.nslots <= 1 ? 0 : na1;   /* (JIT never runs pac) */
       rc = sn_jit_launch(jfn, jgrid, (const sn_dev_batch *)db_dev,
                          (const sn_dev_tile *)tl_dev, ntiles, e->scratch,
                          dp.jkeys, dp.jpayload, dp.jlut,
@@ -2052,18 +2073,19 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
     /* hash-aggregate results: compacted (key, accumulator-row) pairs.
      * Keys surface as decimal text like the dense integer-key path (the
      * partial-block and result formats stay shared). */
-    const int naggs1 = q->dev_naggs + 1;
+    const int naggs1 = (q->pac ? 2 : 1) * q->dev_naggs + 1;
     const bool packed = p.ngroup == 2;
     for (size_t i = 0; i < q->sparse_keys.size(); i++) {
       const double *row = &q->sparse_rows[i * naggs1];
-      double rowcount = row[q->dev_naggs];
+      double rowcount = row[naggs1 - 1];
       if (rowcount == 0.0) continue;
       GroupOut g;
       g.rowcount = rowcount;
       for (int a = 0; a < p.naggs; a++) {
         int di = q->agg_map[a];
         g.sums[a] = di < 0 ? rowcount : row[di];
-        g.counts[a] = rowcount;
+        g.counts[a] = (di < 0 || !q->pac) ? rowcount
+                                          : row[q->dev_naggs + di];
       }
       long long key = q->sparse_keys[i];
       if (packed) {
@@ -2091,11 +2113,13 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
         g.sums[a] = row[a];
         g.counts[a] = row[q->na_t + a];
       } else {
-        /* grouped: deduped device sweeps; COUNT(*) = rowcount; non-null
-         * inputs enforced -> per-agg count = rowcount */
+        /* grouped: deduped device sweeps; COUNT(*) = rowcount; per-agg
+         * counts live at [dev_naggs + di] when the pac layout ran
+         * (nullable aggregate inputs), else counts == rowcount */
         int di = q->agg_map[a];
         g.sums[a] = di < 0 ? rowcount : row[di];
-        g.counts[a] = rowcount;
+        g.counts[a] = (di < 0 || !q->pac) ? rowcount
+                                          : row[q->dev_naggs + di];
       }
     }
     if (q->join_group) {

@@ -147,6 +147,22 @@ typedef struct {
  * key array (a REAL key of -1 routes to the reserved row at index cap) */
 #define SN_HASH_EMPTY (-1ll)
 
+#define SN_SCAN_CHUNK 1024     /* rows per LDS conversion chunk (kernels.hip) */
+
+/* shared host/launcher routing predicate: does the grouped LDS-accumulator
+ * kernel fit, or must the plan take the global-atomic route?  Used by BOTH
+ * sn_launch_scan_agg and the engine (which must zero the global accumulator
+ * before launch) so the two can never disagree. */
+static inline int sn_grouped_needs_global(int nused, int nslots, int na1) {
+  unsigned long long lds =
+      (unsigned long long)nused * SN_SCAN_CHUNK * 8 +
+      (unsigned long long)nused * (SN_SCAN_CHUNK / 64) * 8 +
+      2ull * (SN_SCAN_CHUNK / 64) * 8 + sizeof(sn_dev_plan) + 512 +
+      (SN_SCAN_CHUNK / 64) * 8 + SN_SCAN_CHUNK * 2 +
+      (unsigned long long)nslots * na1 * 8 + 64;
+  return lds > 160ull * 1024;
+}
+
 #define SN_GRID_CAP 2048
 #define SN_RESULT_PAGE 1024      /* == SN_MAX_GROUP_SLOTS (result page) */
 #define SN_BIG_GROUP_CAP (1 << 20) /* dense-slot cap of the global-atomic

@@ -1004,7 +1004,10 @@ void k_grouped_lds(sn_dev_plan plan,
     unsigned *dst = (unsigned *)P;
     for (unsigned i = tid; i < sizeof(sn_dev_plan) / 4; i += WG) dst[i] = src[i];
   }
-  const int NV = nslots * (naggs + 1);
+  /* pac: accumulator rows widen to [sums][per-agg counts][rowcount] */
+  const int pac = plan.pac;
+  const int na1 = pac ? 2 * naggs + 1 : naggs + 1;
+  const int NV = nslots * na1;
   for (int i = tid; i < NV; i += WG) bacc[i] = 0.0;
   __syncthreads();
 
@@ -1042,10 +1045,17 @@ void k_grouped_lds(sn_dev_plan plan,
             slot += (int)((long long)sval[(size_t)gc1 * CHUNK + r] -
                           P->gbase[1]);
         }
-        double *row_acc = bacc + (size_t)slot * (naggs + 1);
-        for (int a = 0; a < naggs; a++)
-          atomicAdd(&row_acc[a], eval_agg(P, P->aggs[a], sval, r));
-        atomicAdd(&row_acc[naggs], 1.0);
+        double *row_acc = bacc + (size_t)slot * na1;
+        for (int a = 0; a < naggs; a++) {
+          const sn_dev_agg &A = P->aggs[a];
+          if (pac) {
+            const int av = clean ? 1 : agg_valid(A, svalid, r);
+            if (!av) continue;
+            atomicAdd(&row_acc[naggs + a], 1.0);
+          }
+          atomicAdd(&row_acc[a], eval_agg(P, A, sval, r));
+        }
+        atomicAdd(&row_acc[na1 - 1], 1.0);
       }
       __syncthreads();
     }
@@ -1229,8 +1239,10 @@ void k_grouped_global(sn_dev_plan plan,
   __syncthreads();
 
   /* 8-way XCD privatization (see the JIT twin): fold happens in k_reduce */
+  const int pac = plan.pac;
+  const int na1 = pac ? 2 * naggs + 1 : naggs + 1;
   GAS double *acc = (GAS double *)(uintptr_t)gacc +
-                    (size_t)(blockIdx.x & 7) * (size_t)plan.nslots * (naggs + 1);
+                    (size_t)(blockIdx.x & 7) * (size_t)plan.nslots * na1;
   const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
@@ -1264,10 +1276,17 @@ void k_grouped_global(sn_dev_plan plan,
           if (ngroup >= 2)
             slot += (long long)sval[(size_t)gc1 * CHUNK + r] - P->gbase[1];
         }
-        GAS double *row_acc = acc + (size_t)slot * (naggs + 1);
-        for (int a = 0; a < naggs; a++)
-          (void)atomicAdd((double *)&row_acc[a], eval_agg(P, P->aggs[a], sval, r));
-        (void)atomicAdd((double *)&row_acc[naggs], 1.0);
+        GAS double *row_acc = acc + (size_t)slot * na1;
+        for (int a = 0; a < naggs; a++) {
+          const sn_dev_agg &A = P->aggs[a];
+          if (pac) {
+            const int av = clean ? 1 : agg_valid(A, svalid, r);
+            if (!av) continue;
+            (void)atomicAdd((double *)&row_acc[naggs + a], 1.0);
+          }
+          (void)atomicAdd((double *)&row_acc[a], eval_agg(P, A, sval, r));
+        }
+        (void)atomicAdd((double *)&row_acc[na1 - 1], 1.0);
       }
       __syncthreads();
     }
@@ -1344,6 +1363,8 @@ void k_grouped_hash(sn_dev_plan plan,
   }
   __syncthreads();
 
+  const int pac = plan.pac;
+  const int na1 = pac ? 2 * naggs + 1 : naggs + 1;
   GAS double *acc = (GAS double *)(uintptr_t)plan.hacc;
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
@@ -1371,10 +1392,17 @@ void k_grouped_hash(sn_dev_plan plan,
         const int slot = hash_probe(key, plan.hkeys, plan.hcap_log2,
                                     plan.hflags);
         if (slot < 0) continue;                     /* overflow: host retries */
-        GAS double *row_acc = acc + (size_t)slot * (naggs + 1);
-        for (int a = 0; a < naggs; a++)
-          (void)atomicAdd((double *)&row_acc[a], eval_agg(P, P->aggs[a], sval, r));
-        (void)atomicAdd((double *)&row_acc[naggs], 1.0);
+        GAS double *row_acc = acc + (size_t)slot * na1;
+        for (int a = 0; a < naggs; a++) {
+          const sn_dev_agg &A = P->aggs[a];
+          if (pac) {
+            const int av = clean ? 1 : agg_valid(A, svalid, r);
+            if (!av) continue;
+            (void)atomicAdd((double *)&row_acc[naggs + a], 1.0);
+          }
+          (void)atomicAdd((double *)&row_acc[a], eval_agg(P, A, sval, r));
+        }
+        (void)atomicAdd((double *)&row_acc[na1 - 1], 1.0);
       }
       __syncthreads();
     }
@@ -1534,11 +1562,12 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
                sizeof(sn_dev_plan) + 512;        /* plan mirror + keyless bacc */
   hipError_t err;
   const bool nc4 = plan->nused <= 4;
+  const int na1 = plan->pac ? 2 * na + 1 : na + 1;
   int nv, naggs1, out_stride;
   if (ns <= 1) {
     nv = 2 * na_t + 1; naggs1 = 0; out_stride = nv;
   } else {
-    nv = ns * (na + 1); naggs1 = na + 1; out_stride = 2 * na_t + 1;
+    nv = ns * na1; naggs1 = na1; out_stride = 2 * na_t + 1;
   }
 #define KL(A, NCv) hipLaunchKernelGGL((k_keyless<A, NCv>), dim3(grid), dim3(WG), lds, s, \
         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_scratch)
@@ -1548,20 +1577,23 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
     if (na <= 2) { if (nc4) KL(2, 4); else KL(2, 8); }
     else if (na <= 4) { if (nc4) KL(4, 4); else KL(4, 8); }
     else { if (nc4) KL(12, 4); else KL(12, 8); }
-  } else if (ns > SN_RESULT_PAGE) {
-    /* unbounded cardinality: global f64 atomics into the (host-zeroed)
-     * scratch accumulator — one "block" of partials for k_reduce */
+  } else if (ns > SN_RESULT_PAGE ||
+             (plan->pac &&
+              sn_grouped_needs_global(plan->nused, ns, na1))) {
+    /* unbounded cardinality (or a pac accumulator too wide for LDS):
+     * global f64 atomics into the (host-zeroed) scratch accumulator */
     lds += (CHUNK / 64) * 8 + CHUNK * 2 + sizeof(sn_dev_plan) + 64;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     hipLaunchKernelGGL(k_grouped_global, dim3(grid), dim3(WG), lds, s,
                        *plan, dev_plan, dev_batches, dev_tiles, ntiles,
                        dev_scratch);
     grid = 8;   /* k_reduce folds the 8 XCD-private copies */
-  } else if (ns > 16) {
-    /* large-cardinality LDS hash-aggregate path: LDS accumulator bounds the
-     * grid so scratch rows stay small */
+  } else if (ns > 16 || plan->pac) {
+    /* large-cardinality LDS hash-aggregate path (also the per-agg-count
+     * path for nullable aggregate inputs at any slot count): the LDS
+     * accumulator bounds the grid so scratch rows stay small */
     if (grid > SN_GRID_BIGSLOT) grid = SN_GRID_BIGSLOT;
-    lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)ns * (na + 1) * 8 + 64;
+    lds += (CHUNK / 64) * 8 + CHUNK * 2 + (size_t)ns * na1 * 8 + 64;
     if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
     hipLaunchKernelGGL(k_grouped_lds, dim3(grid), dim3(WG), lds, s,
                        *plan, dev_plan, dev_batches, dev_tiles, ntiles,

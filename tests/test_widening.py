@@ -179,6 +179,88 @@ def test_long_group_key_rejected(eng):
     assert rows[0][1][0] == float(n)
 
 
+@pytest.mark.gpu
+def test_nullable_grouped_agg_inputs(eng):
+    """Grouped aggregates over nullable inputs (previously rejected): rows
+    with a null factor skip that aggregate but still count in COUNT(*) and
+    rowcount (Spark Sum/Average semantics); AVG divides by the per-agg
+    non-null count.  Routed through the per-agg-count (pac) layout."""
+    n = 400_000
+    rng = np.random.default_rng(71)
+    keys = [b"G%02d" % v for v in rng.integers(0, 30, n)]
+    a = rng.random(n)
+    avalid = (rng.random(n) > 0.15).astype(np.uint8)
+    b = rng.random(n)
+    bvalid = (rng.random(n) > 0.4).astype(np.uint8)
+    cols = [po.encode(po.T_STRING, po.ENC_DICT, keys),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, a, valid=avalid),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, b, valid=bvalid)]
+    t = eng.table_define("tnullagg", [(abi.T_STRING, False), (abi.T_DOUBLE, True),
+                                      (abi.T_DOUBLE, True)])
+    for st in range(0, n, 100_000):
+        en = min(n, st + 100_000)
+        sub = [po.encode(po.T_STRING, po.ENC_DICT, keys[st:en]),
+               po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, a[st:en],
+                         valid=avalid[st:en]),
+               po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, b[st:en],
+                         valid=bvalid[st:en])]
+        eng.batch_put(t, st, st // 100_000, en - st, sub)
+    plan_kw = dict(group_cols=[0],
+                   aggs=[("sum", [(1, 0.0, 1.0)]),
+                         ("avg", [(1, 0.0, 1.0)]),
+                         ("sum", [(1, 0.0, 1.0), (2, 0.0, 1.0)]),
+                         ("avg", [(2, 0.0, 1.0)]),
+                         ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_STRING, po.T_DOUBLE, po.T_DOUBLE],
+                        nullable=[False, True, True])
+    ot.add_batch(n, cols)
+    orows = po.result_rows(ot.query(po.make_plan(**plan_kw)))
+    assert_close_rows(grows, orows, count_aggs={4})
+    # numpy cross-check of one group: sum skips nulls, count(*) does not
+    g0 = np.array([k == b"G00" for k in keys])
+    m = g0 & (avalid == 1)
+    got = {k[0]: v for k, v in grows}
+    assert abs(got["G00"][0] - a[m].sum()) <= REL * max(1.0, abs(a[m].sum()))
+    assert got["G00"][4] == float(g0.sum())
+
+
+@pytest.mark.gpu
+def test_nullable_grouped_sparse_keys(eng):
+    """pac layout on the sparse hash-aggregate path: int64 keys + nullable
+    aggregate input."""
+    n = 300_000
+    rng = np.random.default_rng(73)
+    keys = rng.integers(0, 4_000, n).astype(np.int64) * (1 << 35)
+    w = rng.random(n)
+    valid = (rng.random(n) > 0.2).astype(np.uint8)
+    cols = [po.encode(po.T_INT64, po.ENC_UNCOMPRESSED, keys),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w, valid=valid)]
+    t = eng.table_define("tnullsparse", [(abi.T_INT64, False), (abi.T_DOUBLE, True)])
+    eng.batch_put(t, 1, 0, n, cols)
+    plan_kw = dict(group_cols=[0],
+                   aggs=[("sum", [(1, 0.0, 1.0)]), ("avg", [(1, 0.0, 1.0)]),
+                         ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_INT64, po.T_DOUBLE], nullable=[False, True])
+    ot.add_batch(n, cols)
+    orows = ot.query_groups(po.make_plan(**plan_kw), nthreads=16)
+    assert_close_rows(grows, orows, count_aggs={2})
+
+
+def assert_close_rows(grows, orows, count_aggs=()):
+    assert len(grows) == len(orows)
+    for (gk, gv), (ok_, ov) in zip(grows, orows):
+        assert gk == ok_
+        for i, (g, o) in enumerate(zip(gv, ov)):
+            if i in count_aggs:
+                assert g == o, (gk, i, g, o)
+            elif o is None:
+                assert g is None, (gk, i, g)
+            else:
+                assert abs(g - o) <= REL * max(1.0, abs(o)), (gk, i, g, o)
+
+
 def test_oracle_int64_delta_exact_cpu():
     """Oracle-side: int64 update deltas keep exactness beyond 2^53."""
     n = 10_000
