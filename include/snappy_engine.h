@@ -165,12 +165,15 @@ typedef struct {
 #define SN_AGG_SUM        0
 #define SN_AGG_COUNT_STAR 1
 #define SN_AGG_AVG        2
+#define SN_AGG_MIN        3
+#define SN_AGG_MAX        4
 
-/* aggregate: SUM/AVG over a product of factors, or COUNT(*).
- * Null semantics follow Spark Sum/Average/Count
- * (SnappyHashAggregateExec.scala:450-500 accumulate rules): a row
- * contributes iff every referenced column is non-null; COUNT(*) counts
- * every surviving row. */
+/* aggregate: SUM/AVG/MIN/MAX over a product of factors, or COUNT(*).
+ * Null semantics follow Spark Sum/Average/Count and the Min/Max
+ * DeclarativeAggregates (SnappyHashAggregateExec.scala:450-500 accumulate
+ * rules): a row contributes iff every referenced column is non-null;
+ * COUNT(*) counts every surviving row; MIN/MAX of an empty/all-null
+ * input group is NULL. */
 typedef struct {
   int32_t kind;
   int32_t nfactors;

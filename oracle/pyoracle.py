@@ -23,7 +23,7 @@ SN_KEY_MAX = 48
 T_INT32, T_INT64, T_DOUBLE, T_STRING, T_BOOL, T_INT16, T_INT8, T_FLOAT = range(8)
 # encodings
 ENC_UNCOMPRESSED, ENC_RLE, ENC_DICT, ENC_BIGDICT, ENC_BOOLBITSET = range(5)
-AGG_SUM, AGG_COUNT_STAR, AGG_AVG = range(3)
+AGG_SUM, AGG_COUNT_STAR, AGG_AVG, AGG_MIN, AGG_MAX = range(5)
 
 
 class SnBuf(C.Structure):
@@ -409,7 +409,8 @@ def make_plan(preds=(), group_cols=(), aggs=(), join=None):
     p.naggs = len(aggs)
     for i, (kind, factors) in enumerate(aggs):
         ag = p.aggs[i]
-        ag.kind = {"sum": AGG_SUM, "count": AGG_COUNT_STAR, "avg": AGG_AVG}[kind]
+        ag.kind = {"sum": AGG_SUM, "count": AGG_COUNT_STAR,
+                   "avg": AGG_AVG, "min": AGG_MIN, "max": AGG_MAX}[kind]
         ag.nfactors = len(factors)
         for j, (col, add, mul) in enumerate(factors):
             ag.factors[j].col = col

@@ -1086,7 +1086,15 @@ static int eval_batch(const sno_table *t, const sno_batch *b, const sn_plan *p,
         if (val_null[c]) { anynull = 1; break; }
         v *= ag->factors[f].add + ag->factors[f].mul * val_d[c];
       }
-      if (!anynull) { grp->sums[a] += v; grp->counts[a] += 1.0; }
+      if (!anynull) {
+        if (ag->kind == SN_AGG_MIN)
+          grp->sums[a] = grp->counts[a] > 0 ? fmin(grp->sums[a], v) : v;
+        else if (ag->kind == SN_AGG_MAX)
+          grp->sums[a] = grp->counts[a] > 0 ? fmax(grp->sums[a], v) : v;
+        else
+          grp->sums[a] += v;
+        grp->counts[a] += 1.0;
+      }
     }
   }
 
@@ -1112,6 +1120,8 @@ static int eval_batch_fast(const sno_table *t, const sno_batch *b,
                            int64_t *rows_scanned, int64_t *rows_passed) {
   if (b->has_deltas || b->del) return 0;
   if (p->npreds > SN_MAX_PREDS || p->naggs > SN_MAX_AGGS) return 0;
+  for (int a = 0; a < p->naggs; a++)   /* MIN/MAX keep the generic path */
+    if (p->aggs[a].kind == SN_AGG_MIN || p->aggs[a].kind == SN_AGG_MAX) return 0;
   const int has_join = p->join_dim != SN_JOIN_NONE && t->dim_hk != NULL;
   if (has_join) return 0;   /* join keeps the generic path */
 
@@ -1364,7 +1374,16 @@ static int32_t sno_run(sno_table *t, const sn_plan *p, int32_t nthreads,
         if (!dst) { rc = SN_ERR_OVERFLOW; break; }
         dst->rowcount += src->rowcount;
         for (int a = 0; a < p->naggs; a++) {
-          dst->sums[a] += src->sums[a];
+          int k = p->aggs[a].kind;
+          if (k == SN_AGG_MIN || k == SN_AGG_MAX) {
+            if (src->counts[a] > 0)
+              dst->sums[a] = dst->counts[a] > 0
+                  ? (k == SN_AGG_MIN ? fmin(dst->sums[a], src->sums[a])
+                                     : fmax(dst->sums[a], src->sums[a]))
+                  : src->sums[a];
+          } else {
+            dst->sums[a] += src->sums[a];
+          }
           dst->counts[a] += src->counts[a];
         }
       }

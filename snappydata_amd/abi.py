@@ -13,7 +13,7 @@ SN_MAX_GROUP_SLOTS = 1024
 SN_KEY_MAX = 48
 
 T_INT32, T_INT64, T_DOUBLE, T_STRING, T_BOOL, T_INT16, T_INT8, T_FLOAT = range(8)
-AGG_SUM, AGG_COUNT_STAR, AGG_AVG = range(3)
+AGG_SUM, AGG_COUNT_STAR, AGG_AVG, AGG_MIN, AGG_MAX = range(5)
 
 OK = 0
 ERR_NOGPU = -6
@@ -122,7 +122,8 @@ def make_plan(table=0, preds=(), group_cols=(), aggs=(), join=None):
     p.naggs = len(aggs)
     for i, (kind, factors) in enumerate(aggs):
         ag = p.aggs[i]
-        ag.kind = {"sum": AGG_SUM, "count": AGG_COUNT_STAR, "avg": AGG_AVG}[kind]
+        ag.kind = {"sum": AGG_SUM, "count": AGG_COUNT_STAR,
+                   "avg": AGG_AVG, "min": AGG_MIN, "max": AGG_MAX}[kind]
         ag.nfactors = len(factors)
         for j, (col, add, mul) in enumerate(factors):
             ag.factors[j].col = col

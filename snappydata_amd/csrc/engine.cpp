@@ -1172,6 +1172,7 @@ struct sn_query {
   int g1cap = 0, g2cap = 0;             /* per-group-col slot counts */
   bool sparse = false;                  /* open-address hash-aggregate mode */
   bool pac = false;                     /* per-agg counts (nullable agg inputs) */
+  bool mm = false;                      /* plan has MIN/MAX aggregates */
   std::vector<long long> sparse_keys;   /* compacted group keys */
   std::vector<double> sparse_rows;      /* [n][na1] accumulator rows */
   bool gint[2] = { false, false };      /* integer group key (stats-ranged) */
@@ -1556,7 +1557,13 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       da.m0 = da.m1 = da.m2 = 0.0;
       da.c0 = da.c1 = da.c2 = 0;
       da.nf = 0;
+      da._p2 = 0;
       const sn_agg &sa = plan->aggs[a];
+      da.op = sa.kind == SN_AGG_MIN ? 1 : sa.kind == SN_AGG_MAX ? 2 : 0;
+      if (da.op != 0 && sa.nfactors < 1) {
+        fail(SN_ERR_BADARG, "MIN/MAX needs at least one factor");
+        return nullptr;
+      }
       if (sa.kind != SN_AGG_COUNT_STAR) {
         /* neutral factors get c = c0 below so their LDS reads CSE away */
         da.nf = sa.nfactors;
@@ -1620,6 +1627,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     q->dev_naggs = grouped ? ndev : plan->naggs;
     dp.naggs = q->dev_naggs;
   }
+  for (int a = 0; a < q->dev_naggs; a++)
+    if (dp.aggs[a].op != 0) q->mm = true;
 
   q->na_t = template_naggs(q->dev_naggs > 0 ? q->dev_naggs : 1);
   q->out_stride = 2 * (size_t)q->na_t + 1;
@@ -1835,8 +1844,11 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     }
   }
   {
+    /* MIN/MAX plans always take the pac layout (per-agg counts carry the
+     * empty-group NULL semantics; routing goes through the op-aware
+     * grouped kernels, keyless included) */
     const bool grouped_mode = plan->ngroup > 0 || q->join_group;
-    q->pac = grouped_mode && (hit ? hit->pac != 0 : agg_nulls);
+    q->pac = (grouped_mode && (hit ? hit->pac != 0 : agg_nulls)) || q->mm;
     dp.pac = q->pac ? 1 : 0;
   }
   if (q->sparse && ntiles > 0) {
@@ -1951,8 +1963,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     int grid = ntiles < SN_GRID_CAP ? ntiles : SN_GRID_CAP;
     if (dp.nslots > 16 && grid > SN_GRID_BIGSLOT) grid = SN_GRID_BIGSLOT;
     const int na1 = dp.pac ? 2 * q->dev_naggs + 1 : q->dev_naggs + 1;
-    size_t nv = dp.nslots <= 1 ? (size_t)(2 * q->na_t + 1)
-                               : (size_t)dp.nslots * na1;
+    size_t nv = (dp.nslots <= 1 && !dp.pac)
+                    ? (size_t)(2 * q->na_t + 1)
+                    : (size_t)(dp.nslots < 1 ? 1 : dp.nslots) * na1;
     const bool big_groups =
         dp.nslots > SN_RESULT_PAGE ||
         (dp.pac && sn_grouped_needs_global(dp.nused, dp.nslots, na1));
@@ -1965,6 +1978,11 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (big_groups &&
         hipMemsetAsync(e->scratch, 0, 8 * nv * 8, e->stream) != hipSuccess) {
       fail(SN_ERR_GENERIC, "accumulator zero"); return nullptr;
+    }
+    if (big_groups && q->mm &&
+        sn_launch_acc_init(e->scratch, 8ll * dp.nslots, q->dev_naggs, na1,
+                           dp_dev, e->stream) != 0) {
+      fail(SN_ERR_GENERIC, "accumulator min/max init"); return nullptr;
     }
     q->ev_start = e->ev_acquire();
     q->ev_stop = e->ev_acquire();
@@ -1982,7 +2000,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     else
       /* global-atomic mode: only the LDS image constrains */
       jit_shape_ok = dp.nslots <= SN_BIG_GROUP_CAP;
-    if (e->jit && jit_shape_ok &&
+    if (e->jit && jit_shape_ok && !q->mm &&
         (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first))) {
       const int *jk = hit ? hit->jit_kinds : jit_kinds;
       int jdel = hit ? hit->jit_del : (jit_any_del ? 1 : 0);
@@ -2008,7 +2026,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       if (rc == 0)
         rc = sn_launch_reduce(e->scratch, big_groups ? 8 : jgrid, (int)nv,
                               q->dev_out, naggs1, (int)q->out_stride,
-                              e->stream);
+                              dp_dev, e->stream);
       q->used_jit = rc == 0;
       if (rc != 0) jfn = nullptr;   /* interpreted kernels take over */
     }
@@ -2083,7 +2101,15 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
       g.rowcount = rowcount;
       for (int a = 0; a < p.naggs; a++) {
         int di = q->agg_map[a];
-        g.sums[a] = di < 0 ? rowcount : row[di];
+        double s = di < 0 ? rowcount : row[di];
+        int k = p.aggs[a].kind;
+        if (di >= 0 && (k == SN_AGG_MIN || k == SN_AGG_MAX)) {
+          /* sparse rows bypass k_reduce: decode the ord-u64 cell here */
+          unsigned long long o;
+          memcpy(&o, &row[di], 8);
+          s = sn_ord_f64_h(o);
+        }
+        g.sums[a] = s;
         g.counts[a] = (di < 0 || !q->pac) ? rowcount
                                           : row[q->dev_naggs + di];
       }
@@ -2110,8 +2136,10 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
     g.rowcount = rowcount;
     for (int a = 0; a < p.naggs; a++) {
       if (p.ngroup == 0 && !q->join_group) {
+        /* keyless MIN/MAX plans run the grouped (pac) layout with 1 slot:
+         * counts land at [naggs + a] instead of the keyless [na_t + a] */
         g.sums[a] = row[a];
-        g.counts[a] = row[q->na_t + a];
+        g.counts[a] = q->pac ? row[p.naggs + a] : row[q->na_t + a];
       } else {
         /* grouped: deduped device sweeps; COUNT(*) = rowcount; per-agg
          * counts live at [dev_naggs + di] when the pac layout ran
@@ -2268,7 +2296,10 @@ extern "C" int32_t sn_query_partials2(sn_query *q, void *dst,
   if (p.ngroup == 0 && !q->join_group) {
     double *o = (double *)block.data();
     const double *row = q->host_out.data();
-    for (int a = 0; a < p.naggs; a++) { o[a] = row[a]; o[p.naggs + a] = row[q->na_t + a]; }
+    for (int a = 0; a < p.naggs; a++) {
+      o[a] = row[a];
+      o[p.naggs + a] = q->pac ? row[p.naggs + a] : row[q->na_t + a];
+    }
     o[2 * p.naggs] = row[2 * q->na_t];
   } else {
     std::vector<GroupOut> groups;
@@ -2369,11 +2400,25 @@ extern "C" int32_t sn_query_merge(sn_query *q, const void *blocks, int64_t strid
   if (!q || !blocks || n_blocks <= 0) return SN_ERR_BADARG;
   const sn_plan &p = q->plan;
   std::vector<GroupOut> merged;
+  /* op-aware fold of one partial contribution into a GroupOut slot */
+  auto fold = [&](GroupOut &g, int a, double s, double c) {
+    int k = p.aggs[a].kind;
+    if (k == SN_AGG_MIN || k == SN_AGG_MAX) {
+      if (c > 0)
+        g.sums[a] = g.counts[a] > 0
+                        ? (k == SN_AGG_MIN ? std::min(g.sums[a], s)
+                                           : std::max(g.sums[a], s))
+                        : s;
+    } else {
+      g.sums[a] += s;
+    }
+    g.counts[a] += c;
+  };
   if (p.ngroup == 0 && !q->join_group) {
     GroupOut g;
     for (int32_t bi = 0; bi < n_blocks; bi++) {
       const double *o = (const double *)((const uint8_t *)blocks + bi * stride);
-      for (int a = 0; a < p.naggs; a++) { g.sums[a] += o[a]; g.counts[a] += o[p.naggs + a]; }
+      for (int a = 0; a < p.naggs; a++) fold(g, a, o[a], o[p.naggs + a]);
       g.rowcount += o[2 * p.naggs];
     }
     merged.push_back(g);
@@ -2396,10 +2441,8 @@ extern "C" int32_t sn_query_merge(sn_query *q, const void *blocks, int64_t strid
           g.keys[k] = slots[i].keys[k];
           g.key_null[k] = slots[i].key_null[k] != 0;
         }
-        for (int a = 0; a < p.naggs; a++) {
-          g.sums[a] += slots[i].sums[a];
-          g.counts[a] += slots[i].counts[a];
-        }
+        for (int a = 0; a < p.naggs; a++)
+          fold(g, a, slots[i].sums[a], slots[i].counts[a]);
         g.rowcount += slots[i].rowcount;
       }
     }

@@ -87,12 +87,33 @@ typedef struct { int64_t lo, hi; int32_t cslot; int32_t _p; } sn_dev_pred_i;
 /* Canonical aggregate input: value = (a0+m0*x0)*(a1+m1*x1)*(a2+m2*x2) with
  * unused factors neutralized to (1 + 0*x_c0) — three fmas, no branches.
  * COUNT(*) is all-neutral (value 1).  nf/c-slots retained for the general
- * path's per-factor null checks. */
+ * path's per-factor null checks.  op: 0 = sum-accumulate (SUM/AVG/COUNT),
+ * 1 = MIN, 2 = MAX — min/max accumulators store the ORDER-PRESERVING u64
+ * encoding (sn_f64_ord) so device atomics are integer atomicMin/Max;
+ * k_reduce (dense) or the host (sparse readback) decodes. */
 typedef struct {
   double a0, m0, a1, m1, a2, m2;
   int32_t c0, c1, c2;
   int32_t nf;                  /* real factors (0 for COUNT(*)) */
+  int32_t op;
+  int32_t _p2;
 } sn_dev_agg;
+
+/* order-preserving bijection double -> u64 (non-NaN): u64 min/max of the
+ * encoding equals double min/max.  Identity elements: min = enc(+inf),
+ * max = enc(-inf). */
+static inline unsigned long long sn_f64_ord_h(double x) {
+  unsigned long long b;
+  __builtin_memcpy(&b, &x, 8);
+  return (b & 0x8000000000000000ull) ? ~b : (b | 0x8000000000000000ull);
+}
+static inline double sn_ord_f64_h(unsigned long long o) {
+  unsigned long long b = (o & 0x8000000000000000ull)
+      ? (o ^ 0x8000000000000000ull) : ~o;
+  double d;
+  __builtin_memcpy(&d, &b, 8);
+  return d;
+}
 
 typedef struct {
   int32_t npreds_d, npreds_i, naggs, ngroup;
@@ -201,10 +222,16 @@ int sn_launch_hash_compact(const long long *hk, const double *hacc,
                            double *orows, int *counter, void *stream);
 
 /* launches only the partial-fold (k_reduce): scratch[nblocks][nv] -> out.
- * Used by the JIT path, whose scan kernel writes the same scratch rows. */
+ * Used by the JIT path, whose scan kernel writes the same scratch rows.
+ * plan_dev (nullable) supplies per-agg ops for MIN/MAX folding. */
 int sn_launch_reduce(const double *dev_scratch, int nblocks, int nv,
                      double *dev_out, int naggs1, int out_stride,
-                     void *stream);
+                     const void *plan_dev, void *stream);
+
+/* initialize a global accumulator's MIN/MAX cells to their ord-encoding
+ * identities (rows x na1 cells; sum/count cells keep the host's memset 0) */
+int sn_launch_acc_init(double *acc, long long rows, int naggs, int na1,
+                       const void *plan_dev, void *stream);
 
 /* query-compiled scan kernels (jit.cpp, hipRTC).  sn_jit_get returns an
  * opaque hipFunction_t for the plan (compiling + caching on first use) or

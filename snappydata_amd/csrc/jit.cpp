@@ -92,7 +92,7 @@ struct sn_dev_batch { int num_rows; int clean; const u64 *del_bm; sn_dev_col col
 struct sn_dev_tile { int batch; int row_start; };
 struct sn_dev_pred_d { double lo, hi; int cslot, _p; };
 struct sn_dev_pred_i { i64 lo, hi; int cslot, _p; };
-struct sn_dev_agg { double a0, m0, a1, m1, a2, m2; int c0, c1, c2, nf; };
+struct sn_dev_agg { double a0, m0, a1, m1, a2, m2; int c0, c1, c2, nf, op, _p2; };
 struct sn_dev_plan {
   int npreds_d, npreds_i, naggs, ngroup, nslots, nused;
   unsigned i64_mask; int gcol[2];

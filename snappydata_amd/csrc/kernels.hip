@@ -491,6 +491,36 @@ __device__ __forceinline__ double eval_agg(const sn_dev_plan *P,
          (A.a2 + A.m2 * sv_agg(P, sval, A.c2, r));
 }
 
+/* order-preserving double<->u64 for MIN/MAX accumulators (device twin of
+ * sn_f64_ord_h): integer atomicMin/Max implement f64 min/max */
+__device__ __forceinline__ unsigned long long f64_ord(double x) {
+  unsigned long long b = (unsigned long long)__double_as_longlong(x);
+  return (b & 0x8000000000000000ull) ? ~b : (b | 0x8000000000000000ull);
+}
+#define ORD_MIN_IDENT 0xFFF8000000000000ull   /* enc(+inf) < this; safe top */
+#define ORD_MAX_IDENT 0x0000000000000000ull   /* below enc(-inf) */
+
+/* op-aware accumulate into an accumulator cell (LDS or global).
+ * Sum cells hold plain doubles; min/max cells hold the ord-u64 encoding
+ * bit-cast into the double slot. */
+template <typename PTR>
+__device__ __forceinline__ void acc_cell(PTR *cell, int op, double va) {
+  if (op == 0) {
+    (void)atomicAdd((double *)cell, va);
+  } else if (op == 1) {
+    (void)atomicMin((unsigned long long *)cell, f64_ord(va));
+  } else {
+    (void)atomicMax((unsigned long long *)cell, f64_ord(va));
+  }
+}
+
+/* identity value (as raw double bits) for an accumulator cell */
+__device__ __forceinline__ double acc_ident(int op) {
+  if (op == 0) return 0.0;
+  return __longlong_as_double(
+      (long long)(op == 1 ? ORD_MIN_IDENT : ORD_MAX_IDENT));
+}
+
 /* per-agg factor validity on general batches: 1 when every referenced
  * factor column is non-null for row r (Spark Sum/Average skip null
  * inputs; COUNT(*) has nf == 0 and is always valid) */
@@ -1004,11 +1034,16 @@ void k_grouped_lds(sn_dev_plan plan,
     unsigned *dst = (unsigned *)P;
     for (unsigned i = tid; i < sizeof(sn_dev_plan) / 4; i += WG) dst[i] = src[i];
   }
-  /* pac: accumulator rows widen to [sums][per-agg counts][rowcount] */
+  /* pac: accumulator rows widen to [sums][per-agg counts][rowcount];
+   * min/max cells initialize to their ord-encoding identity (read ops from
+   * the by-value plan — the LDS mirror copy has no barrier yet) */
   const int pac = plan.pac;
   const int na1 = pac ? 2 * naggs + 1 : naggs + 1;
   const int NV = nslots * na1;
-  for (int i = tid; i < NV; i += WG) bacc[i] = 0.0;
+  for (int i = tid; i < NV; i += WG) {
+    const int a = i % na1;
+    bacc[i] = a < naggs ? acc_ident(plan.aggs[a].op) : 0.0;
+  }
   __syncthreads();
 
   const int gc0 = plan.gcol[0], gc1 = plan.gcol[1];
@@ -1053,7 +1088,7 @@ void k_grouped_lds(sn_dev_plan plan,
             if (!av) continue;
             atomicAdd(&row_acc[naggs + a], 1.0);
           }
-          atomicAdd(&row_acc[a], eval_agg(P, A, sval, r));
+          acc_cell(&row_acc[a], A.op, eval_agg(P, A, sval, r));
         }
         atomicAdd(&row_acc[na1 - 1], 1.0);
       }
@@ -1284,7 +1319,7 @@ void k_grouped_global(sn_dev_plan plan,
             if (!av) continue;
             (void)atomicAdd((double *)&row_acc[naggs + a], 1.0);
           }
-          (void)atomicAdd((double *)&row_acc[a], eval_agg(P, A, sval, r));
+          acc_cell(&row_acc[a], A.op, eval_agg(P, A, sval, r));
         }
         (void)atomicAdd((double *)&row_acc[na1 - 1], 1.0);
       }
@@ -1400,7 +1435,7 @@ void k_grouped_hash(sn_dev_plan plan,
             if (!av) continue;
             (void)atomicAdd((double *)&row_acc[naggs + a], 1.0);
           }
-          (void)atomicAdd((double *)&row_acc[a], eval_agg(P, A, sval, r));
+          acc_cell(&row_acc[a], A.op, eval_agg(P, A, sval, r));
         }
         (void)atomicAdd((double *)&row_acc[na1 - 1], 1.0);
       }
@@ -1471,16 +1506,59 @@ extern "C" int sn_launch_hash_compact(const long long *hk, const double *hacc,
 
 /* fold per-block partial rows into the final output.
  * keyless: final[i] = sum_b scratch[b][i]  (NV = 2*NA_t+1, identical layout)
- * grouped: scratch rows are [slot][naggs+1]; final is [slot][out_stride]
- * with rowcount at out_stride-1. */
+ * grouped: scratch rows are [slot][na1]; final is [slot][out_stride]
+ * with rowcount at out_stride-1.  MIN/MAX cells (op from plan_g, NULL =
+ * all-sum) hold the ord-u64 encoding in the scratch rows; they fold with
+ * integer min/max and DECODE to plain doubles here. */
+__device__ __forceinline__ double ord_f64(unsigned long long o) {
+  unsigned long long b = (o & 0x8000000000000000ull)
+      ? (o ^ 0x8000000000000000ull) : ~o;
+  return __longlong_as_double((long long)b);
+}
+
 __global__ void k_reduce(const double *__restrict__ scratch, int nblocks,
                          int nv, double *__restrict__ out, int naggs1,
-                         int out_stride) {
+                         int out_stride,
+                         const sn_dev_plan *__restrict__ plan_g) {
   /* one block per output value; 256 threads stride the partial rows */
   __shared__ double red[4];
+  __shared__ unsigned long long redu[4];
   const GAS double *src = (const GAS double *)(uintptr_t)scratch;
   const int i = blockIdx.x;
   if (i >= nv) return;
+  int op = 0;
+  if (plan_g && naggs1 > 0) {
+    const GAS sn_dev_plan *P = (const GAS sn_dev_plan *)(uintptr_t)plan_g;
+    const int naggs = P->naggs;
+    const int a = i % naggs1;
+    if (a < naggs) op = P->aggs[a].op;
+  }
+  if (op != 0) {
+    unsigned long long m = op == 1 ? ORD_MIN_IDENT : ORD_MAX_IDENT;
+    for (int b = threadIdx.x; b < nblocks; b += blockDim.x) {
+      const unsigned long long v =
+          (unsigned long long)__double_as_longlong(src[(size_t)b * nv + i]);
+      m = op == 1 ? (v < m ? v : m) : (v > m ? v : m);
+    }
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      const unsigned long long o2 =
+          (unsigned long long)__shfl_down((long long)m, off, 64);
+      m = op == 1 ? (o2 < m ? o2 : m) : (o2 > m ? o2 : m);
+    }
+    if ((threadIdx.x & 63) == 0) redu[threadIdx.x >> 6] = m;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      for (int w = 1; w < 4; w++)
+        m = op == 1 ? (redu[w] < redu[0] ? redu[w] : redu[0])
+                    : (redu[w] > redu[0] ? redu[w] : redu[0]),
+        redu[0] = m;
+      const int slot = i / naggs1, a = i % naggs1;
+      out[(size_t)slot * out_stride + (a < naggs1 - 1 ? a : out_stride - 1)] =
+          ord_f64(redu[0]);
+    }
+    return;
+  }
   double s = 0.0;
   for (int b = threadIdx.x; b < nblocks; b += blockDim.x)
     s += src[(size_t)b * nv + i];
@@ -1496,6 +1574,31 @@ __global__ void k_reduce(const double *__restrict__ scratch, int nblocks,
       out[(size_t)slot * out_stride + (a < naggs1 - 1 ? a : out_stride - 1)] = t;
     }
   }
+}
+
+/* initialize a global accumulator's MIN/MAX cells to their identities
+ * (host memset covers only the sum/count cells' zero) */
+__global__ void k_acc_init(double *__restrict__ acc, long long rows,
+                           int naggs, int na1,
+                           const sn_dev_plan *__restrict__ plan_g) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= rows * na1) return;
+  const int a = (int)(i % na1);
+  if (a >= naggs) return;
+  const GAS sn_dev_plan *P = (const GAS sn_dev_plan *)(uintptr_t)plan_g;
+  const int op = P->aggs[a].op;
+  if (op == 0) return;
+  ((GAS double *)(uintptr_t)acc)[i] = acc_ident(op);
+}
+
+extern "C" int sn_launch_acc_init(double *acc, long long rows, int naggs,
+                                  int na1, const void *plan_dev,
+                                  void *stream) {
+  long long n = rows * na1;
+  hipLaunchKernelGGL(k_acc_init, dim3((unsigned)((n + 255) / 256)), dim3(256),
+                     0, (hipStream_t)stream, acc, rows, naggs, na1,
+                     (const sn_dev_plan *)plan_dev);
+  return (int)hipGetLastError();
 }
 
 /* put-time patch materialization: write value-only update patches straight
@@ -1532,9 +1635,11 @@ extern "C" int sn_launch_patch_apply(void *body, const int32_t *pos,
 
 extern "C" int sn_launch_reduce(const double *dev_scratch, int nblocks,
                                 int nv, double *dev_out, int naggs1,
-                                int out_stride, void *stream) {
+                                int out_stride, const void *plan_dev,
+                                void *stream) {
   hipLaunchKernelGGL(k_reduce, dim3(nv), dim3(WG), 0, (hipStream_t)stream,
-                     dev_scratch, nblocks, nv, dev_out, naggs1, out_stride);
+                     dev_scratch, nblocks, nv, dev_out, naggs1, out_stride,
+                     (const sn_dev_plan *)plan_dev);
   return (int)hipGetLastError();
 }
 
@@ -1564,16 +1669,18 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
   const bool nc4 = plan->nused <= 4;
   const int na1 = plan->pac ? 2 * na + 1 : na + 1;
   int nv, naggs1, out_stride;
-  if (ns <= 1) {
+  if (ns <= 1 && !plan->pac) {
     nv = 2 * na_t + 1; naggs1 = 0; out_stride = nv;
   } else {
-    nv = ns * na1; naggs1 = na1; out_stride = 2 * na_t + 1;
+    /* grouped layout (also keyless MIN/MAX plans routed through the
+     * grouped kernels with one slot) */
+    nv = (ns < 1 ? 1 : ns) * na1; naggs1 = na1; out_stride = 2 * na_t + 1;
   }
 #define KL(A, NCv) hipLaunchKernelGGL((k_keyless<A, NCv>), dim3(grid), dim3(WG), lds, s, \
         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_scratch)
 #define KG(S, NCv) hipLaunchKernelGGL((k_grouped<S, NCv>), dim3(grid), dim3(WG), lds, s, \
         *plan, dev_plan, dev_batches, dev_tiles, ntiles, dev_scratch, out_stride)
-  if (ns <= 1) {
+  if (ns <= 1 && !plan->pac) {
     if (na <= 2) { if (nc4) KL(2, 4); else KL(2, 8); }
     else if (na <= 4) { if (nc4) KL(4, 4); else KL(4, 8); }
     else { if (nc4) KL(12, 4); else KL(12, 8); }
@@ -1621,7 +1728,8 @@ extern "C" int sn_launch_scan_agg(const sn_dev_plan *plan,
 #undef KL
 #undef KG
   hipLaunchKernelGGL(k_reduce, dim3(nv), dim3(WG), 0, s,
-                     dev_scratch, grid, nv, dev_out, naggs1, out_stride);
+                     dev_scratch, grid, nv, dev_out, naggs1, out_stride,
+                     dev_plan);
   err = hipGetLastError();
   return (int)err;
 }

@@ -1,0 +1,167 @@
+"""MIN/MAX aggregates (the reference's Min/Max DeclarativeAggregates,
+SnappyHashAggregateExec.scala:450-500 family): keyless, grouped (dense dict
+keys, big dense slots, sparse hash keys), nullable inputs, and the
+partial-block merge.  Device accumulation uses integer atomicMin/Max over an
+order-preserving f64<->u64 encoding; MIN/MAX of an empty/all-null group is
+NULL."""
+import numpy as np
+import pytest
+
+from oracle import pyoracle as po
+from snappydata_amd import abi, engine as se
+
+REL = 1e-12   # min/max pick EXACT input values — no summation-order slack
+
+
+@pytest.fixture
+def eng():
+    e = se.Engine(device=0)
+    yield e
+    e.close()
+
+
+def test_oracle_minmax_cpu():
+    n = 100_000
+    rng = np.random.default_rng(83)
+    keys = [b"K%d" % v for v in rng.integers(0, 7, n)]
+    w = rng.standard_normal(n) * 100
+    t = po.OracleTable([po.T_STRING, po.T_DOUBLE])
+    t.add_batch(n, [po.encode(po.T_STRING, po.ENC_DICT, keys),
+                    po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)])
+    rows = po.result_rows(t.query(po.make_plan(
+        group_cols=[0], aggs=[("min", [(1, 0.0, 1.0)]),
+                              ("max", [(1, 0.0, 1.0)])])))
+    got = {k[0]: v for k, v in rows}
+    for g in range(7):
+        m = np.array([k == b"K%d" % g for k in keys])
+        assert got[f"K{g}"][0] == w[m].min()
+        assert got[f"K{g}"][1] == w[m].max()
+    # OpenMP thread-merge path folds min/max correctly too
+    rows_mt = po.result_rows(t.query(po.make_plan(
+        group_cols=[0], aggs=[("min", [(1, 0.0, 1.0)]),
+                              ("max", [(1, 0.0, 1.0)])]), nthreads=4))
+    assert rows_mt == rows
+
+
+@pytest.mark.gpu
+def test_keyless_minmax(eng):
+    n = 500_000
+    rng = np.random.default_rng(89)
+    v = rng.standard_normal(n) * 1e6
+    d = rng.random(n)
+    t = eng.table_define("tmm0", [(abi.T_DOUBLE, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": v}, {"data": d}], n, batch_rows=100_000)
+    plan_kw = dict(preds=[dict(col=1, is_double=True, lo=0.5)],
+                   aggs=[("min", [(0, 0.0, 1.0)]), ("max", [(0, 0.0, 1.0)]),
+                         ("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    q = eng.query(abi.make_plan(table=t, **plan_kw))
+    rows = q.rows()
+    m = d >= 0.5
+    assert rows[0][1][0] == v[m].min()
+    assert rows[0][1][1] == v[m].max()
+    assert abs(rows[0][1][2] - v[m].sum()) <= 1e-6 * max(1.0, abs(v[m].sum()))
+    assert rows[0][1][3] == float(m.sum())
+
+
+@pytest.mark.gpu
+def test_grouped_minmax_dict_keys(eng):
+    n = 400_000
+    rng = np.random.default_rng(97)
+    keys = [b"G%02d" % v for v in rng.integers(0, 40, n)]
+    w = rng.standard_normal(n) * 50
+    cols = [po.encode(po.T_STRING, po.ENC_DICT, keys),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
+    t = eng.table_define("tmmg", [(abi.T_STRING, False), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 1, 0, n, cols)
+    plan_kw = dict(group_cols=[0],
+                   aggs=[("min", [(1, 0.0, 1.0)]), ("max", [(1, 0.0, 1.0)]),
+                         ("avg", [(1, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_STRING, po.T_DOUBLE])
+    ot.add_batch(n, cols)
+    orows = po.result_rows(ot.query(po.make_plan(**plan_kw)))
+    assert len(grows) == len(orows) == 40
+    for (gk, gv), (ok_, ov) in zip(grows, orows):
+        assert gk == ok_
+        assert gv[0] == ov[0] and gv[1] == ov[1]   # min/max bit-exact
+        assert abs(gv[2] - ov[2]) <= 1e-9 * max(1.0, abs(ov[2]))
+        assert gv[3] == ov[3]
+
+
+@pytest.mark.gpu
+def test_sparse_minmax_int64_keys(eng):
+    n = 300_000
+    rng = np.random.default_rng(101)
+    keys = rng.integers(0, 3_000, n).astype(np.int64) * (1 << 34)
+    w = rng.standard_normal(n)
+    t = eng.table_define("tmmsp", [(abi.T_INT64, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": keys}, {"data": w}], n, batch_rows=100_000)
+    plan_kw = dict(group_cols=[0],
+                   aggs=[("min", [(1, 0.0, 1.0)]), ("max", [(1, 0.0, 1.0)]),
+                         ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_INT64, po.T_DOUBLE])
+    ot.add_batch(n, [po.encode(po.T_INT64, po.ENC_UNCOMPRESSED, keys),
+                     po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)])
+    orows = ot.query_groups(po.make_plan(**plan_kw), nthreads=16)
+    assert len(grows) == len(orows)
+    for (gk, gv), (ok_, ov) in zip(grows, orows):
+        assert gk == ok_ and gv[0] == ov[0] and gv[1] == ov[1] and gv[2] == ov[2]
+
+
+@pytest.mark.gpu
+def test_minmax_nullable_inputs_and_empty_group_null(eng):
+    """All-null inputs for one group yield NULL min/max (Spark semantics)."""
+    n = 60_000
+    rng = np.random.default_rng(103)
+    keys = [b"A" if i % 3 else b"B" for i in range(n)]
+    w = rng.random(n)
+    valid = np.array([0 if k == b"B" else 1 for k in keys], dtype=np.uint8)
+    cols = [po.encode(po.T_STRING, po.ENC_DICT, keys),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w, valid=valid)]
+    t = eng.table_define("tmmnull", [(abi.T_STRING, False), (abi.T_DOUBLE, True)])
+    eng.batch_put(t, 1, 0, n, cols)
+    grows = eng.query(abi.make_plan(
+        table=t, group_cols=[0],
+        aggs=[("min", [(1, 0.0, 1.0)]), ("count", [])])).rows()
+    got = {k[0]: v for k, v in grows}
+    m = (valid == 1)
+    assert got["A"][0] == w[m].min()
+    assert got["B"][0] is None                 # all-null group -> NULL
+    assert got["B"][1] == float((valid == 0).sum())
+
+
+@pytest.mark.gpu
+def test_minmax_partial_merge(eng):
+    """Two shard engines' grouped partials min/max-merge correctly."""
+    n = 200_000
+    rng = np.random.default_rng(107)
+    keys = [b"K%d" % v for v in rng.integers(0, 5, n)]
+    w = rng.standard_normal(n)
+
+    def load(e2):
+        t2 = e2.table_define("t", [(abi.T_STRING, False), (abi.T_DOUBLE, False)])
+        for st in range(0, n, 50_000):
+            en = min(n, st + 50_000)
+            sub = [po.encode(po.T_STRING, po.ENC_DICT, keys[st:en]),
+                   po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w[st:en])]
+            e2.batch_put(t2, st, st // 50_000, en - st, sub)
+        return t2
+
+    plan_kw = dict(group_cols=[0],
+                   aggs=[("min", [(1, 0.0, 1.0)]), ("max", [(1, 0.0, 1.0)]),
+                         ("count", [])])
+    ref = eng.query(abi.make_plan(table=load(eng), **plan_kw)).rows()
+
+    e0 = se.Engine(device=0, shard_rank=0, shard_count=2)
+    e1 = se.Engine(device=0, shard_rank=1, shard_count=2)
+    try:
+        q0 = e0.query(abi.make_plan(table=load(e0), **plan_kw))
+        q1 = e1.query(abi.make_plan(table=load(e1), **plan_kw))
+        bb = q0.partial_bytes()
+        blocks = np.concatenate([q0.partials_host(), q1.partials_host()])
+        q0.merge_host(np.ascontiguousarray(blocks), bb, 2)
+        assert q0.rows() == ref
+    finally:
+        e0.close()
+        e1.close()
