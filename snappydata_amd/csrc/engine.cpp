@@ -75,21 +75,35 @@ static inline float   rd_f32(const uint8_t *p) { float v; memcpy(&v, p, 4); retu
 static inline int16_t rd_i16(const uint8_t *p) { int16_t v; memcpy(&v, p, 2); return v; }
 
 /* ------------------------------------------------------------------ */
-/* device memory arena (bump allocator over big HBM slabs)             */
+/* device memory arena: bump allocator over big HBM slabs, plus a
+ * size-bucketed free list so per-query transients (result buffers,
+ * uncached descriptor sets, sparse plan copies) recycle instead of
+ * leaking ~100 KB/query in a long-lived executor */
 struct Arena {
   int device = -1;               /* -1: host-only shadow mode */
   std::vector<void *> slabs;
   std::vector<void *> host_allocs;   /* host-only mode: freed on destroy */
   size_t slab_sz = 512ull << 20;
   size_t off = 0;
+  std::multimap<size_t, void *> freelist;   /* rounded size -> block */
   std::mutex mu;
+
+  static size_t rnd(size_t n) { return (n + 255) & ~size_t(255); }
 
   void *alloc(size_t n) {
     std::lock_guard<std::mutex> g(mu);
-    n = (n + 255) & ~size_t(255);
+    n = rnd(n);
     if (device < 0) {
       void *p = malloc(n);
       if (p) host_allocs.push_back(p);
+      return p;
+    }
+    /* exact-bucket reuse (blocks are only ever freed at the size they
+     * were allocated with, so exact match is the common case) */
+    auto it = freelist.find(n);
+    if (it != freelist.end()) {
+      void *p = it->second;
+      freelist.erase(it);
       return p;
     }
     if (slabs.empty() || off + n > slab_sz) {
@@ -103,6 +117,13 @@ struct Arena {
     void *p = (char *)slabs.back() + off;
     off += n;
     return p;
+  }
+  /* return a block for reuse (device mode only; host blocks die with the
+   * arena).  n must be the original request size. */
+  void release(void *p, size_t n) {
+    if (!p || device < 0) return;
+    std::lock_guard<std::mutex> g(mu);
+    freelist.emplace(rnd(n), p);
   }
   ~Arena() {
     for (void *p : slabs) (void)hipFree(p);
@@ -158,6 +179,11 @@ struct Batch {
   std::vector<const int32_t *> rle_ends_dev;
   std::vector<const double *> rle_vals_dev;
   std::vector<int32_t> rle_n;
+  /* premultiplied dictionary maps, cached per (mul, null_gid): profiles
+   * showed ~2000 per-batch map uploads on a first Q1 SF=100 submit —
+   * local2global is fixed per batch, so the device map is reusable
+   * across queries with the same group geometry */
+  std::vector<std::map<std::pair<int32_t, int32_t>, const int32_t *>> dictmap_cache;
   std::vector<ColMeta> cols;
   /* stats (parsed) */
   bool stats_valid = false;
@@ -176,6 +202,7 @@ struct DescCache {
   size_t batch_count = 0;       /* invalidation: table grew */
   const void *db_dev = nullptr;
   const void *tl_dev = nullptr;
+  size_t db_bytes = 0, tl_bytes = 0;   /* released to the free list on evict */
   int32_t ntiles = 0;
   int64_t rows = 0;
   /* JIT eligibility of this descriptor set: every batch clean with these
@@ -199,6 +226,9 @@ struct Table {
    * bytes cannot travel in results/partial blocks — queries grouping on
    * such a column fail loudly instead of silently merging truncated keys */
   std::vector<int32_t> gdict_maxlen;
+  /* last adequate sparse hash-table capacity (log2): later queries start
+   * there instead of rediscovering it through grow-and-retry */
+  int sparse_cap_hint = 0;
   int64_t total_rows = 0;
   std::mutex mu;
 };
@@ -886,6 +916,7 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
   b.rle_ends_dev.resize(nc, nullptr);
   b.rle_vals_dev.resize(nc, nullptr);
   b.rle_n.resize(nc, 0);
+  b.dictmap_cache.resize(nc);
 
   std::lock_guard<std::mutex> g(t->mu);
   std::vector<uint8_t> decomp;
@@ -1106,6 +1137,10 @@ extern "C" int32_t sn_batch_mutate(sn_engine *e, int32_t table, int64_t uuid,
   if (stats) apply_stats(t, b[0], stats);
   else b->stats_valid = false;   /* old bounds no longer trustworthy */
   /* cached descriptor sets reference the old patch/delete device data */
+  for (auto &dc : t->desc_caches) {
+    e->arena.release((void *)dc.db_dev, dc.db_bytes);
+    e->arena.release((void *)dc.tl_dev, dc.tl_bytes);
+  }
   t->desc_caches.clear();
   return SN_OK;
 }
@@ -1193,6 +1228,8 @@ struct sn_query {
   int64_t rows_scanned = 0;             /* host metric: rows in unskipped batches */
   int64_t batches_seen = 0, batches_skipped = 0;
   hipEvent_t ev_start = nullptr, ev_stop = nullptr;  /* brackets the scan kernel */
+  /* per-query arena blocks, recycled via the free list at destroy */
+  std::vector<std::pair<void *, size_t>> owned;
   float kernel_ms = -1.0f;
   bool used_jit = false;                /* query-compiled kernel ran */
   bool done = false;
@@ -1735,19 +1772,27 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
             return nullptr;
           }
           dc.kind = m.type_id == SN_ENC_DICTIONARY ? SN_K_DICT16 : SN_K_DICT32;
-          /* build the local->premultiplied-global map for this query */
+          /* local->premultiplied-global map, cached per (mul, null_gid) —
+           * local2global is fixed per batch, so queries with the same
+           * group geometry reuse one device map */
           bool is_g2 = plan->ngroup == 2 && plan->group_cols[1] == c;
           int mul = is_g2 ? 1 : (q->g2cap > 0 ? q->g2cap : 1);
-          std::vector<int32_t> map(m.local2global.size() + 1);
-          for (size_t i = 0; i < m.local2global.size(); i++)
-            map[i] = m.local2global[i] * mul;
-          /* index == numElements denotes null (DictionaryEncoding.scala:90);
-           * only nullable key columns reserve a null slot */
           int gn = is_g2 ? q->gnull2 : q->gnull1;
           int null_gid_local = gn >= 0 ? (is_g2 ? gn : gn * q->g2cap) : 0;
-          map[m.local2global.size()] = null_gid_local;
           dc.null_gid = null_gid_local;
-          dc.dictmap = (const int32_t *)up(e, map.data(), map.size() * 4);
+          auto &mcache = b.dictmap_cache[c];
+          auto mit = mcache.find({ mul, null_gid_local });
+          if (mit != mcache.end()) {
+            dc.dictmap = mit->second;
+          } else {
+            std::vector<int32_t> map(m.local2global.size() + 1);
+            for (size_t i = 0; i < m.local2global.size(); i++)
+              map[i] = m.local2global[i] * mul;
+            /* index == numElements denotes null (DictionaryEncoding.scala:90) */
+            map[m.local2global.size()] = null_gid_local;
+            dc.dictmap = (const int32_t *)up(e, map.data(), map.size() * 4);
+            mcache.emplace(std::make_pair(mul, null_gid_local), dc.dictmap);
+          }
           break;
         }
         default:
@@ -1807,6 +1852,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   if (!q->sparse) {
     q->dev_out = (double *)e->arena.alloc(out_n * 8);
     if (!q->dev_out) { fail(SN_ERR_NOMEM, "out alloc"); return nullptr; }
+    q->owned.push_back({ q->dev_out, out_n * 8 });
     if (hipMemsetAsync(q->dev_out, 0, out_n * 8, e->stream) != hipSuccess) {
       fail(SN_ERR_GENERIC, "memset out"); return nullptr;
     }
@@ -1830,11 +1876,26 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     }
     db_dev = dbp; tl_dev = tlp;
     ntiles = (int32_t)htiles.size();
+    if (skip_count != 0) {
+      /* not cacheable (stats-skip removed batches): this query owns the
+       * descriptor blocks; they recycle at sn_query_destroy */
+      q->owned.push_back({ dbp, hbatches.size() * sizeof(sn_dev_batch) });
+      q->owned.push_back({ tlp, htiles.size() * sizeof(sn_dev_tile) });
+    }
     if (skip_count == 0) {
-      if (t->desc_caches.size() > 16) t->desc_caches.clear();
+      if (t->desc_caches.size() > 16) {
+        /* evict the oldest cached set (FIFO); stream ordering guarantees
+         * any in-flight kernel reading it finishes before a reuse write */
+        DescCache &old = t->desc_caches.front();
+        e->arena.release((void *)old.db_dev, old.db_bytes);
+        e->arena.release((void *)old.tl_dev, old.tl_bytes);
+        t->desc_caches.erase(t->desc_caches.begin());
+      }
       DescCache dc;
       dc.sig = sig; dc.batch_count = t->batches.size();
       dc.db_dev = db_dev; dc.tl_dev = tl_dev; dc.ntiles = ntiles;
+      dc.db_bytes = hbatches.size() * sizeof(sn_dev_batch);
+      dc.tl_bytes = htiles.size() * sizeof(sn_dev_tile);
       dc.rows = q->rows_scanned;
       dc.jit_ok = (jit_ok_b && !jit_first) ? 1 : 0;
       dc.jit_del = jit_any_del ? 1 : 0;
@@ -1857,9 +1918,14 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
      * are compacted on device and read back HERE (inside submit, stream
      * synchronized), so concurrent queries never share the live table. */
     const int naggs1 = (q->pac ? 2 : 1) * q->dev_naggs + 1;
-    int cap_log2 = 20;
-    while (cap_log2 > 12 && (1ll << (cap_log2 - 1)) >= 4 * q->rows_scanned)
+    /* capacity: 2x headroom over the worst-case distinct count keeps the
+     * linear probe short (load factor <= 0.5 at full distinctness); the
+     * fill counter grows the table before probing turns pathological,
+     * and the per-table hint skips the rediscovery next query */
+    int cap_log2 = 21;
+    while (cap_log2 > 12 && (1ll << (cap_log2 - 1)) >= 2 * q->rows_scanned)
       cap_log2--;                      /* small tables: smaller table */
+    if (t->sparse_cap_hint > cap_log2) cap_log2 = t->sparse_cap_hint;
     q->ev_start = e->ev_acquire();
     q->ev_stop = e->ev_acquire();
     bool done_h = false;
@@ -1879,11 +1945,17 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
           hipMemcpy(dps_dev, &dps, sizeof(dps), hipMemcpyHostToDevice) != hipSuccess) {
         fail(SN_ERR_NOMEM, "sparse plan upload"); return nullptr;
       }
+      q->owned.push_back({ dps_dev, sizeof(dps) });
       if (hipMemsetAsync(e->hws_keys, 0xff, cap * 8, e->stream) != hipSuccess ||
           hipMemsetAsync(e->hws_acc, 0, (cap + 1) * (size_t)naggs1 * 8,
                          e->stream) != hipSuccess ||
-          hipMemsetAsync(e->hws_flags, 0, 8, e->stream) != hipSuccess) {
+          hipMemsetAsync(e->hws_flags, 0, 16, e->stream) != hipSuccess) {
         fail(SN_ERR_GENERIC, "sparse workspace zero"); return nullptr;
+      }
+      if (q->mm &&
+          sn_launch_acc_init(e->hws_acc, (long long)cap + 1, q->dev_naggs,
+                             naggs1, dps_dev, e->stream) != 0) {
+        fail(SN_ERR_GENERIC, "sparse min/max init"); return nullptr;
       }
       if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
       int rc = sn_launch_hash_scan(&dps, (const sn_dev_plan *)dps_dev,
@@ -1899,10 +1971,14 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       if (hipStreamSynchronize(e->stream) != hipSuccess) {
         fail(SN_ERR_GENERIC, "hash-agg sync"); return nullptr;
       }
-      int32_t ovf = 0;
-      (void)hipMemcpy(&ovf, e->hws_flags, 4, hipMemcpyDeviceToHost);
-      if (ovf) {
-        if (cap_log2 >= 24) {
+      int32_t fl[4] = { 0, 0, 0, 0 };
+      (void)hipMemcpy(fl, e->hws_flags, 16, hipMemcpyDeviceToHost);
+      const int32_t ovf = fl[0];
+      const int64_t fill = fl[2];
+      /* grow on hard overflow OR load factor > 0.6 (probe chains degrade
+       * sharply past that); results so far are correct either way */
+      if (ovf || (fill * 5 > (int64_t)cap * 3 && cap_log2 < 24)) {
+        if (ovf && cap_log2 >= 24) {
           fail(SN_ERR_OVERFLOW,
                "sparse group cardinality exceeds the 2^24 hash table");
           return nullptr;
@@ -1910,6 +1986,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         cap_log2 = std::min(cap_log2 + 2, 24);
         continue;
       }
+      t->sparse_cap_hint = std::max(t->sparse_cap_hint, cap_log2);
       if (hipMemsetAsync(e->hws_flags + 1, 0, 4, e->stream) != hipSuccess ||
           sn_launch_hash_compact(e->hws_keys, e->hws_acc, (int)cap, naggs1,
                                  e->hws_okeys, e->hws_orows,
@@ -1955,7 +2032,11 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
             hipMemcpy(dp_dev, &dp, sizeof(dp), hipMemcpyHostToDevice) != hipSuccess) {
           fail(SN_ERR_NOMEM, "plan upload"); return nullptr;
         }
-        if (e->dp_cache.size() > 64) e->dp_cache.clear();  /* dev copies stay in arena */
+        if (e->dp_cache.size() > 64) {
+          for (auto &kv : e->dp_cache)
+            e->arena.release(kv.second, sizeof(sn_dev_plan));
+          e->dp_cache.clear();
+        }
         e->dp_cache[dph] = dp_dev;
       }
     }
@@ -2258,6 +2339,9 @@ static void sn_detach_queries(sn_engine *e) {
 
 extern "C" void sn_query_destroy(sn_query *q) {
   if (q && q->e) {
+    /* the stream may still reference this query's buffers */
+    (void)sn_query_wait(q);
+    for (auto &pr : q->owned) q->e->arena.release(pr.first, pr.second);
     std::lock_guard<std::mutex> g(q->e->aux_mu);
     q->e->live_q.erase(q);
   }

@@ -1366,7 +1366,13 @@ __device__ __forceinline__ int hash_probe(long long key, long long *hk_,
       const long long old = (long long)atomicCAS(
           (unsigned long long *)&hk[h], (unsigned long long)SN_HASH_EMPTY,
           (unsigned long long)key);
-      if (old == SN_HASH_EMPTY || old == key) return (int)h;
+      if (old == SN_HASH_EMPTY) {
+        /* we inserted: count the fill so the host can grow the table
+         * BEFORE load factor makes linear probing pathological */
+        (void)atomicAdd((int *)(uintptr_t)(ovf + 2), 1);
+        return (int)h;
+      }
+      if (old == key) return (int)h;
       /* lost the insert race to a DIFFERENT key: step on */
     }
     h = (h + 1) & mask;
