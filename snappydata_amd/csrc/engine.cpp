@@ -1957,11 +1957,37 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
                              naggs1, dps_dev, e->stream) != 0) {
         fail(SN_ERR_GENERIC, "sparse min/max init"); return nullptr;
       }
+      /* query-compiled twin first (plan structure compile-time, capacity
+       * tokenized); any miss falls back to the interpreted hash kernel */
+      void *jfn = nullptr;
+      if (e->jit && !q->mm && !dp.pac && q->dev_naggs <= 12 &&
+          (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first))) {
+        const int *jk = hit ? hit->jit_kinds : jit_kinds;
+        int jdel = hit ? hit->jit_del : (jit_any_del ? 1 : 0);
+        jfn = sn_jit_get(e->jit, &dps, jk, 0, q->na_t, jdel);
+      }
       if (q->ev_start) (void)hipEventRecord(q->ev_start, e->stream);
-      int rc = sn_launch_hash_scan(&dps, (const sn_dev_plan *)dps_dev,
-                                   (const sn_dev_batch *)db_dev,
-                                   (const sn_dev_tile *)tl_dev, ntiles,
-                                   e->stream);
+      int rc = -1;
+      if (jfn) {
+        int jgrid;
+        if (ntiles <= SN_GRID_CAP) jgrid = ntiles;
+        else {
+          int rounds = (ntiles + SN_GRID_CAP - 1) / SN_GRID_CAP;
+          jgrid = (ntiles + rounds - 1) / rounds;
+        }
+        rc = sn_jit_launch(jfn, jgrid, (const sn_dev_batch *)db_dev,
+                           (const sn_dev_tile *)tl_dev, ntiles,
+                           e->hws_acc, (const int64_t *)e->hws_keys,
+                           (const int32_t *)e->hws_flags, nullptr,
+                           (const sn_dev_plan *)dps_dev, e->stream);
+        q->used_jit = rc == 0;
+        if (rc != 0) jfn = nullptr;
+      }
+      if (!jfn)
+        rc = sn_launch_hash_scan(&dps, (const sn_dev_plan *)dps_dev,
+                                 (const sn_dev_batch *)db_dev,
+                                 (const sn_dev_tile *)tl_dev, ntiles,
+                                 e->stream);
       if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);
       if (rc != 0) {
         fail(SN_ERR_GENERIC, "hash-agg launch: %s",

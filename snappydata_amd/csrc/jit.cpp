@@ -69,7 +69,12 @@ static std::string gen_source(const sn_dev_plan *p, const int *kinds,
                               int nslots, int na_t, int has_del) {
   const int NC = p->nused;
   const int NA = p->naggs;
-  const int grouped = nslots > 1;
+  /* sparse: open-address hash aggregate — accumulator IS `out` (hacc),
+   * key table and flag words arrive via the jkeys/jpayload params, and the
+   * capacity is tokenized (read from the device plan) so grow-and-retry
+   * reuses one compiled kernel */
+  const int sparse_mode = p->sparse != 0;
+  const int grouped = nslots > 1 || sparse_mode;
   std::string o;
   o += R"(
 typedef double double2_t __attribute__((ext_vector_type(2)));
@@ -123,7 +128,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   /* >8 slots: block-level LDS accumulators (f64 LDS atomics) instead of
    * per-lane registers — the high-cardinality SHAMap analogue.  Occupancy 1
    * (the LDS image + accumulator array leave no room for a second group). */
-  const int glob_mode = grouped && nslots > 1024;   /* HBM accumulator */
+  const int glob_mode = (grouped && nslots > 1024) || sparse_mode;
   const int lds_mode = grouped && nslots > 8 && !glob_mode;
   const int wbin_pre = grouped && !lds_mode && !glob_mode && NA <= 2 &&
                        nslots * (NA + 1) >= 12;
@@ -178,7 +183,16 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   o += "  const int tid = threadIdx.x;\n";
 
   /* accumulators */
-  if (glob_mode) {
+  if (sparse_mode) {
+    /* single HBM accumulator (sparse keys -> low per-address contention;
+     * probe position indexes the rows); capacity tokenized from the plan */
+    emitf(o, "  GAS double *gacc = (GAS double *)(u64)out;\n"
+             "  GAS i64 *hkeys = (GAS i64 *)(u64)jkeys_p;\n"
+             "  GAS int *hflags = (GAS int *)(u64)jpayload_p;\n"
+             "  const int hcl = P->hcap_log2;\n"
+             "  const unsigned hmask = (1u << hcl) - 1;\n"
+             "  const int hcap = 1 << hcl;\n");
+  } else if (glob_mode) {
     /* HBM accumulator IS the out/scratch pointer (host-zeroed), privatized
      * 8 ways by XCD (blockIdx & 7 matches the dispatch round-robin) so
      * same-slot atomics from different XCDs never contend and stay in the
@@ -444,7 +458,43 @@ __device__ __forceinline__ u64 mix64(u64 x) {
             (1u << p->jcap_log2) - 1, (1u << p->jcap_log2) - 1);
     }
   }
-  if (grouped) {
+  if (sparse_mode) {
+    /* open-address probe (ByteBufferHashMap.putBufferIfAbsent shape):
+     * key geometry is compile-time, capacity tokenized */
+    if (p->ngroup == 1) {
+      if ((p->i64_mask >> p->gcol[0]) & 1u)
+        emitf(o, "        const i64 skey = __double_as_longlong(sval[%d][r]);\n",
+              p->gcol[0]);
+      else
+        emitf(o, "        const i64 skey = (i64)sval[%d][r];\n", p->gcol[0]);
+    } else {
+      emitf(o, "        const i64 skey = (i64)(((u64)(unsigned)(int)sval[%d][r]"
+               " << 32) | (unsigned)(int)sval[%d][r]);\n",
+            p->gcol[0], p->gcol[1]);
+    }
+    o += R"(        int slot = -1;
+        if (ok) {
+          if (skey == -1ll) slot = hcap;     /* sentinel-valued key */
+          else {
+            unsigned h = (unsigned)mix64((u64)skey) & hmask;
+            for (unsigned it = 0; it <= hmask; ++it) {
+              const i64 k0 = hkeys[h];
+              if (k0 == skey) { slot = (int)h; break; }
+              if (k0 == -1ll) {
+                const i64 old = (i64)atomicCAS((unsigned long long *)&hkeys[h],
+                                               (unsigned long long)-1ll,
+                                               (unsigned long long)skey);
+                if (old == -1ll) { atomicAdd((int *)(hflags + 2), 1); slot = (int)h; break; }
+                if (old == skey) { slot = (int)h; break; }
+              }
+              h = (h + 1) & hmask;
+            }
+            if (slot < 0) atomicOr((int *)hflags, 1);
+          }
+        }
+        ok &= slot >= 0;
+)";
+  } else if (grouped) {
     if (p->jkeys && p->jmode == 1) {
       o += "        const int slot = pay > 0 ? pay : 0;\n";
     } else {
@@ -590,6 +640,16 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
    * so every literal value of the same shape reuses one compiled kernel
    * (the reference's tokenized plan cache) */
   sn_dev_plan shape = *p;
+  /* pointers travel as kernel params; only their NULLness shapes the
+   * source.  Capacity (hcap_log2) is tokenized, so grow-and-retry and
+   * fresh workspaces reuse one compiled kernel. */
+  shape.jkeys = shape.jkeys ? (const int64_t *)1 : nullptr;
+  shape.jpayload = shape.jpayload ? (const int32_t *)1 : nullptr;
+  shape.jlut = shape.jlut ? (const int32_t *)1 : nullptr;
+  shape.hkeys = nullptr;
+  shape.hacc = nullptr;
+  shape.hflags = nullptr;
+  shape.hcap_log2 = 0;
   auto trivm = [](double a, double m) { return (a == 0.0 && m == 1.0) ? 1.0 : 0.0; };
   for (int i = 0; i < 8; i++) { shape.preds_d[i].lo = shape.preds_d[i].hi = 0.0; }
   for (int i = 0; i < 4; i++) { shape.preds_i[i].lo = shape.preds_i[i].hi = 0; }
