@@ -918,7 +918,11 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
   b.rle_n.resize(nc, 0);
   b.dictmap_cache.resize(nc);
 
-  std::lock_guard<std::mutex> g(t->mu);
+  /* Phase 1 — NO table lock: decompress, parse, validate, materialize and
+   * upload are all batch-local (the arena has its own mutex), so concurrent
+   * puts from many ingest threads overlap their heavy work.  Only the
+   * global-dictionary interning, delta decode (which interns) and the
+   * batches-vector append need t->mu (phase 2 below). */
   std::vector<uint8_t> decomp;
   for (int c = 0; c < nc; c++) {
     const uint8_t *blob = (const uint8_t *)columns[c].data;
@@ -928,22 +932,7 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
     if (dec == 1) { blob = decomp.data(); len = (int64_t)decomp.size(); }
     int rc = parse_blob(blob, len, t->schema[c].dtype, &b.cols[c]);
     if (rc != SN_OK) return fail(rc, "column %d blob parse failed", c);
-    /* intern dictionary into table-global dict (string dict cols) */
     if (t->schema[c].dtype == SN_TYPE_STRING && !b.cols[c].dict.empty()) {
-      auto &l2g = b.cols[c].local2global;
-      l2g.reserve(b.cols[c].dict.size());
-      for (auto &s : b.cols[c].dict) {
-        auto it = t->gdict_idx[c].find(s);
-        int32_t gid;
-        if (it == t->gdict_idx[c].end()) {
-          gid = (int32_t)t->gdict[c].size();
-          t->gdict[c].push_back(s);
-          t->gdict_idx[c].emplace(s, gid);
-          if ((int32_t)s.size() > t->gdict_maxlen[c])
-            t->gdict_maxlen[c] = (int32_t)s.size();
-        } else gid = it->second;
-        l2g.push_back(gid);
-      }
       /* validate the index array: a corrupt index would walk the kernel
        * off the end of the local->global map (index == dict size is the
        * NULL sentinel, DictionaryEncoding.scala:90) */
@@ -1093,9 +1082,31 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
 
   { int32_t rc_ = apply_delete_mask(e, b, delete_mask); if (rc_ != SN_OK) return rc_; }
 
-  { int32_t rc_ = apply_deltas(e, t, b, deltas); if (rc_ != SN_OK) return rc_; }
-
   apply_stats(t, b, stats);
+
+  /* Phase 2 — under t->mu: global-dictionary interning, delta decode
+   * (interns new entries) and the append */
+  std::lock_guard<std::mutex> g(t->mu);
+  for (int c = 0; c < nc; c++) {
+    if (t->schema[c].dtype != SN_TYPE_STRING || b.cols[c].dict.empty())
+      continue;
+    auto &l2g = b.cols[c].local2global;
+    l2g.reserve(b.cols[c].dict.size());
+    for (auto &s : b.cols[c].dict) {
+      auto it = t->gdict_idx[c].find(s);
+      int32_t gid;
+      if (it == t->gdict_idx[c].end()) {
+        gid = (int32_t)t->gdict[c].size();
+        t->gdict[c].push_back(s);
+        t->gdict_idx[c].emplace(s, gid);
+        if ((int32_t)s.size() > t->gdict_maxlen[c])
+          t->gdict_maxlen[c] = (int32_t)s.size();
+      } else gid = it->second;
+      l2g.push_back(gid);
+    }
+  }
+
+  { int32_t rc_ = apply_deltas(e, t, b, deltas); if (rc_ != SN_OK) return rc_; }
 
   t->total_rows += b.num_rows;
   t->batches.push_back(std::move(b));
