@@ -145,7 +145,9 @@ __device__ __forceinline__ u64 mix64(u64 x) {
            "  const GAS int *jlut = (const GAS int *)(u64)jlut_p;\n"
            "  const GAS sn_dev_plan *P = (const GAS sn_dev_plan *)(u64)plan_p;\n"
            "  (void)jkeys; (void)jpayload; (void)jlut; (void)P;\n",
-        (lds_mode || glob_mode) ? 1 : (wbin_pre ? 4 : 2));
+        /* sparse is random-probe latency-bound: more waves cover the
+         * dependent loads (its LDS footprint is only the sval image) */
+        sparse_mode ? 4 : (lds_mode || glob_mode) ? 1 : (wbin_pre ? 4 : 2));
   /* tokenized plan values (the reference's ParamLiteral tokenization,
    * TokenizationTest / SnappySession plan cache): predicate bounds and
    * aggregate coefficients load once per wave from the cached device plan
