@@ -178,7 +178,15 @@ __device__ __forceinline__ u64 mix64(u64 x) {
    * atomics per row) — per-wave LDS bins shift the work to the LDS pipe,
    * which the row phase barely uses. */
   const int wbin_mode = wbin_pre;
-  emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
+  /* fused register pass (the star-join shape): predicate-free dense-LUT
+   * group-by-attr joins skip the LDS image ENTIRELY — probe, mask and
+   * accumulate run straight from the staged registers with no barriers in
+   * the chunk loop, so s_waitcnt overlaps the next chunk's loads under the
+   * current chunk's gathers/atomics instead of a block-wide pipeline. */
+  const int fuse_mode = wbin_mode && p->jkeys && p->jlut && p->jmode == 1 &&
+                        p->npreds_d + p->npreds_i == 0;
+  if (!fuse_mode)
+    emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
   if (!lds_mode && !wbin_mode && !glob_mode)
     emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
           grouped ? nslots * (NA + 1) : 2 * na_t + 1);
@@ -226,7 +234,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emitf(o, "#pragma unroll\n  for (int a = 0; a < %d; a++) { sums[a] = 0; cnts[a] = 0; }\n", NA);
   }
 
-  if (p->jkeys && p->jlut)
+  if (p->jkeys && p->jlut && !fuse_mode)
     o += "  __shared__ int spay[CHUNK];\n";
 
   /* staged register buffers: per column, by width class */
@@ -238,6 +246,204 @@ __device__ __forceinline__ u64 mix64(u64 x) {
       emitf(o, "  int2_t st%d_0, st%d_1;\n", c, c);
     else /* I16/DICT16 */
       emitf(o, "  unsigned st%d_0, st%d_1;\n", c, c);
+  }
+
+  /* stage_load body, emitted in the prologue and inside the chunk loop */
+  auto emit_load = [&](const char *base_expr, const char *ind) {
+    for (int c = 0; c < NC; c++) {
+      int k = kinds[c];
+      if (k == SN_K_F64 || k == SN_K_I64) {
+        emitf(o, "%sst%d_0 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid];\n", ind, c, c, base_expr);
+        emitf(o, "%sst%d_1 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid + WG];\n", ind, c, c, base_expr);
+      } else if (k == SN_K_I32 || k == SN_K_F32 || k == SN_K_DICT32) {
+        emitf(o, "%sst%d_0 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid];\n", ind, c, c, base_expr);
+        emitf(o, "%sst%d_1 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid + WG];\n", ind, c, c, base_expr);
+      } else {
+        emitf(o, "%sst%d_0 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid];\n", ind, c, c, base_expr);
+        emitf(o, "%sst%d_1 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid + WG];\n", ind, c, c, base_expr);
+      }
+    }
+  };
+
+  if (fuse_mode) {
+    /* value of column c at staged-register row j (j: 0 -> st_0.x row
+     * base+2*tid, 1 -> st_0.y, 2 -> st_1.x row base+2*(tid+WG), 3 -> .y) */
+    auto rexpr = [&](int c, int j) -> std::string {
+      char buf[96];
+      const int h = j >> 1, lo = !(j & 1);
+      const char *xy = (j & 1) ? "y" : "x";
+      switch (kinds[c]) {
+        case SN_K_F64:
+          snprintf(buf, 96, "st%d_%d.%s", c, h, xy); break;
+        case SN_K_I64:
+          snprintf(buf, 96, "(double)__double_as_longlong(st%d_%d.%s)", c, h, xy); break;
+        case SN_K_I32:
+          snprintf(buf, 96, "(double)st%d_%d.%s", c, h, xy); break;
+        case SN_K_F32:
+          snprintf(buf, 96, "(double)((const float *)&st%d_%d)[%d]", c, h, j & 1); break;
+        case SN_K_DICT32:
+          snprintf(buf, 96, "(double)dm%d[st%d_%d.%s]", c, c, h, xy); break;
+        case SN_K_DICT16:
+          snprintf(buf, 96, lo ? "(double)dm%d[st%d_%d & 0xffff]"
+                               : "(double)dm%d[st%d_%d >> 16]", c, c, h); break;
+        default: /* I16 */
+          snprintf(buf, 96, lo ? "(double)(short)(st%d_%d & 0xffff)"
+                               : "(double)(short)(st%d_%d >> 16)", c, h); break;
+      }
+      return buf;
+    };
+    auto kexpr_reg = [&](int c, int j) -> std::string {
+      char buf[96];
+      const int h = j >> 1, lo = !(j & 1);
+      const char *xy = (j & 1) ? "y" : "x";
+      const int is_i64 = (p->i64_mask >> c) & 1u;
+      switch (kinds[c]) {
+        case SN_K_I32: snprintf(buf, 96, "(i64)st%d_%d.%s", c, h, xy); break;
+        case SN_K_F64: case SN_K_I64:
+          snprintf(buf, 96, "%s(st%d_%d.%s)",
+                   is_i64 || kinds[c] == SN_K_I64 ? "__double_as_longlong" : "(i64)",
+                   c, h, xy);
+          break;
+        default:
+          snprintf(buf, 96, lo ? "(i64)(short)(st%d_%d & 0xffff)"
+                               : "(i64)(short)(st%d_%d >> 16)", c, h); break;
+      }
+      return buf;
+    };
+    auto va_fused = [&](int a, const std::string (&f)[3]) -> std::string {
+      const sn_dev_agg &A = p->aggs[a];
+      if (A.nf < 1) return "1.0";
+      char t0[192], t1[192], t2[192];
+      if (ftriv(A.a0, A.m0)) snprintf(t0, 192, "%s", f[0].c_str());
+      else snprintf(t0, 192, "__builtin_fma(ag%d_m0, %s, ag%d_a0)", a, f[0].c_str(), a);
+      std::string r = t0;
+      if (A.nf >= 2) {
+        if (ftriv(A.a1, A.m1)) snprintf(t1, 192, "%s", f[1].c_str());
+        else snprintf(t1, 192, "__builtin_fma(ag%d_m1, %s, ag%d_a1)", a, f[1].c_str(), a);
+        r += " * "; r += t1;
+      }
+      if (A.nf >= 3) {
+        if (ftriv(A.a2, A.m2)) snprintf(t2, 192, "%s", f[2].c_str());
+        else snprintf(t2, 192, "__builtin_fma(ag%d_m2, %s, ag%d_a2)", a, f[2].c_str(), a);
+        r += " * "; r += t2;
+      }
+      return r;
+    };
+    /* one row's probe + accumulate */
+    auto emit_row = [&](const std::string &key, const char *rowi,
+                        const std::string (&vas)[12]) {
+      emitf(o, "      { const i64 fk = %s;\n"
+               "        const int inr = (fk >= %lldll) & (fk <= %lldll);\n"
+               "        const i64 ck = fk < %lldll ? %lldll : (fk > %lldll ? %lldll : fk);\n"
+               "        int pay = inr ? jlut[ck - %lldll] : -1;\n"
+               "        int okj = pay >= 0;\n",
+            key.c_str(),
+            (long long)p->jlut_min, (long long)p->jlut_max,
+            (long long)p->jlut_min, (long long)p->jlut_min,
+            (long long)p->jlut_max, (long long)p->jlut_max,
+            (long long)p->jlut_min);
+      if (has_del)
+        emitf(o, "        if (del) { const int gr = %s;\n"
+                 "          okj &= (int)(~(del[(u64)gr >> 6] >> (gr & 63)) & 1ull); }\n",
+              rowi);
+      emitf(o, "        if (okj) {\n"
+               "          double *rw = &wbin[tid >> 4][(pay > 0 ? pay : 0) * %d];\n"
+               "          atomicAdd(&rw[%d], 1.0);\n", NA + 1, NA);
+      for (int a = 0; a < NA; a++)
+        emitf(o, "          atomicAdd(&rw[%d], %s);\n", a, vas[a].c_str());
+      o += "        }\n      }\n";
+    };
+
+    o += R"(
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const sn_dev_tile tile = tiles[t];
+    const sn_dev_batch &b = batches[tile.batch];
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + TILE, num_rows);
+)";
+    if (has_del)
+      o += "    const GAS u64 *del = (const GAS u64 *)(u64)b.del_bm;\n";
+    for (int c = 0; c < NC; c++) {
+      emitf(o, "    const GAS char *body%d = (const GAS char *)(unsigned long long)b.cols[%d].body;\n", c, c);
+      if (kinds[c] == SN_K_DICT16 || kinds[c] == SN_K_DICT32)
+        emitf(o, "    const GAS int *dm%d = (const GAS int *)(unsigned long long)b.cols[%d].dictmap;\n", c, c);
+    }
+    o += "    for (int base = tile.row_start; base < tile_end; base += CHUNK) {\n"
+         "      const int rows = min(CHUNK, tile_end - base);\n"
+         "      if (rows == CHUNK) {\n";
+    emit_load("base", "        ");
+    for (int j = 0; j < 4; j++) {
+      char rowi[48];
+      if (j < 2) snprintf(rowi, 48, "base + 2 * tid + %d", j);
+      else snprintf(rowi, 48, "base + 2 * (tid + WG) + %d", j & 1);
+      std::string fac[3] = { "1.0", "1.0", "1.0" };
+      std::string vas[12];
+      for (int a = 0; a < NA; a++) {
+        const sn_dev_agg &A = p->aggs[a];
+        std::string f3[3];
+        f3[0] = A.nf >= 1 ? rexpr(A.c0, j) : "1.0";
+        f3[1] = A.nf >= 2 ? rexpr(A.c1, j) : "1.0";
+        f3[2] = A.nf >= 3 ? rexpr(A.c2, j) : "1.0";
+        vas[a] = va_fused(a, f3);
+      }
+      (void)fac;
+      emit_row(kexpr_reg(p->jcslot, j), rowi, vas);
+    }
+    /* scalar tail: direct global reads, same probe/accumulate */
+    o += "      } else {\n"
+         "        for (int r = tid; r < rows; r += WG) {\n"
+         "          const int gr = base + r;\n";
+    auto sexpr = [&](int c) -> std::string {
+      char buf[96];
+      switch (kinds[c]) {
+        case SN_K_F64: snprintf(buf, 96, "((const GAS double *)body%d)[gr]", c); break;
+        case SN_K_I64: snprintf(buf, 96, "(double)((const GAS i64 *)body%d)[gr]", c); break;
+        case SN_K_I32: snprintf(buf, 96, "(double)((const GAS int *)body%d)[gr]", c); break;
+        case SN_K_F32: snprintf(buf, 96, "(double)((const GAS float *)body%d)[gr]", c); break;
+        case SN_K_DICT32: snprintf(buf, 96, "(double)dm%d[((const GAS int *)body%d)[gr]]", c, c); break;
+        case SN_K_DICT16: snprintf(buf, 96, "(double)dm%d[(int)((const GAS unsigned short *)body%d)[gr]]", c, c); break;
+        default: snprintf(buf, 96, "(double)((const GAS short *)body%d)[gr]", c); break;
+      }
+      return buf;
+    };
+    {
+      char kb[96];
+      const int jc = p->jcslot;
+      const int is_i64 = (p->i64_mask >> jc) & 1u;
+      switch (kinds[jc]) {
+        case SN_K_I32: snprintf(kb, 96, "(i64)((const GAS int *)body%d)[gr]", jc); break;
+        case SN_K_F64: case SN_K_I64:
+          snprintf(kb, 96, "%s((const GAS double *)body%d)[gr]%s",
+                   is_i64 || kinds[jc] == SN_K_I64 ? "__double_as_longlong(" : "(i64)(",
+                   jc, ")");
+          break;
+        default: snprintf(kb, 96, "(i64)((const GAS short *)body%d)[gr]", jc); break;
+      }
+      std::string vas[12];
+      for (int a = 0; a < NA; a++) {
+        const sn_dev_agg &A = p->aggs[a];
+        std::string f3[3];
+        f3[0] = A.nf >= 1 ? sexpr(A.c0) : "1.0";
+        f3[1] = A.nf >= 2 ? sexpr(A.c1) : "1.0";
+        f3[2] = A.nf >= 3 ? sexpr(A.c2) : "1.0";
+        vas[a] = va_fused(a, f3);
+      }
+      emit_row(kb, "gr", vas);
+    }
+    o += "        }\n"
+         "      }\n"
+         "    }\n"
+         "  }\n";
+    int nvf = nslots * (NA + 1);
+    emitf(o, "  __syncthreads();\n"
+             "  for (int i = tid; i < %d; i += WG) {\n"
+             "    double acc = 0.0;\n"
+             "#pragma unroll\n"
+             "    for (int w = 0; w < 16; w++) acc += wbin[w][i];\n"
+             "    out[(u64)blockIdx.x * %d + i] = acc;\n"
+             "  }\n"
+             "}\n", nvf, nvf);
+    return o;
   }
 
   o += R"(
@@ -256,22 +462,6 @@ __device__ __forceinline__ u64 mix64(u64 x) {
       emitf(o, "    const GAS int *dm%d = (const GAS int *)(unsigned long long)b.cols[%d].dictmap;\n", c, c);
   }
 
-  /* stage_load macro body as a lambda-ish emitted twice (prologue + in-loop) */
-  auto emit_load = [&](const char *base_expr, const char *ind) {
-    for (int c = 0; c < NC; c++) {
-      int k = kinds[c];
-      if (k == SN_K_F64 || k == SN_K_I64) {
-        emitf(o, "%sst%d_0 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid];\n", ind, c, c, base_expr);
-        emitf(o, "%sst%d_1 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid + WG];\n", ind, c, c, base_expr);
-      } else if (k == SN_K_I32 || k == SN_K_F32 || k == SN_K_DICT32) {
-        emitf(o, "%sst%d_0 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid];\n", ind, c, c, base_expr);
-        emitf(o, "%sst%d_1 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid + WG];\n", ind, c, c, base_expr);
-      } else {
-        emitf(o, "%sst%d_0 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid];\n", ind, c, c, base_expr);
-        emitf(o, "%sst%d_1 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid + WG];\n", ind, c, c, base_expr);
-      }
-    }
-  };
   auto emit_write = [&](const char *ind) {
     for (int c = 0; c < NC; c++) {
       int k = kinds[c];
