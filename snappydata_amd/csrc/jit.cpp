@@ -237,32 +237,38 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   if (p->jkeys && p->jlut && !fuse_mode)
     o += "  __shared__ int spay[CHUNK];\n";
 
-  /* staged register buffers: per column, by width class */
+  /* staged register buffers: per column, by width class (fuse mode adds a
+   * second set, B, so the previous chunk processes under the current
+   * chunk's in-flight loads) */
   for (int c = 0; c < NC; c++) {
     int k = kinds[c];
-    if (k == SN_K_F64 || k == SN_K_I64)
-      emitf(o, "  double2_t st%d_0, st%d_1;\n", c, c);
-    else if (k == SN_K_I32 || k == SN_K_F32 || k == SN_K_DICT32)
-      emitf(o, "  int2_t st%d_0, st%d_1;\n", c, c);
-    else /* I16/DICT16 */
-      emitf(o, "  unsigned st%d_0, st%d_1;\n", c, c);
+    const char *ty = (k == SN_K_F64 || k == SN_K_I64) ? "double2_t"
+                     : (k == SN_K_I32 || k == SN_K_F32 || k == SN_K_DICT32)
+                           ? "int2_t" : "unsigned";
+    emitf(o, "  %s st%d_0, st%d_1;\n", ty, c, c);
+    if (fuse_mode) emitf(o, "  %s stB%d_0, stB%d_1;\n", ty, c, c);
   }
 
-  /* stage_load body, emitted in the prologue and inside the chunk loop */
-  auto emit_load = [&](const char *base_expr, const char *ind) {
+  /* stage_load body, emitted in the prologue and inside the chunk loop;
+   * `pre` picks the destination register set ("st" or fuse-mode "stB") */
+  auto emit_load_pre = [&](const char *base_expr, const char *ind,
+                           const char *pre) {
     for (int c = 0; c < NC; c++) {
       int k = kinds[c];
       if (k == SN_K_F64 || k == SN_K_I64) {
-        emitf(o, "%sst%d_0 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid];\n", ind, c, c, base_expr);
-        emitf(o, "%sst%d_1 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid + WG];\n", ind, c, c, base_expr);
+        emitf(o, "%s%s%d_0 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid];\n", ind, pre, c, c, base_expr);
+        emitf(o, "%s%s%d_1 = ((const GAS double2_t *)(body%d + (u64)(%s) * 8))[tid + WG];\n", ind, pre, c, c, base_expr);
       } else if (k == SN_K_I32 || k == SN_K_F32 || k == SN_K_DICT32) {
-        emitf(o, "%sst%d_0 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid];\n", ind, c, c, base_expr);
-        emitf(o, "%sst%d_1 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid + WG];\n", ind, c, c, base_expr);
+        emitf(o, "%s%s%d_0 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid];\n", ind, pre, c, c, base_expr);
+        emitf(o, "%s%s%d_1 = ((const GAS int2_t *)(body%d + (u64)(%s) * 4))[tid + WG];\n", ind, pre, c, c, base_expr);
       } else {
-        emitf(o, "%sst%d_0 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid];\n", ind, c, c, base_expr);
-        emitf(o, "%sst%d_1 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid + WG];\n", ind, c, c, base_expr);
+        emitf(o, "%s%s%d_0 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid];\n", ind, pre, c, c, base_expr);
+        emitf(o, "%s%s%d_1 = ((const GAS unsigned *)(body%d + (u64)(%s) * 2))[tid + WG];\n", ind, pre, c, c, base_expr);
       }
     }
+  };
+  auto emit_load = [&](const char *base_expr, const char *ind) {
+    emit_load_pre(base_expr, ind, "st");
   };
 
   if (fuse_mode) {
@@ -368,29 +374,50 @@ __device__ __forceinline__ u64 mix64(u64 x) {
       if (kinds[c] == SN_K_DICT16 || kinds[c] == SN_K_DICT32)
         emitf(o, "    const GAS int *dm%d = (const GAS int *)(unsigned long long)b.cols[%d].dictmap;\n", c, c);
     }
-    o += "    for (int base = tile.row_start; base < tile_end; base += CHUNK) {\n"
+    /* the per-chunk process block, generated once and spliced at the three
+     * pipeline drain points (steady state, pre-tail, epilogue) */
+    std::string proc;
+    {
+      std::string keep = std::move(o);
+      o.clear();
+      for (int j = 0; j < 4; j++) {
+        char rowi[48];
+        if (j < 2) snprintf(rowi, 48, "pbase + 2 * tid + %d", j);
+        else snprintf(rowi, 48, "pbase + 2 * (tid + WG) + %d", j & 1);
+        std::string vas[12];
+        for (int a = 0; a < NA; a++) {
+          const sn_dev_agg &A = p->aggs[a];
+          std::string f3[3];
+          f3[0] = A.nf >= 1 ? rexpr(A.c0, j) : "1.0";
+          f3[1] = A.nf >= 2 ? rexpr(A.c1, j) : "1.0";
+          f3[2] = A.nf >= 3 ? rexpr(A.c2, j) : "1.0";
+          vas[a] = va_fused(a, f3);
+        }
+        emit_row(kexpr_reg(p->jcslot, j), rowi, vas);
+      }
+      proc = std::move(o);
+      o = std::move(keep);
+    }
+    std::string bcopy;
+    for (int c = 0; c < NC; c++) {
+      char cb[96];
+      snprintf(cb, 96, "        st%d_0 = stB%d_0; st%d_1 = stB%d_1;\n",
+               c, c, c, c);
+      bcopy += cb;
+    }
+    o += "    int pbase = -1;\n"
+         "    for (int base = tile.row_start; base < tile_end; base += CHUNK) {\n"
          "      const int rows = min(CHUNK, tile_end - base);\n"
          "      if (rows == CHUNK) {\n";
-    emit_load("base", "        ");
-    for (int j = 0; j < 4; j++) {
-      char rowi[48];
-      if (j < 2) snprintf(rowi, 48, "base + 2 * tid + %d", j);
-      else snprintf(rowi, 48, "base + 2 * (tid + WG) + %d", j & 1);
-      std::string fac[3] = { "1.0", "1.0", "1.0" };
-      std::string vas[12];
-      for (int a = 0; a < NA; a++) {
-        const sn_dev_agg &A = p->aggs[a];
-        std::string f3[3];
-        f3[0] = A.nf >= 1 ? rexpr(A.c0, j) : "1.0";
-        f3[1] = A.nf >= 2 ? rexpr(A.c1, j) : "1.0";
-        f3[2] = A.nf >= 3 ? rexpr(A.c2, j) : "1.0";
-        vas[a] = va_fused(a, f3);
-      }
-      (void)fac;
-      emit_row(kexpr_reg(p->jcslot, j), rowi, vas);
-    }
-    /* scalar tail: direct global reads, same probe/accumulate */
+    /* issue this chunk's loads into B, process the PREVIOUS chunk from A
+     * underneath them, then rotate B into A */
+    emit_load_pre("base", "        ", "stB");
+    o += "        if (pbase >= 0) {\n" + proc + "        }\n" + bcopy +
+         "        pbase = base;\n";
+    /* scalar tail: drain the pipeline, then direct global reads */
     o += "      } else {\n"
+         "        if (pbase >= 0) {\n" + proc + "          pbase = -1;\n"
+         "        }\n"
          "        for (int r = tid; r < rows; r += WG) {\n"
          "          const int gr = base + r;\n";
     auto sexpr = [&](int c) -> std::string {
@@ -433,6 +460,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     o += "        }\n"
          "      }\n"
          "    }\n"
+         "    if (pbase >= 0) {\n" + proc + "    }\n"
          "  }\n";
     int nvf = nslots * (NA + 1);
     emitf(o, "  __syncthreads();\n"
