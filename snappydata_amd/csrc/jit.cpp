@@ -374,8 +374,41 @@ __device__ __forceinline__ u64 mix64(u64 x) {
       if (kinds[c] == SN_K_DICT16 || kinds[c] == SN_K_DICT32)
         emitf(o, "    const GAS int *dm%d = (const GAS int *)(unsigned long long)b.cols[%d].dictmap;\n", c, c);
     }
-    /* the per-chunk process block, generated once and spliced at the three
-     * pipeline drain points (steady state, pre-tail, epilogue) */
+    /* probe block: LUT gathers issued from the B set's key registers RIGHT
+     * AFTER the loads, so they fly a full pipeline stage before the
+     * accumulate consumes them (the round-1 measured optimum, minus LDS) */
+    std::string probe;
+    {
+      std::string keep = std::move(o);
+      o.clear();
+      for (int j = 0; j < 4; j++) {
+        /* kexpr against the B set: build from kexpr_reg then prefix-swap is
+         * fragile; emit directly with the helper on a temp name */
+        std::string kx = kexpr_reg(p->jcslot, j);
+        /* retarget st{c}_ -> stB{c}_ (the key expr references exactly one
+         * register name, emitted by kexpr_reg as "st<digit>") */
+        size_t ppos = 0;
+        while ((ppos = kx.find("st", ppos)) != std::string::npos) {
+          if (ppos + 2 < kx.size() && kx[ppos + 2] >= '0' && kx[ppos + 2] <= '9')
+            kx.insert(ppos + 2, "B");
+          ppos += 3;
+        }
+        char pd[32];
+        snprintf(pd, 32, "payB%d", j);
+        emitf(o, "        { const i64 fk = %s;\n"
+                 "          const int inr = (fk >= %lldll) & (fk <= %lldll);\n"
+                 "          const i64 ck = fk < %lldll ? %lldll : (fk > %lldll ? %lldll : fk);\n"
+                 "          %s = inr ? jlut[ck - %lldll] : -1; }\n",
+              kx.c_str(),
+              (long long)p->jlut_min, (long long)p->jlut_max,
+              (long long)p->jlut_min, (long long)p->jlut_min,
+              (long long)p->jlut_max, (long long)p->jlut_max,
+              pd, (long long)p->jlut_min);
+      }
+      probe = std::move(o);
+      o = std::move(keep);
+    }
+    /* accumulate block: consumes the A set + carried payA payloads */
     std::string proc;
     {
       std::string keep = std::move(o);
@@ -393,7 +426,18 @@ __device__ __forceinline__ u64 mix64(u64 x) {
           f3[2] = A.nf >= 3 ? rexpr(A.c2, j) : "1.0";
           vas[a] = va_fused(a, f3);
         }
-        emit_row(kexpr_reg(p->jcslot, j), rowi, vas);
+        emitf(o, "      { const int pay = payA%d;\n"
+                 "        int okj = pay >= 0;\n", j);
+        if (has_del)
+          emitf(o, "        if (del) { const int gr = %s;\n"
+                   "          okj &= (int)(~(del[(u64)gr >> 6] >> (gr & 63)) & 1ull); }\n",
+                rowi);
+        emitf(o, "        if (okj) {\n"
+                 "          double *rw = &wbin[tid >> 4][(pay > 0 ? pay : 0) * %d];\n"
+                 "          atomicAdd(&rw[%d], 1.0);\n", NA + 1, NA);
+        for (int a = 0; a < NA; a++)
+          emitf(o, "          atomicAdd(&rw[%d], %s);\n", a, vas[a].c_str());
+        o += "        }\n      }\n";
       }
       proc = std::move(o);
       o = std::move(keep);
@@ -405,14 +449,19 @@ __device__ __forceinline__ u64 mix64(u64 x) {
                c, c, c, c);
       bcopy += cb;
     }
+    bcopy += "        payA0 = payB0; payA1 = payB1;"
+             " payA2 = payB2; payA3 = payB3;\n";
+    o += "  int payA0 = -1, payA1 = -1, payA2 = -1, payA3 = -1;\n"
+         "  int payB0, payB1, payB2, payB3;\n";
     o += "    int pbase = -1;\n"
          "    for (int base = tile.row_start; base < tile_end; base += CHUNK) {\n"
          "      const int rows = min(CHUNK, tile_end - base);\n"
          "      if (rows == CHUNK) {\n";
-    /* issue this chunk's loads into B, process the PREVIOUS chunk from A
-     * underneath them, then rotate B into A */
+    /* issue this chunk's loads into B, accumulate the PREVIOUS chunk from
+     * A (probes already in flight since last iteration), then probe B's
+     * keys (gathers fly through the next load+accumulate) and rotate */
     emit_load_pre("base", "        ", "stB");
-    o += "        if (pbase >= 0) {\n" + proc + "        }\n" + bcopy +
+    o += "        if (pbase >= 0) {\n" + proc + "        }\n" + probe + bcopy +
          "        pbase = base;\n";
     /* scalar tail: drain the pipeline, then direct global reads */
     o += "      } else {\n"
