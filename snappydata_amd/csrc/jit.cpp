@@ -26,6 +26,7 @@
 #include <cstdarg>
 #include <cstdio>
 #include <cstring>
+#include <cstdlib>
 #include <map>
 #include <mutex>
 #include <string>
@@ -181,10 +182,19 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   /* fused register pass (the star-join shape): predicate-free dense-LUT
    * group-by-attr joins skip the LDS image ENTIRELY — probe, mask and
    * accumulate run straight from the staged registers with no barriers in
-   * the chunk loop, so s_waitcnt overlaps the next chunk's loads under the
-   * current chunk's gathers/atomics instead of a block-wide pipeline. */
-  const int fuse_mode = wbin_mode && p->jkeys && p->jlut && p->jmode == 1 &&
-                        p->npreds_d + p->npreds_i == 0;
+   * the chunk loop.  BUILT AND MEASURED SLOWER than the staged-LDS scheme
+   * on star-join SF10 (2.39 TB/s staged-LDS vs 1.84 single-buffered /
+   * 1.72 value-pipelined / 1.76 probe-pipelined fused): without the LDS
+   * stage the per-chunk dependency chain load->probe->atomic exposes
+   * whichever leg is not double-buffered, and the block-wide barrier the
+   * fusion removes was evidently not the bottleneck.  Kept behind
+   * SN_JIT_FUSE=1 for re-evaluation; default off. */
+  static const int fuse_env = [] {
+    const char *v = getenv("SN_JIT_FUSE");
+    return v && v[0] == '1';
+  }();
+  const int fuse_mode = fuse_env && wbin_mode && p->jkeys && p->jlut &&
+                        p->jmode == 1 && p->npreds_d + p->npreds_i == 0;
   if (!fuse_mode)
     emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
   if (!lds_mode && !wbin_mode && !glob_mode)
