@@ -131,6 +131,11 @@ __device__ __forceinline__ u64 mix64(u64 x) {
    * (the LDS image + accumulator array leave no room for a second group). */
   const int glob_mode = (grouped && nslots > 1024) || sparse_mode;
   const int lds_mode = grouped && nslots > 8 && !glob_mode;
+  /* wbin eligibility: NA<=2 only.  Widening to Q1's 6-slot/5-agg shape was
+   * MEASURED SLOWER (3.51 vs 5.13 TB/s on Q1 SF10): 6 LDS f64 atomics/row
+   * across 16-lane bins conflict harder than the 36 slot-predicated fmas
+   * cost, so the register accumulators stay the measured optimum for
+   * multi-aggregate small-slot shapes. */
   const int wbin_pre = grouped && !lds_mode && !glob_mode && NA <= 2 &&
                        nslots * (NA + 1) >= 12;
   emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, %d)\n"
