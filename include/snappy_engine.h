@@ -270,6 +270,22 @@ int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
                    const int64_t *keys, const char *attr_payload,
                    const int32_t *attr_lens);
 
+/* ---- partitioned-partitioned (colocated) join build ----
+ * Populate an EMPTY dimension handle from a resident COLUMN TABLE's rows,
+ * built on device: the HashJoinExec per-task build of the probe map from
+ * the build side (HashJoinExec.scala:285-520) with HashedObjectCache reuse
+ * (:449-470).  The reference's partitioned-partitioned joins run
+ * bucket-LOCALLY when tables are colocated on the join key (GemFire
+ * colocation) — both sides sharded identically means no exchange step, so
+ * each shard builds from ITS build-side rows and probes ITS fact rows.
+ * key_col: int32/int64 column with UNIQUE values per shard (duplicate keys
+ * with conflicting payloads fail with SN_ERR_BADARG); attr_col: optional
+ * dictionary-string column whose values become the SN_JOIN_GROUP
+ * attributes (-1 = semi-join only).  Rebuild after the build table
+ * changes; deleted/patched rows are honored at build time. */
+int32_t sn_dim_from_table(sn_engine *e, int32_t dim, int32_t table,
+                          int32_t key_col, int32_t attr_col);
+
 /* attach the CURRENT (cumulative) mutation state to an existing batch —
  * the UPDATE/DELETE seam: delete_mask and the per-column delta pairs
  * replace any previous state for that batch; stats (if given) replace the

@@ -249,6 +249,8 @@ struct Dim {
   const int32_t *dev_lut = nullptr;
   int64_t lut_min = 0, lut_max = -1;
   int32_t attr_maxlen = 0;
+  bool from_table = false;   /* device-built from a column table (f3):
+                                host keys unknown, sn_dim_put forbidden */
   /* guards keys/attrs/device-table pointers: sn_dim_put may run while a
    * submit builds/reads the device table (the reference's replicated
    * region puts are similarly concurrent with task-side get()s) */
@@ -369,6 +371,8 @@ extern "C" int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
     return fail(SN_ERR_BADARG, "bad dim args");
   Dim *d = e->dims[dim].get();
   std::lock_guard<std::mutex> gd(d->mu);
+  if (d->from_table)
+    return fail(SN_ERR_BADARG, "dimension was built from a column table");
   int64_t off = 0;
   for (int64_t i = 0; i < nkeys; i++) {
     d->keys.push_back(keys[i]);
@@ -394,12 +398,208 @@ extern "C" int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
 }
 
 static void *up(sn_engine *e, const void *host, size_t n);
+static Table *get_table(sn_engine *e, int32_t t);
 
 static inline uint64_t mix64h(uint64_t x) {
   x += 0x9E3779B97f4A7C15ull;
   x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
   x = (x ^ (x >> 27)) * 0x94D049BB133111EBull;
   return x ^ (x >> 31);
+}
+
+/* ---- partitioned-partitioned (colocated) join build: populate a Dim from
+ * a resident column table, built ON DEVICE (k_join_build) — the
+ * HashJoinExec per-task ObjectHashSet build + HashedObjectCache reuse
+ * (HashJoinExec.scala:285-520, :449-470).  Colocated tables join
+ * bucket-locally, so each shard builds from its own build-side rows. ---- */
+extern "C" int32_t sn_dim_from_table(sn_engine *e, int32_t dim, int32_t table,
+                                     int32_t key_col, int32_t attr_col) {
+  if (!e || dim < 0 || dim >= (int32_t)e->dims.size())
+    return fail(SN_ERR_BADARG, "bad dim handle");
+  Table *t = get_table(e, table);
+  if (!t) return fail(SN_ERR_BADARG, "bad table");
+  if (!e->has_gpu)
+    return fail(SN_ERR_NOGPU, "join build runs on device");
+  Dim *d = e->dims[dim].get();
+  std::lock_guard<std::mutex> gd(d->mu);
+  if (d->dev_keys || !d->keys.empty())
+    return fail(SN_ERR_BADARG, "dimension is not empty");
+  std::lock_guard<std::mutex> g(t->mu);
+  const int nc = (int)t->schema.size();
+  if (key_col < 0 || key_col >= nc) return fail(SN_ERR_BADARG, "bad key col");
+  sn_type_t kdt = t->schema[key_col].dtype;
+  if (kdt != SN_TYPE_INT32 && kdt != SN_TYPE_INT64 && kdt != SN_TYPE_INT16)
+    return fail(SN_ERR_UNSUPPORTED, "join build key must be int16/int32/int64");
+  const bool has_attr = attr_col >= 0;
+  if (has_attr) {
+    if (attr_col >= nc || t->schema[attr_col].dtype != SN_TYPE_STRING)
+      return fail(SN_ERR_UNSUPPORTED, "attr col must be a string column");
+    if (t->gdict_maxlen[attr_col] > SN_KEY_MAX - 1)
+      return fail(SN_ERR_UNSUPPORTED, "attr exceeds the group-key limit");
+  }
+  if (t->total_rows > (1ll << 23))
+    return fail(SN_ERR_UNSUPPORTED,
+                "build side exceeds 2^23 rows (HashJoinSize-class bound)");
+
+  sn_dev_plan dp;
+  memset(&dp, 0, sizeof(dp));
+  dp.nused = has_attr ? 2 : 1;
+  dp.i64_mask = kdt == SN_TYPE_INT64 ? 1u : 0u;
+  dp.naggs = 1;
+
+  std::vector<sn_dev_batch> hb;
+  std::vector<sn_dev_tile> ht;
+  const int used[2] = { key_col, attr_col };
+  for (auto &b : t->batches) {
+    sn_dev_batch db;
+    memset(&db, 0, sizeof(db));
+    db.num_rows = b.num_rows;
+    db.del_bm = b.del_bm_dev;
+    bool clean = !b.has_deletes;
+    for (int ui = 0; ui < dp.nused; ui++) {
+      int c = used[ui];
+      const ColMeta &m = b.cols[c];
+      sn_dev_col &dc = db.cols[ui];
+      memset(&dc, 0, sizeof(dc));
+      sn_type_t dt = t->schema[c].dtype;
+      dc.body = (const uint8_t *)b.col_dev[c] + m.body_off;
+      dc.has_nulls = m.num_null_words > 0;
+      if (dc.has_nulls) {
+        dc.nullw = (const uint64_t *)((const uint8_t *)b.col_dev[c] + m.null_off);
+        dc.nullpfx = b.nullpfx_dev[c];
+        clean = false;
+      }
+      switch (m.type_id) {
+        case SN_ENC_UNCOMPRESSED:
+          switch (dt) {
+            case SN_TYPE_INT32: dc.kind = SN_K_I32; break;
+            case SN_TYPE_INT64: dc.kind = SN_K_I64; break;
+            case SN_TYPE_INT16: dc.kind = SN_K_I16; break;
+            default:
+              return fail(SN_ERR_UNSUPPORTED, "build col %d dtype", c);
+          }
+          break;
+        case SN_ENC_RUNLENGTH:
+          if (b.rle_n[c] <= 0 && b.num_rows > 0)
+            return fail(SN_ERR_UNSUPPORTED, "RLE col %d without run aux", c);
+          dc.kind = dt == SN_TYPE_INT64 ? SN_K_RLE_I64 : SN_K_RLE;
+          dc.rle_ends = b.rle_ends_dev[c];
+          dc.rle_vals = b.rle_vals_dev[c];
+          dc.rle_n = b.rle_n[c];
+          break;
+        case SN_ENC_DICTIONARY:
+        case SN_ENC_BIG_DICTIONARY: {
+          if (dt != SN_TYPE_STRING)
+            return fail(SN_ERR_UNSUPPORTED, "unexpected dict col %d", c);
+          dc.kind = m.type_id == SN_ENC_DICTIONARY ? SN_K_DICT16 : SN_K_DICT32;
+          /* payload gid = table-global dictionary id (mul 1, null slot 0) */
+          auto &mcache = b.dictmap_cache[c];
+          auto mit = mcache.find({ 1, 0 });
+          if (mit != mcache.end()) {
+            dc.dictmap = mit->second;
+          } else {
+            std::vector<int32_t> map(m.local2global.size() + 1);
+            for (size_t i = 0; i < m.local2global.size(); i++)
+              map[i] = m.local2global[i];
+            map[m.local2global.size()] = 0;
+            dc.dictmap = (const int32_t *)up(e, map.data(), map.size() * 4);
+            if (!dc.dictmap) return fail(SN_ERR_NOMEM, "dictmap upload");
+            mcache.emplace(std::make_pair(1, 0), dc.dictmap);
+          }
+          dc.null_gid = 0;
+          break;
+        }
+        default:
+          return fail(SN_ERR_UNSUPPORTED, "encoding %d on build path", m.type_id);
+      }
+      if (b.patch_dev[c].n > 0) {
+        clean = false;
+        const Batch::PatchDev &pd = b.patch_dev[c];
+        dc.patch_bm = pd.bm;
+        dc.patch_pos = pd.pos;
+        dc.patch_nullbm = pd.nullbm;
+        dc.patch_n = pd.n;
+        dc.patch_val = pd.val;   /* string patches carry global ids (mul 1) */
+      }
+    }
+    db.clean = clean ? 1 : 0;
+    int32_t bi = (int32_t)hb.size();
+    hb.push_back(db);
+    for (int32_t r = 0; r < b.num_rows; r += SN_TILE_ROWS)
+      ht.push_back({ bi, r });
+  }
+
+  int cap_log2 = 10;
+  while ((1ll << cap_log2) < 2 * t->total_rows && cap_log2 < 25) cap_log2++;
+  const size_t cap = 1ull << cap_log2;
+  long long *hk = (long long *)e->arena.alloc(cap * 8);
+  int32_t *hp = (int32_t *)e->arena.alloc(cap * 4);
+  int32_t *flags = (int32_t *)e->arena.alloc(64);
+  void *dp_dev = e->arena.alloc(sizeof(dp));
+  void *db_dev = hb.empty() ? nullptr : e->arena.alloc(hb.size() * sizeof(sn_dev_batch));
+  void *tl_dev = ht.empty() ? nullptr : e->arena.alloc(ht.size() * sizeof(sn_dev_tile));
+  if (!hk || !hp || !flags || !dp_dev || (!hb.empty() && (!db_dev || !tl_dev)))
+    return fail(SN_ERR_NOMEM, "join build alloc");
+  if (hipMemsetAsync(flags, 0, 16, e->stream) != hipSuccess ||
+      hipMemcpy(dp_dev, &dp, sizeof(dp), hipMemcpyHostToDevice) != hipSuccess)
+    return fail(SN_ERR_GENERIC, "join build upload");
+  if (!hb.empty() &&
+      (hipMemcpy(db_dev, hb.data(), hb.size() * sizeof(sn_dev_batch),
+                 hipMemcpyHostToDevice) != hipSuccess ||
+       hipMemcpy(tl_dev, ht.data(), ht.size() * sizeof(sn_dev_tile),
+                 hipMemcpyHostToDevice) != hipSuccess))
+    return fail(SN_ERR_GENERIC, "join build upload");
+  int rc = sn_launch_join_build(&dp, (const sn_dev_plan *)dp_dev,
+                                (const sn_dev_batch *)db_dev,
+                                (const sn_dev_tile *)tl_dev,
+                                (int32_t)ht.size(), hk, hp, cap_log2,
+                                has_attr ? 1 : 0, flags, e->stream);
+  if (rc != 0)
+    return fail(SN_ERR_GENERIC, "join build: %s",
+                hipGetErrorString((hipError_t)rc));
+  if (hipStreamSynchronize(e->stream) != hipSuccess)
+    return fail(SN_ERR_GENERIC, "join build sync");
+  int32_t fl[2] = { 0, 0 };
+  (void)hipMemcpy(fl, flags, 8, hipMemcpyDeviceToHost);
+  if (fl[0])
+    return fail(SN_ERR_BADARG,
+                "duplicate build keys with conflicting payloads (or a "
+                "sentinel-valued key)");
+  if (fl[1]) return fail(SN_ERR_OVERFLOW, "join build table full");
+
+  d->dev_keys = (const int64_t *)hk;
+  d->dev_payload = hp;
+  d->cap_log2 = cap_log2;
+  d->from_table = true;
+  if (has_attr) {
+    d->attr_dict = t->gdict[attr_col];
+    d->attr_maxlen = t->gdict_maxlen[attr_col];
+  } else {
+    d->attr_dict.assign(1, std::string());
+  }
+  /* dense LUT when the key span (from stats bounds) is small */
+  bool have_span = !t->batches.empty();
+  int64_t mn = INT64_MAX, mx = INT64_MIN;
+  for (auto &b : t->batches) {
+    if (!b.stats_valid || b.bounds_null.size() <= (size_t)key_col ||
+        b.bounds_null[key_col]) { have_span = false; break; }
+    mn = std::min(mn, b.lo_i[key_col]);
+    mx = std::max(mx, b.hi_i[key_col]);
+  }
+  if (have_span && mx >= mn && mx - mn + 1 <= (1ll << 24)) {
+    int64_t span = mx - mn + 1;
+    int32_t *lut = (int32_t *)e->arena.alloc((size_t)span * 4);
+    if (lut &&
+        hipMemsetAsync(lut, 0xff, (size_t)span * 4, e->stream) == hipSuccess &&
+        sn_launch_hash_to_lut(hk, hp, (long long)cap, lut, mn,
+                              e->stream) == 0 &&
+        hipStreamSynchronize(e->stream) == hipSuccess) {
+      d->dev_lut = lut;
+      d->lut_min = mn;
+      d->lut_max = mx;
+    }
+  }
+  return SN_OK;
 }
 
 /* build (or reuse) the device open-address table: the broadcast into HBM */
@@ -1377,7 +1577,9 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       fail(SN_ERR_BADARG, "unknown dimension %d", plan->join_dim); return nullptr;
     }
     jd = e->dims[plan->join_dim].get();
-    if (jd->keys.empty()) { fail(SN_ERR_BADARG, "empty dimension"); return nullptr; }
+    if (jd->keys.empty() && !jd->dev_keys) {
+      fail(SN_ERR_BADARG, "empty dimension"); return nullptr;
+    }
     sn_type_t kt = t->schema[plan->join_fact_col].dtype;
     if (kt != SN_TYPE_INT32 && kt != SN_TYPE_INT64) {
       fail(SN_ERR_UNSUPPORTED, "join key must be int32/int64"); return nullptr;

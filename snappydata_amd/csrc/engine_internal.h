@@ -221,6 +221,18 @@ int sn_launch_hash_compact(const long long *hk, const double *hacc,
                            int cap, int naggs1, long long *okeys,
                            double *orows, int *counter, void *stream);
 
+/* device-side join-table build from a column table (colocated
+ * partitioned-partitioned join): fills an open-address (key, payload)
+ * table in the probe_sweep layout; optional LUT densification */
+int sn_launch_join_build(const sn_dev_plan *plan, const sn_dev_plan *dev_plan,
+                         const sn_dev_batch *dev_batches,
+                         const sn_dev_tile *dev_tiles, int32_t ntiles,
+                         long long *hk, int32_t *hp, int cap_log2,
+                         int has_attr, int32_t *flags, void *stream);
+int sn_launch_hash_to_lut(const long long *hk, const int32_t *hp,
+                          long long cap, int32_t *lut, long long lmin,
+                          void *stream);
+
 /* launches only the partial-fold (k_reduce): scratch[nblocks][nv] -> out.
  * Used by the JIT path, whose scan kernel writes the same scratch rows.
  * plan_dev (nullable) supplies per-agg ops for MIN/MAX folding. */

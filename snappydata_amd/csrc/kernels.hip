@@ -1450,6 +1450,157 @@ void k_grouped_hash(sn_dev_plan plan,
   }
 }
 
+/* ---- device-side join-table build (the partitioned-partitioned /
+ * colocated join's HashedObjectCache analogue, HashJoinExec.scala:285-520,
+ * ObjectHashSet.scala:107-135): scan the BUILD-side column table's batches
+ * and insert (key -> payload gid) into an open-address table with the SAME
+ * layout probe_sweep consumes (sentinel LLONG_MIN).  Colocated partitioned
+ * tables join bucket-locally in the reference (GemFire colocation), so no
+ * exchange step exists on this path. ---- */
+__global__ void k_fill_i64(long long *dst, long long n, long long v) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) ((GAS long long *)(uintptr_t)dst)[i] = v;
+}
+
+/* flags: [0] = duplicate key with conflicting payload / key == sentinel,
+ * [1] = table full */
+__global__ __launch_bounds__(WG, 2)
+void k_join_build(sn_dev_plan plan,
+                  const sn_dev_plan *__restrict__ plan_g,
+                  const sn_dev_batch *__restrict__ batches,
+                  const sn_dev_tile *__restrict__ tiles, int ntiles,
+                  long long *__restrict__ hk_, int32_t *__restrict__ hp_,
+                  int cap_log2, int has_attr, int32_t *__restrict__ flags_) {
+  const int tid = threadIdx.x;
+  const int nused = plan.nused;
+  GAS long long *hk = (GAS long long *)(uintptr_t)hk_;
+  GAS int32_t *hp = (GAS int32_t *)(uintptr_t)hp_;
+  GAS int32_t *flags = (GAS int32_t *)(uintptr_t)flags_;
+  const unsigned mask = (1u << cap_log2) - 1;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  double *sval = (double *)smem;
+  uint64_t *svalid = (uint64_t *)(smem + (size_t)nused * CHUNK * 8);
+  uint64_t *sdead = svalid + (size_t)nused * (CHUNK / 64);
+  uint64_t *salive = sdead + CHUNK / 64;
+
+  for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
+    const sn_dev_tile tile = tiles[t];
+    const sn_dev_batch &b = batches[tile.batch];
+    const int num_rows = b.num_rows;
+    const int tile_end = min(tile.row_start + SN_TILE_ROWS, num_rows);
+    const int clean = b.clean;
+
+    for (int base = tile.row_start; base < tile_end; base += CHUNK) {
+      const int rows = min(CHUNK, tile_end - base);
+      convert_chunk(b, nused, base, rows, num_rows, sval, svalid, sdead);
+      __syncthreads();
+      alive_init(salive, sdead, rows, clean);
+      if (!clean) {
+        /* null build keys never join (SQL equality) */
+        const int tid64 = tid & 63;
+        (void)tid64;
+#pragma unroll
+        for (int k = 0; k < CHUNK / WG; k++) {
+          const int r = tid + k * WG;
+          uint64_t w = svalid[(size_t)0 * (CHUNK / 64) + (r >> 6)];
+          if ((tid & 63) == 0) salive[r >> 6] &= w;
+        }
+      }
+#pragma unroll 2
+      for (int k = 0; k < CHUNK / WG; k++) {
+        const int r = tid + k * WG;
+        const uint64_t w = salive[r >> 6];
+        if (w == 0) continue;
+        if (!((w >> (tid & 63)) & 1ull)) continue;
+        const double kx = sval[r];
+        const long long key = ((plan.i64_mask >> 0) & 1u)
+                                  ? __double_as_longlong(kx)
+                                  : (long long)kx;
+        const int32_t pay = has_attr ? (int32_t)sval[(size_t)1 * CHUNK + r] : 0;
+        if (key == LLONG_MIN) { atomicOr((int *)&flags[0], 1); continue; }
+        unsigned h = (unsigned)mix64((unsigned long long)key) & mask;
+        int done = 0;
+        for (unsigned it = 0; it <= mask && !done; ++it) {
+          const long long k0 = hk[h];
+          if (k0 == key) {
+            /* duplicate build key: allowed iff payload agrees (the dim
+             * contract requires unique keys; equal rows are idempotent) */
+            if (hp[h] != pay) atomicOr((int *)&flags[0], 1);
+            done = 1;
+            break;
+          }
+          if (k0 == LLONG_MIN) {
+            const long long old = (long long)atomicCAS(
+                (unsigned long long *)&hk[h], (unsigned long long)LLONG_MIN,
+                (unsigned long long)key);
+            if (old == LLONG_MIN || old == key) {
+              if (old == LLONG_MIN) hp[h] = pay;
+              else if (hp[h] != pay) atomicOr((int *)&flags[0], 1);
+              done = 1;
+              break;
+            }
+          }
+          h = (h + 1) & mask;
+        }
+        if (!done) atomicOr((int *)&flags[1], 1);
+      }
+      __syncthreads();
+    }
+  }
+}
+
+/* densify the open-address table into a payload LUT over [lmin, lmax] */
+__global__ void k_hash_to_lut(const long long *__restrict__ hk_,
+                              const int32_t *__restrict__ hp_,
+                              long long cap, int32_t *__restrict__ lut_,
+                              long long lmin) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= cap) return;
+  const long long key = ((const GAS long long *)(uintptr_t)hk_)[i];
+  if (key == LLONG_MIN) return;
+  ((GAS int32_t *)(uintptr_t)lut_)[key - lmin] =
+      ((const GAS int32_t *)(uintptr_t)hp_)[i];
+}
+
+extern "C" int sn_launch_join_build(const sn_dev_plan *plan,
+                                    const sn_dev_plan *dev_plan,
+                                    const sn_dev_batch *dev_batches,
+                                    const sn_dev_tile *dev_tiles,
+                                    int32_t ntiles, long long *hk,
+                                    int32_t *hp, int cap_log2, int has_attr,
+                                    int32_t *flags, void *stream) {
+  hipStream_t s = (hipStream_t)stream;
+  {
+    long long cap = 1ll << cap_log2;
+    hipLaunchKernelGGL(k_fill_i64, dim3((unsigned)((cap + 255) / 256)),
+                       dim3(256), 0, s, hk, cap, LLONG_MIN);
+  }
+  int grid;
+  if (ntiles <= 0) return (int)hipGetLastError();
+  if (ntiles <= SN_GRID_CAP) grid = ntiles;
+  else {
+    int rounds = (ntiles + SN_GRID_CAP - 1) / SN_GRID_CAP;
+    grid = (ntiles + rounds - 1) / rounds;
+  }
+  size_t lds = (size_t)plan->nused * CHUNK * 8 +
+               (size_t)plan->nused * (CHUNK / 64) * 8 +
+               2 * (CHUNK / 64) * 8 + 64;
+  if (lds > 160 * 1024) return (int)hipErrorInvalidValue;
+  hipLaunchKernelGGL(k_join_build, dim3(grid), dim3(WG), lds, s,
+                     *plan, dev_plan, dev_batches, dev_tiles, ntiles,
+                     hk, hp, cap_log2, has_attr, flags);
+  return (int)hipGetLastError();
+}
+
+extern "C" int sn_launch_hash_to_lut(const long long *hk, const int32_t *hp,
+                                     long long cap, int32_t *lut,
+                                     long long lmin, void *stream) {
+  hipLaunchKernelGGL(k_hash_to_lut, dim3((unsigned)((cap + 255) / 256)),
+                     dim3(256), 0, (hipStream_t)stream, hk, hp, cap, lut, lmin);
+  return (int)hipGetLastError();
+}
+
 /* compact the used hash-table rows into dense (key, accumulator-row) pairs
  * so the host reads back only the live groups, not the whole table */
 __global__ void k_hash_compact(const long long *__restrict__ hk_,

@@ -123,6 +123,9 @@ def lib():
         L.sn_dim_put.argtypes = [C.c_void_p, C.c_int32, C.c_int64,
                                  C.POINTER(C.c_int64), C.c_char_p,
                                  C.POINTER(C.c_int32)]
+        L.sn_dim_from_table.restype = C.c_int32
+        L.sn_dim_from_table.argtypes = [C.c_void_p, C.c_int32, C.c_int32,
+                                        C.c_int32, C.c_int32]
         L.sn_encode_column.restype = C.c_int64
         L.sn_encode_column.argtypes = [C.c_int32, C.c_void_p, C.POINTER(C.c_int32),
                                        C.POINTER(C.c_uint8), C.c_int32,
@@ -455,6 +458,13 @@ class Engine:
             keys.ctypes.data_as(C.POINTER(C.c_int64)), payload,
             lens.ctypes.data_as(C.POINTER(C.c_int32)) if lens is not None else None),
             "dim_put")
+
+    def dim_from_table(self, dim, table, key_col, attr_col=-1):
+        """Populate an empty dimension from a resident column table (the
+        colocated partitioned-partitioned join build, device-side —
+        HashJoinExec per-task build + HashedObjectCache reuse)."""
+        _check(lib().sn_dim_from_table(self._h, dim, table, key_col, attr_col),
+               "dim_from_table")
 
     def query(self, plan):
         h = lib().sn_query_submit(self._h, C.byref(plan))
