@@ -277,6 +277,8 @@ struct sn_engine {
   int32_t *hws_flags = nullptr;   /* [0] overflow, [1] compact counter */
   int hws_cap_log2 = 0;
   size_t hws_acc_bytes = 0;
+  int32_t *lz4_err_dev = nullptr; /* device error word for the LZ4 decode */
+  std::mutex lz4_mu;              /* serializes decode launch+sync+readback */
   /* per-engine cache of query-compiled kernels (jit.cpp) */
   void *jit = nullptr;
   /* steady-state submit caches: device plan copies by content hash, and a
@@ -1219,17 +1221,49 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
 
     /* upload blob, placed so the BODY is 16-byte aligned (the scan kernel
      * issues 16 B/lane vector loads on the body; the 8-byte blob header
-     * would otherwise leave it 8-aligned) */
+     * would otherwise leave it 8-aligned).
+     * f1 compressed-upload: LZ4-wrapped blobs ship their COMPRESSED bytes
+     * over PCIe (the host decompressed copy serves only parse/validation)
+     * and decode wave-cooperatively on device into the blob slot —
+     * byte-identical output, ~2-4x less bus traffic per blob. */
     {
       int64_t pad = (16 - (b.cols[c].body_off & 15)) & 15;
       char *base = (char *)e->arena.alloc((size_t)len + 16);
       if (!base) return fail(SN_ERR_NOMEM, "HBM upload failed");
       char *dst = base + pad;
-      if (e->arena.device >= 0) {
-        if (hipMemcpy(dst, blob, (size_t)len, hipMemcpyHostToDevice) != hipSuccess)
-          return fail(SN_ERR_NOMEM, "HBM upload failed");
-      } else {
-        memcpy(dst, blob, (size_t)len);
+      bool dev_decoded = false;
+      if (e->arena.device >= 0 && dec == 1 &&
+          rd_i32((const uint8_t *)columns[c].data) == -1 /* LZ4 */) {
+        const int64_t clen = columns[c].len - 8;
+        /* the engine stream + error word are shared: one decode at a time
+         * (concurrent puts overlap their host work; device decodes queue) */
+        std::lock_guard<std::mutex> glz(e->lz4_mu);
+        void *cdev = e->arena.alloc((size_t)clen);
+        if (!e->lz4_err_dev)
+          e->lz4_err_dev = (int32_t *)e->arena.alloc(64);
+        if (cdev && e->lz4_err_dev &&
+            hipMemcpy(cdev, (const uint8_t *)columns[c].data + 8,
+                      (size_t)clen, hipMemcpyHostToDevice) == hipSuccess &&
+            hipMemsetAsync(e->lz4_err_dev, 0, 4, e->stream) == hipSuccess &&
+            sn_launch_lz4_decompress(cdev, clen, dst, len, e->lz4_err_dev,
+                                     e->stream) == 0 &&
+            hipStreamSynchronize(e->stream) == hipSuccess) {
+          int32_t derr = -1;
+          (void)hipMemcpy(&derr, e->lz4_err_dev, 4, hipMemcpyDeviceToHost);
+          if (derr != 0)
+            return fail(SN_ERR_BADFORMAT,
+                        "device LZ4 decode failed (%d) col %d", derr, c);
+          dev_decoded = true;
+        }
+        if (cdev) e->arena.release(cdev, (size_t)clen);
+      }
+      if (!dev_decoded) {
+        if (e->arena.device >= 0) {
+          if (hipMemcpy(dst, blob, (size_t)len, hipMemcpyHostToDevice) != hipSuccess)
+            return fail(SN_ERR_NOMEM, "HBM upload failed");
+        } else {
+          memcpy(dst, blob, (size_t)len);
+        }
       }
       b.col_dev[c] = dst;
     }

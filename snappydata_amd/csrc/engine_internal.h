@@ -233,6 +233,11 @@ int sn_launch_hash_to_lut(const long long *hk, const int32_t *hp,
                           long long cap, int32_t *lut, long long lmin,
                           void *stream);
 
+/* wave-cooperative raw LZ4 block decode (f1 compressed-upload ingest);
+ * err[0]: 0 ok, 1 malformed, 2 length mismatch */
+int sn_launch_lz4_decompress(const void *src, long long slen, void *dst,
+                             long long dlen, int32_t *err, void *stream);
+
 /* launches only the partial-fold (k_reduce): scratch[nblocks][nv] -> out.
  * Used by the JIT path, whose scan kernel writes the same scratch rows.
  * plan_dev (nullable) supplies per-agg ops for MIN/MAX folding. */

@@ -1661,6 +1661,95 @@ extern "C" int sn_launch_hash_compact(const long long *hk, const double *hacc,
   return (int)hipGetLastError();
 }
 
+/* ---- wave-cooperative LZ4 block decode (f1: compressed-upload ingest) ----
+ * The reference wraps each column blob as ONE raw LZ4 block
+ * (CompressionUtils.scala:132-160): the token chain is strictly
+ * sequential, so lane 0 parses while all 64 lanes of the wave copy the
+ * literal/match runs — wave lockstep makes the copy loop barrier-free.
+ * Overlapping matches (offset < length) replicate periodically:
+ * dst[i] = dst[start - off + i % off], every read landing before the
+ * segment start.  One wave per blob; concurrency comes from decoding
+ * many blobs in flight (one launch per put on the engine stream). */
+__global__ __launch_bounds__(64, 8)
+void k_lz4_decompress(const uint8_t *__restrict__ src_, long long slen,
+                      uint8_t *__restrict__ dst_, long long dlen,
+                      int32_t *__restrict__ err_) {
+  const GAS uint8_t *src = (const GAS uint8_t *)(uintptr_t)src_;
+  GAS uint8_t *dst = (GAS uint8_t *)(uintptr_t)dst_;
+  GAS int32_t *err = (GAS int32_t *)(uintptr_t)err_;
+  const int lane = threadIdx.x;
+  long long ip = 0, op = 0;
+  while (true) {
+    /* lane 0 parses one sequence; broadcast (lit_src, lit_len, m_off,
+     * m_len, next_ip); next_ip -2 = clean end, -3 = malformed */
+    long long v0 = 0, v1 = 0, v2 = 0, v3 = 0, v4 = -3;
+    if (lane == 0) {
+      if (ip >= slen) {
+        v4 = -2;
+      } else {
+        const unsigned tok = src[ip++];
+        long long ll = tok >> 4;
+        if (ll == 15) {
+          unsigned b;
+          do { b = (ip < slen) ? src[ip++] : 0; ll += b; }
+          while (b == 255 && ip < slen);
+        }
+        v0 = ip;                                /* literal source */
+        v1 = ll;
+        ip += ll;
+        if (ip > slen || op + ll > dlen) {
+          v4 = -3;
+        } else if (ip == slen) {
+          v4 = ip;                              /* final literals-only seq */
+        } else if (ip + 2 > slen) {
+          v4 = -3;
+        } else {
+          long long off = (long long)src[ip] | ((long long)src[ip + 1] << 8);
+          ip += 2;
+          long long ml = tok & 15;
+          if (ml == 15) {
+            unsigned b;
+            do { b = (ip < slen) ? src[ip++] : 0; ml += b; }
+            while (b == 255 && ip < slen);
+          }
+          ml += 4;
+          if (off == 0 || off > op + ll || op + ll + ml > dlen) v4 = -3;
+          else { v2 = off; v3 = ml; v4 = ip; }
+        }
+      }
+    }
+    v0 = __shfl(v0, 0, 64);
+    v1 = __shfl(v1, 0, 64);
+    v2 = __shfl(v2, 0, 64);
+    v3 = __shfl(v3, 0, 64);
+    v4 = __shfl(v4, 0, 64);
+    if (v4 == -2) break;                        /* end of input */
+    if (v4 == -3) { if (lane == 0 && err) err[0] = 1; return; }
+    for (long long i = lane; i < v1; i += 64) dst[op + i] = src[v0 + i];
+    op += v1;
+    if (v3 > 0) {
+      /* match copy: direct when the source window clears the segment,
+       * periodic replication when it overlaps (off < len) — every read
+       * lands strictly before the segment start either way */
+      const long long mo = v2;
+      for (long long i = lane; i < v3; i += 64)
+        dst[op + i] = dst[op - mo + (mo >= v3 ? i : i % mo)];
+      op += v3;
+    }
+    ip = v4;
+  }
+  if (lane == 0 && err) err[0] = (op == dlen) ? 0 : 2;
+}
+
+extern "C" int sn_launch_lz4_decompress(const void *src, long long slen,
+                                        void *dst, long long dlen,
+                                        int32_t *err, void *stream) {
+  hipLaunchKernelGGL(k_lz4_decompress, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (const uint8_t *)src, slen,
+                     (uint8_t *)dst, dlen, err);
+  return (int)hipGetLastError();
+}
+
 /* fold per-block partial rows into the final output.
  * keyless: final[i] = sum_b scratch[b][i]  (NV = 2*NA_t+1, identical layout)
  * grouped: scratch rows are [slot][na1]; final is [slot][out_stride]

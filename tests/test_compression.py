@@ -206,3 +206,42 @@ def test_engine_gpu_snappy_parity():
     assert rows[0][1][1] == float(m.sum())
     assert abs(rows[0][1][0] - f64[m].sum()) <= 1e-6 * abs(f64[m].sum())
     eng.close()
+
+
+@pytest.mark.gpu
+def test_gpu_lz4_device_decode_stress():
+    """f1 compressed-upload: LZ4 blobs now ship compressed over PCIe and
+    decode wave-cooperatively on device (k_lz4_decompress).  Stress the
+    decoder with highly repetitive data (long matches + overlapping
+    offsets + extended length bytes) and a dictionary string column, then
+    parity-check the scan against the plain-blob table."""
+    n = 400_000
+    rng = np.random.default_rng(23)
+    # long runs -> matches with small offsets and 15+ extended lengths
+    i32 = np.repeat(rng.integers(0, 5, 2000), 200)[:n].astype(np.int32)
+    f64 = np.round(rng.random(n), 1)            # repetitive doubles
+    keys = [b"K%d" % v for v in rng.integers(0, 3, n)]
+    plain = [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, i32),
+             po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, f64),
+             po.encode(po.T_STRING, po.ENC_DICT, keys)]
+    wrapped = [wrap_lz4(b) for b in plain]
+    assert sum(map(len, wrapped)) < sum(map(len, plain)) // 2
+
+    eng = se.Engine(device=0)
+    try:
+        schema = [(abi.T_INT32, False), (abi.T_DOUBLE, False),
+                  (abi.T_STRING, False)]
+        t1 = eng.table_define("lzplain", schema)
+        eng.batch_put(t1, 0, 0, n, plain)
+        t2 = eng.table_define("lzwrap", schema)
+        eng.batch_put(t2, 0, 0, n, wrapped)
+        plan_kw = dict(preds=[dict(col=0, hi=2)],
+                       group_cols=[2],
+                       aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+        r1 = eng.query(abi.make_plan(table=t1, **plan_kw)).rows()
+        r2 = eng.query(abi.make_plan(table=t2, **plan_kw)).rows()
+        assert r1 == r2 and len(r1) == 3
+        m = i32 <= 2
+        assert r1[0][1][1] + r1[1][1][1] + r1[2][1][1] == float(m.sum())
+    finally:
+        eng.close()
