@@ -282,6 +282,8 @@ extern "C" int64_t sn_datagen_lineitem(sn_engine *e, int32_t table,
   if (ti.ncols != 7) return SN_ERR_BADARG;
   if (batch_rows <= 0) batch_rows = 600000;   /* ~24 MB across 7 columns */
   if (nthreads <= 0) nthreads = (int32_t)std::thread::hardware_concurrency();
+  /* oversubscription measured counterproductive (allocator/copy contention) */
+  if (nthreads > 64) nthreads = 64;
   int64_t nbatches = (total_rows + batch_rows - 1) / batch_rows;
   std::atomic<int64_t> next(0), put_rows(0);
   std::atomic<int32_t> err(SN_OK);
