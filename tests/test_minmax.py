@@ -165,3 +165,31 @@ def test_minmax_partial_merge(eng):
     finally:
         e0.close()
         e1.close()
+
+
+@pytest.mark.gpu
+def test_minmax_big_dense_groups(eng):
+    """MIN/MAX on >1024 dense slots: the global-atomic route with 8-way XCD
+    privatization needs per-copy identity init (k_acc_init) and an op-aware
+    8-block fold in k_reduce."""
+    n = 1_500_000
+    ngroups = 5_000
+    rng = np.random.default_rng(109)
+    keys = rng.integers(0, ngroups, n).astype(np.int32)   # dense span
+    w = rng.standard_normal(n) * 10
+    t = eng.table_define("tmmbig", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": keys}, {"data": w}], n, batch_rows=300_000)
+    q = eng.query(abi.make_plan(
+        table=t, group_cols=[0],
+        aggs=[("min", [(1, 0.0, 1.0)]), ("max", [(1, 0.0, 1.0)]),
+              ("count", [])]))
+    rows = q.rows()
+    assert len(rows) == len(np.unique(keys))
+    got = {int(k[0]): v for k, v in rows}
+    for g in rng.integers(0, ngroups, 40):
+        m = keys == g
+        if not m.any():
+            continue
+        assert got[int(g)][0] == w[m].min()
+        assert got[int(g)][1] == w[m].max()
+        assert got[int(g)][2] == float(m.sum())

@@ -302,3 +302,41 @@ def test_engine_int64_delta_exact(eng):
     merged[pos] = newv
     assert (merged == cut).sum() == 1    # collision-free by construction
     assert grows[0][1][0] == 1.0
+
+
+@pytest.mark.gpu
+def test_nullable_grouped_wide_pac_global_route(eng):
+    """pac accumulator too wide for LDS (800 slots x 12 aggs -> ~160KB+):
+    the shared routing predicate must send BOTH the kernel and the
+    engine's zeroing to the global-atomic path."""
+    n = 500_000
+    nslots = 800
+    rng = np.random.default_rng(79)
+    keys = rng.integers(0, nslots, n).astype(np.int32)
+    vals = [rng.random(n) for _ in range(3)]
+    valids = [(rng.random(n) > 0.2).astype(np.uint8) for _ in range(3)]
+    cols = [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, keys)] + \
+        [po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, v, valid=vd)
+         for v, vd in zip(vals, valids)]
+    t = eng.table_define("tpacglob", [(abi.T_INT32, False)] +
+                         [(abi.T_DOUBLE, True)] * 3)
+    eng.batch_put(t, 1, 0, n, cols)
+    aggs = []
+    for c in range(1, 4):
+        aggs += [("sum", [(c, 0.0, 1.0)]), ("avg", [(c, 0.0, 1.0)]),
+                 ("sum", [(c, 1.0, 2.0)]), ("avg", [(c, 2.0, -1.0)])]
+    grows = eng.query(abi.make_plan(table=t, group_cols=[0],
+                                    aggs=aggs)).rows()
+    ot = po.OracleTable([po.T_INT32] + [po.T_DOUBLE] * 3,
+                        nullable=[False, True, True, True])
+    ot.add_batch(n, cols)
+    orows = ot.query_groups(po.make_plan(group_cols=[0], aggs=aggs),
+                            nthreads=8)
+    assert len(grows) == len(orows)
+    for (gk, gv), (ok_, ov) in zip(grows, orows):
+        assert gk == ok_
+        for g, o in zip(gv, ov):
+            if o is None:
+                assert g is None
+            else:
+                assert abs(g - o) <= 1e-6 * max(1.0, abs(o)), (gk, g, o)
