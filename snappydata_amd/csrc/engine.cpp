@@ -1190,13 +1190,77 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
                       ix, dn, (long long)i, c);
       }
     }
+    /* Var-width (non-dictionary) STRING bodies (Uncompressed.scala:117-160:
+     * [int32 size][bytes] sequential per non-null row) have no random-access
+     * layout a data-parallel kernel can consume; transcode them to
+     * Dictionary encoding at put (the same re-encode-on-put idea as the
+     * LZ4 wrapper and int-dict materialization below) so the existing
+     * dictionary machinery — global interning, premultiplied maps, pushdown
+     * equality, group keys — applies wholesale. */
+    std::vector<uint8_t> synth;
+    if (t->schema[c].dtype == SN_TYPE_STRING &&
+        b.cols[c].type_id == SN_ENC_UNCOMPRESSED) {
+      const ColMeta dm = b.cols[c];
+      const int32_t rows = b.num_rows;
+      const int32_t nwords = dm.num_null_words;
+      std::map<std::string, int32_t> didx;
+      std::vector<std::string> dict;
+      std::vector<int32_t> idx;
+      idx.reserve(rows);
+      int64_t cur = dm.body_off;
+      int64_t nnull = 0;
+      for (int32_t wi = 0; wi < nwords; wi++)
+        nnull += __builtin_popcountll(
+            (unsigned long long)rd_i64(blob + dm.null_off + (int64_t)wi * 8));
+      const int64_t nn_rows = rows - nnull;
+      int64_t dict_bytes = 0;
+      for (int64_t i = 0; i < nn_rows; i++) {
+        if (cur + 4 > len)
+          return fail(SN_ERR_BADFORMAT, "string body truncated col %d", c);
+        int32_t sz = rd_i32(blob + cur);
+        cur += 4;
+        if (sz < 0 || cur + sz > len)
+          return fail(SN_ERR_BADFORMAT, "string body truncated col %d", c);
+        std::string s((const char *)blob + cur, (size_t)sz);
+        cur += sz;
+        auto it = didx.find(s);
+        int32_t di;
+        if (it == didx.end()) {
+          di = (int32_t)dict.size();
+          dict_bytes += 4 + sz;
+          didx.emplace(std::move(s), di);
+          dict.push_back(std::string((const char *)blob + cur - sz, (size_t)sz));
+        } else di = it->second;
+        idx.push_back(di);
+      }
+      const int64_t nb = (int64_t)nwords * 8;
+      synth.resize(8 + nb + 4 + dict_bytes + (int64_t)idx.size() * 4);
+      int32_t tid3 = SN_ENC_BIG_DICTIONARY, nb32 = (int32_t)nb;
+      memcpy(synth.data(), &tid3, 4);
+      memcpy(synth.data() + 4, &nb32, 4);
+      if (nb) memcpy(synth.data() + 8, blob + dm.null_off, (size_t)nb);
+      uint8_t *w = synth.data() + 8 + nb;
+      int32_t dn = (int32_t)dict.size();
+      memcpy(w, &dn, 4); w += 4;
+      for (auto &s : dict) {
+        int32_t sz = (int32_t)s.size();
+        memcpy(w, &sz, 4); w += 4;
+        memcpy(w, s.data(), s.size()); w += s.size();
+      }
+      memcpy(w, idx.data(), idx.size() * 4);
+      blob = synth.data();
+      len = (int64_t)synth.size();
+      b.cols[c] = ColMeta();
+      int rc2 = parse_blob(blob, len, SN_TYPE_STRING, &b.cols[c]);
+      if (rc2 != SN_OK) return fail(rc2, "transcoded string col %d", c);
+    }
+
     /* int-typed dictionary columns (DictionaryEncoding over int32/int64,
      * DictionaryEncoding.scala:85-137): materialize the values into a plain
      * fixed-width body at put — decompress-on-put, like the LZ4 wrapper —
      * so the scan runs the plain I32/I64 path (JIT-eligible).  Index ==
      * numElements is the null sentinel (DictionaryEncoding.scala:90): such
      * rows are folded into the synthesized null bitset. */
-    std::vector<uint8_t> synth;
     if ((b.cols[c].type_id == SN_ENC_DICTIONARY ||
          b.cols[c].type_id == SN_ENC_BIG_DICTIONARY) &&
         t->schema[c].dtype != SN_TYPE_STRING) {

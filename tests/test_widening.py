@@ -340,3 +340,35 @@ def test_nullable_grouped_wide_pac_global_route(eng):
                 assert g is None
             else:
                 assert abs(g - o) <= 1e-6 * max(1.0, abs(o)), (gk, g, o)
+
+
+@pytest.mark.gpu
+def test_varwidth_string_columns_transcoded(eng):
+    """Uncompressed (non-dictionary) string bodies — [len][bytes] sequential,
+    no random access — transcode to Dictionary at put, so grouping and
+    equality pushdown work on them (previously SN_ERR_UNSUPPORTED)."""
+    n = 120_000
+    rng = np.random.default_rng(137)
+    words = [b"alpha", b"beta", b"gamma", b"delta", b"epsilon"]
+    keys = [words[v] for v in rng.integers(0, 5, n)]
+    w = rng.random(n)
+    valid = (rng.random(n) > 0.1).astype(np.uint8)
+    cols = [po.encode(po.T_STRING, po.ENC_UNCOMPRESSED, keys, valid=valid),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
+    t = eng.table_define("tvarstr", [(abi.T_STRING, True), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 1, 0, n, cols)
+    # group by the var-width string column
+    plan_kw = dict(group_cols=[0], aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_STRING, po.T_DOUBLE], nullable=[True, False])
+    ot.add_batch(n, cols)
+    orows = po.result_rows(ot.query(po.make_plan(**plan_kw)))
+    assert len(grows) == len(orows) == 6       # 5 words + NULL key group
+    for (gk, gv), (ok_, ov) in zip(grows, orows):
+        assert gk == ok_ and gv[1] == ov[1]
+        assert abs(gv[0] - ov[0]) <= REL * max(1.0, abs(ov[0]))
+    # equality pushdown on the transcoded column
+    q = eng.query(abi.make_plan(table=t, preds=[dict(col=0, eq=b"gamma")],
+                                aggs=[("count", [])]))
+    m = np.array([k == b"gamma" for k in keys]) & (valid == 1)
+    assert q.rows()[0][1][0] == float(m.sum())
