@@ -312,6 +312,14 @@ int32_t   sn_query_result_page(sn_query *q, int64_t offset, sn_result *out);
 int64_t   sn_query_num_groups(sn_query *q);
 /* 1 when the query ran a query-compiled (hipRTC) kernel, 0 interpreted */
 int32_t   sn_query_used_jit(sn_query *q);
+/* ORDER BY <aggregate value> [DESC] LIMIT k epilogue (SnappySortExec /
+ * TakeOrderedAndProject semantics, core/.../SnappySortExec.scala): reorders
+ * the finalized group rows by aggregate `agg_idx`'s value (NULLs last,
+ * ties broken by the key ordering) and truncates to k (k <= 0: no limit).
+ * Call after any partial merge; subsequent sn_query_result/`_page` calls
+ * return the reordered rows.  agg_idx < 0 restores the key ordering. */
+int32_t   sn_query_order_by(sn_query *q, int32_t agg_idx, int32_t descending,
+                            int64_t k);
 
 /* ---- multi-GPU partial exchange (caller runs the RCCL collective) ----
  * The caller (one process per GPU, torch.distributed over RCCL/xGMI)

@@ -2686,6 +2686,49 @@ extern "C" int32_t sn_query_result_page(sn_query *q, int64_t offset,
   return result_fill_page(q, offset, out);
 }
 
+/* ORDER BY <aggregate value> [DESC] LIMIT k epilogue — the SnappySortExec /
+ * TakeOrderedAndProject analogue over the finalized group rows.  The sort
+ * key is the RESULT value of aggregate agg_idx (AVG divides, COUNT(*)
+ * counts, NULL groups order last); ties keep the key ordering (stable). */
+extern "C" int32_t sn_query_order_by(sn_query *q, int32_t agg_idx,
+                                     int32_t descending, int64_t k) {
+  if (!q) return SN_ERR_BADARG;
+  int rc = sn_query_wait(q);
+  if (rc != SN_OK) return rc;
+  if (q->final_groups.empty() && !q->merged) {
+    std::vector<GroupOut> groups;
+    local_groups(q, &groups);
+    finalize_groups(q, groups);
+  }
+  const sn_plan &p = q->plan;
+  if (agg_idx >= p.naggs) return fail(SN_ERR_BADARG, "agg_idx out of range");
+  if (agg_idx >= 0) {
+    const sn_agg &ag = p.aggs[agg_idx];
+    auto val_of = [&](const GroupOut &g, double *v) -> bool {
+      if (ag.kind == SN_AGG_COUNT_STAR) { *v = g.sums[agg_idx]; return true; }
+      if (g.counts[agg_idx] <= 0) return false;                 /* NULL */
+      *v = ag.kind == SN_AGG_AVG ? g.sums[agg_idx] / g.counts[agg_idx]
+                                 : g.sums[agg_idx];
+      return true;
+    };
+    std::stable_sort(q->final_groups.begin(), q->final_groups.end(),
+                     [&](const GroupOut &a, const GroupOut &b) {
+                       double va, vb;
+                       bool ha = val_of(a, &va), hb = val_of(b, &vb);
+                       if (ha != hb) return ha;                 /* NULLs last */
+                       if (!ha) return false;
+                       return descending ? va > vb : va < vb;
+                     });
+  } else {
+    /* restore key order */
+    std::vector<GroupOut> groups = q->final_groups;
+    finalize_groups(q, groups);
+  }
+  if (k > 0 && (int64_t)q->final_groups.size() > k)
+    q->final_groups.resize((size_t)k);
+  return SN_OK;
+}
+
 /* total group rows of the finalized result */
 extern "C" int64_t sn_query_num_groups(sn_query *q) {
   if (!q) return SN_ERR_BADARG;

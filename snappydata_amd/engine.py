@@ -83,6 +83,9 @@ def lib():
         L.sn_query_kernel_ms.argtypes = [C.c_void_p]
         L.sn_query_used_jit.restype = C.c_int32
         L.sn_query_used_jit.argtypes = [C.c_void_p]
+        L.sn_query_order_by.restype = C.c_int32
+        L.sn_query_order_by.argtypes = [C.c_void_p, C.c_int32, C.c_int32,
+                                        C.c_int64]
         L.sn_query_result_page.restype = C.c_int32
         L.sn_query_result_page.argtypes = [C.c_void_p, C.c_int64,
                                            C.POINTER(abi.SnResult)]
@@ -256,6 +259,13 @@ class Query:
     def used_jit(self):
         """True when the query ran a query-compiled (hipRTC) kernel."""
         return bool(lib().sn_query_used_jit(self._h))
+
+    def order_by(self, agg_idx, descending=False, k=0):
+        """ORDER BY <aggregate value> [DESC] LIMIT k epilogue
+        (SnappySortExec / TakeOrderedAndProject semantics)."""
+        _check(lib().sn_query_order_by(self._h, agg_idx,
+                                       1 if descending else 0, k), "order_by")
+        return self
 
     def num_groups(self):
         """Total group rows of the local (pre-merge) result."""

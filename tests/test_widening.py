@@ -372,3 +372,43 @@ def test_varwidth_string_columns_transcoded(eng):
                                 aggs=[("count", [])]))
     m = np.array([k == b"gamma" for k in keys]) & (valid == 1)
     assert q.rows()[0][1][0] == float(m.sum())
+
+
+@pytest.mark.gpu
+def test_order_by_topk_epilogue(eng):
+    """ORDER BY <agg> [DESC] LIMIT k (SnappySortExec/TakeOrderedAndProject):
+    reorders finalized groups by aggregate value, NULLs last, stable ties,
+    truncates to k; agg_idx=-1 restores key order."""
+    n = 500_000
+    rng = np.random.default_rng(139)
+    keys = rng.integers(0, 3_000, n).astype(np.int64) * 7919   # sparse path
+    w = rng.random(n)
+    t = eng.table_define("ttopk", [(abi.T_INT64, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": keys}, {"data": w}], n, batch_rows=100_000)
+    q = eng.query(abi.make_plan(table=t, group_cols=[0],
+                                aggs=[("sum", [(1, 0.0, 1.0)]),
+                                      ("count", [])]))
+    q.wait()
+    # numpy reference: top-10 groups by sum desc
+    import collections
+    sums = collections.defaultdict(float)
+    cnts = collections.defaultdict(int)
+    for kk, vv in zip(keys, w):
+        sums[int(kk)] += vv
+        cnts[int(kk)] += 1
+    top = sorted(sums.items(), key=lambda kv: -kv[1])[:10]
+    q.order_by(0, descending=True, k=10)
+    rows = q.rows()
+    assert len(rows) == 10
+    for (gk, gv), (ek, ev) in zip(rows, top):
+        assert int(gk[0]) == ek
+        assert abs(gv[0] - ev) <= 1e-6 * max(1.0, abs(ev))
+        assert gv[1] == float(cnts[ek])
+    # ascending with count tie-breaking stability + key-order restore
+    q2 = eng.query(abi.make_plan(table=t, group_cols=[0],
+                                 aggs=[("count", [])]))
+    q2.wait()
+    q2.order_by(0, descending=False, k=5)
+    asc = q2.rows()
+    counts_sorted = sorted(cnts.values())
+    assert [v[0] for _, v in asc] == [float(c) for c in counts_sorted[:5]]
