@@ -59,6 +59,48 @@ def main():
         join=dict(dim=dim, fact_col=6)))
     print("rows shipping in 1994:", int(qj.rows()[0][1][0]))
 
+    # -- sparse group-by (open-address hash aggregate) + ORDER BY / TOP-K --
+    rng = np.random.default_rng(7)
+    sk = rng.integers(0, 50_000, 1_000_000).astype(np.int64) * (1 << 30)
+    sv = rng.random(1_000_000)
+    ts = eng.table_define("events", [(abi.T_INT64, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(ts, [{"data": sk}, {"data": sv}], len(sk),
+                       batch_rows=250_000)
+    qsp = eng.query(abi.make_plan(table=ts, group_cols=[0],
+                                  aggs=[("sum", [(1, 0.0, 1.0)]),
+                                        ("max", [(1, 0.0, 1.0)]),
+                                        ("count", [])]))
+    qsp.wait()
+    total = qsp.num_groups()
+    qsp.order_by(0, descending=True, k=3)   # TOP-3 groups by SUM
+    print(f"sparse group-by: {total} groups; top-3 by sum:")
+    for key, vals in qsp.rows():
+        print("   key", key[0], "sum", round(vals[0], 2),
+              "max", round(vals[1], 4), "rows", int(vals[2]))
+
+    # -- colocated partitioned-partitioned join (build side on device) -----
+    bt = eng.table_define("suppliers", [(abi.T_INT32, False), (abi.T_STRING, False)])
+    bkeys = np.arange(0, 40_000, 2, dtype=np.int32)
+    battrs = [b"N%d" % (int(k) % 4) for k in bkeys]
+    eng.ingest_columns(bt, [{"data": bkeys},
+                            {"data": b"".join(battrs),
+                             "lens": np.array([len(a) for a in battrs],
+                                              dtype=np.int32)}],
+                       len(bkeys), batch_rows=10_000)
+    fact = eng.table_define("orders", [(abi.T_INT32, False), (abi.T_DOUBLE, False)])
+    fk = rng.integers(0, 40_000, 500_000).astype(np.int32)
+    fv = rng.random(500_000)
+    eng.ingest_columns(fact, [{"data": fk}, {"data": fv}], len(fk),
+                       batch_rows=125_000)
+    d2 = eng.dim_define("suppliers_as_dim")
+    eng.dim_from_table(d2, bt, key_col=0, attr_col=1)
+    qc = eng.query(abi.make_plan(table=fact,
+                                 aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])],
+                                 join=dict(dim=d2, fact_col=0, group=True)))
+    print("colocated join, orders by supplier nation:")
+    for key, vals in qc.rows():
+        print("  ", key[0], "->", round(vals[0], 2), int(vals[1]))
+
     # -- UPDATE then DELETE against a live batch ---------------------------
     tiny = eng.table_define("tiny", [(abi.T_DOUBLE, False)])
     base = np.arange(10, dtype=np.float64)
