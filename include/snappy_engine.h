@@ -254,6 +254,18 @@ int32_t sn_batch_put(sn_engine *e, int32_t table,
                      const sn_buf *columns, const sn_buf *stats,
                      const sn_buf *delete_mask, const sn_buf *deltas);
 
+/* f2 fast ingest (the ColumnBatchCreator rollover with encode + stats
+ * offloaded to the GPU): fixed-width NON-NULL columns arrive as RAW value
+ * arrays (the Uncompressed encoder for numerics is the identity — the
+ * engine uploads the bytes once and computes the ColumnStatsSchema bounds
+ * ON DEVICE, async on the ingest stream); other columns arrive as encoded
+ * blobs in `encoded`.  Per column exactly one of raw[c].data /
+ * encoded[c].data is set.  Queries resolve the pending device bounds with
+ * one stream sync. */
+int32_t sn_batch_put_raw(sn_engine *e, int32_t table, int64_t uuid,
+                         int32_t bucket_id, int32_t num_rows,
+                         const sn_buf *raw, const sn_buf *encoded);
+
 /* number of resident batches / rows for a table on this shard */
 int64_t sn_table_num_batches(sn_engine *e, int32_t table);
 int64_t sn_table_num_rows(sn_engine *e, int32_t table);

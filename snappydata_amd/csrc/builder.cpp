@@ -30,6 +30,8 @@
 extern "C" int32_t sn_batch_put(sn_engine *, int32_t, int64_t, int32_t, int32_t,
                                 const sn_buf *, const sn_buf *, const sn_buf *,
                                 const sn_buf *);
+extern "C" int32_t sn_batch_put_raw(sn_engine *, int32_t, int64_t, int32_t,
+                                    int32_t, const sn_buf *, const sn_buf *);
 
 /* ---------------- little-endian writers ---------------- */
 static inline void put_i32(std::vector<uint8_t> &b, int32_t v) {
@@ -164,16 +166,13 @@ extern "C" int64_t sn_ingest_columns(sn_engine *e, int32_t table, int64_t nrows,
   for (int64_t s = 0, bi = 0; s < nrows; s += batch_rows, bi++) {
     int32_t n = (int32_t)std::min<int64_t>(batch_rows, nrows - s);
     std::vector<std::vector<uint8_t>> blobs(nc);
-    std::vector<sn_buf> bufs(nc);
-    std::vector<sn_type_t> dtypes(nc);
-    std::vector<double> lo_d(nc, 0), hi_d(nc, 0);
-    std::vector<int64_t> lo_i(nc, 0), hi_i(nc, 0);
-    std::vector<int32_t> ncount(nc, 0);
-    std::vector<uint8_t> hb(nc, 0);
+    std::vector<sn_buf> raw(nc), enc(nc);
     for (int c = 0; c < nc; c++) {
-      dtypes[c] = ti.dtypes[c];
+      raw[c].data = nullptr; raw[c].len = 0;
+      enc[c].data = nullptr; enc[c].len = 0;
       sn_ingest_col view = cols[c];
       const uint8_t *base = (const uint8_t *)cols[c].data;
+      if (cols[c].valid) view.valid = cols[c].valid + s;
       if (ti.dtypes[c] == SN_TYPE_STRING) {
         view.data = base + str_off[c];
         view.str_lens = cols[c].str_lens + s;
@@ -185,44 +184,26 @@ extern "C" int64_t sn_ingest_columns(sn_engine *e, int32_t table, int64_t nrows,
                 (ti.dtypes[c] == SN_TYPE_INT16) ? 2 :
                 (ti.dtypes[c] == SN_TYPE_INT8 || ti.dtypes[c] == SN_TYPE_BOOL) ? 1 : 4;
         view.data = base + (int64_t)s * w;
-      }
-      if (cols[c].valid) view.valid = cols[c].valid + s;
-      int rc = encode_column(ti.dtypes[c], view, n, blobs[c]);
-      if (rc != SN_OK) return rc;
-      /* stats: min/max over non-null values */
-      bool first = true;
-      for (int32_t i = 0; i < n; i++) {
-        if (view.valid && !view.valid[i]) { ncount[c]++; continue; }
-        switch (ti.dtypes[c]) {
-          case SN_TYPE_DOUBLE: {
-            double v = ((const double *)view.data)[i];
-            if (first || v < lo_d[c]) lo_d[c] = v;
-            if (first || v > hi_d[c]) hi_d[c] = v;
-            hb[c] = 1; first = false; break;
-          }
-          case SN_TYPE_INT32: {
-            int32_t v = ((const int32_t *)view.data)[i];
-            if (first || v < lo_i[c]) lo_i[c] = v;
-            if (first || v > hi_i[c]) hi_i[c] = v;
-            hb[c] = 1; first = false; break;
-          }
-          case SN_TYPE_INT64: {
-            int64_t v = ((const int64_t *)view.data)[i];
-            if (first || v < lo_i[c]) lo_i[c] = v;
-            if (first || v > hi_i[c]) hi_i[c] = v;
-            hb[c] = 1; first = false; break;
-          }
-          default: break;   /* strings/other: no bounds in stats (round 1) */
+        bool rawable = !view.valid &&
+            (ti.dtypes[c] == SN_TYPE_DOUBLE || ti.dtypes[c] == SN_TYPE_FLOAT ||
+             ti.dtypes[c] == SN_TYPE_INT64 || ti.dtypes[c] == SN_TYPE_INT32 ||
+             ti.dtypes[c] == SN_TYPE_INT16);
+        if (rawable) {
+          /* f2 fast path: the Uncompressed encoder for non-null numerics is
+           * the identity — hand the raw slice to the engine, which uploads
+           * it once and computes the stats bounds ON DEVICE */
+          raw[c].data = view.data;
+          raw[c].len = (int64_t)n * w;
+          continue;
         }
       }
-      bufs[c].data = blobs[c].data();
-      bufs[c].len = (int64_t)blobs[c].size();
+      int rc = encode_column(ti.dtypes[c], view, n, blobs[c]);
+      if (rc != SN_OK) return rc;
+      enc[c].data = blobs[c].data();
+      enc[c].len = (int64_t)blobs[c].size();
     }
-    std::vector<uint8_t> stats;
-    encode_stats(dtypes, n, lo_d, hi_d, lo_i, hi_i, ncount, hb, stats);
-    sn_buf sbuf = { stats.data(), (int64_t)stats.size() };
-    int32_t rc = sn_batch_put(e, table, bi, first_bucket + (int32_t)bi, n,
-                              bufs.data(), &sbuf, nullptr, nullptr);
+    int32_t rc = sn_batch_put_raw(e, table, bi, first_bucket + (int32_t)bi, n,
+                                  raw.data(), enc.data());
     if (rc != SN_OK) return rc;
     put_rows += n;
   }
@@ -308,40 +289,27 @@ extern "C" int64_t sn_datagen_lineitem(sn_engine *e, int32_t table,
       sn_gen_lineitem_arrays(rows_before, n, seed, qty.data(), ep.data(),
                              disc.data(), tax.data(), rf.data(), ls.data(),
                              ship.data());
-      /* encode the 7 columns */
-      std::vector<std::vector<uint8_t>> blobs(7);
+      /* f2 fast path: numerics ship RAW (encode = identity, stats on
+       * device); only the two 1-char dictionary string columns encode */
+      std::vector<std::vector<uint8_t>> blobs(2);
       sn_ingest_col c;
-      c.valid = nullptr; c.str_lens = nullptr;
-      c.data = qty.data();  encode_column(SN_TYPE_DOUBLE, c, n, blobs[0]);
-      c.data = ep.data();   encode_column(SN_TYPE_DOUBLE, c, n, blobs[1]);
-      c.data = disc.data(); encode_column(SN_TYPE_DOUBLE, c, n, blobs[2]);
-      c.data = tax.data();  encode_column(SN_TYPE_DOUBLE, c, n, blobs[3]);
-      c.data = rf.data();   c.str_lens = len1.data();
-      encode_column(SN_TYPE_STRING, c, n, blobs[4]);
-      c.data = ls.data();
-      encode_column(SN_TYPE_STRING, c, n, blobs[5]);
-      c.str_lens = nullptr;
-      c.data = ship.data(); encode_column(SN_TYPE_INT32, c, n, blobs[6]);
-      /* stats */
-      std::vector<double> lo_d(7, 0), hi_d(7, 0);
-      std::vector<int64_t> lo_i(7, 0), hi_i(7, 0);
-      std::vector<int32_t> ncnt(7, 0);
-      std::vector<uint8_t> hb(7, 0);
-      auto mm_d = [&](int ci, const double *v) {
-        double lo = v[0], hi = v[0];
-        for (int32_t i = 1; i < n; i++) { if (v[i] < lo) lo = v[i]; if (v[i] > hi) hi = v[i]; }
-        lo_d[ci] = lo; hi_d[ci] = hi; hb[ci] = 1;
-      };
-      mm_d(0, qty.data()); mm_d(1, ep.data()); mm_d(2, disc.data()); mm_d(3, tax.data());
-      int32_t slo = ship[0], shi = ship[0];
-      for (int32_t i = 1; i < n; i++) { if (ship[i] < slo) slo = ship[i]; if (ship[i] > shi) shi = ship[i]; }
-      lo_i[6] = slo; hi_i[6] = shi; hb[6] = 1;
-      std::vector<uint8_t> stats;
-      encode_stats(dtypes, n, lo_d, hi_d, lo_i, hi_i, ncnt, hb, stats);
-      sn_buf bufs[7], sbuf = { stats.data(), (int64_t)stats.size() };
-      for (int i = 0; i < 7; i++) { bufs[i].data = blobs[i].data(); bufs[i].len = (int64_t)blobs[i].size(); }
-      int32_t rc = sn_batch_put(e, table, bi, (int32_t)bi, n, bufs, &sbuf,
-                                nullptr, nullptr);
+      c.valid = nullptr;
+      c.str_lens = len1.data();
+      c.data = rf.data(); encode_column(SN_TYPE_STRING, c, n, blobs[0]);
+      c.data = ls.data(); encode_column(SN_TYPE_STRING, c, n, blobs[1]);
+      sn_buf raw[7], enc[7];
+      for (int i = 0; i < 7; i++) {
+        raw[i].data = nullptr; raw[i].len = 0;
+        enc[i].data = nullptr; enc[i].len = 0;
+      }
+      raw[0] = { qty.data(), (int64_t)n * 8 };
+      raw[1] = { ep.data(), (int64_t)n * 8 };
+      raw[2] = { disc.data(), (int64_t)n * 8 };
+      raw[3] = { tax.data(), (int64_t)n * 8 };
+      enc[4] = { blobs[0].data(), (int64_t)blobs[0].size() };
+      enc[5] = { blobs[1].data(), (int64_t)blobs[1].size() };
+      raw[6] = { ship.data(), (int64_t)n * 4 };
+      int32_t rc = sn_batch_put_raw(e, table, bi, (int32_t)bi, n, raw, enc);
       if (rc != SN_OK) { err.store(rc); break; }
       put_rows += n;
     }

@@ -184,6 +184,12 @@ struct Batch {
    * local2global is fixed per batch, so the device map is reusable
    * across queries with the same group geometry */
   std::vector<std::map<std::pair<int32_t, int32_t>, const int32_t *>> dictmap_cache;
+  /* f2 raw ingest: device-computed stats pending a sync.  Layout:
+   * [nc mins][nc maxs] as order-preserving u64 encodings (f64 ord / int
+   * sign-bias); is_raw marks which columns the device bounds cover. */
+  const unsigned long long *stats_dev = nullptr;
+  std::vector<uint8_t> is_raw;
+  bool stats_pending = false;
   std::vector<ColMeta> cols;
   /* stats (parsed) */
   bool stats_valid = false;
@@ -432,6 +438,7 @@ extern "C" int32_t sn_dim_put(sn_engine *e, int32_t dim, int64_t nkeys,
 
 static void *up(sn_engine *e, const void *host, size_t n);
 static Table *get_table(sn_engine *e, int32_t t);
+static void sync_pending_stats(sn_engine *e, Table *t);
 
 static inline uint64_t mix64h(uint64_t x) {
   x += 0x9E3779B97f4A7C15ull;
@@ -473,6 +480,7 @@ extern "C" int32_t sn_dim_from_table(sn_engine *e, int32_t dim, int32_t table,
   if (t->total_rows > (1ll << 23))
     return fail(SN_ERR_UNSUPPORTED,
                 "build side exceeds 2^23 rows (HashJoinSize-class bound)");
+  sync_pending_stats(e, t);   /* the LUT span reads stats bounds */
 
   sn_dev_plan dp;
   memset(&dp, 0, sizeof(dp));
@@ -1124,42 +1132,14 @@ static void apply_stats(Table *t, Batch &b, const sn_buf *stats) {
 
 }
 
-extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
-                                int64_t uuid, int32_t bucket_id, int32_t num_rows,
-                                const sn_buf *columns, const sn_buf *stats,
-                                const sn_buf *delete_mask, const sn_buf *deltas) {
-  Table *t = get_table(e, table);
-  if (!t || !columns || num_rows == 0) return fail(SN_ERR_BADARG, "bad batch args");
-  /* shard filter: batches whose bucket belongs to another rank are ignored */
-  if (e->cfg.shard_count > 1 &&
-      (bucket_id % e->cfg.shard_count) != e->cfg.shard_rank)
-    return SN_OK;
-
-  const int nc = (int)t->schema.size();
-  Batch b;
-  b.uuid = uuid; b.bucket = bucket_id;
-  b.has_deltas = num_rows < 0;
-  b.num_rows = num_rows < 0 ? -num_rows : num_rows;
-  b.cols.resize(nc);
-  b.col_dev.resize(nc);
-  b.nullpfx_dev.resize(nc, nullptr);
-  b.patch_dev.resize(nc);
-  b.patch_host.resize(nc);
-  b.had_patches.assign(nc, 0);
-  b.rle_ends_dev.resize(nc, nullptr);
-  b.rle_vals_dev.resize(nc, nullptr);
-  b.rle_n.resize(nc, 0);
-  b.dictmap_cache.resize(nc);
-
-  /* Phase 1 — NO table lock: decompress, parse, validate, materialize and
-   * upload are all batch-local (the arena has its own mutex), so concurrent
-   * puts from many ingest threads overlap their heavy work.  Only the
-   * global-dictionary interning, delta decode (which interns) and the
-   * batches-vector append need t->mu (phase 2 below). */
+/* one encoded column blob -> device-resident state (decompress, parse,
+ * transcode/materialize, validate, upload, aux) — batch-local, no table
+ * lock; shared by sn_batch_put and sn_batch_put_raw */
+static int32_t process_encoded_col(sn_engine *e, Table *t, Batch &b, int c,
+                                   const sn_buf &column) {
   std::vector<uint8_t> decomp;
-  for (int c = 0; c < nc; c++) {
-    const uint8_t *blob = (const uint8_t *)columns[c].data;
-    int64_t len = columns[c].len;
+    const uint8_t *blob = (const uint8_t *)column.data;
+    int64_t len = column.len;
     int dec = maybe_decompress(blob, len, &decomp);
     if (dec < 0) return dec;
     if (dec == 1) { blob = decomp.data(); len = (int64_t)decomp.size(); }
@@ -1328,8 +1308,8 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
       char *dst = base + pad;
       bool dev_decoded = false;
       if (e->arena.device >= 0 && dec == 1 &&
-          rd_i32((const uint8_t *)columns[c].data) == -1 /* LZ4 */) {
-        const int64_t clen = columns[c].len - 8;
+          rd_i32((const uint8_t *)column.data) == -1 /* LZ4 */) {
+        const int64_t clen = column.len - 8;
         /* the engine stream + error word are shared: one decode at a time
          * (concurrent puts overlap their host work; device decodes queue) */
         std::lock_guard<std::mutex> glz(e->lz4_mu);
@@ -1337,7 +1317,7 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
         if (!e->lz4_err_dev)
           e->lz4_err_dev = (int32_t *)e->arena.alloc(64);
         if (cdev && e->lz4_err_dev &&
-            hipMemcpy(cdev, (const uint8_t *)columns[c].data + 8,
+            hipMemcpy(cdev, (const uint8_t *)column.data + 8,
                       (size_t)clen, hipMemcpyHostToDevice) == hipSuccess &&
             hipMemsetAsync(e->lz4_err_dev, 0, 4, e->stream) == hipSuccess &&
             sn_launch_lz4_decompress(cdev, clen, dst, len, e->lz4_err_dev,
@@ -1407,7 +1387,231 @@ extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
       }
       b.nullpfx_dev[c] = (const uint32_t *)up(e, pfx.data(), pfx.size() * 4);
     }
+    return SN_OK;
+}
+
+/* f2 fast ingest: fixed-width NON-NULL columns arrive as RAW value arrays
+ * (the Uncompressed encoder for numerics is the identity), other columns as
+ * encoded blobs.  Bounds for the raw columns compute ON DEVICE (async
+ * min/max on the engine stream — ColumnEncoder's lowerLong/upperLong
+ * tracking, ColumnEncoding.scala:188-251, moved to the GPU); queries force
+ * the pending sync.  This is the ColumnBatchCreator rollover with encode +
+ * stats offloaded — ingest threads only touch the bytes once. */
+extern "C" int32_t sn_batch_put_raw(sn_engine *e, int32_t table,
+                                    int64_t uuid, int32_t bucket_id,
+                                    int32_t num_rows, const sn_buf *raw,
+                                    const sn_buf *encoded) {
+  Table *t = get_table(e, table);
+  if (!t || !raw || !encoded || num_rows <= 0)
+    return fail(SN_ERR_BADARG, "bad raw batch args");
+  if (e->cfg.shard_count > 1 &&
+      (bucket_id % e->cfg.shard_count) != e->cfg.shard_rank)
+    return SN_OK;
+  const int nc = (int)t->schema.size();
+  Batch b;
+  b.uuid = uuid;
+  b.bucket = bucket_id;
+  b.num_rows = num_rows;
+  b.cols.resize(nc);
+  b.col_dev.resize(nc);
+  b.nullpfx_dev.resize(nc, nullptr);
+  b.patch_dev.resize(nc);
+  b.patch_host.resize(nc);
+  b.had_patches.assign(nc, 0);
+  b.rle_ends_dev.resize(nc, nullptr);
+  b.rle_vals_dev.resize(nc, nullptr);
+  b.rle_n.resize(nc, 0);
+  b.dictmap_cache.resize(nc);
+  b.is_raw.assign(nc, 0);
+
+  unsigned long long *sd = nullptr;
+  if (e->has_gpu) {
+    sd = (unsigned long long *)e->arena.alloc((size_t)nc * 16);
+    if (!sd) return fail(SN_ERR_NOMEM, "stats scratch");
+    if (hipMemsetAsync(sd, 0xff, (size_t)nc * 8, e->stream) != hipSuccess ||
+        hipMemsetAsync(sd + nc, 0, (size_t)nc * 8, e->stream) != hipSuccess)
+      return fail(SN_ERR_GENERIC, "stats scratch init");
   }
+  bool any_raw = false;
+  for (int c = 0; c < nc; c++) {
+    if (!raw[c].data) {
+      if (!encoded[c].data)
+        return fail(SN_ERR_BADARG, "column %d has neither raw nor encoded", c);
+      int32_t rc_pc = process_encoded_col(e, t, b, c, encoded[c]);
+      if (rc_pc != SN_OK) return rc_pc;
+      continue;
+    }
+    sn_type_t dt = t->schema[c].dtype;
+    int w, kind;
+    switch (dt) {
+      case SN_TYPE_DOUBLE: w = 8; kind = SN_K_F64; break;
+      case SN_TYPE_INT64:  w = 8; kind = SN_K_I64; break;
+      case SN_TYPE_INT32:  w = 4; kind = SN_K_I32; break;
+      case SN_TYPE_FLOAT:  w = 4; kind = SN_K_F32; break;
+      case SN_TYPE_INT16:  w = 2; kind = SN_K_I16; break;
+      default:
+        return fail(SN_ERR_UNSUPPORTED, "raw column %d dtype %d", c, (int)dt);
+    }
+    if (raw[c].len != (int64_t)num_rows * w)
+      return fail(SN_ERR_BADARG, "raw column %d length", c);
+    /* device blob = [typeId=0][nullBytes=0][body]: header at base+8
+     * (8-aligned), body at base+16 (16-aligned — arena blocks are
+     * 256-aligned); blob pointer = base+8, body_off = 8 */
+    char *base = (char *)e->arena.alloc((size_t)raw[c].len + 16);
+    if (!base) return fail(SN_ERR_NOMEM, "HBM upload failed");
+    char *body = base + 16;
+    b.cols[c].type_id = SN_ENC_UNCOMPRESSED;
+    if (e->arena.device >= 0) {
+      int32_t hdr[2] = { SN_ENC_UNCOMPRESSED, 0 };
+      if (hipMemcpy(base + 8, hdr, 8, hipMemcpyHostToDevice) != hipSuccess ||
+          h2d_copy(e, body, raw[c].data, (size_t)raw[c].len) != hipSuccess)
+        return fail(SN_ERR_NOMEM, "HBM upload failed");
+      if (sn_launch_col_minmax(body, num_rows, kind, sd + c, sd + nc + c,
+                               e->stream) != 0)
+        return fail(SN_ERR_GENERIC, "stats kernel");
+      b.is_raw[c] = 1;
+      any_raw = true;
+    } else {
+      int32_t hdr[2] = { SN_ENC_UNCOMPRESSED, 0 };
+      memcpy(base + 8, hdr, 8);
+      memcpy(body, raw[c].data, (size_t)raw[c].len);
+      b.host_blobs.emplace_back((const uint8_t *)base + 8,
+                                (const uint8_t *)body + raw[c].len);
+      /* host-only: cheap scalar bounds so stats-skip still works */
+      b.lo_i.resize(nc); b.hi_i.resize(nc);
+      b.lo_d.resize(nc); b.hi_d.resize(nc);
+      b.null_count.resize(nc, 0);
+      b.bounds_null.resize(nc, 1);
+      const uint8_t *p8 = (const uint8_t *)raw[c].data;
+      if (dt == SN_TYPE_DOUBLE || dt == SN_TYPE_FLOAT) {
+        double mn = 0, mx = 0;
+        for (int32_t i = 0; i < num_rows; i++) {
+          double v = dt == SN_TYPE_DOUBLE ? rd_f64(p8 + (int64_t)i * 8)
+                                          : rd_f32(p8 + (int64_t)i * 4);
+          if (i == 0 || v < mn) mn = v;
+          if (i == 0 || v > mx) mx = v;
+        }
+        b.lo_d[c] = mn; b.hi_d[c] = mx;
+      } else {
+        int64_t mn = 0, mx = 0;
+        for (int32_t i = 0; i < num_rows; i++) {
+          int64_t v = w == 8 ? rd_i64(p8 + (int64_t)i * 8)
+                    : w == 4 ? rd_i32(p8 + (int64_t)i * 4)
+                             : rd_i16(p8 + (int64_t)i * 2);
+          if (i == 0 || v < mn) mn = v;
+          if (i == 0 || v > mx) mx = v;
+        }
+        b.lo_i[c] = mn; b.hi_i[c] = mx;
+      }
+      b.bounds_null[c] = 0;
+      b.stats_valid = true;
+    }
+    b.col_dev[c] = base + 8;
+    b.cols[c].body_off = 8;
+  }
+  if (any_raw) {
+    b.stats_dev = sd;
+    b.stats_pending = true;
+  }
+
+  std::lock_guard<std::mutex> g(t->mu);
+  for (int c = 0; c < nc; c++) {
+    if (t->schema[c].dtype != SN_TYPE_STRING || b.cols[c].dict.empty())
+      continue;
+    auto &l2g = b.cols[c].local2global;
+    l2g.reserve(b.cols[c].dict.size());
+    for (auto &sstr : b.cols[c].dict) {
+      auto it = t->gdict_idx[c].find(sstr);
+      int32_t gid;
+      if (it == t->gdict_idx[c].end()) {
+        gid = (int32_t)t->gdict[c].size();
+        t->gdict[c].push_back(sstr);
+        t->gdict_idx[c].emplace(sstr, gid);
+        if ((int32_t)sstr.size() > t->gdict_maxlen[c])
+          t->gdict_maxlen[c] = (int32_t)sstr.size();
+      } else gid = it->second;
+      l2g.push_back(gid);
+    }
+  }
+  t->total_rows += b.num_rows;
+  t->batches.push_back(std::move(b));
+  return SN_OK;
+}
+
+/* resolve device-computed bounds for raw-ingested batches (one stream sync
+ * for ALL pending batches; called under t->mu before any bounds are read) */
+static void sync_pending_stats(sn_engine *e, Table *t) {
+  bool any = false;
+  for (auto &b : t->batches) any |= b.stats_pending;
+  if (!any) return;
+  (void)hipStreamSynchronize(e->stream);
+  const int nc = (int)t->schema.size();
+  std::vector<unsigned long long> h((size_t)nc * 2);
+  for (auto &b : t->batches) {
+    if (!b.stats_pending) continue;
+    b.stats_pending = false;
+    if (hipMemcpy(h.data(), b.stats_dev, (size_t)nc * 16,
+                  hipMemcpyDeviceToHost) != hipSuccess)
+      continue;                         /* bounds stay absent: conservative */
+    b.lo_d.resize(nc); b.hi_d.resize(nc);
+    b.lo_i.resize(nc); b.hi_i.resize(nc);
+    b.null_count.assign(nc, 0);
+    b.bounds_null.assign(nc, 1);
+    for (int c = 0; c < nc; c++) {
+      if (!b.is_raw[c]) continue;
+      sn_type_t dt = t->schema[c].dtype;
+      if (dt == SN_TYPE_DOUBLE || dt == SN_TYPE_FLOAT) {
+        b.lo_d[c] = sn_ord_f64_h(h[c]);
+        b.hi_d[c] = sn_ord_f64_h(h[nc + c]);
+      } else {
+        b.lo_i[c] = (int64_t)(h[c] ^ 0x8000000000000000ull);
+        b.hi_i[c] = (int64_t)(h[nc + c] ^ 0x8000000000000000ull);
+      }
+      b.bounds_null[c] = 0;
+    }
+    b.stats_valid = true;
+    e->arena.release((void *)b.stats_dev, (size_t)nc * 16);
+    b.stats_dev = nullptr;
+  }
+}
+
+extern "C" int32_t sn_batch_put(sn_engine *e, int32_t table,
+                                int64_t uuid, int32_t bucket_id, int32_t num_rows,
+                                const sn_buf *columns, const sn_buf *stats,
+                                const sn_buf *delete_mask, const sn_buf *deltas) {
+  Table *t = get_table(e, table);
+  if (!t || !columns || num_rows == 0) return fail(SN_ERR_BADARG, "bad batch args");
+  /* shard filter: batches whose bucket belongs to another rank are ignored */
+  if (e->cfg.shard_count > 1 &&
+      (bucket_id % e->cfg.shard_count) != e->cfg.shard_rank)
+    return SN_OK;
+
+  const int nc = (int)t->schema.size();
+  Batch b;
+  b.uuid = uuid; b.bucket = bucket_id;
+  b.has_deltas = num_rows < 0;
+  b.num_rows = num_rows < 0 ? -num_rows : num_rows;
+  b.cols.resize(nc);
+  b.col_dev.resize(nc);
+  b.nullpfx_dev.resize(nc, nullptr);
+  b.patch_dev.resize(nc);
+  b.patch_host.resize(nc);
+  b.had_patches.assign(nc, 0);
+  b.rle_ends_dev.resize(nc, nullptr);
+  b.rle_vals_dev.resize(nc, nullptr);
+  b.rle_n.resize(nc, 0);
+  b.dictmap_cache.resize(nc);
+
+  /* Phase 1 — NO table lock: decompress, parse, validate, materialize and
+   * upload are all batch-local (the arena has its own mutex), so concurrent
+   * puts from many ingest threads overlap their heavy work.  Only the
+   * global-dictionary interning, delta decode (which interns) and the
+   * batches-vector append need t->mu (phase 2 below). */
+  for (int c = 0; c < nc; c++) {
+    int32_t rc_pc = process_encoded_col(e, t, b, c, columns[c]);
+    if (rc_pc != SN_OK) return rc_pc;
+  }
+
 
   { int32_t rc_ = apply_delete_mask(e, b, delete_mask); if (rc_ != SN_OK) return rc_; }
 
@@ -1473,6 +1677,15 @@ extern "C" int32_t sn_batch_mutate(sn_engine *e, int32_t table, int64_t uuid,
   if (deltas) {
     int32_t rc = apply_deltas(e, t, *b, deltas);
     if (rc != SN_OK) return rc;
+  }
+  if (b->stats_pending) {
+    /* caller-provided (or absent) stats supersede the device-computed ones */
+    b->stats_pending = false;
+    if (b->stats_dev) {
+      e->arena.release((void *)b->stats_dev,
+                       t->schema.size() * 16);
+      b->stats_dev = nullptr;
+    }
   }
   if (stats) apply_stats(t, b[0], stats);
   else b->stats_valid = false;   /* old bounds no longer trustworthy */
@@ -1739,6 +1952,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
    * the same consistent table state or grouped kernels write accumulator
    * rows out of bounds. */
   std::lock_guard<std::mutex> g(t->mu);
+  sync_pending_stats(e, t);   /* raw-ingested batches: resolve device bounds */
 
   /* group slot space: global dict sizes, plus a null slot only when the
    * schema allows null keys (non-nullable key columns waste no slots —

@@ -238,6 +238,13 @@ int sn_launch_hash_to_lut(const long long *hk, const int32_t *hp,
 int sn_launch_lz4_decompress(const void *src, long long slen, void *dst,
                              long long dlen, int32_t *err, void *stream);
 
+/* f2: device min/max over one raw fixed-width column; omin memset 0xFF,
+ * omax memset 0x00 (values land as order-preserving u64 encodings:
+ * f64 ord / integer sign-bias) */
+int sn_launch_col_minmax(const void *body, long long n, int kind,
+                         unsigned long long *omin, unsigned long long *omax,
+                         void *stream);
+
 /* launches only the partial-fold (k_reduce): scratch[nblocks][nv] -> out.
  * Used by the JIT path, whose scan kernel writes the same scratch rows.
  * plan_dev (nullable) supplies per-agg ops for MIN/MAX folding. */

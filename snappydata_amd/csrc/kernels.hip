@@ -1661,6 +1661,107 @@ extern "C" int sn_launch_hash_compact(const long long *hk, const double *hacc,
   return (int)hipGetLastError();
 }
 
+/* ---- device-side batch stats (f2: GPU batch building) ----
+ * min/max over one raw fixed-width column (the ColumnStatsSchema bounds
+ * ColumnEncoder tracks during encode, ColumnEncoding.scala:188-251),
+ * accumulated into out[0]=min, out[1]=max with ord-encoded u64 atomics
+ * (doubles) or native i64 atomics — async on the put stream, so ingest
+ * never syncs per batch. */
+__device__ __forceinline__ unsigned long long wave_umin(unsigned long long x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const unsigned long long y =
+        (unsigned long long)__shfl_down((long long)x, off, 64);
+    x = y < x ? y : x;
+  }
+  return x;
+}
+__device__ __forceinline__ unsigned long long wave_umax(unsigned long long x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    const unsigned long long y =
+        (unsigned long long)__shfl_down((long long)x, off, 64);
+    x = y > x ? y : x;
+  }
+  return x;
+}
+
+/* omin slots memset to 0xFF (>= every encoding), omax to 0x00: f64 values
+ * use the ord encoding, integers the sign-bias (v ^ 1<<63) — both
+ * order-preserving into unsigned, so the memset identities work for all */
+__global__ __launch_bounds__(256, 4)
+void k_col_minmax(const void *__restrict__ body_, long long n, int kind,
+                  unsigned long long *__restrict__ omin_,
+                  unsigned long long *__restrict__ omax_) {
+  GAS unsigned long long *omin = (GAS unsigned long long *)(uintptr_t)omin_;
+  GAS unsigned long long *omax = (GAS unsigned long long *)(uintptr_t)omax_;
+  const long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  unsigned long long emn = ~0ull, emx = 0ull;
+  if (kind == SN_K_F64 || kind == SN_K_F32) {
+    double mn = __longlong_as_double(0x7FF0000000000000ll);   /* +inf */
+    double mx = -mn;
+    bool any = false;
+    if (kind == SN_K_F64) {
+      const GAS double *b = (const GAS double *)(uintptr_t)body_;
+      for (long long i = i0; i < n; i += stride) {
+        const double v = b[i];
+        mn = fmin(mn, v); mx = fmax(mx, v); any = true;
+      }
+    } else {
+      const GAS float *b = (const GAS float *)(uintptr_t)body_;
+      for (long long i = i0; i < n; i += stride) {
+        const double v = (double)b[i];
+        mn = fmin(mn, v); mx = fmax(mx, v); any = true;
+      }
+    }
+    if (any) { emn = f64_ord(mn); emx = f64_ord(mx); }
+  } else {
+    long long mn = 0x7FFFFFFFFFFFFFFFll, mx = (long long)0x8000000000000000ll;
+    bool any = false;
+    if (kind == SN_K_I64) {
+      const GAS long long *b = (const GAS long long *)(uintptr_t)body_;
+      for (long long i = i0; i < n; i += stride) {
+        const long long v = b[i];
+        mn = v < mn ? v : mn; mx = v > mx ? v : mx; any = true;
+      }
+    } else if (kind == SN_K_I32) {
+      const GAS int *b = (const GAS int *)(uintptr_t)body_;
+      for (long long i = i0; i < n; i += stride) {
+        const long long v = b[i];
+        mn = v < mn ? v : mn; mx = v > mx ? v : mx; any = true;
+      }
+    } else {  /* I16 */
+      const GAS short *b = (const GAS short *)(uintptr_t)body_;
+      for (long long i = i0; i < n; i += stride) {
+        const long long v = b[i];
+        mn = v < mn ? v : mn; mx = v > mx ? v : mx; any = true;
+      }
+    }
+    if (any) {
+      emn = (unsigned long long)mn ^ 0x8000000000000000ull;
+      emx = (unsigned long long)mx ^ 0x8000000000000000ull;
+    }
+  }
+  emn = wave_umin(emn);
+  emx = wave_umax(emx);
+  if ((threadIdx.x & 63) == 0) {
+    (void)atomicMin((unsigned long long *)omin, emn);
+    (void)atomicMax((unsigned long long *)omax, emx);
+  }
+}
+
+extern "C" int sn_launch_col_minmax(const void *body, long long n, int kind,
+                                    unsigned long long *omin,
+                                    unsigned long long *omax, void *stream) {
+  long long blocks = (n + 256 * 16 - 1) / (256 * 16);
+  if (blocks < 1) blocks = 1;
+  if (blocks > 1024) blocks = 1024;
+  hipLaunchKernelGGL(k_col_minmax, dim3((unsigned)blocks), dim3(256), 0,
+                     (hipStream_t)stream, body, n, kind, omin, omax);
+  return (int)hipGetLastError();
+}
+
 /* ---- wave-cooperative LZ4 block decode (f1: compressed-upload ingest) ----
  * The reference wraps each column blob as ONE raw LZ4 block
  * (CompressionUtils.scala:132-160): the token chain is strictly
