@@ -829,3 +829,62 @@ def test_fullsize_checksum_of_checksums(eng):
     gcnt = sum(v[1] for _, v in grows)
     assert gcnt == krows[0][1][1] == float(n)
     assert abs(gsum - krows[0][1][0]) <= 1e-9 * abs(gsum)
+
+
+def test_concurrent_raw_ingest_and_scan(eng):
+    """Config-5 concurrency over the f2 RAW ingest path: device-computed
+    stats are resolved lazily (sync_pending_stats under the table lock)
+    while another thread queries with a stats-skippable predicate — every
+    result must still be a consistent batch prefix."""
+    import threading
+    n_batches, rows = 30, 20_000
+    rng = np.random.default_rng(19)
+    data = [np.round(rng.random(rows), 3) + bi for bi in range(n_batches)]
+    t = eng.table_define("tconcraw", [(abi.T_DOUBLE, False)])
+    # predicate (col >= 0) exercises the stats path on every submit
+    plan = abi.make_plan(table=t, preds=[dict(col=0, is_double=True, lo=0.0)],
+                         aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])])
+    stop = threading.Event()
+    errs = []
+
+    def ingest():
+        try:
+            for bi in range(n_batches):
+                eng.ingest_columns(t, [{"data": data[bi]}], rows,
+                                   batch_rows=rows, first_bucket=bi)
+        except Exception as ex:  # pragma: no cover
+            errs.append(ex)
+        finally:
+            stop.set()
+
+    prefix_sums = np.cumsum([d.sum() for d in data])
+
+    def scan():
+        try:
+            seen_any = False
+            while not stop.is_set() or not seen_any:
+                res = eng.query(plan).rows()
+                cnt = res[0][1][1]
+                if cnt == 0.0:
+                    continue
+                seen_any = True
+                k = int(cnt // rows)
+                assert cnt == k * float(rows), "partial batch visible"
+                assert abs(res[0][1][0] - prefix_sums[k - 1]) <= \
+                    1e-9 * prefix_sums[k - 1]
+        except Exception as ex:  # pragma: no cover
+            errs.append(ex)
+
+    ti = threading.Thread(target=ingest)
+    ts = threading.Thread(target=scan)
+    ti.start(); ts.start()
+    ti.join(timeout=120); ts.join(timeout=120)
+    assert not errs, errs
+    final = eng.query(plan).rows()
+    assert final[0][1][1] == float(n_batches * rows)
+    # stats landed: a selective predicate skips provably-excluded batches
+    hi = eng.query(abi.make_plan(
+        table=t, preds=[dict(col=0, is_double=True, lo=float(n_batches + 5))],
+        aggs=[("count", [])]))
+    res = hi.result()
+    assert res.batches_skipped == res.batches_seen  # all excluded by bounds
