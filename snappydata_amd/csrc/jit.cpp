@@ -138,6 +138,14 @@ __device__ __forceinline__ u64 mix64(u64 x) {
    * multi-aggregate small-slot shapes. */
   const int wbin_pre = grouped && !lds_mode && !glob_mode && NA <= 2 &&
                        nslots * (NA + 1) >= 12;
+  /* predicate folding: evaluate every predicate AT STAGE TIME from the
+   * just-loaded registers and ballot the verdicts into a 16-word LDS
+   * bitmap (CHUNK rows); the row pass then reads ONE broadcast bit per
+   * row instead of re-reading each predicate column from LDS and
+   * comparing — Q6 drops 3 LDS reads + 6 compares per row, Q1 one read +
+   * two compares.  Bit layout: word = (row>>7)*2 + (row&1), bit =
+   * (row>>1)&63 — exactly what per-register-row wave ballots produce. */
+  const int fold_preds = !sparse_mode && (p->npreds_d + p->npreds_i) > 0;
   emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, %d)\n"
            "void jit_scan(const sn_dev_batch *__restrict__ batches,\n"
            "              const sn_dev_tile *__restrict__ tiles, int ntiles,\n"
@@ -200,8 +208,11 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   }();
   const int fuse_mode = fuse_env && wbin_mode && p->jkeys && p->jlut &&
                         p->jmode == 1 && p->npreds_d + p->npreds_i == 0;
-  if (!fuse_mode)
+  if (!fuse_mode) {
     emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
+    if (fold_preds)
+      o += "  __shared__ unsigned long long sbits[16];\n";
+  }
   if (!lds_mode && !wbin_mode && !glob_mode)
     emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
           grouped ? nslots * (NA + 1) : 2 * na_t + 1);
@@ -286,10 +297,9 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emit_load_pre(base_expr, ind, "st");
   };
 
-  if (fuse_mode) {
-    /* value of column c at staged-register row j (j: 0 -> st_0.x row
-     * base+2*tid, 1 -> st_0.y, 2 -> st_1.x row base+2*(tid+WG), 3 -> .y) */
-    auto rexpr = [&](int c, int j) -> std::string {
+  /* value of column c at staged-register row j (j: 0 -> st_0.x row
+   * base+2*tid, 1 -> st_0.y, 2 -> st_1.x row base+2*(tid+WG), 3 -> .y) */
+  auto rexpr = [&](int c, int j) -> std::string {
       char buf[96];
       const int h = j >> 1, lo = !(j & 1);
       const char *xy = (j & 1) ? "y" : "x";
@@ -330,7 +340,9 @@ __device__ __forceinline__ u64 mix64(u64 x) {
                                : "(i64)(short)(st%d_%d >> 16)", c, h); break;
       }
       return buf;
-    };
+  };
+
+  if (fuse_mode) {
     auto va_fused = [&](int a, const std::string (&f)[3]) -> std::string {
       const sn_dev_agg &A = p->aggs[a];
       if (A.nf < 1) return "1.0";
@@ -647,8 +659,32 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emit_lut_probe(b3, "spay[2 * (tid + WG) + 1]", "        ");
   }
   emit_write("        ");
+  if (fold_preds) {
+    /* evaluate every predicate from the staged registers, one wave ballot
+     * per register row into the chunk bitmap */
+    for (int j = 0; j < 4; j++) {
+      emitf(o, "        { int okp = 1;\n");
+      for (int i = 0; i < p->npreds_d; i++) {
+        emitf(o, "          { const double x = %s;\n"
+                 "            okp &= (x >= pd%d_lo) & (x <= pd%d_hi); }\n",
+              rexpr(p->preds_d[i].cslot, j).c_str(), i, i);
+      }
+      for (int i = 0; i < p->npreds_i; i++) {
+        emitf(o, "          { const i64 x = %s;\n"
+                 "            okp &= (x >= pi%d_lo) & (x <= pi%d_hi); }\n",
+              kexpr_reg(p->preds_i[i].cslot, j).c_str(), i, i);
+      }
+      const int w0 = (j >= 2 ? 1 : 0);   /* +4 blocks for st_1 rows */
+      emitf(o, "          const u64 bl = __ballot(okp);\n"
+               "          if ((tid & 63) == 0) sbits[((tid >> 6) + %d) * 2 + %d] = bl;\n"
+               "        }\n", w0 * 4, j & 1);
+    }
+  }
   o += "      } else {\n"
        "        /* scalar tail conversion */\n";
+  if (fold_preds)
+    o += "        for (int i = tid; i < 16; i += WG) sbits[i] = 0ull;\n"
+         "        __syncthreads();\n";
   for (int c = 0; c < NC; c++) {
     int k = kinds[c];
     emitf(o, "        for (int r = tid; r < rows; r += WG) sval[%d][r] = ", c);
@@ -668,6 +704,23 @@ __device__ __forceinline__ u64 mix64(u64 x) {
       default: /* I16 */
         emitf(o, "(double)((const GAS short *)(body%d))[base + r];\n", c); break;
     }
+  }
+  if (fold_preds) {
+    /* same thread wrote these sval rows above — no barrier needed */
+    o += "        for (int r = tid; r < rows; r += WG) {\n"
+         "          int okp = 1;\n";
+    for (int i = 0; i < p->npreds_d; i++)
+      emitf(o, "          { const double x = sval[%d][r];\n"
+               "            okp &= (x >= pd%d_lo) & (x <= pd%d_hi); }\n",
+            p->preds_d[i].cslot, i, i);
+    for (int i = 0; i < p->npreds_i; i++)
+      emitf(o, "          { const i64 x = __double_as_longlong(sval[%d][r]);\n"
+               "            okp &= (x >= pi%d_lo) & (x <= pi%d_hi); }\n",
+            p->preds_i[i].cslot, i, i);
+    o += "          if (okp)\n"
+         "            atomicOr(&sbits[((r >> 7) << 1) | (r & 1)],\n"
+         "                     1ull << ((r >> 1) & 63));\n"
+         "        }\n";
   }
   o += "      }\n"
        "      __syncthreads();\n"
@@ -694,24 +747,33 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   }
 
   /* fused row pass */
-  o += "#pragma unroll 2\n"
-       "      for (int k = 0; k < CHUNK / WG; k++) {\n"
-       "        const int r = tid + k * WG;\n"
-       "        int ok = r < rows;\n";
+  if (fold_preds)
+    o += "#pragma unroll 2\n"
+         "      for (int k = 0; k < CHUNK / WG; k++) {\n"
+         "        const int r = tid + k * WG;\n"
+         "        int ok = (int)((sbits[((r >> 7) << 1) | (r & 1)]\n"
+         "                        >> ((r >> 1) & 63)) & 1ull);\n";
+  else
+    o += "#pragma unroll 2\n"
+         "      for (int k = 0; k < CHUNK / WG; k++) {\n"
+         "        const int r = tid + k * WG;\n"
+         "        int ok = r < rows;\n";
   if (has_del)
     o += "        if (del) {\n"
          "          const int gr = base + r;\n"
          "          ok &= (int)(~(del[(u64)gr >> 6] >> (gr & 63)) & 1ull);\n"
          "        }\n";
-  for (int i = 0; i < p->npreds_d; i++) {
-    emitf(o, "        { const double x = sval[%d][r];\n"
-             "          ok &= (x >= pd%d_lo) & (x <= pd%d_hi); }\n",
-          p->preds_d[i].cslot, i, i);
-  }
-  for (int i = 0; i < p->npreds_i; i++) {
-    emitf(o, "        { const i64 x = __double_as_longlong(sval[%d][r]);\n"
-             "          ok &= (x >= pi%d_lo) & (x <= pi%d_hi); }\n",
-          p->preds_i[i].cslot, i, i);
+  if (!fold_preds) {
+    for (int i = 0; i < p->npreds_d; i++) {
+      emitf(o, "        { const double x = sval[%d][r];\n"
+               "          ok &= (x >= pd%d_lo) & (x <= pd%d_hi); }\n",
+            p->preds_d[i].cslot, i, i);
+    }
+    for (int i = 0; i < p->npreds_i; i++) {
+      emitf(o, "        { const i64 x = __double_as_longlong(sval[%d][r]);\n"
+               "          ok &= (x >= pi%d_lo) & (x <= pi%d_hi); }\n",
+            p->preds_i[i].cslot, i, i);
+    }
   }
   o += "        if (__popcll(__ballot(ok)) == 0) continue;\n";
   if (p->jkeys) {
