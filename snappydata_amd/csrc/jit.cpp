@@ -140,12 +140,19 @@ __device__ __forceinline__ u64 mix64(u64 x) {
                        nslots * (NA + 1) >= 12;
   /* predicate folding: evaluate every predicate AT STAGE TIME from the
    * just-loaded registers and ballot the verdicts into a 16-word LDS
-   * bitmap (CHUNK rows); the row pass then reads ONE broadcast bit per
-   * row instead of re-reading each predicate column from LDS and
-   * comparing — Q6 drops 3 LDS reads + 6 compares per row, Q1 one read +
-   * two compares.  Bit layout: word = (row>>7)*2 + (row&1), bit =
-   * (row>>1)&63 — exactly what per-register-row wave ballots produce. */
-  const int fold_preds = !sparse_mode && (p->npreds_d + p->npreds_i) > 0;
+   * bitmap; the row pass then reads ONE broadcast bit per row instead of
+   * re-reading each predicate column from LDS and comparing.  BUILT AND
+   * MEASURED SLOWER (same-box A/B, 2 rounds each): Q6 SF100 74.9 vs
+   * 75.8%, Q1 SF100 71.9 vs 73.6% — the saved LDS reads/compares were
+   * already hidden under HBM latency, while the stage-time ballots add
+   * v_cmp+SGPR traffic on the critical staging path.  Kept behind
+   * SN_JIT_FOLD=1; default off. */
+  static const int fold_env = [] {
+    const char *v = getenv("SN_JIT_FOLD");
+    return v && v[0] == '1';
+  }();
+  const int fold_preds = fold_env && !sparse_mode &&
+                         (p->npreds_d + p->npreds_i) > 0;
   emitf(o, "extern \"C\" __global__ __launch_bounds__(WG, %d)\n"
            "void jit_scan(const sn_dev_batch *__restrict__ batches,\n"
            "              const sn_dev_tile *__restrict__ tiles, int ntiles,\n"
