@@ -412,3 +412,34 @@ def test_order_by_topk_epilogue(eng):
     asc = q2.rows()
     counts_sorted = sorted(cnts.values())
     assert [v[0] for _, v in asc] == [float(c) for c in counts_sorted[:5]]
+
+
+@pytest.mark.gpu
+def test_mutate_after_raw_ingest(eng):
+    """UPDATE/DELETE against a batch that arrived through the f2 RAW path:
+    patch materialization writes into the device-built body and the
+    device-computed stats are superseded."""
+    n = 30_000
+    rng = np.random.default_rng(149)
+    v = rng.random(n)
+    t = eng.table_define("trawmut", [(abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": v}], n, batch_rows=n)   # uuid 0, bucket 0
+    pos = np.arange(0, n, 13, dtype=np.int32)
+    newv = v[pos] + 10.0
+    d1 = se.encode_update_delta(abi.T_DOUBLE, pos, n, newv)
+    dels = np.arange(0, n, 97, dtype=np.int32)
+    eng.batch_mutate(t, 0, 0, deltas=[(d1, None)],
+                     delete_mask=se.encode_delete_mask(dels, n))
+    merged = v.copy()
+    merged[pos] = newv
+    alive = np.ones(n, dtype=bool)
+    alive[dels] = False
+    q = eng.query(abi.make_plan(table=t,
+                                aggs=[("sum", [(0, 0.0, 1.0)]),
+                                      ("max", [(0, 0.0, 1.0)]),
+                                      ("count", [])]))
+    rows = q.rows()
+    exp = merged[alive].sum()
+    assert abs(rows[0][1][0] - exp) <= 1e-9 * max(1.0, exp)
+    assert rows[0][1][1] == merged[alive].max()
+    assert rows[0][1][2] == float(alive.sum())
