@@ -474,9 +474,14 @@ extern "C" int32_t sn_dim_from_table(sn_engine *e, int32_t dim, int32_t table,
   if (has_attr) {
     if (attr_col >= nc || t->schema[attr_col].dtype != SN_TYPE_STRING)
       return fail(SN_ERR_UNSUPPORTED, "attr col must be a string column");
+    if (t->schema[attr_col].nullable)
+      return fail(SN_ERR_UNSUPPORTED,
+                  "nullable attribute columns not supported in join build");
     if (t->gdict_maxlen[attr_col] > SN_KEY_MAX - 1)
       return fail(SN_ERR_UNSUPPORTED, "attr exceeds the group-key limit");
   }
+  /* (key columns may be nullable: null keys never join and are masked at
+   * build time via the validity bitmap) */
   if (t->total_rows > (1ll << 23))
     return fail(SN_ERR_UNSUPPORTED,
                 "build side exceeds 2^23 rows (HashJoinSize-class bound)");

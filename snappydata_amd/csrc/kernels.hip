@@ -1521,25 +1521,31 @@ void k_join_build(sn_dev_plan plan,
         if (key == LLONG_MIN) { atomicOr((int *)&flags[0], 1); continue; }
         unsigned h = (unsigned)mix64((unsigned long long)key) & mask;
         int done = 0;
+        /* payload publication: the inserter CASes the key then atomically
+         * exchanges the payload over the INT32_MIN sentinel; a concurrent
+         * duplicate spins (bounded) until the payload is visible before
+         * checking for a conflict */
         for (unsigned it = 0; it <= mask && !done; ++it) {
-          const long long k0 = hk[h];
-          if (k0 == key) {
-            /* duplicate build key: allowed iff payload agrees (the dim
-             * contract requires unique keys; equal rows are idempotent) */
-            if (hp[h] != pay) atomicOr((int *)&flags[0], 1);
-            done = 1;
-            break;
-          }
+          long long k0 = hk[h];
           if (k0 == LLONG_MIN) {
-            const long long old = (long long)atomicCAS(
+            k0 = (long long)atomicCAS(
                 (unsigned long long *)&hk[h], (unsigned long long)LLONG_MIN,
                 (unsigned long long)key);
-            if (old == LLONG_MIN || old == key) {
-              if (old == LLONG_MIN) hp[h] = pay;
-              else if (hp[h] != pay) atomicOr((int *)&flags[0], 1);
+            if (k0 == LLONG_MIN) {              /* we inserted */
+              (void)atomicExch((int *)&hp[h], pay);
               done = 1;
               break;
             }
+          }
+          if (k0 == key) {
+            int pv;
+            int spins = 0;
+            do {
+              pv = atomicOr((int *)&hp[h], 0);  /* atomic read */
+            } while (pv == INT_MIN && ++spins < (1 << 20));
+            if (pv != pay) atomicOr((int *)&flags[0], 1);
+            done = 1;
+            break;
           }
           h = (h + 1) & mask;
         }
@@ -1575,6 +1581,10 @@ extern "C" int sn_launch_join_build(const sn_dev_plan *plan,
     long long cap = 1ll << cap_log2;
     hipLaunchKernelGGL(k_fill_i64, dim3((unsigned)((cap + 255) / 256)),
                        dim3(256), 0, s, hk, cap, LLONG_MIN);
+    /* payloads init to the INT32_MIN publication sentinel (two per i64) */
+    hipLaunchKernelGGL(k_fill_i64, dim3((unsigned)((cap / 2 + 255) / 256)),
+                       dim3(256), 0, s, (long long *)hp, cap / 2,
+                       (long long)0x8000000080000000ull);
   }
   int grid;
   if (ntiles <= 0) return (int)hipGetLastError();
