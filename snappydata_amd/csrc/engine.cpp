@@ -2521,7 +2521,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       /* query-compiled twin first (plan structure compile-time, capacity
        * tokenized); any miss falls back to the interpreted hash kernel */
       void *jfn = nullptr;
-      if (e->jit && !q->mm && !dp.pac && q->dev_naggs <= 12 &&
+      if (e->jit && q->dev_naggs <= 12 &&
           (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first))) {
         const int *jk = hit ? hit->jit_kinds : jit_kinds;
         int jdel = hit ? hit->jit_del : (jit_any_del ? 1 : 0);
@@ -2659,16 +2659,18 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
      * interpreted runtime-plan kernel.  Any miss falls back, still on GPU. */
     void *jfn = nullptr;
     bool jit_shape_ok;
-    if (dp.nslots <= 1) jit_shape_ok = dp.naggs <= 12;
+    if (dp.nslots <= 1 && !dp.pac) jit_shape_ok = dp.naggs <= 12;
     else if (dp.nslots <= 8) jit_shape_ok = dp.naggs <= 6;
     else if (dp.nslots <= 1024)
       /* LDS-accumulator mode: static shared = LDS image + gacc must fit */
       jit_shape_ok = (size_t)dp.nused * 8192 +
-                     (size_t)dp.nslots * (dp.naggs + 1) * 8 + 1024 <= 160 * 1024;
+                     (size_t)dp.nslots * na1 * 8 + 1024 <= 160 * 1024;
     else
       /* global-atomic mode: only the LDS image constrains */
       jit_shape_ok = dp.nslots <= SN_BIG_GROUP_CAP;
-    if (e->jit && jit_shape_ok && !q->mm &&
+    /* pac from ACTUAL nulls never reaches here (jit_ok excludes null
+     * batches); pac from MIN/MAX compiles op-aware kernels */
+    if (e->jit && jit_shape_ok &&
         (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first))) {
       const int *jk = hit ? hit->jit_kinds : jit_kinds;
       int jdel = hit ? hit->jit_del : (jit_any_del ? 1 : 0);
@@ -2686,7 +2688,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       /* >16-slot scratch rows are wide; mirror the interpreted grid cap
        * (scratch was sized with the same bound above) */
       if (dp.nslots > 16 && jgrid > SN_GRID_BIGSLOT) jgrid = SN_GRID_BIGSLOT;
-      int naggs1 = dp.nslots <= 1 ? 0 : na1;   /* (JIT never runs pac) */
+      int naggs1 = (dp.nslots <= 1 && !dp.pac) ? 0 : na1;
       rc = sn_jit_launch(jfn, jgrid, (const sn_dev_batch *)db_dev,
                          (const sn_dev_tile *)tl_dev, ntiles, e->scratch,
                          dp.jkeys, dp.jpayload, dp.jlut,

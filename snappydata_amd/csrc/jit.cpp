@@ -75,7 +75,17 @@ static std::string gen_source(const sn_dev_plan *p, const int *kinds,
    * capacity is tokenized (read from the device plan) so grow-and-retry
    * reuses one compiled kernel */
   const int sparse_mode = p->sparse != 0;
-  const int grouped = nslots > 1 || sparse_mode;
+  /* pac: per-agg-count accumulator rows ([sums][counts][rowcount]).  In
+   * the JIT this occurs only for MIN/MAX plans (null-carrying batches are
+   * never JIT-eligible, so counts == rowcount and the count cells simply
+   * mirror the rowcount); mm plans route through the grouped layouts,
+   * keyless included (one slot). */
+  const int pac_mode = p->pac != 0;
+  int mm_any = 0;
+  for (int a = 0; a < NA; a++) mm_any |= p->aggs[a].op != 0;
+  const int NS = nslots < 1 ? 1 : nslots;
+  const int NA1 = pac_mode ? 2 * NA + 1 : NA + 1;
+  const int grouped = nslots > 1 || sparse_mode || pac_mode;
   std::string o;
   o += R"(
 typedef double double2_t __attribute__((ext_vector_type(2)));
@@ -117,6 +127,20 @@ __device__ __forceinline__ double wsum(double x) {
   for (int off = 32; off > 0; off >>= 1) x += __shfl_down(x, off, 64);
   return x;
 }
+__device__ __forceinline__ u64 f64ord(double x) {
+  u64 b = (u64)__double_as_longlong(x);
+  return (b & 0x8000000000000000ull) ? ~b : (b | 0x8000000000000000ull);
+}
+__device__ __forceinline__ double wmin(double x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmin(x, __shfl_down(x, off, 64));
+  return x;
+}
+__device__ __forceinline__ double wmax(double x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmax(x, __shfl_down(x, off, 64));
+  return x;
+}
 __device__ __forceinline__ u64 mix64(u64 x) {
   x += 0x9E3779B97f4A7C15ull;
   x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ull;
@@ -137,7 +161,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
    * cost, so the register accumulators stay the measured optimum for
    * multi-aggregate small-slot shapes. */
   const int wbin_pre = grouped && !lds_mode && !glob_mode && NA <= 2 &&
-                       nslots * (NA + 1) >= 12;
+                       !mm_any && nslots * (NA + 1) >= 12;
   /* predicate folding: evaluate every predicate AT STAGE TIME from the
    * just-loaded registers and ballot the verdicts into a 16-word LDS
    * bitmap; the row pass then reads ONE broadcast bit per row instead of
@@ -222,7 +246,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   }
   if (!lds_mode && !wbin_mode && !glob_mode)
     emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
-          grouped ? nslots * (NA + 1) : 2 * na_t + 1);
+          grouped ? NS * NA1 : 2 * na_t + 1);
   o += "  const int tid = threadIdx.x;\n";
 
   /* accumulators */
@@ -241,13 +265,27 @@ __device__ __forceinline__ u64 mix64(u64 x) {
      * same-slot atomics from different XCDs never contend and stay in the
      * local L2; k_reduce folds the 8 copies */
     emitf(o, "  GAS double *gacc = (GAS double *)(u64)out"
-             " + (u64)(blockIdx.x & 7) * %d;\n", nslots * (NA + 1));
+             " + (u64)(blockIdx.x & 7) * %d;\n", NS * NA1);
   } else if (lds_mode) {
-    /* block-level LDS accumulator, zeroed once, flushed once at the end */
-    emitf(o, "  __shared__ __attribute__((aligned(16))) double gacc[%d];\n"
-             "  for (int i = tid; i < %d; i += WG) gacc[i] = 0.0;\n"
-             "  __syncthreads();\n",
-          nslots * (NA + 1), nslots * (NA + 1));
+    /* block-level LDS accumulator, initialized once (min/max cells take
+     * their ord-encoding identities), flushed once at the end */
+    emitf(o, "  __shared__ __attribute__((aligned(16))) double gacc[%d];\n",
+          NS * NA1);
+    if (!mm_any) {
+      emitf(o, "  for (int i = tid; i < %d; i += WG) gacc[i] = 0.0;\n",
+            NS * NA1);
+    } else {
+      emitf(o, "  for (int i = tid; i < %d; i += WG) {\n"
+               "    const int a = i %% %d;\n"
+               "    double iv = 0.0;\n", NS * NA1, NA1);
+      for (int a = 0; a < NA; a++)
+        if (p->aggs[a].op != 0)
+          emitf(o, "    if (a == %d) iv = __longlong_as_double(%s);\n", a,
+                p->aggs[a].op == 1 ? "0xFFF8000000000000ll"
+                                   : "0x0000000000000000ll");
+      o += "    gacc[i] = iv;\n  }\n";
+    }
+    o += "  __syncthreads();\n";
   } else if (wbin_mode) {
     /* 16 bins of 16 lanes each: ~2-way LDS atomic conflicts instead of the
      * 8-way a per-wave bin sees (measured: issue-stall bound, 58% of wave
@@ -256,12 +294,18 @@ __device__ __forceinline__ u64 mix64(u64 x) {
              "  for (int i = tid; i < 16 * %d; i += WG)\n"
              "    ((double *)wbin)[i] = 0.0;\n"
              "  __syncthreads();\n",
-          nslots * (NA + 1), nslots * (NA + 1));
+          NS * NA1, NS * NA1);
   } else if (grouped) {
-    emitf(o, "  double sums[%d][%d]; double rc[%d];\n", nslots, NA, nslots);
-    emitf(o, "#pragma unroll\n  for (int s = 0; s < %d; s++) { rc[s] = 0;\n"
-             "#pragma unroll\n    for (int a = 0; a < %d; a++) sums[s][a] = 0; }\n",
-          nslots, NA);
+    emitf(o, "  double sums[%d][%d]; double rc[%d];\n", NS, NA, NS);
+    emitf(o, "#pragma unroll\n  for (int s = 0; s < %d; s++) {\n"
+             "    rc[s] = 0;\n", NS);
+    for (int a = 0; a < NA; a++) {
+      const int op = p->aggs[a].op;
+      if (op == 0) emitf(o, "    sums[s][%d] = 0.0;\n", a);
+      else emitf(o, "    sums[s][%d] = __longlong_as_double(%s);\n", a,
+                 op == 1 ? "0x7FF0000000000000ll" : "0xFFF0000000000000ll");
+    }
+    o += "  }\n";
   } else {
     emitf(o, "  double sums[%d], cnts[%d], rcnt = 0.0;\n", NA, NA);
     emitf(o, "#pragma unroll\n  for (int a = 0; a < %d; a++) { sums[a] = 0; cnts[a] = 0; }\n", NA);
@@ -850,6 +894,9 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   } else if (grouped) {
     if (p->jkeys && p->jmode == 1) {
       o += "        const int slot = pay > 0 ? pay : 0;\n";
+    } else if (p->ngroup == 0) {
+      /* keyless plan in the grouped (pac) layout: one slot */
+      o += "        const int slot = 0;\n";
     } else {
       /* slot = (v0 - base0)*mul0 + (v1 - base1), all literals; dictionary
        * keys arrive premultiplied (base 0, mul 1) so this folds to the
@@ -889,9 +936,17 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emitf(o, "        if (ok) {\n"
              "          %sdouble *row = &gacc[(u64)slot * %d];\n"
              "          atomicAdd((double *)&row[%d], 1.0);\n",
-          glob_mode ? "GAS " : "", NA + 1, NA);
-    for (int a = 0; a < NA; a++)
-      emitf(o, "          atomicAdd((double *)&row[%d], va%d);\n", a, a);
+          glob_mode ? "GAS " : "", NA1, NA1 - 1);
+    for (int a = 0; a < NA; a++) {
+      const int op = p->aggs[a].op;
+      if (op == 0)
+        emitf(o, "          atomicAdd((double *)&row[%d], va%d);\n", a, a);
+      else
+        emitf(o, "          %s((unsigned long long *)&row[%d], f64ord(va%d));\n",
+              op == 1 ? "atomicMin" : "atomicMax", a, a);
+      if (pac_mode)
+        emitf(o, "          atomicAdd((double *)&row[%d], 1.0);\n", NA + a);
+    }
     o += "        }\n";
   } else if (wbin_mode) {
     emitf(o, "        if (ok) {\n"
@@ -906,10 +961,18 @@ __device__ __forceinline__ u64 mix64(u64 x) {
      * VALU op per aggregate instead of two */
     emitf(o, "#pragma unroll\n        for (int s = 0; s < %d; s++) {\n"
              "          const double m = (ok && slot == s) ? 1.0 : 0.0;\n"
-             "          rc[s] += m;\n", nslots);
-    for (int a = 0; a < NA; a++)
-      emitf(o, "          sums[s][%d] = __builtin_fma(m, va%d, sums[s][%d]);\n",
-            a, a, a);
+             "          rc[s] += m;\n", NS);
+    for (int a = 0; a < NA; a++) {
+      const int op = p->aggs[a].op;
+      if (op == 0)
+        emitf(o, "          sums[s][%d] = __builtin_fma(m, va%d, sums[s][%d]);\n",
+              a, a, a);
+      else
+        emitf(o, "          sums[s][%d] = %s(sums[s][%d],\n"
+                 "              (ok && slot == s) ? va%d : __longlong_as_double(%s));\n",
+              a, op == 1 ? "fmin" : "fmax", a, a,
+              op == 1 ? "0x7FF0000000000000ll" : "0xFFF0000000000000ll");
+    }
     o += "        }\n";
   } else {
     for (int a = 0; a < NA; a++) {
@@ -925,7 +988,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
        "  }\n";
 
   /* block reduce into LDS bacc then scratch row */
-  int nv = grouped ? nslots * (NA + 1) : 2 * na_t + 1;
+  int nv = grouped ? NS * NA1 : 2 * na_t + 1;
   if (glob_mode) {
     /* nothing to flush — the HBM accumulator holds the single partial set */
     o += "}\n";
@@ -950,18 +1013,42 @@ __device__ __forceinline__ u64 mix64(u64 x) {
              "}\n", nv, nv);
     return o;
   }
-  emitf(o, "  for (int i = tid; i < %d; i += WG) bacc[i] = 0.0;\n"
-           "  __syncthreads();\n", nv);
-  if (grouped) {
-    emitf(o, "#pragma unroll\n  for (int s = 0; s < %d; s++) {\n", nslots);
+  if (!mm_any || !grouped) {
+    emitf(o, "  for (int i = tid; i < %d; i += WG) bacc[i] = 0.0;\n"
+             "  __syncthreads();\n", nv);
+  } else {
+    emitf(o, "  for (int i = tid; i < %d; i += WG) {\n"
+             "    const int a = i %% %d;\n"
+             "    double iv = 0.0;\n", nv, NA1);
     for (int a = 0; a < NA; a++)
-      emitf(o, "    { double x = wsum(sums[s][%d]);\n"
-               "      if ((tid & 63) == 0 && x != 0.0) atomicAdd(&bacc[s * %d + %d], x); }\n",
-            a, NA + 1, a);
+      if (p->aggs[a].op != 0)
+        emitf(o, "    if (a == %d) iv = __longlong_as_double(%s);\n", a,
+              p->aggs[a].op == 1 ? "0xFFF8000000000000ll"
+                                 : "0x0000000000000000ll");
+    o += "    bacc[i] = iv;\n  }\n  __syncthreads();\n";
+  }
+  if (grouped) {
+    emitf(o, "#pragma unroll\n  for (int s = 0; s < %d; s++) {\n", NS);
+    for (int a = 0; a < NA; a++) {
+      const int op = p->aggs[a].op;
+      if (op == 0)
+        emitf(o, "    { double x = wsum(sums[s][%d]);\n"
+                 "      if ((tid & 63) == 0 && x != 0.0) atomicAdd(&bacc[s * %d + %d], x); }\n",
+              a, NA1, a);
+      else
+        emitf(o, "    { double x = %s(sums[s][%d]);\n"
+                 "      if ((tid & 63) == 0)\n"
+                 "        %s((unsigned long long *)&bacc[s * %d + %d], f64ord(x)); }\n",
+              op == 1 ? "wmin" : "wmax", a,
+              op == 1 ? "atomicMin" : "atomicMax", NA1, a);
+    }
     emitf(o, "    { double x = wsum(rc[s]);\n"
-             "      if ((tid & 63) == 0 && x != 0.0) atomicAdd(&bacc[s * %d + %d], x); }\n",
-          NA + 1, NA);
-    o += "  }\n";
+             "      if ((tid & 63) == 0 && x != 0.0) {\n"
+             "        atomicAdd(&bacc[s * %d + %d], x);\n", NA1, NA1 - 1);
+    if (pac_mode)
+      for (int a = 0; a < NA; a++)
+        emitf(o, "        atomicAdd(&bacc[s * %d + %d], x);\n", NA1, NA + a);
+    o += "      } }\n  }\n";
   } else {
     for (int a = 0; a < NA; a++) {
       emitf(o, "  { double x = wsum(sums[%d]);\n"
