@@ -1768,6 +1768,7 @@ struct sn_query {
   bool mm = false;                      /* plan has MIN/MAX aggregates */
   std::vector<long long> sparse_keys;   /* compacted group keys */
   std::vector<double> sparse_rows;      /* [n][na1] accumulator rows */
+  std::vector<double> sparse_null_row;  /* NULL-key group accumulator */
   bool gint[2] = { false, false };      /* integer group key (stats-ranged) */
   int64_t gmin[2] = { 0, 0 };           /* integer key minimum (slot base) */
   int gnull1 = -1, gnull2 = -1;         /* null slot index per group col (-1: none) */
@@ -1845,7 +1846,7 @@ static int ensure_sparse_ws(sn_engine *e, int cap_log2, int naggs1) {
     e->hws_acc_bytes = 0;
     if (!e->hws_keys || !e->hws_okeys) return SN_ERR_NOMEM;
   }
-  size_t accb = ((1ull << e->hws_cap_log2) + 1) * (size_t)naggs1 * 8;
+  size_t accb = ((1ull << e->hws_cap_log2) + 2) * (size_t)naggs1 * 8;
   if (e->hws_acc_bytes < accb) {
     e->hws_acc = (double *)e->arena.alloc(accb);
     e->hws_orows = (double *)e->arena.alloc(accb);
@@ -1903,8 +1904,12 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
            "group-by supports dictionary string and int16/int32/int64 key columns");
       return nullptr;
     }
-    if (int_key && t->schema[c].nullable) {
-      fail(SN_ERR_UNSUPPORTED, "nullable integer group keys not supported");
+    if (int_key && t->schema[c].nullable && plan->ngroup != 1) {
+      /* a NULL in one key of a pair is a distinct composite group the
+       * packed sparse key cannot represent */
+      fail(SN_ERR_UNSUPPORTED,
+           "nullable integer group keys supported for single-column "
+           "group-bys only");
       return nullptr;
     }
     if (use_col(c) < 0) { fail(SN_ERR_BADARG, "bad group col"); return nullptr; }
@@ -2015,7 +2020,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         caps[i] = (int)t->gdict[c].size() + (t->schema[c].nullable ? 1 : 0);
         continue;
       }
-      int64_t span = gdt == SN_TYPE_INT64 ? -1 : int_key_span(c, &q->gmin[i]);
+      int64_t span = (gdt == SN_TYPE_INT64 || t->schema[c].nullable)
+                         ? -1 : int_key_span(c, &q->gmin[i]);
       if (span < 0 || span > SN_BIG_GROUP_CAP) { col_sparse[i] = true; continue; }
       q->gint[i] = true;
       caps[i] = (int)span;
@@ -2508,13 +2514,13 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       }
       q->owned.push_back({ dps_dev, sizeof(dps) });
       if (hipMemsetAsync(e->hws_keys, 0xff, cap * 8, e->stream) != hipSuccess ||
-          hipMemsetAsync(e->hws_acc, 0, (cap + 1) * (size_t)naggs1 * 8,
+          hipMemsetAsync(e->hws_acc, 0, (cap + 2) * (size_t)naggs1 * 8,
                          e->stream) != hipSuccess ||
           hipMemsetAsync(e->hws_flags, 0, 16, e->stream) != hipSuccess) {
         fail(SN_ERR_GENERIC, "sparse workspace zero"); return nullptr;
       }
       if (q->mm &&
-          sn_launch_acc_init(e->hws_acc, (long long)cap + 1, q->dev_naggs,
+          sn_launch_acc_init(e->hws_acc, (long long)cap + 2, q->dev_naggs,
                              naggs1, dps_dev, e->stream) != 0) {
         fail(SN_ERR_GENERIC, "sparse min/max init"); return nullptr;
       }
@@ -2583,6 +2589,12 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       }
       int32_t ngrp = 0;
       (void)hipMemcpy(&ngrp, e->hws_flags + 1, 4, hipMemcpyDeviceToHost);
+      /* the NULL-key group accumulates in the extra row at cap+1 (the
+       * compaction covers rows [0, cap]) */
+      q->sparse_null_row.assign((size_t)naggs1, 0.0);
+      (void)hipMemcpy(q->sparse_null_row.data(),
+                      e->hws_acc + (size_t)(cap + 1) * naggs1,
+                      (size_t)naggs1 * 8, hipMemcpyDeviceToHost);
       q->sparse_keys.resize((size_t)ngrp);
       q->sparse_rows.resize((size_t)ngrp * naggs1);
       if (ngrp > 0) {
@@ -2763,15 +2775,12 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
      * partial-block and result formats stay shared). */
     const int naggs1 = (q->pac ? 2 : 1) * q->dev_naggs + 1;
     const bool packed = p.ngroup == 2;
-    for (size_t i = 0; i < q->sparse_keys.size(); i++) {
-      const double *row = &q->sparse_rows[i * naggs1];
-      double rowcount = row[naggs1 - 1];
-      if (rowcount == 0.0) continue;
+    auto fill = [&](const double *row) -> GroupOut {
       GroupOut g;
-      g.rowcount = rowcount;
+      g.rowcount = row[naggs1 - 1];
       for (int a = 0; a < p.naggs; a++) {
         int di = q->agg_map[a];
-        double s = di < 0 ? rowcount : row[di];
+        double s = di < 0 ? g.rowcount : row[di];
         int k = p.aggs[a].kind;
         if (di >= 0 && (k == SN_AGG_MIN || k == SN_AGG_MAX)) {
           /* sparse rows bypass k_reduce: decode the ord-u64 cell here */
@@ -2780,9 +2789,15 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
           s = sn_ord_f64_h(o);
         }
         g.sums[a] = s;
-        g.counts[a] = (di < 0 || !q->pac) ? rowcount
+        g.counts[a] = (di < 0 || !q->pac) ? g.rowcount
                                           : row[q->dev_naggs + di];
       }
+      return g;
+    };
+    for (size_t i = 0; i < q->sparse_keys.size(); i++) {
+      const double *row = &q->sparse_rows[i * naggs1];
+      if (row[naggs1 - 1] == 0.0) continue;
+      GroupOut g = fill(row);
       long long key = q->sparse_keys[i];
       if (packed) {
         g.keys[0] = std::to_string((int32_t)((unsigned long long)key >> 32));
@@ -2790,6 +2805,12 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
       } else {
         g.keys[0] = std::to_string(key);
       }
+      out->push_back(std::move(g));
+    }
+    if (!q->sparse_null_row.empty() &&
+        q->sparse_null_row[naggs1 - 1] > 0.0) {
+      GroupOut g = fill(q->sparse_null_row.data());
+      g.key_null[0] = true;               /* the NULL-key group */
       out->push_back(std::move(g));
     }
     return;

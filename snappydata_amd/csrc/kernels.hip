@@ -1429,10 +1429,20 @@ void k_grouped_hash(sn_dev_plan plan,
         if (w == 0) continue;
         const int m = (int)((w >> (tid & 63)) & 1ull);
         if (!m) continue;
-        const long long key = sparse_key(P, sval, r);
-        const int slot = hash_probe(key, plan.hkeys, plan.hcap_log2,
-                                    plan.hflags);
-        if (slot < 0) continue;                     /* overflow: host retries */
+        /* a NULL key routes to the dedicated null-group row at cap+1
+         * (nullable keys are single-column by the engine contract) */
+        int kvalid = 1;
+        if (!clean)
+          kvalid = (int)((svalid[(size_t)plan.gcol[0] * (CHUNK / 64) +
+                                 (r >> 6)] >> (r & 63)) & 1ull);
+        int slot;
+        if (!kvalid) {
+          slot = (1 << plan.hcap_log2) + 1;
+        } else {
+          const long long key = sparse_key(P, sval, r);
+          slot = hash_probe(key, plan.hkeys, plan.hcap_log2, plan.hflags);
+          if (slot < 0) continue;                   /* overflow: host retries */
+        }
         GAS double *row_acc = acc + (size_t)slot * na1;
         for (int a = 0; a < naggs; a++) {
           const sn_dev_agg &A = P->aggs[a];

@@ -235,3 +235,29 @@ def test_sparse_with_deletes_general_path(eng):
     ot.add_batch(n, cols, delete_mask=dmask)
     orows = ot.query_groups(po.make_plan(**plan_kw), nthreads=16)
     assert_rows_match(grows, orows, count_aggs={1})
+
+
+@pytest.mark.gpu
+def test_sparse_nullable_int64_keys(eng):
+    """Nullable int64 group keys (single-column): NULL keys form their own
+    group (Spark GROUP BY null semantics), accumulated in the dedicated
+    null row of the open-address table."""
+    n = 250_000
+    rng = np.random.default_rng(151)
+    keys = rng.integers(0, 2_000, n).astype(np.int64) * (1 << 32)
+    valid = (rng.random(n) > 0.15).astype(np.uint8)
+    w = rng.random(n)
+    cols = [po.encode(po.T_INT64, po.ENC_UNCOMPRESSED, keys, valid=valid),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
+    t = eng.table_define("tsparsenull", [(abi.T_INT64, True), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 1, 0, n, cols)
+    plan_kw = dict(group_cols=[0],
+                   aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_INT64, po.T_DOUBLE], nullable=[True, False])
+    ot.add_batch(n, cols)
+    orows = ot.query_groups(po.make_plan(**plan_kw), nthreads=16)
+    assert_rows_match(grows, orows, count_aggs={1})
+    # the NULL group exists and counts exactly the invalid rows
+    nulls = [v for k, v in grows if k[0] is None]
+    assert len(nulls) == 1 and nulls[0][1] == float((valid == 0).sum())
