@@ -152,6 +152,17 @@ typedef struct {
                            valid for the duration of the submit call */
   int32_t str_len;
   int32_t _pad3;
+  /* IN-list membership (the reference's Q12/Q19-class `col IN (...)`
+   * filters, compiled into the generated loop by ColumnTableScan):
+   * integer/date columns take in_i[in_n]; dictionary string columns take
+   * in_s/in_s_len[in_n] (resolved to dictionary ids at submit, like
+   * str_eq).  in_n == 0 disables; all pointers valid only during submit.
+   * An IN pred may not also carry range bounds. */
+  const int64_t *in_i;
+  const char *const *in_s;
+  const int32_t *in_s_len;
+  int32_t in_n;
+  int32_t _pad6;
 } sn_pred;
 
 /* one multiplicative factor of an aggregate input: (add + mul * col) */

@@ -1018,6 +1018,22 @@ static int eval_batch(const sno_table *t, const sno_batch *b, const sn_plan *p,
       const sn_pred *pr = &p->preds[i];
       int c = pr->col;
       if (val_null[c]) { row_ok = 0; break; }
+      if (pr->in_n > 0) {
+        /* IN-list membership (Q12/Q19-class) */
+        int hit = 0;
+        if (t->dtypes[c] == SN_TYPE_STRING) {
+          for (int32_t ii = 0; ii < pr->in_n && !hit; ii++)
+            if (pr->in_s && pr->in_s_len &&
+                val_slen[c] == pr->in_s_len[ii] &&
+                memcmp(val_s[c], pr->in_s[ii], (size_t)val_slen[c]) == 0)
+              hit = 1;
+        } else {
+          for (int32_t ii = 0; ii < pr->in_n && !hit; ii++)
+            if (pr->in_i && pr->in_i[ii] == val_i[c]) hit = 1;
+        }
+        if (!hit) row_ok = 0;
+        continue;
+      }
       if (pr->str_eq && pr->str_len > 0) {
         /* dictionary string equality (engine pushes this down to a dict-id
          * compare; the oracle compares the decoded bytes) */
@@ -1122,6 +1138,8 @@ static int eval_batch_fast(const sno_table *t, const sno_batch *b,
   if (p->npreds > SN_MAX_PREDS || p->naggs > SN_MAX_AGGS) return 0;
   for (int a = 0; a < p->naggs; a++)   /* MIN/MAX keep the generic path */
     if (p->aggs[a].kind == SN_AGG_MIN || p->aggs[a].kind == SN_AGG_MAX) return 0;
+  for (int i = 0; i < p->npreds; i++)  /* IN-lists keep the generic path */
+    if (p->preds[i].in_n > 0) return 0;
   const int has_join = p->join_dim != SN_JOIN_NONE && t->dim_hk != NULL;
   if (has_join) return 0;   /* join keeps the generic path */
 

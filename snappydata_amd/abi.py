@@ -45,7 +45,11 @@ class SnPred(C.Structure):
                 ("lo_strict", C.c_uint8), ("hi_strict", C.c_uint8),
                 ("_pad2", C.c_uint8 * 4),
                 ("str_eq", C.c_char_p), ("str_len", C.c_int32),
-                ("_pad3", C.c_int32)]
+                ("_pad3", C.c_int32),
+                ("in_i", C.POINTER(C.c_int64)),
+                ("in_s", C.POINTER(C.c_char_p)),
+                ("in_s_len", C.POINTER(C.c_int32)),
+                ("in_n", C.c_int32), ("_pad6", C.c_int32)]
 
 
 class SnFactor(C.Structure):
@@ -111,6 +115,20 @@ def make_plan(table=0, preds=(), group_cols=(), aggs=(), join=None):
             lit = pr["eq"] if isinstance(pr["eq"], bytes) else pr["eq"].encode()
             sp.str_eq = lit          # ctypes keeps the bytes alive via _objects
             sp.str_len = len(lit)
+        if "in" in pr:
+            vals = pr["in"]
+            sp.in_n = len(vals)
+            if vals and isinstance(vals[0], (bytes, str)):
+                lits = [v if isinstance(v, bytes) else v.encode() for v in vals]
+                arr = (C.c_char_p * len(lits))(*lits)
+                lens = (C.c_int32 * len(lits))(*[len(v) for v in lits])
+                sp.in_s = arr
+                sp.in_s_len = lens
+                p._keep = getattr(p, "_keep", []) + [arr, lens, lits]
+            else:
+                arr = (C.c_int64 * len(vals))(*[int(v) for v in vals])
+                sp.in_i = arr
+                p._keep = getattr(p, "_keep", []) + [arr]
     p.ngroup = len(group_cols)
     for i, c in enumerate(group_cols):
         p.group_cols[i] = c

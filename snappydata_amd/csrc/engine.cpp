@@ -1888,9 +1888,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     const sn_pred &pr = plan->preds[i];
     if (use_col(pr.col) < 0) { fail(SN_ERR_BADARG, "bad pred col"); return nullptr; }
     if (t->schema[pr.col].dtype == SN_TYPE_STRING &&
-        (!pr.str_eq || pr.str_len <= 0)) {
+        (!pr.str_eq || pr.str_len <= 0) && pr.in_n <= 0) {
       fail(SN_ERR_UNSUPPORTED,
-           "string predicate supports dictionary equality only (str_eq)");
+           "string predicate supports dictionary equality (str_eq) and "
+           "IN-lists (in_s) only");
       return nullptr;
     }
   }
@@ -2113,6 +2114,68 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     const sn_pred &s = plan->preds[i];
     sn_type_t dt = t->schema[s.col].dtype;
     int cslot = q->cslot_of_col[s.col];
+    if (s.in_n > 0) {
+      /* IN-list membership (Q12/Q19-class): integer values directly,
+       * dictionary strings resolved to premultiplied global ids (absent
+       * literals drop out).  Dense value spans build a bitmap LUT; wide
+       * spans keep a sorted list (binary-searched; interpreted kernels). */
+      if (dp.npreds_in >= 2) { fail(SN_ERR_BADARG, "too many IN predicates"); return nullptr; }
+      if (s.has_lo || s.has_hi) {
+        fail(SN_ERR_BADARG, "an IN predicate may not also carry range bounds");
+        return nullptr;
+      }
+      std::vector<int64_t> vals;
+      if (dt == SN_TYPE_STRING) {
+        if (!s.in_s || !s.in_s_len) { fail(SN_ERR_BADARG, "IN needs in_s"); return nullptr; }
+        bool is_g2 = plan->ngroup == 2 && plan->group_cols[1] == s.col;
+        int64_t mul = is_g2 ? 1 : (q->g2cap > 0 ? q->g2cap : 1);
+        for (int32_t vi = 0; vi < s.in_n; vi++) {
+          auto it = t->gdict_idx[s.col].find(
+              std::string(s.in_s[vi], (size_t)s.in_s_len[vi]));
+          if (it != t->gdict_idx[s.col].end())
+            vals.push_back((int64_t)it->second * mul);
+        }
+      } else {
+        if (!s.in_i) { fail(SN_ERR_BADARG, "IN needs in_i"); return nullptr; }
+        vals.assign(s.in_i, s.in_i + s.in_n);
+      }
+      std::sort(vals.begin(), vals.end());
+      vals.erase(std::unique(vals.begin(), vals.end()), vals.end());
+      if (vals.empty()) {
+        /* no literal matches anything: impossible range */
+        sn_dev_pred_d &d = dp.preds_d[dp.npreds_d++];
+        d.cslot = cslot; d.lo = 1.0; d.hi = 0.0;
+        continue;
+      }
+      auto &ip = dp.inp[dp.npreds_in++];
+      memset(&ip, 0, sizeof(ip));
+      ip.cslot = cslot;
+      const int64_t mn = vals.front(), mx = vals.back();
+      if (mx - mn < (1ll << 20)) {
+        const int64_t nwords = ((mx - mn) >> 6) + 1;
+        std::vector<uint64_t> bm((size_t)nwords, 0);
+        for (int64_t v : vals) bm[(v - mn) >> 6] |= 1ull << ((v - mn) & 63);
+        void *bmd = e->arena.alloc(bm.size() * 8);
+        if (!bmd || hipMemcpy(bmd, bm.data(), bm.size() * 8,
+                              hipMemcpyHostToDevice) != hipSuccess) {
+          fail(SN_ERR_NOMEM, "IN bitmap upload"); return nullptr;
+        }
+        q->owned.push_back({ bmd, bm.size() * 8 });
+        ip.bm = (const uint64_t *)bmd;
+        ip.base = mn;
+        ip.nwords = (int32_t)nwords;
+      } else {
+        void *ld = e->arena.alloc(vals.size() * 8);
+        if (!ld || hipMemcpy(ld, vals.data(), vals.size() * 8,
+                             hipMemcpyHostToDevice) != hipSuccess) {
+          fail(SN_ERR_NOMEM, "IN list upload"); return nullptr;
+        }
+        q->owned.push_back({ ld, vals.size() * 8 });
+        ip.list = (const int64_t *)ld;
+        ip.n = (int32_t)vals.size();
+      }
+      continue;
+    }
     if (dt == SN_TYPE_STRING) {
       /* dictionary pushdown: literal -> global dict id, compared against
        * the (possibly premultiplied) id the conversion pass writes */
@@ -2527,7 +2590,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       /* query-compiled twin first (plan structure compile-time, capacity
        * tokenized); any miss falls back to the interpreted hash kernel */
       void *jfn = nullptr;
-      if (e->jit && q->dev_naggs <= 12 &&
+      const bool jit_in_ok_s =
+          dp.npreds_in == 0 ||
+          (dp.inp[0].bm && (dp.npreds_in < 2 || dp.inp[1].bm));
+      if (e->jit && jit_in_ok_s && q->dev_naggs <= 12 &&
           (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first))) {
         const int *jk = hit ? hit->jit_kinds : jit_kinds;
         int jdel = hit ? hit->jit_del : (jit_any_del ? 1 : 0);
@@ -2681,8 +2747,12 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       /* global-atomic mode: only the LDS image constrains */
       jit_shape_ok = dp.nslots <= SN_BIG_GROUP_CAP;
     /* pac from ACTUAL nulls never reaches here (jit_ok excludes null
-     * batches); pac from MIN/MAX compiles op-aware kernels */
-    if (e->jit && jit_shape_ok &&
+     * batches); pac from MIN/MAX compiles op-aware kernels.  IN-lists run
+     * compiled only in bitmap form (sorted-list search stays interpreted). */
+    const bool jit_in_ok =
+        dp.npreds_in == 0 ||
+        (dp.inp[0].bm && (dp.npreds_in < 2 || dp.inp[1].bm));
+    if (e->jit && jit_shape_ok && jit_in_ok &&
         (hit ? hit->jit_ok != 0 : (jit_ok_b && !jit_first))) {
       const int *jk = hit ? hit->jit_kinds : jit_kinds;
       int jdel = hit ? hit->jit_del : (jit_any_del ? 1 : 0);

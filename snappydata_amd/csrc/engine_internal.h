@@ -141,6 +141,20 @@ typedef struct {
   sn_dev_pred_d preds_d[8];
   sn_dev_pred_i preds_i[4];
   sn_dev_agg aggs[12];
+  /* IN-list membership predicates: bitmap LUT over [base, base+nwords*64)
+   * when the value span is dense enough (dictionary ids always are), else
+   * a sorted list binary-searched per row (interpreted kernels only) */
+  struct {
+    const uint64_t *bm;
+    const int64_t *list;
+    int64_t base;
+    int32_t nwords;
+    int32_t n;
+    int32_t cslot;
+    int32_t _p;
+  } inp[2];
+  int32_t npreds_in;
+  int32_t _pad4b;
   /* sparse-key open-address hash aggregate (the ByteBufferHashMap /
    * SHAMapAccessor analogue, ByteBufferHashMap.scala:140-183,
    * SHAMapAccessor.scala:716-830): integer group keys WITHOUT dense-slot

@@ -119,6 +119,9 @@ struct sn_dev_plan {
   sn_dev_pred_d preds_d[8];
   sn_dev_pred_i preds_i[4];
   sn_dev_agg aggs[12];
+  struct { const u64 *bm; const i64 *list; i64 base;
+           int nwords, n, cslot, _p; } inp[2];
+  int npreds_in, _pad4b;
   i64 *hkeys; double *hacc; int *hflags;
   int hcap_log2, sparse, pac, _pad3;
 };
@@ -206,6 +209,11 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   for (int i = 0; i < p->npreds_i; i++)
     emitf(o, "  const i64 pi%d_lo = P->preds_i[%d].lo, pi%d_hi = P->preds_i[%d].hi;\n",
           i, i, i, i);
+  for (int i = 0; i < p->npreds_in && i < 2; i++)
+    emitf(o, "  const GAS u64 *inbm%d = (const GAS u64 *)(u64)P->inp[%d].bm;\n"
+             "  const i64 inb%d_base = P->inp[%d].base;\n"
+             "  const i64 inb%d_lim = (i64)P->inp[%d].nwords * 64;\n",
+          i, i, i, i, i, i);
   /* trivial factors (a=0, m=1 — plain column reads, the common case) fold
    * back to the raw product; their triviality is part of the shape hash */
   auto ftriv = [](double a, double m) { return a == 0.0 && m == 1.0; };
@@ -826,6 +834,16 @@ __device__ __forceinline__ u64 mix64(u64 x) {
             p->preds_i[i].cslot, i, i);
     }
   }
+  for (int i = 0; i < p->npreds_in && i < 2; i++) {
+    const int cs = p->inp[i].cslot;
+    const int is64 = (p->i64_mask >> cs) & 1u;
+    emitf(o, "        { const i64 v = %ssval[%d][r]%s;\n"
+             "          const i64 ix = v - inb%d_base;\n"
+             "          ok &= (ix >= 0) & (ix < inb%d_lim) &\n"
+             "                (int)((inbm%d[ix >> 6] >> (ix & 63)) & 1ull); }\n",
+          is64 ? "__double_as_longlong(" : "(i64)", cs, is64 ? ")" : "",
+          i, i, i);
+  }
   o += "        if (__popcll(__ballot(ok)) == 0) continue;\n";
   if (p->jkeys) {
     /* broadcast-dimension probe with literal table shape (mask, key slot,
@@ -1090,6 +1108,13 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
   shape.hacc = nullptr;
   shape.hflags = nullptr;
   shape.hcap_log2 = 0;
+  for (int i = 0; i < 2; i++) {
+    shape.inp[i].bm = shape.inp[i].bm ? (const uint64_t *)1 : nullptr;
+    shape.inp[i].list = shape.inp[i].list ? (const int64_t *)1 : nullptr;
+    shape.inp[i].base = 0;
+    shape.inp[i].nwords = 0;
+    shape.inp[i].n = 0;
+  }
   auto trivm = [](double a, double m) { return (a == 0.0 && m == 1.0) ? 1.0 : 0.0; };
   for (int i = 0; i < 8; i++) { shape.preds_d[i].lo = shape.preds_d[i].hi = 0.0; }
   for (int i = 0; i < 4; i++) { shape.preds_i[i].lo = shape.preds_i[i].hi = 0; }

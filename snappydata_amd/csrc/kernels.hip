@@ -595,6 +595,48 @@ __device__ __forceinline__ void pred_sweep_fused(const sn_dev_plan *P, int npd,
   }
 }
 
+/* IN-list membership sweep (bitmap LUT or sorted-list binary search) */
+__device__ __forceinline__ void in_sweeps(const sn_dev_plan *P, int clean,
+                                          const double *sval,
+                                          const uint64_t *svalid,
+                                          uint64_t *salive) {
+  const int tid = threadIdx.x;
+  for (int i = 0; i < P->npreds_in && i < 2; i++) {
+    const int cs = P->inp[i].cslot;
+    const int is64 = (int)((P->i64_mask >> cs) & 1u);
+    const uint64_t *bm = P->inp[i].bm;
+    const int64_t *list = P->inp[i].list;
+    const long long base = P->inp[i].base;
+    const long long lim = (long long)P->inp[i].nwords * 64;
+    const int n = P->inp[i].n;
+#pragma unroll
+    for (int k = 0; k < CHUNK / WG; k++) {
+      const int r = tid + k * WG;
+      const double x = sval[(size_t)cs * CHUNK + r];
+      const long long v = is64 ? __double_as_longlong(x) : (long long)x;
+      int ok;
+      if (bm) {
+        const long long idx = v - base;
+        ok = idx >= 0 && idx < lim &&
+             (int)((as_global(bm)[idx >> 6] >> (idx & 63)) & 1ull);
+      } else {
+        const GAS int64_t *ls = as_global(list);
+        int lo = 0, hi = n - 1;
+        ok = 0;
+        while (lo <= hi) {
+          const int mid = (lo + hi) >> 1;
+          const long long m = (long long)ls[mid];
+          if (m == v) { ok = 1; break; }
+          if (m < v) lo = mid + 1; else hi = mid - 1;
+        }
+      }
+      uint64_t w = __ballot(ok);
+      if (!clean) w &= svalid[(size_t)cs * (CHUNK / 64) + (r >> 6)];
+      if ((tid & 63) == 0) salive[r >> 6] &= w;
+    }
+  }
+}
+
 /* one sweep per predicate, params hoisted; wave-owned word updates */
 __device__ __forceinline__ void pred_sweeps(const sn_dev_plan *P, int npd,
                                             int npi, int clean,
@@ -602,6 +644,7 @@ __device__ __forceinline__ void pred_sweeps(const sn_dev_plan *P, int npd,
                                             const uint64_t *svalid,
                                             uint64_t *salive) {
   const int tid = threadIdx.x;
+  if (P->npreds_in > 0) in_sweeps(P, clean, sval, svalid, salive);
   if (npd <= 3 && npi == 0) {
     if (npd > 0) pred_sweep_fused(P, npd, clean, sval, svalid, salive);
     return;
@@ -743,7 +786,8 @@ __global__ void k_keyless(sn_dev_plan plan,
       /* fully-fused single pass (clean chunks, <=3 double preds, no join):
        * one traversal computes predicates, aggregates and counts — the
        * minimal LDS-read structure (matches the membw probe's shape) */
-      if (clean && npd <= 3 && npi == 0 && !plan.jkeys) {
+      if (clean && npd <= 3 && npi == 0 && !plan.jkeys &&
+          plan.npreds_in == 0) {
         double lo0 = -1e308, hi0 = 1e308, lo1 = lo0, hi1 = hi0, lo2 = lo0, hi2 = hi0;
         int c0 = 0, c1 = 0, c2 = 0;
         if (npd >= 1) { lo0 = P->preds_d[0].lo; hi0 = P->preds_d[0].hi; c0 = P->preds_d[0].cslot; }
@@ -1172,7 +1216,8 @@ __global__ void k_grouped_reg(sn_dev_plan plan,
       const int next_staged = pipe && nbase + CHUNK <= tile_end;
       if (STAGED && next_staged) { if constexpr (STAGED) stage_load(cr, nused, nbase, st); }
 
-      const int inline_preds = clean && npi == 0 && !plan.jkeys;
+      const int inline_preds = clean && npi == 0 && !plan.jkeys &&
+                               plan.npreds_in == 0;
       if (!inline_preds) {
         alive_init(salive, sdead, rows, clean);
         pred_sweeps(P, npd, npi, clean, sval, svalid, salive);
