@@ -1517,6 +1517,8 @@ extern "C" int32_t sn_batch_put_raw(sn_engine *e, int32_t table,
   if (any_raw) {
     b.stats_dev = sd;
     b.stats_pending = true;
+  } else if (sd) {
+    e->arena.release(sd, (size_t)nc * 16);
   }
 
   std::lock_guard<std::mutex> g(t->mu);
