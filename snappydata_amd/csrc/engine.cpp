@@ -280,9 +280,16 @@ struct sn_engine {
   long long *hws_okeys = nullptr;
   double *hws_acc = nullptr;
   double *hws_orows = nullptr;
-  int32_t *hws_flags = nullptr;   /* [0] overflow, [1] compact counter */
+  int32_t *hws_flags = nullptr;   /* [0] overflow, [1] compact counter,
+                                     [2] fill, [3] radix record overflow */
   int hws_cap_log2 = 0;
   size_t hws_acc_bytes = 0;
+  /* radix two-pass record buffer + per-partition fill counters (engine-
+   * cached like the table workspace; sized to the largest query seen) */
+  double *rws_recs = nullptr;
+  size_t rws_bytes = 0;
+  int32_t *rws_pcount = nullptr;
+  int rws_npart = 0;
   int32_t *lz4_err_dev = nullptr; /* device error word for the LZ4 decode */
   std::mutex lz4_mu;              /* serializes decode launch+sync+readback */
   /* pinned staging for big blob uploads: pageable hipMemcpy bounces through
@@ -2560,6 +2567,18 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (t->sparse_cap_hint > cap_log2) cap_log2 = t->sparse_cap_hint;
     q->ev_start = e->ev_acquire();
     q->ev_stop = e->ev_acquire();
+    /* radix two-pass: the single-pass probe touches ~3 random
+     * 64 B lines per row across a table far bigger than L2 (measured
+     * 182 B/row of HBM traffic on 1M keys).  For big tables the compiled
+     * pass 1 scatters (key, values) records into cap/4096 hash partitions
+     * (streaming traffic) and k_radix_agg aggregates each partition into
+     * its private L2-resident table segment.  JIT-only (clean batches);
+     * pac excluded (records carry values, not per-agg validity). */
+    static const bool radix_env = [] {
+      const char *v = getenv("SN_RADIX");
+      return !v || v[0] != '0';
+    }();
+    bool radix_ok = radix_env && !q->pac && q->dev_naggs <= 4;
     bool done_h = false;
     for (int attempt = 0; attempt < 4 && !done_h; attempt++) {
       if (ensure_sparse_ws(e, cap_log2, naggs1) != SN_OK) {
@@ -2572,6 +2591,41 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       dps.hacc = e->hws_acc;
       dps.hflags = e->hws_flags;
       dps.hcap_log2 = cap_log2;
+      bool radix = radix_ok && cap_log2 >= 17;
+      if (radix) {
+        const int npart = 1 << (cap_log2 - SN_RADIX_SUB_LOG2);
+        const long long percap =
+            2 * q->rows_scanned / npart + 4096;
+        const size_t recb = (size_t)npart * (size_t)percap *
+                            (size_t)(1 + q->dev_naggs) * 8;
+        if (percap > INT32_MAX / 2 || recb > (48ull << 30)) {
+          radix = false;
+        } else {
+          if (!e->rws_recs || e->rws_bytes < recb) {
+            if (e->rws_recs) e->arena.release(e->rws_recs, e->rws_bytes);
+            e->rws_recs = (double *)e->arena.alloc(recb);
+            e->rws_bytes = e->rws_recs ? recb : 0;
+          }
+          if (e->rws_npart < npart) {
+            if (e->rws_pcount)
+              e->arena.release(e->rws_pcount, (size_t)e->rws_npart * 4);
+            e->rws_pcount = (int32_t *)e->arena.alloc((size_t)npart * 4);
+            e->rws_npart = e->rws_pcount ? npart : 0;
+          }
+          if (!e->rws_recs || !e->rws_pcount) {
+            radix = false;               /* fall back to the single pass */
+          } else {
+            dps.radix = 1;
+            dps.precs = e->rws_recs;
+            dps.pcount = e->rws_pcount;
+            dps.percap = (int32_t)percap;
+            if (hipMemsetAsync(e->rws_pcount, 0, (size_t)npart * 4,
+                               e->stream) != hipSuccess) {
+              fail(SN_ERR_GENERIC, "radix counters zero"); return nullptr;
+            }
+          }
+        }
+      }
       void *dps_dev = e->arena.alloc(sizeof(dps));
       if (!dps_dev ||
           hipMemcpy(dps_dev, &dps, sizeof(dps), hipMemcpyHostToDevice) != hipSuccess) {
@@ -2623,6 +2677,12 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
                                  (const sn_dev_batch *)db_dev,
                                  (const sn_dev_tile *)tl_dev, ntiles,
                                  e->stream);
+      /* radix pass 2: aggregate the partitioned records into the table
+       * segments (only meaningful after the compiled pass 1) */
+      const bool radix_ran = radix && jfn;
+      if (rc == 0 && radix_ran)
+        rc = sn_launch_radix_agg(&dps, (const sn_dev_plan *)dps_dev,
+                                 e->stream);
       if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);
       if (rc != 0) {
         fail(SN_ERR_GENERIC, "hash-agg launch: %s",
@@ -2634,6 +2694,14 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       }
       int32_t fl[4] = { 0, 0, 0, 0 };
       (void)hipMemcpy(fl, e->hws_flags, 16, hipMemcpyDeviceToHost);
+      if (radix_ran && fl[3]) {
+        /* record buffer overflow (pathological key skew): redo this
+         * attempt through the single-pass probe — nothing was lost, the
+         * table is re-zeroed per attempt */
+        radix_ok = false;
+        attempt--;
+        continue;
+      }
       const int32_t ovf = fl[0];
       const int64_t fill = fl[2];
       /* grow on hard overflow OR load factor > 0.6 (probe chains degrade

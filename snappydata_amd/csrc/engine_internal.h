@@ -176,7 +176,23 @@ typedef struct {
    * counts == rowcount (non-null inputs) */
   int32_t pac;
   int32_t _pad3;
+  /* radix-partitioned two-pass hash aggregate (query-compiled pass 1 only;
+   * big tables, clean batches).  Pass 1 scatters (key, agg values) records
+   * into 1 << (hcap_log2 - SN_RADIX_SUB_LOG2) hash partitions instead of
+   * probing the table; pass 2 (k_radix_agg) aggregates each partition into
+   * its own 4096-slot SEGMENT of hkeys/hacc, so the probe + accumulate
+   * working set per workgroup is ~100 KB and stays in the XCD's L2 instead
+   * of thrashing HBM with 64 B random lines.  precs = record buffer
+   * [npart][percap][(1+naggs) doubles] (slot 0 = key bits); pcount[npart]
+   * fill counters; hflags[3] = partition overflow (host retries without
+   * radix).  radix == 0 leaves the single-pass probe behavior. */
+  double *precs;
+  int32_t *pcount;
+  int32_t percap;
+  int32_t radix;
 } sn_dev_plan;
+
+#define SN_RADIX_SUB_LOG2 12   /* 4096-slot per-partition table segments */
 
 /* sparse hash-aggregate empty-slot sentinel: -1 so the host can memset the
  * key array (a REAL key of -1 routes to the reserved row at index cap) */
@@ -234,6 +250,11 @@ int sn_launch_hash_scan(const sn_dev_plan *plan, const sn_dev_plan *dev_plan,
 int sn_launch_hash_compact(const long long *hk, const double *hacc,
                            int cap, int naggs1, long long *okeys,
                            double *orows, int *counter, void *stream);
+
+/* radix pass 2: aggregate each partition's records into its table segment
+ * (plan carries precs/pcount/percap and the table pointers) */
+int sn_launch_radix_agg(const sn_dev_plan *plan, const sn_dev_plan *dev_plan,
+                        void *stream);
 
 /* device-side join-table build from a column table (colocated
  * partitioned-partitioned join): fills an open-address (key, payload)

@@ -75,6 +75,12 @@ static std::string gen_source(const sn_dev_plan *p, const int *kinds,
    * capacity is tokenized (read from the device plan) so grow-and-retry
    * reuses one compiled kernel */
   const int sparse_mode = p->sparse != 0;
+  /* radix two-pass: instead of probing the global table per row (64 B
+   * random lines across tens of MB — the measured 182 B/row), pass 1
+   * scatters (key, agg values) records into 1 << (hcap_log2-12) hash
+   * partitions via an LDS-histogram multi-split; k_radix_agg then
+   * aggregates each partition into its private L2-resident table segment */
+  const int radix_mode = sparse_mode && p->radix != 0;
   /* pac: per-agg-count accumulator rows ([sums][counts][rowcount]).  In
    * the JIT this occurs only for MIN/MAX plans (null-carrying batches are
    * never JIT-eligible, so counts == rowcount and the count cells simply
@@ -124,6 +130,7 @@ struct sn_dev_plan {
   int npreds_in, _pad4b;
   i64 *hkeys; double *hacc; int *hflags;
   int hcap_log2, sparse, pac, _pad3;
+  double *precs; int *pcount; int percap, radix;
 };
 __device__ __forceinline__ double wsum(double x) {
 #pragma unroll
@@ -252,6 +259,8 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     if (fold_preds)
       o += "  __shared__ unsigned long long sbits[16];\n";
   }
+  if (radix_mode)
+    o += "  __shared__ int phist[4096];\n";   /* npart <= 4096 (cap 2^24) */
   if (!lds_mode && !wbin_mode && !glob_mode)
     emitf(o, "  __shared__ __attribute__((aligned(16))) double bacc[%d];\n",
           grouped ? NS * NA1 : 2 * na_t + 1);
@@ -266,7 +275,13 @@ __device__ __forceinline__ u64 mix64(u64 x) {
              "  GAS int *hflags = (GAS int *)(u64)jpayload_p;\n"
              "  const int hcl = P->hcap_log2;\n"
              "  const unsigned hmask = (1u << hcl) - 1;\n"
-             "  const int hcap = 1 << hcl;\n");
+             "  const int hcap = 1 << hcl;\n"
+             "  (void)gacc; (void)hkeys; (void)hmask; (void)hcap;\n");
+    if (radix_mode)
+      o += "  GAS double *precs = (GAS double *)(u64)P->precs;\n"
+           "  GAS int *pcount = (GAS int *)(u64)P->pcount;\n"
+           "  const int percap = P->percap;\n"
+           "  const int npart = 1 << (hcl - 12);\n";
   } else if (glob_mode) {
     /* HBM accumulator IS the out/scratch pointer (host-zeroed), privatized
      * 8 ways by XCD (blockIdx & 7 matches the dispatch round-robin) so
@@ -692,8 +707,10 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   emit_load("tile.row_start", "      ");
   o += "      staged = 1;\n    }\n"
        "    for (int base = tile.row_start; base < tile_end; base += CHUNK) {\n"
-       "      const int rows = min(CHUNK, tile_end - base);\n"
-       "      if (staged) {\n";
+       "      const int rows = min(CHUNK, tile_end - base);\n";
+  if (radix_mode)   /* previous chunk's scatter is behind the tail barrier */
+    o += "      for (int i = tid; i < npart; i += WG) phist[i] = 0;\n";
+  o += "      if (staged) {\n";
   if (use_spay) {
     const int jc = p->jcslot;
     const int jk = kinds[jc];
@@ -812,11 +829,23 @@ __device__ __forceinline__ u64 mix64(u64 x) {
          "        const int r = tid + k * WG;\n"
          "        int ok = (int)((sbits[((r >> 7) << 1) | (r & 1)]\n"
          "                        >> ((r >> 1) & 63)) & 1ull);\n";
-  else
-    o += "#pragma unroll 2\n"
-         "      for (int k = 0; k < CHUNK / WG; k++) {\n"
+  else {
+    if (radix_mode) {
+      /* per-k register stash: pass A computes (partition, rank, key,
+       * values) per row; pass B scatters after the per-chunk base
+       * reservation.  Fully unrolled so the arrays stay in registers. */
+      emitf(o, "      i64 rkey[CHUNK / WG];\n"
+               "      int rok[CHUNK / WG], rpk[CHUNK / WG], rrnk[CHUNK / WG];\n");
+      for (int a = 0; a < NA; a++)
+        emitf(o, "      double rva%d[CHUNK / WG];\n", a);
+      o += "#pragma unroll\n";
+    } else {
+      o += "#pragma unroll 2\n";
+    }
+    o += "      for (int k = 0; k < CHUNK / WG; k++) {\n"
          "        const int r = tid + k * WG;\n"
          "        int ok = r < rows;\n";
+  }
   if (has_del)
     o += "        if (del) {\n"
          "          const int gr = base + r;\n"
@@ -844,6 +873,8 @@ __device__ __forceinline__ u64 mix64(u64 x) {
           is64 ? "__double_as_longlong(" : "(i64)", cs, is64 ? ")" : "",
           i, i, i);
   }
+  if (radix_mode)   /* before the wave skip: pass B keys off rok alone */
+    o += "        rok[k] = ok;\n";
   o += "        if (__popcll(__ballot(ok)) == 0) continue;\n";
   if (p->jkeys) {
     /* broadcast-dimension probe with literal table shape (mask, key slot,
@@ -887,7 +918,18 @@ __device__ __forceinline__ u64 mix64(u64 x) {
                " << 32) | (unsigned)(int)sval[%d][r]);\n",
             p->gcol[0], p->gcol[1]);
     }
-    o += R"(        int slot = -1;
+    if (radix_mode) {
+      /* pass A: partition by mix64 HIGH bits (pass 2 probes the segment
+       * with the LOW bits — independent), rank via the LDS histogram */
+      o += "        rkey[k] = skey;\n"
+           "        int rpk_ = 0, rrnk_ = 0;\n"
+           "        if (ok) {\n"
+           "          rpk_ = (int)((mix64((u64)skey) >> 44) & (u64)(npart - 1));\n"
+           "          rrnk_ = atomicAdd(&phist[rpk_], 1);\n"
+           "        }\n"
+           "        rpk[k] = rpk_; rrnk[k] = rrnk_;\n";
+    } else {
+      o += R"(        int slot = -1;
         if (ok) {
           if (skey == -1ll) slot = hcap;     /* sentinel-valued key */
           else {
@@ -909,6 +951,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
         }
         ok &= slot >= 0;
 )";
+    }
   } else if (grouped) {
     if (p->jkeys && p->jmode == 1) {
       o += "        const int slot = pay > 0 ? pay : 0;\n";
@@ -950,7 +993,10 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     if (A.nf >= 3) emitf(o, " * %s", t2);
     o += ";\n";
   }
-  if (lds_mode || glob_mode) {
+  if (radix_mode) {
+    for (int a = 0; a < NA; a++)
+      emitf(o, "        rva%d[k] = va%d;\n", a, a);
+  } else if (lds_mode || glob_mode) {
     emitf(o, "        if (ok) {\n"
              "          %sdouble *row = &gacc[(u64)slot * %d];\n"
              "          atomicAdd((double *)&row[%d], 1.0);\n",
@@ -999,8 +1045,29 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     }
     o += "        rcnt += ok ? 1.0 : 0.0;\n";
   }
-  o += "      }\n"
-       "      __syncthreads();\n"
+  o += "      }\n";
+  if (radix_mode) {
+    /* reserve contiguous global space per (chunk, partition) — one global
+     * atomic per NON-EMPTY partition per chunk, not one per row — then
+     * scatter the stashed records to [base, base+count) */
+    emitf(o, "      __syncthreads();\n"
+             "      for (int i = tid; i < npart; i += WG) {\n"
+             "        const int c = phist[i];\n"
+             "        phist[i] = c ? atomicAdd(&pcount[i], c) : 0;\n"
+             "      }\n"
+             "      __syncthreads();\n"
+             "#pragma unroll\n"
+             "      for (int k = 0; k < CHUNK / WG; k++) {\n"
+             "        if (!rok[k]) continue;\n"
+             "        const u64 off = (u64)phist[rpk[k]] + (u64)rrnk[k];\n"
+             "        if (off >= (u64)percap) { atomicOr(hflags + 3, 1); continue; }\n"
+             "        GAS double *rec = precs + ((u64)rpk[k] * (u64)percap + off) * %d;\n"
+             "        ((GAS i64 *)rec)[0] = rkey[k];\n", 1 + NA);
+    for (int a = 0; a < NA; a++)
+      emitf(o, "        rec[%d] = rva%d[k];\n", 1 + a, a);
+    o += "      }\n";
+  }
+  o += "      __syncthreads();\n"
        "      staged = next_staged;\n"
        "    }\n"
        "  }\n";
@@ -1108,6 +1175,11 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
   shape.hacc = nullptr;
   shape.hflags = nullptr;
   shape.hcap_log2 = 0;
+  /* radix stays in the hash (it changes the source); the record-buffer
+   * geometry is tokenized */
+  shape.precs = nullptr;
+  shape.pcount = nullptr;
+  shape.percap = 0;
   for (int i = 0; i < 2; i++) {
     shape.inp[i].bm = shape.inp[i].bm ? (const uint64_t *)1 : nullptr;
     shape.inp[i].list = shape.inp[i].list ? (const int64_t *)1 : nullptr;
