@@ -2592,8 +2592,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       dps.hflags = e->hws_flags;
       dps.hcap_log2 = cap_log2;
       bool radix = radix_ok && cap_log2 >= 17;
+      const int sub_log2 =
+          std::max(SN_RADIX_SUB_MIN, cap_log2 - SN_RADIX_NPART_MAX_LOG2);
       if (radix) {
-        const int npart = 1 << (cap_log2 - SN_RADIX_SUB_LOG2);
+        const int npart = 1 << (cap_log2 - sub_log2);
         const long long percap =
             2 * q->rows_scanned / npart + 4096;
         const size_t recb = (size_t)npart * (size_t)percap *
@@ -2615,7 +2617,7 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
           if (!e->rws_recs || !e->rws_pcount) {
             radix = false;               /* fall back to the single pass */
           } else {
-            dps.radix = 1;
+            dps.radix = sub_log2;
             dps.precs = e->rws_recs;
             dps.pcount = e->rws_pcount;
             dps.percap = (int32_t)percap;
@@ -2677,12 +2679,15 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
                                  (const sn_dev_batch *)db_dev,
                                  (const sn_dev_tile *)tl_dev, ntiles,
                                  e->stream);
-      /* radix pass 2: aggregate the partitioned records into the table
-       * segments (only meaningful after the compiled pass 1) */
+      /* radix pass 2: aggregate the partitioned records (only meaningful
+       * after the compiled pass 1).  The LDS variant compacts straight
+       * into okeys/orows via the counter at hws_flags[1]. */
       const bool radix_ran = radix && jfn;
+      const bool radix_direct = radix_ran && sn_radix_direct(sub_log2, naggs1);
       if (rc == 0 && radix_ran)
         rc = sn_launch_radix_agg(&dps, (const sn_dev_plan *)dps_dev,
-                                 e->stream);
+                                 e->hws_okeys, e->hws_orows,
+                                 (int *)(e->hws_flags + 1), e->stream);
       if (q->ev_stop) (void)hipEventRecord(q->ev_stop, e->stream);
       if (rc != 0) {
         fail(SN_ERR_GENERIC, "hash-agg launch: %s",
@@ -2716,11 +2721,14 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
         continue;
       }
       t->sparse_cap_hint = std::max(t->sparse_cap_hint, cap_log2);
-      if (hipMemsetAsync(e->hws_flags + 1, 0, 4, e->stream) != hipSuccess ||
-          sn_launch_hash_compact(e->hws_keys, e->hws_acc, (int)cap, naggs1,
-                                 e->hws_okeys, e->hws_orows,
-                                 (int *)(e->hws_flags + 1), e->stream) != 0 ||
-          hipStreamSynchronize(e->stream) != hipSuccess) {
+      /* the LDS radix pass already compacted into okeys/orows (counter at
+       * hws_flags[1]); otherwise compact the table here */
+      if (!radix_direct &&
+          (hipMemsetAsync(e->hws_flags + 1, 0, 4, e->stream) != hipSuccess ||
+           sn_launch_hash_compact(e->hws_keys, e->hws_acc, (int)cap, naggs1,
+                                  e->hws_okeys, e->hws_orows,
+                                  (int *)(e->hws_flags + 1), e->stream) != 0 ||
+           hipStreamSynchronize(e->stream) != hipSuccess)) {
         fail(SN_ERR_GENERIC, "hash-agg compact"); return nullptr;
       }
       int32_t ngrp = 0;
@@ -2740,6 +2748,19 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
                       (size_t)ngrp * naggs1 * 8,
                       hipMemcpyDeviceToHost) != hipSuccess) {
           fail(SN_ERR_GENERIC, "hash-agg readback"); return nullptr;
+        }
+      }
+      if (radix_direct) {
+        /* the LDS compaction never scans the (untouched) global table, so
+         * append the reserved sentinel-key row (a REAL key of -1) here —
+         * the same row k_hash_compact emits from index cap */
+        std::vector<double> rrow((size_t)naggs1, 0.0);
+        (void)hipMemcpy(rrow.data(), e->hws_acc + (size_t)cap * naggs1,
+                        (size_t)naggs1 * 8, hipMemcpyDeviceToHost);
+        if (rrow[naggs1 - 1] != 0.0) {
+          q->sparse_keys.push_back(SN_HASH_EMPTY);
+          q->sparse_rows.insert(q->sparse_rows.end(), rrow.begin(),
+                                rrow.end());
         }
       }
       done_h = true;

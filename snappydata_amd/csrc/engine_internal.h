@@ -192,7 +192,21 @@ typedef struct {
   int32_t radix;
 } sn_dev_plan;
 
-#define SN_RADIX_SUB_LOG2 12   /* 4096-slot per-partition table segments */
+/* radix partition shift (plan->radix): slots per partition table.  11
+ * (2048 slots) keeps the LDS table at 2 workgroups/CU for the common
+ * sum+count row; capped so npart = cap >> radix never exceeds the 4096
+ * the pass-1 histogram sizes for. */
+#define SN_RADIX_SUB_MIN 11
+#define SN_RADIX_NPART_MAX_LOG2 12
+#define SN_RADIX_LDS_MAX (144u * 1024)
+static inline unsigned sn_radix_lds_bytes(int sub_log2, int naggs1) {
+  return (unsigned)((1u << sub_log2) * (8u + 8u * (unsigned)naggs1) +
+                    256 /*WG*/ * 4u);
+}
+/* does pass 2 compact DIRECTLY from LDS (host then skips k_hash_compact)? */
+static inline int sn_radix_direct(int sub_log2, int naggs1) {
+  return sn_radix_lds_bytes(sub_log2, naggs1) <= SN_RADIX_LDS_MAX;
+}
 
 /* sparse hash-aggregate empty-slot sentinel: -1 so the host can memset the
  * key array (a REAL key of -1 routes to the reserved row at index cap) */
@@ -251,9 +265,12 @@ int sn_launch_hash_compact(const long long *hk, const double *hacc,
                            int cap, int naggs1, long long *okeys,
                            double *orows, int *counter, void *stream);
 
-/* radix pass 2: aggregate each partition's records into its table segment
- * (plan carries precs/pcount/percap and the table pointers) */
+/* radix pass 2: aggregate each partition's records in an LDS table and
+ * compact straight into (okeys, orows) — or, for rows too wide for LDS,
+ * into per-partition segments of the global table (host runs
+ * k_hash_compact as usual; sn_radix_direct says which) */
 int sn_launch_radix_agg(const sn_dev_plan *plan, const sn_dev_plan *dev_plan,
+                        long long *okeys, double *orows, int *counter,
                         void *stream);
 
 /* device-side join-table build from a column table (colocated

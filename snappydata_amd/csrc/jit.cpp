@@ -281,7 +281,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
       o += "  GAS double *precs = (GAS double *)(u64)P->precs;\n"
            "  GAS int *pcount = (GAS int *)(u64)P->pcount;\n"
            "  const int percap = P->percap;\n"
-           "  const int npart = 1 << (hcl - 12);\n";
+           "  const int npart = 1 << (hcl - P->radix);\n";
   } else if (glob_mode) {
     /* HBM accumulator IS the out/scratch pointer (host-zeroed), privatized
      * 8 ways by XCD (blockIdx & 7 matches the dispatch round-robin) so
@@ -1175,11 +1175,13 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
   shape.hacc = nullptr;
   shape.hflags = nullptr;
   shape.hcap_log2 = 0;
-  /* radix stays in the hash (it changes the source); the record-buffer
-   * geometry is tokenized */
+  /* radix stays in the hash as a BOOLEAN (it changes the source; the
+   * partition shift itself is read from the plan at run time) and the
+   * record-buffer geometry is tokenized */
   shape.precs = nullptr;
   shape.pcount = nullptr;
   shape.percap = 0;
+  shape.radix = shape.radix ? 1 : 0;
   for (int i = 0; i < 2; i++) {
     shape.inp[i].bm = shape.inp[i].bm ? (const uint64_t *)1 : nullptr;
     shape.inp[i].list = shape.inp[i].list ? (const int64_t *)1 : nullptr;

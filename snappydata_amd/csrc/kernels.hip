@@ -1726,21 +1726,137 @@ extern "C" int sn_launch_hash_compact(const long long *hk, const double *hacc,
   return (int)hipGetLastError();
 }
 
-/* ---- radix pass 2: per-partition aggregation into table SEGMENTS ----
+/* ---- radix pass 2: per-partition aggregation in LDS ----
  * Pass 1 (the query-compiled scatter) partitioned the passing rows by
  * mix64(key) high bits into plan->precs.  Here one workgroup owns one
- * partition and aggregates its records into the partition's private
- * 4096-slot segment of hkeys/hacc — probe + accumulate touch a ~100 KB
- * region that lives in the XCD's L2, instead of 64 B random lines across
- * the whole table in HBM (the measured 182 B/row of the single-pass
- * kernel).  Layout, sentinel, fill counter, overflow flag, reserved row
- * and the compact/readback epilogue are IDENTICAL to the single-pass
- * path, so grow-and-retry and the result plumbing need no radix cases. */
-__global__ __launch_bounds__(WG, 2)
-void k_radix_agg(const sn_dev_plan *__restrict__ plan_g) {
+ * partition and aggregates its records into an LDS-RESIDENT open-address
+ * table (the measured fix: a first version kept per-partition segments of
+ * the global table — "L2-resident" probes still cost 6.9 ms on 60M rows
+ * at 1 workgroup/partition, nearly the whole single-pass time, because
+ * the streaming record reads evict the segments and the dependent L2
+ * atomic chains sit at 8 waves/CU; LDS f64 atomics are the same pipe the
+ * 5 TB/s dense grouped kernels run on).  The filled slots then compact
+ * DIRECTLY from LDS into the (okeys, orows) result arrays — the global
+ * hkeys table is never touched and k_hash_compact is skipped; only the
+ * reserved sentinel-key row still lives in hacc[cap].  Fill counting and
+ * the overflow flag keep the single-pass contract, so grow-and-retry is
+ * unchanged. */
+__global__ __launch_bounds__(WG, 1)
+void k_radix_agg_lds(const sn_dev_plan *__restrict__ plan_g,
+                     long long *__restrict__ okeys_,
+                     double *__restrict__ orows_,
+                     int *__restrict__ counter_) {
+  extern __shared__ __attribute__((aligned(16))) char lsm[];
   const GAS sn_dev_plan *P = (const GAS sn_dev_plan *)(uintptr_t)plan_g;
   const int naggs = P->naggs;
   const int na1 = naggs + 1;              /* radix requires pac == 0 */
+  const int sub = P->radix;               /* partition shift (slots log2) */
+  const int subcap = 1 << sub;
+  const unsigned smask = (unsigned)subcap - 1;
+  const int tid = threadIdx.x;
+  const int p = blockIdx.x;
+  long long *skeys = (long long *)lsm;
+  double *sacc = (double *)(lsm + (size_t)subcap * 8);
+  int *scnt = (int *)(lsm + (size_t)subcap * 8 + (size_t)subcap * na1 * 8);
+  __shared__ int basew;
+
+  int ops[12];                            /* SN_MAX_AGGS */
+  double idents[12];
+  for (int a = 0; a < naggs; a++) {
+    ops[a] = P->aggs[a].op;
+    idents[a] = acc_ident(ops[a]);
+  }
+  for (int i = tid; i < subcap; i += WG) {
+    skeys[i] = SN_HASH_EMPTY;
+    double *row = sacc + (size_t)i * na1;
+    for (int a = 0; a < naggs; a++) row[a] = idents[a];
+    row[na1 - 1] = 0.0;
+  }
+  __syncthreads();
+
+  const int percap = P->percap;
+  const int n = min(((const GAS int *)(uintptr_t)P->pcount)[p], percap);
+  const GAS double *recs = (const GAS double *)(uintptr_t)P->precs +
+                           (size_t)p * percap * (1 + naggs);
+  GAS double *acc = (GAS double *)(uintptr_t)P->hacc;
+  GAS int32_t *flags = (GAS int32_t *)(uintptr_t)P->hflags;
+  const int cap = 1 << P->hcap_log2;
+
+  for (int i = tid; i < n; i += WG) {
+    const GAS double *rec = recs + (size_t)i * (1 + naggs);
+    const long long key = __double_as_longlong(rec[0]);
+    if (key == SN_HASH_EMPTY) {
+      /* a REAL key of -1: reserved global row at cap (rare) */
+      GAS double *row = acc + (size_t)cap * na1;
+      for (int a = 0; a < naggs; a++) acc_cell(&row[a], ops[a], rec[1 + a]);
+      (void)atomicAdd((double *)&row[na1 - 1], 1.0);
+      continue;
+    }
+    unsigned h = (unsigned)mix64((unsigned long long)key) & smask;
+    int slot = -1;
+    for (unsigned it = 0; it <= smask; ++it) {
+      const long long k0 = skeys[h];
+      if (k0 == key) { slot = (int)h; break; }
+      if (k0 == SN_HASH_EMPTY) {
+        const long long old = (long long)atomicCAS(
+            (unsigned long long *)&skeys[h], (unsigned long long)SN_HASH_EMPTY,
+            (unsigned long long)key);
+        if (old == SN_HASH_EMPTY || old == key) { slot = (int)h; break; }
+      }
+      h = (h + 1) & smask;
+    }
+    if (slot < 0) { atomicOr((int *)flags, 1); continue; }   /* table full */
+    double *row = sacc + (size_t)slot * na1;
+    for (int a = 0; a < naggs; a++) acc_cell(&row[a], ops[a], rec[1 + a]);
+    (void)atomicAdd(&row[na1 - 1], 1.0);
+  }
+  __syncthreads();
+
+  /* compact straight from LDS: thread t owns slots [t*spt, (t+1)*spt);
+   * block prefix-sum places each filled slot at one reserved output
+   * range (ONE global counter add per block, and the filled-slot total
+   * IS the insert count, so the fill counter costs one more add). */
+  const int spt = subcap / WG;
+  int mine = 0;
+  for (int j = 0; j < spt; j++)
+    mine += skeys[tid * spt + j] != SN_HASH_EMPTY;
+  scnt[tid] = mine;
+  __syncthreads();
+  for (int off = 1; off < WG; off <<= 1) {        /* inclusive scan */
+    const int u = tid >= off ? scnt[tid - off] : 0;
+    __syncthreads();
+    scnt[tid] += u;
+    __syncthreads();
+  }
+  if (tid == WG - 1) {
+    const int total = scnt[WG - 1];
+    basew = total ? atomicAdd((int *)(uintptr_t)counter_, total) : 0;
+    if (total) (void)atomicAdd((int *)(flags + 2), total);
+  }
+  __syncthreads();
+  int o = basew + scnt[tid] - mine;
+  GAS long long *okeys = (GAS long long *)(uintptr_t)okeys_;
+  GAS double *orows = (GAS double *)(uintptr_t)orows_;
+  for (int j = 0; j < spt; j++) {
+    const int s = tid * spt + j;
+    if (skeys[s] == SN_HASH_EMPTY) continue;
+    okeys[o] = skeys[s];
+    GAS double *dst = orows + (size_t)o * na1;
+    const double *src = sacc + (size_t)s * na1;
+    for (int a = 0; a < na1; a++) dst[a] = src[a];
+    o++;
+  }
+}
+
+/* fallback for accumulator rows too wide for the LDS table: per-partition
+ * SEGMENTS of the global hkeys/hacc (compact + readback then run as in the
+ * single-pass path) */
+__global__ __launch_bounds__(WG, 2)
+void k_radix_agg_glob(const sn_dev_plan *__restrict__ plan_g) {
+  const GAS sn_dev_plan *P = (const GAS sn_dev_plan *)(uintptr_t)plan_g;
+  const int naggs = P->naggs;
+  const int na1 = naggs + 1;
+  const int sub = P->radix;
   const int p = blockIdx.x;
   const int percap = P->percap;
   const GAS int *pcount = (const GAS int *)(uintptr_t)P->pcount;
@@ -1751,9 +1867,9 @@ void k_radix_agg(const sn_dev_plan *__restrict__ plan_g) {
   GAS double *acc = (GAS double *)(uintptr_t)P->hacc;
   GAS int32_t *flags = (GAS int32_t *)(uintptr_t)P->hflags;
   const int cap = 1 << P->hcap_log2;
-  const int segbase = p << SN_RADIX_SUB_LOG2;
-  const unsigned smask = (1u << SN_RADIX_SUB_LOG2) - 1;
-  int ops[12];                            /* SN_MAX_AGGS */
+  const int segbase = p << sub;
+  const unsigned smask = (1u << sub) - 1;
+  int ops[12];
   for (int a = 0; a < naggs; a++) ops[a] = P->aggs[a].op;
 
   for (int i = threadIdx.x; i < n; i += WG) {
@@ -1790,10 +1906,19 @@ void k_radix_agg(const sn_dev_plan *__restrict__ plan_g) {
 }
 
 extern "C" int sn_launch_radix_agg(const sn_dev_plan *plan,
-                                   const sn_dev_plan *dev_plan, void *stream) {
-  const int npart = 1 << (plan->hcap_log2 - SN_RADIX_SUB_LOG2);
-  hipLaunchKernelGGL(k_radix_agg, dim3(npart), dim3(WG), 0,
-                     (hipStream_t)stream, dev_plan);
+                                   const sn_dev_plan *dev_plan,
+                                   long long *okeys, double *orows,
+                                   int *counter, void *stream) {
+  const int npart = 1 << (plan->hcap_log2 - plan->radix);
+  const int na1 = plan->naggs + 1;
+  const size_t lds = sn_radix_lds_bytes(plan->radix, na1);
+  if (lds <= SN_RADIX_LDS_MAX) {
+    hipLaunchKernelGGL(k_radix_agg_lds, dim3(npart), dim3(WG), lds,
+                       (hipStream_t)stream, dev_plan, okeys, orows, counter);
+  } else {
+    hipLaunchKernelGGL(k_radix_agg_glob, dim3(npart), dim3(WG), 0,
+                       (hipStream_t)stream, dev_plan);
+  }
   return (int)hipGetLastError();
 }
 
