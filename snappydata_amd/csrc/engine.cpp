@@ -40,6 +40,7 @@
 #include <cstdarg>
 #include <cstdio>
 #include <cstring>
+#include <malloc.h>
 #include <map>
 #include <set>
 #include <memory>
@@ -352,7 +353,36 @@ extern "C" sn_engine *sn_engine_create(const sn_config *cfg) {
   } else {
     e->jit = sn_jit_cache_create();
   }
+  /* big per-query result vectors (1M-group readbacks are ~24 MB) must
+   * recycle through the heap, not mmap/munmap per query — the page-fault
+   * churn measured as a bimodal 1.4-3.6 ms of host time per sparse query */
+  (void)mallopt(M_MMAP_THRESHOLD, 256 << 20);
+  (void)mallopt(M_TRIM_THRESHOLD, 256 << 20);
   return e;
+}
+
+/* D2H copy through a pinned bounce slot: pageable hipMemcpy D2H runs at
+ * ~50 GB/s with run-to-run stalls (the sparse 1M-group readback measured
+ * 1.4-3.6 ms of host time per query on identical kernels); the pinned DMA
+ * + one host memcpy is both faster and stable */
+static hipError_t d2h_copy(sn_engine *e, void *dst, const void *src, size_t n) {
+  if (n < (1u << 20))
+    return hipMemcpy(dst, src, n, hipMemcpyDeviceToHost);
+  auto &slot = e->pins[e->pin_rr.fetch_add(1) % sn_engine::PIN_SLOTS];
+  std::lock_guard<std::mutex> g(slot.mu);
+  if (slot.sz < n) {
+    size_t want = std::max<size_t>(n, 32u << 20);
+    void *p = nullptr;
+    if (hipHostMalloc(&p, want) != hipSuccess)
+      return hipMemcpy(dst, src, n, hipMemcpyDeviceToHost);  /* fall back */
+    if (slot.buf) (void)hipHostFree(slot.buf);
+    slot.buf = p;
+    slot.sz = want;
+  }
+  hipError_t rc = hipMemcpy(slot.buf, src, n, hipMemcpyDeviceToHost);
+  if (rc != hipSuccess) return rc;
+  memcpy(dst, slot.buf, n);
+  return hipSuccess;
 }
 
 /* H2D copy through a pinned bounce slot (big transfers; small ones direct) */
@@ -2742,11 +2772,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       q->sparse_keys.resize((size_t)ngrp);
       q->sparse_rows.resize((size_t)ngrp * naggs1);
       if (ngrp > 0) {
-        if (hipMemcpy(q->sparse_keys.data(), e->hws_okeys, (size_t)ngrp * 8,
-                      hipMemcpyDeviceToHost) != hipSuccess ||
-            hipMemcpy(q->sparse_rows.data(), e->hws_orows,
-                      (size_t)ngrp * naggs1 * 8,
-                      hipMemcpyDeviceToHost) != hipSuccess) {
+        if (d2h_copy(e, q->sparse_keys.data(), e->hws_okeys,
+                     (size_t)ngrp * 8) != hipSuccess ||
+            d2h_copy(e, q->sparse_rows.data(), e->hws_orows,
+                     (size_t)ngrp * naggs1 * 8) != hipSuccess) {
           fail(SN_ERR_GENERIC, "hash-agg readback"); return nullptr;
         }
       }
