@@ -1805,8 +1805,12 @@ struct sn_query {
   bool sparse = false;                  /* open-address hash-aggregate mode */
   bool pac = false;                     /* per-agg counts (nullable agg inputs) */
   bool mm = false;                      /* plan has MIN/MAX aggregates */
-  std::vector<long long> sparse_keys;   /* compacted group keys */
-  std::vector<double> sparse_rows;      /* [n][na1] accumulator rows */
+  /* compacted group keys + [n][na1] accumulator rows.  Raw arrays, NOT
+   * vectors: resize() would zero-fill the ~24 MB 1M-group readback before
+   * the copy overwrites it (measured ~1 ms/query of pure host overhead) */
+  std::unique_ptr<long long[]> sparse_keys;
+  std::unique_ptr<double[]> sparse_rows;
+  size_t sparse_n = 0;
   std::vector<double> sparse_null_row;  /* NULL-key group accumulator */
   bool gint[2] = { false, false };      /* integer group key (stats-ranged) */
   int64_t gmin[2] = { 0, 0 };           /* integer key minimum (slot base) */
@@ -2769,27 +2773,32 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       (void)hipMemcpy(q->sparse_null_row.data(),
                       e->hws_acc + (size_t)(cap + 1) * naggs1,
                       (size_t)naggs1 * 8, hipMemcpyDeviceToHost);
-      q->sparse_keys.resize((size_t)ngrp);
-      q->sparse_rows.resize((size_t)ngrp * naggs1);
+      q->sparse_keys.reset(new long long[(size_t)ngrp + 1]);
+      q->sparse_rows.reset(new double[((size_t)ngrp + 1) * naggs1]);
+      q->sparse_n = (size_t)ngrp;
       if (ngrp > 0) {
-        if (d2h_copy(e, q->sparse_keys.data(), e->hws_okeys,
-                     (size_t)ngrp * 8) != hipSuccess ||
-            d2h_copy(e, q->sparse_rows.data(), e->hws_orows,
-                     (size_t)ngrp * naggs1 * 8) != hipSuccess) {
+        /* plain pageable D2H: measured ~55 GB/s and stable once the heap
+         * recycles (mallopt in engine create) — the pinned bounce costs a
+         * second 24 MB host memcpy for nothing */
+        if (hipMemcpy(q->sparse_keys.get(), e->hws_okeys, (size_t)ngrp * 8,
+                      hipMemcpyDeviceToHost) != hipSuccess ||
+            hipMemcpy(q->sparse_rows.get(), e->hws_orows,
+                      (size_t)ngrp * naggs1 * 8,
+                      hipMemcpyDeviceToHost) != hipSuccess) {
           fail(SN_ERR_GENERIC, "hash-agg readback"); return nullptr;
         }
       }
       if (radix_direct) {
         /* the LDS compaction never scans the (untouched) global table, so
          * append the reserved sentinel-key row (a REAL key of -1) here —
-         * the same row k_hash_compact emits from index cap */
-        std::vector<double> rrow((size_t)naggs1, 0.0);
-        (void)hipMemcpy(rrow.data(), e->hws_acc + (size_t)cap * naggs1,
+         * the same row k_hash_compact emits from index cap (the arrays
+         * were sized ngrp+1 for exactly this) */
+        double *rrow = q->sparse_rows.get() + q->sparse_n * naggs1;
+        (void)hipMemcpy(rrow, e->hws_acc + (size_t)cap * naggs1,
                         (size_t)naggs1 * 8, hipMemcpyDeviceToHost);
         if (rrow[naggs1 - 1] != 0.0) {
-          q->sparse_keys.push_back(SN_HASH_EMPTY);
-          q->sparse_rows.insert(q->sparse_rows.end(), rrow.begin(),
-                                rrow.end());
+          q->sparse_keys[q->sparse_n] = SN_HASH_EMPTY;
+          q->sparse_n++;
         }
       }
       done_h = true;
@@ -2984,7 +2993,7 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
       }
       return g;
     };
-    for (size_t i = 0; i < q->sparse_keys.size(); i++) {
+    for (size_t i = 0; i < q->sparse_n; i++) {
       const double *row = &q->sparse_rows[i * naggs1];
       if (row[naggs1 - 1] == 0.0) continue;
       GroupOut g = fill(row);
