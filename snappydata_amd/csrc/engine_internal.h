@@ -22,6 +22,7 @@
 #define SN_ENGINE_INTERNAL_H
 
 #include <stdint.h>
+#include <stdlib.h>
 
 #define SN_DEV_MAX_COLS 8      /* referenced columns per plan */
 
@@ -203,8 +204,14 @@ static inline unsigned sn_radix_lds_bytes(int sub_log2, int naggs1) {
   return (unsigned)((1u << sub_log2) * (8u + 8u * (unsigned)naggs1) +
                     256 /*WG*/ * 4u);
 }
-/* does pass 2 compact DIRECTLY from LDS (host then skips k_hash_compact)? */
+/* does pass 2 compact DIRECTLY from LDS (host then skips k_hash_compact)?
+ * SN_RADIX_GLOB=1 forces the wide-row global-segment variant so its parity
+ * is testable at small sizes (it otherwise only triggers at cap 2^24 with
+ * 4 aggregates).  Shared by the host (compact decision) and the launcher
+ * (kernel choice) so the two can never disagree. */
 static inline int sn_radix_direct(int sub_log2, int naggs1) {
+  const char *fg = getenv("SN_RADIX_GLOB");
+  if (fg && fg[0] == '1') return 0;
   return sn_radix_lds_bytes(sub_log2, naggs1) <= SN_RADIX_LDS_MAX;
 }
 

@@ -1911,8 +1911,8 @@ extern "C" int sn_launch_radix_agg(const sn_dev_plan *plan,
                                    int *counter, void *stream) {
   const int npart = 1 << (plan->hcap_log2 - plan->radix);
   const int na1 = plan->naggs + 1;
-  const size_t lds = sn_radix_lds_bytes(plan->radix, na1);
-  if (lds <= SN_RADIX_LDS_MAX) {
+  if (sn_radix_direct(plan->radix, na1)) {
+    const size_t lds = sn_radix_lds_bytes(plan->radix, na1);
     hipLaunchKernelGGL(k_radix_agg_lds, dim3(npart), dim3(WG), lds,
                        (hipStream_t)stream, dev_plan, okeys, orows, counter);
   } else {

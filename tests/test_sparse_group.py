@@ -261,3 +261,40 @@ def test_sparse_nullable_int64_keys(eng):
     # the NULL group exists and counts exactly the invalid rows
     nulls = [v for k, v in grows if k[0] is None]
     assert len(nulls) == 1 and nulls[0][1] == float((valid == 0).sum())
+
+
+@pytest.mark.gpu
+def test_radix_glob_fallback_parity(eng, monkeypatch):
+    """The radix pass-2 GLOBAL-segment variant (accumulator rows too wide
+    for the LDS table — cap 2^24 with 4 aggregates in production, forced
+    here via SN_RADIX_GLOB so it parity-checks at test size): min/max +
+    sums through per-partition segments of the global table, then the
+    regular k_hash_compact readback."""
+    monkeypatch.setenv("SN_RADIX_GLOB", "1")
+    n = 2_000_000
+    ndistinct = 150_000
+    rng = np.random.default_rng(211)
+    universe = rng.integers(-2**62, 2**62, ndistinct).astype(np.int64)
+    universe[0] = -1                       # sentinel as a REAL key
+    keys = universe[rng.integers(0, ndistinct, n)]
+    w = rng.random(n)
+    v = rng.random(n) * 100 - 50
+    t = eng.table_define("tradixg", [(abi.T_INT64, False), (abi.T_DOUBLE, False),
+                                     (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": keys}, {"data": w}, {"data": v}], n,
+                       batch_rows=500_000)
+    plan_kw = dict(group_cols=[0],
+                   aggs=[("sum", [(1, 0.0, 1.0)]), ("min", [(2, 0.0, 1.0)]),
+                         ("max", [(2, 0.0, 1.0)]), ("count", [])])
+    q = eng.query(abi.make_plan(table=t, **plan_kw))
+    grows = q.rows()
+    assert q.used_jit()                    # radix pass 1 ran compiled
+    ot = po.OracleTable([po.T_INT64, po.T_DOUBLE, po.T_DOUBLE])
+    for st in range(0, n, 500_000):
+        en = min(n, st + 500_000)
+        ot.add_batch(en - st,
+                     [po.encode(po.T_INT64, po.ENC_UNCOMPRESSED, keys[st:en]),
+                      po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w[st:en]),
+                      po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, v[st:en])])
+    orows = ot.query_groups(po.make_plan(**plan_kw), nthreads=32)
+    assert_rows_match(grows, orows, count_aggs={3})
