@@ -2626,8 +2626,13 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
       dps.hflags = e->hws_flags;
       dps.hcap_log2 = cap_log2;
       bool radix = radix_ok && cap_log2 >= 17;
+      static const int sub_env = [] {      /* A/B lever for the sweep */
+        const char *v = getenv("SN_RADIX_SUB");
+        return v ? atoi(v) : 0;
+      }();
       const int sub_log2 =
-          std::max(SN_RADIX_SUB_MIN, cap_log2 - SN_RADIX_NPART_MAX_LOG2);
+          std::max(sub_env ? sub_env : SN_RADIX_SUB_MIN,
+                   cap_log2 - SN_RADIX_NPART_MAX_LOG2);
       if (radix) {
         const int npart = 1 << (cap_log2 - sub_log2);
         const long long percap =

@@ -193,11 +193,13 @@ typedef struct {
   int32_t radix;
 } sn_dev_plan;
 
-/* radix partition shift (plan->radix): slots per partition table.  11
- * (2048 slots) keeps the LDS table at 2 workgroups/CU for the common
- * sum+count row; capped so npart = cap >> radix never exceeds the 4096
- * the pass-1 histogram sizes for. */
-#define SN_RADIX_SUB_MIN 11
+/* radix partition shift (plan->radix): slots per partition table.
+ * Swept on the 1M-key bench: sub 9 (512-slot LDS tables, 13 KB, high
+ * occupancy) beat 10/11/12/13 at 25.7 / 24.9 / 21.2 / 17.0 / 6.5 Grows/s
+ * — pass 2 wants occupancy, not bigger tables.  Clamped below so
+ * npart = cap >> radix never exceeds the 4096 the pass-1 histogram
+ * sizes for. */
+#define SN_RADIX_SUB_MIN 9
 #define SN_RADIX_NPART_MAX_LOG2 12
 #define SN_RADIX_LDS_MAX (144u * 1024)
 static inline unsigned sn_radix_lds_bytes(int sub_log2, int naggs1) {
