@@ -99,9 +99,13 @@ WORKLOADS = {
 # star join, whose PMC wave-cycle profile shows probe issue-stall, not HBM
 # saturation (profiles/star_join_sf10_wavecycles_*.csv) — labeling it
 # "hbm" would overstate headroom against the 8 TB/s peak
-WORKLOAD_BOUND = {"star_join_sf10": "latency",
-                  # random open-address probes + HBM atomics, not streaming
-                  "sparse_group_sf10": "latency"}
+WORKLOAD_BOUND = {"star_join_sf10": "latency"}
+
+# pipeline workloads whose dominant-kernel algorithmic bytes exceed the
+# input bytes: the radix two-pass hash aggregate streams input (16) +
+# record write (16) + record read (16) per row; PMC traffic additionally
+# shows the 16 B-record scatter's sector amplification
+ALG_PIPELINE_BYTES = {"sparse_group_sf10": 48}
 
 
 def build_config1(eng, t, total_rows, seed, batch_rows=600_000):
@@ -476,7 +480,8 @@ def main():
         roofline = None
         if kms:
             avg_ms = float(np.mean(kms))
-            alg_bytes = resident * bytes_per_row     # this rank's launch
+            alg_bytes = resident * ALG_PIPELINE_BYTES.get(
+                args.workload, bytes_per_row)        # this rank's launch
             achieved = alg_bytes / (avg_ms / 1000.0) / 1e9
             roofline = {
                 "bound": WORKLOAD_BOUND.get(args.workload, "hbm"),
