@@ -254,6 +254,22 @@ __device__ __forceinline__ u64 mix64(u64 x) {
   }();
   const int fuse_mode = fuse_env && wbin_mode && p->jkeys && p->jlut &&
                         p->jmode == 1 && p->npreds_d + p->npreds_i == 0;
+  /* LDS-packed dense LUT: the star probe's dependent gathers stall on
+   * L2/HBM latency (measured 58% SQ_WAIT_INST issue-stall).  When the
+   * payload fits 4 bits (group-by-attr gid <= 14, or semi-join presence)
+   * and the packed span fits LDS, each workgroup packs the int32 LUT
+   * into nibbles ONCE at kernel start and probes LDS instead.
+   * SN_JIT_SLUT=0 reverts to the register-staged L2 probe. */
+  static const int slut_env = [] {
+    const char *v = getenv("SN_JIT_SLUT");
+    return !v || v[0] != '0';
+  }();
+  const long long slut_span = (p->jkeys && p->jlut)
+      ? (long long)(p->jlut_max - p->jlut_min + 1) : 0;
+  const long long slut_words = (slut_span + 7) / 8;
+  const int slut_mode = slut_env && !fuse_mode && slut_span > 0 &&
+                        (p->jmode == 0 || nslots <= 14) &&
+                        slut_words * 4 <= 100 * 1024;
   if (!fuse_mode) {
     emitf(o, "  __shared__ __attribute__((aligned(16))) double sval[%d][CHUNK];\n", NC);
     if (fold_preds)
@@ -334,8 +350,10 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     emitf(o, "#pragma unroll\n  for (int a = 0; a < %d; a++) { sums[a] = 0; cnts[a] = 0; }\n", NA);
   }
 
-  if (p->jkeys && p->jlut && !fuse_mode)
+  if (p->jkeys && p->jlut && !fuse_mode && !slut_mode)
     o += "  __shared__ int spay[CHUNK];\n";
+  if (slut_mode)
+    emitf(o, "  __shared__ unsigned slut[%lld];\n", slut_words);
 
   /* staged register buffers: per column, by width class (fuse mode adds a
    * second set, B, so the previous chunk processes under the current
@@ -624,6 +642,24 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     return o;
   }
 
+  if (slut_mode) {
+    /* pack the dense LUT into 4-bit LDS nibbles, once per workgroup:
+     * 15 = absent; group-by-attr payloads are gids <= 14 (structural
+     * gate), semi-join presence packs as 0 */
+    emitf(o, "  for (unsigned i = tid; i < %lldu; i += WG) {\n"
+             "    unsigned wd = 0u;\n"
+             "#pragma unroll\n"
+             "    for (int j = 0; j < 8; j++) {\n"
+             "      const long long ix = (long long)i * 8 + j;\n"
+             "      const int v = ix < %lldll ? jlut[ix] : -1;\n"
+             "      const unsigned nib = v < 0 ? 15u : %s;\n"
+             "      wd |= nib << (4 * j);\n"
+             "    }\n"
+             "    slut[i] = wd;\n"
+             "  }\n"
+             "  __syncthreads();\n",
+          slut_words, slut_span, p->jmode == 0 ? "0u" : "(unsigned)v");
+  }
   o += R"(
   for (int t = blockIdx.x; t < ntiles; t += gridDim.x) {
     const sn_dev_tile tile = tiles[t];
@@ -701,7 +737,7 @@ __device__ __forceinline__ u64 mix64(u64 x) {
           ind, (long long)p->jlut_min,
           ind, dst.c_str());
   };
-  const int use_spay = p->jkeys && p->jlut;
+  const int use_spay = p->jkeys && p->jlut && !slut_mode;
   o += "    int staged = 0;\n"
        "    if (tile.row_start + CHUNK <= tile_end) {\n";
   emit_load("tile.row_start", "      ");
@@ -881,7 +917,23 @@ __device__ __forceinline__ u64 mix64(u64 x) {
      * i64-ness); empty-slot sentinel = INT64_MIN, same as the interpreted
      * probe_sweep */
     int is_i64 = (p->i64_mask >> p->jcslot) & 1u;
-    if (p->jlut) {
+    if (slut_mode) {
+      /* 4-bit LDS probe: range check + one LDS read per row */
+      emitf(o, "        int pay = -1;\n"
+               "        {\n"
+               "          const double jx = sval[%d][r];\n"
+               "          const i64 key = %s;\n"
+               "          const i64 ix = key - %lldll;\n"
+               "          if (ix >= 0 && ix < %lldll) {\n"
+               "            const unsigned nib =\n"
+               "                (slut[ix >> 3] >> (((unsigned)ix & 7u) * 4u)) & 15u;\n"
+               "            pay = nib == 15u ? -1 : (int)nib;\n"
+               "          }\n"
+               "        }\n"
+               "        ok &= pay >= 0;\n",
+            p->jcslot, is_i64 ? "__double_as_longlong(jx)" : "(i64)jx",
+            (long long)p->jlut_min, slut_span);
+    } else if (p->jlut) {
       (void)is_i64;   /* probed from the staged registers into spay */
       o += "        const int pay = spay[r];\n"
            "        ok &= pay >= 0;\n";
