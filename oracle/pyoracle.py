@@ -356,7 +356,7 @@ class OracleTable:
             vnull.ctypes.data_as(C.POINTER(C.c_uint8)))
         assert n >= 0, f"sno_query_groups failed: {n}"
         assert n <= cap, f"{n} groups exceed cap {cap}"
-        ngroup = plan.ngroup if plan.join_dim < 0 or plan.join_mode != 1 else 1
+        ngroup = plan.ngroup if plan.join_dim < 0 or plan.join_mode != 1 else 1 + plan.ngroup
         naggs = plan.naggs
         out = []
         for i in range(n):

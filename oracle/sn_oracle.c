@@ -1069,22 +1069,25 @@ static int eval_batch(const sno_table *t, const sno_batch *b, const sn_plan *p,
     char keys[SN_MAX_GROUPS][SN_KEY_MAX];
     uint8_t knull[SN_MAX_GROUPS];
     memset(knull, 0, sizeof(knull));
+    int kbase = 0;
     if (join_group) {
       int32_t a0 = t->dim_attr_off[dim_ord], a1 = t->dim_attr_off[dim_ord + 1];
       int32_t L = a1 - a0 < SN_KEY_MAX - 1 ? a1 - a0 : SN_KEY_MAX - 1;
       memcpy(keys[0], t->dim_attr_pay + a0, (size_t)L);
       keys[0][L] = 0;
+      kbase = 1;   /* composite GROUP BY dim_attr, fact_col: attr first */
     }
-    for (int i = 0; i < p->ngroup; i++) {
+    for (int i = 0; i < p->ngroup && kbase + i < SN_MAX_GROUPS; i++) {
       int c = p->group_cols[i];
-      if (val_null[c]) { knull[i] = 1; keys[i][0] = 0; }
+      int ki = kbase + i;
+      if (val_null[c]) { knull[ki] = 1; keys[ki][0] = 0; }
       else if (t->dtypes[c] != SN_TYPE_STRING) {
         /* integer group key: decimal text (SHAMapAccessor serializes the
          * raw fixed-width key; text keeps the partial-block format shared) */
-        snprintf(keys[i], SN_KEY_MAX, "%lld", (long long)val_i[c]);
+        snprintf(keys[ki], SN_KEY_MAX, "%lld", (long long)val_i[c]);
       } else {
         int32_t L = val_slen[c] < SN_KEY_MAX - 1 ? val_slen[c] : SN_KEY_MAX - 1;
-        memcpy(keys[i], val_s[c], (size_t)L); keys[i][L] = 0;
+        memcpy(keys[ki], val_s[c], (size_t)L); keys[ki][L] = 0;
       }
     }
     sno_group *grp = gtab_get(g, keys, knull);
@@ -1331,7 +1334,8 @@ static int32_t sno_run(sno_table *t, const sn_plan *p, int32_t nthreads,
   int rc = SN_OK;
   const int join_group = p->join_dim != SN_JOIN_NONE &&
                          p->join_mode == SN_JOIN_GROUP && t->dim_hk != NULL;
-  const int eff_ngroup = join_group ? 1 : p->ngroup;
+  const int eff_ngroup = join_group ? 1 + p->ngroup : p->ngroup;
+  if (eff_ngroup > SN_MAX_GROUPS) return SN_ERR_UNSUPPORTED;
 
   sno_gtab g;
   gtab_init(&g, eff_ngroup, p->naggs);
@@ -1450,7 +1454,7 @@ SNO_EXPORT int32_t sno_query(sno_table *t, const sn_plan *p, sn_result *out,
   if (rc != SN_OK) return rc;
   const int join_group = p->join_dim != SN_JOIN_NONE &&
                          p->join_mode == SN_JOIN_GROUP && t->dim_hk != NULL;
-  const int eff_ngroup = join_group ? 1 : p->ngroup;
+  const int eff_ngroup = join_group ? 1 + p->ngroup : p->ngroup;
   if (g.n > SN_MAX_GROUP_SLOTS) {     /* fixed result page: use sno_query_groups */
     gtab_free(&g);
     return SN_ERR_OVERFLOW;

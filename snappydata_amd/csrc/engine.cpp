@@ -1802,6 +1802,8 @@ struct sn_query {
   std::vector<int32_t> used_cols;       /* cslot -> table col */
   int nslots = 0;
   int g1cap = 0, g2cap = 0;             /* per-group-col slot counts */
+  int fact_slots = 1;                   /* composite dim_attr x fact_col:
+                                           dense fact span (slot minor) */
   bool sparse = false;                  /* open-address hash-aggregate mode */
   bool pac = false;                     /* per-agg counts (nullable agg inputs) */
   bool mm = false;                      /* plan has MIN/MAX aggregates */
@@ -1980,9 +1982,10 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     if (kt != SN_TYPE_INT32 && kt != SN_TYPE_INT64) {
       fail(SN_ERR_UNSUPPORTED, "join key must be int32/int64"); return nullptr;
     }
-    if (plan->join_mode == SN_JOIN_GROUP && plan->ngroup > 0) {
-      fail(SN_ERR_UNSUPPORTED, "dim-attr grouping combined with fact group "
-           "columns is not in the round-1 path");
+    if (plan->join_mode == SN_JOIN_GROUP && plan->ngroup > 1) {
+      fail(SN_ERR_UNSUPPORTED,
+           "dim-attr grouping supports at most ONE fact group column "
+           "(attr + fact key fill the 2-key result row)");
       return nullptr;
     }
     if (use_col(plan->join_fact_col) < 0) {
@@ -2122,6 +2125,24 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   } else {
     q->nslots = 1;
   }
+  if (q->join_group && plan->ngroup >= 1) {
+    /* composite GROUP BY dim_attr, fact_col: attr-major slots over the
+     * dense fact slot space (sparse fact keys were rejected above) */
+    const int attr_cap = (int)q->join_dim->attr_dict.size();
+    if (attr_cap > SN_MAX_GROUP_SLOTS) {
+      fail(SN_ERR_UNSUPPORTED, "dim attr cardinality %d > %d",
+           attr_cap, SN_MAX_GROUP_SLOTS);
+      return nullptr;
+    }
+    q->fact_slots = q->nslots;
+    const int64_t total = (int64_t)attr_cap * q->fact_slots;
+    if (total > SN_BIG_GROUP_CAP) {
+      fail(SN_ERR_UNSUPPORTED, "attr x fact group cardinality %lld > %d",
+           (long long)total, SN_BIG_GROUP_CAP);
+      return nullptr;
+    }
+    q->nslots = (int)total;
+  }
   /* (na_t set after device-aggregate dedup below) */
 
   /* build device plan — canonical branchless forms:
@@ -2149,6 +2170,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     dp.jcap_log2 = jd->cap_log2;
     dp.jcslot = q->cslot_of_col[plan->join_fact_col];
     dp.jmode = plan->join_mode == SN_JOIN_GROUP ? 1 : 0;
+    if (q->join_group && plan->ngroup >= 1)
+      dp.jslot_mul = q->fact_slots;   /* composite dim_attr x fact_col */
     dp.jlut = jd->dev_lut;
     dp.jlut_min = jd->lut_min;
     dp.jlut_max = jd->lut_max;
@@ -3047,7 +3070,17 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
     }
     if (q->join_group) {
       std::lock_guard<std::mutex> gd(q->join_dim->mu);
-      g.keys[0] = q->join_dim->attr_dict[s];
+      if (p.ngroup >= 1) {
+        /* composite: attr-major, fact minor */
+        g.keys[0] = q->join_dim->attr_dict[s / q->fact_slots];
+        const int f = s % q->fact_slots;
+        int c0 = p.group_cols[0];
+        if (q->gint[0]) g.keys[1] = std::to_string(q->gmin[0] + f);
+        else if (f == q->gnull1) g.key_null[1] = true;
+        else g.keys[1] = t->gdict[c0][f];
+      } else {
+        g.keys[0] = q->join_dim->attr_dict[s];
+      }
     } else if (p.ngroup >= 1) {
       int c0 = p.group_cols[0];
       int g1 = s / q->g2cap;
@@ -3090,7 +3123,7 @@ static int32_t result_fill_page(sn_query *q, int64_t offset, sn_result *out) {
   }
   memset(out, 0, sizeof(*out));
   const sn_plan &p = q->plan;
-  out->ngroup = q->join_group ? 1 : p.ngroup;
+  out->ngroup = q->join_group ? 1 + p.ngroup : p.ngroup;
   out->naggs = p.naggs;
   int64_t total = (int64_t)q->final_groups.size();
   int64_t start = offset < total ? offset : total;
@@ -3310,7 +3343,7 @@ extern "C" int32_t sn_query_partials_sharded2(sn_query *q, int32_t world,
   memset(dst, 0, (size_t)bb * world);
   std::vector<GroupOut> groups;
   local_groups(q, &groups);
-  int eff_ngroup = q->join_group ? 1 : p.ngroup;
+  int eff_ngroup = q->join_group ? 1 + p.ngroup : p.ngroup;
   std::vector<int32_t> counts(world, 0);
   for (auto &g : groups) {
     uint64_t h = 1469598103934665603ull;
@@ -3382,7 +3415,7 @@ extern "C" int32_t sn_query_merge(sn_query *q, const void *blocks, int64_t strid
       const uint8_t *bp = (const uint8_t *)blocks + bi * stride;
       int32_t n; memcpy(&n, bp, 4);
       const PartialSlot *slots = (const PartialSlot *)(bp + 8);
-      int eff_ngroup = q->join_group ? 1 : p.ngroup;
+      int eff_ngroup = q->join_group ? 1 + p.ngroup : p.ngroup;
       for (int32_t i = 0; i < n; i++) {
         std::string key;
         for (int k = 0; k < eff_ngroup; k++) {

@@ -127,7 +127,8 @@ typedef struct {
   int32_t jcap_log2;           /* table capacity = 1 << jcap_log2 */
   int32_t jcslot;              /* fact key column slot */
   int32_t jmode;               /* 0 semi, 1 group-by-dim-attr */
-  int32_t _pad;
+  int32_t jslot_mul;           /* >0: composite GROUP BY dim_attr, fact_col —
+                                  slot = pay * jslot_mul + dense fact slot */
   /* dense probe LUT (built when the dim key span fits 16M entries):
    * pay = key in [jlut_min, jlut_max] ? jlut[key - jlut_min] : -1.
    * One dependent load per row instead of the open-address chain. */

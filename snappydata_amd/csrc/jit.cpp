@@ -119,7 +119,7 @@ struct sn_dev_plan {
   int npreds_d, npreds_i, naggs, ngroup, nslots, nused;
   unsigned i64_mask; int gcol[2];
   const i64 *jkeys; const int *jpayload;
-  int jcap_log2, jcslot, jmode, _pad;
+  int jcap_log2, jcslot, jmode, jslot_mul;
   const int *jlut; i64 jlut_min, jlut_max;
   i64 gbase[2]; int gmul0, _pad2;
   sn_dev_pred_d preds_d[8];
@@ -253,7 +253,8 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     return v && v[0] == '1';
   }();
   const int fuse_mode = fuse_env && wbin_mode && p->jkeys && p->jlut &&
-                        p->jmode == 1 && p->npreds_d + p->npreds_i == 0;
+                        p->jmode == 1 && p->jslot_mul == 0 &&
+                        p->npreds_d + p->npreds_i == 0;
   /* LDS-packed dense LUT: the star probe's dependent gathers stall on
    * L2/HBM latency (measured 58% SQ_WAIT_INST issue-stall).  When the
    * payload fits 4 bits (group-by-attr gid <= 14, or semi-join presence)
@@ -1006,7 +1007,13 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     }
   } else if (grouped) {
     if (p->jkeys && p->jmode == 1) {
-      o += "        const int slot = pay > 0 ? pay : 0;\n";
+      if (p->jslot_mul > 0 && p->ngroup >= 1)
+        /* composite GROUP BY dim_attr, fact_col (all literals) */
+        emitf(o, "        const int slot = (pay > 0 ? pay : 0) * %d +"
+                 " (int)((i64)sval[%d][r] - %lldll);\n",
+              p->jslot_mul, p->gcol[0], (long long)p->gbase[0]);
+      else
+        o += "        const int slot = pay > 0 ? pay : 0;\n";
     } else if (p->ngroup == 0) {
       /* keyless plan in the grouped (pac) layout: one slot */
       o += "        const int slot = 0;\n";

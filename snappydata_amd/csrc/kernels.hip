@@ -965,6 +965,18 @@ __global__ void k_grouped(sn_dev_plan plan,
                           P->gbase[1]);
           sslot[r] = (int16_t)slot;
         }
+      } else if (ngroup >= 1) {
+        /* composite GROUP BY dim_attr, fact_col: probe_sweep staged the
+         * attr gid; widen it by the dense fact-slot space (dead rows keep
+         * garbage slots — they are never accumulated) */
+        const int gc0 = plan.gcol[0];
+#pragma unroll
+        for (int k = 0; k < CHUNK / WG; k++) {
+          const int r = tid + k * WG;
+          const int slot = (int)sslot[r] * P->jslot_mul +
+              (int)((long long)sval[(size_t)gc0 * CHUNK + r] - P->gbase[0]);
+          sslot[r] = (int16_t)slot;
+        }
       }
 
       /* rowcount per slot */
@@ -1115,8 +1127,13 @@ void k_grouped_lds(sn_dev_plan plan,
         const int m = (int)((w >> (tid & 63)) & 1ull);
         if (!m) continue;
         int slot = 0;
-        if (jslot) slot = sslot[r];
-        else {
+        if (jslot) {
+          slot = sslot[r];
+          if (ngroup >= 1)   /* composite dim_attr x fact_col */
+            slot = slot * P->jslot_mul +
+                   (int)((long long)sval[(size_t)gc0 * CHUNK + r] -
+                         P->gbase[0]);
+        } else {
           if (ngroup >= 1)
             slot = (int)(((long long)sval[(size_t)gc0 * CHUNK + r] -
                           P->gbase[0]) * P->gmul0);
@@ -1348,8 +1365,12 @@ void k_grouped_global(sn_dev_plan plan,
         const int m = (int)((w >> (tid & 63)) & 1ull);
         if (!m) continue;
         long long slot = 0;
-        if (jslot) slot = sslot[r];
-        else {
+        if (jslot) {
+          slot = sslot[r];
+          if (ngroup >= 1)   /* composite dim_attr x fact_col */
+            slot = slot * P->jslot_mul +
+                   ((long long)sval[(size_t)gc0 * CHUNK + r] - P->gbase[0]);
+        } else {
           if (ngroup >= 1)
             slot = ((long long)sval[(size_t)gc0 * CHUNK + r] -
                     P->gbase[0]) * P->gmul0;

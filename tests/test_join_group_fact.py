@@ -1,0 +1,151 @@
+"""Composite GROUP BY dim_attr, fact_col (HashJoinExec feeding
+SnappyHashAggregateExec with mixed dimension/fact grouping keys,
+HashJoinExec.scala:285-520 + SnappyHashAggregateExec.scala:337-500): the
+probe payload (dim attr gid) widens by the dense fact slot space —
+slot = gid * fact_slots + fact_slot — closing the round-1
+"dim-attr grouping combined with fact group columns" restriction.
+Results carry TWO keys: the attr string first, then the fact key."""
+import numpy as np
+import pytest
+
+from oracle import pyoracle as po
+from snappydata_amd import abi, engine as se
+
+REL = 1e-9
+
+
+@pytest.fixture
+def eng():
+    e = se.Engine(device=0)
+    yield e
+    e.close()
+
+
+def _assert_match(grows, orows, count_idx):
+    assert sorted(k for k, _ in grows) == sorted(k for k, _ in orows)
+    om = {k: v for k, v in orows}
+    for gk, gv in grows:
+        ov = om[gk]
+        for a, (g, o) in enumerate(zip(gv, ov)):
+            if a in count_idx:
+                assert g == o, (gk, a, g, o)
+            else:
+                assert abs(g - o) <= REL * max(1.0, abs(o)), (gk, a, g, o)
+
+
+def test_oracle_composite_join_group_cpu():
+    n = 80_000
+    rng = np.random.default_rng(311)
+    key = rng.integers(0, 100, n).astype(np.int32)
+    cat = rng.integers(0, 5, n).astype(np.int32)
+    w = rng.random(n)
+    t = po.OracleTable([po.T_INT32, po.T_INT32, po.T_DOUBLE])
+    t.add_batch(n, [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, key),
+                    po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, cat),
+                    po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)])
+    dk = np.arange(0, 100, 2, dtype=np.int64)
+    t.set_dim(dk, [b"N%d" % (k % 3) for k in dk])
+    rows = t.query_groups(po.make_plan(
+        group_cols=[1], aggs=[("sum", [(2, 0.0, 1.0)]), ("count", [])],
+        join=dict(dim=0, fact_col=0, group=True)), nthreads=8)
+    mask = key % 2 == 0
+    import collections
+    ref = collections.defaultdict(lambda: [0.0, 0.0])
+    for k, c, v in zip(key[mask], cat[mask], w[mask]):
+        g = ("N%d" % (k % 3), str(c))
+        ref[g][0] += v
+        ref[g][1] += 1
+    assert len(rows) == len(ref) == 15
+    _assert_match(rows, [(k, tuple(v)) for k, v in ref.items()], {1})
+
+
+def _load_fact(eng, key, cat, w, cat_type=abi.T_INT32, batch_rows=200_000):
+    n = len(key)
+    t = eng.table_define("tfact", [(abi.T_INT32, False), (cat_type, False),
+                                   (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": key}, {"data": cat}, {"data": w}], n,
+                       batch_rows=batch_rows)
+    return t
+
+
+@pytest.mark.gpu
+def test_composite_int_fact_key_jit(eng):
+    """attr x dense-int fact key, clean batches: query-compiled path."""
+    n = 1_200_000
+    rng = np.random.default_rng(313)
+    key = rng.integers(0, 100_000, n).astype(np.int32)
+    cat = rng.integers(10, 60, n).astype(np.int32)       # dense span 50
+    w = rng.random(n)
+    t = _load_fact(eng, key, cat, w)
+    dk = np.sort(rng.choice(100_000, size=40_000, replace=False)).astype(np.int64)
+    attrs = [b"NATION_%d" % (int(k) % 8) for k in dk]
+    dim = eng.dim_define("supplier")
+    eng.dim_put(dim, dk, attrs)
+    plan_kw = dict(group_cols=[1],
+                   aggs=[("sum", [(2, 0.0, 1.0)]), ("count", [])],
+                   join=dict(dim=dim, fact_col=0, group=True))
+    q = eng.query(abi.make_plan(table=t, **plan_kw))
+    grows = q.rows()
+    assert q.used_jit()
+    ot = po.OracleTable([po.T_INT32, po.T_INT32, po.T_DOUBLE])
+    for st in range(0, n, 200_000):
+        en = min(n, st + 200_000)
+        ot.add_batch(en - st,
+                     [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, key[st:en]),
+                      po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, cat[st:en]),
+                      po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w[st:en])])
+    ot.set_dim(dk, attrs)
+    orows = ot.query_groups(po.make_plan(**plan_kw), nthreads=16)
+    assert len(grows) == len(orows) > 0
+    _assert_match(grows, orows, {1})
+
+
+@pytest.mark.gpu
+def test_composite_string_fact_key(eng):
+    """attr x dictionary-string fact key (premultiplied gids, mul 1)."""
+    n = 400_000
+    rng = np.random.default_rng(317)
+    key = rng.integers(0, 1_000, n).astype(np.int32)
+    marks = [b"AF", b"BK", b"CN", b"DQ"]
+    cat = [marks[v] for v in rng.integers(0, 4, n)]
+    w = rng.random(n)
+    t = eng.table_define("tfs", [(abi.T_INT32, False), (abi.T_STRING, False),
+                                 (abi.T_DOUBLE, False)])
+    cols = [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, key),
+            po.encode(po.T_STRING, po.ENC_DICT, cat),
+            po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)]
+    eng.batch_put(t, 0, 0, n, cols)
+    dk = np.arange(0, 1_000, 3, dtype=np.int64)
+    attrs = [b"R%d" % (int(k) % 5) for k in dk]
+    dim = eng.dim_define("d2")
+    eng.dim_put(dim, dk, attrs)
+    plan_kw = dict(group_cols=[1],
+                   aggs=[("sum", [(2, 0.0, 1.0)]), ("avg", [(2, 0.0, 1.0)]),
+                         ("count", [])],
+                   join=dict(dim=dim, fact_col=0, group=True))
+    grows = eng.query(abi.make_plan(table=t, **plan_kw)).rows()
+    ot = po.OracleTable([po.T_INT32, po.T_STRING, po.T_DOUBLE])
+    ot.add_batch(n, cols)
+    ot.set_dim(dk, attrs)
+    orows = ot.query_groups(po.make_plan(**plan_kw), nthreads=16)
+    assert len(grows) == len(orows) == 20
+    _assert_match(grows, orows, {2})
+
+
+@pytest.mark.gpu
+def test_composite_two_fact_cols_rejected(eng):
+    n = 10_000
+    rng = np.random.default_rng(331)
+    key = rng.integers(0, 100, n).astype(np.int32)
+    c1 = rng.integers(0, 5, n).astype(np.int32)
+    w = rng.random(n)
+    t = eng.table_define("trej", [(abi.T_INT32, False), (abi.T_INT32, False),
+                                  (abi.T_DOUBLE, False)])
+    eng.ingest_columns(t, [{"data": key}, {"data": c1}, {"data": w}], n)
+    dim = eng.dim_define("d3")
+    eng.dim_put(dim, np.arange(100, dtype=np.int64),
+                [b"A%d" % (i % 2) for i in range(100)])
+    with pytest.raises(se.EngineError):
+        eng.query(abi.make_plan(table=t, group_cols=[1, 2],
+                                aggs=[("count", [])],
+                                join=dict(dim=dim, fact_col=0, group=True)))
