@@ -59,6 +59,15 @@ def main():
         join=dict(dim=dim, fact_col=6)))
     print("rows shipping in 1994:", int(qj.rows()[0][1][0]))
 
+    # -- composite GROUP BY dim_attr, fact_col -----------------------------
+    qjg = eng.query(abi.make_plan(
+        table=t, group_cols=[4],                  # fact returnflag
+        aggs=[("sum", [(0, 0.0, 1.0)]), ("count", [])],
+        join=dict(dim=dim, fact_col=6, group=True)))
+    print("1994 qty by (year, returnflag):")
+    for key, vals in qjg.rows():
+        print("  ", key, "->", round(vals[0], 2), int(vals[1]))
+
     # -- sparse group-by (open-address hash aggregate) + ORDER BY / TOP-K --
     rng = np.random.default_rng(7)
     sk = rng.integers(0, 50_000, 1_000_000).astype(np.int64) * (1 << 30)
