@@ -149,3 +149,61 @@ def test_composite_two_fact_cols_rejected(eng):
         eng.query(abi.make_plan(table=t, group_cols=[1, 2],
                                 aggs=[("count", [])],
                                 join=dict(dim=dim, fact_col=0, group=True)))
+
+
+@pytest.mark.gpu
+def test_composite_partials_split_merge(eng):
+    """Composite (attr, fact) keys through the key-sharded partial
+    exchange: two shard engines export variable-capacity blocks, cross-
+    merging reproduces the unsharded result (PartialSlot carries BOTH
+    keys; the shard hash covers the composite)."""
+    n = 200_000
+    rng = np.random.default_rng(337)
+    key = rng.integers(0, 2_000, n).astype(np.int32)
+    cat = rng.integers(0, 30, n).astype(np.int32)
+    w = rng.random(n)
+    dk = np.arange(0, 2_000, 2, dtype=np.int64)
+    attrs = [b"Z%d" % (int(k) % 7) for k in dk]
+    plan_kw = dict(group_cols=[1],
+                   aggs=[("sum", [(2, 0.0, 1.0)]), ("count", [])])
+
+    def load(e2):
+        t2 = e2.table_define("t", [(abi.T_INT32, False), (abi.T_INT32, False),
+                                   (abi.T_DOUBLE, False)])
+        bi = 0
+        for st in range(0, n, 50_000):
+            en = min(n, st + 50_000)
+            e2.ingest_columns(t2, [{"data": key[st:en]}, {"data": cat[st:en]},
+                                   {"data": w[st:en]}], en - st,
+                              batch_rows=50_000, first_bucket=bi)
+            bi += 1
+        d2 = e2.dim_define("d")
+        e2.dim_put(d2, dk, attrs)
+        return t2, d2
+
+    t_all, d_all = load(eng)
+    ref = eng.query(abi.make_plan(table=t_all, **plan_kw,
+                                  join=dict(dim=d_all, fact_col=0,
+                                            group=True))).rows()
+    assert len(ref) == 7 * 30
+
+    e0 = se.Engine(device=0, shard_rank=0, shard_count=2)
+    e1 = se.Engine(device=0, shard_rank=1, shard_count=2)
+    try:
+        t0, d0 = load(e0)
+        t1, d1 = load(e1)
+        q0 = e0.query(abi.make_plan(table=t0, **plan_kw,
+                                    join=dict(dim=d0, fact_col=0, group=True)))
+        q1 = e1.query(abi.make_plan(table=t1, **plan_kw,
+                                    join=dict(dim=d1, fact_col=0, group=True)))
+        cap = max(1024, q0.num_groups(), q1.num_groups())
+        bb = q0.partial_bytes(cap)
+        s0 = q0.partials_sharded(2, cap)
+        s1 = q1.partials_sharded(2, cap)
+        q0.merge_host(np.ascontiguousarray(np.concatenate([s0[0], s1[0]])), bb, 2)
+        q1.merge_host(np.ascontiguousarray(np.concatenate([s0[1], s1[1]])), bb, 2)
+        merged = sorted(q0.rows() + q1.rows())
+        _assert_match(merged, ref, {1})
+    finally:
+        e0.close()
+        e1.close()
