@@ -201,8 +201,16 @@ __device__ __forceinline__ u64 mix64(u64 x) {
            "  const GAS sn_dev_plan *P = (const GAS sn_dev_plan *)(u64)plan_p;\n"
            "  (void)jkeys; (void)jpayload; (void)jlut; (void)P;\n",
         /* sparse is random-probe latency-bound: more waves cover the
-         * dependent loads (its LDS footprint is only the sval image) */
-        sparse_mode ? 4 : (lds_mode || glob_mode) ? 1 : (wbin_pre ? 4 : 2));
+         * dependent loads (its LDS footprint is only the sval image).
+         * SN_JIT_SPOCC overrides for occupancy sweeps. */
+        [&] {
+          static const int so = [] {
+            const char *v = getenv("SN_JIT_SPOCC");
+            return v ? atoi(v) : 0;
+          }();
+          if (sparse_mode) return so > 0 ? so : 4;
+          return (lds_mode || glob_mode) ? 1 : (wbin_pre ? 4 : 2);
+        }());
   /* tokenized plan values (the reference's ParamLiteral tokenization,
    * TokenizationTest / SnappySession plan cache): predicate bounds and
    * aggregate coefficients load once per wave from the cached device plan
