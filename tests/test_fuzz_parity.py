@@ -1,0 +1,226 @@
+"""Seeded randomized parity fuzz: random schemas (types, nullability),
+random reference-format encodings (uncompressed / RLE / dictionary),
+ragged multi-batch tables with delete masks, and random plans (range +
+IN predicates, SUM/AVG/MIN/MAX/COUNT over 1-2 factor products, dense and
+sparse group keys, broadcast joins incl. composite dim-attr grouping) —
+every case cross-checked GPU engine vs the CPU oracle.
+
+The generator only emits combinations the engine DECLARES supported;
+anything it still rejects (EngineError) is counted and skipped so new
+restrictions surface as a falling `ran` count, not a red test.  Bars per
+the north star: counts exact, doubles <= 1e-6 relative.
+
+Run a bigger sweep with SN_FUZZ_N=<cases> (default 30).
+"""
+import os
+
+import numpy as np
+import pytest
+
+from oracle import pyoracle as po
+from snappydata_amd import abi, engine as se
+
+REL = 1e-6
+T_NUM = [(po.T_INT32, abi.T_INT32), (po.T_INT64, abi.T_INT64),
+         (po.T_DOUBLE, abi.T_DOUBLE), (po.T_FLOAT, abi.T_FLOAT),
+         (po.T_INT16, abi.T_INT16)]
+VOCAB = [b"AAA", b"BETA", b"CC", b"DELTA", b"EVE", b"FOX", b"GOLF", b"HOP"]
+
+
+def _gen_col(rng, dtype, n, nullable):
+    if dtype == po.T_STRING:
+        vals = [VOCAB[v] for v in rng.integers(0, len(VOCAB), n)]
+        if nullable:
+            for i in np.flatnonzero(rng.random(n) < 0.12):
+                vals[i] = None
+        return vals
+    if dtype == po.T_DOUBLE:
+        v = (rng.random(n) * 200 - 100)
+    elif dtype == po.T_FLOAT:
+        v = (rng.random(n) * 50).astype(np.float32)
+    elif dtype == po.T_INT64:
+        v = rng.integers(-(1 << 40), 1 << 40, n)
+    elif dtype == po.T_INT32:
+        v = rng.integers(-5_000, 5_000, n).astype(np.int32)
+    else:
+        v = rng.integers(-300, 300, n).astype(np.int16)
+    valid = None
+    if nullable:
+        valid = (rng.random(n) >= 0.12).astype(np.uint8)
+    return v, valid
+
+
+def _encode(rng, dtype, col):
+    if dtype == po.T_STRING:
+        return po.encode(po.T_STRING, po.ENC_DICT, col)
+    v, valid = col
+    enc = po.ENC_UNCOMPRESSED
+    if dtype in (po.T_INT32, po.T_INT64) and valid is None and rng.random() < 0.3:
+        enc = po.ENC_RLE
+    return po.encode(dtype, enc, v, valid=valid)
+
+
+def _run_case(eng, seed):
+    rng = np.random.default_rng(seed)
+    ncols = int(rng.integers(2, 6))
+    schema = []
+    for c in range(ncols):
+        if c > 0 and rng.random() < 0.25:
+            dtype = po.T_STRING
+        else:
+            dtype = T_NUM[rng.integers(0, len(T_NUM))][0]
+        nullable = bool(rng.random() < 0.3)
+        schema.append((dtype, nullable))
+    a_schema = [(dict(T_NUM)[d] if d != po.T_STRING else abi.T_STRING, nb)
+                for d, nb in schema]
+
+    t = eng.table_define(f"fz{seed}", a_schema)
+    ot = po.OracleTable([d for d, _ in schema])
+    nbatches = int(rng.integers(1, 4))
+    cols_by_batch = []
+    for b in range(nbatches):
+        n = int(rng.integers(1_000, 40_000))
+        raw = [_gen_col(rng, d, n, nb) for d, nb in schema]
+        blobs = [_encode(rng, schema[c][0], raw[c]) for c in range(ncols)]
+        dmask = None
+        if rng.random() < 0.3:
+            dels = np.unique(rng.integers(0, n, max(1, n // 20))).astype(np.int32)
+            dmask = po.encode_delete(dels, n)
+        eng.batch_put(t, 100 + b, b, n, blobs, delete_mask=dmask)
+        ot.add_batch(n, blobs, delete_mask=dmask)
+        cols_by_batch.append((n, raw))
+
+    # ---- random plan over the declared-supported surface ----
+    num_cols = [c for c in range(ncols) if schema[c][0] != po.T_STRING]
+    str_cols = [c for c in range(ncols) if schema[c][0] == po.T_STRING]
+    preds = []
+    for c in rng.permutation(num_cols)[: rng.integers(0, 3)]:
+        d = schema[c][0]
+        is_d = d in (po.T_DOUBLE, po.T_FLOAT)
+        p = dict(col=int(c))
+        if is_d:
+            p["is_double"] = True
+            lo, hi = sorted(rng.random(2) * 200 - 100)
+        elif d == po.T_INT64:
+            lo, hi = sorted(rng.integers(-(1 << 40), 1 << 40, 2).tolist())
+        else:
+            lo, hi = sorted(rng.integers(-4_000, 4_000, 2).tolist())
+        if rng.random() < 0.8:
+            p["lo"] = float(lo) if is_d else int(lo)
+            p["lo_strict"] = bool(rng.random() < 0.3)
+        if rng.random() < 0.8:
+            p["hi"] = float(hi) if is_d else int(hi)
+            p["hi_strict"] = bool(rng.random() < 0.3)
+        if "lo" in p or "hi" in p:
+            preds.append(p)
+    if str_cols and rng.random() < 0.4:
+        c = int(str_cols[0])
+        picks = [VOCAB[i] for i in
+                 rng.choice(len(VOCAB), size=rng.integers(1, 4), replace=False)]
+        if rng.random() < 0.5:
+            preds.append({"col": c, "in": picks})
+        else:
+            preds.append({"col": c, "eq": picks[0]})
+    elif num_cols and rng.random() < 0.25:
+        c = int(rng.choice(num_cols))
+        if schema[c][0] in (po.T_INT16, po.T_INT32):
+            picks = rng.integers(-4_000, 4_000, 4).tolist()
+            preds.append({"col": c, "in": [int(x) for x in picks]})
+
+    aggs = []
+    navg = int(rng.integers(1, 4))
+    for _ in range(navg):
+        kind = ["sum", "avg", "min", "max"][rng.integers(0, 4)]
+        nf = 1 if rng.random() < 0.7 or len(num_cols) < 2 else 2
+        fac = []
+        for fc in rng.choice(num_cols, size=nf, replace=False) if num_cols else []:
+            add = float(rng.integers(-2, 3))
+            mul = float(rng.choice([1.0, 1.0, 0.5, 2.0]))
+            fac.append((int(fc), add, mul))
+        if not fac:
+            continue
+        aggs.append((kind, fac))
+    aggs.append(("count", []))
+
+    group_cols = []
+    join = None
+    r = rng.random()
+    if r < 0.45:
+        cands = str_cols + [c for c in num_cols
+                            if schema[c][0] in (po.T_INT16, po.T_INT32,
+                                                po.T_INT64)]
+        if cands:
+            k = int(rng.choice(cands))
+            group_cols = [k]
+            # second key: only combos the engine declares (no mixed
+            # string+sparse, no int64 pairs, no nullable int pairs)
+            if (rng.random() < 0.3 and schema[k][0] == po.T_STRING and
+                    len(str_cols) > 1):
+                group_cols.append(int(str_cols[1]))
+    elif r < 0.65 and num_cols:
+        jc = next((c for c in num_cols
+                   if schema[c][0] in (po.T_INT32, po.T_INT64) and
+                   not schema[c][1]), None)
+        if jc is not None:
+            dk = np.unique(rng.integers(-5_000, 5_000, 3_000)).astype(np.int64)
+            attrs = [b"G%d" % (int(x) % 6) for x in dk]
+            dim = eng.dim_define(f"dz{seed}")
+            eng.dim_put(dim, dk, attrs)
+            ot.set_dim(dk, attrs)
+            grouped_join = bool(rng.random() < 0.6)
+            join = dict(dim=dim, fact_col=int(jc), group=grouped_join)
+            ojoin = dict(dim=0, fact_col=int(jc), group=grouped_join)
+            if grouped_join and rng.random() < 0.5:
+                fk = next((c for c in str_cols if c != jc), None)
+                if fk is None:
+                    fk = next((c for c in num_cols
+                               if c != jc and
+                               schema[c][0] in (po.T_INT16, po.T_INT32) and
+                               not schema[c][1]), None)
+                if fk is not None:
+                    group_cols = [int(fk)]
+
+    plan_kw = dict(preds=preds, group_cols=group_cols, aggs=aggs)
+    try:
+        q = eng.query(abi.make_plan(table=t, **plan_kw, join=join))
+        grows = q.rows()
+    except se.EngineError:
+        return None                      # declared-unsupported combo: skip
+    o_plan = po.make_plan(**plan_kw, join=None if join is None else ojoin)
+    is_grouped = bool(group_cols) or (join is not None and join["group"])
+    orows = (ot.query_groups(o_plan, nthreads=8) if is_grouped
+             else po.result_rows(ot.query(o_plan)))
+    count_idx = {i for i, (k, _) in enumerate(aggs) if k == "count"}
+
+    gk = sorted(k for k, _ in grows)
+    ok = sorted(k for k, _ in orows)
+    assert gk == ok, f"seed {seed}: key sets differ ({len(gk)} vs {len(ok)})"
+    om = {k: v for k, v in orows}
+    for k, gv in grows:
+        ov = om[k]
+        for a, (gx, ox) in enumerate(zip(gv, ov)):
+            if a in count_idx:
+                assert gx == ox, (seed, k, a, gx, ox)
+            elif ox is None or gx is None:
+                assert gx == ox, (seed, k, a, gx, ox)
+            else:
+                assert abs(gx - ox) <= REL * max(1.0, abs(ox)), (seed, k, a, gx, ox)
+    return len(grows)
+
+
+@pytest.mark.gpu
+def test_fuzz_parity_gpu():
+    n_cases = int(os.environ.get("SN_FUZZ_N", "30"))
+    eng = se.Engine(device=0)
+    try:
+        ran = skipped = 0
+        for seed in range(1000, 1000 + n_cases):
+            r = _run_case(eng, seed)
+            if r is None:
+                skipped += 1
+            else:
+                ran += 1
+        print(f"fuzz: {ran} cases verified, {skipped} skipped-unsupported")
+        assert ran >= max(1, int(0.6 * n_cases))
+    finally:
+        eng.close()
