@@ -192,8 +192,9 @@ def _run_case(eng, seed):
              else po.result_rows(ot.query(o_plan)))
     count_idx = {i for i, (k, _) in enumerate(aggs) if k == "count"}
 
-    gk = sorted(k for k, _ in grows)
-    ok = sorted(k for k, _ in orows)
+    knorm = lambda t: tuple((x is None, x if x is not None else "") for x in t)
+    gk = sorted((k for k, _ in grows), key=knorm)
+    ok = sorted((k for k, _ in orows), key=knorm)
     assert gk == ok, f"seed {seed}: key sets differ ({len(gk)} vs {len(ok)})"
     om = {k: v for k, v in orows}
     for k, gv in grows:
