@@ -1960,6 +1960,14 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
     }
     if (use_col(c) < 0) { fail(SN_ERR_BADARG, "bad group col"); return nullptr; }
   }
+  if (plan->ngroup == 2 && plan->group_cols[0] == plan->group_cols[1]) {
+    /* both keys share one device column slot, which cannot carry two
+     * different premultiplied dictionary images — found by the parity
+     * fuzzer producing off-diagonal keys.  GROUP BY a, a is degenerate;
+     * reject loudly so the planner dedups upstream. */
+    fail(SN_ERR_UNSUPPORTED, "duplicate group column (dedup upstream)");
+    return nullptr;
+  }
   for (int a = 0; a < plan->naggs; a++)
     for (int f = 0; f < plan->aggs[a].nfactors; f++) {
       int c = plan->aggs[a].factors[f].col;

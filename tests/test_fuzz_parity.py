@@ -154,9 +154,10 @@ def _run_case(eng, seed):
             group_cols = [k]
             # second key: only combos the engine declares (no mixed
             # string+sparse, no int64 pairs, no nullable int pairs)
-            if (rng.random() < 0.3 and schema[k][0] == po.T_STRING and
-                    len(str_cols) > 1):
-                group_cols.append(int(str_cols[1]))
+            if rng.random() < 0.3 and schema[k][0] == po.T_STRING:
+                s2 = next((s for s in str_cols if s != k), None)
+                if s2 is not None:
+                    group_cols.append(int(s2))
     elif r < 0.65 and num_cols:
         jc = next((c for c in num_cols
                    if schema[c][0] in (po.T_INT32, po.T_INT64) and
