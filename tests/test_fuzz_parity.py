@@ -74,6 +74,9 @@ def _run_case(eng, seed):
     a_schema = [(dict(T_NUM)[d] if d != po.T_STRING else abi.T_STRING, nb)
                 for d, nb in schema]
 
+    num_cols = [c for c in range(ncols) if schema[c][0] != po.T_STRING]
+    str_cols = [c for c in range(ncols) if schema[c][0] == po.T_STRING]
+
     t = eng.table_define(f"fz{seed}", a_schema)
     ot = po.OracleTable([d for d, _ in schema])
     nbatches = int(rng.integers(1, 4))
@@ -86,13 +89,38 @@ def _run_case(eng, seed):
         if rng.random() < 0.3:
             dels = np.unique(rng.integers(0, n, max(1, n // 20))).astype(np.int32)
             dmask = po.encode_delete(dels, n)
-        eng.batch_put(t, 100 + b, b, n, blobs, delete_mask=dmask)
-        ot.add_batch(n, blobs, delete_mask=dmask)
+        deltas = None
+        if rng.random() < 0.25 and num_cols:
+            # 2-deep update deltas on one numeric column
+            deltas = [(None, None)] * ncols
+            dc = int(rng.choice(num_cols))
+            d = schema[dc][0]
+            for depth in range(1 if rng.random() < 0.5 else 2):
+                upd = np.unique(rng.integers(0, n, max(1, n // 40))).astype(np.int32)
+                if d == po.T_DOUBLE:
+                    dv = rng.random(len(upd)) * 100
+                elif d == po.T_FLOAT:
+                    dv = (rng.random(len(upd)) * 10).astype(np.float32)
+                elif d == po.T_INT64:
+                    dv = rng.integers(-(1 << 40), 1 << 40, len(upd))
+                elif d == po.T_INT32:
+                    dv = rng.integers(-5_000, 5_000, len(upd)).astype(np.int32)
+                else:
+                    dv = rng.integers(-300, 300, len(upd)).astype(np.int16)
+                blob = po.encode_delta(d, po.ENC_UNCOMPRESSED, upd, n, dv)
+                d1, d2 = deltas[dc]
+                deltas[dc] = (blob, d1) if depth else (blob, None)
+        if rng.random() < 0.2:
+            from tests.test_compression import wrap_lz4
+            blobs = [wrap_lz4(bl) if rng.random() < 0.5 else bl
+                     for bl in blobs]
+        eng.batch_put(t, 100 + b, b, -n if deltas else n, blobs,
+                      delete_mask=dmask, deltas=deltas)
+        ot.add_batch(-n if deltas else n, blobs, delete_mask=dmask,
+                     deltas=deltas)
         cols_by_batch.append((n, raw))
 
     # ---- random plan over the declared-supported surface ----
-    num_cols = [c for c in range(ncols) if schema[c][0] != po.T_STRING]
-    str_cols = [c for c in range(ncols) if schema[c][0] == po.T_STRING]
     preds = []
     for c in rng.permutation(num_cols)[: rng.integers(0, 3)]:
         d = schema[c][0]
