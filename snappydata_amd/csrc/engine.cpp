@@ -3064,6 +3064,12 @@ static void local_groups(sn_query *q, std::vector<GroupOut> *out) {
       if (p.ngroup == 0 && !q->join_group) {
         /* keyless MIN/MAX plans run the grouped (pac) layout with 1 slot:
          * counts land at [naggs + a] instead of the keyless [na_t + a] */
+        if (a == 0 && getenv("SN_DBG")) {
+          fprintf(stderr, "[dbg] pac=%d dev_naggs=%d na_t=%d p.naggs=%d stride=%d row=",
+                  q->pac ? 1 : 0, q->dev_naggs, q->na_t, p.naggs, (int)q->out_stride);
+          for (size_t i = 0; i < q->out_stride; i++) fprintf(stderr, "%g ", row[i]);
+          fprintf(stderr, "\n");
+        }
         g.sums[a] = row[a];
         g.counts[a] = q->pac ? row[p.naggs + a] : row[q->na_t + a];
       } else {

@@ -802,7 +802,10 @@ __device__ __forceinline__ u64 mix64(u64 x) {
     }
   }
   o += "      } else {\n"
-       "        /* scalar tail conversion */\n";
+       "        /* scalar tail conversion (the [rows, CHUNK) image tail is\n"
+       "         * zeroed below: raw LDS can hold NaN-pattern bits from the\n"
+       "         * previous kernel, and fma(0, NaN, sum) = NaN would poison\n"
+       "         * the slot-predicated register accumulators) */\n";
   if (fold_preds)
     o += "        for (int i = tid; i < 16; i += WG) sbits[i] = 0ull;\n"
          "        __syncthreads();\n";
@@ -826,6 +829,9 @@ __device__ __forceinline__ u64 mix64(u64 x) {
         emitf(o, "(double)((const GAS short *)(body%d))[base + r];\n", c); break;
     }
   }
+  for (int c = 0; c < NC; c++)
+    emitf(o, "        for (int r = rows + tid; r < CHUNK; r += WG)"
+             " sval[%d][r] = 0.0;\n", c);
   if (fold_preds) {
     /* same thread wrote these sval rows above — no barrier needed */
     o += "        for (int r = tid; r < rows; r += WG) {\n"
@@ -1276,6 +1282,14 @@ extern "C" void *sn_jit_get(void *cache, const sn_dev_plan *p,
   mix(&shape, sizeof(shape));
   mix(kinds, sizeof(int) * SN_DEV_MAX_COLS);
   mix(&nslots, 4); mix(&na_t, 4); mix(&has_del, 4);
+  if (getenv("SN_DBG_JIT")) {
+    fprintf(stderr, "[jit] h=%016llx na=%d ns=%d pac=%d sparse=%d ops=",
+            (unsigned long long)h, p->naggs, nslots, p->pac, p->sparse);
+    for (int a = 0; a < p->naggs; a++) fprintf(stderr, "%d", p->aggs[a].op);
+    fprintf(stderr, " nf=");
+    for (int a = 0; a < p->naggs; a++) fprintf(stderr, "%d", p->aggs[a].nf);
+    fprintf(stderr, "\n");
+  }
   {
     std::lock_guard<std::mutex> g(jc->mu);
     auto it = jc->fns.find(h);

@@ -294,6 +294,19 @@ __device__ __forceinline__ void convert_chunk(
       if ((tid & 63) == 0) sdead[r >> 6] = w;
     }
   }
+  /* short chunks MUST zero the image tail: sval beyond `rows` is raw LDS
+   * left by whatever kernel ran last on this CU.  Dead rows never
+   * contribute through selects, but the slot-predicated fma accumulators
+   * compute fma(0, va, sum) — and fma(0, NaN, s) = NaN, so an ord-ident
+   * bit pattern (0xFFF8...) parked in LDS silently poisons every sum in
+   * the wave (found by the randomized parity fuzzer as a
+   * prior-query-dependent NaN). */
+  if (rows < CHUNK) {
+    for (int c = 0; c < nused; c++) {
+      double *dst = sval + (size_t)c * CHUNK;
+      for (int r = rows + tid; r < CHUNK; r += WG) dst[r] = 0.0;
+    }
+  }
 }
 
 /* ---- staged (software-pipelined) conversion for clean full chunks ----
