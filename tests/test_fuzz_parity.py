@@ -1,5 +1,6 @@
 """Seeded randomized parity fuzz: random schemas (types, nullability),
-random reference-format encodings (uncompressed / RLE / dictionary),
+random reference-format encodings (uncompressed / RLE / dictionary /
+boolean-bitset),
 ragged multi-batch tables with delete masks, and random plans (range +
 IN predicates, SUM/AVG/MIN/MAX/COUNT over 1-2 factor products, dense and
 sparse group keys, broadcast joins incl. composite dim-attr grouping) —
@@ -28,6 +29,12 @@ VOCAB = [b"AAA", b"BETA", b"CC", b"DELTA", b"EVE", b"FOX", b"GOLF", b"HOP"]
 
 
 def _gen_col(rng, dtype, n, nullable):
+    if dtype == po.T_BOOL:
+        v = rng.integers(0, 2, n).astype(np.uint8)
+        valid = None
+        if nullable:
+            valid = (rng.random(n) >= 0.12).astype(np.uint8)
+        return v, valid
     if dtype == po.T_STRING:
         vals = [VOCAB[v] for v in rng.integers(0, len(VOCAB), n)]
         if nullable:
@@ -57,6 +64,8 @@ def _encode(rng, dtype, col):
     enc = po.ENC_UNCOMPRESSED
     if dtype in (po.T_INT32, po.T_INT64) and valid is None and rng.random() < 0.3:
         enc = po.ENC_RLE
+    elif dtype == po.T_BOOL and valid is None and rng.random() < 0.6:
+        enc = po.ENC_BOOLBITSET
     return po.encode(dtype, enc, v, valid=valid)
 
 
@@ -67,12 +76,16 @@ def _run_case(eng, seed):
     for c in range(ncols):
         if c > 0 and rng.random() < 0.25:
             dtype = po.T_STRING
+        elif c > 0 and rng.random() < 0.15:
+            dtype = po.T_BOOL
         else:
             dtype = T_NUM[rng.integers(0, len(T_NUM))][0]
         nullable = bool(rng.random() < 0.3)
         schema.append((dtype, nullable))
-    a_schema = [(dict(T_NUM)[d] if d != po.T_STRING else abi.T_STRING, nb)
-                for d, nb in schema]
+    amap = dict(T_NUM)
+    amap[po.T_STRING] = abi.T_STRING
+    amap[po.T_BOOL] = abi.T_BOOL
+    a_schema = [(amap[d], nb) for d, nb in schema]
 
     num_cols = [c for c in range(ncols) if schema[c][0] != po.T_STRING]
     str_cols = [c for c in range(ncols) if schema[c][0] == po.T_STRING]
@@ -90,10 +103,11 @@ def _run_case(eng, seed):
             dels = np.unique(rng.integers(0, n, max(1, n // 20))).astype(np.int32)
             dmask = po.encode_delete(dels, n)
         deltas = None
-        if rng.random() < 0.25 and num_cols:
+        delta_cands = [c for c in num_cols if schema[c][0] != po.T_BOOL]
+        if rng.random() < 0.25 and delta_cands:
             # 2-deep update deltas on one numeric column
             deltas = [(None, None)] * ncols
-            dc = int(rng.choice(num_cols))
+            dc = int(rng.choice(delta_cands))
             d = schema[dc][0]
             for depth in range(1 if rng.random() < 0.5 else 2):
                 upd = np.unique(rng.integers(0, n, max(1, n // 40))).astype(np.int32)
@@ -131,6 +145,8 @@ def _run_case(eng, seed):
             lo, hi = sorted(rng.random(2) * 200 - 100)
         elif d == po.T_INT64:
             lo, hi = sorted(rng.integers(-(1 << 40), 1 << 40, 2).tolist())
+        elif d == po.T_BOOL:
+            lo, hi = 0, int(rng.integers(0, 2))
         else:
             lo, hi = sorted(rng.integers(-4_000, 4_000, 2).tolist())
         if rng.random() < 0.8:
