@@ -229,7 +229,9 @@ def _run_case(eng, seed):
     try:
         q = eng.query(abi.make_plan(table=t, **plan_kw, join=join))
         grows = q.rows()
-    except se.EngineError:
+    except se.EngineError as ex:
+        if os.environ.get("SN_FUZZ_VERBOSE"):
+            print(f"  skip {seed}: {ex}")
         return None                      # declared-unsupported combo: skip
     o_plan = po.make_plan(**plan_kw, join=None if join is None else ojoin)
     is_grouped = bool(group_cols) or (join is not None and join["group"])
