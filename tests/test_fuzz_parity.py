@@ -259,10 +259,11 @@ def _run_case(eng, seed):
 @pytest.mark.gpu
 def test_fuzz_parity_gpu():
     n_cases = int(os.environ.get("SN_FUZZ_N", "30"))
+    base = int(os.environ.get("SN_FUZZ_BASE", "1000"))
     eng = se.Engine(device=0)
     try:
         ran = skipped = 0
-        for seed in range(1000, 1000 + n_cases):
+        for seed in range(base, base + n_cases):
             r = _run_case(eng, seed)
             if r is None:
                 skipped += 1
