@@ -94,16 +94,15 @@ def _run_case(eng, seed):
     ot = po.OracleTable([d for d, _ in schema])
     nbatches = int(rng.integers(1, 4))
     cols_by_batch = []
-    for b in range(nbatches):
-        n = int(rng.integers(1_000, 40_000))
-        raw = [_gen_col(rng, d, n, nb) for d, nb in schema]
-        blobs = [_encode(rng, schema[c][0], raw[c]) for c in range(ncols)]
+    delta_cands = [c for c in num_cols if schema[c][0] != po.T_BOOL]
+
+    def gen_state(rng, n):
+        """One cumulative mutation state: (delete_mask, deltas)."""
         dmask = None
         if rng.random() < 0.3:
             dels = np.unique(rng.integers(0, n, max(1, n // 20))).astype(np.int32)
             dmask = po.encode_delete(dels, n)
         deltas = None
-        delta_cands = [c for c in num_cols if schema[c][0] != po.T_BOOL]
         if rng.random() < 0.25 and delta_cands:
             # 2-deep update deltas on one numeric column
             deltas = [(None, None)] * ncols
@@ -124,12 +123,26 @@ def _run_case(eng, seed):
                 blob = po.encode_delta(d, po.ENC_UNCOMPRESSED, upd, n, dv)
                 d1, d2 = deltas[dc]
                 deltas[dc] = (blob, d1) if depth else (blob, None)
+        return dmask, deltas
+
+    for b in range(nbatches):
+        n = int(rng.integers(1_000, 40_000))
+        raw = [_gen_col(rng, d, n, nb) for d, nb in schema]
+        blobs = [_encode(rng, schema[c][0], raw[c]) for c in range(ncols)]
+        dmask, deltas = gen_state(rng, n)
         if rng.random() < 0.2:
             from tests.test_compression import wrap_lz4
             blobs = [wrap_lz4(bl) if rng.random() < 0.5 else bl
                      for bl in blobs]
         eng.batch_put(t, 100 + b, b, -n if deltas else n, blobs,
                       delete_mask=dmask, deltas=deltas)
+        if rng.random() < 0.25:
+            # mutate-after-put (the UPDATE/DELETE seam): the engine's
+            # cumulative state is REPLACED, so the oracle equivalently
+            # sees only the final state
+            dmask, deltas = gen_state(rng, n)
+            eng.batch_mutate(t, 100 + b, b, delete_mask=dmask,
+                             deltas=deltas)
         ot.add_batch(-n if deltas else n, blobs, delete_mask=dmask,
                      deltas=deltas)
         cols_by_batch.append((n, raw))
