@@ -137,12 +137,14 @@ def _run_case(eng, seed):
         eng.batch_put(t, 100 + b, b, -n if deltas else n, blobs,
                       delete_mask=dmask, deltas=deltas)
         if rng.random() < 0.25:
-            # mutate-after-put (the UPDATE/DELETE seam): the engine's
-            # cumulative state is REPLACED, so the oracle equivalently
-            # sees only the final state
-            dmask, deltas = gen_state(rng, n)
-            eng.batch_mutate(t, 100 + b, b, delete_mask=dmask,
-                             deltas=deltas)
+            # mutate-after-put (the UPDATE/DELETE seam): each PROVIDED
+            # piece replaces the prior cumulative state, absent pieces
+            # persist — so the oracle sees the merged final state
+            dmask2, deltas2 = gen_state(rng, n)
+            eng.batch_mutate(t, 100 + b, b, delete_mask=dmask2,
+                             deltas=deltas2)
+            dmask = dmask2 if dmask2 is not None else dmask
+            deltas = deltas2 if deltas2 is not None else deltas
         ot.add_batch(-n if deltas else n, blobs, delete_mask=dmask,
                      deltas=deltas)
         cols_by_batch.append((n, raw))
