@@ -96,55 +96,91 @@ def _run_case(eng, seed):
     cols_by_batch = []
     delta_cands = [c for c in num_cols if schema[c][0] != po.T_BOOL]
 
-    def gen_state(rng, n):
-        """One cumulative mutation state: (delete_mask, deltas)."""
-        dmask = None
-        if rng.random() < 0.3:
-            dels = np.unique(rng.integers(0, n, max(1, n // 20))).astype(np.int32)
-            dmask = po.encode_delete(dels, n)
-        deltas = None
-        if rng.random() < 0.25 and delta_cands:
-            # 2-deep update deltas on one numeric column
-            deltas = [(None, None)] * ncols
-            dc = int(rng.choice(delta_cands))
+    def gen_dvals(rng, d, k):
+        if d == po.T_DOUBLE:
+            return rng.random(k) * 100
+        if d == po.T_FLOAT:
+            return (rng.random(k) * 10).astype(np.float32)
+        if d == po.T_INT64:
+            return rng.integers(-(1 << 40), 1 << 40, k)
+        if d == po.T_INT32:
+            return rng.integers(-5_000, 5_000, k).astype(np.int32)
+        return rng.integers(-300, 300, k).astype(np.int16)
+
+    def gen_patch_map(rng, n):
+        """{col: {pos: value}} for 0-1 numeric columns."""
+        if rng.random() >= 0.25 or not delta_cands:
+            return {}
+        dc = int(rng.choice(delta_cands))
+        upd = np.unique(rng.integers(0, n, max(1, n // 40))).astype(np.int32)
+        dv = gen_dvals(rng, schema[dc][0], len(upd))
+        return {dc: {int(p): dv[i] for i, p in enumerate(upd)}}
+
+    def patch_blobs(pm, n, rng):
+        """Patch maps -> per-column (delta1, delta2) blob pairs.  Sometimes
+        split one column's cumulative set into a 2-deep pair (delta1 wins)
+        to exercise the d1-over-d2 decode."""
+        if not pm:
+            return None
+        deltas = [(None, None)] * ncols
+        for dc, m in pm.items():
             d = schema[dc][0]
-            for depth in range(1 if rng.random() < 0.5 else 2):
-                upd = np.unique(rng.integers(0, n, max(1, n // 40))).astype(np.int32)
-                if d == po.T_DOUBLE:
-                    dv = rng.random(len(upd)) * 100
-                elif d == po.T_FLOAT:
-                    dv = (rng.random(len(upd)) * 10).astype(np.float32)
-                elif d == po.T_INT64:
-                    dv = rng.integers(-(1 << 40), 1 << 40, len(upd))
-                elif d == po.T_INT32:
-                    dv = rng.integers(-5_000, 5_000, len(upd)).astype(np.int32)
-                else:
-                    dv = rng.integers(-300, 300, len(upd)).astype(np.int16)
-                blob = po.encode_delta(d, po.ENC_UNCOMPRESSED, upd, n, dv)
-                d1, d2 = deltas[dc]
-                deltas[dc] = (blob, d1) if depth else (blob, None)
-        return dmask, deltas
+            pos = np.array(sorted(m), dtype=np.int32)
+            val = np.array([m[p] for p in sorted(m)],
+                           dtype=np.asarray(gen_dvals(rng, d, 1)).dtype)
+            if len(pos) > 3 and rng.random() < 0.5:
+                cut = len(pos) // 2
+                # delta2 = the older half PLUS stale values for some of the
+                # newer half (delta1 overrides them)
+                stale = gen_dvals(rng, d, len(pos) - cut)
+                b2 = po.encode_delta(d, po.ENC_UNCOMPRESSED,
+                                     pos, n,
+                                     np.concatenate([val[:cut].astype(stale.dtype),
+                                                     stale]))
+                b1 = po.encode_delta(d, po.ENC_UNCOMPRESSED, pos[cut:], n,
+                                     val[cut:])
+                deltas[dc] = (b1, b2)
+            else:
+                deltas[dc] = (po.encode_delta(d, po.ENC_UNCOMPRESSED, pos, n,
+                                              val), None)
+        return deltas
 
     for b in range(nbatches):
         n = int(rng.integers(1_000, 40_000))
         raw = [_gen_col(rng, d, n, nb) for d, nb in schema]
         blobs = [_encode(rng, schema[c][0], raw[c]) for c in range(ncols)]
-        dmask, deltas = gen_state(rng, n)
+        dmask = None
+        if rng.random() < 0.3:
+            dels = np.unique(rng.integers(0, n, max(1, n // 20))).astype(np.int32)
+            dmask = po.encode_delete(dels, n)
+        pm = gen_patch_map(rng, n)
+        deltas = patch_blobs(pm, n, rng)
         if rng.random() < 0.2:
             from tests.test_compression import wrap_lz4
             blobs = [wrap_lz4(bl) if rng.random() < 0.5 else bl
                      for bl in blobs]
         eng.batch_put(t, 100 + b, b, -n if deltas else n, blobs,
                       delete_mask=dmask, deltas=deltas)
-        if rng.random() < 0.25:
-            # mutate-after-put (the UPDATE/DELETE seam): each PROVIDED
-            # piece replaces the prior cumulative state, absent pieces
-            # persist — so the oracle sees the merged final state
-            dmask2, deltas2 = gen_state(rng, n)
+        if rng.random() < 0.3:
+            # mutate-after-put (the UPDATE/DELETE seam): per the reference's
+            # cumulative-delta contract, a column's arriving delta is the
+            # UPWARD MERGE of its prior patches with the new update; absent
+            # columns/pieces persist.  The oracle sees the merged end state.
+            dmask2 = None
+            if rng.random() < 0.4:
+                dels = np.unique(rng.integers(0, n, max(1, n // 20))).astype(np.int32)
+                dmask2 = po.encode_delete(dels, n)
+            pm2 = gen_patch_map(rng, n)
+            for dc, m in pm2.items():
+                merged = dict(pm.get(dc, {}))
+                merged.update(m)
+                pm2[dc] = merged
+            d2blobs = patch_blobs(pm2, n, rng)
             eng.batch_mutate(t, 100 + b, b, delete_mask=dmask2,
-                             deltas=deltas2)
+                             deltas=d2blobs)
             dmask = dmask2 if dmask2 is not None else dmask
-            deltas = deltas2 if deltas2 is not None else deltas
+            pm = {**pm, **pm2}
+            deltas = patch_blobs(pm, n, rng)
         ot.add_batch(-n if deltas else n, blobs, delete_mask=dmask,
                      deltas=deltas)
         cols_by_batch.append((n, raw))
