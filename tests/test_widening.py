@@ -443,3 +443,57 @@ def test_mutate_after_raw_ingest(eng):
     assert abs(rows[0][1][0] - exp) <= 1e-9 * max(1.0, exp)
     assert rows[0][1][1] == merged[alive].max()
     assert rows[0][1][2] == float(alive.sum())
+
+
+@pytest.mark.gpu
+def test_duplicate_group_column_rejected(eng):
+    """GROUP BY a, a: one device column slot cannot carry two premultiplied
+    dictionary images — rejected loudly (fuzzer-found silent wrong keys)."""
+    n = 10_000
+    rng = np.random.default_rng(401)
+    m = [b"A" if x else b"B" for x in rng.integers(0, 2, n)]
+    w = np.ones(n)
+    t = eng.table_define("tdupg", [(abi.T_STRING, False), (abi.T_DOUBLE, False)])
+    eng.batch_put(t, 0, 0, n, [po.encode(po.T_STRING, po.ENC_DICT, m),
+                               po.encode(po.T_DOUBLE, po.ENC_UNCOMPRESSED, w)])
+    with pytest.raises(se.EngineError):
+        eng.query(abi.make_plan(table=t, group_cols=[0, 0],
+                                aggs=[("count", [])]))
+
+
+@pytest.mark.gpu
+def test_short_chunk_lds_tail_not_poisoned(eng):
+    """Regression for the fuzzer-found NaN: a keyless MIN/MAX+AVG plan over
+    a table whose last tile is a SHORT chunk (rows % CHUNK != 0), run right
+    after sparse MIN/MAX queries that park ord-ident NaN bit patterns in
+    LDS.  Pre-fix, the slot-predicated fma accumulators computed
+    fma(0, NaN-from-raw-LDS, sum) and silently poisoned the sums."""
+    rng = np.random.default_rng(409)
+    # step 1: sprinkle ord idents across every CU's LDS (sparse mm queries)
+    ks = rng.integers(0, 50_000, 2_000_000).astype(np.int64) * (1 << 30)
+    vs = rng.random(2_000_000)
+    ts = eng.table_define("tpoison", [(abi.T_INT64, False), (abi.T_DOUBLE, False)])
+    eng.ingest_columns(ts, [{"data": ks}, {"data": vs}], len(ks),
+                       batch_rows=500_000)
+    for _ in range(3):
+        eng.query(abi.make_plan(table=ts, group_cols=[0],
+                                aggs=[("min", [(1, 0.0, 1.0)]),
+                                      ("max", [(1, 0.0, 1.0)]),
+                                      ("count", [])])).wait()
+    # step 2: keyless pac plan over short-tailed batches (n % 1024 != 0)
+    n = 33_007
+    c0 = rng.integers(-5_000, 5_000, n).astype(np.int32)
+    c1 = rng.integers(-300, 300, n).astype(np.int16)
+    t = eng.table_define("tshort", [(abi.T_INT32, False), (abi.T_INT16, False)])
+    eng.batch_put(t, 0, 0, n, [po.encode(po.T_INT32, po.ENC_UNCOMPRESSED, c0),
+                               po.encode(po.T_INT16, po.ENC_UNCOMPRESSED, c1)])
+    aggs = [("avg", [(1, 0.0, 2.0), (0, 2.0, 1.0)]),
+            ("max", [(1, -1.0, 1.0)]),
+            ("avg", [(1, 2.0, 0.5)]),
+            ("count", [])]
+    for _ in range(3):
+        rows = eng.query(abi.make_plan(table=t, aggs=aggs)).rows()
+        vals = rows[0][1]
+        assert all(v is not None and np.isfinite(v) for v in vals), vals
+        assert vals[3] == float(n)
+        assert abs(vals[2] - np.mean(2.0 + 0.5 * c1)) <= 1e-9 * 10
