@@ -293,6 +293,12 @@ struct sn_engine {
   int rws_npart = 0;
   int32_t *lz4_err_dev = nullptr; /* device error word for the LZ4 decode */
   std::mutex lz4_mu;              /* serializes decode launch+sync+readback */
+  /* serializes sn_query_submit: queries on one engine share the stream,
+   * the block-partial scratch and the hash/radix workspaces, so two
+   * submits from different threads (even on DIFFERENT tables — t->mu
+   * does not cover this) must not interleave their enqueue+readback
+   * sections.  Ingest does not take this lock. */
+  std::mutex query_mu;
   /* pinned staging for big blob uploads: pageable hipMemcpy bounces through
    * the runtime's staging path at ~1-2 GB/s with a device sync per call
    * (measured: SF=10 ingest spent ~2 s there, 27 Mrows/s).  A pool of
@@ -1910,6 +1916,8 @@ extern "C" sn_query *sn_query_submit(sn_engine *e, const sn_plan *plan) {
   Table *t = get_table(e, plan->table);
   if (!t) { fail(SN_ERR_BADARG, "unknown table %d", plan->table); return nullptr; }
   if (!e->has_gpu) { fail(SN_ERR_NOGPU, "no HIP device — the engine never falls back to CPU"); return nullptr; }
+  /* one submit at a time per engine (shared stream/scratch/workspaces) */
+  std::lock_guard<std::mutex> qg(e->query_mu);
   if (plan->npreds > SN_MAX_PREDS || plan->naggs > SN_MAX_AGGS ||
       plan->ngroup > SN_MAX_GROUPS || plan->naggs <= 0) {
     fail(SN_ERR_BADARG, "plan limits exceeded"); return nullptr;

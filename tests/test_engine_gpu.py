@@ -888,3 +888,53 @@ def test_concurrent_raw_ingest_and_scan(eng):
         aggs=[("count", [])]))
     res = hi.result()
     assert res.batches_skipped == res.batches_seen  # all excluded by bounds
+
+
+@pytest.mark.gpu
+def test_concurrent_submits_different_tables():
+    """Two threads submitting queries on DIFFERENT tables of one engine:
+    the per-engine submit lock keeps the shared stream/scratch/hash
+    workspaces from interleaving (pre-fix, sparse workspaces and the
+    block-partial scratch could be clobbered between a scan and its
+    reduce)."""
+    import threading
+    e = se.Engine(device=0)
+    try:
+        rng = np.random.default_rng(431)
+        tables, expect = [], []
+        for i in range(2):
+            n = 400_000
+            keys = rng.integers(0, 30_000, n).astype(np.int64) * (1 << 25) + i
+            w = rng.random(n)
+            t = e.table_define(f"tconc{i}", [(abi.T_INT64, False),
+                                             (abi.T_DOUBLE, False)])
+            e.ingest_columns(t, [{"data": keys}, {"data": w}], n,
+                             batch_rows=100_000)
+            tables.append(t)
+            expect.append((len(np.unique(keys)), float(w.sum())))
+        errs = []
+
+        def worker(i):
+            try:
+                for _ in range(6):
+                    q = e.query(abi.make_plan(
+                        table=tables[i], group_cols=[0],
+                        aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])]))
+                    q.wait()
+                    ng = q.num_groups()
+                    assert ng == expect[i][0], (i, ng, expect[i][0])
+                    qa = e.query(abi.make_plan(
+                        table=tables[i], aggs=[("sum", [(1, 0.0, 1.0)])]))
+                    s = qa.rows()[0][1][0]
+                    assert abs(s - expect[i][1]) <= 1e-6 * expect[i][1], (i, s)
+            except Exception as ex:   # noqa: BLE001
+                errs.append((i, repr(ex)))
+
+        th = [threading.Thread(target=worker, args=(i,)) for i in range(2)]
+        for x in th:
+            x.start()
+        for x in th:
+            x.join()
+        assert not errs, errs
+    finally:
+        e.close()
