@@ -240,7 +240,13 @@ def test_gpu_lz4_device_decode_stress():
                        aggs=[("sum", [(1, 0.0, 1.0)]), ("count", [])])
         r1 = eng.query(abi.make_plan(table=t1, **plan_kw)).rows()
         r2 = eng.query(abi.make_plan(table=t2, **plan_kw)).rows()
-        assert r1 == r2 and len(r1) == 3
+        # two separate runs: counts/keys exact; double sums are subject to
+        # the ~1e-13 run-to-run atomicAdd-order wobble DESIGN documents
+        # (an exact == here flaked once in a full-suite run)
+        assert len(r1) == len(r2) == 3
+        for (k1, v1), (k2, v2) in zip(r1, r2):
+            assert k1 == k2 and v1[1] == v2[1]
+            assert abs(v1[0] - v2[0]) <= 1e-12 * max(1.0, abs(v1[0]))
         m = i32 <= 2
         assert r1[0][1][1] + r1[1][1][1] + r1[2][1][1] == float(m.sum())
     finally:
