@@ -193,7 +193,10 @@ typedef struct {
 
 #define SN_JOIN_NONE  -1
 #define SN_JOIN_SEMI   0   /* fact row survives iff key present in dim */
-#define SN_JOIN_GROUP  1   /* additionally GROUP BY the dim attribute */
+#define SN_JOIN_GROUP  1   /* additionally GROUP BY the dim attribute;
+                              may combine with ONE fact group column
+                              (group_cols[0]) — results then carry
+                              (attr, fact key) composite keys */
 
 typedef struct {
   int32_t table;        /* handle from sn_table_define */
